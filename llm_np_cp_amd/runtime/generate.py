@@ -1,0 +1,159 @@
+"""Autoregressive generation loop with streaming output.
+
+Reference parity: ``generate(prompt, tokenizer, model, max_tokens,
+kv_cache, config) -> str`` with per-token streaming prints
+(``/root/reference/llama3.2_model.py:865-902``).  Deliberate fix: later
+steps feed back the sampled token *id* — the reference re-tokenized the
+decoded text of the previous token (``llama3.2_model.py:874-878``), which
+corrupts the stream for any token whose text does not round-trip.
+
+Works with both the NumPy oracle model and the GPU engine: a model here
+is anything with ``.forward(ids, cache, pos0) -> logits`` and
+``.make_cache(max_seq)``.  Cache-less mode (``use_cache=False``) re-runs
+the full prefix each step — a correctness/debug mode, as in the reference
+(SURVEY §3.3).
+"""
+
+from __future__ import annotations
+
+import sys
+import time
+from dataclasses import dataclass, field
+from typing import Callable, List, Optional
+
+import numpy as np
+
+from .sampling import SamplingParams, sample_token
+
+
+@dataclass
+class GenerateResult:
+    text: str
+    token_ids: List[int]
+    prefill_time_s: float = 0.0
+    decode_time_s: float = 0.0
+
+    @property
+    def decode_tokens_per_s(self) -> float:
+        n = max(len(self.token_ids) - 0, 0)
+        return n / self.decode_time_s if self.decode_time_s > 0 else 0.0
+
+
+def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
+             kv_cache=None, params: Optional[SamplingParams] = None,
+             use_cache: bool = True, stream: bool = True,
+             stop_on_eos: bool = True,
+             on_token: Optional[Callable[[str], None]] = None) -> GenerateResult:
+    """Generate up to max_tokens continuation tokens of prompt."""
+    params = params or SamplingParams()
+    rng = np.random.default_rng(params.seed)
+    prompt_ids = list(tokenizer.encode(prompt))
+    eos = getattr(model.config, "eos_token_id", None)
+
+    if kv_cache is None:
+        kv_cache = model.make_cache(len(prompt_ids) + max_tokens + 1)
+
+    emit = on_token
+    if emit is None and stream:
+        emit = lambda s: (sys.stdout.write(s), sys.stdout.flush())
+
+    out_ids: List[int] = []
+    t0 = time.perf_counter()
+    logits = model.forward(np.asarray(prompt_ids, dtype=np.int64), kv_cache, 0)
+    t_prefill = time.perf_counter() - t0
+
+    t1 = time.perf_counter()
+    for _ in range(max_tokens):
+        next_id = sample_token(np.asarray(logits[-1], dtype=np.float32),
+                               params, rng)
+        out_ids.append(next_id)
+        if emit:
+            emit(tokenizer.decode([next_id]))
+        if stop_on_eos and eos is not None and next_id == eos:
+            break
+        if use_cache:
+            logits = model.forward(np.asarray([next_id], dtype=np.int64),
+                                   kv_cache, kv_cache.seq_len)
+        else:
+            full = prompt_ids + out_ids
+            fresh = model.make_cache(len(full) + 1)
+            logits = model.forward(np.asarray(full, dtype=np.int64), fresh, 0)
+    t_decode = time.perf_counter() - t1
+
+    return GenerateResult(
+        text=tokenizer.decode(out_ids),
+        token_ids=out_ids,
+        prefill_time_s=t_prefill,
+        decode_time_s=t_decode,
+    )
+
+
+class ByteTokenizer:
+    """Self-contained fallback tokenizer (byte-level) for environments
+    without HF tokenizer files; real checkpoints use ``AutoTokenizer``."""
+
+    vocab_size = 256
+
+    def encode(self, text: str) -> List[int]:
+        return list(text.encode("utf-8"))
+
+    def decode(self, ids) -> str:
+        return bytes(int(i) % 256 for i in ids).decode("utf-8", errors="replace")
+
+
+def load_model(model_dir_or_preset: str, backend: str = "auto",
+               device: str = "cuda", dtype: str = "bf16",
+               max_seq: int = 4096, seed: int = 0):
+    """Reference-parity entry (``load_model`` -> (tokenizer, model, config),
+    ``llama3.2_model.py:1082-1099``), local-only.
+
+    ``model_dir_or_preset``: a checkpoint directory (config.json +
+    safetensors) or a preset name (random-init synthetic weights).
+    ``backend``: "numpy" (CPU oracle), "gpu" (HIP engine), or "auto".
+    """
+    import os
+
+    from ..core.config import preset_config, PRESETS
+    from ..io.loader import load_config, load_weights_numpy, random_weights
+
+    if os.path.isdir(model_dir_or_preset):
+        config = load_config(model_dir_or_preset)
+        weights = load_weights_numpy(model_dir_or_preset)
+        tok_dir = model_dir_or_preset
+    elif model_dir_or_preset.lower() in PRESETS:
+        config = preset_config(model_dir_or_preset)
+        weights = random_weights(config, seed=seed)
+        tok_dir = None
+    else:
+        raise FileNotFoundError(
+            f"{model_dir_or_preset!r} is neither a directory nor a preset")
+
+    tokenizer = None
+    if tok_dir is not None:
+        try:
+            from transformers import AutoTokenizer
+            tokenizer = AutoTokenizer.from_pretrained(tok_dir)
+        except Exception:
+            tokenizer = None
+    if tokenizer is None:
+        tokenizer = ByteTokenizer()
+
+    if backend == "auto":
+        try:
+            import torch
+            backend = "gpu" if torch.cuda.is_available() else "numpy"
+        except ImportError:
+            backend = "numpy"
+
+    if backend == "numpy":
+        from ..models.numpy_ref import NumpyModel, NumpyKVCache
+
+        model = NumpyModel(config, weights)
+        model.make_cache = lambda n: NumpyKVCache(config, n)
+    elif backend == "gpu":
+        from ..models.engine import GPUModel
+
+        model = GPUModel(config, weights, dtype=dtype, max_seq=max_seq)
+    else:
+        raise ValueError(f"unknown backend {backend!r}")
+    return tokenizer, model, config

@@ -1,0 +1,138 @@
+"""NumPy reference model vs HF transformers on tiny random-init configs.
+
+This is the oracle-validation layer: the reference validated
+itself interactively against ``transformers``; we make that comparison an
+automated test, then every later (HIP) component is tested against the
+NumPy model.
+"""
+
+import numpy as np
+import pytest
+import torch
+
+from llm_np_cp_amd.core.config import preset_config
+from llm_np_cp_amd.models.numpy_ref import NumpyKVCache, NumpyModel
+
+
+def hf_llama(cfg):
+    from transformers import LlamaConfig, LlamaForCausalLM
+
+    hf_cfg = LlamaConfig(
+        vocab_size=cfg.vocab_size, hidden_size=cfg.hidden_size,
+        intermediate_size=cfg.intermediate_size,
+        num_hidden_layers=cfg.num_hidden_layers,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        head_dim=cfg.head_dim, rms_norm_eps=cfg.rms_norm_eps,
+        rope_theta=cfg.rope_theta,
+        max_position_embeddings=cfg.max_position_embeddings,
+        tie_word_embeddings=cfg.tie_word_embeddings,
+        attn_implementation="eager",
+    )
+    torch.manual_seed(7)
+    return LlamaForCausalLM(hf_cfg).eval()
+
+
+def hf_gemma2(cfg):
+    from transformers import Gemma2Config, Gemma2ForCausalLM
+
+    hf_cfg = Gemma2Config(
+        vocab_size=cfg.vocab_size, hidden_size=cfg.hidden_size,
+        intermediate_size=cfg.intermediate_size,
+        num_hidden_layers=cfg.num_hidden_layers,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        head_dim=cfg.head_dim, rms_norm_eps=cfg.rms_norm_eps,
+        rope_theta=cfg.rope_theta,
+        query_pre_attn_scalar=cfg.query_pre_attn_scalar,
+        sliding_window=cfg.sliding_window,
+        attn_logit_softcapping=cfg.attn_logit_softcapping,
+        final_logit_softcapping=cfg.final_logit_softcapping,
+        tie_word_embeddings=True,
+        attn_implementation="eager",
+    )
+    torch.manual_seed(9)
+    return Gemma2ForCausalLM(hf_cfg).eval()
+
+
+def np_weights_from_hf(hf_model):
+    return {k: v.detach().to(torch.float32).numpy()
+            for k, v in hf_model.state_dict().items()}
+
+
+@pytest.mark.parametrize("preset,builder", [
+    ("tiny-llama", hf_llama),
+    ("tiny-gemma2", hf_gemma2),
+])
+def test_forward_matches_transformers(preset, builder):
+    cfg = preset_config(preset)
+    hf = builder(cfg)
+    model = NumpyModel(cfg, np_weights_from_hf(hf))
+
+    rng = np.random.default_rng(0)
+    ids = rng.integers(0, cfg.vocab_size, size=12)
+    with torch.no_grad():
+        ref = hf(torch.tensor(ids[None])).logits[0].numpy()
+
+    cache = NumpyKVCache(cfg, 64)
+    got = model.forward(ids, cache, 0)
+
+    np.testing.assert_allclose(got, ref, rtol=2e-4, atol=2e-4)
+
+
+@pytest.mark.parametrize("preset,builder", [
+    ("tiny-llama", hf_llama),
+    ("tiny-gemma2", hf_gemma2),
+])
+def test_incremental_decode_matches_prefill(preset, builder):
+    """KV-cache decode path == full-prefill logits (cache correctness)."""
+    cfg = preset_config(preset)
+    hf = builder(cfg)
+    model = NumpyModel(cfg, np_weights_from_hf(hf))
+
+    rng = np.random.default_rng(1)
+    ids = rng.integers(0, cfg.vocab_size, size=10)
+
+    cache_full = NumpyKVCache(cfg, 64)
+    full = model.forward(ids, cache_full, 0)
+
+    cache_inc = NumpyKVCache(cfg, 64)
+    model.forward(ids[:4], cache_inc, 0)
+    out = None
+    for t in range(4, 10):
+        out = model.forward(ids[t:t + 1], cache_inc, t)
+    np.testing.assert_allclose(out[0], full[-1], rtol=1e-4, atol=1e-4)
+
+
+def test_sliding_window_limits_attention():
+    """Tokens beyond the sliding window must not influence sliding layers.
+
+    Uses a 1-layer sliding-only Gemma config: logits at the last position
+    must be identical whether the far-past tokens differ, once they fall
+    outside the window."""
+    from llm_np_cp_amd.core.config import ModelConfig
+    from llm_np_cp_amd.io.loader import random_weights
+
+    cfg = ModelConfig(
+        model_type="gemma2", vocab_size=128, hidden_size=32,
+        intermediate_size=64, num_hidden_layers=1, num_attention_heads=2,
+        num_key_value_heads=1, head_dim=16, rope_theta=10000.0,
+        hidden_act="gelu_pytorch_tanh", query_pre_attn_scalar=16,
+        sliding_window=4, attn_logit_softcapping=50.0,
+        final_logit_softcapping=30.0,
+        layer_types=["sliding_attention"],
+    )
+    w = random_weights(cfg, seed=3)
+    model = NumpyModel(cfg, w)
+
+    rng = np.random.default_rng(4)
+    tail = rng.integers(0, 128, size=4)
+    a = np.concatenate([rng.integers(0, 128, size=6), tail])
+    b = np.concatenate([rng.integers(0, 128, size=6), tail])
+
+    la = model.forward(a, NumpyKVCache(cfg, 32), 0)
+    lb = model.forward(b, NumpyKVCache(cfg, 32), 0)
+    # last query position attends only to the last 4 keys -> identical
+    np.testing.assert_allclose(la[-1], lb[-1], rtol=1e-6, atol=1e-6)
+    # sanity: an in-window difference does change logits
+    assert not np.allclose(la[-2], lb[-2])
