@@ -1,0 +1,123 @@
+#!/usr/bin/env python3
+"""Flagship decode benchmark (driver contract).
+
+Measures the BASELINE.json metric: decode tokens/sec, batch 1, synthetic
+prompt, random-init weights — Llama-3.2-1B TP=1 (N=1) or Gemma-2-9B TP=N
+(N>1), bf16.  A "step" is one decoded token; the timed region is K
+hipGraph-replayed decode steps (full forward incl. lm_head + sampling).
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--model M]
+Launched by the driver for N>1 via torch.distributed.run (one rank/GPU).
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=200)
+    ap.add_argument("--warmup", type=int, default=20)
+    ap.add_argument("--model", type=str, default=None,
+                    help="preset override (default: llama-3.2-1b at N=1, "
+                         "gemma-2-9b at N>1)")
+    ap.add_argument("--prompt-len", type=int, default=64)
+    ap.add_argument("--max-seq", type=int, default=None)
+    ap.add_argument("--no-graph", action="store_true")
+    args = ap.parse_args()
+
+    import numpy as np
+    import torch
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    n_gpus = max(world, args.gpus)
+    model_name = args.model or ("llama-3.2-1b" if n_gpus == 1 else "gemma-2-9b")
+
+    from csrc.build import ensure_built
+    ensure_built()
+
+    import llm_np_cp_amd as L
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.engine import GPUModel
+    from llm_np_cp_amd.parallel.tp import init_distributed
+
+    rank, world = init_distributed()
+    cfg = L.preset_config(model_name)
+    max_seq = args.max_seq or min(
+        cfg.max_position_embeddings,
+        args.prompt_len + args.steps + args.warmup + 64)
+    w = random_weights(cfg, seed=0)
+    model = GPUModel(cfg, w, max_seq=max_seq, seed=0)
+
+    rng = np.random.default_rng(0)
+    prompt = rng.integers(0, cfg.vocab_size, size=args.prompt_len)
+
+    def barrier():
+        if world > 1:
+            torch.distributed.barrier()
+
+    # prefill + warmup
+    model.prefill(prompt)
+    use_graph = not args.no_graph
+    if args.warmup > 0:
+        model.decode(args.warmup, greedy=True, use_graph=use_graph,
+                     first_from_logits=True)
+
+    if use_graph:
+        model.capture_decode_graph(True, 0.1)  # no-op if warmup captured
+
+    # timed region: exactly K decode steps
+    barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    if use_graph:
+        for _ in range(args.steps):
+            model._graph.replay()
+    else:
+        for _ in range(args.steps):
+            model._decode_step(True, 0.1)
+    torch.cuda.synchronize()
+    barrier()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    if world > 1:
+        t = torch.tensor([elapsed], device=model.device)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    toks_per_s = args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+
+    if rank == 0:
+        out = {
+            "metric": "decode tokens/sec",
+            "value": toks_per_s,
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": model_name,
+                "global_batch": 1,
+                "seq_len": args.prompt_len + args.warmup + args.steps,
+                "parallelism": f"tp{world}",
+            },
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,684 @@
+// llm_ops.hip — hand-written CDNA4 (gfx950 / MI355X) kernels for the
+// Llama-3.2 / Gemma-2 decode + prefill forward pass.
+//
+// Design notes (MI355X-first, per /opt/skills/guides/cdna_hip_programming.md):
+//  - wave64 everywhere; block sizes are multiples of 64
+//  - bf16 I/O, fp32 accumulation; all bf16 traffic vectorized 16 B/lane
+//  - decode is weight-bandwidth-bound: the GEMV kernel streams W rows with
+//    deep unrolled dwordx4 loads straight to VGPRs (no LDS round trip for
+//    the streamed operand — guide §5 "GEMV / M<=16 decode weights")
+//  - attention never materializes QK^T: online softmax in registers
+//  - KV cache is a preallocated pool, written in place by the RoPE kernel
+//    (replaces the reference's O(T^2) concat cache, llama3.2_model.py:325)
+//  - every kernel is hipGraph-capture-safe (no mallocs/syncs; sequence
+//    position and sampled token ids live in device memory)
+//
+// Replaces (reference parity):
+//  K1 softmax RawKernel      -> folded into k_attn online softmax
+//  K2 cuBLAS GEMM/GEMV       -> k_gemv_bf16 (decode), k_gemm_bf16 (prefill)
+//  K3 elementwise ufuncs     -> k_rmsnorm (+resid), k_rope_cache, k_glu
+//  K4 KV concat              -> in-place pool write in k_rope_cache
+//  K5 repeat_kv              -> k_attn indexes kv_head = q_head / groups
+//  K7 torch.multinomial      -> k_sample (min-p via Gumbel argmax)
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include <cstdio>
+
+typedef unsigned short u16;
+typedef short s8v __attribute__((ext_vector_type(8)));    // 16 B of bf16
+typedef float f4v __attribute__((ext_vector_type(4)));
+
+#define DEVINL __device__ __forceinline__
+
+DEVINL float b2f(u16 u) {
+  union { float f; uint32_t i; } c;
+  c.i = ((uint32_t)u) << 16;
+  return c.f;
+}
+
+DEVINL u16 f2b(float f) {
+  union { float f; uint32_t i; } c;
+  c.f = f;
+  uint32_t u = c.i;
+  if ((u & 0x7fffffffu) > 0x7f800000u) return (u16)0x7fc0;  // NaN
+  u += 0x7fffu + ((u >> 16) & 1u);  // round to nearest even
+  return (u16)(u >> 16);
+}
+
+DEVINL float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int m = 1; m < 64; m <<= 1) v += __shfl_xor(v, m);
+  return v;
+}
+
+// ====================================================================
+// GEMV: y[N] = W[N,K] @ x[K]  (+ res)  (optional softcap, f32 out)
+// W bf16 row-major; one wave per output row; x staged once into LDS.
+// Decode is bound by streaming W once: target ~HBM roofline.
+// ====================================================================
+
+#define GEMV_ROWS_PER_BLOCK 4
+
+extern "C" __global__ void __launch_bounds__(256)
+k_gemv_bf16(const u16* __restrict__ W, const u16* __restrict__ x,
+            void* __restrict__ y, const u16* __restrict__ res,
+            int N, int K, int out_f32, float softcap) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  u16* xs = (u16*)smem;
+  for (int i = threadIdx.x * 8; i < K; i += 256 * 8)
+    *(s8v*)(xs + i) = *(const s8v*)(x + i);
+  __syncthreads();
+
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int row = blockIdx.x * GEMV_ROWS_PER_BLOCK + wave;
+  if (row >= N) return;
+  const u16* Wr = W + (size_t)row * K;
+
+  // 4 independent accumulators; 4 W loads in flight per iteration.
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  int k = lane * 8;
+  for (; k + 1536 < K; k += 2048) {
+    s8v w0 = *(const s8v*)(Wr + k);
+    s8v w1 = *(const s8v*)(Wr + k + 512);
+    s8v w2 = *(const s8v*)(Wr + k + 1024);
+    s8v w3 = *(const s8v*)(Wr + k + 1536);
+    s8v x0 = *(const s8v*)(xs + k);
+    s8v x1 = *(const s8v*)(xs + k + 512);
+    s8v x2 = *(const s8v*)(xs + k + 1024);
+    s8v x3 = *(const s8v*)(xs + k + 1536);
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      a0 += b2f(((u16*)&w0)[j]) * b2f(((u16*)&x0)[j]);
+      a1 += b2f(((u16*)&w1)[j]) * b2f(((u16*)&x1)[j]);
+      a2 += b2f(((u16*)&w2)[j]) * b2f(((u16*)&x2)[j]);
+      a3 += b2f(((u16*)&w3)[j]) * b2f(((u16*)&x3)[j]);
+    }
+  }
+  for (; k < K; k += 512) {
+    s8v w0 = *(const s8v*)(Wr + k);
+    s8v x0 = *(const s8v*)(xs + k);
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      a0 += b2f(((u16*)&w0)[j]) * b2f(((u16*)&x0)[j]);
+  }
+  float acc = wave_reduce_sum((a0 + a1) + (a2 + a3));
+  if (lane == 0) {
+    if (softcap > 0.f) acc = softcap * tanhf(acc / softcap);
+    if (res) acc += b2f(res[row]);
+    if (out_f32) ((float*)y)[row] = acc;
+    else ((u16*)y)[row] = f2b(acc);
+  }
+}
+
+extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x, void* y,
+                                       const void* res, int N, int K,
+                                       int out_f32, float softcap,
+                                       hipStream_t stream) {
+  int blocks = (N + GEMV_ROWS_PER_BLOCK - 1) / GEMV_ROWS_PER_BLOCK;
+  size_t lds = (size_t)K * 2;
+  hipLaunchKernelGGL(k_gemv_bf16, dim3(blocks), dim3(256), lds, stream,
+                     (const u16*)W, (const u16*)x, y, (const u16*)res, N, K,
+                     out_f32, softcap);
+  return hipGetLastError();
+}
+
+// ====================================================================
+// RMSNorm: mode 0: y = norm(x)*g ; mode 1: y = res + norm(x)*g
+// x bf16 [M,H]; g fp32[H] (Gemma's gamma+1 pre-folded on host); one block
+// per row; fused single pass (x kept in registers between reduce+scale).
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_rmsnorm(const u16* __restrict__ x, const float* __restrict__ g,
+          const u16* __restrict__ res, u16* __restrict__ y,
+          int H, float eps, int mode) {
+  __shared__ float warp_sums[8];
+  const int row = blockIdx.x;
+  const u16* xr = x + (size_t)row * H;
+  u16* yr = y + (size_t)row * H;
+  const u16* rr = res ? res + (size_t)row * H : nullptr;
+
+  s8v buf[4];  // up to 8192 elems per row at 256 threads * 8/chunk
+  int nchunk = 0;
+  float ss = 0.f;
+  for (int i = threadIdx.x * 8; i < H; i += 2048) {
+    s8v v = *(const s8v*)(xr + i);
+    buf[nchunk++] = v;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float f = b2f(((u16*)&v)[j]);
+      ss += f * f;
+    }
+  }
+  ss = wave_reduce_sum(ss);
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  if (lane == 0) warp_sums[wave] = ss;
+  __syncthreads();
+  float tot = warp_sums[0] + warp_sums[1] + warp_sums[2] + warp_sums[3];
+  const float rnorm = rsqrtf(tot / (float)H + eps);
+
+  nchunk = 0;
+  for (int i = threadIdx.x * 8; i < H; i += 2048) {
+    s8v v = buf[nchunk++];
+    u16 o[8];
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float f = b2f(((u16*)&v)[j]) * rnorm * g[i + j];
+      if (mode == 1) f += b2f(rr[i + j]);
+      o[j] = f2b(f);
+    }
+    *(s8v*)(yr + i) = *(s8v*)o;
+  }
+}
+
+extern "C" hipError_t launch_rmsnorm(const void* x, const void* g,
+                                     const void* res, void* y, int M, int H,
+                                     float eps, int mode, hipStream_t stream) {
+  hipLaunchKernelGGL(k_rmsnorm, dim3(M), dim3(256), 0, stream, (const u16*)x,
+                     (const float*)g, (const u16*)res, (u16*)y, H, eps, mode);
+  return hipGetLastError();
+}
+
+// ====================================================================
+// RoPE + KV-pool write.
+// q inout [M, nh*hd]; k_in/v_in [M, kvh*hd]; caches [kvh, S, hd].
+// Position of row m is *pos_ptr + m (device scalar: graph-replay-safe).
+// cos/sin tables fp32 [max_seq, hd/2].
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_rope_cache(u16* __restrict__ q, const u16* __restrict__ kin,
+             const u16* __restrict__ vin, u16* __restrict__ kc,
+             u16* __restrict__ vc, const float* __restrict__ cost,
+             const float* __restrict__ sint, const int* __restrict__ pos_ptr,
+             int nh, int kvh, int hd, int S) {
+  const int m = blockIdx.x;
+  const int pos = *pos_ptr + m;
+  const int hd2 = hd / 2;
+  const float* cp = cost + (size_t)pos * hd2;
+  const float* sp = sint + (size_t)pos * hd2;
+
+  // q rotation in place
+  u16* qr = q + (size_t)m * nh * hd;
+  for (int idx = threadIdx.x; idx < nh * hd2; idx += 256) {
+    int h = idx / hd2, i = idx % hd2;
+    float x1 = b2f(qr[h * hd + i]);
+    float x2 = b2f(qr[h * hd + i + hd2]);
+    float c = cp[i], s = sp[i];
+    qr[h * hd + i] = f2b(x1 * c - x2 * s);
+    qr[h * hd + i + hd2] = f2b(x2 * c + x1 * s);
+  }
+  // k rotation -> cache; v copy -> cache
+  const u16* kr = kin + (size_t)m * kvh * hd;
+  const u16* vr = vin + (size_t)m * kvh * hd;
+  for (int idx = threadIdx.x; idx < kvh * hd2; idx += 256) {
+    int h = idx / hd2, i = idx % hd2;
+    float x1 = b2f(kr[h * hd + i]);
+    float x2 = b2f(kr[h * hd + i + hd2]);
+    float c = cp[i], s = sp[i];
+    u16* dst = kc + ((size_t)h * S + pos) * hd;
+    dst[i] = f2b(x1 * c - x2 * s);
+    dst[i + hd2] = f2b(x2 * c + x1 * s);
+  }
+  for (int idx = threadIdx.x; idx < kvh * hd; idx += 256) {
+    int h = idx / hd, i = idx % hd;
+    vc[((size_t)h * S + pos) * hd + i] = vr[h * hd + i];
+  }
+}
+
+extern "C" hipError_t launch_rope_cache(void* q, const void* kin,
+                                        const void* vin, void* kc, void* vc,
+                                        const void* cost, const void* sint,
+                                        const void* pos_ptr, int M, int nh,
+                                        int kvh, int hd, int S,
+                                        hipStream_t stream) {
+  hipLaunchKernelGGL(k_rope_cache, dim3(M), dim3(256), 0, stream, (u16*)q,
+                     (const u16*)kin, (const u16*)vin, (u16*)kc, (u16*)vc,
+                     (const float*)cost, (const float*)sint,
+                     (const int*)pos_ptr, nh, kvh, hd, S);
+  return hipGetLastError();
+}
+
+// ====================================================================
+// GQA attention with online softmax (decode + per-query prefill).
+// grid (nh, M): block = head h, query row m (absolute pos = *len_ptr + m,
+// attends keys [start, pos+1)).  4 waves split the KV range; each lane
+// owns (sub-position, 8-dim chunk); merge within wave then across waves.
+// Sliding window + attention-logit softcap = real Gemma-2 semantics
+// (reference omitted both).
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_attn(const u16* __restrict__ q, const u16* __restrict__ kc,
+       const u16* __restrict__ vc, u16* __restrict__ out,
+       const int* __restrict__ len_ptr, int nh, int kvh, int hd, int S,
+       float scale, float softcap, int window) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* red = (float*)smem;  // [4][hd] acc + [4][2] m,l
+
+  const int h = blockIdx.x, m = blockIdx.y;
+  const int kvhead = h / (nh / kvh);
+  const int pos = *len_ptr + m;
+  const int T = pos + 1;
+  int start = 0;
+  if (window > 0 && T - window > 0) start = T - window;
+
+  const int LP = hd / 8;        // lanes per position
+  const int PP = 64 / LP;       // positions per wave-iteration
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int p = lane / LP, d0 = (lane % LP) * 8;
+
+  // q fragment for this head (8 dims), fp32
+  float qf[8];
+  {
+    s8v v = *(const s8v*)(q + ((size_t)m * nh + h) * hd + d0);
+#pragma unroll
+    for (int j = 0; j < 8; j++) qf[j] = b2f(((u16*)&v)[j]);
+  }
+
+  const u16* K0 = kc + (size_t)kvhead * S * hd;
+  const u16* V0 = vc + (size_t)kvhead * S * hd;
+
+  float mrun = -INFINITY, lrun = 0.f, acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; j++) acc[j] = 0.f;
+
+  for (int t0 = start + wave * PP; t0 < T; t0 += 4 * PP) {
+    int t = t0 + p;
+    bool valid = t < T;
+    int tl = valid ? t : (T - 1);
+    // K row: 16 B per lane; consecutive lanes -> consecutive addresses
+    s8v kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
+    float partial = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; j++) partial += qf[j] * b2f(((u16*)&kv)[j]);
+    // reduce across the LP lanes of this position
+    for (int w = 1; w < LP; w <<= 1) partial += __shfl_xor(partial, w);
+    float score = partial * scale;
+    if (softcap > 0.f) score = softcap * tanhf(score / softcap);
+    if (!valid) score = -INFINITY;
+
+    float mnew = fmaxf(mrun, score);
+    if (mnew == -INFINITY) continue;  // whole sub-iteration invalid
+    float alpha = __expf(mrun - mnew);  // exp(-inf)=0 on first hit
+    float pv = __expf(score - mnew);
+    lrun = lrun * alpha + pv;
+    s8v vv = *(const s8v*)(V0 + (size_t)tl * hd + d0);
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      acc[j] = acc[j] * alpha + pv * b2f(((u16*)&vv)[j]);
+    mrun = mnew;
+  }
+
+  // merge across position-groups within the wave (lanes l, l^LP, l^2LP, ...)
+  for (int w = LP; w < 64; w <<= 1) {
+    float mo = __shfl_xor(mrun, w);
+    float lo = __shfl_xor(lrun, w);
+    float mn = fmaxf(mrun, mo);
+    float sa = (mrun == -INFINITY && mo == -INFINITY) ? 0.f : __expf(mrun - mn);
+    float sb = (mrun == -INFINITY && mo == -INFINITY) ? 0.f : __expf(mo - mn);
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float ao = __shfl_xor(acc[j], w);
+      acc[j] = acc[j] * sa + ao * sb;
+    }
+    lrun = lrun * sa + lo * sb;
+    mrun = mn;
+  }
+
+  // merge across the 4 waves via LDS
+  float* accs = red;             // [4][hd]
+  float* mls = red + 4 * hd;     // [4][2]
+  if (lane < LP) {
+#pragma unroll
+    for (int j = 0; j < 8; j++) accs[wave * hd + d0 + j] = acc[j];
+  }
+  if (lane == 0) {
+    mls[wave * 2] = mrun;
+    mls[wave * 2 + 1] = lrun;
+  }
+  __syncthreads();
+  if (wave == 0) {
+    float mt = fmaxf(fmaxf(mls[0], mls[2]), fmaxf(mls[4], mls[6]));
+    float lt = 0.f, sc[4];
+#pragma unroll
+    for (int w = 0; w < 4; w++) {
+      sc[w] = (mls[w * 2] == -INFINITY) ? 0.f : __expf(mls[w * 2] - mt);
+      lt += mls[w * 2 + 1] * sc[w];
+    }
+    float inv = 1.f / lt;
+    for (int d = lane; d < hd; d += 64) {
+      float v = 0.f;
+#pragma unroll
+      for (int w = 0; w < 4; w++) v += accs[w * hd + d] * sc[w];
+      out[((size_t)m * nh + h) * hd + d] = f2b(v * inv);
+    }
+  }
+}
+
+extern "C" hipError_t launch_attn(const void* q, const void* kc,
+                                  const void* vc, void* out,
+                                  const void* len_ptr, int M, int nh, int kvh,
+                                  int hd, int S, float scale, float softcap,
+                                  int window, hipStream_t stream) {
+  size_t lds = (4 * hd + 8) * sizeof(float);
+  hipLaunchKernelGGL(k_attn, dim3(nh, M), dim3(256), lds, stream,
+                     (const u16*)q, (const u16*)kc, (const u16*)vc, (u16*)out,
+                     (const int*)len_ptr, nh, kvh, hd, S, scale, softcap,
+                     window);
+  return hipGetLastError();
+}
+
+// ====================================================================
+// GLU activations: out = act(gate) * up ; act 0 = SiLU, 1 = tanh-GELU
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_glu(const u16* __restrict__ gate, const u16* __restrict__ up,
+      u16* __restrict__ out, long total, int act) {
+  long i = ((long)blockIdx.x * 256 + threadIdx.x) * 8;
+  if (i >= total) return;
+  s8v g = *(const s8v*)(gate + i);
+  s8v u = *(const s8v*)(up + i);
+  u16 o[8];
+#pragma unroll
+  for (int j = 0; j < 8; j++) {
+    float x = b2f(((u16*)&g)[j]);
+    float a;
+    if (act == 0) {
+      a = x / (1.f + __expf(-x));
+    } else {
+      float c = 0.7978845608028654f * (x + 0.044715f * x * x * x);
+      a = 0.5f * x * (1.f + tanhf(c));
+    }
+    o[j] = f2b(a * b2f(((u16*)&u)[j]));
+  }
+  *(s8v*)(out + i) = *(s8v*)o;
+}
+
+extern "C" hipError_t launch_glu(const void* gate, const void* up, void* out,
+                                 long total, int act, hipStream_t stream) {
+  long blocks = (total / 8 + 255) / 256;
+  hipLaunchKernelGGL(k_glu, dim3((uint32_t)blocks), dim3(256), 0, stream,
+                     (const u16*)gate, (const u16*)up, (u16*)out, total, act);
+  return hipGetLastError();
+}
+
+// ====================================================================
+// Embedding gather: h[m] = embed[ids[m]] * scale  (ids live on device so
+// the decode graph can feed back the sampled token without host sync)
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_embed(const u16* __restrict__ embed, const int* __restrict__ ids,
+        u16* __restrict__ out, int H, float scale) {
+  const int m = blockIdx.x;
+  const int id = ids[m];
+  const u16* src = embed + (size_t)id * H;
+  u16* dst = out + (size_t)m * H;
+  for (int i = threadIdx.x * 8; i < H; i += 2048) {
+    s8v v = *(const s8v*)(src + i);
+    u16 o[8];
+#pragma unroll
+    for (int j = 0; j < 8; j++) o[j] = f2b(b2f(((u16*)&v)[j]) * scale);
+    *(s8v*)(dst + i) = *(s8v*)o;
+  }
+}
+
+extern "C" hipError_t launch_embed(const void* embed, const void* ids,
+                                   void* out, int M, int H, float scale,
+                                   hipStream_t stream) {
+  hipLaunchKernelGGL(k_embed, dim3(M), dim3(256), 0, stream,
+                     (const u16*)embed, (const int*)ids, (u16*)out, H, scale);
+  return hipGetLastError();
+}
+
+// ====================================================================
+// Sampling: greedy argmax or min-p + Gumbel-argmax multinomial.
+//   keep tokens with p >= min_p * p_max  <=>  logit >= max + ln(min_p);
+//   winner = argmax over kept of (logit + Gumbel noise).
+// Writes the token to *next_token (feeds the next decode graph step),
+// appends to out_ring, bumps *len_ptr and *nout — all device-side, so the
+// whole decode step is graph-replayable.
+// ====================================================================
+
+DEVINL uint32_t hash32(uint32_t x) {
+  x ^= x >> 16; x *= 0x7feb352du;
+  x ^= x >> 15; x *= 0x846ca68bu;
+  x ^= x >> 16;
+  return x;
+}
+
+extern "C" __global__ void __launch_bounds__(1024)
+k_sample(const float* __restrict__ logits, int V, float min_p, int greedy,
+         uint64_t seed, uint64_t* __restrict__ ctr,
+         int* __restrict__ next_token, int* __restrict__ out_ring,
+         int* __restrict__ nout, int* __restrict__ len_ptr, int bump_len) {
+  __shared__ float sval[16];
+  __shared__ int sidx[16];
+  const int tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
+
+  // pass 1: global max (value, index)
+  float mv = -INFINITY;
+  int mi = 0;
+  for (int i = tid; i < V; i += 1024) {
+    float v = logits[i];
+    if (v > mv) { mv = v; mi = i; }
+  }
+#pragma unroll
+  for (int w = 1; w < 64; w <<= 1) {
+    float ov = __shfl_xor(mv, w);
+    int oi = __shfl_xor(mi, w);
+    if (ov > mv || (ov == mv && oi < mi)) { mv = ov; mi = oi; }
+  }
+  if (lane == 0) { sval[wave] = mv; sidx[wave] = mi; }
+  __syncthreads();
+  if (tid == 0) {
+    for (int w = 1; w < 16; w++)
+      if (sval[w] > sval[0] || (sval[w] == sval[0] && sidx[w] < sidx[0])) {
+        sval[0] = sval[w]; sidx[0] = sidx[w];
+      }
+  }
+  __syncthreads();
+  const float gmax = sval[0];
+  int winner = sidx[0];
+
+  if (!greedy) {
+    __syncthreads();
+    const float thresh = gmax + __logf(min_p);
+    const uint32_t c = (uint32_t)(*ctr);
+    float bv = -INFINITY;
+    int bi = -1;
+    for (int i = tid; i < V; i += 1024) {
+      float v = logits[i];
+      if (v < thresh) continue;
+      uint32_t r = hash32(hash32(i ^ (c * 0x9e3779b9u)) ^ (uint32_t)(seed));
+      float u = (r + 1.0f) * 2.3283064e-10f;  // (0,1]
+      float gumb = -__logf(-__logf(u));
+      float sc = v + gumb;
+      if (sc > bv) { bv = sc; bi = i; }
+    }
+#pragma unroll
+    for (int w = 1; w < 64; w <<= 1) {
+      float ov = __shfl_xor(bv, w);
+      int oi = __shfl_xor(bi, w);
+      if (ov > bv) { bv = ov; bi = oi; }
+    }
+    if (lane == 0) { sval[wave] = bv; sidx[wave] = bi; }
+    __syncthreads();
+    if (tid == 0) {
+      for (int w = 1; w < 16; w++)
+        if (sval[w] > sval[0]) { sval[0] = sval[w]; sidx[0] = sidx[w]; }
+    }
+    __syncthreads();
+    winner = sidx[0];
+  }
+
+  if (tid == 0) {
+    *next_token = winner;
+    int n = *nout;
+    out_ring[n] = winner;
+    *nout = n + 1;
+    if (bump_len) *len_ptr += 1;
+    if (!greedy) *ctr += 1;
+  }
+}
+
+extern "C" hipError_t launch_sample(const void* logits, int V, float min_p,
+                                    int greedy, uint64_t seed, void* ctr,
+                                    void* next_token, void* out_ring,
+                                    void* nout, void* len_ptr, int bump_len,
+                                    hipStream_t stream) {
+  hipLaunchKernelGGL(k_sample, dim3(1), dim3(1024), 0, stream,
+                     (const float*)logits, V, min_p, greedy, seed,
+                     (uint64_t*)ctr, (int*)next_token, (int*)out_ring,
+                     (int*)nout, (int*)len_ptr, bump_len);
+  return hipGetLastError();
+}
+
+// ====================================================================
+// Prefill GEMM: Y[M,N] = X[M,K] @ W[N,K]^T (+res), bf16 in/out, fp32 acc.
+// MFMA v_mfma_f32_16x16x32_bf16; 128x128 tile, BK=32, 4 waves (2x2),
+// each wave a 64x64 sub-tile (4x4 fragments).  LDS staged, padded rows.
+// Correctness-first structure (guide §5 ladder step ~0-2); prefill only —
+// decode uses k_gemv_bf16.
+// ====================================================================
+
+typedef short bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+#define BM 128
+#define BN 128
+#define BK 32
+#define LDS_STRIDE (BK + 8)  // pad: 40 elems = 80 B, 16B-aligned rows
+
+extern "C" __global__ void __launch_bounds__(256)
+k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
+            u16* __restrict__ Y, const u16* __restrict__ res, int M, int N,
+            int K) {
+  __shared__ u16 As[BM * LDS_STRIDE];
+  __shared__ u16 Bs[BN * LDS_STRIDE];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wrow = wave >> 1, wcol = wave & 1;  // 2x2 waves
+  const int bm = blockIdx.x * BM, bn = blockIdx.y * BN;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; i++)
+#pragma unroll
+    for (int j = 0; j < 4; j++) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // stage thread mapping: 256 threads x 16 B; tile row r = t/4, col (t%4)*8
+  const int sr = tid >> 2, sc = (tid & 3) * 8;
+
+  for (int kt = 0; kt < K; kt += BK) {
+    // A tile: rows clamped to M-1 (stores are masked later)
+#pragma unroll
+    for (int p = 0; p < 2; p++) {
+      int r = sr + p * 64;
+      int gr = bm + r;
+      int grc = gr < M ? gr : (M > 0 ? M - 1 : 0);
+      *(s8v*)(As + r * LDS_STRIDE + sc) =
+          *(const s8v*)(X + (size_t)grc * K + kt + sc);
+      int rb = sr + p * 64;
+      int gb = bn + rb;
+      int gbc = gb < N ? gb : N - 1;  // stores masked on col >= N
+      *(s8v*)(Bs + rb * LDS_STRIDE + sc) =
+          *(const s8v*)(W + (size_t)gbc * K + kt + sc);
+    }
+    __syncthreads();
+
+    // fragments: lane l -> row (l&15), k-chunk (l>>4)*8 of each 16x32 tile
+    const int fr = lane & 15, fk = (lane >> 4) * 8;
+#pragma unroll
+    for (int ks = 0; ks < BK; ks += 32) {
+      bf16x8 a[4], b[4];
+#pragma unroll
+      for (int i = 0; i < 4; i++) {
+        a[i] = *(bf16x8*)(As + (wrow * 64 + i * 16 + fr) * LDS_STRIDE + ks + fk);
+        b[i] = *(bf16x8*)(Bs + (wcol * 64 + i * 16 + fr) * LDS_STRIDE + ks + fk);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; i++)
+#pragma unroll
+        for (int j = 0; j < 4; j++)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: C layout col = lane&15, row = (lane>>4)*4 + reg
+  const int cc = lane & 15, cr = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        int row = bm + wrow * 64 + i * 16 + cr + r;
+        int col = bn + wcol * 64 + j * 16 + cc;
+        if (row < M && col < N) {
+          float v = acc[i][j][r];
+          if (res) v += b2f(res[(size_t)row * N + col]);
+          Y[(size_t)row * N + col] = f2b(v);
+        }
+      }
+    }
+  }
+}
+
+extern "C" hipError_t launch_gemm_bf16(const void* X, const void* W, void* Y,
+                                       const void* res, int M, int N, int K,
+                                       hipStream_t stream) {
+  dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
+  hipLaunchKernelGGL(k_gemm_bf16, grid, dim3(256), 0, stream, (const u16*)X,
+                     (const u16*)W, (u16*)Y, (const u16*)res, M, N, K);
+  return hipGetLastError();
+}
+
+// ====================================================================
+// y += a (bf16, fp32 math).  Used on the TP path where the RCCL
+// all-reduce sits between the row-parallel GEMV and the residual add.
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_addinto(u16* __restrict__ y, const u16* __restrict__ a, long total) {
+  long i = ((long)blockIdx.x * 256 + threadIdx.x) * 8;
+  if (i >= total) return;
+  s8v yv = *(const s8v*)(y + i);
+  s8v av = *(const s8v*)(a + i);
+  u16 o[8];
+#pragma unroll
+  for (int j = 0; j < 8; j++)
+    o[j] = f2b(b2f(((u16*)&yv)[j]) + b2f(((u16*)&av)[j]));
+  *(s8v*)(y + i) = *(s8v*)o;
+}
+
+extern "C" hipError_t launch_addinto(void* y, const void* a, long total,
+                                     hipStream_t stream) {
+  long blocks = (total / 8 + 255) / 256;
+  hipLaunchKernelGGL(k_addinto, dim3((uint32_t)blocks), dim3(256), 0, stream,
+                     (u16*)y, (const u16*)a, total);
+  return hipGetLastError();
+}
+
+// ====================================================================
+// Utility: device-side int32 set/add (for seq-length bookkeeping inside
+// graphs where needed)
+// ====================================================================
+
+extern "C" __global__ void k_i32_set(int* p, int v) { *p = v; }
+extern "C" __global__ void k_i32_add(int* p, int v) { *p += v; }
+
+extern "C" hipError_t launch_i32_set(void* p, int v, hipStream_t stream) {
+  hipLaunchKernelGGL(k_i32_set, dim3(1), dim3(1), 0, stream, (int*)p, v);
+  return hipGetLastError();
+}
+extern "C" hipError_t launch_i32_add(void* p, int v, hipStream_t stream) {
+  hipLaunchKernelGGL(k_i32_add, dim3(1), dim3(1), 0, stream, (int*)p, v);
+  return hipGetLastError();
+}

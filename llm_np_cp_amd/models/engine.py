@@ -1,0 +1,335 @@
+"""GPU inference engine: Llama-3.2 / Gemma-2 forward on hand-written
+CDNA4 HIP kernels, PyTorch tensors as memory containers only.
+
+MI355X-first design (SURVEY §7 stages 2-4):
+  - every compute op is a kernel from ``csrc/llm_ops.hip`` — no ATen math
+    in the forward pass, no eager fallback (ops fail loudly if the
+    extension is missing);
+  - preallocated KV pool per layer (in-place writes, no concat);
+  - decode state (seq length, sampled token, RNG counter, output ring)
+    lives in device memory so a decode step is hipGraph-replayable with
+    zero host round-trips (torch.cuda.CUDAGraph == hipGraph on ROCm);
+  - tensor parallelism: head/row/col-sharded weights per rank, RCCL
+    all-reduce after o_proj and down_proj, all-gather for logits
+    (world_size==1 skips collectives — same code path).
+
+Reference parity: this replaces the CuPy model classes
+(``/root/reference/llama3.2_model.py:334-822``) — same capability
+(forward with KV cache, returns logits), built from the architectural
+spec in SURVEY §2.4 rather than translated.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, Optional
+
+import numpy as np
+import torch
+
+from ..core.config import ModelConfig
+from ..ops import hip_ops as ho
+from ..parallel import tp as tpu
+
+
+class DeviceCacheHandle:
+    """Host-side mirror of the device seq-length (generate()-compatible)."""
+
+    def __init__(self, engine):
+        self.engine = engine
+        self.seq_len = 0
+
+
+class GPUModel:
+    def __init__(self, config: ModelConfig, weights: Dict[str, np.ndarray],
+                 dtype: str = "bf16", max_seq: int = 4096,
+                 prefill_chunk: int = 512, device: Optional[str] = None,
+                 seed: int = 0):
+        if not torch.cuda.is_available():
+            raise RuntimeError("GPUModel requires a GPU (MI355X)")
+        self.config = config
+        self.max_seq = max_seq
+        self.seed = seed
+        self.rank, self.world = tpu.init_distributed()
+        if device is None:
+            device = f"cuda:{self.rank % max(torch.cuda.device_count(), 1)}"
+        self.device = torch.device(device)
+        torch.cuda.set_device(self.device)
+
+        cfg = config
+        tp = self.world
+        assert cfg.num_attention_heads % tp == 0, "nh must divide tp"
+        assert cfg.num_key_value_heads % tp == 0, \
+            f"kv heads {cfg.num_key_value_heads} not divisible by tp {tp}"
+        assert cfg.intermediate_size % tp == 0
+        assert cfg.vocab_size % tp == 0
+        self.nh_l = cfg.num_attention_heads // tp
+        self.kvh_l = cfg.num_key_value_heads // tp
+        self.inter_l = cfg.intermediate_size // tp
+        self.vocab_l = cfg.vocab_size // tp
+        self.hd = cfg.head_dim
+        self.H = cfg.hidden_size
+
+        self._upload_weights(weights)
+        self._alloc_state(prefill_chunk)
+        self._graph = None
+        self._graph_mode = None
+
+    # ------------------------------------------------------------------
+    def _upload_weights(self, w: Dict[str, np.ndarray]):
+        cfg, dev = self.config, self.device
+        tp, r = self.world, self.rank
+        gemma = cfg.model_type == "gemma2"
+
+        def bf16(a: np.ndarray) -> torch.Tensor:
+            return torch.from_numpy(np.ascontiguousarray(a)).to(
+                dev, torch.bfloat16, non_blocking=False)
+
+        def gamma(a: np.ndarray) -> torch.Tensor:
+            g = a.astype(np.float32) + (1.0 if gemma else 0.0)
+            return torch.from_numpy(g).to(dev)
+
+        self.embed = bf16(w["model.embed_tokens.weight"])
+        lm_w = w.get("lm_head.weight", w["model.embed_tokens.weight"])
+        if tp > 1:
+            self.lm_head = bf16(tpu.shard_rows(lm_w, r, tp))
+        elif lm_w is w.get("model.embed_tokens.weight", None) or \
+                cfg.tie_word_embeddings and "lm_head.weight" not in w:
+            self.lm_head = self.embed  # tied, share device memory
+        else:
+            self.lm_head = bf16(lm_w)
+        self.g_final = gamma(w["model.norm.weight"])
+
+        hd = cfg.head_dim
+        self.layers = []
+        for i in range(cfg.num_hidden_layers):
+            p = f"model.layers.{i}"
+            a = f"{p}.self_attn"
+            hrows = lambda x: tpu.shard_rows(x, r, tp) if tp > 1 else x
+            hcols = lambda x: tpu.shard_cols(x, r, tp) if tp > 1 else x
+            lw = {
+                "wq": bf16(hrows(w[f"{a}.q_proj.weight"])),
+                "wk": bf16(hrows(w[f"{a}.k_proj.weight"])),
+                "wv": bf16(hrows(w[f"{a}.v_proj.weight"])),
+                "wo": bf16(hcols(w[f"{a}.o_proj.weight"])),
+                "wgate": bf16(hrows(w[f"{p}.mlp.gate_proj.weight"])),
+                "wup": bf16(hrows(w[f"{p}.mlp.up_proj.weight"])),
+                "wdown": bf16(hcols(w[f"{p}.mlp.down_proj.weight"])),
+                "g_in": gamma(w[f"{p}.input_layernorm.weight"]),
+                "g_post": gamma(w[f"{p}.post_attention_layernorm.weight"]),
+            }
+            if gemma:
+                lw["g_preffn"] = gamma(w[f"{p}.pre_feedforward_layernorm.weight"])
+                lw["g_postffn"] = gamma(w[f"{p}.post_feedforward_layernorm.weight"])
+            self.layers.append(lw)
+
+    def _alloc_state(self, prefill_chunk: int):
+        cfg, dev = self.config, self.device
+        S, hd = self.max_seq, self.hd
+        L = cfg.num_hidden_layers
+        bf = dict(dtype=torch.bfloat16, device=dev)
+        self.k_cache = [torch.zeros(self.kvh_l, S, hd, **bf) for _ in range(L)]
+        self.v_cache = [torch.zeros(self.kvh_l, S, hd, **bf) for _ in range(L)]
+
+        # RoPE tables fp32 (host-precomputed: guide App.B — no device trig)
+        inv = cfg.rope_inv_freq()                      # (hd/2,) fp64
+        t = np.arange(S, dtype=np.float64)
+        freqs = np.outer(t, inv)
+        self.cos_t = torch.from_numpy(np.cos(freqs).astype(np.float32)).to(dev)
+        self.sin_t = torch.from_numpy(np.sin(freqs).astype(np.float32)).to(dev)
+
+        PC = self.PC = prefill_chunk
+        H, I = self.H, self.inter_l
+        self.b_h = torch.zeros(PC, H, **bf)
+        self.b_xn = torch.zeros(PC, H, **bf)
+        self.b_t1 = torch.zeros(PC, H, **bf)
+        self.b_q = torch.zeros(PC, self.nh_l * hd, **bf)
+        self.b_k = torch.zeros(PC, self.kvh_l * hd, **bf)
+        self.b_v = torch.zeros(PC, self.kvh_l * hd, **bf)
+        self.b_att = torch.zeros(PC, self.nh_l * hd, **bf)
+        self.b_gate = torch.zeros(PC, I, **bf)
+        self.b_up = torch.zeros(PC, I, **bf)
+        self.b_logits_l = torch.zeros(self.vocab_l, dtype=torch.float32,
+                                      device=dev)
+        self.b_logits = (self.b_logits_l if self.world == 1 else
+                         torch.zeros(cfg.vocab_size, dtype=torch.float32,
+                                     device=dev))
+        i32 = dict(dtype=torch.int32, device=dev)
+        self.ids_buf = torch.zeros(PC, **i32)
+        self.next_token = torch.zeros(1, **i32)
+        self.out_ring = torch.zeros(65536, **i32)
+        self.nout = torch.zeros(1, **i32)
+        self.len_buf = torch.zeros(1, **i32)
+        self.rng_ctr = torch.zeros(1, dtype=torch.int64, device=dev)
+
+        self.act = 0 if cfg.hidden_act == "silu" else 1
+        self.gemma = cfg.model_type == "gemma2"
+        self.attn_softcap = float(cfg.attn_logit_softcapping or 0.0)
+        self.final_softcap = float(cfg.final_logit_softcapping or 0.0)
+        self.scale = float(cfg.attn_scale)
+
+    # ------------------------------------------------------------------
+    # cache interface (generate()-compatible)
+    # ------------------------------------------------------------------
+    def make_cache(self, n: int) -> DeviceCacheHandle:
+        if n > self.max_seq:
+            raise ValueError(f"requested cache {n} > max_seq {self.max_seq}")
+        self.reset()
+        return DeviceCacheHandle(self)
+
+    def reset(self):
+        self.len_buf.zero_()
+        self.nout.zero_()
+        self.rng_ctr.zero_()
+
+    # ------------------------------------------------------------------
+    # layer stack over rows [0, M) of the scratch buffers
+    # ------------------------------------------------------------------
+    def _linear(self, W, x, y, res=None, M: int = 1, softcap: float = 0.0):
+        if M == 1:
+            ho.gemv(W, x, y, res=res, softcap=softcap)
+        else:
+            ho.gemm(x[:M], W, y, res=res[:M] if res is not None else None)
+
+    def _layers_forward(self, M: int):
+        cfg = self.config
+        eps = cfg.rms_norm_eps
+        h, xn, t1 = self.b_h, self.b_xn, self.b_t1
+        for i, lw in enumerate(self.layers):
+            window = cfg.sliding_window if cfg.is_sliding(i) else 0
+            ho.rmsnorm(h[:M], lw["g_in"], xn[:M], eps=eps)
+            self._linear(lw["wq"], xn, self.b_q, M=M)
+            self._linear(lw["wk"], xn, self.b_k, M=M)
+            self._linear(lw["wv"], xn, self.b_v, M=M)
+            ho.rope_cache(self.b_q, self.b_k, self.b_v, self.k_cache[i],
+                          self.v_cache[i], self.cos_t, self.sin_t,
+                          self.len_buf, M, self.nh_l, self.kvh_l, self.hd)
+            ho.attn(self.b_q, self.k_cache[i], self.v_cache[i], self.b_att,
+                    self.len_buf, M, self.nh_l, self.kvh_l, self.hd,
+                    self.scale, softcap=self.attn_softcap, window=window or 0)
+            if self.gemma:
+                self._linear(lw["wo"], self.b_att, t1, M=M)
+                tpu.all_reduce(t1[:M])
+                ho.rmsnorm(t1[:M], lw["g_post"], h[:M], res=h[:M], eps=eps)
+                ho.rmsnorm(h[:M], lw["g_preffn"], xn[:M], eps=eps)
+            else:
+                if self.world > 1:
+                    self._linear(lw["wo"], self.b_att, t1, M=M)
+                    tpu.all_reduce(t1[:M])
+                    ho.addinto(h[:M], t1[:M])
+                else:
+                    self._linear(lw["wo"], self.b_att, h, res=h, M=M)
+                ho.rmsnorm(h[:M], lw["g_post"], xn[:M], eps=eps)
+            self._linear(lw["wgate"], xn, self.b_gate, M=M)
+            self._linear(lw["wup"], xn, self.b_up, M=M)
+            ho.glu(self.b_gate[:M], self.b_up[:M], self.b_gate[:M], self.act)
+            if self.gemma:
+                self._linear(lw["wdown"], self.b_gate, t1, M=M)
+                tpu.all_reduce(t1[:M])
+                ho.rmsnorm(t1[:M], lw["g_postffn"], h[:M], res=h[:M], eps=eps)
+            else:
+                if self.world > 1:
+                    self._linear(lw["wdown"], self.b_gate, t1, M=M)
+                    tpu.all_reduce(t1[:M])
+                    ho.addinto(h[:M], t1[:M])
+                else:
+                    self._linear(lw["wdown"], self.b_gate, h, res=h, M=M)
+
+    def _lm_head_last(self, M: int):
+        """Final norm + lm_head on the last row -> self.b_logits (f32, V)."""
+        hrow = self.b_h[M - 1] if M > 1 else self.b_h[0]
+        ho.rmsnorm(hrow, self.g_final, self.b_xn[0],
+                   eps=self.config.rms_norm_eps)
+        ho.gemv(self.lm_head, self.b_xn[0], self.b_logits_l,
+                softcap=self.final_softcap)
+        if self.world > 1:
+            tpu.all_gather_into(self.b_logits, self.b_logits_l)
+
+    # ------------------------------------------------------------------
+    # generic forward (oracle-parity / generate()-compatible)
+    # ------------------------------------------------------------------
+    def forward(self, ids: np.ndarray, cache: DeviceCacheHandle,
+                pos0: int) -> np.ndarray:
+        """Returns last-position logits (1, V) fp32 numpy."""
+        ids = np.asarray(ids, dtype=np.int32).ravel()
+        if pos0 != cache.seq_len:
+            raise ValueError("GPU engine requires sequential positions")
+        n = len(ids)
+        done = 0
+        while done < n:
+            M = min(n - done, self.PC)
+            self.ids_buf[:M].copy_(
+                torch.from_numpy(ids[done:done + M].astype(np.int32)))
+            ho.i32_set(self.len_buf, pos0 + done)
+            ho.embed(self.embed, self.ids_buf, self.b_h, M,
+                     self.config.embed_scale)
+            self._layers_forward(M)
+            done += M
+        ho.i32_set(self.len_buf, pos0 + n)
+        self._lm_head_last(M)
+        cache.seq_len = pos0 + n
+        torch.cuda.synchronize()
+        return self.b_logits.float().cpu().numpy()[None, :]
+
+    # ------------------------------------------------------------------
+    # fast device-side decode loop (graph-replayable)
+    # ------------------------------------------------------------------
+    def _decode_step(self, greedy: bool, min_p: float):
+        ho.embed(self.embed, self.next_token, self.b_h, 1,
+                 self.config.embed_scale)
+        self._layers_forward(1)
+        self._lm_head_last(1)
+        ho.sample(self.b_logits, min_p, greedy, self.seed, self.rng_ctr,
+                  self.next_token, self.out_ring, self.nout, self.len_buf,
+                  bump_len=True)
+
+    def capture_decode_graph(self, greedy: bool = True, min_p: float = 0.1):
+        """Capture one decode step into a hipGraph (fixed shapes: KV pool
+        is preallocated and the position is a device scalar — SURVEY §7
+        'graph must be shape-stable').
+
+        NOTE: the warm-up executes ONE REAL decode step (advances the
+        sequence by one token).  Returns the number of real steps taken
+        (1 on fresh capture, 0 when the graph is reused)."""
+        mode = (greedy, min_p)
+        if self._graph_mode == mode:
+            return 0
+        # warm-up (a real step) on a side stream, then capture
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            self._decode_step(greedy, min_p)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._decode_step(greedy, min_p)
+        self._graph = g
+        self._graph_mode = mode
+        return 1
+
+    def prefill(self, ids: np.ndarray, cache: Optional[DeviceCacheHandle] = None):
+        cache = cache or self.make_cache(self.max_seq)
+        logits = self.forward(ids, cache, 0)
+        return cache, logits
+
+    def decode(self, n_tokens: int, greedy: bool = True, min_p: float = 0.1,
+               use_graph: bool = True, first_from_logits: bool = True):
+        """Generate n_tokens ids device-side; returns int32 numpy ids.
+        Assumes prefill() ran (len_buf == prompt length, logits ready)."""
+        if first_from_logits:
+            # sample token 0 from the prefill logits
+            ho.sample(self.b_logits, min_p, greedy, self.seed, self.rng_ctr,
+                      self.next_token, self.out_ring, self.nout,
+                      self.len_buf, bump_len=False)
+        n_steps = n_tokens - (1 if first_from_logits else 0)
+        if use_graph and n_steps > 0:
+            n_steps -= self.capture_decode_graph(greedy, min_p)
+            for _ in range(n_steps):
+                self._graph.replay()
+        else:
+            for _ in range(n_steps):
+                self._decode_step(greedy, min_p)
+        torch.cuda.synchronize()
+        n = int(self.nout.item())
+        return self.out_ring[:n].cpu().numpy()[max(0, n - n_tokens):]

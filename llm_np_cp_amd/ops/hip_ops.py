@@ -1,0 +1,175 @@
+"""ctypes bindings for the hand-written CDNA4 kernel library.
+
+The library (``_libllmops.so``, built in-tree by ``csrc/build.py`` /
+``__graft_entry__.build()``) exposes plain-C launch functions taking raw
+device pointers + the HIP stream; tensors stay ``torch`` tensors (memory
+containers only) and every launch lands on the *current* torch stream, so
+the whole decode step can be captured in a hipGraph via
+``torch.cuda.CUDAGraph``.
+
+This module fails loudly if the extension is missing while a GPU is
+present — there is no silent eager fallback (the HIP path must be the one
+that runs).
+"""
+
+from __future__ import annotations
+
+import ctypes
+import os
+
+import torch
+
+_LIB = None
+_LIB_PATH = os.path.join(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))), "_libllmops.so")
+
+_SIGS = {
+    "launch_gemv_bf16": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 3 +
+                        [ctypes.c_float, ctypes.c_void_p],
+    "launch_rmsnorm": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 2 +
+                      [ctypes.c_float, ctypes.c_int, ctypes.c_void_p],
+    "launch_rope_cache": [ctypes.c_void_p] * 8 + [ctypes.c_int] * 5 +
+                         [ctypes.c_void_p],
+    "launch_attn": [ctypes.c_void_p] * 5 + [ctypes.c_int] * 5 +
+                   [ctypes.c_float, ctypes.c_float, ctypes.c_int,
+                    ctypes.c_void_p],
+    "launch_glu": [ctypes.c_void_p] * 3 + [ctypes.c_long, ctypes.c_int,
+                                           ctypes.c_void_p],
+    "launch_embed": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 +
+                    [ctypes.c_float, ctypes.c_void_p],
+    "launch_sample": [ctypes.c_void_p, ctypes.c_int, ctypes.c_float,
+                      ctypes.c_int, ctypes.c_uint64] + [ctypes.c_void_p] * 5 +
+                     [ctypes.c_int, ctypes.c_void_p],
+    "launch_gemm_bf16": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 3 +
+                        [ctypes.c_void_p],
+    "launch_addinto": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_long,
+                       ctypes.c_void_p],
+    "launch_i32_set": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
+    "launch_i32_add": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
+}
+
+
+def lib():
+    global _LIB
+    if _LIB is None:
+        if not os.path.exists(_LIB_PATH):
+            raise RuntimeError(
+                f"HIP kernel library not built: {_LIB_PATH} missing. "
+                f"Run `python csrc/build.py` (or __graft_entry__.build()).")
+        _LIB = ctypes.CDLL(_LIB_PATH)
+        for name, argtypes in _SIGS.items():
+            fn = getattr(_LIB, name)
+            fn.argtypes = argtypes
+            fn.restype = ctypes.c_int  # hipError_t
+    return _LIB
+
+
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _check(err: int, name: str):
+    if err != 0:
+        raise RuntimeError(f"{name} failed: hipError_t={err}")
+
+
+def _ptr(t) -> int:
+    return 0 if t is None else t.data_ptr()
+
+
+# ----------------------------------------------------------------------
+# op wrappers (shapes validated here; kernels trust their args)
+# ----------------------------------------------------------------------
+
+def gemv(W: torch.Tensor, x: torch.Tensor, y: torch.Tensor,
+         res: torch.Tensor | None = None, softcap: float = 0.0):
+    """y[N] = W[N,K] @ x[K] (+res), bf16 weights/input, bf16 or f32 out."""
+    N, K = W.shape
+    out_f32 = 1 if y.dtype == torch.float32 else 0
+    _check(lib().launch_gemv_bf16(
+        _ptr(W), _ptr(x), _ptr(y), _ptr(res), N, K, out_f32,
+        ctypes.c_float(softcap), _stream()), "gemv")
+
+
+def rmsnorm(x: torch.Tensor, g: torch.Tensor, y: torch.Tensor,
+            res: torch.Tensor | None = None, eps: float = 1e-5):
+    """mode 0: y = norm(x)*g ; with res: y = res + norm(x)*g (Gemma post)."""
+    M = 1 if x.dim() == 1 else x.shape[0]
+    H = x.shape[-1]
+    mode = 0 if res is None else 1
+    _check(lib().launch_rmsnorm(
+        _ptr(x), _ptr(g), _ptr(res), _ptr(y), M, H,
+        ctypes.c_float(eps), mode, _stream()), "rmsnorm")
+
+
+def rope_cache(q: torch.Tensor, k_in: torch.Tensor, v_in: torch.Tensor,
+               k_cache: torch.Tensor, v_cache: torch.Tensor,
+               cos_t: torch.Tensor, sin_t: torch.Tensor,
+               pos_ptr: torch.Tensor, M: int, nh: int, kvh: int, hd: int):
+    S = k_cache.shape[1]
+    _check(lib().launch_rope_cache(
+        _ptr(q), _ptr(k_in), _ptr(v_in), _ptr(k_cache), _ptr(v_cache),
+        _ptr(cos_t), _ptr(sin_t), _ptr(pos_ptr), M, nh, kvh, hd, S,
+        _stream()), "rope_cache")
+
+
+def attn(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
+         out: torch.Tensor, len_ptr: torch.Tensor, M: int, nh: int,
+         kvh: int, hd: int, scale: float, softcap: float = 0.0,
+         window: int = 0):
+    S = k_cache.shape[1]
+    _check(lib().launch_attn(
+        _ptr(q), _ptr(k_cache), _ptr(v_cache), _ptr(out), _ptr(len_ptr),
+        M, nh, kvh, hd, S, ctypes.c_float(scale), ctypes.c_float(softcap),
+        window, _stream()), "attn")
+
+
+def glu(gate: torch.Tensor, up: torch.Tensor, out: torch.Tensor, act: int):
+    """out = act(gate) * up; act 0 = SiLU, 1 = tanh-GELU."""
+    total = gate.numel()
+    assert total % 8 == 0
+    _check(lib().launch_glu(_ptr(gate), _ptr(up), _ptr(out), total, act,
+                            _stream()), "glu")
+
+
+def embed(table: torch.Tensor, ids: torch.Tensor, out: torch.Tensor,
+          M: int, scale: float = 1.0):
+    H = table.shape[1]
+    _check(lib().launch_embed(_ptr(table), _ptr(ids), _ptr(out), M, H,
+                              ctypes.c_float(scale), _stream()), "embed")
+
+
+def sample(logits: torch.Tensor, min_p: float, greedy: bool, seed: int,
+           ctr: torch.Tensor, next_token: torch.Tensor,
+           out_ring: torch.Tensor, nout: torch.Tensor,
+           len_ptr: torch.Tensor, bump_len: bool = True):
+    V = logits.shape[-1]
+    _check(lib().launch_sample(
+        _ptr(logits), V, ctypes.c_float(min_p), 1 if greedy else 0,
+        ctypes.c_uint64(seed), _ptr(ctr), _ptr(next_token), _ptr(out_ring),
+        _ptr(nout), _ptr(len_ptr), 1 if bump_len else 0, _stream()), "sample")
+
+
+def gemm(X: torch.Tensor, W: torch.Tensor, Y: torch.Tensor,
+         res: torch.Tensor | None = None):
+    """Y[M,N] = X[M,K] @ W[N,K]^T (+res), bf16, MFMA prefill path."""
+    M, K = X.shape
+    N = W.shape[0]
+    assert W.shape[1] == K and K % 32 == 0
+    _check(lib().launch_gemm_bf16(_ptr(X), _ptr(W), _ptr(Y), _ptr(res),
+                                  M, N, K, _stream()), "gemm")
+
+
+def addinto(y: torch.Tensor, a: torch.Tensor):
+    """y += a (bf16)."""
+    total = y.numel()
+    assert total % 8 == 0
+    _check(lib().launch_addinto(_ptr(y), _ptr(a), total, _stream()), "addinto")
+
+
+def i32_set(buf: torch.Tensor, v: int):
+    _check(lib().launch_i32_set(_ptr(buf), v, _stream()), "i32_set")
+
+
+def i32_add(buf: torch.Tensor, v: int):
+    _check(lib().launch_i32_add(_ptr(buf), v, _stream()), "i32_add")

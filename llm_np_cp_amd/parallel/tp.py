@@ -1,0 +1,84 @@
+"""Tensor-parallel process-group helpers (RCCL over xGMI).
+
+One process per GPU; ``torch.distributed`` backend "nccl" IS RCCL on
+ROCm.  World size 1 means every collective is a no-op, so the TP=1 path
+is the same code with collectives skipped (doubles as the CPU-CI "fake
+backend" — SURVEY §4 distributed test strategy).
+
+Sharding scheme (SURVEY §7 stage 4):
+  q/k/v projections  : head-sharded rows      (column-parallel)
+  o_proj             : column-sharded         (row-parallel, all-reduce)
+  gate/up            : row-sharded            (column-parallel)
+  down               : column-sharded         (row-parallel, all-reduce)
+  lm_head            : vocab-row-sharded      (all-gather of logits)
+  embeddings, norms  : replicated
+  KV cache           : kv-head-sharded (per-rank pool)
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import numpy as np
+
+
+def init_distributed(backend: Optional[str] = None):
+    """Initialize torch.distributed from torchrun env vars if present.
+    Returns (rank, world_size). Safe to call with no env (1 GPU)."""
+    import torch
+    import torch.distributed as dist
+
+    if dist.is_initialized():
+        return dist.get_rank(), dist.get_world_size()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world == 1:
+        return 0, 1
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29517")
+    dist.init_process_group(backend=backend)
+    return dist.get_rank(), dist.get_world_size()
+
+
+def all_reduce(t):
+    import torch.distributed as dist
+
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        dist.all_reduce(t)
+
+
+def all_gather_into(out, t):
+    import torch.distributed as dist
+
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        dist.all_gather_into_tensor(out, t)
+    else:
+        out.copy_(t.view(out.shape))
+
+
+def broadcast(t, src: int = 0):
+    import torch.distributed as dist
+
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        dist.broadcast(t, src)
+
+
+# ----------------------------------------------------------------------
+# Weight sharding (numpy, done at load time; SURVEY §5 checkpoint scope:
+# per-rank TP sharding happens on the host before the single H2D copy)
+# ----------------------------------------------------------------------
+
+def shard_rows(w: np.ndarray, rank: int, world: int) -> np.ndarray:
+    n = w.shape[0]
+    assert n % world == 0, f"cannot row-shard {w.shape} over {world}"
+    c = n // world
+    return np.ascontiguousarray(w[rank * c:(rank + 1) * c])
+
+
+def shard_cols(w: np.ndarray, rank: int, world: int) -> np.ndarray:
+    n = w.shape[1]
+    assert n % world == 0, f"cannot col-shard {w.shape} over {world}"
+    c = n // world
+    return np.ascontiguousarray(w[:, rank * c:(rank + 1) * c])

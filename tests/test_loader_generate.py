@@ -1,0 +1,89 @@
+"""Checkpoint-directory loading, generate loop, and sampling tests (CPU)."""
+
+import numpy as np
+import pytest
+
+import llm_np_cp_amd as L
+from llm_np_cp_amd.runtime.sampling import SamplingParams, sample_token
+
+
+def test_load_synthetic_checkpoint_dir(tmp_path):
+    from llm_np_cp_amd.io.loader import write_synthetic_checkpoint
+
+    d = str(tmp_path / "ckpt")
+    write_synthetic_checkpoint(d, "tiny-llama", seed=5)
+    tok, model, cfg = L.load_model(d, backend="numpy")
+    assert cfg.model_type == "llama"
+    r = L.generate("hello", tok, model, max_tokens=4, stream=False,
+                   params=SamplingParams(strategy="greedy"), stop_on_eos=False)
+    assert len(r.token_ids) == 4
+
+
+def test_load_preset_and_generate_deterministic():
+    tok, model, cfg = L.load_model("tiny-llama", backend="numpy", seed=11)
+    p = SamplingParams(strategy="min_p", seed=123)
+    r1 = L.generate("Once upon a time", tok, model, max_tokens=6,
+                    stream=False, params=p, stop_on_eos=False)
+    r2 = L.generate("Once upon a time", tok, model, max_tokens=6,
+                    stream=False, params=p, stop_on_eos=False)
+    assert r1.token_ids == r2.token_ids
+
+
+def test_cacheless_mode_matches_cached():
+    tok, model, cfg = L.load_model("tiny-llama", backend="numpy", seed=2)
+    p = SamplingParams(strategy="greedy")
+    a = L.generate("abcd", tok, model, max_tokens=5, stream=False, params=p,
+                   use_cache=True, stop_on_eos=False)
+    b = L.generate("abcd", tok, model, max_tokens=5, stream=False, params=p,
+                   use_cache=False, stop_on_eos=False)
+    assert a.token_ids == b.token_ids
+
+
+def test_streaming_callback_receives_every_token():
+    tok, model, cfg = L.load_model("tiny-llama", backend="numpy", seed=2)
+    pieces = []
+    r = L.generate("xy", tok, model, max_tokens=3, stream=True,
+                   params=SamplingParams(strategy="greedy"),
+                   stop_on_eos=False, on_token=pieces.append)
+    assert len(pieces) == 3
+    assert "".join(pieces) == r.text
+
+
+def test_min_p_masks_low_prob_tokens():
+    logits = np.array([10.0, 9.9, 0.0, -5.0], dtype=np.float32)
+    counts = np.zeros(4, int)
+    rng = np.random.default_rng(0)
+    p = SamplingParams(strategy="min_p", min_p=0.1)
+    for _ in range(200):
+        counts[sample_token(logits, p, rng)] += 1
+    assert counts[2] == 0 and counts[3] == 0
+    assert counts[0] > 0 and counts[1] > 0
+
+
+def test_greedy_picks_argmax():
+    logits = np.array([0.0, 3.0, 2.0], dtype=np.float32)
+    assert sample_token(logits, SamplingParams(strategy="greedy")) == 1
+
+
+def test_top_k_and_top_p():
+    logits = np.array([5.0, 4.0, 3.0, -10.0], dtype=np.float32)
+    rng = np.random.default_rng(1)
+    for _ in range(50):
+        t = sample_token(logits, SamplingParams(strategy="top_k", top_k=2), rng)
+        assert t in (0, 1)
+    for _ in range(50):
+        t = sample_token(logits, SamplingParams(strategy="top_p", top_p=0.5), rng)
+        assert t == 0
+
+
+def test_rope_scaling_llama3_changes_freqs():
+    cfg = L.preset_config("llama-3.2-1b")
+    scaled = cfg.rope_inv_freq()
+    cfg2 = L.preset_config("llama-3.2-1b")
+    cfg2.rope_scaling = None
+    plain = cfg2.rope_inv_freq()
+    assert scaled.shape == plain.shape == (cfg.head_dim // 2,)
+    # low-frequency (long-wavelength) components are divided by factor 32
+    assert np.any(scaled < plain * 0.5)
+    # high-frequency components unchanged
+    assert np.allclose(scaled[0], plain[0])
