@@ -53,21 +53,76 @@ DEVINL float wave_reduce_sum(float v) {
 }
 
 // ====================================================================
-// GEMV: y[N] = W[N,K] @ x[K]  (+ res)  (optional softcap, f32 out)
-// W bf16 row-major; one wave per output row; x staged once into LDS.
+// GEMV: y[N] = W[N,K] @ stage(x)[K]  (+ res)  (optional softcap, f32 out)
+// W bf16 row-major; one wave per output row; the input vector is staged
+// once into LDS with an optional fused pre-op (this is where the decode
+// path's RMSNorm and GLU live — they cost an LDS pass, not a kernel):
+//   stage 0 (RAW):  xs = x1
+//   stage 1 (NORM): xs = rmsnorm(x1) * g     (g fp32, Gemma +1 prefolded)
+//   stage 2 (GLU):  xs = act(x1) * x2        (act 0 = SiLU, 1 = tanh-GELU)
 // Decode is bound by streaming W once: target ~HBM roofline.
 // ====================================================================
 
 #define GEMV_ROWS_PER_BLOCK 4
+#define STAGE_RAW 0
+#define STAGE_NORM 1
+#define STAGE_GLU 2
 
 extern "C" __global__ void __launch_bounds__(256)
 k_gemv_bf16(const u16* __restrict__ W, const u16* __restrict__ x,
+            const u16* __restrict__ x2, const float* __restrict__ g,
             void* __restrict__ y, const u16* __restrict__ res,
-            int N, int K, int out_f32, float softcap) {
+            int N, int K, int stage, int act, float eps, int out_f32,
+            float softcap) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   u16* xs = (u16*)smem;
-  for (int i = threadIdx.x * 8; i < K; i += 256 * 8)
-    *(s8v*)(xs + i) = *(const s8v*)(x + i);
+  if (stage == STAGE_NORM) {
+    float ss = 0.f;
+    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+      s8v v = *(const s8v*)(x + i);
+      *(s8v*)(xs + i) = v;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = b2f(((u16*)&v)[j]);
+        ss += f * f;
+      }
+    }
+    float* red = (float*)(smem + (size_t)K * 2);
+    ss = wave_reduce_sum(ss);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+    __syncthreads();
+    float rnorm = rsqrtf((red[0] + red[1] + red[2] + red[3]) / (float)K + eps);
+    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+      s8v v = *(s8v*)(xs + i);
+      u16 o[8];
+#pragma unroll
+      for (int j = 0; j < 8; j++)
+        o[j] = f2b(b2f(((u16*)&v)[j]) * rnorm * g[i + j]);
+      *(s8v*)(xs + i) = *(s8v*)o;
+    }
+  } else if (stage == STAGE_GLU) {
+    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+      s8v gv = *(const s8v*)(x + i);
+      s8v uv = *(const s8v*)(x2 + i);
+      u16 o[8];
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float xx = b2f(((u16*)&gv)[j]);
+        float a;
+        if (act == 0) {
+          a = xx / (1.f + __expf(-xx));
+        } else {
+          float c = 0.7978845608028654f * (xx + 0.044715f * xx * xx * xx);
+          a = 0.5f * xx * (1.f + tanhf(c));
+        }
+        o[j] = f2b(a * b2f(((u16*)&uv)[j]));
+      }
+      *(s8v*)(xs + i) = *(s8v*)o;
+    }
+  } else {
+    for (int i = threadIdx.x * 8; i < K; i += 256 * 8)
+      *(s8v*)(xs + i) = *(const s8v*)(x + i);
+  }
   __syncthreads();
 
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
@@ -111,15 +166,18 @@ k_gemv_bf16(const u16* __restrict__ W, const u16* __restrict__ x,
   }
 }
 
-extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x, void* y,
+extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
+                                       const void* x2, const void* g, void* y,
                                        const void* res, int N, int K,
+                                       int stage, int act, float eps,
                                        int out_f32, float softcap,
                                        hipStream_t stream) {
   int blocks = (N + GEMV_ROWS_PER_BLOCK - 1) / GEMV_ROWS_PER_BLOCK;
-  size_t lds = (size_t)K * 2;
+  size_t lds = (size_t)K * 2 + 16;
   hipLaunchKernelGGL(k_gemv_bf16, dim3(blocks), dim3(256), lds, stream,
-                     (const u16*)W, (const u16*)x, y, (const u16*)res, N, K,
-                     out_f32, softcap);
+                     (const u16*)W, (const u16*)x, (const u16*)x2,
+                     (const float*)g, y, (const u16*)res, N, K, stage, act,
+                     eps, out_f32, softcap);
   return hipGetLastError();
 }
 
@@ -357,6 +415,176 @@ k_attn(const u16* __restrict__ q, const u16* __restrict__ kc,
   }
 }
 
+// ====================================================================
+// Fused decode attention (M=1): RoPE(q,k) + KV-pool write + online-
+// softmax GQA in ONE kernel.  Input is the raw fused-QKV GEMV output
+// [nh*hd | kvh*hd | kvh*hd]; each head-block rotates its own q head and
+// (redundantly, cheaply) its kv-head's new k/v — so no block depends on
+// this step's cache write; the first block of each kv-head group
+// persists k/v to the pool for future steps.
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
+           u16* __restrict__ vc, u16* __restrict__ out,
+           const int* __restrict__ len_ptr, const float* __restrict__ cost,
+           const float* __restrict__ sint, int nh, int kvh, int hd, int S,
+           float scale, float softcap, int window) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* red = (float*)smem;
+
+  const int h = blockIdx.x;
+  const int grp = nh / kvh;
+  const int kvhead = h / grp;
+  const int pos = *len_ptr;
+  const int T = pos + 1;
+  int start = 0;
+  if (window > 0 && T - window > 0) start = T - window;
+
+  const int LP = hd / 8;
+  const int PP = 64 / LP;
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int p = lane / LP, d0 = (lane % LP) * 8;
+  const int hd2 = hd / 2;
+  const float* cp = cost + (size_t)pos * hd2;
+  const float* sp = sint + (size_t)pos * hd2;
+
+  // rotated q fragment for this head at dims [d0, d0+8)
+  const u16* qh = qkv + (size_t)h * hd;
+  const u16* kh = qkv + (size_t)(nh + kvhead) * hd;
+  const u16* vh = qkv + (size_t)(nh + kvh + kvhead) * hd;
+  float qf[8], kn[8], vn[8];
+#pragma unroll
+  for (int j = 0; j < 8; j++) {
+    int d = d0 + j;
+    if (d < hd2) {
+      float c = cp[d], s = sp[d];
+      qf[j] = b2f(qh[d]) * c - b2f(qh[d + hd2]) * s;
+      kn[j] = b2f(kh[d]) * c - b2f(kh[d + hd2]) * s;
+    } else {
+      float c = cp[d - hd2], s = sp[d - hd2];
+      qf[j] = b2f(qh[d]) * c + b2f(qh[d - hd2]) * s;
+      kn[j] = b2f(kh[d]) * c + b2f(kh[d - hd2]) * s;
+    }
+    vn[j] = b2f(vh[d0 + j]);
+  }
+  // persist this step's k/v once per kv-head (wave 0, one lane-group)
+  if (h == kvhead * grp && wave == 0 && p == 0) {
+    u16 ko[8], vo[8];
+#pragma unroll
+    for (int j = 0; j < 8; j++) { ko[j] = f2b(kn[j]); vo[j] = f2b(vn[j]); }
+    *(s8v*)(kc + ((size_t)kvhead * S + pos) * hd + d0) = *(s8v*)ko;
+    *(s8v*)(vc + ((size_t)kvhead * S + pos) * hd + d0) = *(s8v*)vo;
+  }
+
+  const u16* K0 = kc + (size_t)kvhead * S * hd;
+  const u16* V0 = vc + (size_t)kvhead * S * hd;
+
+  float mrun = -INFINITY, lrun = 0.f, acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; j++) acc[j] = 0.f;
+
+  // history [start, pos) from the pool
+  for (int t0 = start + wave * PP; t0 < pos; t0 += 4 * PP) {
+    int t = t0 + p;
+    bool valid = t < pos;
+    int tl = valid ? t : (pos > 0 ? pos - 1 : 0);
+    s8v kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
+    float partial = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; j++) partial += qf[j] * b2f(((u16*)&kv)[j]);
+    for (int w = 1; w < LP; w <<= 1) partial += __shfl_xor(partial, w);
+    float score = partial * scale;
+    if (softcap > 0.f) score = softcap * tanhf(score / softcap);
+    if (!valid) score = -INFINITY;
+    float mnew = fmaxf(mrun, score);
+    if (mnew == -INFINITY) continue;
+    float alpha = __expf(mrun - mnew);
+    float pv = __expf(score - mnew);
+    lrun = lrun * alpha + pv;
+    s8v vv = *(const s8v*)(V0 + (size_t)tl * hd + d0);
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      acc[j] = acc[j] * alpha + pv * b2f(((u16*)&vv)[j]);
+    mrun = mnew;
+  }
+
+  // current position from registers (wave 0, p == 0 lanes only)
+  if (wave == 0 && p == 0) {
+    float partial = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; j++) partial += qf[j] * kn[j];
+    for (int w = 1; w < LP; w <<= 1) partial += __shfl_xor(partial, w);
+    float score = partial * scale;
+    if (softcap > 0.f) score = softcap * tanhf(score / softcap);
+    float mnew = fmaxf(mrun, score);
+    float alpha = __expf(mrun - mnew);
+    float pv = __expf(score - mnew);
+    lrun = lrun * alpha + pv;
+#pragma unroll
+    for (int j = 0; j < 8; j++) acc[j] = acc[j] * alpha + pv * vn[j];
+    mrun = mnew;
+  }
+
+  for (int w = LP; w < 64; w <<= 1) {
+    float mo = __shfl_xor(mrun, w);
+    float lo = __shfl_xor(lrun, w);
+    float mn = fmaxf(mrun, mo);
+    float sa = (mrun == -INFINITY && mo == -INFINITY) ? 0.f : __expf(mrun - mn);
+    float sb = (mrun == -INFINITY && mo == -INFINITY) ? 0.f : __expf(mo - mn);
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float ao = __shfl_xor(acc[j], w);
+      acc[j] = acc[j] * sa + ao * sb;
+    }
+    lrun = lrun * sa + lo * sb;
+    mrun = mn;
+  }
+
+  float* accs = red;
+  float* mls = red + 4 * hd;
+  if (lane < LP) {
+#pragma unroll
+    for (int j = 0; j < 8; j++) accs[wave * hd + d0 + j] = acc[j];
+  }
+  if (lane == 0) {
+    mls[wave * 2] = mrun;
+    mls[wave * 2 + 1] = lrun;
+  }
+  __syncthreads();
+  if (wave == 0) {
+    float mt = fmaxf(fmaxf(mls[0], mls[2]), fmaxf(mls[4], mls[6]));
+    float lt = 0.f, sc[4];
+#pragma unroll
+    for (int w = 0; w < 4; w++) {
+      sc[w] = (mls[w * 2] == -INFINITY) ? 0.f : __expf(mls[w * 2] - mt);
+      lt += mls[w * 2 + 1] * sc[w];
+    }
+    float inv = 1.f / lt;
+    for (int d = lane; d < hd; d += 64) {
+      float v = 0.f;
+#pragma unroll
+      for (int w = 0; w < 4; w++) v += accs[w * hd + d] * sc[w];
+      out[(size_t)h * hd + d] = f2b(v * inv);
+    }
+  }
+}
+
+extern "C" hipError_t launch_attn_dec(const void* qkv, void* kc, void* vc,
+                                      void* out, const void* len_ptr,
+                                      const void* cost, const void* sint,
+                                      int nh, int kvh, int hd, int S,
+                                      float scale, float softcap, int window,
+                                      hipStream_t stream) {
+  size_t lds = (4 * hd + 8) * sizeof(float);
+  hipLaunchKernelGGL(k_attn_dec, dim3(nh), dim3(256), lds, stream,
+                     (const u16*)qkv, (u16*)kc, (u16*)vc, (u16*)out,
+                     (const int*)len_ptr, (const float*)cost,
+                     (const float*)sint, nh, kvh, hd, S, scale, softcap,
+                     window);
+  return hipGetLastError();
+}
+
 extern "C" hipError_t launch_attn(const void* q, const void* kc,
                                   const void* vc, void* out,
                                   const void* len_ptr, int M, int nh, int kvh,
@@ -450,90 +678,125 @@ DEVINL uint32_t hash32(uint32_t x) {
   return x;
 }
 
-extern "C" __global__ void __launch_bounds__(1024)
-k_sample(const float* __restrict__ logits, int V, float min_p, int greedy,
-         uint64_t seed, uint64_t* __restrict__ ctr,
-         int* __restrict__ next_token, int* __restrict__ out_ring,
-         int* __restrict__ nout, int* __restrict__ len_ptr, int bump_len) {
-  __shared__ float sval[16];
-  __shared__ int sidx[16];
-  const int tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
+// order-preserving float->u32 key (monotone for max); invertible
+DEVINL uint32_t fkey(float f) {
+  union { float f; uint32_t u; } c; c.f = f;
+  return (c.u & 0x80000000u) ? ~c.u : (c.u ^ 0x80000000u);
+}
+DEVINL float fkey_inv(uint32_t k) {
+  union { float f; uint32_t u; } c;
+  c.u = (k & 0x80000000u) ? (k ^ 0x80000000u) : ~k;
+  return c.f;
+}
 
-  // pass 1: global max (value, index)
+// pack (key, idx): ties prefer the SMALLEST index (matches np.argmax)
+DEVINL uint64_t pack_ki(uint32_t key, int idx) {
+  return ((uint64_t)key << 32) | (uint32_t)(0x7fffffff - idx);
+}
+DEVINL int unpack_idx(uint64_t p) { return 0x7fffffff - (int)(uint32_t)p; }
+
+// pass 1 (min-p only): global max logit -> *gmax (packed)
+extern "C" __global__ void __launch_bounds__(256)
+k_logit_max(const float* __restrict__ logits, int V,
+            unsigned long long* __restrict__ gmax) {
   float mv = -INFINITY;
   int mi = 0;
-  for (int i = tid; i < V; i += 1024) {
+  for (int i = blockIdx.x * 256 + threadIdx.x; i < V; i += gridDim.x * 256) {
     float v = logits[i];
-    if (v > mv) { mv = v; mi = i; }
+    if (v > mv || (v == mv && i < mi)) { mv = v; mi = i; }
   }
+  uint64_t pk = pack_ki(fkey(mv), mi);
 #pragma unroll
   for (int w = 1; w < 64; w <<= 1) {
-    float ov = __shfl_xor(mv, w);
-    int oi = __shfl_xor(mi, w);
-    if (ov > mv || (ov == mv && oi < mi)) { mv = ov; mi = oi; }
+    uint64_t o = __shfl_xor((unsigned long long)pk, w);
+    if (o > pk) pk = o;
   }
-  if (lane == 0) { sval[wave] = mv; sidx[wave] = mi; }
+  __shared__ unsigned long long ws[4];
+  if ((threadIdx.x & 63) == 0) ws[threadIdx.x >> 6] = pk;
   __syncthreads();
-  if (tid == 0) {
-    for (int w = 1; w < 16; w++)
-      if (sval[w] > sval[0] || (sval[w] == sval[0] && sidx[w] < sidx[0])) {
-        sval[0] = sval[w]; sidx[0] = sidx[w];
-      }
+  if (threadIdx.x == 0) {
+    uint64_t b = ws[0];
+    for (int w = 1; w < 4; w++) if (ws[w] > b) b = ws[w];
+    atomicMax(gmax, (unsigned long long)b);
   }
-  __syncthreads();
-  const float gmax = sval[0];
-  int winner = sidx[0];
+}
 
-  if (!greedy) {
-    __syncthreads();
-    const float thresh = gmax + __logf(min_p);
-    const uint32_t c = (uint32_t)(*ctr);
-    float bv = -INFINITY;
-    int bi = -1;
-    for (int i = tid; i < V; i += 1024) {
-      float v = logits[i];
-      if (v < thresh) continue;
-      uint32_t r = hash32(hash32(i ^ (c * 0x9e3779b9u)) ^ (uint32_t)(seed));
+// pass 2: winner = argmax over kept tokens of (logit [+ Gumbel]) -> *pick
+extern "C" __global__ void __launch_bounds__(256)
+k_sample_pick(const float* __restrict__ logits, int V, float min_p,
+              int greedy, uint64_t seed,
+              const unsigned long long* __restrict__ gmax,
+              const uint64_t* __restrict__ ctr,
+              unsigned long long* __restrict__ pick) {
+  float thresh = -INFINITY;
+  if (!greedy) thresh = fkey_inv((uint32_t)(*gmax >> 32)) + __logf(min_p);
+  const uint32_t c = (uint32_t)(*ctr);
+  float bv = -INFINITY;
+  int bi = 0x7fffffff;
+  for (int i = blockIdx.x * 256 + threadIdx.x; i < V; i += gridDim.x * 256) {
+    float v = logits[i];
+    if (v < thresh) continue;
+    float sc = v;
+    if (!greedy) {
+      uint32_t r = hash32(hash32((uint32_t)i ^ (c * 0x9e3779b9u)) ^
+                          (uint32_t)seed);
       float u = (r + 1.0f) * 2.3283064e-10f;  // (0,1]
-      float gumb = -__logf(-__logf(u));
-      float sc = v + gumb;
-      if (sc > bv) { bv = sc; bi = i; }
+      sc = v - __logf(-__logf(u));
     }
+    if (sc > bv || (sc == bv && i < bi)) { bv = sc; bi = i; }
+  }
+  uint64_t pk = pack_ki(fkey(bv), bi);
 #pragma unroll
-    for (int w = 1; w < 64; w <<= 1) {
-      float ov = __shfl_xor(bv, w);
-      int oi = __shfl_xor(bi, w);
-      if (ov > bv) { bv = ov; bi = oi; }
-    }
-    if (lane == 0) { sval[wave] = bv; sidx[wave] = bi; }
-    __syncthreads();
-    if (tid == 0) {
-      for (int w = 1; w < 16; w++)
-        if (sval[w] > sval[0]) { sval[0] = sval[w]; sidx[0] = sidx[w]; }
-    }
-    __syncthreads();
-    winner = sidx[0];
+  for (int w = 1; w < 64; w <<= 1) {
+    uint64_t o = __shfl_xor((unsigned long long)pk, w);
+    if (o > pk) pk = o;
   }
+  __shared__ unsigned long long ws[4];
+  if ((threadIdx.x & 63) == 0) ws[threadIdx.x >> 6] = pk;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint64_t b = ws[0];
+    for (int w = 1; w < 4; w++) if (ws[w] > b) b = ws[w];
+    atomicMax(pick, (unsigned long long)b);
+  }
+}
 
-  if (tid == 0) {
-    *next_token = winner;
-    int n = *nout;
-    out_ring[n] = winner;
-    *nout = n + 1;
-    if (bump_len) *len_ptr += 1;
-    if (!greedy) *ctr += 1;
-  }
+// pass 3: commit the winner, reset scratch, advance device-side state
+extern "C" __global__ void k_sample_fin(
+    unsigned long long* __restrict__ pick,
+    unsigned long long* __restrict__ gmax, uint64_t* __restrict__ ctr,
+    int* __restrict__ next_token, int* __restrict__ out_ring,
+    int* __restrict__ nout, int* __restrict__ len_ptr, int bump_len,
+    int greedy) {
+  int winner = unpack_idx(*pick);
+  *next_token = winner;
+  int n = *nout;
+  out_ring[n] = winner;
+  *nout = n + 1;
+  if (bump_len) *len_ptr += 1;
+  if (!greedy) *ctr += 1;
+  *pick = 0ull;
+  *gmax = 0ull;
 }
 
 extern "C" hipError_t launch_sample(const void* logits, int V, float min_p,
                                     int greedy, uint64_t seed, void* ctr,
-                                    void* next_token, void* out_ring,
-                                    void* nout, void* len_ptr, int bump_len,
-                                    hipStream_t stream) {
-  hipLaunchKernelGGL(k_sample, dim3(1), dim3(1024), 0, stream,
+                                    void* gmax, void* pick, void* next_token,
+                                    void* out_ring, void* nout, void* len_ptr,
+                                    int bump_len, hipStream_t stream) {
+  int blocks = (V + 255) / 256;
+  if (blocks > 512) blocks = 512;
+  if (!greedy)
+    hipLaunchKernelGGL(k_logit_max, dim3(blocks), dim3(256), 0, stream,
+                       (const float*)logits, V, (unsigned long long*)gmax);
+  hipLaunchKernelGGL(k_sample_pick, dim3(blocks), dim3(256), 0, stream,
                      (const float*)logits, V, min_p, greedy, seed,
+                     (const unsigned long long*)gmax, (const uint64_t*)ctr,
+                     (unsigned long long*)pick);
+  hipLaunchKernelGGL(k_sample_fin, dim3(1), dim3(1), 0, stream,
+                     (unsigned long long*)pick, (unsigned long long*)gmax,
                      (uint64_t*)ctr, (int*)next_token, (int*)out_ring,
-                     (int*)nout, (int*)len_ptr, bump_len);
+                     (int*)nout, (int*)len_ptr, bump_len, greedy);
   return hipGetLastError();
 }
 

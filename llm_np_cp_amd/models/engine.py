@@ -107,13 +107,24 @@ class GPUModel:
             a = f"{p}.self_attn"
             hrows = lambda x: tpu.shard_rows(x, r, tp) if tp > 1 else x
             hcols = lambda x: tpu.shard_cols(x, r, tp) if tp > 1 else x
+            import numpy as _np
+            wqkv = bf16(_np.concatenate([
+                hrows(w[f"{a}.q_proj.weight"]),
+                hrows(w[f"{a}.k_proj.weight"]),
+                hrows(w[f"{a}.v_proj.weight"])], axis=0))
+            wgu = bf16(_np.concatenate([
+                hrows(w[f"{p}.mlp.gate_proj.weight"]),
+                hrows(w[f"{p}.mlp.up_proj.weight"])], axis=0))
+            nq = self.nh_l * hd
+            nkv = self.kvh_l * hd
+            I = self.inter_l
             lw = {
-                "wq": bf16(hrows(w[f"{a}.q_proj.weight"])),
-                "wk": bf16(hrows(w[f"{a}.k_proj.weight"])),
-                "wv": bf16(hrows(w[f"{a}.v_proj.weight"])),
+                # fused tensors (decode path) + contiguous row views
+                "wqkv": wqkv,
+                "wq": wqkv[:nq], "wk": wqkv[nq:nq + nkv],
+                "wv": wqkv[nq + nkv:],
+                "wgu": wgu, "wgate": wgu[:I], "wup": wgu[I:],
                 "wo": bf16(hcols(w[f"{a}.o_proj.weight"])),
-                "wgate": bf16(hrows(w[f"{p}.mlp.gate_proj.weight"])),
-                "wup": bf16(hrows(w[f"{p}.mlp.up_proj.weight"])),
                 "wdown": bf16(hcols(w[f"{p}.mlp.down_proj.weight"])),
                 "g_in": gamma(w[f"{p}.input_layernorm.weight"]),
                 "g_post": gamma(w[f"{p}.post_attention_layernorm.weight"]),
@@ -154,6 +165,10 @@ class GPUModel:
         self.b_logits = (self.b_logits_l if self.world == 1 else
                          torch.zeros(cfg.vocab_size, dtype=torch.float32,
                                      device=dev))
+        self.b_qkv = torch.zeros((self.nh_l + 2 * self.kvh_l) * hd, **bf)
+        self.b_gu = torch.zeros(2 * I, **bf)
+        self.s_gmax = torch.zeros(1, dtype=torch.int64, device=dev)
+        self.s_pick = torch.zeros(1, dtype=torch.int64, device=dev)
         i32 = dict(dtype=torch.int32, device=dev)
         self.ids_buf = torch.zeros(PC, **i32)
         self.next_token = torch.zeros(1, **i32)
@@ -181,6 +196,8 @@ class GPUModel:
         self.len_buf.zero_()
         self.nout.zero_()
         self.rng_ctr.zero_()
+        self.s_gmax.zero_()
+        self.s_pick.zero_()
 
     # ------------------------------------------------------------------
     # layer stack over rows [0, M) of the scratch buffers
@@ -275,13 +292,59 @@ class GPUModel:
     # fast device-side decode loop (graph-replayable)
     # ------------------------------------------------------------------
     def _decode_step(self, greedy: bool, min_p: float):
-        ho.embed(self.embed, self.next_token, self.b_h, 1,
-                 self.config.embed_scale)
-        self._layers_forward(1)
-        self._lm_head_last(1)
+        """Fused decode path: 4 kernels/layer (llama) or 6 (gemma) —
+        RMSNorm and GLU live inside the GEMV staging pass, RoPE + KV
+        write inside the attention kernel."""
+        cfg = self.config
+        eps = cfg.rms_norm_eps
+        h = self.b_h[0]
+        t1 = self.b_t1[0]
+        ho.embed(self.embed, self.next_token, self.b_h, 1, cfg.embed_scale)
+        for i, lw in enumerate(self.layers):
+            window = cfg.sliding_window if cfg.is_sliding(i) else 0
+            ho.gemv(lw["wqkv"], h, self.b_qkv, stage=ho.STAGE_NORM,
+                    g=lw["g_in"], eps=eps)
+            ho.attn_dec(self.b_qkv, self.k_cache[i], self.v_cache[i],
+                        self.b_att[0], self.len_buf, self.cos_t, self.sin_t,
+                        self.nh_l, self.kvh_l, self.hd, self.scale,
+                        softcap=self.attn_softcap, window=window or 0)
+            if self.gemma:
+                ho.gemv(lw["wo"], self.b_att[0], t1)
+                tpu.all_reduce(t1)
+                ho.rmsnorm(t1, lw["g_post"], h, res=h, eps=eps)
+                ho.gemv(lw["wgu"], h, self.b_gu, stage=ho.STAGE_NORM,
+                        g=lw["g_preffn"], eps=eps)
+                ho.gemv(lw["wdown"], self.b_gu[:self.inter_l], t1,
+                        stage=ho.STAGE_GLU, x2=self.b_gu[self.inter_l:],
+                        act=self.act)
+                tpu.all_reduce(t1)
+                ho.rmsnorm(t1, lw["g_postffn"], h, res=h, eps=eps)
+            else:
+                if self.world > 1:
+                    ho.gemv(lw["wo"], self.b_att[0], t1)
+                    tpu.all_reduce(t1)
+                    ho.addinto(h, t1)
+                else:
+                    ho.gemv(lw["wo"], self.b_att[0], h, res=h)
+                ho.gemv(lw["wgu"], h, self.b_gu, stage=ho.STAGE_NORM,
+                        g=lw["g_post"], eps=eps)
+                if self.world > 1:
+                    ho.gemv(lw["wdown"], self.b_gu[:self.inter_l], t1,
+                            stage=ho.STAGE_GLU, x2=self.b_gu[self.inter_l:],
+                            act=self.act)
+                    tpu.all_reduce(t1)
+                    ho.addinto(h, t1)
+                else:
+                    ho.gemv(lw["wdown"], self.b_gu[:self.inter_l], h,
+                            res=h, stage=ho.STAGE_GLU,
+                            x2=self.b_gu[self.inter_l:], act=self.act)
+        ho.gemv(self.lm_head, h, self.b_logits_l, softcap=self.final_softcap,
+                stage=ho.STAGE_NORM, g=self.g_final, eps=eps)
+        if self.world > 1:
+            tpu.all_gather_into(self.b_logits, self.b_logits_l)
         ho.sample(self.b_logits, min_p, greedy, self.seed, self.rng_ctr,
-                  self.next_token, self.out_ring, self.nout, self.len_buf,
-                  bump_len=True)
+                  self.s_gmax, self.s_pick, self.next_token, self.out_ring,
+                  self.nout, self.len_buf, bump_len=True)
 
     def capture_decode_graph(self, greedy: bool = True, min_p: float = 0.1):
         """Capture one decode step into a hipGraph (fixed shapes: KV pool
@@ -320,8 +383,9 @@ class GPUModel:
         if first_from_logits:
             # sample token 0 from the prefill logits
             ho.sample(self.b_logits, min_p, greedy, self.seed, self.rng_ctr,
-                      self.next_token, self.out_ring, self.nout,
-                      self.len_buf, bump_len=False)
+                      self.s_gmax, self.s_pick, self.next_token,
+                      self.out_ring, self.nout, self.len_buf,
+                      bump_len=False)
         n_steps = n_tokens - (1 if first_from_logits else 0)
         if use_graph and n_steps > 0:
             n_steps -= self.capture_decode_graph(greedy, min_p)

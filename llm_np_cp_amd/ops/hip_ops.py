@@ -24,8 +24,9 @@ _LIB_PATH = os.path.join(os.path.dirname(os.path.dirname(
     os.path.abspath(__file__))), "_libllmops.so")
 
 _SIGS = {
-    "launch_gemv_bf16": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 3 +
-                        [ctypes.c_float, ctypes.c_void_p],
+    "launch_gemv_bf16": [ctypes.c_void_p] * 6 + [ctypes.c_int] * 4 +
+                        [ctypes.c_float, ctypes.c_int, ctypes.c_float,
+                         ctypes.c_void_p],
     "launch_rmsnorm": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 2 +
                       [ctypes.c_float, ctypes.c_int, ctypes.c_void_p],
     "launch_rope_cache": [ctypes.c_void_p] * 8 + [ctypes.c_int] * 5 +
@@ -33,12 +34,15 @@ _SIGS = {
     "launch_attn": [ctypes.c_void_p] * 5 + [ctypes.c_int] * 5 +
                    [ctypes.c_float, ctypes.c_float, ctypes.c_int,
                     ctypes.c_void_p],
+    "launch_attn_dec": [ctypes.c_void_p] * 7 + [ctypes.c_int] * 4 +
+                       [ctypes.c_float, ctypes.c_float, ctypes.c_int,
+                        ctypes.c_void_p],
     "launch_glu": [ctypes.c_void_p] * 3 + [ctypes.c_long, ctypes.c_int,
                                            ctypes.c_void_p],
     "launch_embed": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 +
                     [ctypes.c_float, ctypes.c_void_p],
     "launch_sample": [ctypes.c_void_p, ctypes.c_int, ctypes.c_float,
-                      ctypes.c_int, ctypes.c_uint64] + [ctypes.c_void_p] * 5 +
+                      ctypes.c_int, ctypes.c_uint64] + [ctypes.c_void_p] * 7 +
                      [ctypes.c_int, ctypes.c_void_p],
     "launch_gemm_bf16": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 3 +
                         [ctypes.c_void_p],
@@ -81,13 +85,20 @@ def _ptr(t) -> int:
 # op wrappers (shapes validated here; kernels trust their args)
 # ----------------------------------------------------------------------
 
+STAGE_RAW, STAGE_NORM, STAGE_GLU = 0, 1, 2
+
+
 def gemv(W: torch.Tensor, x: torch.Tensor, y: torch.Tensor,
-         res: torch.Tensor | None = None, softcap: float = 0.0):
-    """y[N] = W[N,K] @ x[K] (+res), bf16 weights/input, bf16 or f32 out."""
+         res: torch.Tensor | None = None, softcap: float = 0.0,
+         stage: int = 0, x2: torch.Tensor | None = None,
+         g: torch.Tensor | None = None, act: int = 0, eps: float = 1e-5):
+    """y[N] = W[N,K] @ stage(x)[K] (+res); stage fuses RMSNorm or GLU
+    into the LDS staging pass (see csrc/llm_ops.hip)."""
     N, K = W.shape
     out_f32 = 1 if y.dtype == torch.float32 else 0
     _check(lib().launch_gemv_bf16(
-        _ptr(W), _ptr(x), _ptr(y), _ptr(res), N, K, out_f32,
+        _ptr(W), _ptr(x), _ptr(x2), _ptr(g), _ptr(y), _ptr(res), N, K,
+        stage, act, ctypes.c_float(eps), out_f32,
         ctypes.c_float(softcap), _stream()), "gemv")
 
 
@@ -124,6 +135,19 @@ def attn(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
         window, _stream()), "attn")
 
 
+def attn_dec(qkv: torch.Tensor, k_cache: torch.Tensor,
+             v_cache: torch.Tensor, out: torch.Tensor,
+             len_ptr: torch.Tensor, cos_t: torch.Tensor,
+             sin_t: torch.Tensor, nh: int, kvh: int, hd: int,
+             scale: float, softcap: float = 0.0, window: int = 0):
+    """Fused decode attention: RoPE(q,k) + KV write + online softmax."""
+    S = k_cache.shape[1]
+    _check(lib().launch_attn_dec(
+        _ptr(qkv), _ptr(k_cache), _ptr(v_cache), _ptr(out), _ptr(len_ptr),
+        _ptr(cos_t), _ptr(sin_t), nh, kvh, hd, S, ctypes.c_float(scale),
+        ctypes.c_float(softcap), window, _stream()), "attn_dec")
+
+
 def glu(gate: torch.Tensor, up: torch.Tensor, out: torch.Tensor, act: int):
     """out = act(gate) * up; act 0 = SiLU, 1 = tanh-GELU."""
     total = gate.numel()
@@ -140,14 +164,18 @@ def embed(table: torch.Tensor, ids: torch.Tensor, out: torch.Tensor,
 
 
 def sample(logits: torch.Tensor, min_p: float, greedy: bool, seed: int,
-           ctr: torch.Tensor, next_token: torch.Tensor,
-           out_ring: torch.Tensor, nout: torch.Tensor,
-           len_ptr: torch.Tensor, bump_len: bool = True):
+           ctr: torch.Tensor, gmax: torch.Tensor, pick: torch.Tensor,
+           next_token: torch.Tensor, out_ring: torch.Tensor,
+           nout: torch.Tensor, len_ptr: torch.Tensor,
+           bump_len: bool = True):
+    """min-p / greedy sampler: parallel max + Gumbel-argmax + commit.
+    gmax/pick are u64 scratch (zeroed once; the commit kernel re-zeros)."""
     V = logits.shape[-1]
     _check(lib().launch_sample(
         _ptr(logits), V, ctypes.c_float(min_p), 1 if greedy else 0,
-        ctypes.c_uint64(seed), _ptr(ctr), _ptr(next_token), _ptr(out_ring),
-        _ptr(nout), _ptr(len_ptr), 1 if bump_len else 0, _stream()), "sample")
+        ctypes.c_uint64(seed), _ptr(ctr), _ptr(gmax), _ptr(pick),
+        _ptr(next_token), _ptr(out_ring), _ptr(nout), _ptr(len_ptr),
+        1 if bump_len else 0, _stream()), "sample")
 
 
 def gemm(X: torch.Tensor, W: torch.Tensor, Y: torch.Tensor,

@@ -242,22 +242,101 @@ def test_sample_greedy_and_minp():
     logits[123] = 10.0
     logits[777] = 9.9
     ctr = torch.zeros(1, dtype=torch.int64, device=dev())
+    gmax = torch.zeros(1, dtype=torch.int64, device=dev())
+    pick = torch.zeros(1, dtype=torch.int64, device=dev())
     nt = torch.zeros(1, dtype=torch.int32, device=dev())
     ring = torch.zeros(64, dtype=torch.int32, device=dev())
     nout = torch.zeros(1, dtype=torch.int32, device=dev())
     ln = torch.zeros(1, dtype=torch.int32, device=dev())
 
-    ho.sample(logits, 0.1, True, 0, ctr, nt, ring, nout, ln, bump_len=True)
+    ho.sample(logits, 0.1, True, 0, ctr, gmax, pick, nt, ring, nout, ln,
+              bump_len=True)
     torch.cuda.synchronize()
     assert nt.item() == 123 and ring[0].item() == 123
     assert nout.item() == 1 and ln.item() == 1
+    assert pick.item() == 0 and gmax.item() == 0  # scratch reset
 
     # min-p: only 123/777 survive the 0.1*pmax cut; both should occur
     seen = set()
     for i in range(40):
-        ho.sample(logits, 0.1, False, 42, ctr, nt, ring, nout, ln,
-                  bump_len=False)
+        ho.sample(logits, 0.1, False, 42, ctr, gmax, pick, nt, ring, nout,
+                  ln, bump_len=False)
         torch.cuda.synchronize()
         seen.add(int(nt.item()))
     assert seen <= {123, 777}
     assert len(seen) == 2
+
+
+def test_gemv_fused_norm_stage():
+    """STAGE_NORM: y = W @ (rmsnorm(x)*g) in one kernel."""
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    N, K, eps = 512, 2048, 1e-5
+    W = randn_bf16(N, K, seed=70, scale=0.05)
+    x = randn_bf16(K, seed=71)
+    g = torch.randn(K, generator=torch.Generator().manual_seed(72)).to(dev())
+    y = torch.empty(N, dtype=torch.bfloat16, device=dev())
+    ho.gemv(W, x, y, stage=ho.STAGE_NORM, g=g, eps=eps)
+    torch.cuda.synchronize()
+    xf = x.float()
+    xn = (xf * torch.rsqrt(xf.pow(2).mean() + eps) * g).to(
+        torch.bfloat16).float()
+    ref = W.float() @ xn
+    assert_close(y, ref)
+
+
+def test_gemv_fused_glu_stage():
+    """STAGE_GLU: y = W @ (act(gate)*up) in one kernel (SiLU + GELU)."""
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    N, K = 256, 4096
+    W = randn_bf16(N, K, seed=80, scale=0.05)
+    gate = randn_bf16(K, seed=81)
+    up = randn_bf16(K, seed=82)
+    y = torch.empty(N, dtype=torch.bfloat16, device=dev())
+    for act, fn in [(0, lambda t: torch.nn.functional.silu(t)),
+                    (1, lambda t: torch.nn.functional.gelu(
+                        t, approximate="tanh"))]:
+        ho.gemv(W, gate, y, stage=ho.STAGE_GLU, x2=up, act=act)
+        torch.cuda.synchronize()
+        xs = (fn(gate.float()) * up.float()).to(torch.bfloat16).float()
+        ref = W.float() @ xs
+        assert_close(y, ref)
+
+
+def test_attn_dec_fused_matches_unfused():
+    """k_attn_dec (RoPE+cache+attn fused) vs rope_cache + attn chain."""
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    nh, kvh, hd, S, pos0 = 4, 2, 64, 128, 11
+    qkv = randn_bf16((nh + 2 * kvh) * hd, seed=90)
+    kc1 = torch.zeros(kvh, S, hd, dtype=torch.bfloat16, device=dev())
+    vc1 = torch.zeros_like(kc1)
+    kc1[:, :pos0] = randn_bf16(kvh, pos0, hd, seed=91)
+    vc1[:, :pos0] = randn_bf16(kvh, pos0, hd, seed=92)
+    kc2, vc2 = kc1.clone(), vc1.clone()
+
+    inv = 1.0 / (10000.0 ** (np.arange(0, hd, 2) / hd))
+    fr = np.outer(np.arange(S, dtype=np.float64), inv)
+    cos_t = torch.from_numpy(np.cos(fr).astype(np.float32)).to(dev())
+    sin_t = torch.from_numpy(np.sin(fr).astype(np.float32)).to(dev())
+    pos = torch.tensor([pos0], dtype=torch.int32, device=dev())
+
+    # fused
+    out1 = torch.empty(nh * hd, dtype=torch.bfloat16, device=dev())
+    ho.attn_dec(qkv, kc1, vc1, out1, pos, cos_t, sin_t, nh, kvh, hd,
+                hd ** -0.5)
+    torch.cuda.synchronize()
+
+    # unfused chain on a copy
+    q = qkv[:nh * hd].clone()
+    k = qkv[nh * hd:(nh + kvh) * hd].clone()
+    v = qkv[(nh + kvh) * hd:].clone()
+    ho.rope_cache(q, k, v, kc2, vc2, cos_t, sin_t, pos, 1, nh, kvh, hd)
+    out2 = torch.empty(nh * hd, dtype=torch.bfloat16, device=dev())
+    ho.attn(q, kc2, vc2, out2, pos, 1, nh, kvh, hd, hd ** -0.5)
+    torch.cuda.synchronize()
+
+    assert_close(out1, out2, rtol=2e-2, atol=2e-2)
+    assert_close(kc1[:, pos0], kc2[:, pos0], rtol=2e-2, atol=2e-2)
+    assert_close(vc1[:, pos0], vc2[:, pos0], rtol=2e-2, atol=2e-2)
