@@ -29,6 +29,8 @@ def main():
                          "gemma-2-9b at N>1)")
     ap.add_argument("--prompt-len", type=int, default=64)
     ap.add_argument("--max-seq", type=int, default=None)
+    ap.add_argument("--dtype", type=str, default="bf16",
+                    choices=["bf16", "fp8"])
     ap.add_argument("--no-graph", action="store_true")
     args = ap.parse_args()
 
@@ -53,7 +55,7 @@ def main():
         cfg.max_position_embeddings,
         args.prompt_len + args.steps + args.warmup + 64)
     w = random_weights(cfg, seed=0)
-    model = GPUModel(cfg, w, max_seq=max_seq, seed=0)
+    model = GPUModel(cfg, w, max_seq=max_seq, seed=0, dtype=args.dtype)
 
     rng = np.random.default_rng(0)
     prompt = rng.integers(0, cfg.vocab_size, size=args.prompt_len)
@@ -107,7 +109,7 @@ def main():
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": args.dtype,
             "data": "synthetic",
             "config": {
                 "model": model_name,

@@ -119,11 +119,9 @@ k_gemv_bf16(const u16* __restrict__ W, const u16* __restrict__ x,
       }
       *(s8v*)(xs + i) = *(s8v*)o;
     }
-  } else {
-    for (int i = threadIdx.x * 8; i < K; i += 256 * 8)
-      *(s8v*)(xs + i) = *(const s8v*)(x + i);
   }
-  __syncthreads();
+  if (stage != STAGE_RAW) __syncthreads();
+  const u16* xv = (stage == STAGE_RAW) ? x : xs;  // RAW: L2-hot, no barrier
 
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int row = blockIdx.x * GEMV_ROWS_PER_BLOCK + wave;
@@ -138,10 +136,10 @@ k_gemv_bf16(const u16* __restrict__ W, const u16* __restrict__ x,
     s8v w1 = *(const s8v*)(Wr + k + 512);
     s8v w2 = *(const s8v*)(Wr + k + 1024);
     s8v w3 = *(const s8v*)(Wr + k + 1536);
-    s8v x0 = *(const s8v*)(xs + k);
-    s8v x1 = *(const s8v*)(xs + k + 512);
-    s8v x2 = *(const s8v*)(xs + k + 1024);
-    s8v x3 = *(const s8v*)(xs + k + 1536);
+    s8v x0 = *(const s8v*)(xv + k);
+    s8v x1 = *(const s8v*)(xv + k + 512);
+    s8v x2 = *(const s8v*)(xv + k + 1024);
+    s8v x3 = *(const s8v*)(xv + k + 1536);
 #pragma unroll
     for (int j = 0; j < 8; j++) {
       a0 += b2f(((u16*)&w0)[j]) * b2f(((u16*)&x0)[j]);
@@ -152,7 +150,7 @@ k_gemv_bf16(const u16* __restrict__ W, const u16* __restrict__ x,
   }
   for (; k < K; k += 512) {
     s8v w0 = *(const s8v*)(Wr + k);
-    s8v x0 = *(const s8v*)(xs + k);
+    s8v x0 = *(const s8v*)(xv + k);
 #pragma unroll
     for (int j = 0; j < 8; j++)
       a0 += b2f(((u16*)&w0)[j]) * b2f(((u16*)&x0)[j]);
@@ -173,11 +171,153 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
                                        int out_f32, float softcap,
                                        hipStream_t stream) {
   int blocks = (N + GEMV_ROWS_PER_BLOCK - 1) / GEMV_ROWS_PER_BLOCK;
-  size_t lds = (size_t)K * 2 + 16;
+  size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 16);
   hipLaunchKernelGGL(k_gemv_bf16, dim3(blocks), dim3(256), lds, stream,
                      (const u16*)W, (const u16*)x, (const u16*)x2,
                      (const float*)g, y, (const u16*)res, N, K, stage, act,
                      eps, out_f32, softcap);
+  return hipGetLastError();
+}
+
+// ====================================================================
+// fp8 GEMV: y[N] = scale[n] * (W8[N,K] @ stage(x)[K]) (+res) (softcap)
+// W8 = OCP e4m3fn, per-output-row scales (absmax/448 quantization at
+// load).  Halves decode weight traffic vs bf16 -> ~2x decode ceiling.
+// Same fused staging modes as the bf16 GEMV; x stays bf16.
+// HW dequant: v_cvt_pk_f32_fp8 (gfx950, OCP not fnuz).
+// ====================================================================
+
+typedef float f2v __attribute__((ext_vector_type(2)));
+typedef uint32_t u4v __attribute__((ext_vector_type(4)));
+
+DEVINL void fp8x16_to_f32(u4v w, float* o) {
+#pragma unroll
+  for (int q = 0; q < 4; q++) {
+    f2v a = __builtin_amdgcn_cvt_pk_f32_fp8(w[q], false);
+    f2v b = __builtin_amdgcn_cvt_pk_f32_fp8(w[q], true);
+    o[q * 4 + 0] = a[0]; o[q * 4 + 1] = a[1];
+    o[q * 4 + 2] = b[0]; o[q * 4 + 3] = b[1];
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+k_gemv_fp8(const uint8_t* __restrict__ W, const float* __restrict__ scales,
+           const u16* __restrict__ x, const u16* __restrict__ x2,
+           const float* __restrict__ g, void* __restrict__ y,
+           const u16* __restrict__ res, int N, int K, int stage, int act,
+           float eps, int out_f32, float softcap) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  u16* xs = (u16*)smem;
+  if (stage == STAGE_NORM) {
+    float ss = 0.f;
+    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+      s8v v = *(const s8v*)(x + i);
+      *(s8v*)(xs + i) = v;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = b2f(((u16*)&v)[j]);
+        ss += f * f;
+      }
+    }
+    float* red = (float*)(smem + (size_t)K * 2);
+    ss = wave_reduce_sum(ss);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+    __syncthreads();
+    float rnorm = rsqrtf((red[0] + red[1] + red[2] + red[3]) / (float)K + eps);
+    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+      s8v v = *(s8v*)(xs + i);
+      u16 o[8];
+#pragma unroll
+      for (int j = 0; j < 8; j++)
+        o[j] = f2b(b2f(((u16*)&v)[j]) * rnorm * g[i + j]);
+      *(s8v*)(xs + i) = *(s8v*)o;
+    }
+  } else if (stage == STAGE_GLU) {
+    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+      s8v gv = *(const s8v*)(x + i);
+      s8v uv = *(const s8v*)(x2 + i);
+      u16 o[8];
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float xx = b2f(((u16*)&gv)[j]);
+        float a;
+        if (act == 0) {
+          a = xx / (1.f + __expf(-xx));
+        } else {
+          float c = 0.7978845608028654f * (xx + 0.044715f * xx * xx * xx);
+          a = 0.5f * xx * (1.f + tanhf(c));
+        }
+        o[j] = f2b(a * b2f(((u16*)&uv)[j]));
+      }
+      *(s8v*)(xs + i) = *(s8v*)o;
+    }
+  }
+  if (stage != STAGE_RAW) __syncthreads();
+  const u16* xv = (stage == STAGE_RAW) ? x : xs;
+
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int row = blockIdx.x * GEMV_ROWS_PER_BLOCK + wave;
+  if (row >= N) return;
+  const uint8_t* Wr = W + (size_t)row * K;
+
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  int k = lane * 16;
+  for (; k + 3072 + 16 <= K; k += 4096) {
+    u4v w0 = *(const u4v*)(Wr + k);
+    u4v w1 = *(const u4v*)(Wr + k + 1024);
+    u4v w2 = *(const u4v*)(Wr + k + 2048);
+    u4v w3 = *(const u4v*)(Wr + k + 3072);
+    s8v x0a = *(const s8v*)(xv + k), x0b = *(const s8v*)(xv + k + 8);
+    s8v x1a = *(const s8v*)(xv + k + 1024), x1b = *(const s8v*)(xv + k + 1032);
+    s8v x2a = *(const s8v*)(xv + k + 2048), x2b = *(const s8v*)(xv + k + 2056);
+    s8v x3a = *(const s8v*)(xv + k + 3072), x3b = *(const s8v*)(xv + k + 3080);
+    float f0[16], f1[16], f2[16], f3[16];
+    fp8x16_to_f32(w0, f0); fp8x16_to_f32(w1, f1);
+    fp8x16_to_f32(w2, f2); fp8x16_to_f32(w3, f3);
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      a0 += f0[j] * b2f(((u16*)&x0a)[j]);
+      a0 += f0[j + 8] * b2f(((u16*)&x0b)[j]);
+      a1 += f1[j] * b2f(((u16*)&x1a)[j]);
+      a1 += f1[j + 8] * b2f(((u16*)&x1b)[j]);
+      a2 += f2[j] * b2f(((u16*)&x2a)[j]);
+      a2 += f2[j + 8] * b2f(((u16*)&x2b)[j]);
+      a3 += f3[j] * b2f(((u16*)&x3a)[j]);
+      a3 += f3[j + 8] * b2f(((u16*)&x3b)[j]);
+    }
+  }
+  for (; k < K; k += 1024) {
+    u4v w0 = *(const u4v*)(Wr + k);
+    s8v xa = *(const s8v*)(xv + k), xb = *(const s8v*)(xv + k + 8);
+    float f0[16];
+    fp8x16_to_f32(w0, f0);
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      a0 += f0[j] * b2f(((u16*)&xa)[j]);
+      a0 += f0[j + 8] * b2f(((u16*)&xb)[j]);
+    }
+  }
+  float acc = wave_reduce_sum((a0 + a1) + (a2 + a3)) * scales[row];
+  if (lane == 0) {
+    if (softcap > 0.f) acc = softcap * tanhf(acc / softcap);
+    if (res) acc += b2f(res[row]);
+    if (out_f32) ((float*)y)[row] = acc;
+    else ((u16*)y)[row] = f2b(acc);
+  }
+}
+
+extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
+                                      const void* x, const void* x2,
+                                      const void* g, void* y, const void* res,
+                                      int N, int K, int stage, int act,
+                                      float eps, int out_f32, float softcap,
+                                      hipStream_t stream) {
+  int blocks = (N + GEMV_ROWS_PER_BLOCK - 1) / GEMV_ROWS_PER_BLOCK;
+  size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 16);
+  hipLaunchKernelGGL(k_gemv_fp8, dim3(blocks), dim3(256), lds, stream,
+                     (const uint8_t*)W, (const float*)scales, (const u16*)x,
+                     (const u16*)x2, (const float*)g, y, (const u16*)res, N,
+                     K, stage, act, eps, out_f32, softcap);
   return hipGetLastError();
 }
 
@@ -454,19 +594,30 @@ k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
   const u16* kh = qkv + (size_t)(nh + kvhead) * hd;
   const u16* vh = qkv + (size_t)(nh + kvh + kvhead) * hd;
   float qf[8], kn[8], vn[8];
+  {
+    // d0 is a multiple of 8, hd2 a multiple of 8: the whole 8-chunk sits
+    // on one side of the rotate-half split -> two 16B loads per tensor.
+    const bool lo = d0 < hd2;
+    const int dp = lo ? d0 + hd2 : d0 - hd2;   // partner chunk
+    const int ci = lo ? d0 : d0 - hd2;         // cos/sin index base
+    s8v qa = *(const s8v*)(qh + d0);
+    s8v qb = *(const s8v*)(qh + dp);
+    s8v ka = *(const s8v*)(kh + d0);
+    s8v kb = *(const s8v*)(kh + dp);
+    s8v va = *(const s8v*)(vh + d0);
+    f4v c0 = *(const f4v*)(cp + ci);
+    f4v c1 = *(const f4v*)(cp + ci + 4);
+    f4v s0 = *(const f4v*)(sp + ci);
+    f4v s1 = *(const f4v*)(sp + ci + 4);
+    const float sgn = lo ? -1.f : 1.f;
 #pragma unroll
-  for (int j = 0; j < 8; j++) {
-    int d = d0 + j;
-    if (d < hd2) {
-      float c = cp[d], s = sp[d];
-      qf[j] = b2f(qh[d]) * c - b2f(qh[d + hd2]) * s;
-      kn[j] = b2f(kh[d]) * c - b2f(kh[d + hd2]) * s;
-    } else {
-      float c = cp[d - hd2], s = sp[d - hd2];
-      qf[j] = b2f(qh[d]) * c + b2f(qh[d - hd2]) * s;
-      kn[j] = b2f(kh[d]) * c + b2f(kh[d - hd2]) * s;
+    for (int j = 0; j < 8; j++) {
+      float c = j < 4 ? c0[j] : c1[j - 4];
+      float s = j < 4 ? s0[j] : s1[j - 4];
+      qf[j] = b2f(((u16*)&qa)[j]) * c + sgn * b2f(((u16*)&qb)[j]) * s;
+      kn[j] = b2f(((u16*)&ka)[j]) * c + sgn * b2f(((u16*)&kb)[j]) * s;
+      vn[j] = b2f(((u16*)&va)[j]);
     }
-    vn[j] = b2f(vh[d0 + j]);
   }
   // persist this step's k/v once per kv-head (wave 0, one lane-group)
   if (h == kvhead * grp && wave == 0 && p == 0) {

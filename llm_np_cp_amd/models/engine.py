@@ -50,6 +50,10 @@ class GPUModel:
         self.config = config
         self.max_seq = max_seq
         self.seed = seed
+        self.dtype = dtype
+        if dtype not in ("bf16", "fp8"):
+            raise ValueError(f"unsupported dtype {dtype!r}")
+        self.fp8 = dtype == "fp8"
         self.rank, self.world = tpu.init_distributed()
         if device is None:
             device = f"cuda:{self.rank % max(torch.cuda.device_count(), 1)}"
@@ -88,6 +92,13 @@ class GPUModel:
         def gamma(a: np.ndarray) -> torch.Tensor:
             g = a.astype(np.float32) + (1.0 if gemma else 0.0)
             return torch.from_numpy(g).to(dev)
+
+        def quant8(a: np.ndarray):
+            """Per-output-row e4m3fn quantization (absmax/448)."""
+            t = torch.from_numpy(np.ascontiguousarray(a)).float()
+            s = t.abs().amax(dim=1).clamp_min(1e-8) / 448.0
+            q = (t / s[:, None]).to(torch.float8_e4m3fn).view(torch.uint8)
+            return q.to(dev), s.to(dev)
 
         self.embed = bf16(w["model.embed_tokens.weight"])
         lm_w = w.get("lm_head.weight", w["model.embed_tokens.weight"])
@@ -132,7 +143,24 @@ class GPUModel:
             if gemma:
                 lw["g_preffn"] = gamma(w[f"{p}.pre_feedforward_layernorm.weight"])
                 lw["g_postffn"] = gamma(w[f"{p}.post_feedforward_layernorm.weight"])
+            if self.fp8:
+                # decode streams fp8 weights; prefill keeps the bf16 copy
+                for name, arr in [
+                    ("wqkv", _np.concatenate([
+                        hrows(w[f"{a}.q_proj.weight"]),
+                        hrows(w[f"{a}.k_proj.weight"]),
+                        hrows(w[f"{a}.v_proj.weight"])], axis=0)),
+                    ("wgu", _np.concatenate([
+                        hrows(w[f"{p}.mlp.gate_proj.weight"]),
+                        hrows(w[f"{p}.mlp.up_proj.weight"])], axis=0)),
+                    ("wo", hcols(w[f"{a}.o_proj.weight"])),
+                    ("wdown", hcols(w[f"{p}.mlp.down_proj.weight"])),
+                ]:
+                    lw[name + "_q"], lw[name + "_s"] = quant8(arr)
             self.layers.append(lw)
+        if self.fp8:
+            lmw = lm_w if tp == 1 else tpu.shard_rows(lm_w, r, tp)
+            self.lm_head_q, self.lm_head_s = quant8(lmw)
 
     def _alloc_state(self, prefill_chunk: int):
         cfg, dev = self.config, self.device
@@ -291,6 +319,13 @@ class GPUModel:
     # ------------------------------------------------------------------
     # fast device-side decode loop (graph-replayable)
     # ------------------------------------------------------------------
+    def _dgemv(self, lw, name, x, y, **kw):
+        """Decode GEMV: fp8 weights when dtype=fp8, else bf16."""
+        if self.fp8:
+            ho.gemv_fp8(lw[name + "_q"], lw[name + "_s"], x, y, **kw)
+        else:
+            ho.gemv(lw[name], x, y, **kw)
+
     def _decode_step(self, greedy: bool, min_p: float):
         """Fused decode path: 4 kernels/layer (llama) or 6 (gemma) —
         RMSNorm and GLU live inside the GEMV staging pass, RoPE + KV
@@ -302,44 +337,49 @@ class GPUModel:
         ho.embed(self.embed, self.next_token, self.b_h, 1, cfg.embed_scale)
         for i, lw in enumerate(self.layers):
             window = cfg.sliding_window if cfg.is_sliding(i) else 0
-            ho.gemv(lw["wqkv"], h, self.b_qkv, stage=ho.STAGE_NORM,
-                    g=lw["g_in"], eps=eps)
+            self._dgemv(lw, "wqkv", h, self.b_qkv, stage=ho.STAGE_NORM,
+                        g=lw["g_in"], eps=eps)
             ho.attn_dec(self.b_qkv, self.k_cache[i], self.v_cache[i],
                         self.b_att[0], self.len_buf, self.cos_t, self.sin_t,
                         self.nh_l, self.kvh_l, self.hd, self.scale,
                         softcap=self.attn_softcap, window=window or 0)
             if self.gemma:
-                ho.gemv(lw["wo"], self.b_att[0], t1)
+                self._dgemv(lw, "wo", self.b_att[0], t1)
                 tpu.all_reduce(t1)
                 ho.rmsnorm(t1, lw["g_post"], h, res=h, eps=eps)
-                ho.gemv(lw["wgu"], h, self.b_gu, stage=ho.STAGE_NORM,
-                        g=lw["g_preffn"], eps=eps)
-                ho.gemv(lw["wdown"], self.b_gu[:self.inter_l], t1,
-                        stage=ho.STAGE_GLU, x2=self.b_gu[self.inter_l:],
-                        act=self.act)
+                self._dgemv(lw, "wgu", h, self.b_gu, stage=ho.STAGE_NORM,
+                            g=lw["g_preffn"], eps=eps)
+                self._dgemv(lw, "wdown", self.b_gu[:self.inter_l], t1,
+                            stage=ho.STAGE_GLU, x2=self.b_gu[self.inter_l:],
+                            act=self.act)
                 tpu.all_reduce(t1)
                 ho.rmsnorm(t1, lw["g_postffn"], h, res=h, eps=eps)
             else:
                 if self.world > 1:
-                    ho.gemv(lw["wo"], self.b_att[0], t1)
+                    self._dgemv(lw, "wo", self.b_att[0], t1)
                     tpu.all_reduce(t1)
                     ho.addinto(h, t1)
                 else:
-                    ho.gemv(lw["wo"], self.b_att[0], h, res=h)
-                ho.gemv(lw["wgu"], h, self.b_gu, stage=ho.STAGE_NORM,
-                        g=lw["g_post"], eps=eps)
+                    self._dgemv(lw, "wo", self.b_att[0], h, res=h)
+                self._dgemv(lw, "wgu", h, self.b_gu, stage=ho.STAGE_NORM,
+                            g=lw["g_post"], eps=eps)
                 if self.world > 1:
-                    ho.gemv(lw["wdown"], self.b_gu[:self.inter_l], t1,
-                            stage=ho.STAGE_GLU, x2=self.b_gu[self.inter_l:],
-                            act=self.act)
+                    self._dgemv(lw, "wdown", self.b_gu[:self.inter_l], t1,
+                                stage=ho.STAGE_GLU,
+                                x2=self.b_gu[self.inter_l:], act=self.act)
                     tpu.all_reduce(t1)
                     ho.addinto(h, t1)
                 else:
-                    ho.gemv(lw["wdown"], self.b_gu[:self.inter_l], h,
-                            res=h, stage=ho.STAGE_GLU,
-                            x2=self.b_gu[self.inter_l:], act=self.act)
-        ho.gemv(self.lm_head, h, self.b_logits_l, softcap=self.final_softcap,
-                stage=ho.STAGE_NORM, g=self.g_final, eps=eps)
+                    self._dgemv(lw, "wdown", self.b_gu[:self.inter_l], h,
+                                res=h, stage=ho.STAGE_GLU,
+                                x2=self.b_gu[self.inter_l:], act=self.act)
+        ho.rmsnorm(h, self.g_final, self.b_xn[0], eps=eps)
+        if self.fp8:
+            ho.gemv_fp8(self.lm_head_q, self.lm_head_s, self.b_xn[0],
+                        self.b_logits_l, softcap=self.final_softcap)
+        else:
+            ho.gemv(self.lm_head, self.b_xn[0], self.b_logits_l,
+                    softcap=self.final_softcap)
         if self.world > 1:
             tpu.all_gather_into(self.b_logits, self.b_logits_l)
         ho.sample(self.b_logits, min_p, greedy, self.seed, self.rng_ctr,

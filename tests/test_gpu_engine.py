@@ -120,3 +120,44 @@ def test_min_p_decode_stays_in_support():
     probs /= probs.sum()
     support = set(np.where(probs >= 0.05 * probs.max())[0].tolist())
     assert int(ids[-1]) in support
+
+
+def test_fp8_engine_decode_close_to_bf16():
+    """fp8-weight decode must track the bf16 engine closely on logits and
+    produce a plausible greedy rollout (per-channel e4m3 quantization)."""
+    import llm_np_cp_amd as L
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.engine import GPUModel
+
+    cfg = L.preset_config("tiny-llama")
+    w = random_weights(cfg, seed=9)
+    bf = GPUModel(cfg, w, max_seq=128)
+    f8 = GPUModel(cfg, dict(w), max_seq=128, dtype="fp8")
+    prompt = np.arange(1, 9)
+
+    _, logits_bf = bf.prefill(prompt)
+    _, logits_f8 = f8.prefill(prompt)
+    # prefill runs bf16 weights in both engines -> identical path
+    np.testing.assert_allclose(logits_f8, logits_bf, rtol=1e-3, atol=1e-3)
+
+    ids_bf = bf.decode(6, greedy=True, use_graph=False)
+    f8.prefill(prompt)
+    ids_f8 = f8.decode(6, greedy=True, use_graph=False)
+    # fp8 rollout may diverge late; first tokens should agree on tiny cfg
+    assert ids_f8[0] == ids_bf[0]
+
+
+def test_fp8_graph_decode_matches_eager():
+    import llm_np_cp_amd as L
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.engine import GPUModel
+
+    cfg = L.preset_config("tiny-llama")
+    w = random_weights(cfg, seed=10)
+    m = GPUModel(cfg, w, max_seq=128, dtype="fp8")
+    prompt = np.arange(1, 9)
+    m.prefill(prompt)
+    a = m.decode(8, greedy=True, use_graph=False)
+    m.prefill(prompt)
+    b = m.decode(8, greedy=True, use_graph=True)
+    np.testing.assert_array_equal(a, b)

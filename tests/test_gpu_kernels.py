@@ -340,3 +340,48 @@ def test_attn_dec_fused_matches_unfused():
     assert_close(out1, out2, rtol=2e-2, atol=2e-2)
     assert_close(kc1[:, pos0], kc2[:, pos0], rtol=2e-2, atol=2e-2)
     assert_close(vc1[:, pos0], vc2[:, pos0], rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.parametrize("N,K", [(2048, 2048), (512, 2304), (1024, 8192)])
+def test_gemv_fp8(N, K):
+    """fp8 GEMV vs fp32 reference of the SAME quantized weights."""
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    W = randn_bf16(N, K, seed=100, scale=0.05).float().cpu()
+    s = W.abs().amax(dim=1).clamp_min(1e-8) / 448.0
+    q = (W / s[:, None]).to(torch.float8_e4m3fn)
+    Wq = q.view(torch.uint8).to(dev())
+    sc = s.to(dev())
+    x = randn_bf16(K, seed=101)
+    y = torch.empty(N, dtype=torch.bfloat16, device=dev())
+    ho.gemv_fp8(Wq, sc, x, y)
+    torch.cuda.synchronize()
+    ref = (q.float() * s[:, None]).to(dev()) @ x.float()
+    assert_close(y, ref, rtol=2e-2, atol=2e-2)
+
+
+def test_gemv_fp8_fused_stages():
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    N, K, eps = 512, 2048, 1e-5
+    W = randn_bf16(N, K, seed=110, scale=0.05).float().cpu()
+    s = W.abs().amax(dim=1).clamp_min(1e-8) / 448.0
+    q = (W / s[:, None]).to(torch.float8_e4m3fn)
+    Wq, sc = q.view(torch.uint8).to(dev()), s.to(dev())
+    Wd = (q.float() * s[:, None]).to(dev())
+
+    x = randn_bf16(K, seed=111)
+    g = torch.randn(K, generator=torch.Generator().manual_seed(112)).to(dev())
+    y = torch.empty(N, dtype=torch.bfloat16, device=dev())
+    ho.gemv_fp8(Wq, sc, x, y, stage=ho.STAGE_NORM, g=g, eps=eps)
+    torch.cuda.synchronize()
+    xf = x.float()
+    xn = (xf * torch.rsqrt(xf.pow(2).mean() + eps) * g).to(
+        torch.bfloat16).float()
+    assert_close(y, Wd @ xn)
+
+    up = randn_bf16(K, seed=113)
+    ho.gemv_fp8(Wq, sc, x, y, stage=ho.STAGE_GLU, x2=up, act=0)
+    torch.cuda.synchronize()
+    xs = (torch.nn.functional.silu(xf) * up.float()).to(torch.bfloat16).float()
+    assert_close(y, Wd @ xs)
