@@ -68,13 +68,10 @@ DEVINL float wave_reduce_sum(float v) {
 #define STAGE_NORM 1
 #define STAGE_GLU 2
 
-extern "C" __global__ void __launch_bounds__(256)
-k_gemv_bf16(const u16* __restrict__ W, const u16* __restrict__ x,
-            const u16* __restrict__ x2, const float* __restrict__ g,
-            void* __restrict__ y, const u16* __restrict__ res,
-            int N, int K, int stage, int act, float eps, int out_f32,
-            float softcap) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
+// staging helper shared by bf16/fp8 GEMV (RMSNorm / GLU fused pre-ops)
+DEVINL const u16* gemv_stage(char* smem, const u16* x, const u16* x2,
+                             const float* g, int K, int stage, int act,
+                             float eps) {
   u16* xs = (u16*)smem;
   if (stage == STAGE_NORM) {
     float ss = 0.f;
@@ -121,47 +118,89 @@ k_gemv_bf16(const u16* __restrict__ W, const u16* __restrict__ x,
     }
   }
   if (stage != STAGE_RAW) __syncthreads();
-  const u16* xv = (stage == STAGE_RAW) ? x : xs;  // RAW: L2-hot, no barrier
+  return (stage == STAGE_RAW) ? x : xs;  // RAW: L2-hot, no barrier
+}
+
+DEVINL void gemv_epilogue(float acc, int row, void* y, const u16* res,
+                          int out_f32, float softcap) {
+  if (softcap > 0.f) acc = softcap * tanhf(acc / softcap);
+  if (res) acc += b2f(res[row]);
+  if (out_f32) ((float*)y)[row] = acc;
+  else ((u16*)y)[row] = f2b(acc);
+}
+
+// NT: non-temporal weight loads (read-once stream, keep L2 for x/KV —
+// MI355X_MICROARCH "nt-weights").  RPW: rows per wave (ILP: RPW x 4
+// weight loads in flight).
+template <bool NT, int RPW>
+__global__ void __launch_bounds__(256)
+k_gemv_bf16_t(const u16* __restrict__ W, const u16* __restrict__ x,
+              const u16* __restrict__ x2, const float* __restrict__ g,
+              void* __restrict__ y, const u16* __restrict__ res,
+              int N, int K, int stage, int act, float eps, int out_f32,
+              float softcap) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const u16* xv = gemv_stage(smem, x, x2, g, K, stage, act, eps);
 
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int row = blockIdx.x * GEMV_ROWS_PER_BLOCK + wave;
-  if (row >= N) return;
-  const u16* Wr = W + (size_t)row * K;
-
-  // 4 independent accumulators; 4 W loads in flight per iteration.
-  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  const int rstride = gridDim.x * 4 * RPW;  // grid-stride rows: amortize
+  for (int row0 = (blockIdx.x * 4 + wave) * RPW; row0 < N;
+       row0 += rstride) {                    // wave setup over many rows
+  const u16* Wr[RPW];
+  float a0[RPW], a1[RPW], a2[RPW], a3[RPW];
+#pragma unroll
+  for (int r = 0; r < RPW; r++) {
+    int rr = row0 + r < N ? row0 + r : N - 1;
+    Wr[r] = W + (size_t)rr * K;
+    a0[r] = a1[r] = a2[r] = a3[r] = 0.f;
+  }
   int k = lane * 8;
   for (; k + 1536 < K; k += 2048) {
-    s8v w0 = *(const s8v*)(Wr + k);
-    s8v w1 = *(const s8v*)(Wr + k + 512);
-    s8v w2 = *(const s8v*)(Wr + k + 1024);
-    s8v w3 = *(const s8v*)(Wr + k + 1536);
     s8v x0 = *(const s8v*)(xv + k);
     s8v x1 = *(const s8v*)(xv + k + 512);
-    s8v x2 = *(const s8v*)(xv + k + 1024);
+    s8v x2v = *(const s8v*)(xv + k + 1024);
     s8v x3 = *(const s8v*)(xv + k + 1536);
 #pragma unroll
-    for (int j = 0; j < 8; j++) {
-      a0 += b2f(((u16*)&w0)[j]) * b2f(((u16*)&x0)[j]);
-      a1 += b2f(((u16*)&w1)[j]) * b2f(((u16*)&x1)[j]);
-      a2 += b2f(((u16*)&w2)[j]) * b2f(((u16*)&x2)[j]);
-      a3 += b2f(((u16*)&w3)[j]) * b2f(((u16*)&x3)[j]);
+    for (int r = 0; r < RPW; r++) {
+      s8v w0, w1, w2, w3;
+      if (NT) {
+        w0 = __builtin_nontemporal_load((const s8v*)(Wr[r] + k));
+        w1 = __builtin_nontemporal_load((const s8v*)(Wr[r] + k + 512));
+        w2 = __builtin_nontemporal_load((const s8v*)(Wr[r] + k + 1024));
+        w3 = __builtin_nontemporal_load((const s8v*)(Wr[r] + k + 1536));
+      } else {
+        w0 = *(const s8v*)(Wr[r] + k);
+        w1 = *(const s8v*)(Wr[r] + k + 512);
+        w2 = *(const s8v*)(Wr[r] + k + 1024);
+        w3 = *(const s8v*)(Wr[r] + k + 1536);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        a0[r] += b2f(((u16*)&w0)[j]) * b2f(((u16*)&x0)[j]);
+        a1[r] += b2f(((u16*)&w1)[j]) * b2f(((u16*)&x1)[j]);
+        a2[r] += b2f(((u16*)&w2)[j]) * b2f(((u16*)&x2v)[j]);
+        a3[r] += b2f(((u16*)&w3)[j]) * b2f(((u16*)&x3)[j]);
+      }
     }
   }
   for (; k < K; k += 512) {
-    s8v w0 = *(const s8v*)(Wr + k);
     s8v x0 = *(const s8v*)(xv + k);
 #pragma unroll
-    for (int j = 0; j < 8; j++)
-      a0 += b2f(((u16*)&w0)[j]) * b2f(((u16*)&x0)[j]);
+    for (int r = 0; r < RPW; r++) {
+      s8v w0 = NT ? __builtin_nontemporal_load((const s8v*)(Wr[r] + k))
+                  : *(const s8v*)(Wr[r] + k);
+#pragma unroll
+      for (int j = 0; j < 8; j++)
+        a0[r] += b2f(((u16*)&w0)[j]) * b2f(((u16*)&x0)[j]);
+    }
   }
-  float acc = wave_reduce_sum((a0 + a1) + (a2 + a3));
-  if (lane == 0) {
-    if (softcap > 0.f) acc = softcap * tanhf(acc / softcap);
-    if (res) acc += b2f(res[row]);
-    if (out_f32) ((float*)y)[row] = acc;
-    else ((u16*)y)[row] = f2b(acc);
+#pragma unroll
+  for (int r = 0; r < RPW; r++) {
+    float acc = wave_reduce_sum((a0[r] + a1[r]) + (a2[r] + a3[r]));
+    if (lane == 0 && row0 + r < N)
+      gemv_epilogue(acc, row0 + r, y, res, out_f32, softcap);
   }
+  }  // row0 grid-stride loop
 }
 
 extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
@@ -169,13 +208,20 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
                                        const void* res, int N, int K,
                                        int stage, int act, float eps,
                                        int out_f32, float softcap,
-                                       hipStream_t stream) {
-  int blocks = (N + GEMV_ROWS_PER_BLOCK - 1) / GEMV_ROWS_PER_BLOCK;
+                                       int nt, int rpw, hipStream_t stream) {
   size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 16);
-  hipLaunchKernelGGL(k_gemv_bf16, dim3(blocks), dim3(256), lds, stream,
-                     (const u16*)W, (const u16*)x, (const u16*)x2,
-                     (const float*)g, y, (const u16*)res, N, K, stage, act,
-                     eps, out_f32, softcap);
+  int blocks = (N + 4 * rpw - 1) / (4 * rpw);
+  if (blocks > 1024) blocks = 1024;  // grid-stride the rest (Guideline 11)
+#define GEMV_CASE(NTV, RPWV)                                                 \
+  hipLaunchKernelGGL((k_gemv_bf16_t<NTV, RPWV>), dim3(blocks), dim3(256),    \
+                     lds, stream, (const u16*)W, (const u16*)x,              \
+                     (const u16*)x2, (const float*)g, y, (const u16*)res, N, \
+                     K, stage, act, eps, out_f32, softcap)
+  if (nt && rpw == 2) GEMV_CASE(true, 2);
+  else if (nt) GEMV_CASE(true, 1);
+  else if (rpw == 2) GEMV_CASE(false, 2);
+  else GEMV_CASE(false, 1);
+#undef GEMV_CASE
   return hipGetLastError();
 }
 
@@ -200,110 +246,73 @@ DEVINL void fp8x16_to_f32(u4v w, float* o) {
   }
 }
 
-extern "C" __global__ void __launch_bounds__(256)
-k_gemv_fp8(const uint8_t* __restrict__ W, const float* __restrict__ scales,
-           const u16* __restrict__ x, const u16* __restrict__ x2,
-           const float* __restrict__ g, void* __restrict__ y,
-           const u16* __restrict__ res, int N, int K, int stage, int act,
-           float eps, int out_f32, float softcap) {
+template <bool NT, int RPW>
+__global__ void __launch_bounds__(256)
+k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
+             const u16* __restrict__ x, const u16* __restrict__ x2,
+             const float* __restrict__ g, void* __restrict__ y,
+             const u16* __restrict__ res, int N, int K, int stage, int act,
+             float eps, int out_f32, float softcap) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  u16* xs = (u16*)smem;
-  if (stage == STAGE_NORM) {
-    float ss = 0.f;
-    for (int i = threadIdx.x * 8; i < K; i += 2048) {
-      s8v v = *(const s8v*)(x + i);
-      *(s8v*)(xs + i) = v;
-#pragma unroll
-      for (int j = 0; j < 8; j++) {
-        float f = b2f(((u16*)&v)[j]);
-        ss += f * f;
-      }
-    }
-    float* red = (float*)(smem + (size_t)K * 2);
-    ss = wave_reduce_sum(ss);
-    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
-    __syncthreads();
-    float rnorm = rsqrtf((red[0] + red[1] + red[2] + red[3]) / (float)K + eps);
-    for (int i = threadIdx.x * 8; i < K; i += 2048) {
-      s8v v = *(s8v*)(xs + i);
-      u16 o[8];
-#pragma unroll
-      for (int j = 0; j < 8; j++)
-        o[j] = f2b(b2f(((u16*)&v)[j]) * rnorm * g[i + j]);
-      *(s8v*)(xs + i) = *(s8v*)o;
-    }
-  } else if (stage == STAGE_GLU) {
-    for (int i = threadIdx.x * 8; i < K; i += 2048) {
-      s8v gv = *(const s8v*)(x + i);
-      s8v uv = *(const s8v*)(x2 + i);
-      u16 o[8];
-#pragma unroll
-      for (int j = 0; j < 8; j++) {
-        float xx = b2f(((u16*)&gv)[j]);
-        float a;
-        if (act == 0) {
-          a = xx / (1.f + __expf(-xx));
-        } else {
-          float c = 0.7978845608028654f * (xx + 0.044715f * xx * xx * xx);
-          a = 0.5f * xx * (1.f + tanhf(c));
-        }
-        o[j] = f2b(a * b2f(((u16*)&uv)[j]));
-      }
-      *(s8v*)(xs + i) = *(s8v*)o;
-    }
-  }
-  if (stage != STAGE_RAW) __syncthreads();
-  const u16* xv = (stage == STAGE_RAW) ? x : xs;
+  const u16* xv = gemv_stage(smem, x, x2, g, K, stage, act, eps);
 
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int row = blockIdx.x * GEMV_ROWS_PER_BLOCK + wave;
-  if (row >= N) return;
-  const uint8_t* Wr = W + (size_t)row * K;
-
-  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  const int rstride = gridDim.x * 4 * RPW;
+  for (int row0 = (blockIdx.x * 4 + wave) * RPW; row0 < N;
+       row0 += rstride) {
+  const uint8_t* Wr[RPW];
+  float a0[RPW], a1[RPW];
+#pragma unroll
+  for (int r = 0; r < RPW; r++) {
+    int rr = row0 + r < N ? row0 + r : N - 1;
+    Wr[r] = W + (size_t)rr * K;
+    a0[r] = a1[r] = 0.f;
+  }
   int k = lane * 16;
-  for (; k + 3072 + 16 <= K; k += 4096) {
-    u4v w0 = *(const u4v*)(Wr + k);
-    u4v w1 = *(const u4v*)(Wr + k + 1024);
-    u4v w2 = *(const u4v*)(Wr + k + 2048);
-    u4v w3 = *(const u4v*)(Wr + k + 3072);
+  for (; k + 1024 + 16 <= K; k += 2048) {
     s8v x0a = *(const s8v*)(xv + k), x0b = *(const s8v*)(xv + k + 8);
     s8v x1a = *(const s8v*)(xv + k + 1024), x1b = *(const s8v*)(xv + k + 1032);
-    s8v x2a = *(const s8v*)(xv + k + 2048), x2b = *(const s8v*)(xv + k + 2056);
-    s8v x3a = *(const s8v*)(xv + k + 3072), x3b = *(const s8v*)(xv + k + 3080);
-    float f0[16], f1[16], f2[16], f3[16];
-    fp8x16_to_f32(w0, f0); fp8x16_to_f32(w1, f1);
-    fp8x16_to_f32(w2, f2); fp8x16_to_f32(w3, f3);
 #pragma unroll
-    for (int j = 0; j < 8; j++) {
-      a0 += f0[j] * b2f(((u16*)&x0a)[j]);
-      a0 += f0[j + 8] * b2f(((u16*)&x0b)[j]);
-      a1 += f1[j] * b2f(((u16*)&x1a)[j]);
-      a1 += f1[j + 8] * b2f(((u16*)&x1b)[j]);
-      a2 += f2[j] * b2f(((u16*)&x2a)[j]);
-      a2 += f2[j + 8] * b2f(((u16*)&x2b)[j]);
-      a3 += f3[j] * b2f(((u16*)&x3a)[j]);
-      a3 += f3[j + 8] * b2f(((u16*)&x3b)[j]);
+    for (int r = 0; r < RPW; r++) {
+      u4v w0 = NT ? __builtin_nontemporal_load((const u4v*)(Wr[r] + k))
+                  : *(const u4v*)(Wr[r] + k);
+      u4v w1 = NT ? __builtin_nontemporal_load((const u4v*)(Wr[r] + k + 1024))
+                  : *(const u4v*)(Wr[r] + k + 1024);
+      float f0[16], f1[16];
+      fp8x16_to_f32(w0, f0);
+      fp8x16_to_f32(w1, f1);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        a0[r] += f0[j] * b2f(((u16*)&x0a)[j]);
+        a0[r] += f0[j + 8] * b2f(((u16*)&x0b)[j]);
+        a1[r] += f1[j] * b2f(((u16*)&x1a)[j]);
+        a1[r] += f1[j + 8] * b2f(((u16*)&x1b)[j]);
+      }
     }
   }
   for (; k < K; k += 1024) {
-    u4v w0 = *(const u4v*)(Wr + k);
     s8v xa = *(const s8v*)(xv + k), xb = *(const s8v*)(xv + k + 8);
-    float f0[16];
-    fp8x16_to_f32(w0, f0);
 #pragma unroll
-    for (int j = 0; j < 8; j++) {
-      a0 += f0[j] * b2f(((u16*)&xa)[j]);
-      a0 += f0[j + 8] * b2f(((u16*)&xb)[j]);
+    for (int r = 0; r < RPW; r++) {
+      u4v w0 = NT ? __builtin_nontemporal_load((const u4v*)(Wr[r] + k))
+                  : *(const u4v*)(Wr[r] + k);
+      float f0[16];
+      fp8x16_to_f32(w0, f0);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        a0[r] += f0[j] * b2f(((u16*)&xa)[j]);
+        a0[r] += f0[j + 8] * b2f(((u16*)&xb)[j]);
+      }
     }
   }
-  float acc = wave_reduce_sum((a0 + a1) + (a2 + a3)) * scales[row];
-  if (lane == 0) {
-    if (softcap > 0.f) acc = softcap * tanhf(acc / softcap);
-    if (res) acc += b2f(res[row]);
-    if (out_f32) ((float*)y)[row] = acc;
-    else ((u16*)y)[row] = f2b(acc);
+#pragma unroll
+  for (int r = 0; r < RPW; r++) {
+    int rr = row0 + r < N ? row0 + r : N - 1;
+    float acc = wave_reduce_sum(a0[r] + a1[r]) * scales[rr];
+    if (lane == 0 && row0 + r < N)
+      gemv_epilogue(acc, row0 + r, y, res, out_f32, softcap);
   }
+  }  // row0 grid-stride loop
 }
 
 extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
@@ -311,13 +320,21 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
                                       const void* g, void* y, const void* res,
                                       int N, int K, int stage, int act,
                                       float eps, int out_f32, float softcap,
-                                      hipStream_t stream) {
-  int blocks = (N + GEMV_ROWS_PER_BLOCK - 1) / GEMV_ROWS_PER_BLOCK;
+                                      int nt, int rpw, hipStream_t stream) {
   size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 16);
-  hipLaunchKernelGGL(k_gemv_fp8, dim3(blocks), dim3(256), lds, stream,
-                     (const uint8_t*)W, (const float*)scales, (const u16*)x,
-                     (const u16*)x2, (const float*)g, y, (const u16*)res, N,
-                     K, stage, act, eps, out_f32, softcap);
+  int blocks = (N + 4 * rpw - 1) / (4 * rpw);
+  if (blocks > 1024) blocks = 1024;
+#define GEMV8_CASE(NTV, RPWV)                                               \
+  hipLaunchKernelGGL((k_gemv_fp8_t<NTV, RPWV>), dim3(blocks), dim3(256),    \
+                     lds, stream, (const uint8_t*)W, (const float*)scales,  \
+                     (const u16*)x, (const u16*)x2, (const float*)g, y,     \
+                     (const u16*)res, N, K, stage, act, eps, out_f32,       \
+                     softcap)
+  if (nt && rpw == 2) GEMV8_CASE(true, 2);
+  else if (nt) GEMV8_CASE(true, 1);
+  else if (rpw == 2) GEMV8_CASE(false, 2);
+  else GEMV8_CASE(false, 1);
+#undef GEMV8_CASE
   return hipGetLastError();
 }
 
@@ -557,29 +574,40 @@ k_attn(const u16* __restrict__ q, const u16* __restrict__ kc,
 
 // ====================================================================
 // Fused decode attention (M=1): RoPE(q,k) + KV-pool write + online-
-// softmax GQA in ONE kernel.  Input is the raw fused-QKV GEMV output
-// [nh*hd | kvh*hd | kvh*hd]; each head-block rotates its own q head and
-// (redundantly, cheaply) its kv-head's new k/v — so no block depends on
-// this step's cache write; the first block of each kv-head group
-// persists k/v to the pool for future steps.
+// softmax GQA, split over the KV range.  grid = (nh, SPLIT): each block
+// scans one chunk of history; partial (m, l, acc) results are combined
+// by the LAST-arriving block of each head via the agent-scope
+// release/acquire + ticket recipe (guide §6 G16) — placement-independent
+// and graph-replay-safe (the merger resets its head's counter).
+// The SPLIT-1 chunk also handles the current token from registers and
+// one block per kv-head persists the new k/v to the pool.
+// Gemma-2 semantics included: sliding window + attn-logit softcap.
 // ====================================================================
 
 extern "C" __global__ void __launch_bounds__(256)
 k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
            u16* __restrict__ vc, u16* __restrict__ out,
            const int* __restrict__ len_ptr, const float* __restrict__ cost,
-           const float* __restrict__ sint, int nh, int kvh, int hd, int S,
+           const float* __restrict__ sint, float* __restrict__ scratch,
+           int* __restrict__ cnt, int nh, int kvh, int hd, int S,
            float scale, float softcap, int window) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* red = (float*)smem;
+  float* red = (float*)smem;              // [4][hd] + [4][2] + flag
+  int* lastflag = (int*)(red + 4 * hd + 8);
 
   const int h = blockIdx.x;
+  const int chunk = blockIdx.y, SPLIT = gridDim.y;
   const int grp = nh / kvh;
   const int kvhead = h / grp;
   const int pos = *len_ptr;
-  const int T = pos + 1;
   int start = 0;
-  if (window > 0 && T - window > 0) start = T - window;
+  if (window > 0 && pos + 1 - window > 0) start = pos + 1 - window;
+  // history chunk [c0, c1) of [start, pos)
+  const int hist = pos - start;
+  const int clen = (hist + SPLIT - 1) / SPLIT;
+  const int c0 = start + chunk * clen;
+  const int c1 = min(c0 + clen, pos);
+  const bool last_chunk = (chunk == SPLIT - 1);
 
   const int LP = hd / 8;
   const int PP = 64 / LP;
@@ -589,38 +617,34 @@ k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
   const float* cp = cost + (size_t)pos * hd2;
   const float* sp = sint + (size_t)pos * hd2;
 
-  // rotated q fragment for this head at dims [d0, d0+8)
   const u16* qh = qkv + (size_t)h * hd;
   const u16* kh = qkv + (size_t)(nh + kvhead) * hd;
   const u16* vh = qkv + (size_t)(nh + kvh + kvhead) * hd;
   float qf[8], kn[8], vn[8];
   {
-    // d0 is a multiple of 8, hd2 a multiple of 8: the whole 8-chunk sits
-    // on one side of the rotate-half split -> two 16B loads per tensor.
     const bool lo = d0 < hd2;
-    const int dp = lo ? d0 + hd2 : d0 - hd2;   // partner chunk
-    const int ci = lo ? d0 : d0 - hd2;         // cos/sin index base
+    const int dp = lo ? d0 + hd2 : d0 - hd2;
+    const int ci = lo ? d0 : d0 - hd2;
     s8v qa = *(const s8v*)(qh + d0);
     s8v qb = *(const s8v*)(qh + dp);
     s8v ka = *(const s8v*)(kh + d0);
     s8v kb = *(const s8v*)(kh + dp);
     s8v va = *(const s8v*)(vh + d0);
-    f4v c0 = *(const f4v*)(cp + ci);
-    f4v c1 = *(const f4v*)(cp + ci + 4);
-    f4v s0 = *(const f4v*)(sp + ci);
-    f4v s1 = *(const f4v*)(sp + ci + 4);
+    f4v c0v = *(const f4v*)(cp + ci);
+    f4v c1v = *(const f4v*)(cp + ci + 4);
+    f4v s0v = *(const f4v*)(sp + ci);
+    f4v s1v = *(const f4v*)(sp + ci + 4);
     const float sgn = lo ? -1.f : 1.f;
 #pragma unroll
     for (int j = 0; j < 8; j++) {
-      float c = j < 4 ? c0[j] : c1[j - 4];
-      float s = j < 4 ? s0[j] : s1[j - 4];
+      float c = j < 4 ? c0v[j] : c1v[j - 4];
+      float s = j < 4 ? s0v[j] : s1v[j - 4];
       qf[j] = b2f(((u16*)&qa)[j]) * c + sgn * b2f(((u16*)&qb)[j]) * s;
       kn[j] = b2f(((u16*)&ka)[j]) * c + sgn * b2f(((u16*)&kb)[j]) * s;
       vn[j] = b2f(((u16*)&va)[j]);
     }
   }
-  // persist this step's k/v once per kv-head (wave 0, one lane-group)
-  if (h == kvhead * grp && wave == 0 && p == 0) {
+  if (last_chunk && h == kvhead * grp && wave == 0 && p == 0) {
     u16 ko[8], vo[8];
 #pragma unroll
     for (int j = 0; j < 8; j++) { ko[j] = f2b(kn[j]); vo[j] = f2b(vn[j]); }
@@ -635,11 +659,10 @@ k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
 #pragma unroll
   for (int j = 0; j < 8; j++) acc[j] = 0.f;
 
-  // history [start, pos) from the pool
-  for (int t0 = start + wave * PP; t0 < pos; t0 += 4 * PP) {
+  for (int t0 = c0 + wave * PP; t0 < c1; t0 += 4 * PP) {
     int t = t0 + p;
-    bool valid = t < pos;
-    int tl = valid ? t : (pos > 0 ? pos - 1 : 0);
+    bool valid = t < c1;
+    int tl = valid ? t : c0;
     s8v kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
     float partial = 0.f;
 #pragma unroll
@@ -660,8 +683,8 @@ k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
     mrun = mnew;
   }
 
-  // current position from registers (wave 0, p == 0 lanes only)
-  if (wave == 0 && p == 0) {
+  // current position from registers (last chunk, p == 0 lanes of wave 0)
+  if (last_chunk && wave == 0 && p == 0) {
     float partial = 0.f;
 #pragma unroll
     for (int j = 0; j < 8; j++) partial += qf[j] * kn[j];
@@ -677,9 +700,10 @@ k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
     mrun = mnew;
   }
 
+  // merge position-groups within the wave
   for (int w = LP; w < 64; w <<= 1) {
     float mo = __shfl_xor(mrun, w);
-    float lo = __shfl_xor(lrun, w);
+    float lo2 = __shfl_xor(lrun, w);
     float mn = fmaxf(mrun, mo);
     float sa = (mrun == -INFINITY && mo == -INFINITY) ? 0.f : __expf(mrun - mn);
     float sb = (mrun == -INFINITY && mo == -INFINITY) ? 0.f : __expf(mo - mn);
@@ -688,10 +712,11 @@ k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
       float ao = __shfl_xor(acc[j], w);
       acc[j] = acc[j] * sa + ao * sb;
     }
-    lrun = lrun * sa + lo * sb;
+    lrun = lrun * sa + lo2 * sb;
     mrun = mn;
   }
 
+  // merge the 4 waves via LDS -> block partial in red[0..hd)+m,l
   float* accs = red;
   float* mls = red + 4 * hd;
   if (lane < LP) {
@@ -703,6 +728,8 @@ k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
     mls[wave * 2 + 1] = lrun;
   }
   __syncthreads();
+
+  float* part = scratch + ((size_t)h * SPLIT + chunk) * (hd + 2);
   if (wave == 0) {
     float mt = fmaxf(fmaxf(mls[0], mls[2]), fmaxf(mls[4], mls[6]));
     float lt = 0.f, sc[4];
@@ -711,28 +738,84 @@ k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
       sc[w] = (mls[w * 2] == -INFINITY) ? 0.f : __expf(mls[w * 2] - mt);
       lt += mls[w * 2 + 1] * sc[w];
     }
+    if (SPLIT == 1) {
+      // single chunk: normalize in registers and write out directly
+      float inv = 1.f / lt;
+      for (int d = lane; d < hd; d += 64) {
+        float v = 0.f;
+#pragma unroll
+        for (int w = 0; w < 4; w++) v += accs[w * hd + d] * sc[w];
+        out[(size_t)h * hd + d] = f2b(v * inv);
+      }
+    } else {
+      for (int d = lane; d < hd; d += 64) {
+        float v = 0.f;
+#pragma unroll
+        for (int w = 0; w < 4; w++) v += accs[w * hd + d] * sc[w];
+        part[2 + d] = v;  // UNnormalized chunk acc at max mt, sum lt
+      }
+      if (lane == 0) { part[0] = mt; part[1] = lt; }
+    }
+  }
+  if (SPLIT == 1) return;
+
+  // publish + ticket (G16 R1: plain stores -> per-wave drain -> barrier
+  // -> one-lane agent release -> asm drain -> relaxed ticket)
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    int t = __hip_atomic_fetch_add(&cnt[h], 1, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+    *lastflag = (t == SPLIT - 1);
+  }
+  __syncthreads();
+  if (!*lastflag) return;
+
+  // last arriver of this head: acquire, merge all chunk partials
+  if (threadIdx.x == 0)
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  __syncthreads();
+  if (wave == 0) {
+    const float* base = scratch + (size_t)h * SPLIT * (hd + 2);
+    float mt = -INFINITY;
+    for (int c = 0; c < SPLIT; c++)
+      mt = fmaxf(mt, base[(size_t)c * (hd + 2)]);
+    float lt = 0.f;
+    for (int c = 0; c < SPLIT; c++) {
+      float mc = base[(size_t)c * (hd + 2)];
+      if (mc != -INFINITY)
+        lt += base[(size_t)c * (hd + 2) + 1] * __expf(mc - mt);
+    }
     float inv = 1.f / lt;
     for (int d = lane; d < hd; d += 64) {
       float v = 0.f;
-#pragma unroll
-      for (int w = 0; w < 4; w++) v += accs[w * hd + d] * sc[w];
+      for (int c = 0; c < SPLIT; c++) {
+        float mc = base[(size_t)c * (hd + 2)];
+        if (mc != -INFINITY)
+          v += base[(size_t)c * (hd + 2) + 2 + d] * __expf(mc - mt);
+      }
       out[(size_t)h * hd + d] = f2b(v * inv);
     }
   }
+  __syncthreads();
+  if (threadIdx.x == 0) cnt[h] = 0;  // re-arm for the next graph replay
 }
 
 extern "C" hipError_t launch_attn_dec(const void* qkv, void* kc, void* vc,
                                       void* out, const void* len_ptr,
                                       const void* cost, const void* sint,
+                                      void* scratch, void* cnt, int split,
                                       int nh, int kvh, int hd, int S,
                                       float scale, float softcap, int window,
                                       hipStream_t stream) {
-  size_t lds = (4 * hd + 8) * sizeof(float);
-  hipLaunchKernelGGL(k_attn_dec, dim3(nh), dim3(256), lds, stream,
+  size_t lds = (4 * hd + 8) * sizeof(float) + 16;
+  hipLaunchKernelGGL(k_attn_dec, dim3(nh, split), dim3(256), lds, stream,
                      (const u16*)qkv, (u16*)kc, (u16*)vc, (u16*)out,
                      (const int*)len_ptr, (const float*)cost,
-                     (const float*)sint, nh, kvh, hd, S, scale, softcap,
-                     window);
+                     (const float*)sint, (float*)scratch, (int*)cnt, nh, kvh,
+                     hd, S, scale, softcap, window);
   return hipGetLastError();
 }
 

@@ -195,6 +195,12 @@ class GPUModel:
                                      device=dev))
         self.b_qkv = torch.zeros((self.nh_l + 2 * self.kvh_l) * hd, **bf)
         self.b_gu = torch.zeros(2 * I, **bf)
+        # split-T decode attention scratch (partials + per-head tickets)
+        self.attn_split = min(16, max(1, (self.max_seq + 1023) // 1024))
+        self.attn_scratch = torch.zeros(
+            self.nh_l * self.attn_split * (hd + 2), dtype=torch.float32,
+            device=dev)
+        self.attn_cnt = torch.zeros(self.nh_l, dtype=torch.int32, device=dev)
         self.s_gmax = torch.zeros(1, dtype=torch.int64, device=dev)
         self.s_pick = torch.zeros(1, dtype=torch.int64, device=dev)
         i32 = dict(dtype=torch.int32, device=dev)
@@ -341,8 +347,10 @@ class GPUModel:
                         g=lw["g_in"], eps=eps)
             ho.attn_dec(self.b_qkv, self.k_cache[i], self.v_cache[i],
                         self.b_att[0], self.len_buf, self.cos_t, self.sin_t,
+                        self.attn_scratch, self.attn_cnt,
                         self.nh_l, self.kvh_l, self.hd, self.scale,
-                        softcap=self.attn_softcap, window=window or 0)
+                        softcap=self.attn_softcap, window=window or 0,
+                        split=self.attn_split)
             if self.gemma:
                 self._dgemv(lw, "wo", self.b_att[0], t1)
                 tpu.all_reduce(t1)

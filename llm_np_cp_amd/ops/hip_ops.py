@@ -26,10 +26,10 @@ _LIB_PATH = os.path.join(os.path.dirname(os.path.dirname(
 _SIGS = {
     "launch_gemv_bf16": [ctypes.c_void_p] * 6 + [ctypes.c_int] * 4 +
                         [ctypes.c_float, ctypes.c_int, ctypes.c_float,
-                         ctypes.c_void_p],
+                         ctypes.c_int, ctypes.c_int, ctypes.c_void_p],
     "launch_gemv_fp8": [ctypes.c_void_p] * 7 + [ctypes.c_int] * 4 +
                        [ctypes.c_float, ctypes.c_int, ctypes.c_float,
-                        ctypes.c_void_p],
+                        ctypes.c_int, ctypes.c_int, ctypes.c_void_p],
     "launch_rmsnorm": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 2 +
                       [ctypes.c_float, ctypes.c_int, ctypes.c_void_p],
     "launch_rope_cache": [ctypes.c_void_p] * 8 + [ctypes.c_int] * 5 +
@@ -37,7 +37,7 @@ _SIGS = {
     "launch_attn": [ctypes.c_void_p] * 5 + [ctypes.c_int] * 5 +
                    [ctypes.c_float, ctypes.c_float, ctypes.c_int,
                     ctypes.c_void_p],
-    "launch_attn_dec": [ctypes.c_void_p] * 7 + [ctypes.c_int] * 4 +
+    "launch_attn_dec": [ctypes.c_void_p] * 9 + [ctypes.c_int] * 5 +
                        [ctypes.c_float, ctypes.c_float, ctypes.c_int,
                         ctypes.c_void_p],
     "launch_glu": [ctypes.c_void_p] * 3 + [ctypes.c_long, ctypes.c_int,
@@ -94,29 +94,30 @@ STAGE_RAW, STAGE_NORM, STAGE_GLU = 0, 1, 2
 def gemv(W: torch.Tensor, x: torch.Tensor, y: torch.Tensor,
          res: torch.Tensor | None = None, softcap: float = 0.0,
          stage: int = 0, x2: torch.Tensor | None = None,
-         g: torch.Tensor | None = None, act: int = 0, eps: float = 1e-5):
+         g: torch.Tensor | None = None, act: int = 0, eps: float = 1e-5,
+         nt: int = 1, rpw: int = 1):
     """y[N] = W[N,K] @ stage(x)[K] (+res); stage fuses RMSNorm or GLU
-    into the LDS staging pass (see csrc/llm_ops.hip)."""
+    into the LDS staging pass; nt = non-temporal weight stream."""
     N, K = W.shape
     out_f32 = 1 if y.dtype == torch.float32 else 0
     _check(lib().launch_gemv_bf16(
         _ptr(W), _ptr(x), _ptr(x2), _ptr(g), _ptr(y), _ptr(res), N, K,
         stage, act, ctypes.c_float(eps), out_f32,
-        ctypes.c_float(softcap), _stream()), "gemv")
+        ctypes.c_float(softcap), nt, rpw, _stream()), "gemv")
 
 
 def gemv_fp8(Wq: torch.Tensor, scales: torch.Tensor, x: torch.Tensor,
              y: torch.Tensor, res: torch.Tensor | None = None,
              softcap: float = 0.0, stage: int = 0,
              x2: torch.Tensor | None = None, g: torch.Tensor | None = None,
-             act: int = 0, eps: float = 1e-5):
+             act: int = 0, eps: float = 1e-5, nt: int = 1, rpw: int = 1):
     """y[N] = scales * (Wq[N,K] @ stage(x)); Wq = e4m3fn bytes."""
     N, K = Wq.shape
     out_f32 = 1 if y.dtype == torch.float32 else 0
     _check(lib().launch_gemv_fp8(
         _ptr(Wq), _ptr(scales), _ptr(x), _ptr(x2), _ptr(g), _ptr(y),
         _ptr(res), N, K, stage, act, ctypes.c_float(eps), out_f32,
-        ctypes.c_float(softcap), _stream()), "gemv_fp8")
+        ctypes.c_float(softcap), nt, rpw, _stream()), "gemv_fp8")
 
 
 def rmsnorm(x: torch.Tensor, g: torch.Tensor, y: torch.Tensor,
@@ -155,13 +156,17 @@ def attn(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
 def attn_dec(qkv: torch.Tensor, k_cache: torch.Tensor,
              v_cache: torch.Tensor, out: torch.Tensor,
              len_ptr: torch.Tensor, cos_t: torch.Tensor,
-             sin_t: torch.Tensor, nh: int, kvh: int, hd: int,
-             scale: float, softcap: float = 0.0, window: int = 0):
-    """Fused decode attention: RoPE(q,k) + KV write + online softmax."""
+             sin_t: torch.Tensor, scratch: torch.Tensor,
+             cnt: torch.Tensor, nh: int, kvh: int, hd: int,
+             scale: float, softcap: float = 0.0, window: int = 0,
+             split: int = 1):
+    """Fused decode attention: RoPE(q,k) + KV write + online softmax,
+    KV range split over `split` blocks/head (last-arriver merge)."""
     S = k_cache.shape[1]
     _check(lib().launch_attn_dec(
         _ptr(qkv), _ptr(k_cache), _ptr(v_cache), _ptr(out), _ptr(len_ptr),
-        _ptr(cos_t), _ptr(sin_t), nh, kvh, hd, S, ctypes.c_float(scale),
+        _ptr(cos_t), _ptr(sin_t), _ptr(scratch), _ptr(cnt), split,
+        nh, kvh, hd, S, ctypes.c_float(scale),
         ctypes.c_float(softcap), window, _stream()), "attn_dec")
 
 

@@ -322,11 +322,21 @@ def test_attn_dec_fused_matches_unfused():
     sin_t = torch.from_numpy(np.sin(fr).astype(np.float32)).to(dev())
     pos = torch.tensor([pos0], dtype=torch.int32, device=dev())
 
-    # fused
-    out1 = torch.empty(nh * hd, dtype=torch.bfloat16, device=dev())
-    ho.attn_dec(qkv, kc1, vc1, out1, pos, cos_t, sin_t, nh, kvh, hd,
-                hd ** -0.5)
-    torch.cuda.synchronize()
+    # fused (exercise both the single-chunk and split-merge paths)
+    outs = []
+    for split in (1, 4):
+        kcx, vcx = kc1.clone(), vc1.clone()
+        scratch = torch.zeros(nh * split * (hd + 2), dtype=torch.float32,
+                              device=dev())
+        cnt = torch.zeros(nh, dtype=torch.int32, device=dev())
+        o = torch.empty(nh * hd, dtype=torch.bfloat16, device=dev())
+        ho.attn_dec(qkv, kcx, vcx, o, pos, cos_t, sin_t, scratch, cnt,
+                    nh, kvh, hd, hd ** -0.5, split=split)
+        torch.cuda.synchronize()
+        assert cnt.sum().item() == 0  # counters re-armed
+        outs.append(o)
+        kc1, vc1 = kcx, vcx
+    out1 = outs[0]
 
     # unfused chain on a copy
     q = qkv[:nh * hd].clone()
@@ -338,6 +348,7 @@ def test_attn_dec_fused_matches_unfused():
     torch.cuda.synchronize()
 
     assert_close(out1, out2, rtol=2e-2, atol=2e-2)
+    assert_close(outs[1], out2, rtol=2e-2, atol=2e-2)
     assert_close(kc1[:, pos0], kc2[:, pos0], rtol=2e-2, atol=2e-2)
     assert_close(vc1[:, pos0], vc2[:, pos0], rtol=2e-2, atol=2e-2)
 

@@ -1,0 +1,125 @@
+#!/usr/bin/env python3
+"""Per-kernel microbenchmark (runs on an MI355X box).
+
+Times each decode-path kernel shape standalone: a hipGraph of REPS
+launches, timed with CUDA events, reported as us/launch and effective
+TB/s of weight traffic.  Used to pick GEMV variants (nt, rows-per-wave)
+— guide §5.4: within-probe comparison, same process, interleaved.
+"""
+
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch  # noqa: E402
+
+
+def time_graph(fn, reps=200, warmup=20):
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        for _ in range(3):
+            fn()
+    torch.cuda.current_stream().wait_stream(s)
+    torch.cuda.synchronize()
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        for _ in range(reps):
+            fn()
+    for _ in range(warmup):
+        pass
+    g.replay()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(5):
+        g.replay()
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / (5 * reps)
+    return dt * 1e6  # us per launch
+
+
+def main():
+    from csrc.build import ensure_built
+    ensure_built()
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    dev = torch.device("cuda:0")
+    torch.cuda.set_device(dev)
+
+    shapes = [
+        ("qkv", 3072, 2048), ("wo", 2048, 2048), ("wgu", 16384, 2048),
+        ("wdown", 2048, 8192), ("lm_head", 128256, 2048),
+        ("9b_qkv", 8192, 3584), ("9b_wgu", 28672, 3584),
+    ]
+    print(f"{'shape':>9} {'N':>7} {'K':>6} | " +
+          " | ".join(f"{v:>12}" for v in
+                     ["bf16", "bf16-nt", "bf16-nt-r2", "fp8-nt", "fp8-nt-r2"]))
+    for name, N, K in shapes:
+        W = torch.randn(N, K, device=dev).to(torch.bfloat16) * 0.05
+        s = W.float().abs().amax(dim=1).clamp_min(1e-8) / 448.0
+        Wq = (W.float() / s[:, None]).to(torch.float8_e4m3fn).view(torch.uint8)
+        x = torch.randn(K, device=dev).to(torch.bfloat16)
+        y = torch.empty(N, dtype=torch.bfloat16, device=dev)
+        res = []
+        for kind, nt, rpw in [("bf16", 0, 1), ("bf16", 1, 1), ("bf16", 1, 2),
+                              ("fp8", 1, 1), ("fp8", 1, 2)]:
+            if kind == "bf16":
+                fn = lambda: ho.gemv(W, x, y, nt=nt, rpw=rpw)
+                nbytes = N * K * 2
+            else:
+                fn = lambda: ho.gemv_fp8(Wq, s, x, y, nt=nt, rpw=rpw)
+                nbytes = N * K
+            us = time_graph(fn)
+            res.append(f"{us:6.2f}us {nbytes/us/1e6:5.2f}T")
+        print(f"{name:>9} {N:>7} {K:>6} | " + " | ".join(f"{r:>12}" for r in res))
+
+    # attention decode at several context lengths
+    print("\nattn_dec (nh=32 kvh=8 hd=64):")
+    nh, kvh, hd, S = 32, 8, 64, 8192
+    kc = torch.randn(kvh, S, hd, device=dev).to(torch.bfloat16)
+    vc = torch.randn_like(kc)
+    qkv = torch.randn((nh + 2 * kvh) * hd, device=dev).to(torch.bfloat16)
+    out = torch.empty(nh * hd, dtype=torch.bfloat16, device=dev)
+    inv = 1.0 / (10000.0 ** (torch.arange(0, hd, 2).float() / hd))
+    fr = torch.outer(torch.arange(S).float(), inv)
+    cos_t = torch.cos(fr).to(dev)
+    sin_t = torch.sin(fr).to(dev)
+    for split in [1, 4, 8, 16]:
+        scratch = torch.zeros(nh * split * (hd + 2), dtype=torch.float32,
+                              device=dev)
+        cnt = torch.zeros(nh, dtype=torch.int32, device=dev)
+        for T in [128, 512, 2048, 8000]:
+            pos = torch.tensor([T - 1], dtype=torch.int32, device=dev)
+            us = time_graph(lambda: ho.attn_dec(
+                qkv, kc, vc, out, pos, cos_t, sin_t, scratch, cnt,
+                nh, kvh, hd, hd ** -0.5, split=split))
+            print(f"  split={split:2d} T={T:5d}: {us:6.2f}us")
+
+    # rmsnorm / sampler
+    H = 2048
+    xh = torch.randn(H, device=dev).to(torch.bfloat16)
+    gh = torch.randn(H, device=dev)
+    yh = torch.empty_like(xh)
+    print(f"\nrmsnorm H={H}: {time_graph(lambda: ho.rmsnorm(xh, gh, yh)):.2f}us")
+
+    V = 128256
+    logits = torch.randn(V, device=dev)
+    ctr = torch.zeros(1, dtype=torch.int64, device=dev)
+    gmax = torch.zeros(1, dtype=torch.int64, device=dev)
+    pick = torch.zeros(1, dtype=torch.int64, device=dev)
+    nt_ = torch.zeros(1, dtype=torch.int32, device=dev)
+    ring = torch.zeros(65536, dtype=torch.int32, device=dev)
+    nout = torch.zeros(1, dtype=torch.int32, device=dev)
+    ln = torch.zeros(1, dtype=torch.int32, device=dev)
+    us = time_graph(lambda: ho.sample(logits, 0.1, True, 0, ctr, gmax, pick,
+                                      nt_, ring, nout, ln))
+    print(f"sample greedy V={V}: {us:.2f}us")
+    us = time_graph(lambda: ho.sample(logits, 0.1, False, 0, ctr, gmax, pick,
+                                      nt_, ring, nout, ln))
+    print(f"sample min-p  V={V}: {us:.2f}us")
+
+
+if __name__ == "__main__":
+    main()
