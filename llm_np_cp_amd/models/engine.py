@@ -196,7 +196,7 @@ class GPUModel:
         self.b_qkv = torch.zeros((self.nh_l + 2 * self.kvh_l) * hd, **bf)
         self.b_gu = torch.zeros(2 * I, **bf)
         # split-T decode attention scratch (partials + per-head tickets)
-        self.attn_split = min(16, max(1, (self.max_seq + 1023) // 1024))
+        self.attn_split = min(16, max(4, (self.max_seq + 255) // 256))
         self.attn_scratch = torch.zeros(
             self.nh_l * self.attn_split * (hd + 2), dtype=torch.float32,
             device=dev)
