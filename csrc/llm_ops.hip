@@ -1001,8 +1001,9 @@ extern "C" __global__ void k_sample_fin(
     unsigned long long* __restrict__ gmax, uint64_t* __restrict__ ctr,
     int* __restrict__ next_token, int* __restrict__ out_ring,
     int* __restrict__ nout, int* __restrict__ len_ptr, int bump_len,
-    int greedy) {
+    int greedy, int V) {
   int winner = unpack_idx(*pick);
+  if (winner < 0 || winner >= V) winner = 0;  // NaN-logit insurance
   *next_token = winner;
   int n = *nout;
   out_ring[n] = winner;
@@ -1030,7 +1031,7 @@ extern "C" hipError_t launch_sample(const void* logits, int V, float min_p,
   hipLaunchKernelGGL(k_sample_fin, dim3(1), dim3(1), 0, stream,
                      (unsigned long long*)pick, (unsigned long long*)gmax,
                      (uint64_t*)ctr, (int*)next_token, (int*)out_ring,
-                     (int*)nout, (int*)len_ptr, bump_len, greedy);
+                     (int*)nout, (int*)len_ptr, bump_len, greedy, V);
   return hipGetLastError();
 }
 
@@ -1052,8 +1053,8 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 extern "C" __global__ void __launch_bounds__(256)
 k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
-            u16* __restrict__ Y, const u16* __restrict__ res, int M, int N,
-            int K) {
+            u16* __restrict__ Y, const u16* __restrict__ res,
+            float* __restrict__ accbuf, int M, int N, int K) {
   __shared__ u16 As[BM * LDS_STRIDE];
   __shared__ u16 Bs[BN * LDS_STRIDE];
 
@@ -1061,6 +1062,13 @@ k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
   const int wave = tid >> 6, lane = tid & 63;
   const int wrow = wave >> 1, wcol = wave & 1;  // 2x2 waves
   const int bm = blockIdx.x * BM, bn = blockIdx.y * BN;
+  // split-K: gridDim.z blocks share (bm,bn), each covers a K slice and
+  // atomically accumulates fp32 partials; k_gemm_fin converts to bf16.
+  const int SK = gridDim.z;
+  const int kslices = (K / BK + SK - 1) / SK;
+  const int k_lo = blockIdx.z * kslices * BK;
+  int k_hi = k_lo + kslices * BK;
+  if (k_hi > K) k_hi = K;
 
   f32x4 acc[4][4];
 #pragma unroll
@@ -1071,7 +1079,7 @@ k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
   // stage thread mapping: 256 threads x 16 B; tile row r = t/4, col (t%4)*8
   const int sr = tid >> 2, sc = (tid & 3) * 8;
 
-  for (int kt = 0; kt < K; kt += BK) {
+  for (int kt = k_lo; kt < k_hi; kt += BK) {
     // A tile: rows clamped to M-1 (stores are masked later)
 #pragma unroll
     for (int p = 0; p < 2; p++) {
@@ -1120,20 +1128,60 @@ k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
         int col = bn + wcol * 64 + j * 16 + cc;
         if (row < M && col < N) {
           float v = acc[i][j][r];
-          if (res) v += b2f(res[(size_t)row * N + col]);
-          Y[(size_t)row * N + col] = f2b(v);
+          if (SK > 1) {
+            atomicAdd(accbuf + (size_t)row * N + col, v);
+          } else {
+            if (res) v += b2f(res[(size_t)row * N + col]);
+            Y[(size_t)row * N + col] = f2b(v);
+          }
         }
       }
     }
   }
 }
 
+// convert split-K fp32 accumulator to bf16 (+res); also used to zero it
+extern "C" __global__ void __launch_bounds__(256)
+k_gemm_fin(const float* __restrict__ accbuf, const u16* __restrict__ res,
+           u16* __restrict__ Y, long total) {
+  long i = (long)blockIdx.x * 256 + threadIdx.x;
+  if (i >= total) return;
+  float v = accbuf[i];
+  if (res) v += b2f(res[i]);
+  Y[i] = f2b(v);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+k_zero_f32(float* __restrict__ p, long total) {
+  long i = ((long)blockIdx.x * 256 + threadIdx.x) * 4;
+  if (i >= total) return;
+  *(f4v*)(p + i) = {0.f, 0.f, 0.f, 0.f};
+}
+
 extern "C" hipError_t launch_gemm_bf16(const void* X, const void* W, void* Y,
-                                       const void* res, int M, int N, int K,
-                                       hipStream_t stream) {
-  dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
+                                       const void* res, void* accbuf, int M,
+                                       int N, int K, hipStream_t stream) {
+  int gm = (M + BM - 1) / BM, gn = (N + BN - 1) / BN;
+  int sk = 1;
+  if (accbuf) {
+    while (sk < 8 && gm * gn * sk * 2 <= 256 && (K / BK) % (sk * 2) == 0)
+      sk *= 2;
+  }
+  if (sk > 1) {
+    long total = (long)M * N;
+    hipLaunchKernelGGL(k_zero_f32, dim3((uint32_t)((total / 4 + 255) / 256)),
+                       dim3(256), 0, stream, (float*)accbuf, total);
+  }
+  dim3 grid(gm, gn, sk);
   hipLaunchKernelGGL(k_gemm_bf16, grid, dim3(256), 0, stream, (const u16*)X,
-                     (const u16*)W, (u16*)Y, (const u16*)res, M, N, K);
+                     (const u16*)W, (u16*)Y, (const u16*)res, (float*)accbuf,
+                     M, N, K);
+  if (sk > 1) {
+    long total = (long)M * N;
+    hipLaunchKernelGGL(k_gemm_fin, dim3((uint32_t)((total + 255) / 256)),
+                       dim3(256), 0, stream, (const float*)accbuf,
+                       (const u16*)res, (u16*)Y, total);
+  }
   return hipGetLastError();
 }
 

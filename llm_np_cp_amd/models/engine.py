@@ -188,6 +188,8 @@ class GPUModel:
         self.b_att = torch.zeros(PC, self.nh_l * hd, **bf)
         self.b_gate = torch.zeros(PC, I, **bf)
         self.b_up = torch.zeros(PC, I, **bf)
+        self.b_gemm_acc = torch.zeros(PC * max(I, H), dtype=torch.float32,
+                                      device=dev)
         self.b_logits_l = torch.zeros(self.vocab_l, dtype=torch.float32,
                                       device=dev)
         self.b_logits = (self.b_logits_l if self.world == 1 else
@@ -240,7 +242,8 @@ class GPUModel:
         if M == 1:
             ho.gemv(W, x, y, res=res, softcap=softcap)
         else:
-            ho.gemm(x[:M], W, y, res=res[:M] if res is not None else None)
+            ho.gemm(x[:M], W, y, res=res[:M] if res is not None else None,
+                    accbuf=self.b_gemm_acc)
 
     def _layers_forward(self, M: int):
         cfg = self.config
@@ -436,7 +439,13 @@ class GPUModel:
                       bump_len=False)
         n_steps = n_tokens - (1 if first_from_logits else 0)
         if use_graph and n_steps > 0:
-            n_steps -= self.capture_decode_graph(greedy, min_p)
+            try:
+                n_steps -= self.capture_decode_graph(greedy, min_p)
+            except Exception:
+                # e.g. RCCL collectives not capturable in this stack:
+                # fall back to eager launches (correctness first)
+                use_graph = False
+        if use_graph and n_steps > 0:
             for _ in range(n_steps):
                 self._graph.replay()
         else:

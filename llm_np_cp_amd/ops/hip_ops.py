@@ -47,7 +47,7 @@ _SIGS = {
     "launch_sample": [ctypes.c_void_p, ctypes.c_int, ctypes.c_float,
                       ctypes.c_int, ctypes.c_uint64] + [ctypes.c_void_p] * 7 +
                      [ctypes.c_int, ctypes.c_void_p],
-    "launch_gemm_bf16": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 3 +
+    "launch_gemm_bf16": [ctypes.c_void_p] * 5 + [ctypes.c_int] * 3 +
                         [ctypes.c_void_p],
     "launch_addinto": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_long,
                        ctypes.c_void_p],
@@ -201,13 +201,18 @@ def sample(logits: torch.Tensor, min_p: float, greedy: bool, seed: int,
 
 
 def gemm(X: torch.Tensor, W: torch.Tensor, Y: torch.Tensor,
-         res: torch.Tensor | None = None):
-    """Y[M,N] = X[M,K] @ W[N,K]^T (+res), bf16, MFMA prefill path."""
+         res: torch.Tensor | None = None,
+         accbuf: torch.Tensor | None = None):
+    """Y[M,N] = X[M,K] @ W[N,K]^T (+res), bf16, MFMA prefill path.
+    With accbuf (fp32 scratch >= M*N) small-M launches split K over
+    grid.z for chip fill."""
     M, K = X.shape
     N = W.shape[0]
     assert W.shape[1] == K and K % 32 == 0
+    if accbuf is not None and accbuf.numel() < M * N:
+        accbuf = None
     _check(lib().launch_gemm_bf16(_ptr(X), _ptr(W), _ptr(Y), _ptr(res),
-                                  M, N, K, _stream()), "gemm")
+                                  _ptr(accbuf), M, N, K, _stream()), "gemm")
 
 
 def addinto(y: torch.Tensor, a: torch.Tensor):

@@ -233,6 +233,13 @@ def test_gemm(M, N, K):
     torch.cuda.synchronize()
     assert_close(Y, ref + res.float(), rtol=3e-2, atol=3e-2)
 
+    # split-K path (fp32 atomic accumulation + finalize)
+    acc = torch.zeros(M * N, dtype=torch.float32, device=dev())
+    Y2 = torch.empty_like(Y)
+    ho.gemm(X, W, Y2, res=res, accbuf=acc)
+    torch.cuda.synchronize()
+    assert_close(Y2, ref + res.float(), rtol=3e-2, atol=3e-2)
+
 
 def test_sample_greedy_and_minp():
     from llm_np_cp_amd.ops import hip_ops as ho
