@@ -344,20 +344,20 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
 // per row; fused single pass (x kept in registers between reduce+scale).
 // ====================================================================
 
-extern "C" __global__ void __launch_bounds__(256)
+extern "C" __global__ void __launch_bounds__(1024)
 k_rmsnorm(const u16* __restrict__ x, const float* __restrict__ g,
           const u16* __restrict__ res, u16* __restrict__ y,
           int H, float eps, int mode) {
-  __shared__ float warp_sums[8];
+  __shared__ float warp_sums[16];
   const int row = blockIdx.x;
   const u16* xr = x + (size_t)row * H;
   u16* yr = y + (size_t)row * H;
   const u16* rr = res ? res + (size_t)row * H : nullptr;
 
-  s8v buf[4];  // up to 8192 elems per row at 256 threads * 8/chunk
+  s8v buf[2];  // up to 16384 elems per row at 1024 threads * 8/chunk
   int nchunk = 0;
   float ss = 0.f;
-  for (int i = threadIdx.x * 8; i < H; i += 2048) {
+  for (int i = threadIdx.x * 8; i < H; i += 8192) {
     s8v v = *(const s8v*)(xr + i);
     buf[nchunk++] = v;
 #pragma unroll
@@ -370,17 +370,23 @@ k_rmsnorm(const u16* __restrict__ x, const float* __restrict__ g,
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
   if (lane == 0) warp_sums[wave] = ss;
   __syncthreads();
-  float tot = warp_sums[0] + warp_sums[1] + warp_sums[2] + warp_sums[3];
+  float tot = 0.f;
+#pragma unroll
+  for (int w = 0; w < 16; w++) tot += warp_sums[w];
   const float rnorm = rsqrtf(tot / (float)H + eps);
 
   nchunk = 0;
-  for (int i = threadIdx.x * 8; i < H; i += 2048) {
+  for (int i = threadIdx.x * 8; i < H; i += 8192) {
     s8v v = buf[nchunk++];
+    f4v g0 = *(const f4v*)(g + i);
+    f4v g1 = *(const f4v*)(g + i + 4);
+    s8v rv;
+    if (mode == 1) rv = *(const s8v*)(rr + i);
     u16 o[8];
 #pragma unroll
     for (int j = 0; j < 8; j++) {
-      float f = b2f(((u16*)&v)[j]) * rnorm * g[i + j];
-      if (mode == 1) f += b2f(rr[i + j]);
+      float f = b2f(((u16*)&v)[j]) * rnorm * (j < 4 ? g0[j] : g1[j - 4]);
+      if (mode == 1) f += b2f(((u16*)&rv)[j]);
       o[j] = f2b(f);
     }
     *(s8v*)(yr + i) = *(s8v*)o;
@@ -390,8 +396,9 @@ k_rmsnorm(const u16* __restrict__ x, const float* __restrict__ g,
 extern "C" hipError_t launch_rmsnorm(const void* x, const void* g,
                                      const void* res, void* y, int M, int H,
                                      float eps, int mode, hipStream_t stream) {
-  hipLaunchKernelGGL(k_rmsnorm, dim3(M), dim3(256), 0, stream, (const u16*)x,
-                     (const float*)g, (const u16*)res, (u16*)y, H, eps, mode);
+  hipLaunchKernelGGL(k_rmsnorm, dim3(M), dim3(1024), 0, stream,
+                     (const u16*)x, (const float*)g, (const u16*)res, (u16*)y,
+                     H, eps, mode);
   return hipGetLastError();
 }
 
