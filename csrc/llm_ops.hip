@@ -506,31 +506,43 @@ k_attn(const u16* __restrict__ q, const u16* __restrict__ kc,
 #pragma unroll
   for (int j = 0; j < 8; j++) acc[j] = 0.f;
 
-  for (int t0 = start + wave * PP; t0 < T; t0 += 4 * PP) {
-    int t = t0 + p;
-    bool valid = t < T;
-    int tl = valid ? t : (T - 1);
-    // K row: 16 B per lane; consecutive lanes -> consecutive addresses
-    s8v kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
-    float partial = 0.f;
+  {
+    int t0 = start + wave * PP;
+    if (t0 < T) {
+      int t = t0 + p;
+      bool valid = t < T;
+      int tl = valid ? t : start;
+      s8v kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
+      s8v vv = *(const s8v*)(V0 + (size_t)tl * hd + d0);
+      for (; t0 < T; t0 += 4 * PP) {
+        int t0n = t0 + 4 * PP;
+        s8v kvn, vvn;
+        bool validn = false;
+        if (t0n < T) {
+          int tn = t0n + p;
+          validn = tn < T;
+          int tln = validn ? tn : start;
+          kvn = *(const s8v*)(K0 + (size_t)tln * hd + d0);
+          vvn = *(const s8v*)(V0 + (size_t)tln * hd + d0);
+        }
+        float partial = 0.f;
 #pragma unroll
-    for (int j = 0; j < 8; j++) partial += qf[j] * b2f(((u16*)&kv)[j]);
-    // reduce across the LP lanes of this position
-    for (int w = 1; w < LP; w <<= 1) partial += __shfl_xor(partial, w);
-    float score = partial * scale;
-    if (softcap > 0.f) score = softcap * tanhf(score / softcap);
-    if (!valid) score = -INFINITY;
-
-    float mnew = fmaxf(mrun, score);
-    if (mnew == -INFINITY) continue;  // whole sub-iteration invalid
-    float alpha = __expf(mrun - mnew);  // exp(-inf)=0 on first hit
-    float pv = __expf(score - mnew);
-    lrun = lrun * alpha + pv;
-    s8v vv = *(const s8v*)(V0 + (size_t)tl * hd + d0);
+        for (int j = 0; j < 8; j++) partial += qf[j] * b2f(((u16*)&kv)[j]);
+        for (int w = 1; w < LP; w <<= 1) partial += __shfl_xor(partial, w);
+        float score = partial * scale;
+        if (softcap > 0.f) score = softcap * tanhf(score / softcap);
+        if (!valid) score = -INFINITY;
+        float mnew = fmaxf(mrun, score);
+        float alpha = (mnew == -INFINITY) ? 0.f : __expf(mrun - mnew);
+        float pv = (mnew == -INFINITY) ? 0.f : __expf(score - mnew);
+        lrun = lrun * alpha + pv;
 #pragma unroll
-    for (int j = 0; j < 8; j++)
-      acc[j] = acc[j] * alpha + pv * b2f(((u16*)&vv)[j]);
-    mrun = mnew;
+        for (int j = 0; j < 8; j++)
+          acc[j] = acc[j] * alpha + pv * b2f(((u16*)&vv)[j]);
+        if (mnew != -INFINITY) mrun = mnew;
+        kv = kvn; vv = vvn; valid = validn;
+      }
+    }
   }
 
   // merge across position-groups within the wave (lanes l, l^LP, l^2LP, ...)
@@ -666,28 +678,46 @@ k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
 #pragma unroll
   for (int j = 0; j < 8; j++) acc[j] = 0.f;
 
-  for (int t0 = c0 + wave * PP; t0 < c1; t0 += 4 * PP) {
-    int t = t0 + p;
-    bool valid = t < c1;
-    int tl = valid ? t : c0;
-    s8v kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
-    float partial = 0.f;
+  // software-pipelined history scan: K[i+1]/V[i+1] issue while the
+  // score/online-update of iteration i computes (the serial
+  // K->dot->softmax->V chain was the latency bottleneck at small T)
+  {
+    int t0 = c0 + wave * PP;
+    if (t0 < c1) {
+      int t = t0 + p;
+      bool valid = t < c1;
+      int tl = valid ? t : c0;
+      s8v kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
+      s8v vv = *(const s8v*)(V0 + (size_t)tl * hd + d0);
+      for (; t0 < c1; t0 += 4 * PP) {
+        int t0n = t0 + 4 * PP;
+        s8v kvn, vvn;
+        bool validn = false;
+        if (t0n < c1) {
+          int tn = t0n + p;
+          validn = tn < c1;
+          int tln = validn ? tn : c0;
+          kvn = *(const s8v*)(K0 + (size_t)tln * hd + d0);
+          vvn = *(const s8v*)(V0 + (size_t)tln * hd + d0);
+        }
+        float partial = 0.f;
 #pragma unroll
-    for (int j = 0; j < 8; j++) partial += qf[j] * b2f(((u16*)&kv)[j]);
-    for (int w = 1; w < LP; w <<= 1) partial += __shfl_xor(partial, w);
-    float score = partial * scale;
-    if (softcap > 0.f) score = softcap * tanhf(score / softcap);
-    if (!valid) score = -INFINITY;
-    float mnew = fmaxf(mrun, score);
-    if (mnew == -INFINITY) continue;
-    float alpha = __expf(mrun - mnew);
-    float pv = __expf(score - mnew);
-    lrun = lrun * alpha + pv;
-    s8v vv = *(const s8v*)(V0 + (size_t)tl * hd + d0);
+        for (int j = 0; j < 8; j++) partial += qf[j] * b2f(((u16*)&kv)[j]);
+        for (int w = 1; w < LP; w <<= 1) partial += __shfl_xor(partial, w);
+        float score = partial * scale;
+        if (softcap > 0.f) score = softcap * tanhf(score / softcap);
+        if (!valid) score = -INFINITY;
+        float mnew = fmaxf(mrun, score);
+        float alpha = (mnew == -INFINITY) ? 0.f : __expf(mrun - mnew);
+        float pv = (mnew == -INFINITY) ? 0.f : __expf(score - mnew);
+        lrun = lrun * alpha + pv;
 #pragma unroll
-    for (int j = 0; j < 8; j++)
-      acc[j] = acc[j] * alpha + pv * b2f(((u16*)&vv)[j]);
-    mrun = mnew;
+        for (int j = 0; j < 8; j++)
+          acc[j] = acc[j] * alpha + pv * b2f(((u16*)&vv)[j]);
+        if (mnew != -INFINITY) mrun = mnew;
+        kv = kvn; vv = vvn; valid = validn;
+      }
+    }
   }
 
   // current position from registers (last chunk, p == 0 lanes of wave 0)
