@@ -29,8 +29,10 @@ def main():
                          "gemma-2-9b at N>1)")
     ap.add_argument("--prompt-len", type=int, default=64)
     ap.add_argument("--max-seq", type=int, default=None)
-    ap.add_argument("--dtype", type=str, default="bf16",
-                    choices=["bf16", "fp8"])
+    ap.add_argument("--dtype", type=str, default=None,
+                    choices=["bf16", "fp8"],
+                    help="default: bf16 at N=1, fp8 at N>1 (BASELINE "
+                         "config 4: Gemma-2-9B TP=8 fp8 weights)")
     ap.add_argument("--no-graph", action="store_true")
     args = ap.parse_args()
 
@@ -45,17 +47,18 @@ def main():
     ensure_built()
 
     import llm_np_cp_amd as L
-    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.io.loader import LazyRandomWeights
     from llm_np_cp_amd.models.engine import GPUModel
     from llm_np_cp_amd.parallel.tp import init_distributed
 
     rank, world = init_distributed()
+    dtype = args.dtype or ("bf16" if world == 1 else "fp8")
     cfg = L.preset_config(model_name)
     max_seq = args.max_seq or min(
         cfg.max_position_embeddings,
         args.prompt_len + args.steps + args.warmup + 64)
-    w = random_weights(cfg, seed=0)
-    model = GPUModel(cfg, w, max_seq=max_seq, seed=0, dtype=args.dtype)
+    w = LazyRandomWeights(cfg, seed=0)
+    model = GPUModel(cfg, w, max_seq=max_seq, seed=0, dtype=dtype)
 
     rng = np.random.default_rng(0)
     prompt = rng.integers(0, cfg.vocab_size, size=args.prompt_len)
@@ -109,7 +112,7 @@ def main():
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": args.dtype,
+            "dtype": dtype,
             "data": "synthetic",
             "config": {
                 "model": model_name,

@@ -116,6 +116,44 @@ def random_weights(cfg: ModelConfig, seed: int = 0,
     return w
 
 
+class LazyRandomWeights:
+    """Mapping that generates each HF-named tensor on demand (fp32,
+    uniform [-scale, scale), deterministic per name+seed) and holds no
+    reference afterward.  Keeps host memory at one-tensor peak — with
+    8 TP ranks of a 9B model on one node, an eager fp32 dict would cost
+    ~36 GB per rank.  Norm gammas follow random_weights conventions."""
+
+    def __init__(self, cfg: ModelConfig, seed: int = 0, scale: float = 0.02):
+        self.cfg = cfg
+        self.seed = seed
+        self.scale = scale
+        self.shapes = hf_weight_shapes(cfg)
+
+    def __contains__(self, name):
+        return name in self.shapes
+
+    def keys(self):
+        return self.shapes.keys()
+
+    def __getitem__(self, name) -> np.ndarray:
+        if name not in self.shapes:
+            raise KeyError(name)
+        shape = self.shapes[name]
+        import zlib
+        h = (zlib.crc32(name.encode()) ^ self.seed) & 0x7FFFFFFF
+        rng = np.random.default_rng(h)  # deterministic ACROSS processes
+        # (python hash() is per-process salted: TP ranks must agree)
+        if "norm" in name:
+            base = 0.0 if self.cfg.model_type == "gemma2" else 1.0
+            return (base + self.scale *
+                    rng.standard_normal(shape)).astype(np.float32)
+        a = rng.random(np.prod(shape), dtype=np.float32).reshape(shape)
+        return (a - 0.5) * (2.0 * self.scale)
+
+    def get(self, name, default=None):
+        return self[name] if name in self.shapes else default
+
+
 def write_synthetic_checkpoint(model_dir: str, preset: str, seed: int = 0):
     """Write a random-init single-file safetensors checkpoint + config.json
     for a preset architecture (for tests of the directory-loading path)."""
