@@ -208,10 +208,12 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
                                        const void* res, int N, int K,
                                        int stage, int act, float eps,
                                        int out_f32, float softcap,
-                                       int nt, int rpw, hipStream_t stream) {
+                                       int nt, int rpw, int maxblocks,
+                                       hipStream_t stream) {
   size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 16);
   int blocks = (N + 4 * rpw - 1) / (4 * rpw);
-  if (blocks > 1024) blocks = 1024;  // grid-stride the rest (Guideline 11)
+  int cap = maxblocks > 0 ? maxblocks : 1024;
+  if (blocks > cap) blocks = cap;  // grid-stride the rest (Guideline 11)
 #define GEMV_CASE(NTV, RPWV)                                                 \
   hipLaunchKernelGGL((k_gemv_bf16_t<NTV, RPWV>), dim3(blocks), dim3(256),    \
                      lds, stream, (const u16*)W, (const u16*)x,              \
@@ -320,10 +322,12 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
                                       const void* g, void* y, const void* res,
                                       int N, int K, int stage, int act,
                                       float eps, int out_f32, float softcap,
-                                      int nt, int rpw, hipStream_t stream) {
+                                      int nt, int rpw, int maxblocks,
+                                      hipStream_t stream) {
   size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 16);
   int blocks = (N + 4 * rpw - 1) / (4 * rpw);
-  if (blocks > 1024) blocks = 1024;
+  int cap = maxblocks > 0 ? maxblocks : 1024;
+  if (blocks > cap) blocks = cap;
 #define GEMV8_CASE(NTV, RPWV)                                               \
   hipLaunchKernelGGL((k_gemv_fp8_t<NTV, RPWV>), dim3(blocks), dim3(256),    \
                      lds, stream, (const uint8_t*)W, (const float*)scales,  \

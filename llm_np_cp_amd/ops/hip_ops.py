@@ -26,10 +26,12 @@ _LIB_PATH = os.path.join(os.path.dirname(os.path.dirname(
 _SIGS = {
     "launch_gemv_bf16": [ctypes.c_void_p] * 6 + [ctypes.c_int] * 4 +
                         [ctypes.c_float, ctypes.c_int, ctypes.c_float,
-                         ctypes.c_int, ctypes.c_int, ctypes.c_void_p],
+                         ctypes.c_int, ctypes.c_int, ctypes.c_int,
+                         ctypes.c_void_p],
     "launch_gemv_fp8": [ctypes.c_void_p] * 7 + [ctypes.c_int] * 4 +
                        [ctypes.c_float, ctypes.c_int, ctypes.c_float,
-                        ctypes.c_int, ctypes.c_int, ctypes.c_void_p],
+                        ctypes.c_int, ctypes.c_int, ctypes.c_int,
+                        ctypes.c_void_p],
     "launch_rmsnorm": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 2 +
                       [ctypes.c_float, ctypes.c_int, ctypes.c_void_p],
     "launch_rope_cache": [ctypes.c_void_p] * 8 + [ctypes.c_int] * 5 +
@@ -95,7 +97,7 @@ def gemv(W: torch.Tensor, x: torch.Tensor, y: torch.Tensor,
          res: torch.Tensor | None = None, softcap: float = 0.0,
          stage: int = 0, x2: torch.Tensor | None = None,
          g: torch.Tensor | None = None, act: int = 0, eps: float = 1e-5,
-         nt: int = 1, rpw: int = 1):
+         nt: int = 1, rpw: int = 1, maxblocks: int = 0):
     """y[N] = W[N,K] @ stage(x)[K] (+res); stage fuses RMSNorm or GLU
     into the LDS staging pass; nt = non-temporal weight stream."""
     N, K = W.shape
@@ -103,21 +105,22 @@ def gemv(W: torch.Tensor, x: torch.Tensor, y: torch.Tensor,
     _check(lib().launch_gemv_bf16(
         _ptr(W), _ptr(x), _ptr(x2), _ptr(g), _ptr(y), _ptr(res), N, K,
         stage, act, ctypes.c_float(eps), out_f32,
-        ctypes.c_float(softcap), nt, rpw, _stream()), "gemv")
+        ctypes.c_float(softcap), nt, rpw, maxblocks, _stream()), "gemv")
 
 
 def gemv_fp8(Wq: torch.Tensor, scales: torch.Tensor, x: torch.Tensor,
              y: torch.Tensor, res: torch.Tensor | None = None,
              softcap: float = 0.0, stage: int = 0,
              x2: torch.Tensor | None = None, g: torch.Tensor | None = None,
-             act: int = 0, eps: float = 1e-5, nt: int = 1, rpw: int = 1):
+             act: int = 0, eps: float = 1e-5, nt: int = 1, rpw: int = 1,
+             maxblocks: int = 0):
     """y[N] = scales * (Wq[N,K] @ stage(x)); Wq = e4m3fn bytes."""
     N, K = Wq.shape
     out_f32 = 1 if y.dtype == torch.float32 else 0
     _check(lib().launch_gemv_fp8(
         _ptr(Wq), _ptr(scales), _ptr(x), _ptr(x2), _ptr(g), _ptr(y),
         _ptr(res), N, K, stage, act, ctypes.c_float(eps), out_f32,
-        ctypes.c_float(softcap), nt, rpw, _stream()), "gemv_fp8")
+        ctypes.c_float(softcap), nt, rpw, maxblocks, _stream()), "gemv_fp8")
 
 
 def rmsnorm(x: torch.Tensor, g: torch.Tensor, y: torch.Tensor,

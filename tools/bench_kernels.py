@@ -122,5 +122,54 @@ def main():
     print(f"sample min-p  V={V}: {us:.2f}us")
 
 
-if __name__ == "__main__":
+if __name__ == "__main__" and "--layersim" not in sys.argv:
     main()
+
+
+def layersim():
+    """In-context layer simulation: a graph over 16 DISTINCT layers'
+    GEMV weights (2.4 GB total > L3) — honest per-layer time without the
+    full engine; sweeps grid caps and nt."""
+    from csrc.build import ensure_built
+    ensure_built()
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    dev = torch.device("cuda:0")
+    torch.cuda.set_device(dev)
+    H, I, NKV = 2048, 8192, 1024
+    L = 16
+    layers = []
+    for _ in range(L):
+        layers.append(dict(
+            wqkv=torch.randn(H + NKV, H, device=dev).to(torch.bfloat16),
+            wo=torch.randn(H, H, device=dev).to(torch.bfloat16),
+            wgu=torch.randn(2 * I, H, device=dev).to(torch.bfloat16),
+            wdown=torch.randn(H, I, device=dev).to(torch.bfloat16),
+        ))
+    h = torch.randn(H, device=dev).to(torch.bfloat16)
+    qkv = torch.empty(H + NKV, dtype=torch.bfloat16, device=dev)
+    att = torch.randn(H, device=dev).to(torch.bfloat16)
+    gu = torch.empty(2 * I, dtype=torch.bfloat16, device=dev)
+    g1 = torch.randn(H, device=dev)
+
+    def one_pass(nt, cap):
+        for lw in layers:
+            ho.gemv(lw["wqkv"], h, qkv, stage=ho.STAGE_NORM, g=g1,
+                    nt=nt, maxblocks=cap)
+            ho.gemv(lw["wo"], att, h, res=h, nt=nt, maxblocks=cap)
+            ho.gemv(lw["wgu"], h, gu, stage=ho.STAGE_NORM, g=g1,
+                    nt=nt, maxblocks=cap)
+            ho.gemv(lw["wdown"], gu[:I], h, res=h, stage=ho.STAGE_GLU,
+                    x2=gu[I:], nt=nt, maxblocks=cap)
+
+    print("layersim: 16 llama-1b layers of GEMVs (121.6 MB/layer bf16)")
+    for nt in (1, 0):
+        for cap in (256, 512, 1024, 2048):
+            us = time_graph(lambda: one_pass(nt, cap), reps=5, warmup=5)
+            print(f"  nt={nt} cap={cap:5d}: {us/L:7.2f}us/layer "
+                  f"({121.6e6*L/us/1e6:4.2f} TB/s)")
+
+
+if __name__ == "__main__" and "--layersim" in sys.argv:
+    layersim()
+    sys.exit(0)
