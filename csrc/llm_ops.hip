@@ -304,7 +304,7 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
   int blocks = (N + 4 * rpw - 1) / (4 * rpw);
   int cap = maxblocks > 0 ? maxblocks : 1024;
   if (blocks > cap) blocks = cap;  // grid-stride the rest (Guideline 11)
-  if (K <= 4096 && K % 8 == 0 && rpw == 1) {
+  if (false && K <= 4096 && K % 8 == 0 && rpw == 1) {  // measured slower: keep LDS path
     if (nt)
       hipLaunchKernelGGL((k_gemv_bf16_reg<true>), dim3(blocks), dim3(256), 0,
                          stream, (const u16*)W, (const u16*)x, (const u16*)x2,
@@ -537,7 +537,7 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
   int blocks = (N + 4 * rpw - 1) / (4 * rpw);
   int cap = maxblocks > 0 ? maxblocks : 1024;
   if (blocks > cap) blocks = cap;
-  if (K <= 4096 && K % 16 == 0 && rpw == 1) {
+  if (false && K <= 4096 && K % 16 == 0 && rpw == 1) {  // measured slower: keep LDS path
     if (nt)
       hipLaunchKernelGGL((k_gemv_fp8_reg<true>), dim3(blocks), dim3(256), 0,
                          stream, (const uint8_t*)W, (const float*)scales,
