@@ -203,96 +203,6 @@ k_gemv_bf16_t(const u16* __restrict__ W, const u16* __restrict__ x,
   }  // row0 grid-stride loop
 }
 
-// Register-staged GEMV for K <= 4096: one wave's 64 lanes x 8-elem
-// chunks cover the whole input vector, so x (with the fused RMSNorm /
-// GLU pre-op applied) lives in VGPRs for the wave's whole row loop —
-// no LDS, no block barrier, no per-row x reads.  NORM's sum-of-squares
-// reduces within the wave (it sees all of x).
-template <bool NT>
-__global__ void __launch_bounds__(256)
-k_gemv_bf16_reg(const u16* __restrict__ W, const u16* __restrict__ x,
-                const u16* __restrict__ x2, const float* __restrict__ g,
-                void* __restrict__ y, const u16* __restrict__ res,
-                int N, int K, int stage, int act, float eps, int out_f32,
-                float softcap) {
-  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int kbase = lane * 8;
-
-  s8v xr[8];
-#pragma unroll
-  for (int c = 0; c < 8; c++) {
-    int k = kbase + c * 512;
-    xr[c] = (k < K) ? *(const s8v*)(x + k) : s8v{0, 0, 0, 0, 0, 0, 0, 0};
-  }
-  if (stage == STAGE_NORM) {
-    float ss = 0.f;
-#pragma unroll
-    for (int c = 0; c < 8; c++)
-#pragma unroll
-      for (int j = 0; j < 8; j++) {
-        float f = b2f(((u16*)&xr[c])[j]);
-        ss += f * f;
-      }
-    ss = wave_reduce_sum(ss);
-    float rnorm = rsqrtf(ss / (float)K + eps);
-#pragma unroll
-    for (int c = 0; c < 8; c++) {
-      int k = kbase + c * 512;
-      if (k < K) {
-        f4v g0 = *(const f4v*)(g + k);
-        f4v g1 = *(const f4v*)(g + k + 4);
-        u16 o[8];
-#pragma unroll
-        for (int j = 0; j < 8; j++)
-          o[j] = f2b(b2f(((u16*)&xr[c])[j]) * rnorm *
-                     (j < 4 ? g0[j] : g1[j - 4]));
-        xr[c] = *(s8v*)o;
-      }
-    }
-  } else if (stage == STAGE_GLU) {
-#pragma unroll
-    for (int c = 0; c < 8; c++) {
-      int k = kbase + c * 512;
-      if (k < K) {
-        s8v uv = *(const s8v*)(x2 + k);
-        u16 o[8];
-#pragma unroll
-        for (int j = 0; j < 8; j++) {
-          float xx = b2f(((u16*)&xr[c])[j]);
-          float a;
-          if (act == 0) {
-            a = xx / (1.f + __expf(-xx));
-          } else {
-            float cc = 0.7978845608028654f * (xx + 0.044715f * xx * xx * xx);
-            a = 0.5f * xx * (1.f + tanhf(cc));
-          }
-          o[j] = f2b(a * b2f(((u16*)&uv)[j]));
-        }
-        xr[c] = *(s8v*)o;
-      }
-    }
-  }
-
-  const int rstride = gridDim.x * 4;
-  for (int row = blockIdx.x * 4 + wave; row < N; row += rstride) {
-    const u16* Wr = W + (size_t)row * K;
-    float acc = 0.f;
-#pragma unroll
-    for (int c = 0; c < 8; c++) {
-      int k = kbase + c * 512;
-      if (k < K) {
-        s8v w = NT ? __builtin_nontemporal_load((const s8v*)(Wr + k))
-                   : *(const s8v*)(Wr + k);
-#pragma unroll
-        for (int j = 0; j < 8; j++)
-          acc += b2f(((u16*)&w)[j]) * b2f(((u16*)&xr[c])[j]);
-      }
-    }
-    acc = wave_reduce_sum(acc);
-    if (lane == 0) gemv_epilogue(acc, row, y, res, out_f32, softcap);
-  }
-}
-
 extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
                                        const void* x2, const void* g, void* y,
                                        const void* res, int N, int K,
@@ -304,19 +214,6 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
   int blocks = (N + 4 * rpw - 1) / (4 * rpw);
   int cap = maxblocks > 0 ? maxblocks : 1024;
   if (blocks > cap) blocks = cap;  // grid-stride the rest (Guideline 11)
-  if (false && K <= 4096 && K % 8 == 0 && rpw == 1) {  // measured slower: keep LDS path
-    if (nt)
-      hipLaunchKernelGGL((k_gemv_bf16_reg<true>), dim3(blocks), dim3(256), 0,
-                         stream, (const u16*)W, (const u16*)x, (const u16*)x2,
-                         (const float*)g, y, (const u16*)res, N, K, stage,
-                         act, eps, out_f32, softcap);
-    else
-      hipLaunchKernelGGL((k_gemv_bf16_reg<false>), dim3(blocks), dim3(256), 0,
-                         stream, (const u16*)W, (const u16*)x, (const u16*)x2,
-                         (const float*)g, y, (const u16*)res, N, K, stage,
-                         act, eps, out_f32, softcap);
-    return hipGetLastError();
-  }
 #define GEMV_CASE(NTV, RPWV)                                                 \
   hipLaunchKernelGGL((k_gemv_bf16_t<NTV, RPWV>), dim3(blocks), dim3(256),    \
                      lds, stream, (const u16*)W, (const u16*)x,              \
@@ -420,112 +317,6 @@ k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
   }  // row0 grid-stride loop
 }
 
-// Register-staged fp8 GEMV for K <= 4096 (see bf16 twin above): x in
-// VGPRs (16 bf16 per chunk to pair with 16-fp8 dwordx4 weight loads).
-template <bool NT>
-__global__ void __launch_bounds__(256)
-k_gemv_fp8_reg(const uint8_t* __restrict__ W,
-               const float* __restrict__ scales, const u16* __restrict__ x,
-               const u16* __restrict__ x2, const float* __restrict__ g,
-               void* __restrict__ y, const u16* __restrict__ res, int N,
-               int K, int stage, int act, float eps, int out_f32,
-               float softcap) {
-  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int kbase = lane * 16;
-
-  s8v xa[4], xb[4];
-#pragma unroll
-  for (int c = 0; c < 4; c++) {
-    int k = kbase + c * 1024;
-    if (k < K) {
-      xa[c] = *(const s8v*)(x + k);
-      xb[c] = *(const s8v*)(x + k + 8);
-    } else {
-      xa[c] = s8v{0, 0, 0, 0, 0, 0, 0, 0};
-      xb[c] = s8v{0, 0, 0, 0, 0, 0, 0, 0};
-    }
-  }
-  if (stage == STAGE_NORM) {
-    float ss = 0.f;
-#pragma unroll
-    for (int c = 0; c < 4; c++)
-#pragma unroll
-      for (int j = 0; j < 8; j++) {
-        float fa = b2f(((u16*)&xa[c])[j]);
-        float fb = b2f(((u16*)&xb[c])[j]);
-        ss += fa * fa + fb * fb;
-      }
-    ss = wave_reduce_sum(ss);
-    float rnorm = rsqrtf(ss / (float)K + eps);
-#pragma unroll
-    for (int c = 0; c < 4; c++) {
-      int k = kbase + c * 1024;
-      if (k < K) {
-        u16 oa[8], ob[8];
-#pragma unroll
-        for (int j = 0; j < 8; j++) {
-          oa[j] = f2b(b2f(((u16*)&xa[c])[j]) * rnorm * g[k + j]);
-          ob[j] = f2b(b2f(((u16*)&xb[c])[j]) * rnorm * g[k + 8 + j]);
-        }
-        xa[c] = *(s8v*)oa;
-        xb[c] = *(s8v*)ob;
-      }
-    }
-  } else if (stage == STAGE_GLU) {
-#pragma unroll
-    for (int c = 0; c < 4; c++) {
-      int k = kbase + c * 1024;
-      if (k < K) {
-        s8v ua = *(const s8v*)(x2 + k);
-        s8v ub = *(const s8v*)(x2 + k + 8);
-        u16 oa[8], ob[8];
-#pragma unroll
-        for (int j = 0; j < 8; j++) {
-          float g1 = b2f(((u16*)&xa[c])[j]);
-          float g2 = b2f(((u16*)&xb[c])[j]);
-          float a1, a2;
-          if (act == 0) {
-            a1 = g1 / (1.f + __expf(-g1));
-            a2 = g2 / (1.f + __expf(-g2));
-          } else {
-            float c1 = 0.7978845608028654f * (g1 + 0.044715f * g1 * g1 * g1);
-            float c2 = 0.7978845608028654f * (g2 + 0.044715f * g2 * g2 * g2);
-            a1 = 0.5f * g1 * (1.f + tanhf(c1));
-            a2 = 0.5f * g2 * (1.f + tanhf(c2));
-          }
-          oa[j] = f2b(a1 * b2f(((u16*)&ua)[j]));
-          ob[j] = f2b(a2 * b2f(((u16*)&ub)[j]));
-        }
-        xa[c] = *(s8v*)oa;
-        xb[c] = *(s8v*)ob;
-      }
-    }
-  }
-
-  const int rstride = gridDim.x * 4;
-  for (int row = blockIdx.x * 4 + wave; row < N; row += rstride) {
-    const uint8_t* Wr = W + (size_t)row * K;
-    float acc = 0.f;
-#pragma unroll
-    for (int c = 0; c < 4; c++) {
-      int k = kbase + c * 1024;
-      if (k < K) {
-        u4v w = NT ? __builtin_nontemporal_load((const u4v*)(Wr + k))
-                   : *(const u4v*)(Wr + k);
-        float f[16];
-        fp8x16_to_f32(w, f);
-#pragma unroll
-        for (int j = 0; j < 8; j++) {
-          acc += f[j] * b2f(((u16*)&xa[c])[j]);
-          acc += f[j + 8] * b2f(((u16*)&xb[c])[j]);
-        }
-      }
-    }
-    acc = wave_reduce_sum(acc) * scales[row];
-    if (lane == 0) gemv_epilogue(acc, row, y, res, out_f32, softcap);
-  }
-}
-
 extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
                                       const void* x, const void* x2,
                                       const void* g, void* y, const void* res,
@@ -537,21 +328,6 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
   int blocks = (N + 4 * rpw - 1) / (4 * rpw);
   int cap = maxblocks > 0 ? maxblocks : 1024;
   if (blocks > cap) blocks = cap;
-  if (false && K <= 4096 && K % 16 == 0 && rpw == 1) {  // measured slower: keep LDS path
-    if (nt)
-      hipLaunchKernelGGL((k_gemv_fp8_reg<true>), dim3(blocks), dim3(256), 0,
-                         stream, (const uint8_t*)W, (const float*)scales,
-                         (const u16*)x, (const u16*)x2, (const float*)g, y,
-                         (const u16*)res, N, K, stage, act, eps, out_f32,
-                         softcap);
-    else
-      hipLaunchKernelGGL((k_gemv_fp8_reg<false>), dim3(blocks), dim3(256), 0,
-                         stream, (const uint8_t*)W, (const float*)scales,
-                         (const u16*)x, (const u16*)x2, (const float*)g, y,
-                         (const u16*)res, N, K, stage, act, eps, out_f32,
-                         softcap);
-    return hipGetLastError();
-  }
 #define GEMV8_CASE(NTV, RPWV)                                               \
   hipLaunchKernelGGL((k_gemv_fp8_t<NTV, RPWV>), dim3(blocks), dim3(256),    \
                      lds, stream, (const uint8_t*)W, (const float*)scales,  \
