@@ -44,7 +44,7 @@ class GPUModel:
     def __init__(self, config: ModelConfig, weights: Dict[str, np.ndarray],
                  dtype: str = "bf16", max_seq: int = 4096,
                  prefill_chunk: int = 512, device: Optional[str] = None,
-                 seed: int = 0):
+                 seed: int = 0, force_tp_path: bool = False):
         if not torch.cuda.is_available():
             raise RuntimeError("GPUModel requires a GPU (MI355X)")
         self.config = config
@@ -55,6 +55,9 @@ class GPUModel:
             raise ValueError(f"unsupported dtype {dtype!r}")
         self.fp8 = dtype == "fp8"
         self.rank, self.world = tpu.init_distributed()
+        # exercise the TP code path on 1 GPU (collectives no-op at
+        # world=1, partial sums are then exact) — used by tests
+        self.tp_branch = self.world > 1 or force_tp_path
         if device is None:
             device = f"cuda:{self.rank % max(torch.cuda.device_count(), 1)}"
         self.device = torch.device(device)
@@ -192,7 +195,7 @@ class GPUModel:
                                       device=dev)
         self.b_logits_l = torch.zeros(self.vocab_l, dtype=torch.float32,
                                       device=dev)
-        self.b_logits = (self.b_logits_l if self.world == 1 else
+        self.b_logits = (self.b_logits_l if not self.tp_branch else
                          torch.zeros(cfg.vocab_size, dtype=torch.float32,
                                      device=dev))
         self.b_qkv = torch.zeros((self.nh_l + 2 * self.kvh_l) * hd, **bf)
@@ -267,7 +270,7 @@ class GPUModel:
                 ho.rmsnorm(t1[:M], lw["g_post"], h[:M], res=h[:M], eps=eps)
                 ho.rmsnorm(h[:M], lw["g_preffn"], xn[:M], eps=eps)
             else:
-                if self.world > 1:
+                if self.tp_branch:
                     self._linear(lw["wo"], self.b_att, t1, M=M)
                     tpu.all_reduce(t1[:M])
                     ho.addinto(h[:M], t1[:M])
@@ -282,7 +285,7 @@ class GPUModel:
                 tpu.all_reduce(t1[:M])
                 ho.rmsnorm(t1[:M], lw["g_postffn"], h[:M], res=h[:M], eps=eps)
             else:
-                if self.world > 1:
+                if self.tp_branch:
                     self._linear(lw["wdown"], self.b_gate, t1, M=M)
                     tpu.all_reduce(t1[:M])
                     ho.addinto(h[:M], t1[:M])
@@ -296,7 +299,7 @@ class GPUModel:
                    eps=self.config.rms_norm_eps)
         ho.gemv(self.lm_head, self.b_xn[0], self.b_logits_l,
                 softcap=self.final_softcap)
-        if self.world > 1:
+        if self.tp_branch:
             tpu.all_gather_into(self.b_logits, self.b_logits_l)
 
     # ------------------------------------------------------------------
@@ -366,7 +369,7 @@ class GPUModel:
                 tpu.all_reduce(t1)
                 ho.rmsnorm(t1, lw["g_postffn"], h, res=h, eps=eps)
             else:
-                if self.world > 1:
+                if self.tp_branch:
                     self._dgemv(lw, "wo", self.b_att[0], t1)
                     tpu.all_reduce(t1)
                     ho.addinto(h, t1)
@@ -374,7 +377,7 @@ class GPUModel:
                     self._dgemv(lw, "wo", self.b_att[0], h, res=h)
                 self._dgemv(lw, "wgu", h, self.b_gu, stage=ho.STAGE_NORM,
                             g=lw["g_post"], eps=eps)
-                if self.world > 1:
+                if self.tp_branch:
                     self._dgemv(lw, "wdown", self.b_gu[:self.inter_l], t1,
                                 stage=ho.STAGE_GLU,
                                 x2=self.b_gu[self.inter_l:], act=self.act)
@@ -391,7 +394,7 @@ class GPUModel:
         else:
             ho.gemv(self.lm_head, self.b_xn[0], self.b_logits_l,
                     softcap=self.final_softcap)
-        if self.world > 1:
+        if self.tp_branch:
             tpu.all_gather_into(self.b_logits, self.b_logits_l)
         ho.sample(self.b_logits, min_p, greedy, self.seed, self.rng_ctr,
                   self.s_gmax, self.s_pick, self.next_token, self.out_ring,

@@ -161,3 +161,47 @@ def test_fp8_graph_decode_matches_eager():
     m.prefill(prompt)
     b = m.decode(8, greedy=True, use_graph=True)
     np.testing.assert_array_equal(a, b)
+
+
+def test_tp_branch_matches_plain_on_one_gpu():
+    """The TP code path (partial GEMV -> all-reduce -> add, vocab shard +
+    gather) must produce identical greedy ids at world=1 where the
+    collectives are no-ops — catches TP-branch buffer/shape bugs before
+    a real multi-GPU run."""
+    import llm_np_cp_amd as L
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.engine import GPUModel
+
+    cfg = L.preset_config("tiny-llama")
+    w = random_weights(cfg, seed=21)
+    plain = GPUModel(cfg, dict(w), max_seq=128)
+    tp = GPUModel(cfg, dict(w), max_seq=128, force_tp_path=True)
+    prompt = np.arange(1, 10)
+
+    plain.prefill(prompt)
+    a = plain.decode(8, greedy=True, use_graph=False)
+    tp.prefill(prompt)
+    b = tp.decode(8, greedy=True, use_graph=False)
+    np.testing.assert_array_equal(a, b)
+
+    # and through the graph path
+    tp.prefill(prompt)
+    c = tp.decode(8, greedy=True, use_graph=True)
+    np.testing.assert_array_equal(a, c)
+
+
+def test_tp_branch_gemma():
+    import llm_np_cp_amd as L
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.engine import GPUModel
+
+    cfg = L.preset_config("tiny-gemma2")
+    w = random_weights(cfg, seed=22)
+    plain = GPUModel(cfg, dict(w), max_seq=128)
+    tp = GPUModel(cfg, dict(w), max_seq=128, force_tp_path=True)
+    prompt = np.arange(1, 8)
+    plain.prefill(prompt)
+    a = plain.decode(6, greedy=True, use_graph=False)
+    tp.prefill(prompt)
+    b = tp.decode(6, greedy=True, use_graph=False)
+    np.testing.assert_array_equal(a, b)
