@@ -67,12 +67,78 @@ DEVINL float wave_reduce_sum(float v) {
 #define STAGE_RAW 0
 #define STAGE_NORM 1
 #define STAGE_GLU 2
+// NORM2 (Gemma sandwich fusion): h' = h + rmsnorm(x)*g  (post-norm +
+// residual), block 0 persists h' to global, then xs = rmsnorm(h')*g2
+// (pre-norm of the projection this GEMV computes).  Removes the two
+// standalone k_rmsnorm launches per Gemma layer.
+#define STAGE_NORM2 3
 
 // staging helper shared by bf16/fp8 GEMV (RMSNorm / GLU fused pre-ops)
 DEVINL const u16* gemv_stage(char* smem, const u16* x, const u16* x2,
-                             const float* g, int K, int stage, int act,
-                             float eps) {
+                             const float* g, const float* g2, u16* hout,
+                             int K, int stage, int act, float eps) {
   u16* xs = (u16*)smem;
+  if (stage == STAGE_NORM2) {
+    // x2 = h_in (read-only here); block 0 persists h' into hout — a
+    // DIFFERENT buffer (ping-pong), since other blocks still read h_in
+    const u16* h = x2;
+    float* red = (float*)(smem + (size_t)K * 2);
+    // pass 1: sumsq of x (the un-normed projection output t)
+    float ss = 0.f;
+    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+      s8v v = *(const s8v*)(x + i);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = b2f(((u16*)&v)[j]);
+        ss += f * f;
+      }
+    }
+    ss = wave_reduce_sum(ss);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+    __syncthreads();
+    float rnorm_a = rsqrtf((red[0] + red[1] + red[2] + red[3]) / (float)K +
+                           eps);
+    __syncthreads();
+    // pass 2: h' = h + norm_a(x)*g (bf16-rounded, matching the
+    // standalone k_rmsnorm mode 1), stash h' in LDS, sumsq of h'
+    float ss2 = 0.f;
+    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+      s8v v = *(const s8v*)(x + i);
+      s8v hv = *(const s8v*)(h + i);
+      f4v ga = *(const f4v*)(g + i);
+      f4v gb = *(const f4v*)(g + i + 4);
+      u16 o[8];
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = b2f(((u16*)&v)[j]) * rnorm_a * (j < 4 ? ga[j] : gb[j - 4])
+                  + b2f(((u16*)&hv)[j]);
+        o[j] = f2b(f);
+        float fr = b2f(o[j]);
+        ss2 += fr * fr;
+      }
+      *(s8v*)(xs + i) = *(s8v*)o;
+      if (blockIdx.x == 0) *(s8v*)(hout + i) = *(s8v*)o;  // persist h'
+    }
+    ss2 = wave_reduce_sum(ss2);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss2;
+    __syncthreads();
+    float rnorm_b = rsqrtf((red[0] + red[1] + red[2] + red[3]) / (float)K +
+                           eps);
+    // pass 3: xs = norm_b(h')*g2
+    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+      s8v v = *(s8v*)(xs + i);
+      f4v ga = *(const f4v*)(g2 + i);
+      f4v gb = *(const f4v*)(g2 + i + 4);
+      u16 o[8];
+#pragma unroll
+      for (int j = 0; j < 8; j++)
+        o[j] = f2b(b2f(((u16*)&v)[j]) * rnorm_b *
+                   (j < 4 ? ga[j] : gb[j - 4]));
+      *(s8v*)(xs + i) = *(s8v*)o;
+    }
+    __syncthreads();
+    return xs;
+  }
   if (stage == STAGE_NORM) {
     float ss = 0.f;
     for (int i = threadIdx.x * 8; i < K; i += 2048) {
@@ -136,11 +202,13 @@ template <bool NT, int RPW>
 __global__ void __launch_bounds__(256)
 k_gemv_bf16_t(const u16* __restrict__ W, const u16* __restrict__ x,
               const u16* __restrict__ x2, const float* __restrict__ g,
-              void* __restrict__ y, const u16* __restrict__ res,
-              int N, int K, int stage, int act, float eps, int out_f32,
-              float softcap) {
+              const float* __restrict__ g2, void* __restrict__ y,
+              const u16* __restrict__ res, int N, int K, int stage, int act,
+              float eps, int out_f32, float softcap) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const u16* xv = gemv_stage(smem, x, x2, g, K, stage, act, eps);
+  const u16* xv = gemv_stage(smem, x, x2, g, g2, (u16*)res, K, stage, act,
+                             eps);
+  const u16* eres = (stage == STAGE_NORM2) ? nullptr : res;  // res = hout
 
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int rstride = gridDim.x * 4 * RPW;  // grid-stride rows: amortize
@@ -198,13 +266,14 @@ k_gemv_bf16_t(const u16* __restrict__ W, const u16* __restrict__ x,
   for (int r = 0; r < RPW; r++) {
     float acc = wave_reduce_sum((a0[r] + a1[r]) + (a2[r] + a3[r]));
     if (lane == 0 && row0 + r < N)
-      gemv_epilogue(acc, row0 + r, y, res, out_f32, softcap);
+      gemv_epilogue(acc, row0 + r, y, eres, out_f32, softcap);
   }
   }  // row0 grid-stride loop
 }
 
 extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
-                                       const void* x2, const void* g, void* y,
+                                       const void* x2, const void* g,
+                                       const void* g2, void* y,
                                        const void* res, int N, int K,
                                        int stage, int act, float eps,
                                        int out_f32, float softcap,
@@ -217,8 +286,9 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
 #define GEMV_CASE(NTV, RPWV)                                                 \
   hipLaunchKernelGGL((k_gemv_bf16_t<NTV, RPWV>), dim3(blocks), dim3(256),    \
                      lds, stream, (const u16*)W, (const u16*)x,              \
-                     (const u16*)x2, (const float*)g, y, (const u16*)res, N, \
-                     K, stage, act, eps, out_f32, softcap)
+                     (const u16*)x2, (const float*)g, (const float*)g2, y,   \
+                     (const u16*)res, N, K, stage, act, eps, out_f32,        \
+                     softcap)
   if (nt && rpw == 2) GEMV_CASE(true, 2);
   else if (nt) GEMV_CASE(true, 1);
   else if (rpw == 2) GEMV_CASE(false, 2);
@@ -252,11 +322,13 @@ template <bool NT, int RPW>
 __global__ void __launch_bounds__(256)
 k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
              const u16* __restrict__ x, const u16* __restrict__ x2,
-             const float* __restrict__ g, void* __restrict__ y,
-             const u16* __restrict__ res, int N, int K, int stage, int act,
-             float eps, int out_f32, float softcap) {
+             const float* __restrict__ g, const float* __restrict__ g2,
+             void* __restrict__ y, const u16* __restrict__ res, int N, int K,
+             int stage, int act, float eps, int out_f32, float softcap) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const u16* xv = gemv_stage(smem, x, x2, g, K, stage, act, eps);
+  const u16* xv = gemv_stage(smem, x, x2, g, g2, (u16*)res, K, stage, act,
+                             eps);
+  const u16* eres = (stage == STAGE_NORM2) ? nullptr : res;
 
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int rstride = gridDim.x * 4 * RPW;
@@ -312,17 +384,18 @@ k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
     int rr = row0 + r < N ? row0 + r : N - 1;
     float acc = wave_reduce_sum(a0[r] + a1[r]) * scales[rr];
     if (lane == 0 && row0 + r < N)
-      gemv_epilogue(acc, row0 + r, y, res, out_f32, softcap);
+      gemv_epilogue(acc, row0 + r, y, eres, out_f32, softcap);
   }
   }  // row0 grid-stride loop
 }
 
 extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
                                       const void* x, const void* x2,
-                                      const void* g, void* y, const void* res,
-                                      int N, int K, int stage, int act,
-                                      float eps, int out_f32, float softcap,
-                                      int nt, int rpw, int maxblocks,
+                                      const void* g, const void* g2, void* y,
+                                      const void* res, int N, int K,
+                                      int stage, int act, float eps,
+                                      int out_f32, float softcap, int nt,
+                                      int rpw, int maxblocks,
                                       hipStream_t stream) {
   size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 16);
   int blocks = (N + 4 * rpw - 1) / (4 * rpw);
@@ -331,9 +404,9 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
 #define GEMV8_CASE(NTV, RPWV)                                               \
   hipLaunchKernelGGL((k_gemv_fp8_t<NTV, RPWV>), dim3(blocks), dim3(256),    \
                      lds, stream, (const uint8_t*)W, (const float*)scales,  \
-                     (const u16*)x, (const u16*)x2, (const float*)g, y,     \
-                     (const u16*)res, N, K, stage, act, eps, out_f32,       \
-                     softcap)
+                     (const u16*)x, (const u16*)x2, (const float*)g,        \
+                     (const float*)g2, y, (const u16*)res, N, K, stage,     \
+                     act, eps, out_f32, softcap)
   if (nt && rpw == 2) GEMV8_CASE(true, 2);
   else if (nt) GEMV8_CASE(true, 1);
   else if (rpw == 2) GEMV8_CASE(false, 2);

@@ -199,6 +199,8 @@ class GPUModel:
                          torch.zeros(cfg.vocab_size, dtype=torch.float32,
                                      device=dev))
         self.b_qkv = torch.zeros((self.nh_l + 2 * self.kvh_l) * hd, **bf)
+        self.b_hb = torch.zeros(H, **bf)   # gemma residual ping-pong
+        self.b_t2 = torch.zeros(H, **bf)   # gemma ffn-out delta
         self.b_gu = torch.zeros(2 * I, **bf)
         # split-T decode attention scratch (partials + per-head tickets)
         self.attn_split = min(16, max(4, (self.max_seq + 255) // 256))
@@ -345,12 +347,24 @@ class GPUModel:
         cfg = self.config
         eps = cfg.rms_norm_eps
         h = self.b_h[0]
+        hnext = self.b_hb
         t1 = self.b_t1[0]
+        t2 = self.b_t2
         ho.embed(self.embed, self.next_token, self.b_h, 1, cfg.embed_scale)
+        prev = None  # previous gemma layer (its postffn norm fused here)
         for i, lw in enumerate(self.layers):
             window = cfg.sliding_window if cfg.is_sliding(i) else 0
-            self._dgemv(lw, "wqkv", h, self.b_qkv, stage=ho.STAGE_NORM,
-                        g=lw["g_in"], eps=eps)
+            if self.gemma and prev is not None:
+                # h' = h + norm(t2_prev)*g_postffn_prev; stage norm(h')*g_in
+                self._dgemv(lw, "wqkv", t2, self.b_qkv,
+                            stage=ho.STAGE_NORM2, x2=h,
+                            g=prev["g_postffn"], g2=lw["g_in"], res=hnext,
+                            eps=eps)
+                h, hnext = hnext, h
+            else:
+                self._dgemv(lw, "wqkv", h, self.b_qkv, stage=ho.STAGE_NORM,
+                            g=lw["g_in"], eps=eps)
+            prev = lw
             ho.attn_dec(self.b_qkv, self.k_cache[i], self.v_cache[i],
                         self.b_att[0], self.len_buf, self.cos_t, self.sin_t,
                         self.attn_scratch, self.attn_cnt,
@@ -360,14 +374,16 @@ class GPUModel:
             if self.gemma:
                 self._dgemv(lw, "wo", self.b_att[0], t1)
                 tpu.all_reduce(t1)
-                ho.rmsnorm(t1, lw["g_post"], h, res=h, eps=eps)
-                self._dgemv(lw, "wgu", h, self.b_gu, stage=ho.STAGE_NORM,
-                            g=lw["g_preffn"], eps=eps)
-                self._dgemv(lw, "wdown", self.b_gu[:self.inter_l], t1,
+                # NORM2: h' = h + norm(t1)*g_post (persisted to hnext),
+                # stage = norm(h')*g_preffn — both sandwich norms fused
+                self._dgemv(lw, "wgu", t1, self.b_gu, stage=ho.STAGE_NORM2,
+                            x2=h, g=lw["g_post"], g2=lw["g_preffn"],
+                            res=hnext, eps=eps)
+                h, hnext = hnext, h
+                self._dgemv(lw, "wdown", self.b_gu[:self.inter_l], t2,
                             stage=ho.STAGE_GLU, x2=self.b_gu[self.inter_l:],
                             act=self.act)
-                tpu.all_reduce(t1)
-                ho.rmsnorm(t1, lw["g_postffn"], h, res=h, eps=eps)
+                tpu.all_reduce(t2)
             else:
                 if self.tp_branch:
                     self._dgemv(lw, "wo", self.b_att[0], t1)
@@ -387,6 +403,10 @@ class GPUModel:
                     self._dgemv(lw, "wdown", self.b_gu[:self.inter_l], h,
                                 res=h, stage=ho.STAGE_GLU,
                                 x2=self.b_gu[self.inter_l:], act=self.act)
+        if self.gemma:
+            # last layer's post-ffn sandwich norm (no following projection
+            # to fold it into)
+            ho.rmsnorm(t2, self.layers[-1]["g_postffn"], h, res=h, eps=eps)
         ho.rmsnorm(h, self.g_final, self.b_xn[0], eps=eps)
         if self.fp8:
             ho.gemv_fp8(self.lm_head_q, self.lm_head_s, self.b_xn[0],

@@ -24,11 +24,11 @@ _LIB_PATH = os.path.join(os.path.dirname(os.path.dirname(
     os.path.abspath(__file__))), "_libllmops.so")
 
 _SIGS = {
-    "launch_gemv_bf16": [ctypes.c_void_p] * 6 + [ctypes.c_int] * 4 +
+    "launch_gemv_bf16": [ctypes.c_void_p] * 7 + [ctypes.c_int] * 4 +
                         [ctypes.c_float, ctypes.c_int, ctypes.c_float,
                          ctypes.c_int, ctypes.c_int, ctypes.c_int,
                          ctypes.c_void_p],
-    "launch_gemv_fp8": [ctypes.c_void_p] * 7 + [ctypes.c_int] * 4 +
+    "launch_gemv_fp8": [ctypes.c_void_p] * 8 + [ctypes.c_int] * 4 +
                        [ctypes.c_float, ctypes.c_int, ctypes.c_float,
                         ctypes.c_int, ctypes.c_int, ctypes.c_int,
                         ctypes.c_void_p],
@@ -90,21 +90,23 @@ def _ptr(t) -> int:
 # op wrappers (shapes validated here; kernels trust their args)
 # ----------------------------------------------------------------------
 
-STAGE_RAW, STAGE_NORM, STAGE_GLU = 0, 1, 2
+STAGE_RAW, STAGE_NORM, STAGE_GLU, STAGE_NORM2 = 0, 1, 2, 3
 
 
 def gemv(W: torch.Tensor, x: torch.Tensor, y: torch.Tensor,
          res: torch.Tensor | None = None, softcap: float = 0.0,
          stage: int = 0, x2: torch.Tensor | None = None,
          g: torch.Tensor | None = None, act: int = 0, eps: float = 1e-5,
-         nt: int = 1, rpw: int = 1, maxblocks: int = 0):
-    """y[N] = W[N,K] @ stage(x)[K] (+res); stage fuses RMSNorm or GLU
-    into the LDS staging pass; nt = non-temporal weight stream."""
+         nt: int = 1, rpw: int = 1, maxblocks: int = 0,
+         g2: torch.Tensor | None = None):
+    """y[N] = W[N,K] @ stage(x)[K] (+res); stage fuses RMSNorm / GLU /
+    the Gemma sandwich (NORM2: x2 = h_in, g/g2 = post/pre gammas, res =
+    h_out ping-pong) into the LDS staging pass; nt = non-temporal W."""
     N, K = W.shape
     out_f32 = 1 if y.dtype == torch.float32 else 0
     _check(lib().launch_gemv_bf16(
-        _ptr(W), _ptr(x), _ptr(x2), _ptr(g), _ptr(y), _ptr(res), N, K,
-        stage, act, ctypes.c_float(eps), out_f32,
+        _ptr(W), _ptr(x), _ptr(x2), _ptr(g), _ptr(g2), _ptr(y), _ptr(res),
+        N, K, stage, act, ctypes.c_float(eps), out_f32,
         ctypes.c_float(softcap), nt, rpw, maxblocks, _stream()), "gemv")
 
 
@@ -113,13 +115,13 @@ def gemv_fp8(Wq: torch.Tensor, scales: torch.Tensor, x: torch.Tensor,
              softcap: float = 0.0, stage: int = 0,
              x2: torch.Tensor | None = None, g: torch.Tensor | None = None,
              act: int = 0, eps: float = 1e-5, nt: int = 1, rpw: int = 1,
-             maxblocks: int = 0):
+             maxblocks: int = 0, g2: torch.Tensor | None = None):
     """y[N] = scales * (Wq[N,K] @ stage(x)); Wq = e4m3fn bytes."""
     N, K = Wq.shape
     out_f32 = 1 if y.dtype == torch.float32 else 0
     _check(lib().launch_gemv_fp8(
-        _ptr(Wq), _ptr(scales), _ptr(x), _ptr(x2), _ptr(g), _ptr(y),
-        _ptr(res), N, K, stage, act, ctypes.c_float(eps), out_f32,
+        _ptr(Wq), _ptr(scales), _ptr(x), _ptr(x2), _ptr(g), _ptr(g2),
+        _ptr(y), _ptr(res), N, K, stage, act, ctypes.c_float(eps), out_f32,
         ctypes.c_float(softcap), nt, rpw, maxblocks, _stream()), "gemv_fp8")
 
 

@@ -403,3 +403,29 @@ def test_gemv_fp8_fused_stages():
     torch.cuda.synchronize()
     xs = (torch.nn.functional.silu(xf) * up.float()).to(torch.bfloat16).float()
     assert_close(y, Wd @ xs)
+
+
+def test_gemv_fused_norm2_stage():
+    """STAGE_NORM2 (Gemma sandwich): h' = h + rmsnorm(t)*g_a persisted to
+    hout; y = W @ (rmsnorm(h')*g_b)."""
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    N, K, eps = 512, 2304, 1e-6
+    W = randn_bf16(N, K, seed=120, scale=0.05)
+    t = randn_bf16(K, seed=121)
+    hin = randn_bf16(K, seed=122)
+    hout = torch.zeros_like(hin)
+    ga = torch.randn(K, generator=torch.Generator().manual_seed(123)).to(dev())
+    gb = torch.randn(K, generator=torch.Generator().manual_seed(124)).to(dev())
+    y = torch.empty(N, dtype=torch.bfloat16, device=dev())
+    ho.gemv(W, t, y, stage=ho.STAGE_NORM2, x2=hin, g=ga, g2=gb, res=hout,
+            eps=eps)
+    torch.cuda.synchronize()
+
+    tf = t.float()
+    tn = tf * torch.rsqrt(tf.pow(2).mean() + eps) * ga
+    hp = (tn + hin.float()).to(torch.bfloat16).float()
+    xn = (hp * torch.rsqrt(hp.pow(2).mean() + eps) * gb).to(
+        torch.bfloat16).float()
+    assert_close(y, W.float() @ xn)
+    assert_close(hout, hp)
