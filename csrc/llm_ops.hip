@@ -1300,6 +1300,32 @@ extern "C" hipError_t launch_gemm_bf16(const void* X, const void* W, void* Y,
 }
 
 // ====================================================================
+// elementwise tanh soft-cap: y = cap * tanh(x / cap)  (bf16)
+// (Gemma-2 final-logit capping on the all-positions GEMM path; the
+// decode GEMV fuses this into its epilogue instead)
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_softcap(u16* __restrict__ y, long total, float cap) {
+  long i = ((long)blockIdx.x * 256 + threadIdx.x) * 8;
+  if (i >= total) return;
+  s8v v = *(const s8v*)(y + i);
+  u16 o[8];
+#pragma unroll
+  for (int j = 0; j < 8; j++)
+    o[j] = f2b(cap * tanhf(b2f(((u16*)&v)[j]) / cap));
+  *(s8v*)(y + i) = *(s8v*)o;
+}
+
+extern "C" hipError_t launch_softcap(void* y, long total, float cap,
+                                     hipStream_t stream) {
+  long blocks = (total / 8 + 255) / 256;
+  hipLaunchKernelGGL(k_softcap, dim3((uint32_t)blocks), dim3(256), 0, stream,
+                     (u16*)y, total, cap);
+  return hipGetLastError();
+}
+
+// ====================================================================
 // y += a (bf16, fp32 math).  Used on the TP path where the RCCL
 // all-reduce sits between the row-parallel GEMV and the residual add.
 // ====================================================================

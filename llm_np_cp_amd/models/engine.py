@@ -304,6 +304,47 @@ class GPUModel:
         if self.tp_branch:
             tpu.all_gather_into(self.b_logits, self.b_logits_l)
 
+    def forward_full(self, ids: np.ndarray) -> np.ndarray:
+        """All-positions logits (M, V) — reference-parity API (the HF
+        tuple shape).  Prefill-style pass; vocab GEMM per
+        chunk.  Resets the cache."""
+        ids = np.asarray(ids, dtype=np.int32).ravel()
+        self.reset()
+        n = len(ids)
+        out = np.empty((n, self.config.vocab_size), dtype=np.float32)
+        logits_buf = torch.empty(self.PC, self.vocab_l,
+                                 dtype=torch.bfloat16, device=self.device)
+        done = 0
+        while done < n:
+            M = min(n - done, self.PC)
+            self.ids_buf[:M].copy_(
+                torch.from_numpy(ids[done:done + M].astype(np.int32)))
+            ho.i32_set(self.len_buf, done)
+            ho.embed(self.embed, self.ids_buf, self.b_h, M,
+                     self.config.embed_scale)
+            self._layers_forward(M)
+            ho.rmsnorm(self.b_h[:M], self.g_final, self.b_xn[:M],
+                       eps=self.config.rms_norm_eps)
+            ho.gemm(self.b_xn[:M], self.lm_head, logits_buf[:M],
+                    accbuf=self.b_gemm_acc)
+            if self.final_softcap:
+                ho.softcap(logits_buf[:M], self.final_softcap)
+            torch.cuda.synchronize()
+            loc = logits_buf[:M].float().cpu().numpy()
+            if self.world > 1:
+                t = torch.from_numpy(loc).to(self.device)
+                full = torch.zeros(M, self.config.vocab_size,
+                                   device=self.device)
+                import torch.distributed as dist
+                chunks = [torch.empty_like(t) for _ in range(self.world)]
+                dist.all_gather(chunks, t)
+                full = torch.cat(chunks, dim=1)
+                loc = full.cpu().numpy()
+            out[done:done + M] = loc
+            done += M
+        ho.i32_set(self.len_buf, n)
+        return out
+
     # ------------------------------------------------------------------
     # generic forward (oracle-parity / generate()-compatible)
     # ------------------------------------------------------------------

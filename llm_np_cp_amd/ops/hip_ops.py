@@ -51,6 +51,8 @@ _SIGS = {
                      [ctypes.c_int, ctypes.c_void_p],
     "launch_gemm_bf16": [ctypes.c_void_p] * 5 + [ctypes.c_int] * 3 +
                         [ctypes.c_void_p],
+    "launch_softcap": [ctypes.c_void_p, ctypes.c_long, ctypes.c_float,
+                       ctypes.c_void_p],
     "launch_addinto": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_long,
                        ctypes.c_void_p],
     "launch_i32_set": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
@@ -218,6 +220,14 @@ def gemm(X: torch.Tensor, W: torch.Tensor, Y: torch.Tensor,
         accbuf = None
     _check(lib().launch_gemm_bf16(_ptr(X), _ptr(W), _ptr(Y), _ptr(res),
                                   _ptr(accbuf), M, N, K, _stream()), "gemm")
+
+
+def softcap(y: torch.Tensor, cap: float):
+    """in-place y = cap * tanh(y / cap) (bf16)."""
+    total = y.numel()
+    assert total % 8 == 0
+    _check(lib().launch_softcap(_ptr(y), total, ctypes.c_float(cap),
+                                _stream()), "softcap")
 
 
 def addinto(y: torch.Tensor, a: torch.Tensor):
