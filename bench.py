@@ -24,15 +24,17 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=200)
     ap.add_argument("--warmup", type=int, default=20)
-    ap.add_argument("--model", type=str, default=None,
-                    help="preset override (default: llama-3.2-1b at N=1, "
-                         "gemma-2-9b at N>1)")
+    ap.add_argument("--model", type=str, default="llama-3.2-1b",
+                    help="preset or checkpoint dir (default llama-3.2-1b "
+                         "at every N so the driver's 1/2/4/8-GPU scaling "
+                         "curve compares like with like; gemma-2-9b &c "
+                         "via this flag)")
     ap.add_argument("--prompt-len", type=int, default=64)
     ap.add_argument("--max-seq", type=int, default=None)
-    ap.add_argument("--dtype", type=str, default=None,
+    ap.add_argument("--dtype", type=str, default="bf16",
                     choices=["bf16", "fp8"],
-                    help="default: bf16 at N=1, fp8 at N>1 (BASELINE "
-                         "config 4: Gemma-2-9B TP=8 fp8 weights)")
+                    help="weight dtype (constant across N for honest "
+                         "scaling; fp8 = BASELINE config 4)")
     ap.add_argument("--no-graph", action="store_true")
     args = ap.parse_args()
 
@@ -41,7 +43,7 @@ def main():
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     n_gpus = max(world, args.gpus)
-    model_name = args.model or ("llama-3.2-1b" if n_gpus == 1 else "gemma-2-9b")
+    model_name = args.model
 
     from csrc.build import ensure_built
     ensure_built()
@@ -52,7 +54,7 @@ def main():
     from llm_np_cp_amd.parallel.tp import init_distributed
 
     rank, world = init_distributed()
-    dtype = args.dtype or ("bf16" if world == 1 else "fp8")
+    dtype = args.dtype
     cfg = L.preset_config(model_name)
     max_seq = args.max_seq or min(
         cfg.max_position_embeddings,
