@@ -1300,6 +1300,31 @@ extern "C" hipError_t launch_gemm_bf16(const void* X, const void* W, void* Y,
 }
 
 // ====================================================================
+// Weight prefetcher (side-stream): stream a tensor through L2/MALL with
+// plain (retaining) loads so the 256 MB Infinity Cache holds the NEXT
+// layer's weights before its compute kernel issues.  The sink write
+// keeps the loads alive; modest grid so compute keeps its CUs.
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_prefetch(const u16* __restrict__ p, long n, float* __restrict__ sink) {
+  float acc = 0.f;
+  for (long i = ((long)blockIdx.x * 256 + threadIdx.x) * 8; i < n;
+       i += (long)gridDim.x * 256 * 8) {
+    s8v v = *(const s8v*)(p + i);
+    acc += b2f(((u16*)&v)[0]);  // one live use per 16 B line segment
+  }
+  if (acc == 1e30f) sink[blockIdx.x] = acc;  // never true; keeps loads
+}
+
+extern "C" hipError_t launch_prefetch(const void* p, long n_elems,
+                                      void* sink, hipStream_t stream) {
+  hipLaunchKernelGGL(k_prefetch, dim3(128), dim3(256), 0, stream,
+                     (const u16*)p, n_elems, (float*)sink);
+  return hipGetLastError();
+}
+
+// ====================================================================
 // elementwise tanh soft-cap: y = cap * tanh(x / cap)  (bf16)
 // (Gemma-2 final-logit capping on the all-positions GEMM path; the
 // decode GEMV fuses this into its epilogue instead)

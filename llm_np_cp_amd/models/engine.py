@@ -81,6 +81,14 @@ class GPUModel:
         self._alloc_state(prefill_chunk)
         self._graph = None
         self._graph_mode = None
+        # side-stream weight prefetcher (warms Infinity Cache with layer
+        # i+1's weights while layer i computes); off via LLM_PREFETCH=0
+        import os as _os
+        self.prefetch_on = _os.environ.get("LLM_PREFETCH", "1") != "0"
+        self._pf_stream = torch.cuda.Stream(device=self.device)
+        self._pf_sink = torch.zeros(256, dtype=torch.float32,
+                                    device=self.device)
+        self._pf_events = []
 
     # ------------------------------------------------------------------
     def _upload_weights(self, w: Dict[str, np.ndarray]):
@@ -381,19 +389,36 @@ class GPUModel:
         else:
             ho.gemv(lw[name], x, y, **kw)
 
+    def _pf_tensors(self, lw):
+        if self.fp8:
+            return (lw["wqkv_q"], lw["wo_q"], lw["wgu_q"], lw["wdown_q"])
+        return (lw["wqkv"], lw["wo"], lw["wgu"], lw["wdown"])
+
     def _decode_step(self, greedy: bool, min_p: float):
-        """Fused decode path: 4 kernels/layer (llama) or 6 (gemma) —
+        """Fused decode path: 4 kernels/layer (llama) or 5 (gemma) —
         RMSNorm and GLU live inside the GEMV staging pass, RoPE + KV
-        write inside the attention kernel."""
+        write inside the attention kernel; a side stream prefetches the
+        next layer's weights through MALL, paced by per-layer events."""
         cfg = self.config
         eps = cfg.rms_norm_eps
         h = self.b_h[0]
         hnext = self.b_hb
         t1 = self.b_t1[0]
         t2 = self.b_t2
+        s0 = torch.cuda.current_stream()
+        pf = self.prefetch_on
+        self._pf_events = []
         ho.embed(self.embed, self.next_token, self.b_h, 1, cfg.embed_scale)
         prev = None  # previous gemma layer (its postffn norm fused here)
         for i, lw in enumerate(self.layers):
+            if pf and i + 1 < len(self.layers):
+                ev = torch.cuda.Event()
+                ev.record(s0)
+                self._pf_events.append(ev)
+                self._pf_stream.wait_event(ev)
+                with torch.cuda.stream(self._pf_stream):
+                    for t in self._pf_tensors(self.layers[i + 1]):
+                        ho.prefetch(t, self._pf_sink)
             window = cfg.sliding_window if cfg.is_sliding(i) else 0
             if self.gemma and prev is not None:
                 # h' = h + norm(t2_prev)*g_postffn_prev; stage norm(h')*g_in
@@ -444,6 +469,11 @@ class GPUModel:
                     self._dgemv(lw, "wdown", self.b_gu[:self.inter_l], h,
                                 res=h, stage=ho.STAGE_GLU,
                                 x2=self.b_gu[self.inter_l:], act=self.act)
+        if pf:
+            ev = torch.cuda.Event()
+            ev.record(self._pf_stream)
+            self._pf_events.append(ev)
+            s0.wait_event(ev)  # join the fork before the step ends
         if self.gemma:
             # last layer's post-ffn sandwich norm (no following projection
             # to fold it into)

@@ -51,6 +51,8 @@ _SIGS = {
                      [ctypes.c_int, ctypes.c_void_p],
     "launch_gemm_bf16": [ctypes.c_void_p] * 5 + [ctypes.c_int] * 3 +
                         [ctypes.c_void_p],
+    "launch_prefetch": [ctypes.c_void_p, ctypes.c_long, ctypes.c_void_p,
+                        ctypes.c_void_p],
     "launch_softcap": [ctypes.c_void_p, ctypes.c_long, ctypes.c_float,
                        ctypes.c_void_p],
     "launch_addinto": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_long,
@@ -220,6 +222,13 @@ def gemm(X: torch.Tensor, W: torch.Tensor, Y: torch.Tensor,
         accbuf = None
     _check(lib().launch_gemm_bf16(_ptr(X), _ptr(W), _ptr(Y), _ptr(res),
                                   _ptr(accbuf), M, N, K, _stream()), "gemm")
+
+
+def prefetch(t: torch.Tensor, sink: torch.Tensor):
+    """Stream t through the cache hierarchy (side-stream warmer)."""
+    n = t.numel() * t.element_size() // 2  # treat as u16 elements
+    _check(lib().launch_prefetch(_ptr(t), n, _ptr(sink), _stream()),
+           "prefetch")
 
 
 def softcap(y: torch.Tensor, cap: float):
