@@ -81,10 +81,12 @@ class GPUModel:
         self._alloc_state(prefill_chunk)
         self._graph = None
         self._graph_mode = None
-        # side-stream weight prefetcher (warms Infinity Cache with layer
-        # i+1's weights while layer i computes); off via LLM_PREFETCH=0
+        # side-stream weight prefetcher — measured NEGATIVE (the fork's
+        # join makes the sequential prefetch chain the critical path:
+        # fp8 1391 -> 764 tok/s), so off by default; LLM_PREFETCH=1 to
+        # experiment
         import os as _os
-        self.prefetch_on = _os.environ.get("LLM_PREFETCH", "1") != "0"
+        self.prefetch_on = _os.environ.get("LLM_PREFETCH", "0") == "1"
         self._pf_stream = torch.cuda.Stream(device=self.device)
         self._pf_sink = torch.zeros(256, dtype=torch.float32,
                                     device=self.device)
