@@ -74,10 +74,18 @@ DEVINL float wave_reduce_sum(float v) {
 #define STAGE_NORM2 3
 
 // staging helper shared by bf16/fp8 GEMV (RMSNorm / GLU fused pre-ops)
+DEVINL float stage_red_sum(float* red) {
+  const int nw = blockDim.x >> 6;
+  float t = 0.f;
+  for (int w = 0; w < nw; w++) t += red[w];
+  return t;
+}
+
 DEVINL const u16* gemv_stage(char* smem, const u16* x, const u16* x2,
                              const float* g, const float* g2, u16* hout,
                              int K, int stage, int act, float eps) {
   u16* xs = (u16*)smem;
+  const int STRIDE = blockDim.x * 8;
   if (stage == STAGE_NORM2) {
     // x2 = h_in (read-only here); block 0 persists h' into hout — a
     // DIFFERENT buffer (ping-pong), since other blocks still read h_in
@@ -85,7 +93,7 @@ DEVINL const u16* gemv_stage(char* smem, const u16* x, const u16* x2,
     float* red = (float*)(smem + (size_t)K * 2);
     // pass 1: sumsq of x (the un-normed projection output t)
     float ss = 0.f;
-    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
       s8v v = *(const s8v*)(x + i);
 #pragma unroll
       for (int j = 0; j < 8; j++) {
@@ -96,13 +104,12 @@ DEVINL const u16* gemv_stage(char* smem, const u16* x, const u16* x2,
     ss = wave_reduce_sum(ss);
     if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
     __syncthreads();
-    float rnorm_a = rsqrtf((red[0] + red[1] + red[2] + red[3]) / (float)K +
-                           eps);
+    float rnorm_a = rsqrtf(stage_red_sum(red) / (float)K + eps);
     __syncthreads();
     // pass 2: h' = h + norm_a(x)*g (bf16-rounded, matching the
     // standalone k_rmsnorm mode 1), stash h' in LDS, sumsq of h'
     float ss2 = 0.f;
-    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
       s8v v = *(const s8v*)(x + i);
       s8v hv = *(const s8v*)(h + i);
       f4v ga = *(const f4v*)(g + i);
@@ -122,10 +129,9 @@ DEVINL const u16* gemv_stage(char* smem, const u16* x, const u16* x2,
     ss2 = wave_reduce_sum(ss2);
     if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss2;
     __syncthreads();
-    float rnorm_b = rsqrtf((red[0] + red[1] + red[2] + red[3]) / (float)K +
-                           eps);
+    float rnorm_b = rsqrtf(stage_red_sum(red) / (float)K + eps);
     // pass 3: xs = norm_b(h')*g2
-    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
       s8v v = *(s8v*)(xs + i);
       f4v ga = *(const f4v*)(g2 + i);
       f4v gb = *(const f4v*)(g2 + i + 4);
@@ -141,7 +147,7 @@ DEVINL const u16* gemv_stage(char* smem, const u16* x, const u16* x2,
   }
   if (stage == STAGE_NORM) {
     float ss = 0.f;
-    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
       s8v v = *(const s8v*)(x + i);
       *(s8v*)(xs + i) = v;
 #pragma unroll
@@ -154,8 +160,8 @@ DEVINL const u16* gemv_stage(char* smem, const u16* x, const u16* x2,
     ss = wave_reduce_sum(ss);
     if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
     __syncthreads();
-    float rnorm = rsqrtf((red[0] + red[1] + red[2] + red[3]) / (float)K + eps);
-    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+    float rnorm = rsqrtf(stage_red_sum(red) / (float)K + eps);
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
       s8v v = *(s8v*)(xs + i);
       u16 o[8];
 #pragma unroll
@@ -164,7 +170,7 @@ DEVINL const u16* gemv_stage(char* smem, const u16* x, const u16* x2,
       *(s8v*)(xs + i) = *(s8v*)o;
     }
   } else if (stage == STAGE_GLU) {
-    for (int i = threadIdx.x * 8; i < K; i += 2048) {
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
       s8v gv = *(const s8v*)(x + i);
       s8v uv = *(const s8v*)(x2 + i);
       u16 o[8];
@@ -199,7 +205,7 @@ DEVINL void gemv_epilogue(float acc, int row, void* y, const u16* res,
 // MI355X_MICROARCH "nt-weights").  RPW: rows per wave (ILP: RPW x 4
 // weight loads in flight).
 template <bool NT, int RPW>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(512)
 k_gemv_bf16_t(const u16* __restrict__ W, const u16* __restrict__ x,
               const u16* __restrict__ x2, const float* __restrict__ g,
               const float* __restrict__ g2, void* __restrict__ y,
@@ -211,8 +217,9 @@ k_gemv_bf16_t(const u16* __restrict__ W, const u16* __restrict__ x,
   const u16* eres = (stage == STAGE_NORM2) ? nullptr : res;  // res = hout
 
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int rstride = gridDim.x * 4 * RPW;  // grid-stride rows: amortize
-  for (int row0 = (blockIdx.x * 4 + wave) * RPW; row0 < N;
+  const int wpb = blockDim.x >> 6;
+  const int rstride = gridDim.x * wpb * RPW;  // grid-stride rows
+  for (int row0 = (blockIdx.x * wpb + wave) * RPW; row0 < N;
        row0 += rstride) {                    // wave setup over many rows
   const u16* Wr[RPW];
   float a0[RPW], a1[RPW], a2[RPW], a3[RPW];
@@ -279,12 +286,14 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
                                        int out_f32, float softcap,
                                        int nt, int rpw, int maxblocks,
                                        hipStream_t stream) {
-  size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 16);
-  int blocks = (N + 4 * rpw - 1) / (4 * rpw);
+  size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 32);
+  int threads = (N >= 8192) ? 512 : 256;
+  int wpb = threads / 64;
+  int blocks = (N + wpb * rpw - 1) / (wpb * rpw);
   int cap = maxblocks > 0 ? maxblocks : 1024;
   if (blocks > cap) blocks = cap;  // grid-stride the rest (Guideline 11)
 #define GEMV_CASE(NTV, RPWV)                                                 \
-  hipLaunchKernelGGL((k_gemv_bf16_t<NTV, RPWV>), dim3(blocks), dim3(256),    \
+  hipLaunchKernelGGL((k_gemv_bf16_t<NTV, RPWV>), dim3(blocks), dim3(threads),\
                      lds, stream, (const u16*)W, (const u16*)x,              \
                      (const u16*)x2, (const float*)g, (const float*)g2, y,   \
                      (const u16*)res, N, K, stage, act, eps, out_f32,        \
@@ -319,7 +328,7 @@ DEVINL void fp8x16_to_f32(u4v w, float* o) {
 }
 
 template <bool NT, int RPW>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(512)
 k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
              const u16* __restrict__ x, const u16* __restrict__ x2,
              const float* __restrict__ g, const float* __restrict__ g2,
@@ -331,8 +340,9 @@ k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
   const u16* eres = (stage == STAGE_NORM2) ? nullptr : res;
 
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int rstride = gridDim.x * 4 * RPW;
-  for (int row0 = (blockIdx.x * 4 + wave) * RPW; row0 < N;
+  const int wpb = blockDim.x >> 6;
+  const int rstride = gridDim.x * wpb * RPW;
+  for (int row0 = (blockIdx.x * wpb + wave) * RPW; row0 < N;
        row0 += rstride) {
   const uint8_t* Wr[RPW];
   float a0[RPW], a1[RPW];
@@ -397,12 +407,14 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
                                       int out_f32, float softcap, int nt,
                                       int rpw, int maxblocks,
                                       hipStream_t stream) {
-  size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 16);
-  int blocks = (N + 4 * rpw - 1) / (4 * rpw);
+  size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 32);
+  int threads = (N >= 8192) ? 512 : 256;
+  int wpb = threads / 64;
+  int blocks = (N + wpb * rpw - 1) / (wpb * rpw);
   int cap = maxblocks > 0 ? maxblocks : 1024;
   if (blocks > cap) blocks = cap;
 #define GEMV8_CASE(NTV, RPWV)                                               \
-  hipLaunchKernelGGL((k_gemv_fp8_t<NTV, RPWV>), dim3(blocks), dim3(256),    \
+  hipLaunchKernelGGL((k_gemv_fp8_t<NTV, RPWV>), dim3(blocks), dim3(threads),\
                      lds, stream, (const uint8_t*)W, (const float*)scales,  \
                      (const u16*)x, (const u16*)x2, (const float*)g,        \
                      (const float*)g2, y, (const u16*)res, N, K, stage,     \
