@@ -249,6 +249,7 @@ class GPUModel:
         self.rng_ctr.zero_()
         self.s_gmax.zero_()
         self.s_pick.zero_()
+        self._host_len = 0  # host mirror of len_buf (overflow guard)
 
     # ------------------------------------------------------------------
     # layer stack over rows [0, M) of the scratch buffers
@@ -365,6 +366,9 @@ class GPUModel:
         if pos0 != cache.seq_len:
             raise ValueError("GPU engine requires sequential positions")
         n = len(ids)
+        if pos0 + n > self.max_seq:
+            raise ValueError(f"sequence {pos0}+{n} exceeds max_seq "
+                             f"{self.max_seq}")
         done = 0
         while done < n:
             M = min(n - done, self.PC)
@@ -376,6 +380,7 @@ class GPUModel:
             self._layers_forward(M)
             done += M
         ho.i32_set(self.len_buf, pos0 + n)
+        self._host_len = pos0 + n
         self._lm_head_last(M)
         cache.seq_len = pos0 + n
         torch.cuda.synchronize()
@@ -527,6 +532,11 @@ class GPUModel:
                use_graph: bool = True, first_from_logits: bool = True):
         """Generate n_tokens ids device-side; returns int32 numpy ids.
         Assumes prefill() ran (len_buf == prompt length, logits ready)."""
+        if getattr(self, "_host_len", 0) + n_tokens > self.max_seq:
+            raise ValueError(
+                f"decode would overflow the KV pool: len {self._host_len} "
+                f"+ {n_tokens} > max_seq {self.max_seq}")
+        self._host_len += n_tokens
         if first_from_logits:
             # sample token 0 from the prefill logits
             ho.sample(self.b_logits, min_p, greedy, self.seed, self.rng_ctr,
