@@ -945,6 +945,155 @@ extern "C" hipError_t launch_attn_dec(const void* qkv, void* kc, void* vc,
   return hipGetLastError();
 }
 
+// ====================================================================
+// MFMA flash prefill attention: one wave per (head, 16-query tile);
+// iterates 16-position KV tiles with online softmax; never materializes
+// QK^T.  Swapped-operand form (S^T = mfma(K, Q)) keeps each query's
+// max/sum lane-local (guide App.B "swapped QK^T"); PV uses
+// v_mfma_f32_16x16x16bf16_1k with O^T accumulators so the per-query
+// rescale is also lane-local.  Replaces the per-query VALU scan for
+// M > 1 (prefill): ~MFMA-rate QK^T/PV instead of VALU dots.
+// ====================================================================
+
+typedef short b4v __attribute__((ext_vector_type(4)));
+
+template <int HD>
+__global__ void __launch_bounds__(64)
+k_attn_prefill_mfma(const u16* __restrict__ q, const u16* __restrict__ kc,
+                    const u16* __restrict__ vc, u16* __restrict__ out,
+                    const int* __restrict__ len_ptr, int M, int nh, int kvh,
+                    int S, float scale, float softcap, int window) {
+  constexpr int HD32 = HD / 32;
+  constexpr int HD16 = HD / 16;
+  const int h = blockIdx.x;
+  const int qt = blockIdx.y;           // 16-query tile
+  const int kvhead = h / (nh / kvh);
+  const int pos0 = *len_ptr;
+  const int lane = threadIdx.x & 63;
+  const int qcol = lane & 15;          // this lane's query (column)
+  const int krow4 = (lane >> 4) * 4;   // kv rows (QK) / k-index (PV)
+
+  const int m0 = qt * 16;
+  const int my_m = m0 + qcol;          // may be >= M (masked at store)
+  const int qpos = pos0 + my_m;        // absolute position of my query
+
+  // Q fragments (B operand): lane: col = qcol, k-chunk = (lane>>4)*8
+  s8v qf[HD32];
+  {
+    int mr = my_m < M ? my_m : M - 1;
+    const u16* qp = q + ((size_t)mr * nh + h) * HD;
+#pragma unroll
+    for (int c = 0; c < HD32; c++)
+      qf[c] = *(const s8v*)(qp + c * 32 + (lane >> 4) * 8);
+  }
+
+  const u16* K0 = kc + (size_t)kvhead * S * HD;
+  const u16* V0 = vc + (size_t)kvhead * S * HD;
+
+  float mrun = -INFINITY, lrun = 0.f;
+  f4v acc_o[HD16];
+#pragma unroll
+  for (int d = 0; d < HD16; d++) acc_o[d] = {0.f, 0.f, 0.f, 0.f};
+
+  const int T_end = min(pos0 + M, pos0 + m0 + 16);  // causal upper bound
+  int t_start = 0;
+  if (window > 0) {
+    t_start = pos0 + m0 + 1 - window;  // earliest key any tile query sees
+    if (t_start < 0) t_start = 0;
+  }
+
+  for (int t0 = t_start; t0 < T_end; t0 += 16) {
+    // S^T tile: A = K rows (row = lane&15, k-chunk = (lane>>4)*8)
+    f4v st = {0.f, 0.f, 0.f, 0.f};
+    {
+      int tk = t0 + (lane & 15);
+      int tkl = tk < T_end ? tk : T_end - 1;
+      const u16* kp = K0 + (size_t)tkl * HD;
+#pragma unroll
+      for (int c = 0; c < HD32; c++) {
+        s8v kf = *(const s8v*)(kp + c * 32 + (lane >> 4) * 8);
+        st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf, qf[c], st, 0, 0, 0);
+      }
+    }
+    // scale + softcap + causal/window mask; element (kv = t0+krow4+r, qcol)
+    float sv[4];
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      float s = st[r] * scale;
+      if (softcap > 0.f) s = softcap * tanhf(s / softcap);
+      int t = t0 + krow4 + r;
+      bool bad = (t > qpos) || (t >= T_end) ||
+                 (window > 0 && t <= qpos - window);
+      sv[r] = bad ? -INFINITY : s;
+    }
+    // column stats across the 4 row-groups (lanes qcol, qcol+16, ...)
+    float pmax = fmaxf(fmaxf(sv[0], sv[1]), fmaxf(sv[2], sv[3]));
+    pmax = fmaxf(pmax, __shfl_xor(pmax, 16));
+    pmax = fmaxf(pmax, __shfl_xor(pmax, 32));
+    float mnew = fmaxf(mrun, pmax);
+    float alpha = (mnew == -INFINITY) ? 0.f : __expf(mrun - mnew);
+    float psum = 0.f;
+    u16 pb[4];
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      float p = (mnew == -INFINITY) ? 0.f : __expf(sv[r] - mnew);
+      psum += p;
+      pb[r] = f2b(p);
+    }
+    psum += __shfl_xor(psum, 16);
+    psum += __shfl_xor(psum, 32);
+    lrun = lrun * alpha + psum;
+    if (mnew != -INFINITY) mrun = mnew;
+
+    // PV: O^T[d][q] += V^T[d][kv] @ P^T[kv][q]
+    // A = V^T frag: row d = db*16 + (lane&15), k = krow4..+4 (strided V)
+    b4v pfrag = *(b4v*)pb;
+#pragma unroll
+    for (int db = 0; db < HD16; db++) {
+      int dg = db * 16 + (lane & 15);
+      u16 vt[4];
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        int t = t0 + krow4 + r;
+        int tl = t < T_end ? t : T_end - 1;
+        vt[r] = V0[(size_t)tl * HD + dg];  // masked kv rows have p == 0
+      }
+#pragma unroll
+      for (int r = 0; r < 4; r++) acc_o[db][r] *= alpha;
+      acc_o[db] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
+          *(b4v*)vt, pfrag, acc_o[db], 0, 0, 0);
+    }
+  }
+
+  if (my_m < M) {
+    float inv = 1.f / lrun;
+    u16* op = out + ((size_t)my_m * nh + h) * HD;
+#pragma unroll
+    for (int db = 0; db < HD16; db++)
+#pragma unroll
+      for (int r = 0; r < 4; r++)
+        op[db * 16 + krow4 + r] = f2b(acc_o[db][r] * inv);
+  }
+}
+
+extern "C" hipError_t launch_attn_prefill_mfma(
+    const void* q, const void* kc, const void* vc, void* out,
+    const void* len_ptr, int M, int nh, int kvh, int hd, int S, float scale,
+    float softcap, int window, hipStream_t stream) {
+  dim3 grid(nh, (M + 15) / 16);
+#define APF_CASE(HDV)                                                       \
+  hipLaunchKernelGGL((k_attn_prefill_mfma<HDV>), grid, dim3(64), 0, stream, \
+                     (const u16*)q, (const u16*)kc, (const u16*)vc,         \
+                     (u16*)out, (const int*)len_ptr, M, nh, kvh, S, scale,  \
+                     softcap, window)
+  if (hd == 64) APF_CASE(64);
+  else if (hd == 128) APF_CASE(128);
+  else if (hd == 256) APF_CASE(256);
+  else return hipErrorInvalidValue;
+#undef APF_CASE
+  return hipGetLastError();
+}
+
 extern "C" hipError_t launch_attn(const void* q, const void* kc,
                                   const void* vc, void* out,
                                   const void* len_ptr, int M, int nh, int kvh,

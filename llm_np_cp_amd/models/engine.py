@@ -274,9 +274,17 @@ class GPUModel:
             ho.rope_cache(self.b_q, self.b_k, self.b_v, self.k_cache[i],
                           self.v_cache[i], self.cos_t, self.sin_t,
                           self.len_buf, M, self.nh_l, self.kvh_l, self.hd)
-            ho.attn(self.b_q, self.k_cache[i], self.v_cache[i], self.b_att,
+            if self.hd in (64, 128, 256):
+                ho.attn_prefill_mfma(
+                    self.b_q, self.k_cache[i], self.v_cache[i], self.b_att,
                     self.len_buf, M, self.nh_l, self.kvh_l, self.hd,
-                    self.scale, softcap=self.attn_softcap, window=window or 0)
+                    self.scale, softcap=self.attn_softcap,
+                    window=window or 0)
+            else:
+                ho.attn(self.b_q, self.k_cache[i], self.v_cache[i],
+                        self.b_att, self.len_buf, M, self.nh_l, self.kvh_l,
+                        self.hd, self.scale, softcap=self.attn_softcap,
+                        window=window or 0)
             if self.gemma:
                 self._linear(lw["wo"], self.b_att, t1, M=M)
                 tpu.all_reduce(t1[:M])

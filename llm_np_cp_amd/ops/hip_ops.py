@@ -39,6 +39,9 @@ _SIGS = {
     "launch_attn": [ctypes.c_void_p] * 5 + [ctypes.c_int] * 5 +
                    [ctypes.c_float, ctypes.c_float, ctypes.c_int,
                     ctypes.c_void_p],
+    "launch_attn_prefill_mfma": [ctypes.c_void_p] * 5 + [ctypes.c_int] * 5 +
+                                [ctypes.c_float, ctypes.c_float,
+                                 ctypes.c_int, ctypes.c_void_p],
     "launch_attn_dec": [ctypes.c_void_p] * 9 + [ctypes.c_int] * 5 +
                        [ctypes.c_float, ctypes.c_float, ctypes.c_int,
                         ctypes.c_void_p],
@@ -160,6 +163,20 @@ def attn(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
         _ptr(q), _ptr(k_cache), _ptr(v_cache), _ptr(out), _ptr(len_ptr),
         M, nh, kvh, hd, S, ctypes.c_float(scale), ctypes.c_float(softcap),
         window, _stream()), "attn")
+
+
+def attn_prefill_mfma(q: torch.Tensor, k_cache: torch.Tensor,
+                      v_cache: torch.Tensor, out: torch.Tensor,
+                      len_ptr: torch.Tensor, M: int, nh: int, kvh: int,
+                      hd: int, scale: float, softcap: float = 0.0,
+                      window: int = 0):
+    """Flash prefill attention (MFMA, online softmax); q already roped,
+    caches already written for [0, pos0+M)."""
+    S = k_cache.shape[1]
+    _check(lib().launch_attn_prefill_mfma(
+        _ptr(q), _ptr(k_cache), _ptr(v_cache), _ptr(out), _ptr(len_ptr),
+        M, nh, kvh, hd, S, ctypes.c_float(scale), ctypes.c_float(softcap),
+        window, _stream()), "attn_prefill_mfma")
 
 
 def attn_dec(qkv: torch.Tensor, k_cache: torch.Tensor,

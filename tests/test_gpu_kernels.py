@@ -429,3 +429,37 @@ def test_gemv_fused_norm2_stage():
         torch.bfloat16).float()
     assert_close(y, W.float() @ xn)
     assert_close(hout, hp)
+
+
+@pytest.mark.parametrize("hd,window,cap", [(64, 0, 0.0), (128, 0, 0.0),
+                                           (256, 32, 50.0)])
+def test_attn_prefill_mfma_vs_reference(hd, window, cap):
+    """MFMA flash prefill vs fp32 torch reference (and the VALU kernel)."""
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    nh, kvh, S, M, pos0 = 4, 2, 256, 37, 21
+    q = randn_bf16(M, nh * hd, seed=130)
+    kc = randn_bf16(kvh, S, hd, seed=131)
+    vc = randn_bf16(kvh, S, hd, seed=132)
+    pos = torch.tensor([pos0], dtype=torch.int32, device=dev())
+    scale = hd ** -0.5
+
+    out = torch.empty(M, nh * hd, dtype=torch.bfloat16, device=dev())
+    ho.attn_prefill_mfma(q, kc, vc, out, pos, M, nh, kvh, hd, scale,
+                         softcap=cap, window=window)
+    torch.cuda.synchronize()
+
+    Kf, Vf = kc.float(), vc.float()
+    qf = q.float().view(M, nh, hd)
+    ref = torch.empty(M, nh, hd)
+    for m in range(M):
+        qpos = pos0 + m
+        lo = max(0, qpos + 1 - window) if window else 0
+        for h in range(nh):
+            kv = h // (nh // kvh)
+            sc = (Kf[kv, lo:qpos + 1] @ qf[m, h]) * scale
+            if cap:
+                sc = cap * torch.tanh(sc / cap)
+            p = torch.softmax(sc, -1)
+            ref[m, h] = p @ Vf[kv, lo:qpos + 1]
+    assert_close(out.view(M, nh, hd), ref, rtol=4e-2, atol=4e-2)
