@@ -226,6 +226,13 @@ PRESETS = {
         attn_logit_softcapping=50.0, final_logit_softcapping=30.0,
     ),
     # tiny configs for tests
+    "tiny-llama-hd64": dict(    # exercises the hd=64 MFMA prefill path
+        model_type="llama", vocab_size=512, hidden_size=128,
+        intermediate_size=256, num_hidden_layers=2, num_attention_heads=2,
+        num_key_value_heads=1, head_dim=64, rms_norm_eps=1e-5,
+        rope_theta=10000.0, max_position_embeddings=512, hidden_act="silu",
+        tie_word_embeddings=True,
+    ),
     "tiny-llama": dict(
         model_type="llama", vocab_size=512, hidden_size=64,
         intermediate_size=128, num_hidden_layers=2, num_attention_heads=4,

@@ -230,3 +230,37 @@ def test_forward_full_gemma_softcap():
     got = gpu.forward_full(ids)
     np.testing.assert_allclose(got[-1], want[-1], rtol=0.2, atol=0.2)
     assert np.abs(got).max() <= cfg.final_logit_softcapping + 1e-3
+
+
+def test_hd64_engine_matches_oracle():
+    """hd=64 tiny config: the engine's MFMA flash-prefill + decode path
+    (the one real presets use) vs the NumPy oracle."""
+    import llm_np_cp_amd as L
+
+    cfg, gpu, ref = make_pair("tiny-llama-hd64", seed=40)
+    rng = np.random.default_rng(41)
+    ids = rng.integers(0, cfg.vocab_size, size=23)
+
+    from llm_np_cp_amd.models.numpy_ref import NumpyKVCache
+    ref_logits = ref.forward(ids, NumpyKVCache(cfg, 64), 0)
+    got = gpu.forward(ids, gpu.make_cache(64), 0)[0]
+    assert np.argmax(got) == np.argmax(ref_logits[-1])
+
+    # greedy rollout equality
+    tok = L.ByteTokenizer()
+    p = L.SamplingParams(strategy="greedy")
+    a = L.generate("Once upon", tok, ref, max_tokens=10, stream=False,
+                   params=p, stop_on_eos=False)
+    b = L.generate("Once upon", tok, gpu, max_tokens=10, stream=False,
+                   params=p, stop_on_eos=False)
+    assert a.token_ids == b.token_ids
+
+
+def test_decode_overflow_guard():
+    import llm_np_cp_amd as L
+    import pytest as _pytest
+
+    cfg, gpu, ref = make_pair("tiny-llama", seed=50, max_seq=64)
+    gpu.prefill(np.arange(1, 33))
+    with _pytest.raises(ValueError, match="overflow"):
+        gpu.decode(100, greedy=True, use_graph=False)
