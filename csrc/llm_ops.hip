@@ -1323,63 +1323,89 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 #define BM 128
 #define BN 128
-#define BK 32
-#define LDS_STRIDE (BK + 8)  // pad: 40 elems = 80 B, 16B-aligned rows
+#define BK 64
 
+// BM=BN=128, BK=64, 4 waves (2x2), each wave a 64x64 sub-tile of
+// 4x4 16x16 fragments.  Double-buffered LDS filled by
+// global_load_lds (async global->LDS DMA, 16 B/lane), one barrier per
+// K-tile (guide §5 "Minimum 2-phase": STAGE next ahead of ds_read+MFMA).
+// LDS rows are XOR-swizzled st_8x16 (slot ^= row&7) to keep
+// ds_read_b128 fragment reads conflict-free; glds writes lane-linear,
+// so the swizzle is applied to the per-lane GLOBAL source address
+// (guide §5.4 rule 21) and to the read offsets — never the LDS dest.
 extern "C" __global__ void __launch_bounds__(256)
 k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
             u16* __restrict__ Y, const u16* __restrict__ res,
             float* __restrict__ accbuf, int M, int N, int K) {
-  __shared__ u16 As[BM * LDS_STRIDE];
-  __shared__ u16 Bs[BN * LDS_STRIDE];
+  __shared__ u16 As[2][128 * 64];
+  __shared__ u16 Bs[2][128 * 64];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int wrow = wave >> 1, wcol = wave & 1;  // 2x2 waves
   const int bm = blockIdx.x * BM, bn = blockIdx.y * BN;
-  // split-K: gridDim.z blocks share (bm,bn), each covers a K slice and
-  // atomically accumulates fp32 partials; k_gemm_fin converts to bf16.
+
   const int SK = gridDim.z;
   const int kslices = (K / BK + SK - 1) / SK;
   const int k_lo = blockIdx.z * kslices * BK;
   int k_hi = k_lo + kslices * BK;
   if (k_hi > K) k_hi = K;
+  const int nt = (k_hi - k_lo) / BK;
+  if (nt <= 0) return;
 
-  f32x4 acc[4][4];
+  f4v acc[4][4];
 #pragma unroll
   for (int i = 0; i < 4; i++)
 #pragma unroll
     for (int j = 0; j < 4; j++) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // stage thread mapping: 256 threads x 16 B; tile row r = t/4, col (t%4)*8
-  const int sr = tid >> 2, sc = (tid & 3) * 8;
-
-  for (int kt = k_lo; kt < k_hi; kt += BK) {
-    // A tile: rows clamped to M-1 (stores are masked later)
+  // glds staging: per call a wave fills 8 rows x 128 B (lane: row
+  // lane/8, slot lane%8); source col-slot pre-swizzled by row&7.
+  const int g_r = lane >> 3;             // row within the 8-row group
+  const int g_s = lane & 7;              // LDS slot
+  auto stage = [&](int buf, int kt) {
+    const int k0 = k_lo + kt * BK;
 #pragma unroll
-    for (int p = 0; p < 2; p++) {
-      int r = sr + p * 64;
+    for (int i = 0; i < 4; i++) {
+      int r = wave * 32 + i * 8 + g_r;   // tile row 0..127
+      int cs = g_s ^ (r & 7);            // source col-slot (involution)
       int gr = bm + r;
       int grc = gr < M ? gr : (M > 0 ? M - 1 : 0);
-      *(s8v*)(As + r * LDS_STRIDE + sc) =
-          *(const s8v*)(X + (size_t)grc * K + kt + sc);
-      int rb = sr + p * 64;
-      int gb = bn + rb;
-      int gbc = gb < N ? gb : N - 1;  // stores masked on col >= N
-      *(s8v*)(Bs + rb * LDS_STRIDE + sc) =
-          *(const s8v*)(W + (size_t)gbc * K + kt + sc);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(
+              X + (size_t)grc * K + k0 + cs * 8),
+          (__attribute__((address_space(3))) uint32_t*)(
+              &As[buf][(wave * 32 + i * 8) * 64]),
+          16, 0, 0);
+      int gb = bn + r;
+      int gbc = gb < N ? gb : N - 1;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(
+              W + (size_t)gbc * K + k0 + cs * 8),
+          (__attribute__((address_space(3))) uint32_t*)(
+              &Bs[buf][(wave * 32 + i * 8) * 64]),
+          16, 0, 0);
     }
-    __syncthreads();
+  };
 
-    // fragments: lane l -> row (l&15), k-chunk (l>>4)*8 of each 16x32 tile
-    const int fr = lane & 15, fk = (lane >> 4) * 8;
+  stage(0, 0);
+  __syncthreads();
+
+  const int fr = lane & 15, fk = lane >> 4;  // fragment row / k-slot
+  int cur = 0;
+  for (int t = 0; t < nt; t++) {
+    if (t + 1 < nt) stage(cur ^ 1, t + 1);   // issue next tile's DMA
 #pragma unroll
-    for (int ks = 0; ks < BK; ks += 32) {
+    for (int sl = 0; sl < 2; sl++) {         // two 32-deep k-slices
       bf16x8 a[4], b[4];
 #pragma unroll
       for (int i = 0; i < 4; i++) {
-        a[i] = *(bf16x8*)(As + (wrow * 64 + i * 16 + fr) * LDS_STRIDE + ks + fk);
-        b[i] = *(bf16x8*)(Bs + (wcol * 64 + i * 16 + fr) * LDS_STRIDE + ks + fk);
+        int ra = wrow * 64 + i * 16 + fr;
+        int sa = (sl * 4 + fk) ^ (ra & 7);
+        a[i] = *(bf16x8*)(&As[cur][ra * 64 + sa * 8]);
+        int rb = wcol * 64 + i * 16 + fr;
+        int sb = (sl * 4 + fk) ^ (rb & 7);
+        b[i] = *(bf16x8*)(&Bs[cur][rb * 64 + sb * 8]);
       }
 #pragma unroll
       for (int i = 0; i < 4; i++)
@@ -1388,7 +1414,8 @@ k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a[i], b[j], acc[i][j], 0, 0, 0);
     }
-    __syncthreads();
+    __syncthreads();  // drains the in-flight DMA (vmcnt0) + read fence
+    cur ^= 1;
   }
 
   // epilogue: C layout col = lane&15, row = (lane>>4)*4 + reg

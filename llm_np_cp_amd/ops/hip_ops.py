@@ -234,7 +234,7 @@ def gemm(X: torch.Tensor, W: torch.Tensor, Y: torch.Tensor,
     grid.z for chip fill."""
     M, K = X.shape
     N = W.shape[0]
-    assert W.shape[1] == K and K % 32 == 0
+    assert W.shape[1] == K and K % 64 == 0
     if accbuf is not None and accbuf.numel() < M * N:
         accbuf = None
     _check(lib().launch_gemm_bf16(_ptr(X), _ptr(W), _ptr(Y), _ptr(res),
