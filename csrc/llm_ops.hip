@@ -995,6 +995,7 @@ k_attn_prefill_mfma(const u16* __restrict__ q, const u16* __restrict__ kc,
 #pragma unroll
   for (int d = 0; d < HD16; d++) acc_o[d] = {0.f, 0.f, 0.f, 0.f};
 
+  __shared__ u16 vlds[16 * HD];
   const int T_end = min(pos0 + M, pos0 + m0 + 16);  // causal upper bound
   int t_start = 0;
   if (window > 0) {
@@ -1045,19 +1046,28 @@ k_attn_prefill_mfma(const u16* __restrict__ q, const u16* __restrict__ kc,
     lrun = lrun * alpha + psum;
     if (mnew != -INFINITY) mrun = mnew;
 
-    // PV: O^T[d][q] += V^T[d][kv] @ P^T[kv][q]
-    // A = V^T frag: row d = db*16 + (lane&15), k = krow4..+4 (strided V)
+    // PV: O^T[d][q] += V^T[d][kv] @ P^T[kv][q].
+    // Stage the 16-row V tile in LDS with vector loads (single-wave
+    // block: no barrier needed, lgkmcnt orders write->read), then
+    // gather the strided V^T fragments from LDS instead of 2-byte
+    // global loads (those were ~half the kernel's time).
+    {
+      const int vrow = lane >> 2;          // 16 rows, 4 lanes each
+      int tv = t0 + vrow;
+      int tvl = tv < T_end ? tv : T_end - 1;
+#pragma unroll
+      for (int c = 0; c < HD / 32; c++)    // lane covers 8 cols per c
+        *(s8v*)(&vlds[vrow * HD + c * 32 + (lane & 3) * 8]) =
+            *(const s8v*)(V0 + (size_t)tvl * HD + c * 32 + (lane & 3) * 8);
+    }
     b4v pfrag = *(b4v*)pb;
 #pragma unroll
     for (int db = 0; db < HD16; db++) {
       int dg = db * 16 + (lane & 15);
       u16 vt[4];
 #pragma unroll
-      for (int r = 0; r < 4; r++) {
-        int t = t0 + krow4 + r;
-        int tl = t < T_end ? t : T_end - 1;
-        vt[r] = V0[(size_t)tl * HD + dg];  // masked kv rows have p == 0
-      }
+      for (int r = 0; r < 4; r++)
+        vt[r] = vlds[(krow4 + r) * HD + dg];
 #pragma unroll
       for (int r = 0; r < 4; r++) acc_o[db][r] *= alpha;
       acc_o[db] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
@@ -1465,7 +1475,7 @@ extern "C" hipError_t launch_gemm_bf16(const void* X, const void* W, void* Y,
                                        int N, int K, hipStream_t stream) {
   int gm = (M + BM - 1) / BM, gn = (N + BN - 1) / BN;
   int sk = 1;
-  if (accbuf) {
+  if (accbuf && gm * gn < 64) {  // split K only when the grid underfills
     while (sk < 8 && gm * gn * sk * 2 <= 256 && (K / BK) % (sk * 2) == 0)
       sk *= 2;
   }
