@@ -1475,7 +1475,7 @@ extern "C" hipError_t launch_gemm_bf16(const void* X, const void* W, void* Y,
                                        int N, int K, hipStream_t stream) {
   int gm = (M + BM - 1) / BM, gn = (N + BN - 1) / BN;
   int sk = 1;
-  if (accbuf && gm * gn < 64) {  // split K only when the grid underfills
+  if (accbuf && gm * gn < 160) {  // split K while the grid underfills
     while (sk < 8 && gm * gn * sk * 2 <= 256 && (K / BK) % (sk * 2) == 0)
       sk *= 2;
   }
