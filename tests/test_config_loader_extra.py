@@ -1,0 +1,104 @@
+"""Config round-trip, sharded-checkpoint, and sampler-distribution tests."""
+
+import json
+import os
+
+import numpy as np
+import pytest
+
+import llm_np_cp_amd as L
+from llm_np_cp_amd.core.config import ModelConfig, preset_config
+
+
+def test_config_json_roundtrip(tmp_path):
+    for name in ("llama-3.2-1b", "gemma-2-9b"):
+        cfg = preset_config(name)
+        p = tmp_path / f"{name}.json"
+        p.write_text(json.dumps(cfg.to_hf_dict()))
+        back = ModelConfig.from_json(str(p))
+        assert back.to_hf_dict() == cfg.to_hf_dict()
+        assert back.attn_scale == cfg.attn_scale
+        assert back.layer_types == cfg.layer_types
+
+
+def test_preset_shapes_match_hf_published():
+    """Spot-check the preset architecture tables."""
+    c = preset_config("llama-3.2-1b")
+    assert (c.vocab_size, c.hidden_size, c.num_hidden_layers) == \
+        (128256, 2048, 16)
+    assert (c.num_attention_heads, c.num_key_value_heads, c.head_dim) == \
+        (32, 8, 64)
+    g = preset_config("gemma-2-9b")
+    assert (g.vocab_size, g.hidden_size, g.num_hidden_layers) == \
+        (256000, 3584, 42)
+    assert g.head_dim == 256 and g.sliding_window == 4096
+    assert g.layer_types[0] == "sliding_attention"
+    assert g.layer_types[1] == "full_attention"
+
+
+def test_sharded_safetensors_checkpoint(tmp_path):
+    """Two-shard checkpoint with an index.json loads identically to the
+    single-file path (reference weight_map capability)."""
+    import torch
+    from safetensors.torch import save_file
+
+    from llm_np_cp_amd.io.loader import load_weights_numpy, random_weights
+
+    cfg = preset_config("tiny-llama")
+    w = random_weights(cfg, seed=77)
+    d = str(tmp_path / "ck")
+    os.makedirs(d)
+    names = sorted(w)
+    half = len(names) // 2
+    save_file({k: torch.from_numpy(w[k]) for k in names[:half]},
+              os.path.join(d, "model-00001-of-00002.safetensors"))
+    save_file({k: torch.from_numpy(w[k]) for k in names[half:]},
+              os.path.join(d, "model-00002-of-00002.safetensors"))
+    weight_map = {k: ("model-00001-of-00002.safetensors" if i < half else
+                      "model-00002-of-00002.safetensors")
+                  for i, k in enumerate(names)}
+    with open(os.path.join(d, "model.safetensors.index.json"), "w") as f:
+        json.dump({"weight_map": weight_map}, f)
+    with open(os.path.join(d, "config.json"), "w") as f:
+        json.dump(cfg.to_hf_dict(), f)
+
+    got = load_weights_numpy(d)
+    assert set(got) == set(w)
+    for k in names:
+        np.testing.assert_array_equal(got[k], w[k])
+
+
+def test_missing_checkpoint_raises(tmp_path):
+    from llm_np_cp_amd.io.loader import iter_safetensors
+
+    with pytest.raises(FileNotFoundError):
+        list(iter_safetensors(str(tmp_path)))
+
+
+def test_min_p_sampling_distribution():
+    """Sampled frequencies track the renormalized min-p distribution."""
+    from llm_np_cp_amd.runtime.sampling import SamplingParams, sample_token
+
+    logits = np.log(np.array([0.5, 0.3, 0.15, 0.05], dtype=np.float32))
+    # p_max=0.5; min_p=0.1 -> threshold 0.05: all four kept (>=)
+    rng = np.random.default_rng(0)
+    p = SamplingParams(strategy="min_p", min_p=0.1)
+    counts = np.zeros(4)
+    n = 4000
+    for _ in range(n):
+        counts[sample_token(logits, p, rng)] += 1
+    freq = counts / n
+    np.testing.assert_allclose(freq, [0.5, 0.3, 0.15, 0.05], atol=0.03)
+
+
+def test_lazy_random_weights_match_shapes():
+    from llm_np_cp_amd.io.loader import (LazyRandomWeights, hf_weight_shapes)
+
+    cfg = preset_config("tiny-gemma2")
+    lw = LazyRandomWeights(cfg, seed=3)
+    shapes = hf_weight_shapes(cfg)
+    for name, shape in shapes.items():
+        assert lw[name].shape == shape
+    # norm gammas near gemma's stored-gamma convention (~0)
+    g = lw["model.layers.0.input_layernorm.weight"]
+    assert abs(g.mean()) < 0.05
