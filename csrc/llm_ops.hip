@@ -1476,7 +1476,8 @@ extern "C" hipError_t launch_gemm_bf16(const void* X, const void* W, void* Y,
   int gm = (M + BM - 1) / BM, gn = (N + BN - 1) / BN;
   int sk = 1;
   if (accbuf && gm * gn < 160) {  // split K while the grid underfills
-    while (sk < 8 && gm * gn * sk * 2 <= 256 && (K / BK) % (sk * 2) == 0)
+    // target up to 2 blocks/CU (64 KB LDS each -> 2 resident max)
+    while (sk < 8 && gm * gn * sk * 2 <= 512 && (K / BK) % (sk * 2) == 0)
       sk *= 2;
   }
   if (sk > 1) {

@@ -43,7 +43,7 @@ class DeviceCacheHandle:
 class GPUModel:
     def __init__(self, config: ModelConfig, weights: Dict[str, np.ndarray],
                  dtype: str = "bf16", max_seq: int = 4096,
-                 prefill_chunk: int = 512, device: Optional[str] = None,
+                 prefill_chunk: int = 2048, device: Optional[str] = None,
                  seed: int = 0, force_tp_path: bool = False):
         if not torch.cuda.is_available():
             raise RuntimeError("GPUModel requires a GPU (MI355X)")
@@ -190,7 +190,7 @@ class GPUModel:
         self.cos_t = torch.from_numpy(np.cos(freqs).astype(np.float32)).to(dev)
         self.sin_t = torch.from_numpy(np.sin(freqs).astype(np.float32)).to(dev)
 
-        PC = self.PC = prefill_chunk
+        PC = self.PC = min(prefill_chunk, max(256, self.max_seq))
         H, I = self.H, self.inter_l
         self.b_h = torch.zeros(PC, H, **bf)
         self.b_xn = torch.zeros(PC, H, **bf)
