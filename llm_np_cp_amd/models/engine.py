@@ -223,7 +223,7 @@ class GPUModel:
         i32 = dict(dtype=torch.int32, device=dev)
         self.ids_buf = torch.zeros(PC, **i32)
         self.next_token = torch.zeros(1, **i32)
-        self.out_ring = torch.zeros(65536, **i32)
+        self.out_ring = torch.zeros(max(65536, S + 16), **i32)
         self.nout = torch.zeros(1, **i32)
         self.len_buf = torch.zeros(1, **i32)
         self.rng_ctr = torch.zeros(1, dtype=torch.int64, device=dev)
