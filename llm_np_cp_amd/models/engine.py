@@ -30,6 +30,7 @@ import torch
 from ..core.config import ModelConfig
 from ..ops import hip_ops as ho
 from ..parallel import tp as tpu
+from ..utils.tracing import trace_range
 
 
 class DeviceCacheHandle:
@@ -533,7 +534,8 @@ class GPUModel:
 
     def prefill(self, ids: np.ndarray, cache: Optional[DeviceCacheHandle] = None):
         cache = cache or self.make_cache(self.max_seq)
-        logits = self.forward(ids, cache, 0)
+        with trace_range(f"prefill[{len(np.asarray(ids).ravel())}]"):
+            logits = self.forward(ids, cache, 0)
         return cache, logits
 
     def decode(self, n_tokens: int, greedy: bool = True, min_p: float = 0.1,
