@@ -365,6 +365,39 @@ class GPUModel:
         ho.i32_set(self.len_buf, n)
         return out
 
+    def generate_tokens(self, prompt_ids, max_tokens: int,
+                        greedy: bool = True, min_p: float = 0.1,
+                        eos_id=None, chunk: int = 16, on_ids=None):
+        """Fast generate: device-side hipGraph decode in chunks, host
+        sees ids every `chunk` tokens (streaming + EOS stop).  Used by
+        runtime.generate() for greedy/min-p on GPU models."""
+        import time as _time
+        t0 = _time.perf_counter()
+        with trace_range("prefill"):
+            self.prefill(np.asarray(prompt_ids))
+        self.last_prefill_time_s = _time.perf_counter() - t0
+        out = []
+        produced = 0
+        first = True
+        with trace_range("decode"):
+            while produced < max_tokens:
+                n = min(chunk, max_tokens - produced)
+                ids = self.decode(n, greedy=greedy, min_p=min_p,
+                                  use_graph=True, first_from_logits=first)
+                first = False
+                produced += n
+                stop = False
+                take = list(ids)
+                if eos_id is not None and eos_id in take:
+                    take = take[:take.index(eos_id) + 1]
+                    stop = True
+                out.extend(take)
+                if on_ids:
+                    on_ids(take)
+                if stop:
+                    break
+        return out
+
     # ------------------------------------------------------------------
     # generic forward (oracle-parity / generate()-compatible)
     # ------------------------------------------------------------------

@@ -57,6 +57,29 @@ def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
     if emit is None and stream:
         emit = lambda s: (sys.stdout.write(s), sys.stdout.flush())
 
+    # fast path: device-side chunked decode loop (GPU engine, greedy or
+    # min-p, cached mode) — same sampler semantics, device RNG
+    if (use_cache and hasattr(model, "generate_tokens")
+            and params.strategy in ("greedy", "min_p")):
+        t0 = time.perf_counter()
+        pieces: List[int] = []
+
+        def _emit(ids_chunk):
+            if emit:
+                emit(tokenizer.decode(list(ids_chunk)))
+
+        ids = model.generate_tokens(
+            prompt_ids, max_tokens,
+            greedy=params.strategy == "greedy", min_p=params.min_p,
+            eos_id=eos if stop_on_eos else None, on_ids=_emit)
+        dt = time.perf_counter() - t0
+        tp = getattr(model, "last_prefill_time_s", 0.0)
+        res = GenerateResult(text=tokenizer.decode(list(ids)),
+                             token_ids=[int(i) for i in ids],
+                             prefill_time_s=tp,
+                             decode_time_s=max(dt - tp, 1e-9))
+        return res
+
     out_ids: List[int] = []
     t0 = time.perf_counter()
     logits = model.forward(np.asarray(prompt_ids, dtype=np.int64), kv_cache, 0)
