@@ -225,6 +225,28 @@ PRESETS = {
         query_pre_attn_scalar=256, sliding_window=4096,
         attn_logit_softcapping=50.0, final_logit_softcapping=30.0,
     ),
+    "llama-3.1-8b": dict(
+        model_type="llama", vocab_size=128256, hidden_size=4096,
+        intermediate_size=14336, num_hidden_layers=32,
+        num_attention_heads=32, num_key_value_heads=8, head_dim=128,
+        rms_norm_eps=1e-5, rope_theta=500000.0,
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192},
+        max_position_embeddings=131072, hidden_act="silu",
+        tie_word_embeddings=False, bos_token_id=128000,
+        eos_token_id=128001,
+    ),
+    "gemma-2-27b": dict(
+        model_type="gemma2", vocab_size=256000, hidden_size=4608,
+        intermediate_size=36864, num_hidden_layers=46,
+        num_attention_heads=32, num_key_value_heads=16, head_dim=128,
+        rms_norm_eps=1e-6, rope_theta=10000.0,
+        max_position_embeddings=8192, hidden_act="gelu_pytorch_tanh",
+        tie_word_embeddings=True, bos_token_id=2, eos_token_id=1,
+        query_pre_attn_scalar=144, sliding_window=4096,
+        attn_logit_softcapping=50.0, final_logit_softcapping=30.0,
+    ),
     # tiny configs for tests
     "tiny-llama-hd64": dict(    # exercises the hd=64 MFMA prefill path
         model_type="llama", vocab_size=512, hidden_size=128,

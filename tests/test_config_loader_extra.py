@@ -103,3 +103,14 @@ def test_lazy_random_weights_match_shapes():
     # norm gammas near gemma's stored-gamma convention (~0)
     g = lw["model.layers.0.input_layernorm.weight"]
     assert abs(g.mean()) < 0.05
+
+
+def test_additional_presets_construct():
+    from llm_np_cp_amd.io.loader import hf_weight_shapes
+
+    for name in ("llama-3.1-8b", "gemma-2-27b", "llama-3.2-3b"):
+        c = preset_config(name)
+        shapes = hf_weight_shapes(c)
+        n = sum(int(np.prod(s)) for s in shapes.values())
+        assert n > 1e9
+    assert "lm_head.weight" in hf_weight_shapes(preset_config("llama-3.1-8b"))
