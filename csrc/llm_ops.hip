@@ -8,18 +8,22 @@
 //    deep unrolled dwordx4 loads straight to VGPRs (no LDS round trip for
 //    the streamed operand — guide §5 "GEMV / M<=16 decode weights")
 //  - attention never materializes QK^T: online softmax in registers
-//  - KV cache is a preallocated pool, written in place by the RoPE kernel
-//    (replaces the reference's O(T^2) concat cache, llama3.2_model.py:325)
+//  - KV cache is a preallocated pool, written in place (decode: inside
+//    the fused attention kernel; prefill: k_rope_cache) — replaces the
+//    reference's O(T^2) concat cache (llama3.2_model.py:325)
 //  - every kernel is hipGraph-capture-safe (no mallocs/syncs; sequence
 //    position and sampled token ids live in device memory)
 //
 // Replaces (reference parity):
-//  K1 softmax RawKernel      -> folded into k_attn online softmax
-//  K2 cuBLAS GEMM/GEMV       -> k_gemv_bf16 (decode), k_gemm_bf16 (prefill)
-//  K3 elementwise ufuncs     -> k_rmsnorm (+resid), k_rope_cache, k_glu
-//  K4 KV concat              -> in-place pool write in k_rope_cache
-//  K5 repeat_kv              -> k_attn indexes kv_head = q_head / groups
-//  K7 torch.multinomial      -> k_sample (min-p via Gumbel argmax)
+//  K1 softmax RawKernel      -> online softmax in k_attn_dec /
+//                               k_attn_prefill_mfma / k_attn
+//  K2 cuBLAS GEMM/GEMV       -> k_gemv_{bf16,fp8}_t (decode),
+//                               k_gemm_bf16 MFMA (prefill)
+//  K3 elementwise ufuncs     -> fused GEMV staging (NORM/GLU/NORM2),
+//                               k_rmsnorm, k_glu, k_softcap, k_addinto
+//  K4 KV concat              -> in-place pool writes
+//  K5 repeat_kv              -> attention indexes kv_head = q_head/groups
+//  K7 torch.multinomial      -> k_logit_max + k_sample_pick + k_sample_fin
 
 #include <hip/hip_runtime.h>
 #include <cstdint>
