@@ -372,9 +372,14 @@ class GPUModel:
         sees ids every `chunk` tokens (streaming + EOS stop).  Used by
         runtime.generate() for greedy/min-p on GPU models."""
         import time as _time
+        prompt_ids = np.asarray(prompt_ids)
+        room = self.max_seq - len(prompt_ids.ravel()) - 1
+        if room <= 0:
+            raise ValueError(f"prompt fills the {self.max_seq}-token pool")
+        max_tokens = min(max_tokens, room)
         t0 = _time.perf_counter()
         with trace_range("prefill"):
-            self.prefill(np.asarray(prompt_ids))
+            self.prefill(prompt_ids)
         self.last_prefill_time_s = _time.perf_counter() - t0
         out = []
         produced = 0
