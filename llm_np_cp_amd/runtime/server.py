@@ -17,18 +17,8 @@ import threading
 import time
 from typing import Optional
 
-
-def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
-              dtype: str = "bf16", max_seq: int = 4096):
-    from fastapi import FastAPI
+try:  # pydantic model must live at module scope (ForwardRef resolution)
     from pydantic import BaseModel
-
-    import llm_np_cp_amd as L
-
-    tok, model, cfg = L.load_model(model_name, backend=backend,
-                                   dtype=dtype, max_seq=max_seq)
-    lock = threading.Lock()  # batch-1 engine: serialize requests
-    app = FastAPI(title="llm_np_cp_amd", version=L.__version__)
 
     class CompletionRequest(BaseModel):
         prompt: str
@@ -38,6 +28,20 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
         strategy: str = "min_p"   # min_p | greedy | top_k | top_p
         seed: Optional[int] = None
         stop_on_eos: bool = True
+except ImportError:  # pragma: no cover - serving is optional
+    CompletionRequest = None
+
+
+def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
+              dtype: str = "bf16", max_seq: int = 4096):
+    from fastapi import Body, FastAPI
+
+    import llm_np_cp_amd as L
+
+    tok, model, cfg = L.load_model(model_name, backend=backend,
+                                   dtype=dtype, max_seq=max_seq)
+    lock = threading.Lock()  # batch-1 engine: serialize requests
+    app = FastAPI(title="llm_np_cp_amd", version=L.__version__)
 
     @app.get("/health")
     def health():
@@ -45,7 +49,7 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
                 "model_type": cfg.model_type, "backend": type(model).__name__}
 
     @app.post("/v1/completions")
-    def completions(req: CompletionRequest):
+    def completions(req: CompletionRequest = Body(...)):
         params = L.SamplingParams(strategy=req.strategy, min_p=req.min_p,
                                   temperature=req.temperature, seed=req.seed)
         t0 = time.time()
