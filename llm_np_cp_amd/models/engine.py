@@ -214,7 +214,10 @@ class GPUModel:
         self.b_t2 = torch.zeros(H, **bf)   # gemma ffn-out delta
         self.b_gu = torch.zeros(2 * I, **bf)
         # split-T decode attention scratch (partials + per-head tickets)
-        self.attn_split = min(16, max(4, (self.max_seq + 255) // 256))
+        import os as _os
+        _sp = _os.environ.get("LLM_ATTN_SPLIT")
+        self.attn_split = (int(_sp) if _sp else
+                           min(16, max(4, (self.max_seq + 255) // 256)))
         self.attn_scratch = torch.zeros(
             self.nh_l * self.attn_split * (hd + 2), dtype=torch.float32,
             device=dev)
