@@ -262,6 +262,15 @@ PRESETS = {
         rope_theta=10000.0, max_position_embeddings=512, hidden_act="silu",
         tie_word_embeddings=True,
     ),
+    "tiny-gemma2-hd64": dict(   # exercises gemma + MFMA prefill (hd=64)
+        model_type="gemma2", vocab_size=512, hidden_size=128,
+        intermediate_size=256, num_hidden_layers=4, num_attention_heads=2,
+        num_key_value_heads=1, head_dim=64, rms_norm_eps=1e-6,
+        rope_theta=10000.0, max_position_embeddings=512,
+        hidden_act="gelu_pytorch_tanh", tie_word_embeddings=True,
+        query_pre_attn_scalar=64, sliding_window=8,
+        attn_logit_softcapping=50.0, final_logit_softcapping=30.0,
+    ),
     "tiny-gemma2": dict(
         model_type="gemma2", vocab_size=512, hidden_size=64,
         intermediate_size=128, num_hidden_layers=4, num_attention_heads=4,

@@ -264,3 +264,50 @@ def test_decode_overflow_guard():
     gpu.prefill(np.arange(1, 33))
     with _pytest.raises(ValueError, match="overflow"):
         gpu.decode(100, greedy=True, use_graph=False)
+
+
+def test_gemma_hd64_mfma_prefill_matches_oracle():
+    """Gemma semantics (sliding window + softcaps + sandwich norms)
+    through the MFMA prefill path (hd=64) vs the NumPy oracle."""
+    import llm_np_cp_amd as L
+
+    cfg, gpu, ref = make_pair("tiny-gemma2-hd64", seed=60)
+    rng = np.random.default_rng(61)
+    ids = rng.integers(0, cfg.vocab_size, size=21)  # > sliding_window=8
+
+    from llm_np_cp_amd.models.numpy_ref import NumpyKVCache
+    ref_logits = ref.forward(ids, NumpyKVCache(cfg, 64), 0)
+    got = gpu.forward(ids, gpu.make_cache(64), 0)[0]
+    assert np.argmax(got) == np.argmax(ref_logits[-1])
+
+    tok = L.ByteTokenizer()
+    p = L.SamplingParams(strategy="greedy")
+    a = L.generate("Hello there", tok, ref, max_tokens=8, stream=False,
+                   params=p, stop_on_eos=False)
+    b = L.generate("Hello there", tok, gpu, max_tokens=8, stream=False,
+                   params=p, stop_on_eos=False)
+    assert a.token_ids == b.token_ids
+
+
+def test_fast_generate_eos_stop():
+    """generate() fast path stops at EOS."""
+    import llm_np_cp_amd as L
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.engine import GPUModel
+
+    cfg = L.preset_config("tiny-llama")
+    w = random_weights(cfg, seed=70)
+    m = GPUModel(cfg, w, max_seq=128)
+    tok = L.ByteTokenizer()
+    p = L.SamplingParams(strategy="greedy")
+    # learn what greedy produces, then declare token #3 the EOS
+    r = L.generate("abc", tok, m, max_tokens=8, stream=False, params=p,
+                   stop_on_eos=False)
+    assert len(r.token_ids) == 8
+    eos = r.token_ids[2]
+    cfg.eos_token_id = eos
+    r2 = L.generate("abc", tok, m, max_tokens=8, stream=False, params=p,
+                    stop_on_eos=True)
+    assert r2.token_ids[-1] == eos
+    assert len(r2.token_ids) <= 3 + 13  # stops within the eos chunk
+    assert r2.token_ids[:3] == r.token_ids[:3]
