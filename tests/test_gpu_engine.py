@@ -306,8 +306,7 @@ def test_fast_generate_eos_stop():
     assert len(r.token_ids) == 8
     eos = r.token_ids[2]
     cfg.eos_token_id = eos
+    first = r.token_ids.index(eos)  # greedy can repeat tokens
     r2 = L.generate("abc", tok, m, max_tokens=8, stream=False, params=p,
                     stop_on_eos=True)
-    assert r2.token_ids[-1] == eos
-    assert len(r2.token_ids) <= 3 + 13  # stops within the eos chunk
-    assert r2.token_ids[:3] == r.token_ids[:3]
+    assert r2.token_ids == r.token_ids[:first + 1]
