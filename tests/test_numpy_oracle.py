@@ -136,3 +136,36 @@ def test_sliding_window_limits_attention():
     np.testing.assert_allclose(la[-1], lb[-1], rtol=1e-6, atol=1e-6)
     # sanity: an in-window difference does change logits
     assert not np.allclose(la[-2], lb[-2])
+
+
+def test_llama3_rope_scaling_matches_transformers():
+    """llama3 rope_scaling (which the reference ignored) vs HF at
+    positions beyond original_max_position_embeddings/factor."""
+    from transformers import LlamaConfig, LlamaForCausalLM
+
+    cfg = preset_config("tiny-llama")
+    cfg.rope_theta = 500000.0
+    cfg.rope_scaling = {"rope_type": "llama3", "factor": 4.0,
+                        "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                        "original_max_position_embeddings": 16}
+    hf_cfg = LlamaConfig(
+        vocab_size=cfg.vocab_size, hidden_size=cfg.hidden_size,
+        intermediate_size=cfg.intermediate_size,
+        num_hidden_layers=cfg.num_hidden_layers,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        head_dim=cfg.head_dim, rms_norm_eps=cfg.rms_norm_eps,
+        rope_theta=cfg.rope_theta, rope_scaling=dict(cfg.rope_scaling),
+        max_position_embeddings=cfg.max_position_embeddings,
+        tie_word_embeddings=True, attn_implementation="eager",
+    )
+    torch.manual_seed(11)
+    hf = LlamaForCausalLM(hf_cfg).eval()
+    model = NumpyModel(cfg, np_weights_from_hf(hf))
+
+    rng = np.random.default_rng(12)
+    ids = rng.integers(0, cfg.vocab_size, size=48)  # > 16/4 and > 16
+    with torch.no_grad():
+        ref = hf(torch.tensor(ids[None])).logits[0].numpy()
+    got = model.forward(ids, NumpyKVCache(cfg, 64), 0)
+    np.testing.assert_allclose(got, ref, rtol=3e-4, atol=3e-4)
