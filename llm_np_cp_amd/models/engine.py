@@ -21,7 +21,6 @@ spec in SURVEY §2.4 rather than translated.
 
 from __future__ import annotations
 
-import math
 from typing import Dict, Optional
 
 import numpy as np
