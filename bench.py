@@ -77,7 +77,14 @@ def main():
                      first_from_logits=True)
 
     if use_graph:
-        model.capture_decode_graph(True, 0.1)  # no-op if warmup captured
+        try:
+            model.capture_decode_graph(True, 0.1)  # no-op if warmup captured
+        except Exception as e:
+            # e.g. collectives not capturable in this stack: eager timing
+            if rank == 0:
+                print(f"# graph capture unavailable ({type(e).__name__}); "
+                      f"timing eager launches", file=sys.stderr)
+            use_graph = False
 
     # timed region: exactly K decode steps
     barrier()
