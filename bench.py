@@ -25,10 +25,10 @@ def main():
     ap.add_argument("--steps", type=int, default=200)
     ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--model", type=str, default="llama-3.2-1b",
-                    help="preset or checkpoint dir (default llama-3.2-1b "
-                         "at every N so the driver's 1/2/4/8-GPU scaling "
-                         "curve compares like with like; gemma-2-9b &c "
-                         "via this flag)")
+                    help="model preset (default llama-3.2-1b at every N "
+                         "so the driver's 1/2/4/8-GPU scaling curve "
+                         "compares like with like; gemma-2-9b etc. via "
+                         "this flag; synthetic random-init weights)")
     ap.add_argument("--prompt-len", type=int, default=64)
     ap.add_argument("--max-seq", type=int, default=None)
     ap.add_argument("--dtype", type=str, default="bf16",
