@@ -114,3 +114,31 @@ def test_additional_presets_construct():
         n = sum(int(np.prod(s)) for s in shapes.values())
         assert n > 1e9
     assert "lm_head.weight" in hf_weight_shapes(preset_config("llama-3.1-8b"))
+
+
+def test_hf_config_field_variants():
+    """Real checkpoint config.json variants: Gemma-2 uses
+    'hidden_activation'; older Llama rope_scaling uses 'type'."""
+    d = {
+        "model_type": "gemma2", "vocab_size": 256, "hidden_size": 64,
+        "intermediate_size": 128, "num_hidden_layers": 2,
+        "num_attention_heads": 4, "num_key_value_heads": 2,
+        "head_dim": 16, "hidden_activation": "gelu_pytorch_tanh",
+        "query_pre_attn_scalar": 16, "sliding_window": 8,
+    }
+    cfg = ModelConfig.from_hf_dict(d)
+    assert cfg.hidden_act == "gelu_pytorch_tanh"
+    assert cfg.is_sliding(0) and not cfg.is_sliding(1)
+
+    d2 = {
+        "model_type": "llama", "vocab_size": 256, "hidden_size": 64,
+        "intermediate_size": 128, "num_hidden_layers": 1,
+        "num_attention_heads": 4, "rope_theta": 500000.0,
+        "rope_scaling": {"type": "llama3", "factor": 8.0,
+                         "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                         "original_max_position_embeddings": 8192},
+    }
+    cfg2 = ModelConfig.from_hf_dict(d2)
+    assert cfg2.head_dim == 16  # derived hidden/heads
+    plain = ModelConfig.from_hf_dict({**d2, "rope_scaling": None})
+    assert np.any(cfg2.rope_inv_freq() < plain.rope_inv_freq())
