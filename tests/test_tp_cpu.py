@@ -57,17 +57,20 @@ def _worker(rank, world, port, results):
     dist.destroy_process_group()
 
 
-def test_tp2_shard_math_gloo():
-    world = 2
+import pytest
+
+
+@pytest.mark.parametrize("world,port", [(2, 29531), (4, 29532)])
+def test_tp_shard_math_gloo(world, port):
     ctx = mp.get_context("spawn")
     with ctx.Manager() as mgr:
         results = mgr.dict()
-        procs = [ctx.Process(target=_worker, args=(r, world, 29531, results))
+        procs = [ctx.Process(target=_worker, args=(r, world, port, results))
                  for r in range(world)]
         for p in procs:
             p.start()
         for p in procs:
-            p.join(timeout=120)
+            p.join(timeout=180)
         for r in range(world):
             assert results.get(r) == "ok", f"rank {r} failed"
 
