@@ -57,9 +57,6 @@ def _worker(rank, world, port, results):
     dist.destroy_process_group()
 
 
-import pytest
-
-
 @pytest.mark.parametrize("world,port", [(2, 29531), (4, 29532)])
 def test_tp_shard_math_gloo(world, port):
     ctx = mp.get_context("spawn")
