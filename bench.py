@@ -77,12 +77,14 @@ def main():
                      first_from_logits=True)
 
     if use_graph:
-        try:
-            model.capture_decode_graph(True, 0.1)  # no-op if warmup captured
-        except Exception as e:
-            # e.g. collectives not capturable in this stack: eager timing
+        # no-op if the warmup already captured; on failure the step it
+        # executed is harmless here (untimed region)
+        _, ok = model.capture_decode_graph(True, 0.1)
+        if not ok:
             if rank == 0:
-                print(f"# graph capture unavailable ({type(e).__name__}); "
+                err = getattr(model, "_graph_error", None)
+                print(f"# graph capture unavailable "
+                      f"({type(err).__name__ if err else '?'}: {err}); "
                       f"timing eager launches", file=sys.stderr)
             use_graph = False
 

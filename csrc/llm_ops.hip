@@ -1244,15 +1244,19 @@ k_logit_max(const float* __restrict__ logits, int V,
   }
 }
 
-// pass 2: winner = argmax over kept tokens of (logit [+ Gumbel]) -> *pick
+// pass 2: winner = argmax over kept tokens of (logit/T [+ Gumbel]) -> *pick
+// Temperature folds into both the min-p keep-set and the Gumbel score:
+//   p_i(T) >= min_p * p_max(T)  <=>  l_i >= l_max + T*ln(min_p)
+//   Gumbel-argmax over p(T)     <=>  argmax of l_i/T + G_i
 extern "C" __global__ void __launch_bounds__(256)
 k_sample_pick(const float* __restrict__ logits, int V, float min_p,
-              int greedy, uint64_t seed,
+              int greedy, uint64_t seed, float inv_temp,
               const unsigned long long* __restrict__ gmax,
               const uint64_t* __restrict__ ctr,
               unsigned long long* __restrict__ pick) {
   float thresh = -INFINITY;
-  if (!greedy) thresh = fkey_inv((uint32_t)(*gmax >> 32)) + __logf(min_p);
+  if (!greedy)
+    thresh = fkey_inv((uint32_t)(*gmax >> 32)) + __logf(min_p) / inv_temp;
   const uint32_t c = (uint32_t)(*ctr);
   float bv = -INFINITY;
   int bi = 0x7fffffff;
@@ -1264,7 +1268,7 @@ k_sample_pick(const float* __restrict__ logits, int V, float min_p,
       uint32_t r = hash32(hash32((uint32_t)i ^ (c * 0x9e3779b9u)) ^
                           (uint32_t)seed);
       float u = (r + 1.0f) * 2.3283064e-10f;  // (0,1]
-      sc = v - __logf(-__logf(u));
+      sc = v * inv_temp - __logf(-__logf(u));
     }
     if (sc > bv || (sc == bv && i < bi)) { bv = sc; bi = i; }
   }
@@ -1304,9 +1308,10 @@ extern "C" __global__ void k_sample_fin(
 }
 
 extern "C" hipError_t launch_sample(const void* logits, int V, float min_p,
-                                    int greedy, uint64_t seed, void* ctr,
-                                    void* gmax, void* pick, void* next_token,
-                                    void* out_ring, void* nout, void* len_ptr,
+                                    int greedy, uint64_t seed, float inv_temp,
+                                    void* ctr, void* gmax, void* pick,
+                                    void* next_token, void* out_ring,
+                                    void* nout, void* len_ptr,
                                     int bump_len, hipStream_t stream) {
   int blocks = (V + 255) / 256;
   if (blocks > 512) blocks = 512;
@@ -1314,7 +1319,7 @@ extern "C" hipError_t launch_sample(const void* logits, int V, float min_p,
     hipLaunchKernelGGL(k_logit_max, dim3(blocks), dim3(256), 0, stream,
                        (const float*)logits, V, (unsigned long long*)gmax);
   hipLaunchKernelGGL(k_sample_pick, dim3(blocks), dim3(256), 0, stream,
-                     (const float*)logits, V, min_p, greedy, seed,
+                     (const float*)logits, V, min_p, greedy, seed, inv_temp,
                      (const unsigned long long*)gmax, (const uint64_t*)ctr,
                      (unsigned long long*)pick);
   hipLaunchKernelGGL(k_sample_fin, dim3(1), dim3(1), 0, stream,

@@ -365,20 +365,29 @@ class GPUModel:
             out[done:done + M] = loc
             done += M
         ho.i32_set(self.len_buf, n)
+        # mirror the device len so a following decode()'s overflow guard
+        # counts from the right base (KV-pool OOB otherwise)
+        self._host_len = n
         return out
 
     def generate_tokens(self, prompt_ids, max_tokens: int,
                         greedy: bool = True, min_p: float = 0.1,
-                        eos_id=None, chunk: int = 16, on_ids=None):
+                        eos_id=None, chunk: int = 16, on_ids=None,
+                        temperature: float = 1.0):
         """Fast generate: device-side hipGraph decode in chunks, host
         sees ids every `chunk` tokens (streaming + EOS stop).  Used by
-        runtime.generate() for greedy/min-p on GPU models."""
+        runtime.generate() for greedy/min-p on GPU models.
+        ``eos_id`` may be an int or a collection of ints (HF configs
+        often store a list, e.g. Llama-3.2-Instruct)."""
         import time as _time
         prompt_ids = np.asarray(prompt_ids)
         room = self.max_seq - len(prompt_ids.ravel()) - 1
         if room <= 0:
             raise ValueError(f"prompt fills the {self.max_seq}-token pool")
         max_tokens = min(max_tokens, room)
+        eos_set = (set() if eos_id is None else
+                   {int(eos_id)} if np.isscalar(eos_id) else
+                   {int(e) for e in eos_id})
         t0 = _time.perf_counter()
         with trace_range("prefill"):
             self.prefill(prompt_ids)
@@ -390,13 +399,15 @@ class GPUModel:
             while produced < max_tokens:
                 n = min(chunk, max_tokens - produced)
                 ids = self.decode(n, greedy=greedy, min_p=min_p,
-                                  use_graph=True, first_from_logits=first)
+                                  use_graph=True, first_from_logits=first,
+                                  temperature=temperature)
                 first = False
                 produced += n
                 stop = False
                 take = list(ids)
-                if eos_id is not None and eos_id in take:
-                    take = take[:take.index(eos_id) + 1]
+                hit = [j for j, t in enumerate(take) if int(t) in eos_set]
+                if hit:
+                    take = take[:hit[0] + 1]
                     stop = True
                 out.extend(take)
                 if on_ids:
@@ -450,7 +461,8 @@ class GPUModel:
             return (lw["wqkv_q"], lw["wo_q"], lw["wgu_q"], lw["wdown_q"])
         return (lw["wqkv"], lw["wo"], lw["wgu"], lw["wdown"])
 
-    def _decode_step(self, greedy: bool, min_p: float):
+    def _decode_step(self, greedy: bool, min_p: float,
+                     temperature: float = 1.0):
         """Fused decode path: 4 kernels/layer (llama) or 5 (gemma) —
         RMSNorm and GLU live inside the GEMV staging pass, RoPE + KV
         write inside the attention kernel; a side stream prefetches the
@@ -545,32 +557,50 @@ class GPUModel:
             tpu.all_gather_into(self.b_logits, self.b_logits_l)
         ho.sample(self.b_logits, min_p, greedy, self.seed, self.rng_ctr,
                   self.s_gmax, self.s_pick, self.next_token, self.out_ring,
-                  self.nout, self.len_buf, bump_len=True)
+                  self.nout, self.len_buf, bump_len=True,
+                  temperature=temperature)
 
-    def capture_decode_graph(self, greedy: bool = True, min_p: float = 0.1):
+    def capture_decode_graph(self, greedy: bool = True, min_p: float = 0.1,
+                             temperature: float = 1.0):
         """Capture one decode step into a hipGraph (fixed shapes: KV pool
         is preallocated and the position is a device scalar — SURVEY §7
         'graph must be shape-stable').
 
         NOTE: the warm-up executes ONE REAL decode step (advances the
-        sequence by one token).  Returns the number of real steps taken
-        (1 on fresh capture, 0 when the graph is reused)."""
-        mode = (greedy, min_p)
+        sequence by one token).  Returns ``(steps_taken, ok)``:
+        ``steps_taken`` is the number of real decode steps executed here
+        (1 whenever the warm-up ran, 0 when the cached graph is reused)
+        — callers MUST account for it even when ``ok`` is False.  On
+        capture failure ``ok`` is False, ``self._graph_failed`` is set
+        (no further capture attempts) and the exception is kept in
+        ``self._graph_error``."""
+        mode = (greedy, min_p, temperature)
         if self._graph_mode == mode:
-            return 0
+            return 0, True
+        if getattr(self, "_graph_failed", False):
+            return 0, False
         # warm-up (a real step) on a side stream, then capture
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
-            self._decode_step(greedy, min_p)
+            self._decode_step(greedy, min_p, temperature)
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
-        g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
-            self._decode_step(greedy, min_p)
+        try:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._decode_step(greedy, min_p, temperature)
+        except Exception as e:
+            # the warm-up step above DID run (token committed to
+            # out_ring, len_buf advanced) — report it so decode() does
+            # not double-count (ADVICE r1 medium)
+            self._graph_failed = True
+            self._graph_error = e
+            torch.cuda.synchronize()
+            return 1, False
         self._graph = g
         self._graph_mode = mode
-        return 1
+        return 1, True
 
     def prefill(self, ids: np.ndarray, cache: Optional[DeviceCacheHandle] = None):
         cache = cache or self.make_cache(self.max_seq)
@@ -579,7 +609,8 @@ class GPUModel:
         return cache, logits
 
     def decode(self, n_tokens: int, greedy: bool = True, min_p: float = 0.1,
-               use_graph: bool = True, first_from_logits: bool = True):
+               use_graph: bool = True, first_from_logits: bool = True,
+               temperature: float = 1.0):
         """Generate n_tokens ids device-side; returns int32 numpy ids.
         Assumes prefill() ran (len_buf == prompt length, logits ready)."""
         if getattr(self, "_host_len", 0) + n_tokens > self.max_seq:
@@ -592,21 +623,21 @@ class GPUModel:
             ho.sample(self.b_logits, min_p, greedy, self.seed, self.rng_ctr,
                       self.s_gmax, self.s_pick, self.next_token,
                       self.out_ring, self.nout, self.len_buf,
-                      bump_len=False)
+                      bump_len=False, temperature=temperature)
         n_steps = n_tokens - (1 if first_from_logits else 0)
         if use_graph and n_steps > 0:
-            try:
-                n_steps -= self.capture_decode_graph(greedy, min_p)
-            except Exception:
-                # e.g. RCCL collectives not capturable in this stack:
-                # fall back to eager launches (correctness first)
+            # capture_decode_graph reports the warm-up step it executed
+            # even when capture fails, so n_steps stays exact either way
+            taken, ok = self.capture_decode_graph(greedy, min_p, temperature)
+            n_steps -= taken
+            if not ok:
                 use_graph = False
         if use_graph and n_steps > 0:
             for _ in range(n_steps):
                 self._graph.replay()
         else:
             for _ in range(n_steps):
-                self._decode_step(greedy, min_p)
+                self._decode_step(greedy, min_p, temperature)
         torch.cuda.synchronize()
         n = int(self.nout.item())
         return self.out_ring[:n].cpu().numpy()[max(0, n - n_tokens):]

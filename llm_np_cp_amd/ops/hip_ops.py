@@ -50,7 +50,8 @@ _SIGS = {
     "launch_embed": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 +
                     [ctypes.c_float, ctypes.c_void_p],
     "launch_sample": [ctypes.c_void_p, ctypes.c_int, ctypes.c_float,
-                      ctypes.c_int, ctypes.c_uint64] + [ctypes.c_void_p] * 7 +
+                      ctypes.c_int, ctypes.c_uint64, ctypes.c_float] +
+                     [ctypes.c_void_p] * 7 +
                      [ctypes.c_int, ctypes.c_void_p],
     "launch_gemm_bf16": [ctypes.c_void_p] * 5 + [ctypes.c_int] * 3 +
                         [ctypes.c_void_p],
@@ -215,13 +216,17 @@ def sample(logits: torch.Tensor, min_p: float, greedy: bool, seed: int,
            ctr: torch.Tensor, gmax: torch.Tensor, pick: torch.Tensor,
            next_token: torch.Tensor, out_ring: torch.Tensor,
            nout: torch.Tensor, len_ptr: torch.Tensor,
-           bump_len: bool = True):
+           bump_len: bool = True, temperature: float = 1.0):
     """min-p / greedy sampler: parallel max + Gumbel-argmax + commit.
-    gmax/pick are u64 scratch (zeroed once; the commit kernel re-zeros)."""
+    gmax/pick are u64 scratch (zeroed once; the commit kernel re-zeros).
+    Temperature scales device-side (min-p keep-set + Gumbel score), so
+    the GPU fast path matches the CPU sample_token() semantics."""
     V = logits.shape[-1]
+    inv_temp = 1.0 / max(float(temperature), 1e-6)
     _check(lib().launch_sample(
         _ptr(logits), V, ctypes.c_float(min_p), 1 if greedy else 0,
-        ctypes.c_uint64(seed), _ptr(ctr), _ptr(gmax), _ptr(pick),
+        ctypes.c_uint64(seed), ctypes.c_float(inv_temp),
+        _ptr(ctr), _ptr(gmax), _ptr(pick),
         _ptr(next_token), _ptr(out_ring), _ptr(nout), _ptr(len_ptr),
         1 if bump_len else 0, _stream()), "sample")
 

@@ -48,7 +48,11 @@ def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
     params = params or SamplingParams()
     rng = np.random.default_rng(params.seed)
     prompt_ids = list(tokenizer.encode(prompt))
+    # HF configs store eos_token_id as an int OR a list (e.g.
+    # Llama-3.2-Instruct: [128001, 128008, 128009]) — normalize to a set
     eos = getattr(model.config, "eos_token_id", None)
+    eos_set = (set() if eos is None else
+               {int(eos)} if np.isscalar(eos) else {int(e) for e in eos})
 
     if kv_cache is None:
         kv_cache = model.make_cache(len(prompt_ids) + max_tokens + 1)
@@ -71,7 +75,8 @@ def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
         ids = model.generate_tokens(
             prompt_ids, max_tokens,
             greedy=params.strategy == "greedy", min_p=params.min_p,
-            eos_id=eos if stop_on_eos else None, on_ids=_emit)
+            eos_id=eos_set if stop_on_eos else None, on_ids=_emit,
+            temperature=params.temperature)
         dt = time.perf_counter() - t0
         tp = getattr(model, "last_prefill_time_s", 0.0)
         res = GenerateResult(text=tokenizer.decode(list(ids)),
@@ -92,7 +97,7 @@ def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
         out_ids.append(next_id)
         if emit:
             emit(tokenizer.decode([next_id]))
-        if stop_on_eos and eos is not None and next_id == eos:
+        if stop_on_eos and next_id in eos_set:
             break
         if use_cache:
             logits = model.forward(np.asarray([next_id], dtype=np.int64),

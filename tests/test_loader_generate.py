@@ -87,3 +87,26 @@ def test_rope_scaling_llama3_changes_freqs():
     assert np.any(scaled < plain * 0.5)
     # high-frequency components unchanged
     assert np.allclose(scaled[0], plain[0])
+
+
+def test_eos_list_stops_generation():
+    """HF configs may store eos_token_id as a LIST (Llama-3.2-Instruct:
+    [128001, 128008, 128009]); generation must stop on membership."""
+    tok, model, cfg = L.load_model("tiny-llama", backend="numpy", seed=3)
+    p = SamplingParams(strategy="greedy")
+    ref = L.generate("abcd", tok, model, max_tokens=6, stream=False,
+                     params=p, stop_on_eos=False)
+    assert len(ref.token_ids) == 6
+    # declare one of the generated tokens "eos" via a LIST; generation
+    # must stop at its FIRST occurrence
+    eos_tok = int(ref.token_ids[2])
+    first = ref.token_ids.index(eos_tok)
+    cfg.eos_token_id = [999999, eos_tok]
+    r = L.generate("abcd", tok, model, max_tokens=6, stream=False,
+                   params=p, stop_on_eos=True)
+    assert r.token_ids == ref.token_ids[:first + 1]
+    # scalar eos still works
+    cfg.eos_token_id = eos_tok
+    r2 = L.generate("abcd", tok, model, max_tokens=6, stream=False,
+                    params=p, stop_on_eos=True)
+    assert r2.token_ids == ref.token_ids[:first + 1]
