@@ -11,14 +11,18 @@ import subprocess
 import sys
 
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-SRC = os.path.join(ROOT, "csrc", "llm_ops.hip")
+SRCS = [
+    os.path.join(ROOT, "csrc", "llm_ops.hip"),
+    os.path.join(ROOT, "csrc", "xgmi_comm.hip"),
+]
+DEPS = [os.path.join(ROOT, "csrc", "common.h")]
 OUT = os.path.join(ROOT, "llm_np_cp_amd", "_libllmops.so")
 
 
 def build(verbose: bool = True) -> str:
     cmd = [
         "hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17",
-        "-fPIC", "-shared", SRC, "-o", OUT,
+        "-fPIC", "-shared", *SRCS, "-o", OUT,
     ]
     if verbose:
         print("+", " ".join(cmd))
@@ -29,7 +33,8 @@ def build(verbose: bool = True) -> str:
 def needs_rebuild() -> bool:
     if not os.path.exists(OUT):
         return True
-    return os.path.getmtime(SRC) > os.path.getmtime(OUT)
+    out_m = os.path.getmtime(OUT)
+    return any(os.path.getmtime(s) > out_m for s in SRCS + DEPS)
 
 
 def ensure_built(verbose: bool = False) -> str:

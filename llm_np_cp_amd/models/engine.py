@@ -77,6 +77,12 @@ class GPUModel:
         self.hd = cfg.head_dim
         self.H = cfg.hidden_size
 
+        # one-shot xGMI collectives for decode-sized payloads (validated
+        # on hardware at init; silent RCCL fallback when unavailable)
+        if self.world > 1:
+            slot_bytes = max(self.H * 2, self.vocab_l * 4, 1 << 16)
+            tpu.init_xgmi(self.device, slot_bytes)
+
         self._upload_weights(weights)
         self._alloc_state(prefill_chunk)
         self._graph = None
@@ -351,17 +357,17 @@ class GPUModel:
                     accbuf=self.b_gemm_acc)
             if self.final_softcap:
                 ho.softcap(logits_buf[:M], self.final_softcap)
-            torch.cuda.synchronize()
-            loc = logits_buf[:M].float().cpu().numpy()
             if self.world > 1:
-                t = torch.from_numpy(loc).to(self.device)
-                full = torch.zeros(M, self.config.vocab_size,
-                                   device=self.device)
+                # gather vocab shards on-device (each rank holds the
+                # columns [r*vocab_l, (r+1)*vocab_l) of rows [0, M))
                 import torch.distributed as dist
+                t = logits_buf[:M].float().contiguous()
                 chunks = [torch.empty_like(t) for _ in range(self.world)]
                 dist.all_gather(chunks, t)
-                full = torch.cat(chunks, dim=1)
-                loc = full.cpu().numpy()
+                loc = torch.cat(chunks, dim=1).cpu().numpy()
+            else:
+                torch.cuda.synchronize()
+                loc = logits_buf[:M].float().cpu().numpy()
             out[done:done + M] = loc
             done += M
         ho.i32_set(self.len_buf, n)
@@ -639,5 +645,8 @@ class GPUModel:
             for _ in range(n_steps):
                 self._decode_step(greedy, min_p, temperature)
         torch.cuda.synchronize()
+        comm = tpu.xgmi_comm()
+        if comm is not None:
+            comm.check()  # raise if a one-shot collective timed out
         n = int(self.nout.item())
         return self.out_ring[:n].cpu().numpy()[max(0, n - n_tokens):]

@@ -63,6 +63,19 @@ _SIGS = {
                        ctypes.c_void_p],
     "launch_i32_set": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
     "launch_i32_add": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
+    # one-shot xGMI collectives (csrc/xgmi_comm.hip)
+    "launch_xgmi_coll": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                         ctypes.c_int, ctypes.c_int, ctypes.c_long,
+                         ctypes.c_long, ctypes.c_int, ctypes.c_int,
+                         ctypes.c_long, ctypes.c_void_p],
+    "xc_alloc": [ctypes.c_long, ctypes.c_void_p],
+    "xc_free": [ctypes.c_void_p],
+    "xc_memset": [ctypes.c_void_p, ctypes.c_int, ctypes.c_long],
+    "xc_h2d": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_long],
+    "xc_d2h": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_long],
+    "xc_ipc_handle": [ctypes.c_void_p, ctypes.c_void_p],
+    "xc_ipc_open": [ctypes.c_void_p, ctypes.c_void_p],
+    "xc_ipc_close": [ctypes.c_void_p],
 }
 
 
@@ -274,3 +287,72 @@ def i32_set(buf: torch.Tensor, v: int):
 
 def i32_add(buf: torch.Tensor, v: int):
     _check(lib().launch_i32_add(_ptr(buf), v, _stream()), "i32_add")
+
+
+# ----------------------------------------------------------------------
+# one-shot xGMI collectives (csrc/xgmi_comm.hip) — raw-pointer plumbing
+# for the peer-mapped comm buffers (outside torch's caching allocator so
+# the IPC base pointer is stable)
+# ----------------------------------------------------------------------
+
+XC_MODE_AR_BF16, XC_MODE_AR_F32, XC_MODE_GATHER = 0, 1, 2
+XC_OFF_EPOCH, XC_OFF_TICKET, XC_OFF_ERR = 64, 72, 76
+XC_OFF_DATA = 4096
+XC_IPC_HANDLE_BYTES = 64
+
+
+def xc_alloc(nbytes: int) -> int:
+    p = ctypes.c_void_p(0)
+    _check(lib().xc_alloc(ctypes.c_long(nbytes), ctypes.byref(p)), "xc_alloc")
+    return p.value
+
+
+def xc_free(ptr: int):
+    _check(lib().xc_free(ctypes.c_void_p(ptr)), "xc_free")
+
+
+def xc_memset(ptr: int, val: int, nbytes: int):
+    _check(lib().xc_memset(ctypes.c_void_p(ptr), val,
+                           ctypes.c_long(nbytes)), "xc_memset")
+
+
+def xc_h2d(dst: int, data: bytes):
+    buf = (ctypes.c_char * len(data)).from_buffer_copy(data)
+    _check(lib().xc_h2d(ctypes.c_void_p(dst), buf,
+                        ctypes.c_long(len(data))), "xc_h2d")
+
+
+def xc_d2h(src: int, nbytes: int) -> bytes:
+    buf = (ctypes.c_char * nbytes)()
+    _check(lib().xc_d2h(buf, ctypes.c_void_p(src),
+                        ctypes.c_long(nbytes)), "xc_d2h")
+    return bytes(buf)
+
+
+def xc_ipc_handle(ptr: int) -> bytes:
+    h = (ctypes.c_char * XC_IPC_HANDLE_BYTES)()
+    _check(lib().xc_ipc_handle(ctypes.c_void_p(ptr), h), "xc_ipc_handle")
+    return bytes(h)
+
+
+def xc_ipc_open(handle: bytes) -> int:
+    assert len(handle) == XC_IPC_HANDLE_BYTES
+    buf = (ctypes.c_char * XC_IPC_HANDLE_BYTES).from_buffer_copy(handle)
+    p = ctypes.c_void_p(0)
+    _check(lib().xc_ipc_open(buf, ctypes.byref(p)), "xc_ipc_open")
+    return p.value
+
+
+def xc_ipc_close(ptr: int):
+    _check(lib().xc_ipc_close(ctypes.c_void_p(ptr)), "xc_ipc_close")
+
+
+def xgmi_coll(dst_ptr: int, src_ptr: int, mybase: int, rank: int,
+              world: int, nbytes: int, slot_bytes: int, mode: int,
+              nstripes: int, spin_limit: int = 5_000_000):
+    """One one-shot collective on the current torch stream (capturable)."""
+    _check(lib().launch_xgmi_coll(
+        ctypes.c_void_p(dst_ptr), ctypes.c_void_p(src_ptr),
+        ctypes.c_void_p(mybase), rank, world, ctypes.c_long(nbytes),
+        ctypes.c_long(slot_bytes), mode, nstripes,
+        ctypes.c_long(spin_limit), _stream()), "xgmi_coll")

@@ -42,7 +42,50 @@ def init_distributed(backend: Optional[str] = None):
     return dist.get_rank(), dist.get_world_size()
 
 
+# ----------------------------------------------------------------------
+# collective routing: small (decode-sized) payloads go through the
+# one-shot xGMI peer-memory kernel when available (graph-capturable,
+# ~1 hop); everything else through RCCL (prefill-sized payloads are
+# bandwidth-bound — the ring is right for those).
+# ----------------------------------------------------------------------
+
+_XGMI = None
+
+
+def init_xgmi(device, slot_bytes: int) -> bool:
+    """Try to bring up the one-shot communicator (validated on hardware
+    inside XgmiComm.create; all ranks agree on the outcome)."""
+    global _XGMI
+    import torch
+    import torch.distributed as dist
+
+    if _XGMI is not None:
+        return True
+    if not (dist.is_initialized() and dist.get_world_size() > 1
+            and torch.cuda.is_available()):
+        return False
+    from .xgmi import XgmiComm
+
+    _XGMI = XgmiComm.create(dist.get_rank(), dist.get_world_size(),
+                            device, slot_bytes)
+    return _XGMI is not None
+
+
+def xgmi_comm():
+    return _XGMI
+
+
+def shutdown_xgmi():
+    global _XGMI
+    if _XGMI is not None:
+        _XGMI.close()
+        _XGMI = None
+
+
 def all_reduce(t):
+    if _XGMI is not None and _XGMI.fits(t):
+        _XGMI.all_reduce(t)
+        return
     import torch.distributed as dist
 
     if dist.is_initialized() and dist.get_world_size() > 1:
@@ -50,6 +93,9 @@ def all_reduce(t):
 
 
 def all_gather_into(out, t):
+    if _XGMI is not None and _XGMI.fits(t) and out.is_contiguous():
+        _XGMI.all_gather_into(out, t)
+        return
     import torch.distributed as dist
 
     if dist.is_initialized() and dist.get_world_size() > 1:
