@@ -6,6 +6,7 @@
 typedef unsigned short u16;
 typedef short s8v __attribute__((ext_vector_type(8)));    // 16 B of bf16
 typedef float f4v __attribute__((ext_vector_type(4)));
+typedef uint32_t u4v_ __attribute__((ext_vector_type(4)));  // 16 B raw
 
 #define DEVINL __device__ __forceinline__
 

@@ -1483,6 +1483,199 @@ extern "C" hipError_t launch_gemm_bf16(const void* X, const void* W, void* Y,
 }
 
 // ====================================================================
+// fp8 row quantization: Q[N,K] = e4m3(X / s_n), s_n = absmax_n / 448.
+// One block per row.  Used (a) at load time to quantize weights ON
+// DEVICE (drops the round-1 host-quant ~90 s for 9B and the duplicate
+// bf16 copy — VERDICT r1 item 3), (b) per prefill GEMM to quantize the
+// activation rows so BOTH MFMA operands are fp8.
+// HW convert: v_cvt_pk_fp8_f32 (OCP e4m3fn on gfx950, saturating).
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_quant_fp8_rows(const u16* __restrict__ X, uint8_t* __restrict__ Q,
+                 float* __restrict__ S, int K) {
+  __shared__ float wmax[4];
+  const int row = blockIdx.x;
+  const u16* xr = X + (size_t)row * K;
+  uint8_t* qr = Q + (size_t)row * K;
+
+  float amax = 0.f;
+  for (int i = threadIdx.x * 8; i < K; i += 2048) {
+    s8v v = *(const s8v*)(xr + i);
+#pragma unroll
+    for (int j = 0; j < 8; j++) amax = fmaxf(amax, fabsf(b2f(((u16*)&v)[j])));
+  }
+#pragma unroll
+  for (int m = 1; m < 64; m <<= 1) amax = fmaxf(amax, __shfl_xor(amax, m));
+  if ((threadIdx.x & 63) == 0) wmax[threadIdx.x >> 6] = amax;
+  __syncthreads();
+  amax = fmaxf(fmaxf(wmax[0], wmax[1]), fmaxf(wmax[2], wmax[3]));
+  const float s = fmaxf(amax, 1e-8f) / 448.0f;
+  const float rs = 1.0f / s;
+  if (threadIdx.x == 0) S[row] = s;
+
+  for (int i = threadIdx.x * 8; i < K; i += 2048) {
+    s8v v = *(const s8v*)(xr + i);
+    float f[8];
+#pragma unroll
+    for (int j = 0; j < 8; j++) f[j] = b2f(((u16*)&v)[j]) * rs;
+    uint32_t lo = 0, hi = 0;
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(f[0], f[1], lo, false);
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(f[2], f[3], lo, true);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(f[4], f[5], hi, false);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(f[6], f[7], hi, true);
+    *(uint32_t*)(qr + i) = lo;
+    *(uint32_t*)(qr + i + 4) = hi;
+  }
+}
+
+extern "C" hipError_t launch_quant_fp8(const void* X, void* Q, void* S,
+                                       int M, int K, hipStream_t stream) {
+  if (K % 8 != 0) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(k_quant_fp8_rows, dim3(M), dim3(256), 0, stream,
+                     (const u16*)X, (uint8_t*)Q, (float*)S, K);
+  return hipGetLastError();
+}
+
+// ====================================================================
+// fp8 MFMA prefill GEMM: Y[M,N] = (sx_m * sy_n) * Xq[M,K] @ Wq[N,K]^T,
+// both operands OCP e4m3 with per-row scales (north star "CDNA4 fp8
+// MFMA").  v_mfma_f32_16x16x32_fp8_fp8; BM=BN=128, BK=64 (two K=32
+// slices), 4 waves (2x2), 64x64 per wave; LDS rows padded to 80 B so
+// 16-B stores stay aligned and b64 fragment reads spread banks.
+// Same split-K scheme as the bf16 GEMM (scales folded before the
+// atomicAdd so k_gemm_fin stays shared).
+// ====================================================================
+
+typedef long i64frag;
+
+extern "C" __global__ void __launch_bounds__(256)
+k_gemm_fp8(const uint8_t* __restrict__ X, const float* __restrict__ sx,
+           const uint8_t* __restrict__ W, const float* __restrict__ sw,
+           u16* __restrict__ Y, const u16* __restrict__ res,
+           float* __restrict__ accbuf, int M, int N, int K) {
+  constexpr int LDR = 80;  // padded LDS row stride (bytes)
+  __shared__ uint8_t As[2][128 * LDR];
+  __shared__ uint8_t Bs[2][128 * LDR];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wrow = wave >> 1, wcol = wave & 1;
+  const int bm = blockIdx.x * 128, bn = blockIdx.y * 128;
+
+  const int SK = gridDim.z;
+  const int kslices = (K / 64 + SK - 1) / SK;
+  const int k_lo = blockIdx.z * kslices * 64;
+  int k_hi = k_lo + kslices * 64;
+  if (k_hi > K) k_hi = K;
+  const int nt = (k_hi - k_lo) / 64;
+  if (nt <= 0) return;
+
+  f4v acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; i++)
+#pragma unroll
+    for (int j = 0; j < 4; j++) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // stage: 128 rows x 64 B per operand; 256 threads x 2 iters x 16 B
+  auto stage = [&](int buf, int kt) {
+    const int k0 = k_lo + kt * 64;
+    for (int i = tid; i < 128 * 4; i += 256) {
+      int r = i >> 2, seg = i & 3;
+      int gr = bm + r;
+      int grc = gr < M ? gr : (M > 0 ? M - 1 : 0);
+      *(u4v_*)(&As[buf][r * LDR + seg * 16]) =
+          *(const u4v_*)(X + (size_t)grc * K + k0 + seg * 16);
+      int gb = bn + r;
+      int gbc = gb < N ? gb : N - 1;
+      *(u4v_*)(&Bs[buf][r * LDR + seg * 16]) =
+          *(const u4v_*)(W + (size_t)gbc * K + k0 + seg * 16);
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+
+  const int fr = lane & 15, fk8 = (lane >> 4) * 8;  // frag row / k-offset
+  int cur = 0;
+  for (int t = 0; t < nt; t++) {
+    if (t + 1 < nt) stage(cur ^ 1, t + 1);  // fill other buf during MFMA
+#pragma unroll
+    for (int sl = 0; sl < 2; sl++) {  // two K=32 slices per 64-B tile
+      i64frag a[4], b[4];
+#pragma unroll
+      for (int i = 0; i < 4; i++) {
+        int ra = wrow * 64 + i * 16 + fr;
+        a[i] = *(const i64frag*)(&As[cur][ra * LDR + sl * 32 + fk8]);
+        int rb = wcol * 64 + i * 16 + fr;
+        b[i] = *(const i64frag*)(&Bs[cur][rb * LDR + sl * 32 + fk8]);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; i++)
+#pragma unroll
+        for (int j = 0; j < 4; j++)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();  // next iter writes cur / reads cur^1
+    cur ^= 1;
+  }
+
+  // epilogue: C col = lane&15, row = (lane>>4)*4 + reg (same map as bf16)
+  const int cc = lane & 15, cr = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        int row = bm + wrow * 64 + i * 16 + cr + r;
+        int col = bn + wcol * 64 + j * 16 + cc;
+        if (row < M && col < N) {
+          float v = acc[i][j][r] * sx[row] * sw[col];
+          if (SK > 1) {
+            atomicAdd(accbuf + (size_t)row * N + col, v);
+          } else {
+            if (res) v += b2f(res[(size_t)row * N + col]);
+            Y[(size_t)row * N + col] = f2b(v);
+          }
+        }
+      }
+    }
+  }
+}
+
+extern "C" hipError_t launch_gemm_fp8(const void* X, const void* sx,
+                                      const void* W, const void* sw, void* Y,
+                                      const void* res, void* accbuf, int M,
+                                      int N, int K, hipStream_t stream) {
+  if (K % 64 != 0) return hipErrorInvalidValue;
+  int gm = (M + 127) / 128, gn = (N + 127) / 128;
+  int sk = 1;
+  if (accbuf && gm * gn < 160) {
+    while (sk < 8 && gm * gn * sk * 2 <= 512 && (K / 64) % (sk * 2) == 0)
+      sk *= 2;
+  }
+  if (sk > 1) {
+    long total = (long)M * N;
+    hipLaunchKernelGGL(k_zero_f32, dim3((uint32_t)((total / 4 + 255) / 256)),
+                       dim3(256), 0, stream, (float*)accbuf, total);
+  }
+  dim3 grid(gm, gn, sk);
+  hipLaunchKernelGGL(k_gemm_fp8, grid, dim3(256), 0, stream,
+                     (const uint8_t*)X, (const float*)sx, (const uint8_t*)W,
+                     (const float*)sw, (u16*)Y, (const u16*)res,
+                     (float*)accbuf, M, N, K);
+  if (sk > 1) {
+    long total = (long)M * N;
+    hipLaunchKernelGGL(k_gemm_fin, dim3((uint32_t)((total + 255) / 256)),
+                       dim3(256), 0, stream, (const float*)accbuf,
+                       (const u16*)res, (u16*)Y, total);
+  }
+  return hipGetLastError();
+}
+
+// ====================================================================
 // Weight prefetcher (side-stream): stream a tensor through L2/MALL with
 // plain (retaining) loads so the 256 MB Infinity Cache holds the NEXT
 // layer's weights before its compute kernel issues.  The sink write

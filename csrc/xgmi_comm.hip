@@ -51,7 +51,6 @@
 #define XC_OFF_DATA 4096
 
 typedef unsigned long long u64t;
-typedef uint32_t u4v_ __attribute__((ext_vector_type(4)));
 
 // MODE: 0 = all-reduce bf16 (fp32 accum), 1 = all-reduce fp32,
 //       2 = all-gather (raw bytes; dst holds world*nbytes)

@@ -113,15 +113,25 @@ class GPUModel:
             return torch.from_numpy(g).to(dev)
 
         def quant8(a: np.ndarray):
-            """Per-output-row e4m3fn quantization (absmax/448)."""
-            t = torch.from_numpy(np.ascontiguousarray(a)).float()
-            s = t.abs().amax(dim=1).clamp_min(1e-8) / 448.0
-            q = (t / s[:, None]).to(torch.float8_e4m3fn).view(torch.uint8)
-            return q.to(dev), s.to(dev)
+            """Per-output-row e4m3fn quantization ON DEVICE (k_quant_
+            fp8_rows): one bf16 upload, quantize, drop the bf16 copy —
+            fp8 models hold a single weight copy in HBM and load in
+            seconds (round-1 host quant took ~90 s for 9B)."""
+            t = bf16(a)
+            N, K = t.shape
+            q = torch.empty(N, K, dtype=torch.uint8, device=dev)
+            s = torch.empty(N, dtype=torch.float32, device=dev)
+            ho.quant_fp8(t, q, s)
+            torch.cuda.synchronize()  # t freed right after
+            return q, s
 
         self.embed = bf16(w["model.embed_tokens.weight"])
         lm_w = w.get("lm_head.weight", w["model.embed_tokens.weight"])
-        if tp > 1:
+        if self.fp8:
+            lmw = lm_w if tp == 1 else tpu.shard_rows(lm_w, r, tp)
+            self.lm_head_q, self.lm_head_s = quant8(lmw)
+            self.lm_head = None  # single-copy: no bf16 duplicate
+        elif tp > 1:
             self.lm_head = bf16(tpu.shard_rows(lm_w, r, tp))
         elif lm_w is w.get("model.embed_tokens.weight", None) or \
                 cfg.tie_word_embeddings and "lm_head.weight" not in w:
@@ -138,48 +148,53 @@ class GPUModel:
             hrows = lambda x: tpu.shard_rows(x, r, tp) if tp > 1 else x
             hcols = lambda x: tpu.shard_cols(x, r, tp) if tp > 1 else x
             import numpy as _np
-            wqkv = bf16(_np.concatenate([
+            qkv_np = _np.concatenate([
                 hrows(w[f"{a}.q_proj.weight"]),
                 hrows(w[f"{a}.k_proj.weight"]),
-                hrows(w[f"{a}.v_proj.weight"])], axis=0))
-            wgu = bf16(_np.concatenate([
+                hrows(w[f"{a}.v_proj.weight"])], axis=0)
+            gu_np = _np.concatenate([
                 hrows(w[f"{p}.mlp.gate_proj.weight"]),
-                hrows(w[f"{p}.mlp.up_proj.weight"])], axis=0))
+                hrows(w[f"{p}.mlp.up_proj.weight"])], axis=0)
+            o_np = hcols(w[f"{a}.o_proj.weight"])
+            down_np = hcols(w[f"{p}.mlp.down_proj.weight"])
             nq = self.nh_l * hd
             nkv = self.kvh_l * hd
             I = self.inter_l
             lw = {
-                # fused tensors (decode path) + contiguous row views
-                "wqkv": wqkv,
-                "wq": wqkv[:nq], "wk": wqkv[nq:nq + nkv],
-                "wv": wqkv[nq + nkv:],
-                "wgu": wgu, "wgate": wgu[:I], "wup": wgu[I:],
-                "wo": bf16(hcols(w[f"{a}.o_proj.weight"])),
-                "wdown": bf16(hcols(w[f"{p}.mlp.down_proj.weight"])),
                 "g_in": gamma(w[f"{p}.input_layernorm.weight"]),
                 "g_post": gamma(w[f"{p}.post_attention_layernorm.weight"]),
             }
+            if self.fp8:
+                # SINGLE fp8 copy: decode GEMV and prefill fp8-MFMA GEMM
+                # both read these (no bf16 duplicate in HBM)
+                for name, arr in [("wqkv", qkv_np), ("wgu", gu_np),
+                                  ("wo", o_np), ("wdown", down_np)]:
+                    lw[name + "_q"], lw[name + "_s"] = quant8(arr)
+                lw.update({
+                    "wq_q": lw["wqkv_q"][:nq], "wq_s": lw["wqkv_s"][:nq],
+                    "wk_q": lw["wqkv_q"][nq:nq + nkv],
+                    "wk_s": lw["wqkv_s"][nq:nq + nkv],
+                    "wv_q": lw["wqkv_q"][nq + nkv:],
+                    "wv_s": lw["wqkv_s"][nq + nkv:],
+                    "wgate_q": lw["wgu_q"][:I], "wgate_s": lw["wgu_s"][:I],
+                    "wup_q": lw["wgu_q"][I:], "wup_s": lw["wgu_s"][I:],
+                })
+            else:
+                wqkv = bf16(qkv_np)
+                wgu = bf16(gu_np)
+                lw.update({
+                    # fused tensors (decode path) + contiguous row views
+                    "wqkv": wqkv,
+                    "wq": wqkv[:nq], "wk": wqkv[nq:nq + nkv],
+                    "wv": wqkv[nq + nkv:],
+                    "wgu": wgu, "wgate": wgu[:I], "wup": wgu[I:],
+                    "wo": bf16(o_np),
+                    "wdown": bf16(down_np),
+                })
             if gemma:
                 lw["g_preffn"] = gamma(w[f"{p}.pre_feedforward_layernorm.weight"])
                 lw["g_postffn"] = gamma(w[f"{p}.post_feedforward_layernorm.weight"])
-            if self.fp8:
-                # decode streams fp8 weights; prefill keeps the bf16 copy
-                for name, arr in [
-                    ("wqkv", _np.concatenate([
-                        hrows(w[f"{a}.q_proj.weight"]),
-                        hrows(w[f"{a}.k_proj.weight"]),
-                        hrows(w[f"{a}.v_proj.weight"])], axis=0)),
-                    ("wgu", _np.concatenate([
-                        hrows(w[f"{p}.mlp.gate_proj.weight"]),
-                        hrows(w[f"{p}.mlp.up_proj.weight"])], axis=0)),
-                    ("wo", hcols(w[f"{a}.o_proj.weight"])),
-                    ("wdown", hcols(w[f"{p}.mlp.down_proj.weight"])),
-                ]:
-                    lw[name + "_q"], lw[name + "_s"] = quant8(arr)
             self.layers.append(lw)
-        if self.fp8:
-            lmw = lm_w if tp == 1 else tpu.shard_rows(lm_w, r, tp)
-            self.lm_head_q, self.lm_head_s = quant8(lmw)
 
     def _alloc_state(self, prefill_chunk: int):
         cfg, dev = self.config, self.device
@@ -209,6 +224,11 @@ class GPUModel:
         self.b_up = torch.zeros(PC, I, **bf)
         self.b_gemm_acc = torch.zeros(PC * max(I, H), dtype=torch.float32,
                                       device=dev)
+        if self.fp8:
+            # activation-quant scratch for the fp8 MFMA prefill GEMMs
+            kmax = max(H, I, self.nh_l * hd)
+            self.b_xq = torch.zeros(PC * kmax, dtype=torch.uint8, device=dev)
+            self.b_sx = torch.zeros(PC, dtype=torch.float32, device=dev)
         self.b_logits_l = torch.zeros(self.vocab_l, dtype=torch.float32,
                                       device=dev)
         self.b_logits = (self.b_logits_l if not self.tp_branch else
@@ -263,11 +283,26 @@ class GPUModel:
     # ------------------------------------------------------------------
     # layer stack over rows [0, M) of the scratch buffers
     # ------------------------------------------------------------------
-    def _linear(self, W, x, y, res=None, M: int = 1, softcap: float = 0.0):
+    def _linear(self, lw, name, x, y, res=None, M: int = 1,
+                softcap: float = 0.0):
+        """Projection by weight NAME so dtype routing stays in one
+        place: bf16 GEMV/MFMA-GEMM, or (fp8) GEMV / fp8-MFMA GEMM with
+        on-the-fly activation row quantization."""
         if M == 1:
-            ho.gemv(W, x, y, res=res, softcap=softcap)
+            if self.fp8:
+                ho.gemv_fp8(lw[name + "_q"], lw[name + "_s"], x, y,
+                            res=res, softcap=softcap)
+            else:
+                ho.gemv(lw[name], x, y, res=res, softcap=softcap)
+        elif self.fp8:
+            K = x.shape[-1]
+            ho.quant_fp8(x[:M], self.b_xq, self.b_sx)
+            ho.gemm_fp8(self.b_xq, self.b_sx, lw[name + "_q"],
+                        lw[name + "_s"], y[:M], M, K,
+                        res=res[:M] if res is not None else None,
+                        accbuf=self.b_gemm_acc)
         else:
-            ho.gemm(x[:M], W, y, res=res[:M] if res is not None else None,
+            ho.gemm(x[:M], lw[name], y, res=res[:M] if res is not None else None,
                     accbuf=self.b_gemm_acc)
 
     def _layers_forward(self, M: int):
@@ -277,9 +312,9 @@ class GPUModel:
         for i, lw in enumerate(self.layers):
             window = cfg.sliding_window if cfg.is_sliding(i) else 0
             ho.rmsnorm(h[:M], lw["g_in"], xn[:M], eps=eps)
-            self._linear(lw["wq"], xn, self.b_q, M=M)
-            self._linear(lw["wk"], xn, self.b_k, M=M)
-            self._linear(lw["wv"], xn, self.b_v, M=M)
+            self._linear(lw, "wq", xn, self.b_q, M=M)
+            self._linear(lw, "wk", xn, self.b_k, M=M)
+            self._linear(lw, "wv", xn, self.b_v, M=M)
             ho.rope_cache(self.b_q, self.b_k, self.b_v, self.k_cache[i],
                           self.v_cache[i], self.cos_t, self.sin_t,
                           self.len_buf, M, self.nh_l, self.kvh_l, self.hd)
@@ -295,40 +330,44 @@ class GPUModel:
                         self.hd, self.scale, softcap=self.attn_softcap,
                         window=window or 0)
             if self.gemma:
-                self._linear(lw["wo"], self.b_att, t1, M=M)
+                self._linear(lw, "wo", self.b_att, t1, M=M)
                 tpu.all_reduce(t1[:M])
                 ho.rmsnorm(t1[:M], lw["g_post"], h[:M], res=h[:M], eps=eps)
                 ho.rmsnorm(h[:M], lw["g_preffn"], xn[:M], eps=eps)
             else:
                 if self.tp_branch:
-                    self._linear(lw["wo"], self.b_att, t1, M=M)
+                    self._linear(lw, "wo", self.b_att, t1, M=M)
                     tpu.all_reduce(t1[:M])
                     ho.addinto(h[:M], t1[:M])
                 else:
-                    self._linear(lw["wo"], self.b_att, h, res=h, M=M)
+                    self._linear(lw, "wo", self.b_att, h, res=h, M=M)
                 ho.rmsnorm(h[:M], lw["g_post"], xn[:M], eps=eps)
-            self._linear(lw["wgate"], xn, self.b_gate, M=M)
-            self._linear(lw["wup"], xn, self.b_up, M=M)
+            self._linear(lw, "wgate", xn, self.b_gate, M=M)
+            self._linear(lw, "wup", xn, self.b_up, M=M)
             ho.glu(self.b_gate[:M], self.b_up[:M], self.b_gate[:M], self.act)
             if self.gemma:
-                self._linear(lw["wdown"], self.b_gate, t1, M=M)
+                self._linear(lw, "wdown", self.b_gate, t1, M=M)
                 tpu.all_reduce(t1[:M])
                 ho.rmsnorm(t1[:M], lw["g_postffn"], h[:M], res=h[:M], eps=eps)
             else:
                 if self.tp_branch:
-                    self._linear(lw["wdown"], self.b_gate, t1, M=M)
+                    self._linear(lw, "wdown", self.b_gate, t1, M=M)
                     tpu.all_reduce(t1[:M])
                     ho.addinto(h[:M], t1[:M])
                 else:
-                    self._linear(lw["wdown"], self.b_gate, h, res=h, M=M)
+                    self._linear(lw, "wdown", self.b_gate, h, res=h, M=M)
 
     def _lm_head_last(self, M: int):
         """Final norm + lm_head on the last row -> self.b_logits (f32, V)."""
         hrow = self.b_h[M - 1] if M > 1 else self.b_h[0]
         ho.rmsnorm(hrow, self.g_final, self.b_xn[0],
                    eps=self.config.rms_norm_eps)
-        ho.gemv(self.lm_head, self.b_xn[0], self.b_logits_l,
-                softcap=self.final_softcap)
+        if self.fp8:
+            ho.gemv_fp8(self.lm_head_q, self.lm_head_s, self.b_xn[0],
+                        self.b_logits_l, softcap=self.final_softcap)
+        else:
+            ho.gemv(self.lm_head, self.b_xn[0], self.b_logits_l,
+                    softcap=self.final_softcap)
         if self.tp_branch:
             tpu.all_gather_into(self.b_logits, self.b_logits_l)
 
@@ -353,8 +392,14 @@ class GPUModel:
             self._layers_forward(M)
             ho.rmsnorm(self.b_h[:M], self.g_final, self.b_xn[:M],
                        eps=self.config.rms_norm_eps)
-            ho.gemm(self.b_xn[:M], self.lm_head, logits_buf[:M],
-                    accbuf=self.b_gemm_acc)
+            if self.fp8:
+                ho.quant_fp8(self.b_xn[:M], self.b_xq, self.b_sx)
+                ho.gemm_fp8(self.b_xq, self.b_sx, self.lm_head_q,
+                            self.lm_head_s, logits_buf[:M], M, self.H,
+                            accbuf=self.b_gemm_acc)
+            else:
+                ho.gemm(self.b_xn[:M], self.lm_head, logits_buf[:M],
+                        accbuf=self.b_gemm_acc)
             if self.final_softcap:
                 ho.softcap(logits_buf[:M], self.final_softcap)
             if self.world > 1:

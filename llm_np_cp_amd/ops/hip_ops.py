@@ -63,6 +63,10 @@ _SIGS = {
                        ctypes.c_void_p],
     "launch_i32_set": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
     "launch_i32_add": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
+    "launch_quant_fp8": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 +
+                        [ctypes.c_void_p],
+    "launch_gemm_fp8": [ctypes.c_void_p] * 7 + [ctypes.c_int] * 3 +
+                       [ctypes.c_void_p],
     # one-shot xGMI collectives (csrc/xgmi_comm.hip)
     "launch_xgmi_coll": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                          ctypes.c_int, ctypes.c_int, ctypes.c_long,
@@ -257,6 +261,32 @@ def gemm(X: torch.Tensor, W: torch.Tensor, Y: torch.Tensor,
         accbuf = None
     _check(lib().launch_gemm_bf16(_ptr(X), _ptr(W), _ptr(Y), _ptr(res),
                                   _ptr(accbuf), M, N, K, _stream()), "gemm")
+
+
+def quant_fp8(x: torch.Tensor, q: torch.Tensor, s: torch.Tensor):
+    """Per-row e4m3fn quantization on device: q = fp8(x / s_row),
+    s_row = absmax/448.  x: (M,K) bf16 contiguous; q: >= M*K uint8;
+    s: >= M fp32.  Used for weights at load and activations per GEMM."""
+    M = 1 if x.dim() == 1 else x.shape[0]
+    K = x.shape[-1]
+    assert q.numel() >= M * K and s.numel() >= M
+    _check(lib().launch_quant_fp8(_ptr(x), _ptr(q), _ptr(s), M, K,
+                                  _stream()), "quant_fp8")
+
+
+def gemm_fp8(xq: torch.Tensor, sx: torch.Tensor, Wq: torch.Tensor,
+             sw: torch.Tensor, y: torch.Tensor, M: int, K: int,
+             res: torch.Tensor | None = None,
+             accbuf: torch.Tensor | None = None):
+    """Y[M,N] = (sx_m * sw_n) * Xq[M,K] @ Wq[N,K]^T (+res): fp8 MFMA
+    prefill path (both operands e4m3 with per-row scales)."""
+    N = Wq.shape[0]
+    assert Wq.shape[-1] == K and K % 64 == 0
+    if accbuf is not None and accbuf.numel() < M * N:
+        accbuf = None
+    _check(lib().launch_gemm_fp8(
+        _ptr(xq), _ptr(sx), _ptr(Wq), _ptr(sw), _ptr(y), _ptr(res),
+        _ptr(accbuf), M, N, K, _stream()), "gemm_fp8")
 
 
 def prefetch(t: torch.Tensor, sink: torch.Tensor):

@@ -5,6 +5,13 @@ import pytest
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
+# The simulated-multi-rank comm tests run 8 spin-waiting kernels on 8
+# streams of ONE process; ROCm's default of 4 hardware queues would
+# serialize them into a deadlock.  Must be set before HIP runtime init
+# (conftest imports before any CUDA call).  Real TP runs one process
+# per GPU and needs no such setting.
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "10")
+
 
 def pytest_configure(config):
     config.addinivalue_line(

@@ -29,6 +29,20 @@ def _ho():
     return ho
 
 
+# ONE shared stream pool for every simulated-rank test: spinning kernels
+# must be co-resident, and HIP maps streams round-robin onto a bounded
+# set of hardware queues (GPU_MAX_HW_QUEUES, raised in conftest.py) — a
+# fresh 8 streams per test would eventually collide two spinners onto
+# one queue and serialize them into a timeout.
+_STREAMS = []
+
+
+def _streams(world):
+    while len(_STREAMS) < world:
+        _STREAMS.append(torch.cuda.Stream(device=torch.device("cuda:0")))
+    return _STREAMS[:world]
+
+
 def _sim_world(ho, world, slot_bytes):
     total = ho.XC_OFF_DATA + 2 * 8 * slot_bytes
     bases = [ho.xc_alloc(total) for _ in range(world)]
@@ -57,7 +71,7 @@ def test_oneshot_allreduce_bf16_simulated(world, n, nstripes):
                 for _ in range(world)]
         expect = sum(s.float() for s in srcs)
         bufs = [s.clone() for s in srcs]
-        streams = [torch.cuda.Stream(device=dev) for _ in range(world)]
+        streams = _streams(world)
         for r in range(world):
             with torch.cuda.stream(streams[r]):
                 ho.xgmi_coll(bufs[r].data_ptr(), bufs[r].data_ptr(),
@@ -86,7 +100,7 @@ def test_oneshot_f32_and_gather_simulated():
         expect = sum(srcs)
         outs = [s.clone() for s in srcs]
         gouts = [torch.zeros(world * n, device=dev) for _ in range(world)]
-        streams = [torch.cuda.Stream(device=dev) for _ in range(world)]
+        streams = _streams(world)
         # f32 all-reduce then gather back-to-back on each stream:
         # exercises epoch parity alternation within one submission
         for r in range(world):
@@ -126,7 +140,7 @@ def test_oneshot_graph_replay_simulated():
                 for _ in range(world)]
         bufs = [torch.empty_like(s) for s in srcs]
         expect = sum(s.float() for s in srcs)
-        streams = [torch.cuda.Stream(device=dev) for _ in range(world)]
+        streams = _streams(world)
 
         def _coll(r):
             ho.xgmi_coll(bufs[r].data_ptr(), bufs[r].data_ptr(), bases[r],
@@ -244,7 +258,7 @@ def _tp2_worker(rank, world, port, q):
         from llm_np_cp_amd.models.engine import GPUModel
         from llm_np_cp_amd.parallel import tp
 
-        cfg = L.preset_config("tiny-llama")
+        cfg = L.preset_config("tiny-llama-tp")
         w = random_weights(cfg, seed=0)
         model = GPUModel(cfg, w, max_seq=128, device="cuda:0")
         assert tp.xgmi_comm() is not None, \
@@ -270,7 +284,7 @@ def test_tp2_engine_graph_decode_one_gpu():
     from llm_np_cp_amd.models.engine import GPUModel
 
     # TP=1 reference in the parent (same weights)
-    cfg = L.preset_config("tiny-llama")
+    cfg = L.preset_config("tiny-llama-tp")
     w = random_weights(cfg, seed=0)
     ref = GPUModel(cfg, w, max_seq=128, device="cuda:0")
     prompt = np.arange(1, 9, dtype=np.int32)

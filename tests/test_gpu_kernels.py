@@ -463,3 +463,76 @@ def test_attn_prefill_mfma_vs_reference(hd, window, cap):
             p = torch.softmax(sc, -1)
             ref[m, h] = p @ Vf[kv, lo:qpos + 1]
     assert_close(out.view(M, nh, hd), ref, rtol=4e-2, atol=4e-2)
+
+
+def test_quant_fp8_rows_matches_torch():
+    """Device per-row e4m3 quant vs torch's float8_e4m3fn cast."""
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    M, K = 64, 2048
+    x = randn_bf16(M, K, seed=200, scale=0.3)
+    q = torch.empty(M * K, dtype=torch.uint8, device=dev())
+    s = torch.empty(M, dtype=torch.float32, device=dev())
+    ho.quant_fp8(x, q, s)
+    torch.cuda.synchronize()
+    xf = x.float().cpu()
+    s_ref = xf.abs().amax(dim=1).clamp_min(1e-8) / 448.0
+    assert_close(s.cpu(), s_ref, rtol=1e-3, atol=1e-8)
+    deq = (q.view(M, K).cpu().view(torch.float8_e4m3fn).float()
+           * s.cpu()[:, None])
+    # both quantizers are RNE e4m3: dequant must match within one LSB
+    # of the fp8 grid (relative 2^-3 at the value's scale)
+    err = (deq - xf).abs()
+    tol = xf.abs() * 0.0705 + s_ref[:, None] * 0.002
+    assert bool((err <= tol).all()), float((err - tol).max())
+
+
+@pytest.mark.parametrize("M,N,K", [(64, 512, 256), (200, 384, 2048),
+                                   (33, 1000, 128), (256, 2048, 2048)])
+def test_gemm_fp8_vs_f32_reference(M, N, K):
+    """fp8 MFMA GEMM (both operands quantized on device) vs the fp32
+    product of the SAME dequantized operands -> tolerance covers only
+    fp32-accum rounding, not quantization."""
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    X = randn_bf16(M, K, seed=300, scale=0.2)
+    W = randn_bf16(N, K, seed=301, scale=0.05)
+    xq = torch.empty(M * K, dtype=torch.uint8, device=dev())
+    sx = torch.empty(M, dtype=torch.float32, device=dev())
+    wq = torch.empty(N, K, dtype=torch.uint8, device=dev())
+    sw = torch.empty(N, dtype=torch.float32, device=dev())
+    ho.quant_fp8(X, xq, sx)
+    ho.quant_fp8(W, wq, sw)
+    y = torch.empty(M, N, dtype=torch.bfloat16, device=dev())
+    acc = torch.zeros(M * N, dtype=torch.float32, device=dev())
+    ho.gemm_fp8(xq, sx, wq, sw, y, M, K, accbuf=acc)
+    torch.cuda.synchronize()
+    Xd = (xq[:M * K].view(M, K).view(torch.float8_e4m3fn).float()
+          * sx[:, None])
+    Wd = wq.view(torch.float8_e4m3fn).float() * sw[:, None]
+    ref = Xd @ Wd.T
+    assert_close(y, ref, rtol=2e-2, atol=2e-2)
+
+
+def test_gemm_fp8_residual():
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    M, N, K = 32, 256, 192
+    X = randn_bf16(M, K, seed=310, scale=0.2)
+    W = randn_bf16(N, K, seed=311, scale=0.05)
+    res = randn_bf16(M, N, seed=312)
+    xq = torch.empty(M * K, dtype=torch.uint8, device=dev())
+    sx = torch.empty(M, dtype=torch.float32, device=dev())
+    wq = torch.empty(N, K, dtype=torch.uint8, device=dev())
+    sw = torch.empty(N, dtype=torch.float32, device=dev())
+    ho.quant_fp8(X, xq, sx)
+    ho.quant_fp8(W, wq, sw)
+    y = torch.empty(M, N, dtype=torch.bfloat16, device=dev())
+    acc = torch.zeros(M * N, dtype=torch.float32, device=dev())
+    ho.gemm_fp8(xq, sx, wq, sw, y, M, K, res=res, accbuf=acc)
+    torch.cuda.synchronize()
+    Xd = (xq[:M * K].view(M, K).view(torch.float8_e4m3fn).float()
+          * sx[:, None])
+    Wd = wq.view(torch.float8_e4m3fn).float() * sw[:, None]
+    ref = Xd @ Wd.T + res.float()
+    assert_close(y, ref, rtol=2e-2, atol=2e-2)
