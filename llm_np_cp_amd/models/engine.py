@@ -82,6 +82,14 @@ class GPUModel:
         if self.world > 1:
             slot_bytes = max(self.H * 2, self.vocab_l * 4, 1 << 16)
             tpu.init_xgmi(self.device, slot_bytes)
+        # prefill-sized all-reduces: RCCL async on its own stream,
+        # overlapped with the next row-chunk's GEMM (north star: comm
+        # overlap on a second HIP stream).  Needs the nccl(=RCCL)
+        # backend; gloo (CPU CI) takes the inline path.
+        import torch.distributed as dist
+        self._async_ar = bool(
+            self.world > 1 and dist.is_initialized()
+            and "nccl" in str(dist.get_backend()).lower())
 
         self._upload_weights(weights)
         self._alloc_state(prefill_chunk)
@@ -305,6 +313,35 @@ class GPUModel:
             ho.gemm(x[:M], lw[name], y, res=res[:M] if res is not None else None,
                     accbuf=self.b_gemm_acc)
 
+    def _rowpar_ar(self, lw, name, x, y, M: int):
+        """Row-parallel projection followed by all-reduce over TP ranks.
+        For prefill-sized M the rows are split in two: the all-reduce of
+        chunk A runs on RCCL's comm stream while chunk B's GEMM occupies
+        the compute stream (SURVEY §2.3 'overlap comm on 2nd stream' —
+        the reference has no distributed code at all)."""
+        if not (self._async_ar and M >= 512):
+            self._linear(lw, name, x, y, M=M)
+            tpu.all_reduce(y[:M])
+            return
+        import torch.distributed as dist
+        K = x.shape[-1]
+        if self.fp8:
+            ho.quant_fp8(x[:M], self.b_xq, self.b_sx)
+        works = []
+        M2 = M // 2
+        for a, b in ((0, M2), (M2, M)):
+            if self.fp8:
+                ho.gemm_fp8(self.b_xq[a * K:], self.b_sx[a:],
+                            lw[name + "_q"], lw[name + "_s"], y[a:b],
+                            b - a, K, accbuf=self.b_gemm_acc)
+            else:
+                ho.gemm(x[a:b], lw[name], y[a:b], accbuf=self.b_gemm_acc)
+            # async -> lands on the RCCL stream after the chunk's GEMM;
+            # the compute stream proceeds to the next chunk immediately
+            works.append(dist.all_reduce(y[a:b], async_op=True))
+        for wk in works:
+            wk.wait()  # compute stream waits on the comm stream
+
     def _layers_forward(self, M: int):
         cfg = self.config
         eps = cfg.rms_norm_eps
@@ -330,14 +367,12 @@ class GPUModel:
                         self.hd, self.scale, softcap=self.attn_softcap,
                         window=window or 0)
             if self.gemma:
-                self._linear(lw, "wo", self.b_att, t1, M=M)
-                tpu.all_reduce(t1[:M])
+                self._rowpar_ar(lw, "wo", self.b_att, t1, M)
                 ho.rmsnorm(t1[:M], lw["g_post"], h[:M], res=h[:M], eps=eps)
                 ho.rmsnorm(h[:M], lw["g_preffn"], xn[:M], eps=eps)
             else:
                 if self.tp_branch:
-                    self._linear(lw, "wo", self.b_att, t1, M=M)
-                    tpu.all_reduce(t1[:M])
+                    self._rowpar_ar(lw, "wo", self.b_att, t1, M)
                     ho.addinto(h[:M], t1[:M])
                 else:
                     self._linear(lw, "wo", self.b_att, h, res=h, M=M)
@@ -346,13 +381,11 @@ class GPUModel:
             self._linear(lw, "wup", xn, self.b_up, M=M)
             ho.glu(self.b_gate[:M], self.b_up[:M], self.b_gate[:M], self.act)
             if self.gemma:
-                self._linear(lw, "wdown", self.b_gate, t1, M=M)
-                tpu.all_reduce(t1[:M])
+                self._rowpar_ar(lw, "wdown", self.b_gate, t1, M)
                 ho.rmsnorm(t1[:M], lw["g_postffn"], h[:M], res=h[:M], eps=eps)
             else:
                 if self.tp_branch:
-                    self._linear(lw, "wdown", self.b_gate, t1, M=M)
-                    tpu.all_reduce(t1[:M])
+                    self._rowpar_ar(lw, "wdown", self.b_gate, t1, M)
                     ho.addinto(h[:M], t1[:M])
                 else:
                     self._linear(lw, "wdown", self.b_gate, h, res=h, M=M)

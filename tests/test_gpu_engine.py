@@ -14,6 +14,27 @@ def _build():
     ensure_built()
 
 
+def softmax_np(x, axis=-1):
+    x = np.asarray(x, dtype=np.float64)
+    m = x.max(axis=axis, keepdims=True)
+    e = np.exp(x - m)
+    return e / e.sum(axis=axis, keepdims=True)
+
+
+def kl_bits(p_logits, q_logits, axis=-1):
+    """KL(p || q) in bits per position — catches kernel regressions that
+    argmax-only checks miss (VERDICT r1 'tighten verification')."""
+    p = softmax_np(p_logits, axis)
+    q = softmax_np(q_logits, axis)
+    return (p * (np.log2(p + 1e-30) - np.log2(q + 1e-30))).sum(axis)
+
+
+def topk_overlap(a_logits, b_logits, k=8):
+    a = set(np.argsort(a_logits)[-k:].tolist())
+    b = set(np.argsort(b_logits)[-k:].tolist())
+    return len(a & b) / k
+
+
 def make_pair(preset, seed=0, max_seq=256):
     import llm_np_cp_amd as L
     from llm_np_cp_amd.io.loader import random_weights
@@ -43,6 +64,10 @@ def test_prefill_logits_match_oracle(preset):
     # bf16 forward vs fp32 oracle: compare softmax-relevant structure
     np.testing.assert_allclose(got, ref_logits[-1], rtol=0.15, atol=0.15)
     assert np.argmax(got) == np.argmax(ref_logits[-1])
+    # distributional bounds (sharper than argmax): next-token KL and
+    # top-k set agreement vs the fp32 oracle
+    assert kl_bits(ref_logits[-1], got) < 0.02
+    assert topk_overlap(got, ref_logits[-1], k=8) >= 0.875
 
 
 @pytest.mark.parametrize("preset", ["tiny-llama", "tiny-gemma2"])
@@ -137,8 +162,10 @@ def test_fp8_engine_decode_close_to_bf16():
 
     _, logits_bf = bf.prefill(prompt)
     _, logits_f8 = f8.prefill(prompt)
-    # prefill runs bf16 weights in both engines -> identical path
-    np.testing.assert_allclose(logits_f8, logits_bf, rtol=1e-3, atol=1e-3)
+    # single-copy fp8: prefill now runs fp8 MFMA (both operands
+    # quantized per-row) -> differences bounded by e4m3 grid noise
+    np.testing.assert_allclose(logits_f8, logits_bf, rtol=5e-2, atol=8e-2)
+    assert np.argmax(logits_f8) == np.argmax(logits_bf)
 
     ids_bf = bf.decode(6, greedy=True, use_graph=False)
     f8.prefill(prompt)
@@ -219,6 +246,10 @@ def test_forward_full_matches_oracle_all_positions():
     assert got.shape == want.shape
     # bf16 gemm logits vs fp32 oracle: argmax agreement per position
     assert (got.argmax(1) == want.argmax(1)).mean() > 0.9
+    # per-position KL bound (mean over positions) — a ±1-ulp-class
+    # kernel regression shows up here long before argmax flips
+    kl = kl_bits(want, got, axis=1)
+    assert kl.mean() < 0.02 and kl.max() < 0.08, (kl.mean(), kl.max())
 
 
 def test_forward_full_gemma_softcap():

@@ -536,3 +536,45 @@ def test_gemm_fp8_residual():
     Wd = wq.view(torch.float8_e4m3fn).float() * sw[:, None]
     ref = Xd @ Wd.T + res.float()
     assert_close(y, ref, rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.parametrize("temperature", [1.0, 0.7])
+def test_sampler_min_p_distribution_chi_square(temperature):
+    """The device Gumbel-argmax sampler must MATCH the min-p categorical
+    distribution (not just its support): chi-square over ~4000 draws vs
+    the analytic probabilities, including device-side temperature."""
+    from scipy.stats import chi2
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    V = 32
+    g = torch.Generator().manual_seed(77)
+    logits = (2.0 * torch.randn(V, generator=g)).to(dev(), torch.float32)
+    lf = logits.cpu().numpy().astype(np.float64)
+    min_p = 0.1
+    # analytic min-p distribution at temperature T
+    z = lf / temperature
+    p = np.exp(z - z.max()); p /= p.sum()
+    keep = p >= min_p * p.max()
+    q = np.where(keep, p, 0.0); q /= q.sum()
+
+    ctr = torch.zeros(1, dtype=torch.int64, device=dev())
+    gmax = torch.zeros(1, dtype=torch.int64, device=dev())
+    pick = torch.zeros(1, dtype=torch.int64, device=dev())
+    nt = torch.zeros(1, dtype=torch.int32, device=dev())
+    ring = torch.zeros(8192, dtype=torch.int32, device=dev())
+    nout = torch.zeros(1, dtype=torch.int32, device=dev())
+    ln = torch.zeros(1, dtype=torch.int32, device=dev())
+    n = 4000
+    for i in range(n):
+        ho.sample(logits, min_p, False, 1234, ctr, gmax, pick, nt, ring,
+                  nout, ln, bump_len=False, temperature=temperature)
+    torch.cuda.synchronize()
+    draws = ring[:n].cpu().numpy()
+    counts = np.bincount(draws, minlength=V).astype(np.float64)
+    assert counts[~keep].sum() == 0, "sampled outside the min-p keep-set"
+    exp = q * n
+    mask = exp > 0
+    stat = (((counts - exp) ** 2) / np.maximum(exp, 1e-9))[mask].sum()
+    df = int(mask.sum()) - 1
+    thresh = chi2.ppf(0.999, df)
+    assert stat < thresh, (stat, thresh, counts[mask], exp[mask])
