@@ -342,10 +342,12 @@ class GPUModel:
         for wk in works:
             wk.wait()  # compute stream waits on the comm stream
 
-    def _layers_forward(self, M: int):
+    def _layers_forward(self, M: int, layer_hook=None):
         cfg = self.config
         eps = cfg.rms_norm_eps
         h, xn, t1 = self.b_h, self.b_xn, self.b_t1
+        if layer_hook is not None:
+            layer_hook(-1, h[:M])  # embedding output
         for i, lw in enumerate(self.layers):
             window = cfg.sliding_window if cfg.is_sliding(i) else 0
             ho.rmsnorm(h[:M], lw["g_in"], xn[:M], eps=eps)
@@ -389,6 +391,8 @@ class GPUModel:
                     ho.addinto(h[:M], t1[:M])
                 else:
                     self._linear(lw, "wdown", self.b_gate, h, res=h, M=M)
+            if layer_hook is not None:
+                layer_hook(i, h[:M])
 
     def _lm_head_last(self, M: int):
         """Final norm + lm_head on the last row -> self.b_logits (f32, V)."""
@@ -404,14 +408,21 @@ class GPUModel:
         if self.tp_branch:
             tpu.all_gather_into(self.b_logits, self.b_logits_l)
 
-    def forward_full(self, ids: np.ndarray) -> np.ndarray:
+    def forward_full(self, ids: np.ndarray,
+                     return_hidden_states: bool = False):
         """All-positions logits (M, V) — reference-parity API (the HF
         tuple shape).  Prefill-style pass; vocab GEMM per
-        chunk.  Resets the cache."""
+        chunk.  Resets the cache.  With ``return_hidden_states`` also
+        returns the L+1 per-layer hidden states (embedding output +
+        each decoder layer, reference ``all_hidden_states``
+        llama3.2_model.py:682-716) as fp32 numpy arrays."""
         ids = np.asarray(ids, dtype=np.int32).ravel()
         self.reset()
         n = len(ids)
         out = np.empty((n, self.config.vocab_size), dtype=np.float32)
+        L = self.config.num_hidden_layers
+        hidden = ([np.empty((n, self.H), dtype=np.float32)
+                   for _ in range(L + 1)] if return_hidden_states else None)
         logits_buf = torch.empty(self.PC, self.vocab_l,
                                  dtype=torch.bfloat16, device=self.device)
         done = 0
@@ -422,7 +433,16 @@ class GPUModel:
             ho.i32_set(self.len_buf, done)
             ho.embed(self.embed, self.ids_buf, self.b_h, M,
                      self.config.embed_scale)
-            self._layers_forward(M)
+            if return_hidden_states:
+                d0 = done
+
+                def hook(i, h):
+                    torch.cuda.synchronize()
+                    hidden[i + 1][d0:d0 + h.shape[0]] = \
+                        h.float().cpu().numpy()
+                self._layers_forward(M, layer_hook=hook)
+            else:
+                self._layers_forward(M)
             ho.rmsnorm(self.b_h[:M], self.g_final, self.b_xn[:M],
                        eps=self.config.rms_norm_eps)
             if self.fp8:
@@ -452,7 +472,21 @@ class GPUModel:
         # mirror the device len so a following decode()'s overflow guard
         # counts from the right base (KV-pool OOB otherwise)
         self._host_len = n
+        if return_hidden_states:
+            return out, hidden
         return out
+
+    def forward_hf(self, ids: np.ndarray):
+        """Reference-parity output tuple ``(loss, logits, kv_cache,
+        hidden_states, attentions)`` (llama3.2_model.py:726-822).
+        ``loss`` is None (inference-only, as in the reference) and
+        ``attentions`` is None by design: the fused attention kernels
+        use online softmax and never materialize the probability
+        matrix."""
+        logits, hidden = self.forward_full(ids, return_hidden_states=True)
+        cache = DeviceCacheHandle(self)
+        cache.seq_len = len(np.ravel(ids))
+        return None, logits, cache, hidden, None
 
     def generate_tokens(self, prompt_ids, max_tokens: int,
                         greedy: bool = True, min_p: float = 0.1,

@@ -158,14 +158,20 @@ class NumpyModel:
 
     # -- forward ---------------------------------------------------------
     def forward(self, input_ids: np.ndarray, cache: NumpyKVCache,
-                pos0: Optional[int] = None) -> np.ndarray:
+                pos0: Optional[int] = None,
+                collect_hidden: Optional[list] = None) -> np.ndarray:
         """input_ids: (q_len,) int array.  Returns logits (q_len, vocab).
-        Writes K/V at positions [pos0, pos0+q_len) and sets cache.seq_len."""
+        Writes K/V at positions [pos0, pos0+q_len) and sets cache.seq_len.
+        ``collect_hidden``: optional list that receives the embedding
+        output plus each layer's output (L+1 arrays — the reference's
+        ``all_hidden_states``, llama3.2_model.py:682-716)."""
         cfg = self.config
         if pos0 is None:
             pos0 = cache.seq_len
         h = self.w["model.embed_tokens.weight"][input_ids].astype(np.float32)
         h = h * np.float32(cfg.embed_scale)
+        if collect_hidden is not None:
+            collect_hidden.append(h.copy())
         gemma = cfg.model_type == "gemma2"
         for i in range(cfg.num_hidden_layers):
             pl = f"model.layers.{i}"
@@ -185,6 +191,8 @@ class NumpyModel:
                 res = h
                 x = self._rmsnorm(h, f"{pl}.post_attention_layernorm.weight")
                 h = res + self._mlp(i, x)
+            if collect_hidden is not None:
+                collect_hidden.append(h.copy())
         h = self._rmsnorm(h, "model.norm.weight")
         logits = h @ self.w["lm_head.weight"].T
         if cfg.final_logit_softcapping:
@@ -192,3 +200,15 @@ class NumpyModel:
             logits = c * np.tanh(logits / c)
         cache.seq_len = pos0 + len(input_ids)
         return logits
+
+    def forward_hf(self, input_ids: np.ndarray, cache: NumpyKVCache = None):
+        """Reference-parity output tuple ``(loss, logits, kv_cache,
+        hidden_states, attentions)`` (llama3.2_model.py:726-822).
+        ``attentions`` is None by design: attention here is computed with
+        online softmax and the probability matrix never materializes."""
+        cache = cache or NumpyKVCache(self.config,
+                                      len(np.ravel(input_ids)) + 1)
+        hidden = []
+        logits = self.forward(np.ravel(input_ids), cache, 0,
+                              collect_hidden=hidden)
+        return None, logits, cache, hidden, None

@@ -116,6 +116,28 @@ def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
     )
 
 
+def _hub_download(repo_id: str) -> str:
+    """Fetch a checkpoint from the HF hub (reference parity:
+    ``snapshot_download``, llama3.2_model.py:1090).  Raises a clear
+    FileNotFoundError when offline or the repo is unavailable."""
+    try:
+        from huggingface_hub import snapshot_download
+    except ImportError as e:
+        raise FileNotFoundError(
+            f"{repo_id!r} looks like an HF repo id but huggingface_hub "
+            f"is not installed") from e
+    try:
+        return snapshot_download(
+            repo_id,
+            allow_patterns=["*.json", "*.safetensors", "tokenizer*",
+                            "*.model"])
+    except Exception as e:
+        raise FileNotFoundError(
+            f"could not download {repo_id!r} from the HF hub (no network "
+            f"access, auth, or unknown repo): {type(e).__name__}: {e}"
+        ) from e
+
+
 class ByteTokenizer:
     """Self-contained fallback tokenizer (byte-level) for environments
     without HF tokenizer files; real checkpoints use ``AutoTokenizer``."""
@@ -133,10 +155,14 @@ def load_model(model_dir_or_preset: str, backend: str = "auto",
                device: str = "cuda", dtype: str = "bf16",
                max_seq: int = 4096, seed: int = 0):
     """Reference-parity entry (``load_model`` -> (tokenizer, model, config),
-    ``llama3.2_model.py:1082-1099``), local-only.
+    ``llama3.2_model.py:1082-1099``).
 
     ``model_dir_or_preset``: a checkpoint directory (config.json +
-    safetensors) or a preset name (random-init synthetic weights).
+    safetensors), a preset name (random-init synthetic weights), or an
+    HF hub repo id like ``"meta-llama/Llama-3.2-1B"`` — the hub path
+    mirrors the reference's ``snapshot_download`` flow
+    (``llama3.2_model.py:1090``) and fails with a clear error when the
+    machine has no network access.
     ``backend``: "numpy" (CPU oracle), "gpu" (HIP engine), or "auto".
     """
     import os
@@ -152,9 +178,15 @@ def load_model(model_dir_or_preset: str, backend: str = "auto",
         config = preset_config(model_dir_or_preset)
         weights = random_weights(config, seed=seed)
         tok_dir = None
+    elif "/" in model_dir_or_preset:
+        local = _hub_download(model_dir_or_preset)
+        config = load_config(local)
+        weights = load_weights_numpy(local)
+        tok_dir = local
     else:
         raise FileNotFoundError(
-            f"{model_dir_or_preset!r} is neither a directory nor a preset")
+            f"{model_dir_or_preset!r} is neither a directory, a preset "
+            f"({sorted(PRESETS)}), nor an HF repo id")
 
     tokenizer = None
     if tok_dir is not None:

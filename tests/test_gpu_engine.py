@@ -341,3 +341,20 @@ def test_fast_generate_eos_stop():
     r2 = L.generate("abc", tok, m, max_tokens=8, stream=False, params=p,
                     stop_on_eos=True)
     assert r2.token_ids == r.token_ids[:first + 1]
+
+
+def test_gpu_forward_hf_hidden_states_match_oracle():
+    """GPU forward_hf: reference 5-tuple surface with per-layer hidden
+    states matching the NumPy oracle."""
+    cfg, gpu, ref = make_pair("tiny-llama", seed=40)
+    ids = np.arange(1, 10)
+    loss, logits, cache, hidden, attn = gpu.forward_hf(ids)
+    assert loss is None and attn is None
+    assert cache.seq_len == len(ids)
+    assert len(hidden) == cfg.num_hidden_layers + 1
+    r_loss, r_logits, r_cache, r_hidden, r_attn = ref.forward_hf(ids)
+    for i, (a, b) in enumerate(zip(hidden, r_hidden)):
+        assert a.shape == b.shape
+        scale = np.abs(b).max() + 1e-6
+        assert np.abs(a - b).max() < 0.05 * scale, f"layer {i}"
+    assert (logits.argmax(1) == r_logits.argmax(1)).mean() > 0.9

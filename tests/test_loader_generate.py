@@ -110,3 +110,58 @@ def test_eos_list_stops_generation():
     r2 = L.generate("abcd", tok, model, max_tokens=6, stream=False,
                     params=p, stop_on_eos=True)
     assert r2.token_ids == ref.token_ids[:first + 1]
+
+
+def test_load_model_by_hub_repo_id_mocked(tmp_path, monkeypatch):
+    """Reference parity: load_model('org/name') goes through
+    snapshot_download (llama3.2_model.py:1082-1099).  Mocked hub: the
+    download lands in a local dir written by the synthetic-checkpoint
+    writer; offline failure raises a clear FileNotFoundError."""
+    from llm_np_cp_amd.io.loader import write_synthetic_checkpoint
+    import llm_np_cp_amd.runtime.generate as G
+
+    d = str(tmp_path / "hub_snapshot")
+    write_synthetic_checkpoint(d, "tiny-llama", seed=7)
+    calls = {}
+
+    def fake_snapshot_download(repo_id, **kw):
+        calls["repo"] = repo_id
+        return d
+
+    import huggingface_hub
+    monkeypatch.setattr(huggingface_hub, "snapshot_download",
+                        fake_snapshot_download)
+    tok, model, cfg = G.load_model("fake-org/tiny-llama", backend="numpy")
+    assert calls["repo"] == "fake-org/tiny-llama"
+    r = L.generate("hi", tok, model, max_tokens=3, stream=False,
+                   params=SamplingParams(strategy="greedy"),
+                   stop_on_eos=False)
+    assert len(r.token_ids) == 3
+
+    def failing_download(repo_id, **kw):
+        raise OSError("offline")
+
+    monkeypatch.setattr(huggingface_hub, "snapshot_download",
+                        failing_download)
+    with pytest.raises(FileNotFoundError, match="no network|offline"):
+        G.load_model("meta-llama/Llama-3.2-1B", backend="numpy")
+
+
+def test_numpy_forward_hf_tuple_surface():
+    """Reference output surface: (loss, logits, kv_cache, hidden_states,
+    attentions) — llama3.2_model.py:726-822."""
+    tok, model, cfg = L.load_model("tiny-llama", backend="numpy", seed=5)
+    ids = np.arange(1, 8)
+    loss, logits, cache, hidden, attn = model.forward_hf(ids)
+    assert loss is None and attn is None
+    assert logits.shape == (7, cfg.vocab_size)
+    assert len(hidden) == cfg.num_hidden_layers + 1
+    assert all(h.shape == (7, cfg.hidden_size) for h in hidden)
+    assert cache.seq_len == 7
+    # hidden[0] is the (scaled) embedding output
+    emb = model.w["model.embed_tokens.weight"][ids] * cfg.embed_scale
+    np.testing.assert_allclose(hidden[0], emb, rtol=1e-5, atol=1e-6)
+    # logits equal a plain forward
+    from llm_np_cp_amd.models.numpy_ref import NumpyKVCache
+    ref = model.forward(ids, NumpyKVCache(cfg, 16), 0)
+    np.testing.assert_allclose(logits, ref, rtol=1e-5, atol=1e-6)
