@@ -172,6 +172,116 @@ DEVINL const u16* gemv_stage(char* smem, const u16* x, const u16* x2,
   return (stage == STAGE_RAW) ? x : xs;  // RAW: L2-hot, no barrier
 }
 
+// f32 variant of the staging pass, used by the fp8 GEMV: x lands in LDS
+// as fp32 PAIRS so the inner loop can feed v_pk_fma_f32 directly from
+// the v_cvt_pk_f32_fp8 outputs (the round-1 fp8 stream was
+// conversion-VALU-bound at 4.3 TB/s — this halves the VALU work per
+// weight byte).  Semantics match gemv_stage: the NORM2-persisted h' is
+// still bf16-rounded (it feeds the residual stream), only the staged
+// vector skips the final bf16 round-trip (within GEMV tolerance).
+DEVINL const float* gemv_stage_f32(char* smem, const u16* x, const u16* x2,
+                                   const float* g, const float* g2,
+                                   u16* hout, int K, int stage, int act,
+                                   float eps) {
+  float* xs = (float*)smem;
+  const int STRIDE = blockDim.x * 8;
+  if (stage == STAGE_RAW) {
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
+      s8v v = *(const s8v*)(x + i);
+#pragma unroll
+      for (int j = 0; j < 8; j++) xs[i + j] = b2f(((u16*)&v)[j]);
+    }
+    __syncthreads();
+    return xs;
+  }
+  float* red = (float*)(smem + (size_t)K * 4);
+  if (stage == STAGE_NORM2) {
+    const u16* h = x2;
+    float ss = 0.f;
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
+      s8v v = *(const s8v*)(x + i);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = b2f(((u16*)&v)[j]);
+        ss += f * f;
+      }
+    }
+    ss = wave_reduce_sum(ss);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+    __syncthreads();
+    float rnorm_a = rsqrtf(stage_red_sum(red) / (float)K + eps);
+    __syncthreads();
+    float ss2 = 0.f;
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
+      s8v v = *(const s8v*)(x + i);
+      s8v hv = *(const s8v*)(h + i);
+      f4v ga = *(const f4v*)(g + i);
+      f4v gb = *(const f4v*)(g + i + 4);
+      u16 o[8];
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = b2f(((u16*)&v)[j]) * rnorm_a * (j < 4 ? ga[j] : gb[j - 4])
+                  + b2f(((u16*)&hv)[j]);
+        o[j] = f2b(f);           // bf16-rounded h' (residual stream)
+        float fr = b2f(o[j]);
+        xs[i + j] = fr;
+        ss2 += fr * fr;
+      }
+      if (blockIdx.x == 0) *(s8v*)(hout + i) = *(s8v*)o;
+    }
+    ss2 = wave_reduce_sum(ss2);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss2;
+    __syncthreads();
+    float rnorm_b = rsqrtf(stage_red_sum(red) / (float)K + eps);
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
+#pragma unroll
+      for (int j = 0; j < 8; j++)
+        xs[i + j] = xs[i + j] * rnorm_b * g2[i + j];
+    }
+    __syncthreads();
+    return xs;
+  }
+  if (stage == STAGE_NORM) {
+    float ss = 0.f;
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
+      s8v v = *(const s8v*)(x + i);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = b2f(((u16*)&v)[j]);
+        xs[i + j] = f;
+        ss += f * f;
+      }
+    }
+    ss = wave_reduce_sum(ss);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+    __syncthreads();
+    float rnorm = rsqrtf(stage_red_sum(red) / (float)K + eps);
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
+#pragma unroll
+      for (int j = 0; j < 8; j++) xs[i + j] *= rnorm * g[i + j];
+    }
+  } else if (stage == STAGE_GLU) {
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
+      s8v gv = *(const s8v*)(x + i);
+      s8v uv = *(const s8v*)(x2 + i);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float xx = b2f(((u16*)&gv)[j]);
+        float a;
+        if (act == 0) {
+          a = xx / (1.f + __expf(-xx));
+        } else {
+          float c = 0.7978845608028654f * (xx + 0.044715f * xx * xx * xx);
+          a = 0.5f * xx * (1.f + tanhf(c));
+        }
+        xs[i + j] = a * b2f(((u16*)&uv)[j]);
+      }
+    }
+  }
+  __syncthreads();
+  return xs;
+}
+
 DEVINL void gemv_epilogue(float acc, int row, void* y, const u16* res,
                           int out_f32, float softcap) {
   if (softcap > 0.f) acc = softcap * tanhf(acc / softcap);
@@ -306,6 +416,11 @@ DEVINL void fp8x16_to_f32(u4v w, float* o) {
   }
 }
 
+// Inner loop is PACKED: x is staged as fp32 in LDS (gemv_stage_f32),
+// each v_cvt_pk_f32_fp8 output pair feeds one v_pk_fma_f32 against an
+// LDS fp32 pair — ~16 VALU per 16 weight bytes vs ~40 for the round-1
+// scalar form (the fp8 stream was conversion-bound at 4.3 TB/s while
+// bf16 hit 6.4; see profiles/decode_kernels_r01.md).
 template <bool NT, int RPW>
 __global__ void __launch_bounds__(512)
 k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
@@ -314,8 +429,8 @@ k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
              void* __restrict__ y, const u16* __restrict__ res, int N, int K,
              int stage, int act, float eps, int out_f32, float softcap) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const u16* xv = gemv_stage(smem, x, x2, g, g2, (u16*)res, K, stage, act,
-                             eps);
+  const float* xv = gemv_stage_f32(smem, x, x2, g, g2, (u16*)res, K, stage,
+                                   act, eps);
   const u16* eres = (stage == STAGE_NORM2) ? nullptr : res;
 
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
@@ -324,54 +439,54 @@ k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
   for (int row0 = (blockIdx.x * wpb + wave) * RPW; row0 < N;
        row0 += rstride) {
   const uint8_t* Wr[RPW];
-  float a0[RPW], a1[RPW];
+  f2v a0[RPW], a1[RPW];
 #pragma unroll
   for (int r = 0; r < RPW; r++) {
     int rr = row0 + r < N ? row0 + r : N - 1;
     Wr[r] = W + (size_t)rr * K;
-    a0[r] = a1[r] = 0.f;
+    a0[r] = (f2v){0.f, 0.f};
+    a1[r] = (f2v){0.f, 0.f};
   }
   int k = lane * 16;
   for (; k + 1024 + 16 <= K; k += 2048) {
-    s8v x0a = *(const s8v*)(xv + k), x0b = *(const s8v*)(xv + k + 8);
-    s8v x1a = *(const s8v*)(xv + k + 1024), x1b = *(const s8v*)(xv + k + 1032);
 #pragma unroll
     for (int r = 0; r < RPW; r++) {
       u4v w0 = NT ? __builtin_nontemporal_load((const u4v*)(Wr[r] + k))
                   : *(const u4v*)(Wr[r] + k);
       u4v w1 = NT ? __builtin_nontemporal_load((const u4v*)(Wr[r] + k + 1024))
                   : *(const u4v*)(Wr[r] + k + 1024);
-      float f0[16], f1[16];
-      fp8x16_to_f32(w0, f0);
-      fp8x16_to_f32(w1, f1);
 #pragma unroll
-      for (int j = 0; j < 8; j++) {
-        a0[r] += f0[j] * b2f(((u16*)&x0a)[j]);
-        a0[r] += f0[j + 8] * b2f(((u16*)&x0b)[j]);
-        a1[r] += f1[j] * b2f(((u16*)&x1a)[j]);
-        a1[r] += f1[j + 8] * b2f(((u16*)&x1b)[j]);
+      for (int q = 0; q < 4; q++) {
+        f2v ca = __builtin_amdgcn_cvt_pk_f32_fp8(w0[q], false);
+        f2v cb = __builtin_amdgcn_cvt_pk_f32_fp8(w0[q], true);
+        a0[r] += ca * *(const f2v*)(xv + k + q * 4);
+        a1[r] += cb * *(const f2v*)(xv + k + q * 4 + 2);
+        f2v da = __builtin_amdgcn_cvt_pk_f32_fp8(w1[q], false);
+        f2v db = __builtin_amdgcn_cvt_pk_f32_fp8(w1[q], true);
+        a0[r] += da * *(const f2v*)(xv + k + 1024 + q * 4);
+        a1[r] += db * *(const f2v*)(xv + k + 1024 + q * 4 + 2);
       }
     }
   }
   for (; k < K; k += 1024) {
-    s8v xa = *(const s8v*)(xv + k), xb = *(const s8v*)(xv + k + 8);
 #pragma unroll
     for (int r = 0; r < RPW; r++) {
       u4v w0 = NT ? __builtin_nontemporal_load((const u4v*)(Wr[r] + k))
                   : *(const u4v*)(Wr[r] + k);
-      float f0[16];
-      fp8x16_to_f32(w0, f0);
 #pragma unroll
-      for (int j = 0; j < 8; j++) {
-        a0[r] += f0[j] * b2f(((u16*)&xa)[j]);
-        a0[r] += f0[j + 8] * b2f(((u16*)&xb)[j]);
+      for (int q = 0; q < 4; q++) {
+        f2v ca = __builtin_amdgcn_cvt_pk_f32_fp8(w0[q], false);
+        f2v cb = __builtin_amdgcn_cvt_pk_f32_fp8(w0[q], true);
+        a0[r] += ca * *(const f2v*)(xv + k + q * 4);
+        a1[r] += cb * *(const f2v*)(xv + k + q * 4 + 2);
       }
     }
   }
 #pragma unroll
   for (int r = 0; r < RPW; r++) {
     int rr = row0 + r < N ? row0 + r : N - 1;
-    float acc = wave_reduce_sum(a0[r] + a1[r]) * scales[rr];
+    f2v s = a0[r] + a1[r];
+    float acc = wave_reduce_sum(s[0] + s[1]) * scales[rr];
     if (lane == 0 && row0 + r < N)
       gemv_epilogue(acc, row0 + r, y, eres, out_f32, softcap);
   }
@@ -386,7 +501,23 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
                                       int out_f32, float softcap, int nt,
                                       int rpw, int maxblocks,
                                       hipStream_t stream) {
-  size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 32);
+  // fp32 staging for packed math (every stage incl. RAW)
+  size_t lds = (size_t)K * 4 + 32;
+  if (lds > 65536) {
+    // gfx950 allows up to 160 KB dynamic LDS with an explicit opt-in
+    // (e.g. Gemma-27B down-proj K=36864)
+    static bool raised = false;
+    if (!raised) {
+#define GEMV8_RAISE(NTV, RPWV)                                              \
+      hipFuncSetAttribute((const void*)&k_gemv_fp8_t<NTV, RPWV>,            \
+                          hipFuncAttributeMaxDynamicSharedMemorySize,       \
+                          160 * 1024)
+      GEMV8_RAISE(true, 1); GEMV8_RAISE(true, 2);
+      GEMV8_RAISE(false, 1); GEMV8_RAISE(false, 2);
+#undef GEMV8_RAISE
+      raised = true;
+    }
+  }
   int threads = 256;  // 512 measured slower (fp8 1391->1328)
   int wpb = threads / 64;
   int blocks = (N + wpb * rpw - 1) / (wpb * rpw);
