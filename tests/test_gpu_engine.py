@@ -347,7 +347,7 @@ def test_gpu_forward_hf_hidden_states_match_oracle():
     """GPU forward_hf: reference 5-tuple surface with per-layer hidden
     states matching the NumPy oracle."""
     cfg, gpu, ref = make_pair("tiny-llama", seed=40)
-    ids = np.arange(1, 10)
+    ids = np.random.default_rng(41).integers(0, cfg.vocab_size, size=9)
     loss, logits, cache, hidden, attn = gpu.forward_hf(ids)
     assert loss is None and attn is None
     assert cache.seq_len == len(ids)
@@ -357,4 +357,5 @@ def test_gpu_forward_hf_hidden_states_match_oracle():
         assert a.shape == b.shape
         scale = np.abs(b).max() + 1e-6
         assert np.abs(a - b).max() < 0.05 * scale, f"layer {i}"
-    assert (logits.argmax(1) == r_logits.argmax(1)).mean() > 0.9
+    kl = kl_bits(r_logits, logits, axis=1)
+    assert kl.mean() < 0.02, kl.mean()
