@@ -51,6 +51,10 @@
 // (pre-norm of the projection this GEMV computes).  Removes the two
 // standalone k_rmsnorm launches per Gemma layer.
 #define STAGE_NORM2 3
+// NORM_EMBED: like NORM but x = embedding TABLE base; the row is
+// gathered at *tok (device token id) and scaled by escale first —
+// fuses the k_embed launch into the first QKV GEMV of the decode step.
+#define STAGE_NORM_EMBED 4
 
 // staging helper shared by bf16/fp8 GEMV (RMSNorm / GLU fused pre-ops)
 DEVINL float stage_red_sum(float* red) {
@@ -62,7 +66,8 @@ DEVINL float stage_red_sum(float* red) {
 
 DEVINL const u16* gemv_stage(char* smem, const u16* x, const u16* x2,
                              const float* g, const float* g2, u16* hout,
-                             int K, int stage, int act, float eps) {
+                             int K, int stage, int act, float eps,
+                             float escale) {
   u16* xs = (u16*)smem;
   const int STRIDE = blockDim.x * 8;
   if (stage == STAGE_NORM2) {
@@ -124,10 +129,24 @@ DEVINL const u16* gemv_stage(char* smem, const u16* x, const u16* x2,
     __syncthreads();
     return xs;
   }
-  if (stage == STAGE_NORM) {
+  if (stage == STAGE_NORM || stage == STAGE_NORM_EMBED) {
+    // NORM_EMBED: x is the embedding TABLE; gather row *x2 (token id),
+    // scale by escale (bf16-rounded, matching the standalone k_embed),
+    // and have block 0 persist it as the residual-stream h
+    const u16* xsrc = x;
+    if (stage == STAGE_NORM_EMBED)
+      xsrc = x + (size_t)(*(const int*)x2) * (size_t)K;
     float ss = 0.f;
     for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
-      s8v v = *(const s8v*)(x + i);
+      s8v v = *(const s8v*)(xsrc + i);
+      if (stage == STAGE_NORM_EMBED) {
+        u16 o[8];
+#pragma unroll
+        for (int j = 0; j < 8; j++)
+          o[j] = f2b(b2f(((u16*)&v)[j]) * escale);
+        v = *(s8v*)o;
+        if (blockIdx.x == 0) *(s8v*)(hout + i) = v;
+      }
       *(s8v*)(xs + i) = v;
 #pragma unroll
       for (int j = 0; j < 8; j++) {
@@ -182,7 +201,7 @@ DEVINL const u16* gemv_stage(char* smem, const u16* x, const u16* x2,
 DEVINL const float* gemv_stage_f32(char* smem, const u16* x, const u16* x2,
                                    const float* g, const float* g2,
                                    u16* hout, int K, int stage, int act,
-                                   float eps) {
+                                   float eps, float escale) {
   float* xs = (float*)smem;
   const int STRIDE = blockDim.x * 8;
   if (stage == STAGE_RAW) {
@@ -241,10 +260,21 @@ DEVINL const float* gemv_stage_f32(char* smem, const u16* x, const u16* x2,
     __syncthreads();
     return xs;
   }
-  if (stage == STAGE_NORM) {
+  if (stage == STAGE_NORM || stage == STAGE_NORM_EMBED) {
+    const u16* xsrc = x;
+    if (stage == STAGE_NORM_EMBED)
+      xsrc = x + (size_t)(*(const int*)x2) * (size_t)K;
     float ss = 0.f;
     for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
-      s8v v = *(const s8v*)(x + i);
+      s8v v = *(const s8v*)(xsrc + i);
+      if (stage == STAGE_NORM_EMBED) {
+        u16 o[8];
+#pragma unroll
+        for (int j = 0; j < 8; j++)
+          o[j] = f2b(b2f(((u16*)&v)[j]) * escale);
+        v = *(s8v*)o;
+        if (blockIdx.x == 0) *(s8v*)(hout + i) = v;
+      }
 #pragma unroll
       for (int j = 0; j < 8; j++) {
         float f = b2f(((u16*)&v)[j]);
@@ -299,11 +329,13 @@ k_gemv_bf16_t(const u16* __restrict__ W, const u16* __restrict__ x,
               const u16* __restrict__ x2, const float* __restrict__ g,
               const float* __restrict__ g2, void* __restrict__ y,
               const u16* __restrict__ res, int N, int K, int stage, int act,
-              float eps, int out_f32, float softcap) {
+              float eps, int out_f32, float softcap, float escale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const u16* xv = gemv_stage(smem, x, x2, g, g2, (u16*)res, K, stage, act,
-                             eps);
-  const u16* eres = (stage == STAGE_NORM2) ? nullptr : res;  // res = hout
+                             eps, escale);
+  // NORM2 / NORM_EMBED borrow `res` as the persisted-h output
+  const u16* eres = (stage == STAGE_NORM2 || stage == STAGE_NORM_EMBED)
+                        ? nullptr : res;
 
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int wpb = blockDim.x >> 6;
@@ -374,7 +406,7 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
                                        int stage, int act, float eps,
                                        int out_f32, float softcap,
                                        int nt, int rpw, int maxblocks,
-                                       hipStream_t stream) {
+                                       float escale, hipStream_t stream) {
   size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 32);
   int threads = 256;  // 512 measured slower (fp8 1391->1328)
   int wpb = threads / 64;
@@ -386,7 +418,7 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
                      lds, stream, (const u16*)W, (const u16*)x,              \
                      (const u16*)x2, (const float*)g, (const float*)g2, y,   \
                      (const u16*)res, N, K, stage, act, eps, out_f32,        \
-                     softcap)
+                     softcap, escale)
   if (nt && rpw == 2) GEMV_CASE(true, 2);
   else if (nt) GEMV_CASE(true, 1);
   else if (rpw == 2) GEMV_CASE(false, 2);
@@ -427,11 +459,13 @@ k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
              const u16* __restrict__ x, const u16* __restrict__ x2,
              const float* __restrict__ g, const float* __restrict__ g2,
              void* __restrict__ y, const u16* __restrict__ res, int N, int K,
-             int stage, int act, float eps, int out_f32, float softcap) {
+             int stage, int act, float eps, int out_f32, float softcap,
+             float escale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const float* xv = gemv_stage_f32(smem, x, x2, g, g2, (u16*)res, K, stage,
-                                   act, eps);
-  const u16* eres = (stage == STAGE_NORM2) ? nullptr : res;
+                                   act, eps, escale);
+  const u16* eres = (stage == STAGE_NORM2 || stage == STAGE_NORM_EMBED)
+                        ? nullptr : res;
 
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int wpb = blockDim.x >> 6;
@@ -499,7 +533,7 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
                                       const void* res, int N, int K,
                                       int stage, int act, float eps,
                                       int out_f32, float softcap, int nt,
-                                      int rpw, int maxblocks,
+                                      int rpw, int maxblocks, float escale,
                                       hipStream_t stream) {
   // fp32 staging for packed math (every stage incl. RAW)
   size_t lds = (size_t)K * 4 + 32;
@@ -528,7 +562,7 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
                      lds, stream, (const uint8_t*)W, (const float*)scales,  \
                      (const u16*)x, (const u16*)x2, (const float*)g,        \
                      (const float*)g2, y, (const u16*)res, N, K, stage,     \
-                     act, eps, out_f32, softcap)
+                     act, eps, out_f32, softcap, escale)
   if (nt && rpw == 2) GEMV8_CASE(true, 2);
   else if (nt) GEMV8_CASE(true, 1);
   else if (rpw == 2) GEMV8_CASE(false, 2);
@@ -1350,16 +1384,22 @@ k_logit_max(const float* __restrict__ logits, int V,
   }
 }
 
-// pass 2: winner = argmax over kept tokens of (logit/T [+ Gumbel]) -> *pick
+// pass 2: winner = argmax over kept tokens of (logit/T [+ Gumbel]) -> *pick,
+// and the LAST-arriving block (G16 ticket) commits it: write *next_token,
+// append to out_ring, bump len/ctr, re-zero the scratch — the former
+// k_sample_fin's 1-thread launch (~4 us) folded away.
 // Temperature folds into both the min-p keep-set and the Gumbel score:
 //   p_i(T) >= min_p * p_max(T)  <=>  l_i >= l_max + T*ln(min_p)
 //   Gumbel-argmax over p(T)     <=>  argmax of l_i/T + G_i
 extern "C" __global__ void __launch_bounds__(256)
 k_sample_pick(const float* __restrict__ logits, int V, float min_p,
               int greedy, uint64_t seed, float inv_temp,
-              const unsigned long long* __restrict__ gmax,
-              const uint64_t* __restrict__ ctr,
-              unsigned long long* __restrict__ pick) {
+              unsigned long long* __restrict__ gmax,
+              uint64_t* __restrict__ ctr,
+              unsigned long long* __restrict__ pick,
+              int* __restrict__ cnt, int* __restrict__ next_token,
+              int* __restrict__ out_ring, int* __restrict__ nout,
+              int* __restrict__ len_ptr, int bump_len) {
   float thresh = -INFINITY;
   if (!greedy)
     thresh = fkey_inv((uint32_t)(*gmax >> 32)) + __logf(min_p) / inv_temp;
@@ -1385,23 +1425,25 @@ k_sample_pick(const float* __restrict__ logits, int V, float min_p,
     if (o > pk) pk = o;
   }
   __shared__ unsigned long long ws[4];
+  __shared__ int lastflag;
   if ((threadIdx.x & 63) == 0) ws[threadIdx.x >> 6] = pk;
   __syncthreads();
   if (threadIdx.x == 0) {
     uint64_t b = ws[0];
     for (int w = 1; w < 4; w++) if (ws[w] > b) b = ws[w];
     atomicMax(pick, (unsigned long long)b);
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    int t = __hip_atomic_fetch_add(cnt, 1, __ATOMIC_ACQ_REL,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+    lastflag = (t == (int)gridDim.x - 1);
   }
-}
-
-// pass 3: commit the winner, reset scratch, advance device-side state
-extern "C" __global__ void k_sample_fin(
-    unsigned long long* __restrict__ pick,
-    unsigned long long* __restrict__ gmax, uint64_t* __restrict__ ctr,
-    int* __restrict__ next_token, int* __restrict__ out_ring,
-    int* __restrict__ nout, int* __restrict__ len_ptr, int bump_len,
-    int greedy, int V) {
-  int winner = unpack_idx(*pick);
+  __syncthreads();
+  if (!lastflag || threadIdx.x != 0) return;
+  // last arriver: commit the winner, reset scratch, advance state
+  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  *cnt = 0;
+  int winner = unpack_idx(
+      __hip_atomic_load(pick, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT));
   if (winner < 0 || winner >= V) winner = 0;  // NaN-logit insurance
   *next_token = winner;
   int n = *nout;
@@ -1416,9 +1458,10 @@ extern "C" __global__ void k_sample_fin(
 extern "C" hipError_t launch_sample(const void* logits, int V, float min_p,
                                     int greedy, uint64_t seed, float inv_temp,
                                     void* ctr, void* gmax, void* pick,
-                                    void* next_token, void* out_ring,
-                                    void* nout, void* len_ptr,
-                                    int bump_len, hipStream_t stream) {
+                                    void* cnt, void* next_token,
+                                    void* out_ring, void* nout,
+                                    void* len_ptr, int bump_len,
+                                    hipStream_t stream) {
   int blocks = (V + 255) / 256;
   if (blocks > 512) blocks = 512;
   if (!greedy)
@@ -1426,12 +1469,9 @@ extern "C" hipError_t launch_sample(const void* logits, int V, float min_p,
                        (const float*)logits, V, (unsigned long long*)gmax);
   hipLaunchKernelGGL(k_sample_pick, dim3(blocks), dim3(256), 0, stream,
                      (const float*)logits, V, min_p, greedy, seed, inv_temp,
-                     (const unsigned long long*)gmax, (const uint64_t*)ctr,
-                     (unsigned long long*)pick);
-  hipLaunchKernelGGL(k_sample_fin, dim3(1), dim3(1), 0, stream,
-                     (unsigned long long*)pick, (unsigned long long*)gmax,
-                     (uint64_t*)ctr, (int*)next_token, (int*)out_ring,
-                     (int*)nout, (int*)len_ptr, bump_len, greedy, V);
+                     (unsigned long long*)gmax, (uint64_t*)ctr,
+                     (unsigned long long*)pick, (int*)cnt, (int*)next_token,
+                     (int*)out_ring, (int*)nout, (int*)len_ptr, bump_len);
   return hipGetLastError();
 }
 

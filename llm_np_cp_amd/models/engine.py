@@ -257,6 +257,7 @@ class GPUModel:
         self.attn_cnt = torch.zeros(self.nh_l, dtype=torch.int32, device=dev)
         self.s_gmax = torch.zeros(1, dtype=torch.int64, device=dev)
         self.s_pick = torch.zeros(1, dtype=torch.int64, device=dev)
+        self.s_cnt = torch.zeros(1, dtype=torch.int32, device=dev)
         i32 = dict(dtype=torch.int32, device=dev)
         self.ids_buf = torch.zeros(PC, **i32)
         self.next_token = torch.zeros(1, **i32)
@@ -286,6 +287,7 @@ class GPUModel:
         self.rng_ctr.zero_()
         self.s_gmax.zero_()
         self.s_pick.zero_()
+        self.s_cnt.zero_()
         self._host_len = 0  # host mirror of len_buf (overflow guard)
 
     # ------------------------------------------------------------------
@@ -395,16 +397,16 @@ class GPUModel:
                 layer_hook(i, h[:M])
 
     def _lm_head_last(self, M: int):
-        """Final norm + lm_head on the last row -> self.b_logits (f32, V)."""
+        """Final norm + lm_head on the last row -> self.b_logits (f32, V);
+        the norm rides in the GEMV staging pass (no separate launch)."""
         hrow = self.b_h[M - 1] if M > 1 else self.b_h[0]
-        ho.rmsnorm(hrow, self.g_final, self.b_xn[0],
-                   eps=self.config.rms_norm_eps)
+        kw = dict(stage=ho.STAGE_NORM, g=self.g_final,
+                  eps=self.config.rms_norm_eps, softcap=self.final_softcap)
         if self.fp8:
-            ho.gemv_fp8(self.lm_head_q, self.lm_head_s, self.b_xn[0],
-                        self.b_logits_l, softcap=self.final_softcap)
+            ho.gemv_fp8(self.lm_head_q, self.lm_head_s, hrow,
+                        self.b_logits_l, **kw)
         else:
-            ho.gemv(self.lm_head, self.b_xn[0], self.b_logits_l,
-                    softcap=self.final_softcap)
+            ho.gemv(self.lm_head, hrow, self.b_logits_l, **kw)
         if self.tp_branch:
             tpu.all_gather_into(self.b_logits, self.b_logits_l)
 
@@ -594,7 +596,6 @@ class GPUModel:
         s0 = torch.cuda.current_stream()
         pf = self.prefetch_on
         self._pf_events = []
-        ho.embed(self.embed, self.next_token, self.b_h, 1, cfg.embed_scale)
         prev = None  # previous gemma layer (its postffn norm fused here)
         for i, lw in enumerate(self.layers):
             if pf and i + 1 < len(self.layers):
@@ -613,6 +614,13 @@ class GPUModel:
                             g=prev["g_postffn"], g2=lw["g_in"], res=hnext,
                             eps=eps)
                 h, hnext = hnext, h
+            elif i == 0:
+                # layer 0: embed gather fused into the QKV staging pass
+                # (x = table, x2 = sampled token id, res = persisted h)
+                self._dgemv(lw, "wqkv", self.embed, self.b_qkv,
+                            stage=ho.STAGE_NORM_EMBED, x2=self.next_token,
+                            g=lw["g_in"], res=h, eps=eps,
+                            escale=cfg.embed_scale)
             else:
                 self._dgemv(lw, "wqkv", h, self.b_qkv, stage=ho.STAGE_NORM,
                             g=lw["g_in"], eps=eps)
@@ -661,22 +669,32 @@ class GPUModel:
             self._pf_events.append(ev)
             s0.wait_event(ev)  # join the fork before the step ends
         if self.gemma:
-            # last layer's post-ffn sandwich norm (no following projection
-            # to fold it into)
-            ho.rmsnorm(t2, self.layers[-1]["g_postffn"], h, res=h, eps=eps)
-        ho.rmsnorm(h, self.g_final, self.b_xn[0], eps=eps)
-        if self.fp8:
-            ho.gemv_fp8(self.lm_head_q, self.lm_head_s, self.b_xn[0],
-                        self.b_logits_l, softcap=self.final_softcap)
+            # last layer's post-ffn sandwich norm + final norm, both
+            # fused into the lm_head staging (NORM2: h' = h + norm(t2)*
+            # g_postffn, then stage norm(h')*g_final)
+            kw = dict(stage=ho.STAGE_NORM2, x2=h,
+                      g=self.layers[-1]["g_postffn"], g2=self.g_final,
+                      res=hnext, eps=eps, softcap=self.final_softcap)
+            if self.fp8:
+                ho.gemv_fp8(self.lm_head_q, self.lm_head_s, t2,
+                            self.b_logits_l, **kw)
+            else:
+                ho.gemv(self.lm_head, t2, self.b_logits_l, **kw)
         else:
-            ho.gemv(self.lm_head, self.b_xn[0], self.b_logits_l,
-                    softcap=self.final_softcap)
+            # final norm fused into the lm_head staging pass
+            kw = dict(stage=ho.STAGE_NORM, g=self.g_final, eps=eps,
+                      softcap=self.final_softcap)
+            if self.fp8:
+                ho.gemv_fp8(self.lm_head_q, self.lm_head_s, h,
+                            self.b_logits_l, **kw)
+            else:
+                ho.gemv(self.lm_head, h, self.b_logits_l, **kw)
         if self.tp_branch:
             tpu.all_gather_into(self.b_logits, self.b_logits_l)
         ho.sample(self.b_logits, min_p, greedy, self.seed, self.rng_ctr,
                   self.s_gmax, self.s_pick, self.next_token, self.out_ring,
                   self.nout, self.len_buf, bump_len=True,
-                  temperature=temperature)
+                  temperature=temperature, cnt=self.s_cnt)
 
     def capture_decode_graph(self, greedy: bool = True, min_p: float = 0.1,
                              temperature: float = 1.0):
@@ -741,7 +759,8 @@ class GPUModel:
             ho.sample(self.b_logits, min_p, greedy, self.seed, self.rng_ctr,
                       self.s_gmax, self.s_pick, self.next_token,
                       self.out_ring, self.nout, self.len_buf,
-                      bump_len=False, temperature=temperature)
+                      bump_len=False, temperature=temperature,
+                      cnt=self.s_cnt)
         n_steps = n_tokens - (1 if first_from_logits else 0)
         if use_graph and n_steps > 0:
             # capture_decode_graph reports the warm-up step it executed

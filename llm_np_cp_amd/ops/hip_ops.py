@@ -27,11 +27,11 @@ _SIGS = {
     "launch_gemv_bf16": [ctypes.c_void_p] * 7 + [ctypes.c_int] * 4 +
                         [ctypes.c_float, ctypes.c_int, ctypes.c_float,
                          ctypes.c_int, ctypes.c_int, ctypes.c_int,
-                         ctypes.c_void_p],
+                         ctypes.c_float, ctypes.c_void_p],
     "launch_gemv_fp8": [ctypes.c_void_p] * 8 + [ctypes.c_int] * 4 +
                        [ctypes.c_float, ctypes.c_int, ctypes.c_float,
                         ctypes.c_int, ctypes.c_int, ctypes.c_int,
-                        ctypes.c_void_p],
+                        ctypes.c_float, ctypes.c_void_p],
     "launch_rmsnorm": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 2 +
                       [ctypes.c_float, ctypes.c_int, ctypes.c_void_p],
     "launch_rope_cache": [ctypes.c_void_p] * 8 + [ctypes.c_int] * 5 +
@@ -51,7 +51,7 @@ _SIGS = {
                     [ctypes.c_float, ctypes.c_void_p],
     "launch_sample": [ctypes.c_void_p, ctypes.c_int, ctypes.c_float,
                       ctypes.c_int, ctypes.c_uint64, ctypes.c_float] +
-                     [ctypes.c_void_p] * 7 +
+                     [ctypes.c_void_p] * 8 +
                      [ctypes.c_int, ctypes.c_void_p],
     "launch_gemm_bf16": [ctypes.c_void_p] * 5 + [ctypes.c_int] * 3 +
                         [ctypes.c_void_p],
@@ -116,6 +116,11 @@ def _ptr(t) -> int:
 # ----------------------------------------------------------------------
 
 STAGE_RAW, STAGE_NORM, STAGE_GLU, STAGE_NORM2 = 0, 1, 2, 3
+# NORM_EMBED: W-input x is the embedding TABLE; the row at *x2 (device
+# token id) is gathered, scaled by `escale` and normed — fuses the
+# decode step's k_embed launch into the first QKV GEMV.  `res` receives
+# the persisted h (residual stream).
+STAGE_NORM_EMBED = 4
 
 
 def gemv(W: torch.Tensor, x: torch.Tensor, y: torch.Tensor,
@@ -123,16 +128,19 @@ def gemv(W: torch.Tensor, x: torch.Tensor, y: torch.Tensor,
          stage: int = 0, x2: torch.Tensor | None = None,
          g: torch.Tensor | None = None, act: int = 0, eps: float = 1e-5,
          nt: int = 1, rpw: int = 1, maxblocks: int = 0,
-         g2: torch.Tensor | None = None):
+         g2: torch.Tensor | None = None, escale: float = 1.0):
     """y[N] = W[N,K] @ stage(x)[K] (+res); stage fuses RMSNorm / GLU /
     the Gemma sandwich (NORM2: x2 = h_in, g/g2 = post/pre gammas, res =
-    h_out ping-pong) into the LDS staging pass; nt = non-temporal W."""
+    h_out ping-pong) / the embed gather (NORM_EMBED: x = embed table,
+    x2 = token-id i32, res = persisted h) into the LDS staging pass;
+    nt = non-temporal W."""
     N, K = W.shape
     out_f32 = 1 if y.dtype == torch.float32 else 0
     _check(lib().launch_gemv_bf16(
         _ptr(W), _ptr(x), _ptr(x2), _ptr(g), _ptr(g2), _ptr(y), _ptr(res),
         N, K, stage, act, ctypes.c_float(eps), out_f32,
-        ctypes.c_float(softcap), nt, rpw, maxblocks, _stream()), "gemv")
+        ctypes.c_float(softcap), nt, rpw, maxblocks,
+        ctypes.c_float(escale), _stream()), "gemv")
 
 
 def gemv_fp8(Wq: torch.Tensor, scales: torch.Tensor, x: torch.Tensor,
@@ -140,14 +148,16 @@ def gemv_fp8(Wq: torch.Tensor, scales: torch.Tensor, x: torch.Tensor,
              softcap: float = 0.0, stage: int = 0,
              x2: torch.Tensor | None = None, g: torch.Tensor | None = None,
              act: int = 0, eps: float = 1e-5, nt: int = 1, rpw: int = 1,
-             maxblocks: int = 0, g2: torch.Tensor | None = None):
+             maxblocks: int = 0, g2: torch.Tensor | None = None,
+             escale: float = 1.0):
     """y[N] = scales * (Wq[N,K] @ stage(x)); Wq = e4m3fn bytes."""
     N, K = Wq.shape
     out_f32 = 1 if y.dtype == torch.float32 else 0
     _check(lib().launch_gemv_fp8(
         _ptr(Wq), _ptr(scales), _ptr(x), _ptr(x2), _ptr(g), _ptr(g2),
         _ptr(y), _ptr(res), N, K, stage, act, ctypes.c_float(eps), out_f32,
-        ctypes.c_float(softcap), nt, rpw, maxblocks, _stream()), "gemv_fp8")
+        ctypes.c_float(softcap), nt, rpw, maxblocks,
+        ctypes.c_float(escale), _stream()), "gemv_fp8")
 
 
 def rmsnorm(x: torch.Tensor, g: torch.Tensor, y: torch.Tensor,
@@ -233,17 +243,31 @@ def sample(logits: torch.Tensor, min_p: float, greedy: bool, seed: int,
            ctr: torch.Tensor, gmax: torch.Tensor, pick: torch.Tensor,
            next_token: torch.Tensor, out_ring: torch.Tensor,
            nout: torch.Tensor, len_ptr: torch.Tensor,
-           bump_len: bool = True, temperature: float = 1.0):
-    """min-p / greedy sampler: parallel max + Gumbel-argmax + commit.
-    gmax/pick are u64 scratch (zeroed once; the commit kernel re-zeros).
+           bump_len: bool = True, temperature: float = 1.0,
+           cnt: torch.Tensor | None = None):
+    """min-p / greedy sampler: parallel max + Gumbel-argmax; the
+    last-arriving pick block commits the winner (no 1-thread fin
+    launch).  gmax/pick are u64 scratch (zeroed once; the commit path
+    re-zeros); cnt is an i32 ticket counter (zeroed once, re-armed).
     Temperature scales device-side (min-p keep-set + Gumbel score), so
     the GPU fast path matches the CPU sample_token() semantics."""
     V = logits.shape[-1]
+    if cnt is None:
+        global _SAMPLE_CNT
+        try:
+            _SAMPLE_CNT
+        except NameError:
+            _SAMPLE_CNT = {}
+        key = logits.device
+        if key not in _SAMPLE_CNT:
+            _SAMPLE_CNT[key] = torch.zeros(1, dtype=torch.int32,
+                                           device=logits.device)
+        cnt = _SAMPLE_CNT[key]
     inv_temp = 1.0 / max(float(temperature), 1e-6)
     _check(lib().launch_sample(
         _ptr(logits), V, ctypes.c_float(min_p), 1 if greedy else 0,
         ctypes.c_uint64(seed), ctypes.c_float(inv_temp),
-        _ptr(ctr), _ptr(gmax), _ptr(pick),
+        _ptr(ctr), _ptr(gmax), _ptr(pick), _ptr(cnt),
         _ptr(next_token), _ptr(out_ring), _ptr(nout), _ptr(len_ptr),
         1 if bump_len else 0, _stream()), "sample")
 
