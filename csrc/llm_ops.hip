@@ -1924,6 +1924,79 @@ extern "C" hipError_t launch_addinto(void* y, const void* a, long total,
 }
 
 // ====================================================================
+// Grid-wide barrier (device-side, graph-replay-safe).  Used by the
+// fused-layer decode path: phases of one layer run inside ONE kernel
+// with barriers instead of kernel boundaries (~4 us launch/ramp floor
+// each — profiles/decode_kernels_r01.md).
+//
+// Protocol: `cnt` is a MONOTONIC u64 arrive counter (never reset — the
+// launch invariant is cnt % nblocks == 0, preserved because every
+// launch/replay runs the same barrier count with the same grid);
+// `seq` is the completed-barrier counter.  Each block reads base_seq
+// once at kernel start (stable: seq cannot change until every block
+// has arrived at barrier 0, which implies every block already read it)
+// and spins for seq > base_seq + bar_idx.  Bounded spin -> err flag,
+// not a hung GPU.
+// ====================================================================
+
+DEVINL int grid_barrier(unsigned long long* cnt, unsigned long long* seq,
+                        unsigned long long base_seq, int bar_idx,
+                        int nblocks, long spin_limit,
+                        uint32_t* err) {
+  __shared__ int ok;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    ok = 1;
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    unsigned long long t = __hip_atomic_fetch_add(
+        cnt, 1ull, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+    if (t % (unsigned long long)nblocks ==
+        (unsigned long long)nblocks - 1ull) {
+      __hip_atomic_store(seq, base_seq + bar_idx + 1, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_AGENT);
+    } else {
+      long spins = 0;
+      while (__hip_atomic_load(seq, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT) <=
+             base_seq + bar_idx) {
+        if (++spins > spin_limit) {
+          __hip_atomic_store(err, 1u, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
+          ok = 0;
+          break;
+        }
+        __builtin_amdgcn_s_sleep(1);
+      }
+    }
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  }
+  __syncthreads();
+  return ok;
+}
+
+// microbenchmark: nbar back-to-back barriers (measures the phase-
+// boundary cost that replaces a kernel launch in the fused path)
+extern "C" __global__ void __launch_bounds__(256)
+k_gbar_bench(unsigned long long* cnt, unsigned long long* seq,
+             uint32_t* err, int nbar, long spin_limit) {
+  const unsigned long long base =
+      __hip_atomic_load(seq, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  for (int b = 0; b < nbar; b++)
+    if (!grid_barrier(cnt, seq, base, b, gridDim.x, spin_limit, err))
+      return;
+}
+
+extern "C" hipError_t launch_gbar_bench(void* cnt, void* seq, void* err,
+                                        int nblocks, int nbar,
+                                        long spin_limit,
+                                        hipStream_t stream) {
+  hipLaunchKernelGGL(k_gbar_bench, dim3(nblocks), dim3(256), 0, stream,
+                     (unsigned long long*)cnt, (unsigned long long*)seq,
+                     (uint32_t*)err, nbar, spin_limit);
+  return hipGetLastError();
+}
+
+// ====================================================================
 // Utility: device-side int32 set/add (for seq-length bookkeeping inside
 // graphs where needed)
 // ====================================================================
