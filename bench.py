@@ -35,6 +35,10 @@ def main():
                     choices=["bf16", "fp8"],
                     help="weight dtype (constant across N for honest "
                          "scaling; fp8 = BASELINE config 4)")
+    ap.add_argument("--kv-dtype", type=str, default="bf16",
+                    choices=["bf16", "fp8"],
+                    help="KV-pool dtype (fp8 = e4m3 with per-head-pos "
+                         "scales; halves the attention read stream)")
     ap.add_argument("--no-graph", action="store_true")
     args = ap.parse_args()
 
@@ -60,7 +64,8 @@ def main():
         cfg.max_position_embeddings,
         args.prompt_len + args.steps + args.warmup + 64)
     w = LazyRandomWeights(cfg, seed=0)
-    model = GPUModel(cfg, w, max_seq=max_seq, seed=0, dtype=dtype)
+    model = GPUModel(cfg, w, max_seq=max_seq, seed=0, dtype=dtype,
+                     kv_dtype=args.kv_dtype)
 
     rng = np.random.default_rng(0)
     prompt = rng.integers(0, cfg.vocab_size, size=args.prompt_len)
@@ -127,6 +132,7 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": model_name,
+                "kv_dtype": args.kv_dtype,
                 "global_batch": 1,
                 "seq_len": args.prompt_len + args.warmup + args.steps,
                 "parallelism": f"tp{world}",

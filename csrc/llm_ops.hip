@@ -435,8 +435,7 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
 // HW dequant: v_cvt_pk_f32_fp8 (gfx950, OCP not fnuz).
 // ====================================================================
 
-typedef float f2v __attribute__((ext_vector_type(2)));
-typedef uint32_t u4v __attribute__((ext_vector_type(4)));
+typedef uint32_t u4v __attribute__((ext_vector_type(4)));  // = u4v_ alias
 
 DEVINL void fp8x16_to_f32(u4v w, float* o) {
 #pragma unroll
@@ -682,16 +681,101 @@ k_rope_cache(u16* __restrict__ q, const u16* __restrict__ kin,
   }
 }
 
+// fp8 (e4m3) KV-pool variant: K/V stored as bytes with one fp32 scale
+// per (kv_head, position) — absmax/448 over the head_dim values, the
+// same scheme as the fp8 weights.  Halves the attention read stream
+// (the long-context decode bound) at exact-representable dequant.
+extern "C" __global__ void __launch_bounds__(256)
+k_rope_cache_fp8(u16* __restrict__ q, const u16* __restrict__ kin,
+                 const u16* __restrict__ vin, uint8_t* __restrict__ kc,
+                 uint8_t* __restrict__ vc, float* __restrict__ kS,
+                 float* __restrict__ vS, const float* __restrict__ cost,
+                 const float* __restrict__ sint,
+                 const int* __restrict__ pos_ptr,
+                 int nh, int kvh, int hd, int S) {
+  __shared__ float row[256];
+  __shared__ float red[4];
+  const int m = blockIdx.x;
+  const int pos = *pos_ptr + m;
+  const int hd2 = hd / 2;
+  const float* cp = cost + (size_t)pos * hd2;
+  const float* sp = sint + (size_t)pos * hd2;
+  const int tid = threadIdx.x;
+
+  // q rotation in place (q stays bf16)
+  u16* qr = q + (size_t)m * nh * hd;
+  for (int idx = tid; idx < nh * hd2; idx += 256) {
+    int h = idx / hd2, i = idx % hd2;
+    float x1 = b2f(qr[h * hd + i]);
+    float x2 = b2f(qr[h * hd + i + hd2]);
+    float c = cp[i], s = sp[i];
+    qr[h * hd + i] = f2b(x1 * c - x2 * s);
+    qr[h * hd + i + hd2] = f2b(x2 * c + x1 * s);
+  }
+
+  const u16* kr = kin + (size_t)m * kvh * hd;
+  const u16* vr = vin + (size_t)m * kvh * hd;
+  for (int h = 0; h < kvh; h++) {
+    // rotated k row -> LDS floats
+    __syncthreads();
+    if (tid < hd2) {
+      float x1 = b2f(kr[h * hd + tid]);
+      float x2 = b2f(kr[h * hd + tid + hd2]);
+      float c = cp[tid], s = sp[tid];
+      row[tid] = x1 * c - x2 * s;
+      row[tid + hd2] = x2 * c + x1 * s;
+    }
+    __syncthreads();
+    float am = 0.f;
+    for (int i = tid; i < hd; i += 256) am = fmaxf(am, fabsf(row[i]));
+#pragma unroll
+    for (int w = 1; w < 64; w <<= 1) am = fmaxf(am, __shfl_xor(am, w));
+    if ((tid & 63) == 0) red[tid >> 6] = am;
+    __syncthreads();
+    am = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+    float sc = fmaxf(am, 1e-8f) / 448.0f;
+    if (tid == 0) kS[(size_t)h * S + pos] = sc;
+    if (tid * 8 < hd)
+      *(unsigned long long*)(kc + ((size_t)h * S + pos) * hd + tid * 8) =
+          f32x8_to_fp8(&row[tid * 8], 1.0f / sc);
+    // v row (no rotation)
+    __syncthreads();
+    for (int i = tid; i < hd; i += 256) row[i] = b2f(vr[h * hd + i]);
+    __syncthreads();
+    am = 0.f;
+    for (int i = tid; i < hd; i += 256) am = fmaxf(am, fabsf(row[i]));
+#pragma unroll
+    for (int w = 1; w < 64; w <<= 1) am = fmaxf(am, __shfl_xor(am, w));
+    if ((tid & 63) == 0) red[tid >> 6] = am;
+    __syncthreads();
+    am = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+    sc = fmaxf(am, 1e-8f) / 448.0f;
+    if (tid == 0) vS[(size_t)h * S + pos] = sc;
+    if (tid * 8 < hd)
+      *(unsigned long long*)(vc + ((size_t)h * S + pos) * hd + tid * 8) =
+          f32x8_to_fp8(&row[tid * 8], 1.0f / sc);
+  }
+}
+
 extern "C" hipError_t launch_rope_cache(void* q, const void* kin,
                                         const void* vin, void* kc, void* vc,
+                                        void* kS, void* vS, int kv8,
                                         const void* cost, const void* sint,
                                         const void* pos_ptr, int M, int nh,
                                         int kvh, int hd, int S,
                                         hipStream_t stream) {
-  hipLaunchKernelGGL(k_rope_cache, dim3(M), dim3(256), 0, stream, (u16*)q,
-                     (const u16*)kin, (const u16*)vin, (u16*)kc, (u16*)vc,
-                     (const float*)cost, (const float*)sint,
-                     (const int*)pos_ptr, nh, kvh, hd, S);
+  if (kv8) {
+    hipLaunchKernelGGL(k_rope_cache_fp8, dim3(M), dim3(256), 0, stream,
+                       (u16*)q, (const u16*)kin, (const u16*)vin,
+                       (uint8_t*)kc, (uint8_t*)vc, (float*)kS, (float*)vS,
+                       (const float*)cost, (const float*)sint,
+                       (const int*)pos_ptr, nh, kvh, hd, S);
+  } else {
+    hipLaunchKernelGGL(k_rope_cache, dim3(M), dim3(256), 0, stream, (u16*)q,
+                       (const u16*)kin, (const u16*)vin, (u16*)kc, (u16*)vc,
+                       (const float*)cost, (const float*)sint,
+                       (const int*)pos_ptr, nh, kvh, hd, S);
+  }
   return hipGetLastError();
 }
 
@@ -704,11 +788,13 @@ extern "C" hipError_t launch_rope_cache(void* q, const void* kin,
 // (reference omitted both).
 // ====================================================================
 
-extern "C" __global__ void __launch_bounds__(256)
-k_attn(const u16* __restrict__ q, const u16* __restrict__ kc,
-       const u16* __restrict__ vc, u16* __restrict__ out,
-       const int* __restrict__ len_ptr, int nh, int kvh, int hd, int S,
-       float scale, float softcap, int window) {
+template <bool KV8>
+__global__ void __launch_bounds__(256)
+k_attn_t(const u16* __restrict__ q, const void* __restrict__ kc,
+         const void* __restrict__ vc, u16* __restrict__ out,
+         const int* __restrict__ len_ptr, const float* __restrict__ kS,
+         const float* __restrict__ vS, int nh, int kvh, int hd, int S,
+         float scale, float softcap, int window) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* red = (float*)smem;  // [4][hd] acc + [4][2] m,l
 
@@ -732,8 +818,12 @@ k_attn(const u16* __restrict__ q, const u16* __restrict__ kc,
     for (int j = 0; j < 8; j++) qf[j] = b2f(((u16*)&v)[j]);
   }
 
-  const u16* K0 = kc + (size_t)kvhead * S * hd;
-  const u16* V0 = vc + (size_t)kvhead * S * hd;
+  const u16* K0 = (const u16*)kc + (size_t)kvhead * S * hd;
+  const u16* V0 = (const u16*)vc + (size_t)kvhead * S * hd;
+  const uint8_t* K08 = (const uint8_t*)kc + (size_t)kvhead * S * hd;
+  const uint8_t* V08 = (const uint8_t*)vc + (size_t)kvhead * S * hd;
+  const float* kS0 = KV8 ? kS + (size_t)kvhead * S : nullptr;
+  const float* vS0 = KV8 ? vS + (size_t)kvhead * S : nullptr;
 
   float mrun = -INFINITY, lrun = 0.f, acc[8];
 #pragma unroll
@@ -745,35 +835,55 @@ k_attn(const u16* __restrict__ q, const u16* __restrict__ kc,
       int t = t0 + p;
       bool valid = t < T;
       int tl = valid ? t : start;
-      s8v kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
-      s8v vv = *(const s8v*)(V0 + (size_t)tl * hd + d0);
+      s8v kv, vv;
+      float ksc = 1.f, vsc = 1.f;
+      if (KV8) {
+        kv = fp8x8_to_bf16(
+            *(const unsigned long long*)(K08 + (size_t)tl * hd + d0));
+        vv = fp8x8_to_bf16(
+            *(const unsigned long long*)(V08 + (size_t)tl * hd + d0));
+        ksc = kS0[tl]; vsc = vS0[tl];
+      } else {
+        kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
+        vv = *(const s8v*)(V0 + (size_t)tl * hd + d0);
+      }
       for (; t0 < T; t0 += 4 * PP) {
         int t0n = t0 + 4 * PP;
         s8v kvn, vvn;
+        float kscn = 1.f, vscn = 1.f;
         bool validn = false;
         if (t0n < T) {
           int tn = t0n + p;
           validn = tn < T;
           int tln = validn ? tn : start;
-          kvn = *(const s8v*)(K0 + (size_t)tln * hd + d0);
-          vvn = *(const s8v*)(V0 + (size_t)tln * hd + d0);
+          if (KV8) {
+            kvn = fp8x8_to_bf16(
+                *(const unsigned long long*)(K08 + (size_t)tln * hd + d0));
+            vvn = fp8x8_to_bf16(
+                *(const unsigned long long*)(V08 + (size_t)tln * hd + d0));
+            kscn = kS0[tln]; vscn = vS0[tln];
+          } else {
+            kvn = *(const s8v*)(K0 + (size_t)tln * hd + d0);
+            vvn = *(const s8v*)(V0 + (size_t)tln * hd + d0);
+          }
         }
         float partial = 0.f;
 #pragma unroll
         for (int j = 0; j < 8; j++) partial += qf[j] * b2f(((u16*)&kv)[j]);
         for (int w = 1; w < LP; w <<= 1) partial += __shfl_xor(partial, w);
-        float score = partial * scale;
+        float score = partial * (KV8 ? ksc : 1.f) * scale;
         if (softcap > 0.f) score = softcap * tanhf(score / softcap);
         if (!valid) score = -INFINITY;
         float mnew = fmaxf(mrun, score);
         float alpha = (mnew == -INFINITY) ? 0.f : __expf(mrun - mnew);
         float pv = (mnew == -INFINITY) ? 0.f : __expf(score - mnew);
         lrun = lrun * alpha + pv;
+        float pvv = KV8 ? pv * vsc : pv;
 #pragma unroll
         for (int j = 0; j < 8; j++)
-          acc[j] = acc[j] * alpha + pv * b2f(((u16*)&vv)[j]);
+          acc[j] = acc[j] * alpha + pvv * b2f(((u16*)&vv)[j]);
         if (mnew != -INFINITY) mrun = mnew;
-        kv = kvn; vv = vvn; valid = validn;
+        kv = kvn; vv = vvn; valid = validn; ksc = kscn; vsc = vscn;
       }
     }
   }
@@ -836,13 +946,15 @@ k_attn(const u16* __restrict__ q, const u16* __restrict__ kc,
 // Gemma-2 semantics included: sliding window + attn-logit softcap.
 // ====================================================================
 
-extern "C" __global__ void __launch_bounds__(256)
-k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
-           u16* __restrict__ vc, u16* __restrict__ out,
-           const int* __restrict__ len_ptr, const float* __restrict__ cost,
-           const float* __restrict__ sint, float* __restrict__ scratch,
-           int* __restrict__ cnt, int nh, int kvh, int hd, int S,
-           float scale, float softcap, int window) {
+template <bool KV8>
+__global__ void __launch_bounds__(256)
+k_attn_dec_t(const u16* __restrict__ qkv, void* __restrict__ kc,
+             void* __restrict__ vc, u16* __restrict__ out,
+             const int* __restrict__ len_ptr, const float* __restrict__ cost,
+             const float* __restrict__ sint, float* __restrict__ kS,
+             float* __restrict__ vS, float* __restrict__ scratch,
+             int* __restrict__ cnt, int nh, int kvh, int hd, int S,
+             float scale, float softcap, int window) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* red = (float*)smem;              // [4][hd] + [4][2] + flag
   int* lastflag = (int*)(red + 4 * hd + 8);
@@ -896,16 +1008,44 @@ k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
       vn[j] = b2f(((u16*)&va)[j]);
     }
   }
-  if (last_chunk && h == kvhead * grp && wave == 0 && p == 0) {
-    u16 ko[8], vo[8];
-#pragma unroll
-    for (int j = 0; j < 8; j++) { ko[j] = f2b(kn[j]); vo[j] = f2b(vn[j]); }
-    *(s8v*)(kc + ((size_t)kvhead * S + pos) * hd + d0) = *(s8v*)ko;
-    *(s8v*)(vc + ((size_t)kvhead * S + pos) * hd + d0) = *(s8v*)vo;
-  }
+  const u16* K0 = (const u16*)kc + (size_t)kvhead * S * hd;
+  const u16* V0 = (const u16*)vc + (size_t)kvhead * S * hd;
+  const uint8_t* K08 = (const uint8_t*)kc + (size_t)kvhead * S * hd;
+  const uint8_t* V08 = (const uint8_t*)vc + (size_t)kvhead * S * hd;
+  const float* kS0 = KV8 ? kS + (size_t)kvhead * S : nullptr;
+  const float* vS0 = KV8 ? vS + (size_t)kvhead * S : nullptr;
 
-  const u16* K0 = kc + (size_t)kvhead * S * hd;
-  const u16* V0 = vc + (size_t)kvhead * S * hd;
+  if (last_chunk && h == kvhead * grp && wave == 0 && p == 0) {
+    if (KV8) {
+      // per-(head,pos) absmax across the LP lanes (each holds 8 dims)
+      float kam = 0.f, vam = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        kam = fmaxf(kam, fabsf(kn[j]));
+        vam = fmaxf(vam, fabsf(vn[j]));
+      }
+      for (int w = 1; w < LP; w <<= 1) {
+        kam = fmaxf(kam, __shfl_xor(kam, w));
+        vam = fmaxf(vam, __shfl_xor(vam, w));
+      }
+      float ksc = fmaxf(kam, 1e-8f) / 448.0f;
+      float vsc = fmaxf(vam, 1e-8f) / 448.0f;
+      *(unsigned long long*)((uint8_t*)kc +
+          ((size_t)kvhead * S + pos) * hd + d0) = f32x8_to_fp8(kn, 1.f / ksc);
+      *(unsigned long long*)((uint8_t*)vc +
+          ((size_t)kvhead * S + pos) * hd + d0) = f32x8_to_fp8(vn, 1.f / vsc);
+      if (lane == 0) {
+        kS[(size_t)kvhead * S + pos] = ksc;
+        vS[(size_t)kvhead * S + pos] = vsc;
+      }
+    } else {
+      u16 ko[8], vo[8];
+#pragma unroll
+      for (int j = 0; j < 8; j++) { ko[j] = f2b(kn[j]); vo[j] = f2b(vn[j]); }
+      *(s8v*)((u16*)kc + ((size_t)kvhead * S + pos) * hd + d0) = *(s8v*)ko;
+      *(s8v*)((u16*)vc + ((size_t)kvhead * S + pos) * hd + d0) = *(s8v*)vo;
+    }
+  }
 
   float mrun = -INFINITY, lrun = 0.f, acc[8];
 #pragma unroll
@@ -920,35 +1060,55 @@ k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
       int t = t0 + p;
       bool valid = t < c1;
       int tl = valid ? t : c0;
-      s8v kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
-      s8v vv = *(const s8v*)(V0 + (size_t)tl * hd + d0);
+      s8v kv, vv;
+      float ksc = 1.f, vsc = 1.f;
+      if (KV8) {
+        kv = fp8x8_to_bf16(
+            *(const unsigned long long*)(K08 + (size_t)tl * hd + d0));
+        vv = fp8x8_to_bf16(
+            *(const unsigned long long*)(V08 + (size_t)tl * hd + d0));
+        ksc = kS0[tl]; vsc = vS0[tl];
+      } else {
+        kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
+        vv = *(const s8v*)(V0 + (size_t)tl * hd + d0);
+      }
       for (; t0 < c1; t0 += 4 * PP) {
         int t0n = t0 + 4 * PP;
         s8v kvn, vvn;
+        float kscn = 1.f, vscn = 1.f;
         bool validn = false;
         if (t0n < c1) {
           int tn = t0n + p;
           validn = tn < c1;
           int tln = validn ? tn : c0;
-          kvn = *(const s8v*)(K0 + (size_t)tln * hd + d0);
-          vvn = *(const s8v*)(V0 + (size_t)tln * hd + d0);
+          if (KV8) {
+            kvn = fp8x8_to_bf16(
+                *(const unsigned long long*)(K08 + (size_t)tln * hd + d0));
+            vvn = fp8x8_to_bf16(
+                *(const unsigned long long*)(V08 + (size_t)tln * hd + d0));
+            kscn = kS0[tln]; vscn = vS0[tln];
+          } else {
+            kvn = *(const s8v*)(K0 + (size_t)tln * hd + d0);
+            vvn = *(const s8v*)(V0 + (size_t)tln * hd + d0);
+          }
         }
         float partial = 0.f;
 #pragma unroll
         for (int j = 0; j < 8; j++) partial += qf[j] * b2f(((u16*)&kv)[j]);
         for (int w = 1; w < LP; w <<= 1) partial += __shfl_xor(partial, w);
-        float score = partial * scale;
+        float score = partial * (KV8 ? ksc : 1.f) * scale;
         if (softcap > 0.f) score = softcap * tanhf(score / softcap);
         if (!valid) score = -INFINITY;
         float mnew = fmaxf(mrun, score);
         float alpha = (mnew == -INFINITY) ? 0.f : __expf(mrun - mnew);
         float pv = (mnew == -INFINITY) ? 0.f : __expf(score - mnew);
         lrun = lrun * alpha + pv;
+        float pvv = KV8 ? pv * vsc : pv;
 #pragma unroll
         for (int j = 0; j < 8; j++)
-          acc[j] = acc[j] * alpha + pv * b2f(((u16*)&vv)[j]);
+          acc[j] = acc[j] * alpha + pvv * b2f(((u16*)&vv)[j]);
         if (mnew != -INFINITY) mrun = mnew;
-        kv = kvn; vv = vvn; valid = validn;
+        kv = kvn; vv = vvn; valid = validn; ksc = kscn; vsc = vscn;
       }
     }
   }
@@ -1076,16 +1236,22 @@ k_attn_dec(const u16* __restrict__ qkv, u16* __restrict__ kc,
 extern "C" hipError_t launch_attn_dec(const void* qkv, void* kc, void* vc,
                                       void* out, const void* len_ptr,
                                       const void* cost, const void* sint,
+                                      void* kS, void* vS, int kv8,
                                       void* scratch, void* cnt, int split,
                                       int nh, int kvh, int hd, int S,
                                       float scale, float softcap, int window,
                                       hipStream_t stream) {
   size_t lds = (4 * hd + 8) * sizeof(float) + 16;
-  hipLaunchKernelGGL(k_attn_dec, dim3(nh, split), dim3(256), lds, stream,
-                     (const u16*)qkv, (u16*)kc, (u16*)vc, (u16*)out,
-                     (const int*)len_ptr, (const float*)cost,
-                     (const float*)sint, (float*)scratch, (int*)cnt, nh, kvh,
-                     hd, S, scale, softcap, window);
+#define ATTN_DEC_CASE(KV8V)                                                  \
+  hipLaunchKernelGGL((k_attn_dec_t<KV8V>), dim3(nh, split), dim3(256), lds, \
+                     stream, (const u16*)qkv, kc, vc, (u16*)out,            \
+                     (const int*)len_ptr, (const float*)cost,               \
+                     (const float*)sint, (float*)kS, (float*)vS,            \
+                     (float*)scratch, (int*)cnt, nh, kvh, hd, S, scale,     \
+                     softcap, window)
+  if (kv8) ATTN_DEC_CASE(true);
+  else ATTN_DEC_CASE(false);
+#undef ATTN_DEC_CASE
   return hipGetLastError();
 }
 
@@ -1101,11 +1267,13 @@ extern "C" hipError_t launch_attn_dec(const void* qkv, void* kc, void* vc,
 
 typedef short b4v __attribute__((ext_vector_type(4)));
 
-template <int HD>
+template <int HD, bool KV8>
 __global__ void __launch_bounds__(64)
-k_attn_prefill_mfma(const u16* __restrict__ q, const u16* __restrict__ kc,
-                    const u16* __restrict__ vc, u16* __restrict__ out,
-                    const int* __restrict__ len_ptr, int M, int nh, int kvh,
+k_attn_prefill_mfma(const u16* __restrict__ q, const void* __restrict__ kc,
+                    const void* __restrict__ vc, u16* __restrict__ out,
+                    const int* __restrict__ len_ptr,
+                    const float* __restrict__ kS,
+                    const float* __restrict__ vS, int M, int nh, int kvh,
                     int S, float scale, float softcap, int window) {
   constexpr int HD32 = HD / 32;
   constexpr int HD16 = HD / 16;
@@ -1131,8 +1299,12 @@ k_attn_prefill_mfma(const u16* __restrict__ q, const u16* __restrict__ kc,
       qf[c] = *(const s8v*)(qp + c * 32 + (lane >> 4) * 8);
   }
 
-  const u16* K0 = kc + (size_t)kvhead * S * HD;
-  const u16* V0 = vc + (size_t)kvhead * S * HD;
+  const u16* K0 = (const u16*)kc + (size_t)kvhead * S * HD;
+  const u16* V0 = (const u16*)vc + (size_t)kvhead * S * HD;
+  const uint8_t* K08 = (const uint8_t*)kc + (size_t)kvhead * S * HD;
+  const uint8_t* V08 = (const uint8_t*)vc + (size_t)kvhead * S * HD;
+  const float* kS0 = KV8 ? kS + (size_t)kvhead * S : nullptr;
+  const float* vS0 = KV8 ? vS + (size_t)kvhead * S : nullptr;
 
   float mrun = -INFINITY, lrun = 0.f;
   f4v acc_o[HD16];
@@ -1153,20 +1325,28 @@ k_attn_prefill_mfma(const u16* __restrict__ q, const u16* __restrict__ kc,
     {
       int tk = t0 + (lane & 15);
       int tkl = tk < T_end ? tk : T_end - 1;
-      const u16* kp = K0 + (size_t)tkl * HD;
 #pragma unroll
       for (int c = 0; c < HD32; c++) {
-        s8v kf = *(const s8v*)(kp + c * 32 + (lane >> 4) * 8);
+        s8v kf;
+        if (KV8)
+          kf = fp8x8_to_bf16(*(const unsigned long long*)(
+              K08 + (size_t)tkl * HD + c * 32 + (lane >> 4) * 8));
+        else
+          kf = *(const s8v*)(K0 + (size_t)tkl * HD + c * 32 +
+                             (lane >> 4) * 8);
         st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf, qf[c], st, 0, 0, 0);
       }
     }
     // scale + softcap + causal/window mask; element (kv = t0+krow4+r, qcol)
+    // KV8: each S^T output ROW is one kv position -> its k-scale folds
+    // here, after the MFMA
     float sv[4];
 #pragma unroll
     for (int r = 0; r < 4; r++) {
-      float s = st[r] * scale;
-      if (softcap > 0.f) s = softcap * tanhf(s / softcap);
       int t = t0 + krow4 + r;
+      int tcl = t < T_end ? t : T_end - 1;
+      float s = st[r] * (KV8 ? kS0[tcl] : 1.f) * scale;
+      if (softcap > 0.f) s = softcap * tanhf(s / softcap);
       bool bad = (t > qpos) || (t >= T_end) ||
                  (window > 0 && t <= qpos - window);
       sv[r] = bad ? -INFINITY : s;
@@ -1183,7 +1363,16 @@ k_attn_prefill_mfma(const u16* __restrict__ q, const u16* __restrict__ kc,
     for (int r = 0; r < 4; r++) {
       float p = (mnew == -INFINITY) ? 0.f : __expf(sv[r] - mnew);
       psum += p;
-      pb[r] = f2b(p);
+      // KV8: the v-scale of kv position t folds into the P fragment
+      // (the PV MFMA's K axis mixes 16 positions, so it cannot fold
+      // after the matmul)
+      if (KV8) {
+        int t = t0 + krow4 + r;
+        int tcl = t < T_end ? t : T_end - 1;
+        pb[r] = f2b(p * vS0[tcl]);
+      } else {
+        pb[r] = f2b(p);
+      }
     }
     psum += __shfl_xor(psum, 16);
     psum += __shfl_xor(psum, 32);
@@ -1200,9 +1389,16 @@ k_attn_prefill_mfma(const u16* __restrict__ q, const u16* __restrict__ kc,
       int tv = t0 + vrow;
       int tvl = tv < T_end ? tv : T_end - 1;
 #pragma unroll
-      for (int c = 0; c < HD / 32; c++)    // lane covers 8 cols per c
-        *(s8v*)(&vlds[vrow * HD + c * 32 + (lane & 3) * 8]) =
-            *(const s8v*)(V0 + (size_t)tvl * HD + c * 32 + (lane & 3) * 8);
+      for (int c = 0; c < HD / 32; c++) {  // lane covers 8 cols per c
+        s8v vf;
+        if (KV8)
+          vf = fp8x8_to_bf16(*(const unsigned long long*)(
+              V08 + (size_t)tvl * HD + c * 32 + (lane & 3) * 8));
+        else
+          vf = *(const s8v*)(V0 + (size_t)tvl * HD + c * 32 +
+                             (lane & 3) * 8);
+        *(s8v*)(&vlds[vrow * HD + c * 32 + (lane & 3) * 8]) = vf;
+      }
     }
     b4v pfrag = *(b4v*)pb;
 #pragma unroll
@@ -1232,17 +1428,19 @@ k_attn_prefill_mfma(const u16* __restrict__ q, const u16* __restrict__ kc,
 
 extern "C" hipError_t launch_attn_prefill_mfma(
     const void* q, const void* kc, const void* vc, void* out,
-    const void* len_ptr, int M, int nh, int kvh, int hd, int S, float scale,
-    float softcap, int window, hipStream_t stream) {
+    const void* len_ptr, const void* kS, const void* vS, int kv8, int M,
+    int nh, int kvh, int hd, int S, float scale, float softcap, int window,
+    hipStream_t stream) {
   dim3 grid(nh, (M + 15) / 16);
-#define APF_CASE(HDV)                                                       \
-  hipLaunchKernelGGL((k_attn_prefill_mfma<HDV>), grid, dim3(64), 0, stream, \
-                     (const u16*)q, (const u16*)kc, (const u16*)vc,         \
-                     (u16*)out, (const int*)len_ptr, M, nh, kvh, S, scale,  \
-                     softcap, window)
-  if (hd == 64) APF_CASE(64);
-  else if (hd == 128) APF_CASE(128);
-  else if (hd == 256) APF_CASE(256);
+#define APF_CASE(HDV, KV8V)                                                 \
+  hipLaunchKernelGGL((k_attn_prefill_mfma<HDV, KV8V>), grid, dim3(64), 0,   \
+                     stream, (const u16*)q, kc, vc, (u16*)out,              \
+                     (const int*)len_ptr, (const float*)kS,                 \
+                     (const float*)vS, M, nh, kvh, S, scale, softcap,       \
+                     window)
+  if (hd == 64) { if (kv8) APF_CASE(64, true); else APF_CASE(64, false); }
+  else if (hd == 128) { if (kv8) APF_CASE(128, true); else APF_CASE(128, false); }
+  else if (hd == 256) { if (kv8) APF_CASE(256, true); else APF_CASE(256, false); }
   else return hipErrorInvalidValue;
 #undef APF_CASE
   return hipGetLastError();
@@ -1250,14 +1448,21 @@ extern "C" hipError_t launch_attn_prefill_mfma(
 
 extern "C" hipError_t launch_attn(const void* q, const void* kc,
                                   const void* vc, void* out,
-                                  const void* len_ptr, int M, int nh, int kvh,
-                                  int hd, int S, float scale, float softcap,
-                                  int window, hipStream_t stream) {
+                                  const void* len_ptr, const void* kS,
+                                  const void* vS, int kv8, int M, int nh,
+                                  int kvh, int hd, int S, float scale,
+                                  float softcap, int window,
+                                  hipStream_t stream) {
   size_t lds = (4 * hd + 8) * sizeof(float);
-  hipLaunchKernelGGL(k_attn, dim3(nh, M), dim3(256), lds, stream,
-                     (const u16*)q, (const u16*)kc, (const u16*)vc, (u16*)out,
-                     (const int*)len_ptr, nh, kvh, hd, S, scale, softcap,
-                     window);
+#define ATTN_CASE(KV8V)                                                     \
+  hipLaunchKernelGGL((k_attn_t<KV8V>), dim3(nh, M), dim3(256), lds, stream, \
+                     (const u16*)q, kc, vc, (u16*)out,                      \
+                     (const int*)len_ptr, (const float*)kS,                 \
+                     (const float*)vS, nh, kvh, hd, S, scale, softcap,      \
+                     window)
+  if (kv8) ATTN_CASE(true);
+  else ATTN_CASE(false);
+#undef ATTN_CASE
   return hipGetLastError();
 }
 

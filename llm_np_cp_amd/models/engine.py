@@ -42,18 +42,25 @@ class DeviceCacheHandle:
 
 class GPUModel:
     def __init__(self, config: ModelConfig, weights: Dict[str, np.ndarray],
-                 dtype: str = "bf16", max_seq: int = 4096,
+                 dtype: str = "bf16", max_seq: Optional[int] = 4096,
                  prefill_chunk: int = 2048, device: Optional[str] = None,
-                 seed: int = 0, force_tp_path: bool = False):
+                 seed: int = 0, force_tp_path: bool = False,
+                 kv_dtype: str = "bf16"):
+        """``max_seq=None`` sizes the KV pool from FREE HBM after the
+        weights land (288 GB per MI355X — SURVEY §5 long-context), capped
+        at config.max_position_embeddings."""
         if not torch.cuda.is_available():
             raise RuntimeError("GPUModel requires a GPU (MI355X)")
         self.config = config
-        self.max_seq = max_seq
         self.seed = seed
         self.dtype = dtype
         if dtype not in ("bf16", "fp8"):
             raise ValueError(f"unsupported dtype {dtype!r}")
         self.fp8 = dtype == "fp8"
+        if kv_dtype not in ("bf16", "fp8"):
+            raise ValueError(f"unsupported kv_dtype {kv_dtype!r}")
+        self.kv_dtype = kv_dtype
+        self.kv8 = kv_dtype == "fp8"  # e4m3 KV pool, per-(head,pos) scales
         self.rank, self.world = tpu.init_distributed()
         # exercise the TP code path on 1 GPU (collectives no-op at
         # world=1, partial sums are then exact) — used by tests
@@ -92,6 +99,9 @@ class GPUModel:
             and "nccl" in str(dist.get_backend()).lower())
 
         self._upload_weights(weights)
+        if max_seq is None:
+            max_seq = self._auto_max_seq()
+        self.max_seq = max_seq
         self._alloc_state(prefill_chunk)
         self._graph = None
         self._graph_mode = None
@@ -107,6 +117,21 @@ class GPUModel:
         self._pf_events = []
 
     # ------------------------------------------------------------------
+    def _auto_max_seq(self) -> int:
+        """KV pool length from free HBM (post-weights): fill the 288 GB
+        MI355X budget instead of a fixed 4k default,
+        capped at the model's max_position_embeddings."""
+        cfg = self.config
+        free, _total = torch.cuda.mem_get_info(self.device)
+        esz = 1 if self.kv8 else 2
+        per_pos = cfg.num_hidden_layers * self.kvh_l * self.hd * 2 * esz
+        if self.kv8:
+            per_pos += cfg.num_hidden_layers * self.kvh_l * 2 * 4  # scales
+        per_pos += self.hd * 4 + 8  # rope tables + ring slot
+        budget = int(free * 0.9) - (2 << 30)  # headroom for scratch/graph
+        n = max(1024, budget // max(per_pos, 1))
+        return int(min(cfg.max_position_embeddings, n))
+
     def _upload_weights(self, w: Dict[str, np.ndarray]):
         cfg, dev = self.config, self.device
         tp, r = self.world, self.rank
@@ -209,8 +234,19 @@ class GPUModel:
         S, hd = self.max_seq, self.hd
         L = cfg.num_hidden_layers
         bf = dict(dtype=torch.bfloat16, device=dev)
-        self.k_cache = [torch.zeros(self.kvh_l, S, hd, **bf) for _ in range(L)]
-        self.v_cache = [torch.zeros(self.kvh_l, S, hd, **bf) for _ in range(L)]
+        kvt = torch.uint8 if self.kv8 else torch.bfloat16
+        self.k_cache = [torch.zeros(self.kvh_l, S, hd, dtype=kvt, device=dev)
+                        for _ in range(L)]
+        self.v_cache = [torch.zeros(self.kvh_l, S, hd, dtype=kvt, device=dev)
+                        for _ in range(L)]
+        if self.kv8:
+            self.k_scale = [torch.zeros(self.kvh_l, S, dtype=torch.float32,
+                                        device=dev) for _ in range(L)]
+            self.v_scale = [torch.zeros(self.kvh_l, S, dtype=torch.float32,
+                                        device=dev) for _ in range(L)]
+        else:
+            self.k_scale = [None] * L
+            self.v_scale = [None] * L
 
         # RoPE tables fp32 (host-precomputed: guide App.B — no device trig)
         inv = cfg.rope_inv_freq()                      # (hd/2,) fp64
@@ -358,18 +394,21 @@ class GPUModel:
             self._linear(lw, "wv", xn, self.b_v, M=M)
             ho.rope_cache(self.b_q, self.b_k, self.b_v, self.k_cache[i],
                           self.v_cache[i], self.cos_t, self.sin_t,
-                          self.len_buf, M, self.nh_l, self.kvh_l, self.hd)
+                          self.len_buf, M, self.nh_l, self.kvh_l, self.hd,
+                          kS=self.k_scale[i], vS=self.v_scale[i])
             if self.hd in (64, 128, 256):
                 ho.attn_prefill_mfma(
                     self.b_q, self.k_cache[i], self.v_cache[i], self.b_att,
                     self.len_buf, M, self.nh_l, self.kvh_l, self.hd,
                     self.scale, softcap=self.attn_softcap,
-                    window=window or 0)
+                    window=window or 0,
+                    kS=self.k_scale[i], vS=self.v_scale[i])
             else:
                 ho.attn(self.b_q, self.k_cache[i], self.v_cache[i],
                         self.b_att, self.len_buf, M, self.nh_l, self.kvh_l,
                         self.hd, self.scale, softcap=self.attn_softcap,
-                        window=window or 0)
+                        window=window or 0,
+                        kS=self.k_scale[i], vS=self.v_scale[i])
             if self.gemma:
                 self._rowpar_ar(lw, "wo", self.b_att, t1, M)
                 ho.rmsnorm(t1[:M], lw["g_post"], h[:M], res=h[:M], eps=eps)
@@ -630,7 +669,8 @@ class GPUModel:
                         self.attn_scratch, self.attn_cnt,
                         self.nh_l, self.kvh_l, self.hd, self.scale,
                         softcap=self.attn_softcap, window=window or 0,
-                        split=self.attn_split)
+                        split=self.attn_split,
+                        kS=self.k_scale[i], vS=self.v_scale[i])
             if self.gemma:
                 self._dgemv(lw, "wo", self.b_att[0], t1)
                 tpu.all_reduce(t1)

@@ -34,15 +34,17 @@ _SIGS = {
                         ctypes.c_float, ctypes.c_void_p],
     "launch_rmsnorm": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 2 +
                       [ctypes.c_float, ctypes.c_int, ctypes.c_void_p],
-    "launch_rope_cache": [ctypes.c_void_p] * 8 + [ctypes.c_int] * 5 +
+    "launch_rope_cache": [ctypes.c_void_p] * 7 + [ctypes.c_int] +
+                         [ctypes.c_void_p] * 3 + [ctypes.c_int] * 5 +
                          [ctypes.c_void_p],
-    "launch_attn": [ctypes.c_void_p] * 5 + [ctypes.c_int] * 5 +
+    "launch_attn": [ctypes.c_void_p] * 7 + [ctypes.c_int] * 6 +
                    [ctypes.c_float, ctypes.c_float, ctypes.c_int,
                     ctypes.c_void_p],
-    "launch_attn_prefill_mfma": [ctypes.c_void_p] * 5 + [ctypes.c_int] * 5 +
+    "launch_attn_prefill_mfma": [ctypes.c_void_p] * 7 + [ctypes.c_int] * 6 +
                                 [ctypes.c_float, ctypes.c_float,
                                  ctypes.c_int, ctypes.c_void_p],
-    "launch_attn_dec": [ctypes.c_void_p] * 9 + [ctypes.c_int] * 5 +
+    "launch_attn_dec": [ctypes.c_void_p] * 9 + [ctypes.c_int] +
+                       [ctypes.c_void_p] * 2 + [ctypes.c_int] * 5 +
                        [ctypes.c_float, ctypes.c_float, ctypes.c_int,
                         ctypes.c_void_p],
     "launch_glu": [ctypes.c_void_p] * 3 + [ctypes.c_long, ctypes.c_int,
@@ -174,10 +176,14 @@ def rmsnorm(x: torch.Tensor, g: torch.Tensor, y: torch.Tensor,
 def rope_cache(q: torch.Tensor, k_in: torch.Tensor, v_in: torch.Tensor,
                k_cache: torch.Tensor, v_cache: torch.Tensor,
                cos_t: torch.Tensor, sin_t: torch.Tensor,
-               pos_ptr: torch.Tensor, M: int, nh: int, kvh: int, hd: int):
+               pos_ptr: torch.Tensor, M: int, nh: int, kvh: int, hd: int,
+               kS: torch.Tensor | None = None,
+               vS: torch.Tensor | None = None):
+    """kS/vS given => fp8 (e4m3) KV pool with per-(head,pos) scales."""
     S = k_cache.shape[1]
     _check(lib().launch_rope_cache(
         _ptr(q), _ptr(k_in), _ptr(v_in), _ptr(k_cache), _ptr(v_cache),
+        _ptr(kS), _ptr(vS), 1 if kS is not None else 0,
         _ptr(cos_t), _ptr(sin_t), _ptr(pos_ptr), M, nh, kvh, hd, S,
         _stream()), "rope_cache")
 
@@ -185,10 +191,12 @@ def rope_cache(q: torch.Tensor, k_in: torch.Tensor, v_in: torch.Tensor,
 def attn(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
          out: torch.Tensor, len_ptr: torch.Tensor, M: int, nh: int,
          kvh: int, hd: int, scale: float, softcap: float = 0.0,
-         window: int = 0):
+         window: int = 0, kS: torch.Tensor | None = None,
+         vS: torch.Tensor | None = None):
     S = k_cache.shape[1]
     _check(lib().launch_attn(
         _ptr(q), _ptr(k_cache), _ptr(v_cache), _ptr(out), _ptr(len_ptr),
+        _ptr(kS), _ptr(vS), 1 if kS is not None else 0,
         M, nh, kvh, hd, S, ctypes.c_float(scale), ctypes.c_float(softcap),
         window, _stream()), "attn")
 
@@ -197,12 +205,14 @@ def attn_prefill_mfma(q: torch.Tensor, k_cache: torch.Tensor,
                       v_cache: torch.Tensor, out: torch.Tensor,
                       len_ptr: torch.Tensor, M: int, nh: int, kvh: int,
                       hd: int, scale: float, softcap: float = 0.0,
-                      window: int = 0):
+                      window: int = 0, kS: torch.Tensor | None = None,
+                      vS: torch.Tensor | None = None):
     """Flash prefill attention (MFMA, online softmax); q already roped,
     caches already written for [0, pos0+M)."""
     S = k_cache.shape[1]
     _check(lib().launch_attn_prefill_mfma(
         _ptr(q), _ptr(k_cache), _ptr(v_cache), _ptr(out), _ptr(len_ptr),
+        _ptr(kS), _ptr(vS), 1 if kS is not None else 0,
         M, nh, kvh, hd, S, ctypes.c_float(scale), ctypes.c_float(softcap),
         window, _stream()), "attn_prefill_mfma")
 
@@ -213,13 +223,17 @@ def attn_dec(qkv: torch.Tensor, k_cache: torch.Tensor,
              sin_t: torch.Tensor, scratch: torch.Tensor,
              cnt: torch.Tensor, nh: int, kvh: int, hd: int,
              scale: float, softcap: float = 0.0, window: int = 0,
-             split: int = 1):
+             split: int = 1, kS: torch.Tensor | None = None,
+             vS: torch.Tensor | None = None):
     """Fused decode attention: RoPE(q,k) + KV write + online softmax,
-    KV range split over `split` blocks/head (last-arriver merge)."""
+    KV range split over `split` blocks/head (last-arriver merge).
+    kS/vS given => fp8 KV pool (quantized write + dequant scan)."""
     S = k_cache.shape[1]
     _check(lib().launch_attn_dec(
         _ptr(qkv), _ptr(k_cache), _ptr(v_cache), _ptr(out), _ptr(len_ptr),
-        _ptr(cos_t), _ptr(sin_t), _ptr(scratch), _ptr(cnt), split,
+        _ptr(cos_t), _ptr(sin_t), _ptr(kS), _ptr(vS),
+        1 if kS is not None else 0,
+        _ptr(scratch), _ptr(cnt), split,
         nh, kvh, hd, S, ctypes.c_float(scale),
         ctypes.c_float(softcap), window, _stream()), "attn_dec")
 

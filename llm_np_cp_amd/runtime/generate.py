@@ -153,7 +153,8 @@ class ByteTokenizer:
 
 def load_model(model_dir_or_preset: str, backend: str = "auto",
                device: str = "cuda", dtype: str = "bf16",
-               max_seq: int = 4096, seed: int = 0):
+               max_seq: int = 4096, seed: int = 0,
+               kv_dtype: str = "bf16"):
     """Reference-parity entry (``load_model`` -> (tokenizer, model, config),
     ``llama3.2_model.py:1082-1099``).
 
@@ -213,7 +214,8 @@ def load_model(model_dir_or_preset: str, backend: str = "auto",
     elif backend == "gpu":
         from ..models.engine import GPUModel
 
-        model = GPUModel(config, weights, dtype=dtype, max_seq=max_seq)
+        model = GPUModel(config, weights, dtype=dtype, max_seq=max_seq,
+                         kv_dtype=kv_dtype)
     else:
         raise ValueError(f"unknown backend {backend!r}")
     return tokenizer, model, config
