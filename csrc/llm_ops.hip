@@ -836,12 +836,11 @@ k_attn_t(const u16* __restrict__ q, const void* __restrict__ kc,
       bool valid = t < T;
       int tl = valid ? t : start;
       s8v kv, vv;
+      unsigned long long kraw = 0, vraw = 0;
       float ksc = 1.f, vsc = 1.f;
       if (KV8) {
-        kv = fp8x8_to_bf16(
-            *(const unsigned long long*)(K08 + (size_t)tl * hd + d0));
-        vv = fp8x8_to_bf16(
-            *(const unsigned long long*)(V08 + (size_t)tl * hd + d0));
+        kraw = *(const unsigned long long*)(K08 + (size_t)tl * hd + d0);
+        vraw = *(const unsigned long long*)(V08 + (size_t)tl * hd + d0);
         ksc = kS0[tl]; vsc = vS0[tl];
       } else {
         kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
@@ -850,6 +849,7 @@ k_attn_t(const u16* __restrict__ q, const void* __restrict__ kc,
       for (; t0 < T; t0 += 4 * PP) {
         int t0n = t0 + 4 * PP;
         s8v kvn, vvn;
+        unsigned long long krawn = 0, vrawn = 0;
         float kscn = 1.f, vscn = 1.f;
         bool validn = false;
         if (t0n < T) {
@@ -857,10 +857,8 @@ k_attn_t(const u16* __restrict__ q, const void* __restrict__ kc,
           validn = tn < T;
           int tln = validn ? tn : start;
           if (KV8) {
-            kvn = fp8x8_to_bf16(
-                *(const unsigned long long*)(K08 + (size_t)tln * hd + d0));
-            vvn = fp8x8_to_bf16(
-                *(const unsigned long long*)(V08 + (size_t)tln * hd + d0));
+            krawn = *(const unsigned long long*)(K08 + (size_t)tln * hd + d0);
+            vrawn = *(const unsigned long long*)(V08 + (size_t)tln * hd + d0);
             kscn = kS0[tln]; vscn = vS0[tln];
           } else {
             kvn = *(const s8v*)(K0 + (size_t)tln * hd + d0);
@@ -868,8 +866,22 @@ k_attn_t(const u16* __restrict__ q, const void* __restrict__ kc,
           }
         }
         float partial = 0.f;
+        if (KV8) {
+          // dequant straight to f32 in the dot (4 cvt + 8 fma per 8
+          // elems — cheaper than the bf16 path's 8 shifts + 8 fma)
+          f2v c0_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)kraw, false);
+          f2v c1_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)kraw, true);
+          f2v c2_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)(kraw >> 32),
+                                                    false);
+          f2v c3_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)(kraw >> 32),
+                                                    true);
+          partial = qf[0] * c0_[0] + qf[1] * c0_[1] + qf[2] * c1_[0] +
+                    qf[3] * c1_[1] + qf[4] * c2_[0] + qf[5] * c2_[1] +
+                    qf[6] * c3_[0] + qf[7] * c3_[1];
+        } else {
 #pragma unroll
-        for (int j = 0; j < 8; j++) partial += qf[j] * b2f(((u16*)&kv)[j]);
+          for (int j = 0; j < 8; j++) partial += qf[j] * b2f(((u16*)&kv)[j]);
+        }
         for (int w = 1; w < LP; w <<= 1) partial += __shfl_xor(partial, w);
         float score = partial * (KV8 ? ksc : 1.f) * scale;
         if (softcap > 0.f) score = softcap * tanhf(score / softcap);
@@ -878,12 +890,27 @@ k_attn_t(const u16* __restrict__ q, const void* __restrict__ kc,
         float alpha = (mnew == -INFINITY) ? 0.f : __expf(mrun - mnew);
         float pv = (mnew == -INFINITY) ? 0.f : __expf(score - mnew);
         lrun = lrun * alpha + pv;
-        float pvv = KV8 ? pv * vsc : pv;
+        if (KV8) {
+          float pvv = pv * vsc;
+          f2v d0_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)vraw, false);
+          f2v d1_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)vraw, true);
+          f2v d2_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)(vraw >> 32),
+                                                    false);
+          f2v d3_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)(vraw >> 32),
+                                                    true);
+          float vf[8] = {d0_[0], d0_[1], d1_[0], d1_[1],
+                         d2_[0], d2_[1], d3_[0], d3_[1]};
 #pragma unroll
-        for (int j = 0; j < 8; j++)
-          acc[j] = acc[j] * alpha + pvv * b2f(((u16*)&vv)[j]);
+          for (int j = 0; j < 8; j++)
+            acc[j] = acc[j] * alpha + pvv * vf[j];
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; j++)
+            acc[j] = acc[j] * alpha + pv * b2f(((u16*)&vv)[j]);
+        }
         if (mnew != -INFINITY) mrun = mnew;
-        kv = kvn; vv = vvn; valid = validn; ksc = kscn; vsc = vscn;
+        kv = kvn; vv = vvn; kraw = krawn; vraw = vrawn;
+        valid = validn; ksc = kscn; vsc = vscn;
       }
     }
   }
@@ -1061,12 +1088,11 @@ k_attn_dec_t(const u16* __restrict__ qkv, void* __restrict__ kc,
       bool valid = t < c1;
       int tl = valid ? t : c0;
       s8v kv, vv;
+      unsigned long long kraw = 0, vraw = 0;
       float ksc = 1.f, vsc = 1.f;
       if (KV8) {
-        kv = fp8x8_to_bf16(
-            *(const unsigned long long*)(K08 + (size_t)tl * hd + d0));
-        vv = fp8x8_to_bf16(
-            *(const unsigned long long*)(V08 + (size_t)tl * hd + d0));
+        kraw = *(const unsigned long long*)(K08 + (size_t)tl * hd + d0);
+        vraw = *(const unsigned long long*)(V08 + (size_t)tl * hd + d0);
         ksc = kS0[tl]; vsc = vS0[tl];
       } else {
         kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
@@ -1075,6 +1101,7 @@ k_attn_dec_t(const u16* __restrict__ qkv, void* __restrict__ kc,
       for (; t0 < c1; t0 += 4 * PP) {
         int t0n = t0 + 4 * PP;
         s8v kvn, vvn;
+        unsigned long long krawn = 0, vrawn = 0;
         float kscn = 1.f, vscn = 1.f;
         bool validn = false;
         if (t0n < c1) {
@@ -1082,10 +1109,8 @@ k_attn_dec_t(const u16* __restrict__ qkv, void* __restrict__ kc,
           validn = tn < c1;
           int tln = validn ? tn : c0;
           if (KV8) {
-            kvn = fp8x8_to_bf16(
-                *(const unsigned long long*)(K08 + (size_t)tln * hd + d0));
-            vvn = fp8x8_to_bf16(
-                *(const unsigned long long*)(V08 + (size_t)tln * hd + d0));
+            krawn = *(const unsigned long long*)(K08 + (size_t)tln * hd + d0);
+            vrawn = *(const unsigned long long*)(V08 + (size_t)tln * hd + d0);
             kscn = kS0[tln]; vscn = vS0[tln];
           } else {
             kvn = *(const s8v*)(K0 + (size_t)tln * hd + d0);
@@ -1093,8 +1118,22 @@ k_attn_dec_t(const u16* __restrict__ qkv, void* __restrict__ kc,
           }
         }
         float partial = 0.f;
+        if (KV8) {
+          // dequant straight to f32 in the dot (4 cvt + 8 fma per 8
+          // elems — cheaper than the bf16 path's 8 shifts + 8 fma)
+          f2v c0_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)kraw, false);
+          f2v c1_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)kraw, true);
+          f2v c2_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)(kraw >> 32),
+                                                    false);
+          f2v c3_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)(kraw >> 32),
+                                                    true);
+          partial = qf[0] * c0_[0] + qf[1] * c0_[1] + qf[2] * c1_[0] +
+                    qf[3] * c1_[1] + qf[4] * c2_[0] + qf[5] * c2_[1] +
+                    qf[6] * c3_[0] + qf[7] * c3_[1];
+        } else {
 #pragma unroll
-        for (int j = 0; j < 8; j++) partial += qf[j] * b2f(((u16*)&kv)[j]);
+          for (int j = 0; j < 8; j++) partial += qf[j] * b2f(((u16*)&kv)[j]);
+        }
         for (int w = 1; w < LP; w <<= 1) partial += __shfl_xor(partial, w);
         float score = partial * (KV8 ? ksc : 1.f) * scale;
         if (softcap > 0.f) score = softcap * tanhf(score / softcap);
@@ -1103,12 +1142,27 @@ k_attn_dec_t(const u16* __restrict__ qkv, void* __restrict__ kc,
         float alpha = (mnew == -INFINITY) ? 0.f : __expf(mrun - mnew);
         float pv = (mnew == -INFINITY) ? 0.f : __expf(score - mnew);
         lrun = lrun * alpha + pv;
-        float pvv = KV8 ? pv * vsc : pv;
+        if (KV8) {
+          float pvv = pv * vsc;
+          f2v d0_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)vraw, false);
+          f2v d1_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)vraw, true);
+          f2v d2_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)(vraw >> 32),
+                                                    false);
+          f2v d3_ = __builtin_amdgcn_cvt_pk_f32_fp8((uint32_t)(vraw >> 32),
+                                                    true);
+          float vf[8] = {d0_[0], d0_[1], d1_[0], d1_[1],
+                         d2_[0], d2_[1], d3_[0], d3_[1]};
 #pragma unroll
-        for (int j = 0; j < 8; j++)
-          acc[j] = acc[j] * alpha + pvv * b2f(((u16*)&vv)[j]);
+          for (int j = 0; j < 8; j++)
+            acc[j] = acc[j] * alpha + pvv * vf[j];
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; j++)
+            acc[j] = acc[j] * alpha + pv * b2f(((u16*)&vv)[j]);
+        }
         if (mnew != -INFINITY) mrun = mnew;
-        kv = kvn; vv = vvn; valid = validn; ksc = kscn; vsc = vscn;
+        kv = kvn; vv = vvn; kraw = krawn; vraw = vrawn;
+        valid = validn; ksc = kscn; vsc = vscn;
       }
     }
   }
