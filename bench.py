@@ -39,6 +39,9 @@ def main():
                     choices=["bf16", "fp8"],
                     help="KV-pool dtype (fp8 = e4m3 with per-head-pos "
                          "scales; halves the attention read stream)")
+    ap.add_argument("--batch", type=int, default=1,
+                    help="lockstep batched decode (throughput mode; "
+                         "value = aggregate tokens/s over the batch)")
     ap.add_argument("--no-graph", action="store_true")
     args = ap.parse_args()
 
@@ -65,7 +68,7 @@ def main():
         args.prompt_len + args.steps + args.warmup + 64)
     w = LazyRandomWeights(cfg, seed=0)
     model = GPUModel(cfg, w, max_seq=max_seq, seed=0, dtype=dtype,
-                     kv_dtype=args.kv_dtype)
+                     kv_dtype=args.kv_dtype, max_batch=args.batch)
 
     rng = np.random.default_rng(0)
     prompt = rng.integers(0, cfg.vocab_size, size=args.prompt_len)
@@ -75,13 +78,22 @@ def main():
             torch.distributed.barrier()
 
     # prefill + warmup
-    model.prefill(prompt)
     use_graph = not args.no_graph
-    if args.warmup > 0:
-        model.decode(args.warmup, greedy=True, use_graph=use_graph,
-                     first_from_logits=True)
+    if args.batch > 1:
+        prompts = rng.integers(0, cfg.vocab_size,
+                               size=(args.batch, args.prompt_len))
+        model.prefill_batch(prompts)
+        model.decode_batch(max(args.warmup, 2), greedy=True,
+                           use_graph=use_graph)
+        if model._graph_mode is None:
+            use_graph = False
+    else:
+        model.prefill(prompt)
+        if args.warmup > 0:
+            model.decode(args.warmup, greedy=True, use_graph=use_graph,
+                         first_from_logits=True)
 
-    if use_graph:
+    if use_graph and args.batch == 1:
         # no-op if the warmup already captured; on failure the step it
         # executed is harmless here (untimed region)
         _, ok = model.capture_decode_graph(True, 0.1)
@@ -100,6 +112,9 @@ def main():
     if use_graph:
         for _ in range(args.steps):
             model._graph.replay()
+    elif args.batch > 1:
+        for _ in range(args.steps):
+            model._decode_batch_step(args.batch, True, 0.1)
     else:
         for _ in range(args.steps):
             model._decode_step(True, 0.1)
@@ -113,7 +128,7 @@ def main():
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    toks_per_s = args.steps / elapsed
+    toks_per_s = args.steps * args.batch / elapsed
     ms_per_step = elapsed / args.steps * 1e3
 
     if rank == 0:
@@ -133,7 +148,7 @@ def main():
             "config": {
                 "model": model_name,
                 "kv_dtype": args.kv_dtype,
-                "global_batch": 1,
+                "global_batch": args.batch,
                 "seq_len": args.prompt_len + args.warmup + args.steps,
                 "parallelism": f"tp{world}",
             },

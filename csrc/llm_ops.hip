@@ -988,6 +988,16 @@ k_attn_dec_t(const u16* __restrict__ qkv, void* __restrict__ kc,
 
   const int h = blockIdx.x;
   const int chunk = blockIdx.y, SPLIT = gridDim.y;
+  // batch axis (lockstep sequences): per-b qkv/out rows, KV pool,
+  // scales, merge scratch and tickets; the position is shared
+  const int b = blockIdx.z;
+  qkv += (size_t)b * (nh + 2 * kvh) * hd;
+  out += (size_t)b * nh * hd;
+  kc = (char*)kc + (size_t)b * kvh * S * hd * (KV8 ? 1 : 2);
+  vc = (char*)vc + (size_t)b * kvh * S * hd * (KV8 ? 1 : 2);
+  if (KV8) { kS += (size_t)b * kvh * S; vS += (size_t)b * kvh * S; }
+  scratch += (size_t)b * nh * SPLIT * (hd + 2);
+  cnt += b * nh;
   const int grp = nh / kvh;
   const int kvhead = h / grp;
   const int pos = *len_ptr;
@@ -1292,12 +1302,14 @@ extern "C" hipError_t launch_attn_dec(const void* qkv, void* kc, void* vc,
                                       const void* cost, const void* sint,
                                       void* kS, void* vS, int kv8,
                                       void* scratch, void* cnt, int split,
+                                      int batch,
                                       int nh, int kvh, int hd, int S,
                                       float scale, float softcap, int window,
                                       hipStream_t stream) {
   size_t lds = (4 * hd + 8) * sizeof(float) + 16;
 #define ATTN_DEC_CASE(KV8V)                                                  \
-  hipLaunchKernelGGL((k_attn_dec_t<KV8V>), dim3(nh, split), dim3(256), lds, \
+  hipLaunchKernelGGL((k_attn_dec_t<KV8V>), dim3(nh, split, batch),          \
+                     dim3(256), lds,                                        \
                      stream, (const u16*)qkv, kc, vc, (u16*)out,            \
                      (const int*)len_ptr, (const float*)cost,               \
                      (const float*)sint, (float*)kS, (float*)vS,            \
@@ -1617,14 +1629,20 @@ DEVINL uint64_t pack_ki(uint32_t key, int idx) {
 }
 DEVINL int unpack_idx(uint64_t p) { return 0x7fffffff - (int)(uint32_t)p; }
 
-// pass 1 (min-p only): global max logit -> *gmax (packed)
+// pass 1 (min-p only): global max logit -> gmax[b] (packed).
+// Batched: grid.y = sequence row b (lbf16: logits stored bf16 — the
+// batched lm_head GEMM path emits bf16 rows).
 extern "C" __global__ void __launch_bounds__(256)
-k_logit_max(const float* __restrict__ logits, int V,
+k_logit_max(const void* __restrict__ logits, int V, int lbf16,
             unsigned long long* __restrict__ gmax) {
+  const int b = blockIdx.y;
+  gmax += b;
+  const float* lf = (const float*)logits + (size_t)b * V;
+  const u16* lh = (const u16*)logits + (size_t)b * V;
   float mv = -INFINITY;
   int mi = 0;
   for (int i = blockIdx.x * 256 + threadIdx.x; i < V; i += gridDim.x * 256) {
-    float v = logits[i];
+    float v = lbf16 ? b2f(lh[i]) : lf[i];
     if (v > mv || (v == mv && i < mi)) { mv = v; mi = i; }
   }
   uint64_t pk = pack_ki(fkey(mv), mi);
@@ -1651,7 +1669,8 @@ k_logit_max(const float* __restrict__ logits, int V,
 //   p_i(T) >= min_p * p_max(T)  <=>  l_i >= l_max + T*ln(min_p)
 //   Gumbel-argmax over p(T)     <=>  argmax of l_i/T + G_i
 extern "C" __global__ void __launch_bounds__(256)
-k_sample_pick(const float* __restrict__ logits, int V, float min_p,
+k_sample_pick(const void* __restrict__ logits, int V, int lbf16,
+              long ring_stride, float min_p,
               int greedy, uint64_t seed, float inv_temp,
               unsigned long long* __restrict__ gmax,
               uint64_t* __restrict__ ctr,
@@ -1659,14 +1678,19 @@ k_sample_pick(const float* __restrict__ logits, int V, float min_p,
               int* __restrict__ cnt, int* __restrict__ next_token,
               int* __restrict__ out_ring, int* __restrict__ nout,
               int* __restrict__ len_ptr, int bump_len) {
+  const int b = blockIdx.y;
+  gmax += b; pick += b; cnt += b; nout += b;
+  out_ring += (size_t)b * ring_stride;
+  const float* lf = (const float*)logits + (size_t)b * V;
+  const u16* lh = (const u16*)logits + (size_t)b * V;
   float thresh = -INFINITY;
   if (!greedy)
     thresh = fkey_inv((uint32_t)(*gmax >> 32)) + __logf(min_p) / inv_temp;
-  const uint32_t c = (uint32_t)(*ctr);
+  const uint32_t c = (uint32_t)(*ctr) ^ ((uint32_t)b * 0x85ebca6bu);
   float bv = -INFINITY;
   int bi = 0x7fffffff;
   for (int i = blockIdx.x * 256 + threadIdx.x; i < V; i += gridDim.x * 256) {
-    float v = logits[i];
+    float v = lbf16 ? b2f(lh[i]) : lf[i];
     if (v < thresh) continue;
     float sc = v;
     if (!greedy) {
@@ -1704,17 +1728,21 @@ k_sample_pick(const float* __restrict__ logits, int V, float min_p,
   int winner = unpack_idx(
       __hip_atomic_load(pick, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT));
   if (winner < 0 || winner >= V) winner = 0;  // NaN-logit insurance
-  *next_token = winner;
+  next_token[b] = winner;
   int n = *nout;
   out_ring[n] = winner;
   *nout = n + 1;
-  if (bump_len) *len_ptr += 1;
-  if (!greedy) *ctr += 1;
+  // shared device state advanced once per step (row 0's committer)
+  if (b == 0) {
+    if (bump_len) *len_ptr += 1;
+    if (!greedy) *ctr += 1;
+  }
   *pick = 0ull;
   *gmax = 0ull;
 }
 
-extern "C" hipError_t launch_sample(const void* logits, int V, float min_p,
+extern "C" hipError_t launch_sample(const void* logits, int V, int lbf16,
+                                    int batch, long ring_stride, float min_p,
                                     int greedy, uint64_t seed, float inv_temp,
                                     void* ctr, void* gmax, void* pick,
                                     void* cnt, void* next_token,
@@ -1724,10 +1752,11 @@ extern "C" hipError_t launch_sample(const void* logits, int V, float min_p,
   int blocks = (V + 255) / 256;
   if (blocks > 512) blocks = 512;
   if (!greedy)
-    hipLaunchKernelGGL(k_logit_max, dim3(blocks), dim3(256), 0, stream,
-                       (const float*)logits, V, (unsigned long long*)gmax);
-  hipLaunchKernelGGL(k_sample_pick, dim3(blocks), dim3(256), 0, stream,
-                     (const float*)logits, V, min_p, greedy, seed, inv_temp,
+    hipLaunchKernelGGL(k_logit_max, dim3(blocks, batch), dim3(256), 0,
+                       stream, logits, V, lbf16, (unsigned long long*)gmax);
+  hipLaunchKernelGGL(k_sample_pick, dim3(blocks, batch), dim3(256), 0,
+                     stream, logits, V, lbf16, ring_stride, min_p, greedy,
+                     seed, inv_temp,
                      (unsigned long long*)gmax, (uint64_t*)ctr,
                      (unsigned long long*)pick, (int*)cnt, (int*)next_token,
                      (int*)out_ring, (int*)nout, (int*)len_ptr, bump_len);

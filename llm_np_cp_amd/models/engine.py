@@ -45,14 +45,19 @@ class GPUModel:
                  dtype: str = "bf16", max_seq: Optional[int] = 4096,
                  prefill_chunk: int = 2048, device: Optional[str] = None,
                  seed: int = 0, force_tp_path: bool = False,
-                 kv_dtype: str = "bf16"):
+                 kv_dtype: str = "bf16", max_batch: int = 1):
         """``max_seq=None`` sizes the KV pool from FREE HBM after the
         weights land (288 GB per MI355X — SURVEY §5 long-context), capped
-        at config.max_position_embeddings."""
+        at config.max_position_embeddings.  ``max_batch>1`` allocates
+        per-sequence KV pools for lockstep batched decode
+        (prefill_batch/decode_batch; beyond-parity serving capability)."""
         if not torch.cuda.is_available():
             raise RuntimeError("GPUModel requires a GPU (MI355X)")
         self.config = config
         self.seed = seed
+        assert 1 <= max_batch <= 256
+        self.max_batch = max_batch
+        self._pb = 0  # current sequence row during per-sequence prefill
         self.dtype = dtype
         if dtype not in ("bf16", "fp8"):
             raise ValueError(f"unsupported dtype {dtype!r}")
@@ -124,9 +129,11 @@ class GPUModel:
         cfg = self.config
         free, _total = torch.cuda.mem_get_info(self.device)
         esz = 1 if self.kv8 else 2
-        per_pos = cfg.num_hidden_layers * self.kvh_l * self.hd * 2 * esz
+        per_pos = (cfg.num_hidden_layers * self.kvh_l * self.hd * 2 * esz
+                   * self.max_batch)
         if self.kv8:
-            per_pos += cfg.num_hidden_layers * self.kvh_l * 2 * 4  # scales
+            per_pos += (cfg.num_hidden_layers * self.kvh_l * 2 * 4
+                        * self.max_batch)  # scales
         per_pos += self.hd * 4 + 8  # rope tables + ring slot
         budget = int(free * 0.9) - (2 << 30)  # headroom for scratch/graph
         n = max(1024, budget // max(per_pos, 1))
@@ -235,14 +242,17 @@ class GPUModel:
         L = cfg.num_hidden_layers
         bf = dict(dtype=torch.bfloat16, device=dev)
         kvt = torch.uint8 if self.kv8 else torch.bfloat16
-        self.k_cache = [torch.zeros(self.kvh_l, S, hd, dtype=kvt, device=dev)
+        B = self.max_batch
+        kv_shape = (self.kvh_l, S, hd) if B == 1 else (B, self.kvh_l, S, hd)
+        self.k_cache = [torch.zeros(*kv_shape, dtype=kvt, device=dev)
                         for _ in range(L)]
-        self.v_cache = [torch.zeros(self.kvh_l, S, hd, dtype=kvt, device=dev)
+        self.v_cache = [torch.zeros(*kv_shape, dtype=kvt, device=dev)
                         for _ in range(L)]
         if self.kv8:
-            self.k_scale = [torch.zeros(self.kvh_l, S, dtype=torch.float32,
+            ks = (self.kvh_l, S) if B == 1 else (B, self.kvh_l, S)
+            self.k_scale = [torch.zeros(*ks, dtype=torch.float32,
                                         device=dev) for _ in range(L)]
-            self.v_scale = [torch.zeros(self.kvh_l, S, dtype=torch.float32,
+            self.v_scale = [torch.zeros(*ks, dtype=torch.float32,
                                         device=dev) for _ in range(L)]
         else:
             self.k_scale = [None] * L
@@ -301,6 +311,25 @@ class GPUModel:
         self.nout = torch.zeros(1, **i32)
         self.len_buf = torch.zeros(1, **i32)
         self.rng_ctr = torch.zeros(1, dtype=torch.int64, device=dev)
+
+        if self.max_batch > 1:
+            B, nhh = self.max_batch, self.nh_l * hd
+            i64 = dict(dtype=torch.int64, device=dev)
+            self.bt_qkv = torch.zeros(B, (self.nh_l + 2 * self.kvh_l) * hd,
+                                      **bf)
+            self.bt_logits = torch.zeros(B, self.vocab_l, **bf)
+            self.bt_logits32 = torch.zeros(B, cfg.vocab_size,
+                                           dtype=torch.float32, device=dev)
+            self.bt_next = torch.zeros(B, **i32)
+            self.bt_ring = torch.zeros(B, S + 16, **i32)
+            self.bt_nout = torch.zeros(B, **i32)
+            self.bt_gmax = torch.zeros(B, **i64)
+            self.bt_pick = torch.zeros(B, **i64)
+            self.bt_cnt = torch.zeros(B, **i32)
+            self.bt_scratch = torch.zeros(
+                B * self.nh_l * self.attn_split * (hd + 2),
+                dtype=torch.float32, device=dev)
+            self.bt_attn_cnt = torch.zeros(B * self.nh_l, **i32)
 
         self.act = 0 if cfg.hidden_act == "silu" else 1
         self.gemma = cfg.model_type == "gemma2"
@@ -380,7 +409,28 @@ class GPUModel:
         for wk in works:
             wk.wait()  # compute stream waits on the comm stream
 
-    def _layers_forward(self, M: int, layer_hook=None):
+    def _kc(self, i):
+        return self.k_cache[i] if self.max_batch == 1 else \
+            self.k_cache[i][self._pb]
+
+    def _vc(self, i):
+        return self.v_cache[i] if self.max_batch == 1 else \
+            self.v_cache[i][self._pb]
+
+    def _ks(self, i):
+        s = self.k_scale[i]
+        return s if (s is None or self.max_batch == 1) else s[self._pb]
+
+    def _vs(self, i):
+        s = self.v_scale[i]
+        return s if (s is None or self.max_batch == 1) else s[self._pb]
+
+    def _layers_forward(self, M: int, layer_hook=None,
+                        batch_attn: bool = False):
+        """Layer stack over rows [0, M).  ``batch_attn``: rows are M
+        INDEPENDENT sequences at the same position (lockstep batched
+        decode) — the fused decode-attention kernel runs per row over
+        per-sequence KV pools instead of the prefill attention."""
         cfg = self.config
         eps = cfg.rms_norm_eps
         h, xn, t1 = self.b_h, self.b_xn, self.b_t1
@@ -389,26 +439,37 @@ class GPUModel:
         for i, lw in enumerate(self.layers):
             window = cfg.sliding_window if cfg.is_sliding(i) else 0
             ho.rmsnorm(h[:M], lw["g_in"], xn[:M], eps=eps)
-            self._linear(lw, "wq", xn, self.b_q, M=M)
-            self._linear(lw, "wk", xn, self.b_k, M=M)
-            self._linear(lw, "wv", xn, self.b_v, M=M)
-            ho.rope_cache(self.b_q, self.b_k, self.b_v, self.k_cache[i],
-                          self.v_cache[i], self.cos_t, self.sin_t,
-                          self.len_buf, M, self.nh_l, self.kvh_l, self.hd,
-                          kS=self.k_scale[i], vS=self.v_scale[i])
-            if self.hd in (64, 128, 256):
-                ho.attn_prefill_mfma(
-                    self.b_q, self.k_cache[i], self.v_cache[i], self.b_att,
-                    self.len_buf, M, self.nh_l, self.kvh_l, self.hd,
-                    self.scale, softcap=self.attn_softcap,
-                    window=window or 0,
-                    kS=self.k_scale[i], vS=self.v_scale[i])
+            if batch_attn:
+                self._linear(lw, "wqkv", xn, self.bt_qkv, M=M)
+                ho.attn_dec(self.bt_qkv, self.k_cache[i], self.v_cache[i],
+                            self.b_att[0], self.len_buf, self.cos_t,
+                            self.sin_t, self.bt_scratch, self.bt_attn_cnt,
+                            self.nh_l, self.kvh_l, self.hd, self.scale,
+                            softcap=self.attn_softcap, window=window or 0,
+                            split=self.attn_split, kS=self.k_scale[i],
+                            vS=self.v_scale[i], batch=M)
             else:
-                ho.attn(self.b_q, self.k_cache[i], self.v_cache[i],
-                        self.b_att, self.len_buf, M, self.nh_l, self.kvh_l,
-                        self.hd, self.scale, softcap=self.attn_softcap,
+                self._linear(lw, "wq", xn, self.b_q, M=M)
+                self._linear(lw, "wk", xn, self.b_k, M=M)
+                self._linear(lw, "wv", xn, self.b_v, M=M)
+                ho.rope_cache(self.b_q, self.b_k, self.b_v, self._kc(i),
+                              self._vc(i), self.cos_t, self.sin_t,
+                              self.len_buf, M, self.nh_l, self.kvh_l,
+                              self.hd, kS=self._ks(i), vS=self._vs(i))
+                if self.hd in (64, 128, 256):
+                    ho.attn_prefill_mfma(
+                        self.b_q, self._kc(i), self._vc(i), self.b_att,
+                        self.len_buf, M, self.nh_l, self.kvh_l, self.hd,
+                        self.scale, softcap=self.attn_softcap,
                         window=window or 0,
-                        kS=self.k_scale[i], vS=self.v_scale[i])
+                        kS=self._ks(i), vS=self._vs(i))
+                else:
+                    ho.attn(self.b_q, self._kc(i), self._vc(i),
+                            self.b_att, self.len_buf, M, self.nh_l,
+                            self.kvh_l, self.hd, self.scale,
+                            softcap=self.attn_softcap,
+                            window=window or 0,
+                            kS=self._ks(i), vS=self._vs(i))
             if self.gemma:
                 self._rowpar_ar(lw, "wo", self.b_att, t1, M)
                 ho.rmsnorm(t1[:M], lw["g_post"], h[:M], res=h[:M], eps=eps)
@@ -735,6 +796,120 @@ class GPUModel:
                   self.s_gmax, self.s_pick, self.next_token, self.out_ring,
                   self.nout, self.len_buf, bump_len=True,
                   temperature=temperature, cnt=self.s_cnt)
+
+    # ------------------------------------------------------------------
+    # lockstep batched decode (throughput mode; beyond-parity capability)
+    # ------------------------------------------------------------------
+    def _decode_batch_step(self, B: int, greedy: bool, min_p: float,
+                           temperature: float = 1.0):
+        """One decode step for B lockstep sequences: the layer stack
+        runs with M=B rows (MFMA GEMMs amortize each weight read over
+        the batch), attention/sampling are per-row.  Graph-replayable
+        (all state device-side)."""
+        cfg = self.config
+        ho.embed(self.embed, self.bt_next, self.b_h, B, cfg.embed_scale)
+        self._layers_forward(B, batch_attn=True)
+        ho.rmsnorm(self.b_h[:B], self.g_final, self.b_xn[:B],
+                   eps=cfg.rms_norm_eps)
+        if self.fp8:
+            ho.quant_fp8(self.b_xn[:B], self.b_xq, self.b_sx)
+            ho.gemm_fp8(self.b_xq, self.b_sx, self.lm_head_q,
+                        self.lm_head_s, self.bt_logits[:B], B, self.H,
+                        accbuf=self.b_gemm_acc)
+        else:
+            ho.gemm(self.b_xn[:B], self.lm_head, self.bt_logits[:B],
+                    accbuf=self.b_gemm_acc)
+        if self.final_softcap:
+            ho.softcap(self.bt_logits[:B], self.final_softcap)
+        ho.sample(self.bt_logits[:B], min_p, greedy, self.seed,
+                  self.rng_ctr, self.bt_gmax, self.bt_pick, self.bt_next,
+                  self.bt_ring, self.bt_nout, self.len_buf, bump_len=True,
+                  temperature=temperature, cnt=self.bt_cnt, batch=B)
+
+    def prefill_batch(self, prompts) -> None:
+        """Prefill B same-length prompts into per-sequence KV pools and
+        sample each row's first token from its prefill logits."""
+        arr = np.asarray(prompts, dtype=np.int32)
+        assert arr.ndim == 2, "prompts must be (B, P) same-length"
+        B, P = arr.shape
+        assert 1 <= B <= self.max_batch and self.max_batch > 1
+        assert self.world == 1, "batched decode is single-GPU for now"
+        if P + 1 >= self.max_seq:
+            raise ValueError(f"prompt {P} fills the {self.max_seq} pool")
+        self.reset()
+        self.bt_nout.zero_()
+        for b in range(B):
+            self._pb = b
+            done = 0
+            while done < P:
+                M = min(P - done, self.PC)
+                self.ids_buf[:M].copy_(
+                    torch.from_numpy(arr[b, done:done + M]))
+                ho.i32_set(self.len_buf, done)
+                ho.embed(self.embed, self.ids_buf, self.b_h, M,
+                         self.config.embed_scale)
+                self._layers_forward(M)
+                done += M
+            self._lm_head_last(M)
+            self.bt_logits32[b].copy_(self.b_logits)
+        self._pb = 0
+        ho.i32_set(self.len_buf, P)
+        self._host_len = P
+        self._batch_n = B
+
+    def decode_batch(self, n_tokens: int, greedy: bool = True,
+                     min_p: float = 0.1, temperature: float = 1.0,
+                     use_graph: bool = True) -> np.ndarray:
+        """Decode n_tokens for every prefilled sequence; returns int32
+        ids of shape (B, n_tokens)."""
+        B = self._batch_n
+        if self._host_len + n_tokens > self.max_seq:
+            raise ValueError("decode_batch would overflow the KV pool")
+        self._host_len += n_tokens
+        # first tokens from the prefill logits (fp32, batch rows)
+        ho.sample(self.bt_logits32[:B], min_p, greedy, self.seed,
+                  self.rng_ctr, self.bt_gmax, self.bt_pick, self.bt_next,
+                  self.bt_ring, self.bt_nout, self.len_buf, bump_len=False,
+                  temperature=temperature, cnt=self.bt_cnt, batch=B)
+        n_steps = n_tokens - 1
+        key = ("batch", B, greedy, min_p, temperature)
+        if use_graph and n_steps > 0 and self._graph_mode != key \
+                and not getattr(self, "_graph_failed", False):
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                self._decode_batch_step(B, greedy, min_p, temperature)
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            n_steps -= 1
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._decode_batch_step(B, greedy, min_p, temperature)
+                self._graph = g
+                self._graph_mode = key
+            except Exception as e:
+                self._graph_failed = True
+                self._graph_error = e
+                use_graph = False
+        if use_graph and self._graph_mode == key:
+            for _ in range(n_steps):
+                self._graph.replay()
+        else:
+            for _ in range(n_steps):
+                self._decode_batch_step(B, greedy, min_p, temperature)
+        torch.cuda.synchronize()
+        n = int(self.bt_nout[0].item())
+        out = self.bt_ring[:B, :n].cpu().numpy()
+        return out[:, max(0, n - n_tokens):]
+
+    def generate_tokens_batch(self, prompts, max_tokens: int,
+                              greedy: bool = True, min_p: float = 0.1,
+                              temperature: float = 1.0) -> np.ndarray:
+        """Convenience: prefill_batch + decode_batch -> (B, max_tokens)."""
+        self.prefill_batch(prompts)
+        return self.decode_batch(max_tokens, greedy=greedy, min_p=min_p,
+                                 temperature=temperature)
 
     def capture_decode_graph(self, greedy: bool = True, min_p: float = 0.1,
                              temperature: float = 1.0):

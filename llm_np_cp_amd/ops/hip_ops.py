@@ -44,14 +44,15 @@ _SIGS = {
                                 [ctypes.c_float, ctypes.c_float,
                                  ctypes.c_int, ctypes.c_void_p],
     "launch_attn_dec": [ctypes.c_void_p] * 9 + [ctypes.c_int] +
-                       [ctypes.c_void_p] * 2 + [ctypes.c_int] * 5 +
+                       [ctypes.c_void_p] * 2 + [ctypes.c_int] * 6 +
                        [ctypes.c_float, ctypes.c_float, ctypes.c_int,
                         ctypes.c_void_p],
     "launch_glu": [ctypes.c_void_p] * 3 + [ctypes.c_long, ctypes.c_int,
                                            ctypes.c_void_p],
     "launch_embed": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 +
                     [ctypes.c_float, ctypes.c_void_p],
-    "launch_sample": [ctypes.c_void_p, ctypes.c_int, ctypes.c_float,
+    "launch_sample": [ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
+                      ctypes.c_int, ctypes.c_long, ctypes.c_float,
                       ctypes.c_int, ctypes.c_uint64, ctypes.c_float] +
                      [ctypes.c_void_p] * 8 +
                      [ctypes.c_int, ctypes.c_void_p],
@@ -224,16 +225,18 @@ def attn_dec(qkv: torch.Tensor, k_cache: torch.Tensor,
              cnt: torch.Tensor, nh: int, kvh: int, hd: int,
              scale: float, softcap: float = 0.0, window: int = 0,
              split: int = 1, kS: torch.Tensor | None = None,
-             vS: torch.Tensor | None = None):
+             vS: torch.Tensor | None = None, batch: int = 1):
     """Fused decode attention: RoPE(q,k) + KV write + online softmax,
     KV range split over `split` blocks/head (last-arriver merge).
-    kS/vS given => fp8 KV pool (quantized write + dequant scan)."""
-    S = k_cache.shape[1]
+    kS/vS given => fp8 KV pool (quantized write + dequant scan).
+    batch>1: grid.z = lockstep sequence rows (per-b qkv/out rows, KV
+    pools, scratch and tickets; shared device position)."""
+    S = k_cache.shape[-2]
     _check(lib().launch_attn_dec(
         _ptr(qkv), _ptr(k_cache), _ptr(v_cache), _ptr(out), _ptr(len_ptr),
         _ptr(cos_t), _ptr(sin_t), _ptr(kS), _ptr(vS),
         1 if kS is not None else 0,
-        _ptr(scratch), _ptr(cnt), split,
+        _ptr(scratch), _ptr(cnt), split, batch,
         nh, kvh, hd, S, ctypes.c_float(scale),
         ctypes.c_float(softcap), window, _stream()), "attn_dec")
 
@@ -258,7 +261,7 @@ def sample(logits: torch.Tensor, min_p: float, greedy: bool, seed: int,
            next_token: torch.Tensor, out_ring: torch.Tensor,
            nout: torch.Tensor, len_ptr: torch.Tensor,
            bump_len: bool = True, temperature: float = 1.0,
-           cnt: torch.Tensor | None = None):
+           cnt: torch.Tensor | None = None, batch: int = 1):
     """min-p / greedy sampler: parallel max + Gumbel-argmax; the
     last-arriving pick block commits the winner (no 1-thread fin
     launch).  gmax/pick are u64 scratch (zeroed once; the commit path
@@ -266,6 +269,12 @@ def sample(logits: torch.Tensor, min_p: float, greedy: bool, seed: int,
     Temperature scales device-side (min-p keep-set + Gumbel score), so
     the GPU fast path matches the CPU sample_token() semantics."""
     V = logits.shape[-1]
+    lbf16 = 1 if logits.dtype == torch.bfloat16 else 0
+    ring_stride = out_ring.shape[-1] if out_ring.dim() > 1 else 0
+    if batch > 1:
+        assert (gmax.numel() >= batch and pick.numel() >= batch
+                and next_token.numel() >= batch and nout.numel() >= batch
+                and cnt is not None and cnt.numel() >= batch)
     if cnt is None:
         global _SAMPLE_CNT
         try:
@@ -279,7 +288,8 @@ def sample(logits: torch.Tensor, min_p: float, greedy: bool, seed: int,
         cnt = _SAMPLE_CNT[key]
     inv_temp = 1.0 / max(float(temperature), 1e-6)
     _check(lib().launch_sample(
-        _ptr(logits), V, ctypes.c_float(min_p), 1 if greedy else 0,
+        _ptr(logits), V, lbf16, batch, ctypes.c_long(ring_stride),
+        ctypes.c_float(min_p), 1 if greedy else 0,
         ctypes.c_uint64(seed), ctypes.c_float(inv_temp),
         _ptr(ctr), _ptr(gmax), _ptr(pick), _ptr(cnt),
         _ptr(next_token), _ptr(out_ring), _ptr(nout), _ptr(len_ptr),
