@@ -1000,7 +1000,10 @@ k_attn_dec_t(const u16* __restrict__ qkv, void* __restrict__ kc,
   cnt += b * nh;
   const int grp = nh / kvh;
   const int kvhead = h / grp;
-  const int pos = *len_ptr;
+  // RAGGED batch: each sequence row sits at its own position
+  // (len_ptr is a per-row array; the single-sequence path passes its
+  // scalar length buffer and reads index 0)
+  const int pos = len_ptr[b];
   int start = 0;
   if (window > 0 && pos + 1 - window > 0) start = pos + 1 - window;
   // history chunk [c0, c1) of [start, pos)
@@ -1732,11 +1735,10 @@ k_sample_pick(const void* __restrict__ logits, int V, int lbf16,
   int n = *nout;
   out_ring[n] = winner;
   *nout = n + 1;
-  // shared device state advanced once per step (row 0's committer)
-  if (b == 0) {
-    if (bump_len) *len_ptr += 1;
-    if (!greedy) *ctr += 1;
-  }
+  // per-row position advances (ragged batch); the RNG counter is
+  // shared and advanced once per step (row 0's committer)
+  if (bump_len) len_ptr[b] += 1;
+  if (b == 0 && !greedy) *ctr += 1;
   *pick = 0ull;
   *gmax = 0ull;
 }

@@ -321,6 +321,7 @@ class GPUModel:
             self.bt_logits32 = torch.zeros(B, cfg.vocab_size,
                                            dtype=torch.float32, device=dev)
             self.bt_next = torch.zeros(B, **i32)
+            self.bt_lens = torch.zeros(B, **i32)  # ragged positions
             self.bt_ring = torch.zeros(B, S + 16, **i32)
             self.bt_nout = torch.zeros(B, **i32)
             self.bt_gmax = torch.zeros(B, **i64)
@@ -442,7 +443,7 @@ class GPUModel:
             if batch_attn:
                 self._linear(lw, "wqkv", xn, self.bt_qkv, M=M)
                 ho.attn_dec(self.bt_qkv, self.k_cache[i], self.v_cache[i],
-                            self.b_att[0], self.len_buf, self.cos_t,
+                            self.b_att[0], self.bt_lens, self.cos_t,
                             self.sin_t, self.bt_scratch, self.bt_attn_cnt,
                             self.nh_l, self.kvh_l, self.hd, self.scale,
                             softcap=self.attn_softcap, window=window or 0,
@@ -823,28 +824,31 @@ class GPUModel:
             ho.softcap(self.bt_logits[:B], self.final_softcap)
         ho.sample(self.bt_logits[:B], min_p, greedy, self.seed,
                   self.rng_ctr, self.bt_gmax, self.bt_pick, self.bt_next,
-                  self.bt_ring, self.bt_nout, self.len_buf, bump_len=True,
+                  self.bt_ring, self.bt_nout, self.bt_lens, bump_len=True,
                   temperature=temperature, cnt=self.bt_cnt, batch=B)
 
     def prefill_batch(self, prompts) -> None:
-        """Prefill B same-length prompts into per-sequence KV pools and
-        sample each row's first token from its prefill logits."""
-        arr = np.asarray(prompts, dtype=np.int32)
-        assert arr.ndim == 2, "prompts must be (B, P) same-length"
-        B, P = arr.shape
+        """Prefill B prompts (RAGGED lengths allowed) into per-sequence
+        KV pools; each row then decodes from its own position
+        (bt_lens[b] device-side)."""
+        seqs = [np.asarray(p, dtype=np.int32).ravel() for p in prompts]
+        B = len(seqs)
         assert 1 <= B <= self.max_batch and self.max_batch > 1
         assert self.world == 1, "batched decode is single-GPU for now"
-        if P + 1 >= self.max_seq:
-            raise ValueError(f"prompt {P} fills the {self.max_seq} pool")
+        lens = [len(sq) for sq in seqs]
+        if max(lens) + 1 >= self.max_seq:
+            raise ValueError(f"prompt {max(lens)} fills the "
+                             f"{self.max_seq} pool")
         self.reset()
         self.bt_nout.zero_()
         for b in range(B):
             self._pb = b
+            P = lens[b]
             done = 0
             while done < P:
                 M = min(P - done, self.PC)
                 self.ids_buf[:M].copy_(
-                    torch.from_numpy(arr[b, done:done + M]))
+                    torch.from_numpy(seqs[b][done:done + M]))
                 ho.i32_set(self.len_buf, done)
                 ho.embed(self.embed, self.ids_buf, self.b_h, M,
                          self.config.embed_scale)
@@ -853,8 +857,8 @@ class GPUModel:
             self._lm_head_last(M)
             self.bt_logits32[b].copy_(self.b_logits)
         self._pb = 0
-        ho.i32_set(self.len_buf, P)
-        self._host_len = P
+        self.bt_lens[:B].copy_(torch.tensor(lens, dtype=torch.int32))
+        self._host_lens = lens
         self._batch_n = B
 
     def decode_batch(self, n_tokens: int, greedy: bool = True,
@@ -863,13 +867,13 @@ class GPUModel:
         """Decode n_tokens for every prefilled sequence; returns int32
         ids of shape (B, n_tokens)."""
         B = self._batch_n
-        if self._host_len + n_tokens > self.max_seq:
+        if max(self._host_lens) + n_tokens > self.max_seq:
             raise ValueError("decode_batch would overflow the KV pool")
-        self._host_len += n_tokens
+        self._host_lens = [p + n_tokens for p in self._host_lens]
         # first tokens from the prefill logits (fp32, batch rows)
         ho.sample(self.bt_logits32[:B], min_p, greedy, self.seed,
                   self.rng_ctr, self.bt_gmax, self.bt_pick, self.bt_next,
-                  self.bt_ring, self.bt_nout, self.len_buf, bump_len=False,
+                  self.bt_ring, self.bt_nout, self.bt_lens, bump_len=False,
                   temperature=temperature, cnt=self.bt_cnt, batch=B)
         n_steps = n_tokens - 1
         key = ("batch", B, greedy, min_p, temperature)

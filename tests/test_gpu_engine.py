@@ -479,3 +479,33 @@ def test_batch_decode_fp8_weights_and_kv():
     # deterministic across runs
     ids2 = m.generate_tokens_batch(prompts, 5, greedy=True)
     np.testing.assert_array_equal(ids, ids2)
+
+
+def test_batch_decode_ragged_lengths():
+    """RAGGED batch: rows prefilled at different lengths decode in
+    lockstep from their own device positions and each reproduces its
+    single-sequence rollout."""
+    import llm_np_cp_amd as L
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.engine import GPUModel
+
+    cfg = L.preset_config("tiny-llama")
+    w = random_weights(cfg, seed=80)
+    rng = np.random.default_rng(81)
+    lens = [5, 11, 8]
+    prompts = [rng.integers(0, cfg.vocab_size, size=p) for p in lens]
+    N = 6
+
+    singles = []
+    single = GPUModel(cfg, dict(w), max_seq=64)
+    for p in prompts:
+        single.prefill(p)
+        singles.append(single.decode(N, greedy=True, use_graph=False))
+    del single
+    torch.cuda.empty_cache()
+
+    m = GPUModel(cfg, dict(w), max_seq=64, max_batch=4)
+    ids = m.generate_tokens_batch(prompts, N, greedy=True)
+    assert ids.shape == (3, N)
+    for b in range(3):
+        np.testing.assert_array_equal(ids[b], singles[b])
