@@ -863,19 +863,23 @@ class GPUModel:
 
     def decode_batch(self, n_tokens: int, greedy: bool = True,
                      min_p: float = 0.1, temperature: float = 1.0,
-                     use_graph: bool = True) -> np.ndarray:
+                     use_graph: bool = True,
+                     first_from_logits: bool = True) -> np.ndarray:
         """Decode n_tokens for every prefilled sequence; returns int32
-        ids of shape (B, n_tokens)."""
+        ids of shape (B, n_tokens).  ``first_from_logits=False``
+        continues a previous decode_batch (chunked serving)."""
         B = self._batch_n
         if max(self._host_lens) + n_tokens > self.max_seq:
             raise ValueError("decode_batch would overflow the KV pool")
         self._host_lens = [p + n_tokens for p in self._host_lens]
-        # first tokens from the prefill logits (fp32, batch rows)
-        ho.sample(self.bt_logits32[:B], min_p, greedy, self.seed,
-                  self.rng_ctr, self.bt_gmax, self.bt_pick, self.bt_next,
-                  self.bt_ring, self.bt_nout, self.bt_lens, bump_len=False,
-                  temperature=temperature, cnt=self.bt_cnt, batch=B)
-        n_steps = n_tokens - 1
+        if first_from_logits:
+            # first tokens from the prefill logits (fp32, batch rows)
+            ho.sample(self.bt_logits32[:B], min_p, greedy, self.seed,
+                      self.rng_ctr, self.bt_gmax, self.bt_pick,
+                      self.bt_next, self.bt_ring, self.bt_nout,
+                      self.bt_lens, bump_len=False,
+                      temperature=temperature, cnt=self.bt_cnt, batch=B)
+        n_steps = n_tokens - (1 if first_from_logits else 0)
         key = ("batch", B, greedy, min_p, temperature)
         if use_graph and n_steps > 0 and self._graph_mode != key \
                 and not getattr(self, "_graph_failed", False):
