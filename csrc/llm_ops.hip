@@ -2188,6 +2188,35 @@ extern "C" hipError_t launch_softcap(void* y, long total, float cap,
 }
 
 // ====================================================================
+// y[M,N] += bias[N] (bf16): Qwen-2 qkv-bias epilogue on the prefill /
+// batched GEMM path (the decode GEMV adds bias via its `res` slot).
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_bias_add(u16* __restrict__ y, const u16* __restrict__ bias, long total,
+           int N) {
+  long i = ((long)blockIdx.x * 256 + threadIdx.x) * 8;
+  if (i >= total) return;
+  s8v yv = *(const s8v*)(y + i);
+  s8v bv = *(const s8v*)(bias + (i % N));
+  u16 o[8];
+#pragma unroll
+  for (int j = 0; j < 8; j++)
+    o[j] = f2b(b2f(((u16*)&yv)[j]) + b2f(((u16*)&bv)[j]));
+  *(s8v*)(y + i) = *(s8v*)o;
+}
+
+extern "C" hipError_t launch_bias_add(void* y, const void* bias, int M,
+                                      int N, hipStream_t stream) {
+  if (N % 8 != 0) return hipErrorInvalidValue;
+  long total = (long)M * N;
+  hipLaunchKernelGGL(k_bias_add, dim3((uint32_t)((total / 8 + 255) / 256)),
+                     dim3(256), 0, stream, (u16*)y, (const u16*)bias, total,
+                     N);
+  return hipGetLastError();
+}
+
+// ====================================================================
 // y += a (bf16, fp32 math).  Used on the TP path where the RCCL
 // all-reduce sits between the row-parallel GEMV and the residual add.
 // ====================================================================

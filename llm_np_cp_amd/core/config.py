@@ -40,6 +40,7 @@ class ModelConfig:
     tie_word_embeddings: bool = True
     bos_token_id: int = 1
     eos_token_id: int = 2
+    attention_bias: bool = False  # Qwen-2(.5): bias on q/k/v projections
     # Gemma-2 specifics
     query_pre_attn_scalar: Optional[float] = None
     sliding_window: Optional[int] = None
@@ -58,6 +59,10 @@ class ModelConfig:
                     "sliding_attention" if i % 2 == 0 else "full_attention"
                     for i in range(self.num_hidden_layers)
                 ]
+            elif self.model_type == "mistral" and self.sliding_window:
+                # Mistral v0.1/v0.2: sliding window on EVERY layer
+                self.layer_types = (["sliding_attention"]
+                                    * self.num_hidden_layers)
             else:
                 self.layer_types = ["full_attention"] * self.num_hidden_layers
 
@@ -109,6 +114,9 @@ class ModelConfig:
             tie_word_embeddings=d.get("tie_word_embeddings", True),
             bos_token_id=d.get("bos_token_id", 1),
             eos_token_id=d.get("eos_token_id", 2),
+            # HF Qwen2Attention hardwires qkv bias=True (no config key)
+            attention_bias=d.get("attention_bias",
+                                 model_type == "qwen2"),
             query_pre_attn_scalar=d.get("query_pre_attn_scalar"),
             sliding_window=d.get("sliding_window"),
             attn_logit_softcapping=d.get("attn_logit_softcapping"),
@@ -140,6 +148,8 @@ class ModelConfig:
             "bos_token_id": self.bos_token_id,
             "eos_token_id": self.eos_token_id,
         }
+        if self.attention_bias:
+            d["attention_bias"] = True
         if self.rope_scaling is not None:
             d["rope_scaling"] = self.rope_scaling
         for k in ("query_pre_attn_scalar", "sliding_window",
@@ -237,6 +247,28 @@ PRESETS = {
         tie_word_embeddings=False, bos_token_id=128000,
         eos_token_id=128001,
     ),
+    # Qwen-2.5: Llama-family arch + qkv bias + untied lm_head
+    # (shape table from the public Qwen/Qwen2.5-7B config.json)
+    "qwen2.5-7b": dict(
+        model_type="qwen2", vocab_size=152064, hidden_size=3584,
+        intermediate_size=18944, num_hidden_layers=28,
+        num_attention_heads=28, num_key_value_heads=4, head_dim=128,
+        rms_norm_eps=1e-6, rope_theta=1000000.0,
+        max_position_embeddings=32768, hidden_act="silu",
+        tie_word_embeddings=False, bos_token_id=151643,
+        eos_token_id=151643, attention_bias=True,
+    ),
+    # Mistral: Llama-family arch + sliding window on every layer
+    # (shape table from the public mistralai/Mistral-7B-v0.1 config.json)
+    "mistral-7b": dict(
+        model_type="mistral", vocab_size=32000, hidden_size=4096,
+        intermediate_size=14336, num_hidden_layers=32,
+        num_attention_heads=32, num_key_value_heads=8, head_dim=128,
+        rms_norm_eps=1e-5, rope_theta=10000.0,
+        max_position_embeddings=32768, hidden_act="silu",
+        tie_word_embeddings=False, bos_token_id=1, eos_token_id=2,
+        sliding_window=4096,
+    ),
     "gemma-2-27b": dict(
         model_type="gemma2", vocab_size=256000, hidden_size=4608,
         intermediate_size=36864, num_hidden_layers=46,
@@ -268,6 +300,20 @@ PRESETS = {
         num_key_value_heads=2, head_dim=16, rms_norm_eps=1e-5,
         rope_theta=10000.0, max_position_embeddings=512, hidden_act="silu",
         tie_word_embeddings=True,
+    ),
+    "tiny-qwen2": dict(         # llama arch + qkv bias
+        model_type="qwen2", vocab_size=512, hidden_size=64,
+        intermediate_size=128, num_hidden_layers=2, num_attention_heads=4,
+        num_key_value_heads=2, head_dim=16, rms_norm_eps=1e-6,
+        rope_theta=10000.0, max_position_embeddings=512, hidden_act="silu",
+        tie_word_embeddings=True, attention_bias=True,
+    ),
+    "tiny-mistral": dict(       # llama arch + ALL-layer sliding window
+        model_type="mistral", vocab_size=512, hidden_size=128,
+        intermediate_size=256, num_hidden_layers=2, num_attention_heads=2,
+        num_key_value_heads=1, head_dim=64, rms_norm_eps=1e-5,
+        rope_theta=10000.0, max_position_embeddings=512, hidden_act="silu",
+        tie_word_embeddings=True, sliding_window=8,
     ),
     "tiny-gemma2-hd64": dict(   # exercises gemma + MFMA prefill (hd=64)
         model_type="gemma2", vocab_size=512, hidden_size=128,

@@ -87,6 +87,10 @@ def hf_weight_shapes(cfg: ModelConfig) -> Dict[str, Tuple[int, ...]]:
         shapes[f"{p}.self_attn.q_proj.weight"] = (nh * hd, h)
         shapes[f"{p}.self_attn.k_proj.weight"] = (kvh * hd, h)
         shapes[f"{p}.self_attn.v_proj.weight"] = (kvh * hd, h)
+        if cfg.attention_bias:
+            shapes[f"{p}.self_attn.q_proj.bias"] = (nh * hd,)
+            shapes[f"{p}.self_attn.k_proj.bias"] = (kvh * hd,)
+            shapes[f"{p}.self_attn.v_proj.bias"] = (kvh * hd,)
         shapes[f"{p}.self_attn.o_proj.weight"] = (h, nh * hd)
         shapes[f"{p}.mlp.gate_proj.weight"] = (im, h)
         shapes[f"{p}.mlp.up_proj.weight"] = (im, h)

@@ -234,6 +234,18 @@ class GPUModel:
             if gemma:
                 lw["g_preffn"] = gamma(w[f"{p}.pre_feedforward_layernorm.weight"])
                 lw["g_postffn"] = gamma(w[f"{p}.post_feedforward_layernorm.weight"])
+            if cfg.attention_bias:
+                # Qwen-2 family: fused qkv bias (decode GEMV adds it via
+                # the res slot; prefill adds it after the GEMM)
+                bias_np = _np.concatenate([
+                    hrows(w[f"{a}.q_proj.bias"]),
+                    hrows(w[f"{a}.k_proj.bias"]),
+                    hrows(w[f"{a}.v_proj.bias"])], axis=0)
+                bqkv = bf16(bias_np.reshape(1, -1))[0]
+                lw["bqkv"] = bqkv
+                lw["bias_q"] = bqkv[:nq]
+                lw["bias_k"] = bqkv[nq:nq + nkv]
+                lw["bias_v"] = bqkv[nq + nkv:]
             self.layers.append(lw)
 
     def _alloc_state(self, prefill_chunk: int):
@@ -442,6 +454,8 @@ class GPUModel:
             ho.rmsnorm(h[:M], lw["g_in"], xn[:M], eps=eps)
             if batch_attn:
                 self._linear(lw, "wqkv", xn, self.bt_qkv, M=M)
+                if "bqkv" in lw:
+                    ho.bias_add(self.bt_qkv, lw["bqkv"], M)
                 ho.attn_dec(self.bt_qkv, self.k_cache[i], self.v_cache[i],
                             self.b_att[0], self.bt_lens, self.cos_t,
                             self.sin_t, self.bt_scratch, self.bt_attn_cnt,
@@ -453,6 +467,10 @@ class GPUModel:
                 self._linear(lw, "wq", xn, self.b_q, M=M)
                 self._linear(lw, "wk", xn, self.b_k, M=M)
                 self._linear(lw, "wv", xn, self.b_v, M=M)
+                if "bqkv" in lw:
+                    ho.bias_add(self.b_q, lw["bias_q"], M)
+                    ho.bias_add(self.b_k, lw["bias_k"], M)
+                    ho.bias_add(self.b_v, lw["bias_v"], M)
                 ho.rope_cache(self.b_q, self.b_k, self.b_v, self._kc(i),
                               self._vc(i), self.cos_t, self.sin_t,
                               self.len_buf, M, self.nh_l, self.kvh_l,
@@ -715,7 +733,7 @@ class GPUModel:
                             g=prev["g_postffn"], g2=lw["g_in"], res=hnext,
                             eps=eps)
                 h, hnext = hnext, h
-            elif i == 0:
+            elif i == 0 and "bqkv" not in lw:
                 # layer 0: embed gather fused into the QKV staging pass
                 # (x = table, x2 = sampled token id, res = persisted h)
                 self._dgemv(lw, "wqkv", self.embed, self.b_qkv,
@@ -723,8 +741,11 @@ class GPUModel:
                             g=lw["g_in"], res=h, eps=eps,
                             escale=cfg.embed_scale)
             else:
+                if i == 0:  # bias models keep the separate embed gather
+                    ho.embed(self.embed, self.next_token, self.b_h, 1,
+                             cfg.embed_scale)
                 self._dgemv(lw, "wqkv", h, self.b_qkv, stage=ho.STAGE_NORM,
-                            g=lw["g_in"], eps=eps)
+                            g=lw["g_in"], res=lw.get("bqkv"), eps=eps)
             prev = lw
             ho.attn_dec(self.b_qkv, self.k_cache[i], self.v_cache[i],
                         self.b_att[0], self.len_buf, self.cos_t, self.sin_t,

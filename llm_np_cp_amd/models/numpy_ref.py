@@ -118,6 +118,10 @@ class NumpyModel:
         q = h @ self.w[f"{p}.q_proj.weight"].T
         k = h @ self.w[f"{p}.k_proj.weight"].T
         v = h @ self.w[f"{p}.v_proj.weight"].T
+        if f"{p}.q_proj.bias" in self.w:  # Qwen-2 family
+            q = q + self.w[f"{p}.q_proj.bias"]
+            k = k + self.w[f"{p}.k_proj.bias"]
+            v = v + self.w[f"{p}.v_proj.bias"]
         nh, kvh, hd = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
         q = q.reshape(q_len, nh, hd).transpose(1, 0, 2)    # (nh, q, hd)
         k = k.reshape(q_len, kvh, hd).transpose(1, 0, 2)   # (kvh, q, hd)

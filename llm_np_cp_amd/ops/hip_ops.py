@@ -64,6 +64,8 @@ _SIGS = {
                        ctypes.c_void_p],
     "launch_addinto": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_long,
                        ctypes.c_void_p],
+    "launch_bias_add": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
+                        ctypes.c_int, ctypes.c_void_p],
     "launch_i32_set": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
     "launch_i32_add": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
     "launch_quant_fp8": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 +
@@ -350,6 +352,13 @@ def softcap(y: torch.Tensor, cap: float):
     assert total % 8 == 0
     _check(lib().launch_softcap(_ptr(y), total, ctypes.c_float(cap),
                                 _stream()), "softcap")
+
+
+def bias_add(y: torch.Tensor, bias: torch.Tensor, M: int):
+    """y[:M, :N] += bias[N] (bf16) — Qwen-2 qkv bias on GEMM outputs."""
+    N = bias.numel()
+    _check(lib().launch_bias_add(_ptr(y), _ptr(bias), M, N, _stream()),
+           "bias_add")
 
 
 def addinto(y: torch.Tensor, a: torch.Tensor):

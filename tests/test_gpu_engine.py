@@ -49,7 +49,8 @@ def make_pair(preset, seed=0, max_seq=256):
     return cfg, gpu, ref
 
 
-@pytest.mark.parametrize("preset", ["tiny-llama", "tiny-gemma2"])
+@pytest.mark.parametrize("preset", ["tiny-llama", "tiny-gemma2",
+                                    "tiny-qwen2", "tiny-mistral"])
 def test_prefill_logits_match_oracle(preset):
     cfg, gpu, ref = make_pair(preset)
     rng = np.random.default_rng(0)
@@ -70,7 +71,8 @@ def test_prefill_logits_match_oracle(preset):
     assert topk_overlap(got, ref_logits[-1], k=8) >= 0.875
 
 
-@pytest.mark.parametrize("preset", ["tiny-llama", "tiny-gemma2"])
+@pytest.mark.parametrize("preset", ["tiny-llama", "tiny-gemma2",
+                                    "tiny-qwen2", "tiny-mistral"])
 def test_greedy_decode_matches_oracle(preset):
     """Token-id equality over a greedy rollout (SURVEY §4 integration)."""
     import llm_np_cp_amd as L

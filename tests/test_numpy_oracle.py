@@ -169,3 +169,48 @@ def test_llama3_rope_scaling_matches_transformers():
         ref = hf(torch.tensor(ids[None])).logits[0].numpy()
     got = model.forward(ids, NumpyKVCache(cfg, 64), 0)
     np.testing.assert_allclose(got, ref, rtol=3e-4, atol=3e-4)
+
+
+def test_qwen2_bias_affects_oracle():
+    """Qwen-2 family: qkv biases are loaded and change the forward."""
+    import llm_np_cp_amd as L
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.numpy_ref import NumpyModel, NumpyKVCache
+
+    cfg = L.preset_config("tiny-qwen2")
+    w = random_weights(cfg, seed=1)
+    assert "model.layers.0.self_attn.q_proj.bias" in w
+    m = NumpyModel(cfg, dict(w))
+    ids = np.arange(1, 6)
+    a = m.forward(ids, NumpyKVCache(cfg, 16), 0)
+
+    w2 = dict(w)
+    for i in range(cfg.num_hidden_layers):
+        for n in ("q", "k", "v"):
+            key = f"model.layers.{i}.self_attn.{n}_proj.bias"
+            w2[key] = np.zeros_like(w2[key])
+    b = NumpyModel(cfg, w2).forward(ids, NumpyKVCache(cfg, 16), 0)
+    assert not np.allclose(a, b)
+
+
+def test_mistral_all_layer_sliding_window():
+    """Mistral: sliding window active on EVERY layer (vs Gemma's
+    alternating); the window changes long-prefix logits."""
+    import llm_np_cp_amd as L
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.numpy_ref import NumpyModel, NumpyKVCache
+
+    cfg = L.preset_config("tiny-mistral")
+    assert all(cfg.is_sliding(i) for i in range(cfg.num_hidden_layers))
+    w = random_weights(cfg, seed=2)
+    m = NumpyModel(cfg, dict(w))
+    rng = np.random.default_rng(3)
+    ids = rng.integers(0, cfg.vocab_size, size=20)  # > window 8
+    a = m.forward(ids, NumpyKVCache(cfg, 32), 0)
+
+    cfg2 = L.preset_config("tiny-mistral")
+    cfg2.sliding_window = None
+    b = NumpyModel(cfg2, dict(w)).forward(ids, NumpyKVCache(cfg2, 32), 0)
+    # early positions (inside the window) agree; late ones differ
+    np.testing.assert_allclose(a[:8], b[:8], rtol=1e-5, atol=1e-6)
+    assert not np.allclose(a[-1], b[-1])
