@@ -28,6 +28,7 @@
 #include <hip/hip_runtime.h>
 #include <cstdint>
 #include <cstdio>
+#include <cstdlib>
 
 #include "common.h"
 
@@ -452,7 +453,7 @@ DEVINL void fp8x16_to_f32(u4v w, float* o) {
 // LDS fp32 pair — ~16 VALU per 16 weight bytes vs ~40 for the round-1
 // scalar form (the fp8 stream was conversion-bound at 4.3 TB/s while
 // bf16 hit 6.4; see profiles/decode_kernels_r01.md).
-template <bool NT, int RPW>
+template <bool NT, int RPW, bool XDIR>
 __global__ void __launch_bounds__(512)
 k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
              const u16* __restrict__ x, const u16* __restrict__ x2,
@@ -461,8 +462,13 @@ k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
              int stage, int act, float eps, int out_f32, float softcap,
              float escale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const float* xv = gemv_stage_f32(smem, x, x2, g, g2, (u16*)res, K, stage,
-                                   act, eps, escale);
+  // XDIR (RAW stage only): x read from global per row iteration — the
+  // f32 LDS staging pass costs ~2K B of LDS+global traffic per BLOCK,
+  // which rivals the weight bytes when a block covers only a few rows
+  // (N~3.5k o/down projections; see profiles/decode_kernels_r02.md)
+  const float* xv = XDIR ? nullptr
+                         : gemv_stage_f32(smem, x, x2, g, g2, (u16*)res, K,
+                                          stage, act, eps, escale);
   const u16* eres = (stage == STAGE_NORM2 || stage == STAGE_NORM_EMBED)
                         ? nullptr : res;
 
@@ -482,6 +488,11 @@ k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
   }
   int k = lane * 16;
   for (; k + 1024 + 16 <= K; k += 2048) {
+    s8v xda, xdb, xdc, xdd;
+    if (XDIR) {
+      xda = *(const s8v*)(x + k); xdb = *(const s8v*)(x + k + 8);
+      xdc = *(const s8v*)(x + k + 1024); xdd = *(const s8v*)(x + k + 1032);
+    }
 #pragma unroll
     for (int r = 0; r < RPW; r++) {
       u4v w0 = NT ? __builtin_nontemporal_load((const u4v*)(Wr[r] + k))
@@ -492,16 +503,35 @@ k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
       for (int q = 0; q < 4; q++) {
         f2v ca = __builtin_amdgcn_cvt_pk_f32_fp8(w0[q], false);
         f2v cb = __builtin_amdgcn_cvt_pk_f32_fp8(w0[q], true);
-        a0[r] += ca * *(const f2v*)(xv + k + q * 4);
-        a1[r] += cb * *(const f2v*)(xv + k + q * 4 + 2);
+        f2v xa, xb, xc, xd;
+        if (XDIR) {
+          const u16* p0 = (q < 2) ? (const u16*)&xda : (const u16*)&xdb;
+          const int o0 = (q & 1) * 4;
+          xa = (f2v){b2f(p0[o0]), b2f(p0[o0 + 1])};
+          xb = (f2v){b2f(p0[o0 + 2]), b2f(p0[o0 + 3])};
+          const u16* p1 = (q < 2) ? (const u16*)&xdc : (const u16*)&xdd;
+          xc = (f2v){b2f(p1[o0]), b2f(p1[o0 + 1])};
+          xd = (f2v){b2f(p1[o0 + 2]), b2f(p1[o0 + 3])};
+        } else {
+          xa = *(const f2v*)(xv + k + q * 4);
+          xb = *(const f2v*)(xv + k + q * 4 + 2);
+          xc = *(const f2v*)(xv + k + 1024 + q * 4);
+          xd = *(const f2v*)(xv + k + 1024 + q * 4 + 2);
+        }
+        a0[r] += ca * xa;
+        a1[r] += cb * xb;
         f2v da = __builtin_amdgcn_cvt_pk_f32_fp8(w1[q], false);
         f2v db = __builtin_amdgcn_cvt_pk_f32_fp8(w1[q], true);
-        a0[r] += da * *(const f2v*)(xv + k + 1024 + q * 4);
-        a1[r] += db * *(const f2v*)(xv + k + 1024 + q * 4 + 2);
+        a0[r] += da * xc;
+        a1[r] += db * xd;
       }
     }
   }
   for (; k < K; k += 1024) {
+    s8v xda, xdb;
+    if (XDIR) {
+      xda = *(const s8v*)(x + k); xdb = *(const s8v*)(x + k + 8);
+    }
 #pragma unroll
     for (int r = 0; r < RPW; r++) {
       u4v w0 = NT ? __builtin_nontemporal_load((const u4v*)(Wr[r] + k))
@@ -510,8 +540,18 @@ k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
       for (int q = 0; q < 4; q++) {
         f2v ca = __builtin_amdgcn_cvt_pk_f32_fp8(w0[q], false);
         f2v cb = __builtin_amdgcn_cvt_pk_f32_fp8(w0[q], true);
-        a0[r] += ca * *(const f2v*)(xv + k + q * 4);
-        a1[r] += cb * *(const f2v*)(xv + k + q * 4 + 2);
+        f2v xa, xb;
+        if (XDIR) {
+          const u16* p0 = (q < 2) ? (const u16*)&xda : (const u16*)&xdb;
+          const int o0 = (q & 1) * 4;
+          xa = (f2v){b2f(p0[o0]), b2f(p0[o0 + 1])};
+          xb = (f2v){b2f(p0[o0 + 2]), b2f(p0[o0 + 3])};
+        } else {
+          xa = *(const f2v*)(xv + k + q * 4);
+          xb = *(const f2v*)(xv + k + q * 4 + 2);
+        }
+        a0[r] += ca * xa;
+        a1[r] += cb * xb;
       }
     }
   }
@@ -534,15 +574,23 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
                                       int out_f32, float softcap, int nt,
                                       int rpw, int maxblocks, float escale,
                                       hipStream_t stream) {
-  // fp32 staging for packed math (every stage incl. RAW)
-  size_t lds = (size_t)K * 4 + 32;
+  static int xdir_raw = -1, rows_min = -1;
+  if (xdir_raw < 0) {
+    const char* e = getenv("LLM_GEMV_XDIR");
+    xdir_raw = e ? atoi(e) : 1;          // default ON for RAW stage
+    const char* r = getenv("LLM_GEMV_ROWSMIN");
+    rows_min = r ? atoi(r) : 1;
+  }
+  const int xdir = (stage == STAGE_RAW) && xdir_raw;
+  // fp32 staging for packed math (all non-RAW stages; RAW reads direct)
+  size_t lds = xdir ? 0 : ((size_t)K * 4 + 32);
   if (lds > 65536) {
     // gfx950 allows up to 160 KB dynamic LDS with an explicit opt-in
     // (e.g. Gemma-27B down-proj K=36864)
     static bool raised = false;
     if (!raised) {
 #define GEMV8_RAISE(NTV, RPWV)                                              \
-      hipFuncSetAttribute((const void*)&k_gemv_fp8_t<NTV, RPWV>,            \
+      hipFuncSetAttribute((const void*)&k_gemv_fp8_t<NTV, RPWV, false>,     \
                           hipFuncAttributeMaxDynamicSharedMemorySize,       \
                           160 * 1024)
       GEMV8_RAISE(true, 1); GEMV8_RAISE(true, 2);
@@ -555,17 +603,31 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
   int wpb = threads / 64;
   int blocks = (N + wpb * rpw - 1) / (wpb * rpw);
   int cap = maxblocks > 0 ? maxblocks : 1024;
+  if (rows_min > 1 && !xdir) {
+    // amortize the staging pass over >= rows_min rows per wave
+    int cap2 = N / (wpb * rpw * rows_min);
+    if (cap2 < 64) cap2 = 64;
+    if (cap > cap2) cap = cap2;
+  }
   if (blocks > cap) blocks = cap;
-#define GEMV8_CASE(NTV, RPWV)                                               \
-  hipLaunchKernelGGL((k_gemv_fp8_t<NTV, RPWV>), dim3(blocks), dim3(threads),\
+#define GEMV8_CASE(NTV, RPWV, XD)                                           \
+  hipLaunchKernelGGL((k_gemv_fp8_t<NTV, RPWV, XD>), dim3(blocks),           \
+                     dim3(threads),                                         \
                      lds, stream, (const uint8_t*)W, (const float*)scales,  \
                      (const u16*)x, (const u16*)x2, (const float*)g,        \
                      (const float*)g2, y, (const u16*)res, N, K, stage,     \
                      act, eps, out_f32, softcap, escale)
-  if (nt && rpw == 2) GEMV8_CASE(true, 2);
-  else if (nt) GEMV8_CASE(true, 1);
-  else if (rpw == 2) GEMV8_CASE(false, 2);
-  else GEMV8_CASE(false, 1);
+  if (xdir) {
+    if (nt && rpw == 2) GEMV8_CASE(true, 2, true);
+    else if (nt) GEMV8_CASE(true, 1, true);
+    else if (rpw == 2) GEMV8_CASE(false, 2, true);
+    else GEMV8_CASE(false, 1, true);
+  } else {
+    if (nt && rpw == 2) GEMV8_CASE(true, 2, false);
+    else if (nt) GEMV8_CASE(true, 1, false);
+    else if (rpw == 2) GEMV8_CASE(false, 2, false);
+    else GEMV8_CASE(false, 1, false);
+  }
 #undef GEMV8_CASE
   return hipGetLastError();
 }
