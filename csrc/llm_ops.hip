@@ -574,13 +574,16 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
                                       int out_f32, float softcap, int nt,
                                       int rpw, int maxblocks, float escale,
                                       hipStream_t stream) {
-  static int xdir_raw = -1, rows_min = -1;
+  static int xdir_raw = -1, rows_min = -1, rpw_env = -1;
   if (xdir_raw < 0) {
     const char* e = getenv("LLM_GEMV_XDIR");
     xdir_raw = e ? atoi(e) : 1;          // default ON for RAW stage
     const char* r = getenv("LLM_GEMV_ROWSMIN");
     rows_min = r ? atoi(r) : 1;
+    const char* p = getenv("LLM_GEMV_RPW");
+    rpw_env = p ? atoi(p) : 0;
   }
+  if (rpw_env > 0) rpw = rpw_env;
   const int xdir = (stage == STAGE_RAW) && xdir_raw;
   // fp32 staging for packed math (all non-RAW stages; RAW reads direct)
   size_t lds = xdir ? 0 : ((size_t)K * 4 + 32);
