@@ -584,6 +584,11 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
     rpw_env = p ? atoi(p) : 0;
   }
   if (rpw_env > 0) rpw = rpw_env;
+  else if (rpw <= 1)
+    // deep-K rows: 2x loads in flight (gemma-9b 240->300, llama-8b
+    // 393->442, qwen-7b 352->442); N floor keeps small-N shapes (1B
+    // down-proj) at full block count
+    rpw = (K >= 3584 && N >= 3072) ? 2 : 1;
   const int xdir = (stage == STAGE_RAW) && xdir_raw;
   // fp32 staging for packed math (all non-RAW stages; RAW reads direct)
   size_t lds = xdir ? 0 : ((size_t)K * 4 + 32);
@@ -621,13 +626,17 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
                      (const float*)g2, y, (const u16*)res, N, K, stage,     \
                      act, eps, out_f32, softcap, escale)
   if (xdir) {
-    if (nt && rpw == 2) GEMV8_CASE(true, 2, true);
+    if (nt && rpw == 4) GEMV8_CASE(true, 4, true);
+    else if (nt && rpw == 2) GEMV8_CASE(true, 2, true);
     else if (nt) GEMV8_CASE(true, 1, true);
+    else if (rpw == 4) GEMV8_CASE(false, 4, true);
     else if (rpw == 2) GEMV8_CASE(false, 2, true);
     else GEMV8_CASE(false, 1, true);
   } else {
-    if (nt && rpw == 2) GEMV8_CASE(true, 2, false);
+    if (nt && rpw == 4) GEMV8_CASE(true, 4, false);
+    else if (nt && rpw == 2) GEMV8_CASE(true, 2, false);
     else if (nt) GEMV8_CASE(true, 1, false);
+    else if (rpw == 4) GEMV8_CASE(false, 4, false);
     else if (rpw == 2) GEMV8_CASE(false, 2, false);
     else GEMV8_CASE(false, 1, false);
   }
