@@ -409,6 +409,14 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
                                        int nt, int rpw, int maxblocks,
                                        float escale, hipStream_t stream) {
   size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 32);
+  static int rpw_env_b = -1;
+  if (rpw_env_b < 0) {
+    const char* p = getenv("LLM_GEMV_RPW");
+    rpw_env_b = p ? atoi(p) : 0;
+  }
+  if (rpw_env_b > 0) rpw = rpw_env_b;
+  else if (rpw <= 1)
+    rpw = (K >= 3584 && N >= 3072) ? 2 : 1;  // same policy as fp8 GEMV
   int threads = 256;  // 512 measured slower (fp8 1391->1328)
   int wpb = threads / 64;
   int blocks = (N + wpb * rpw - 1) / (wpb * rpw);
