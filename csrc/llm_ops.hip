@@ -653,6 +653,243 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
 }
 
 // ====================================================================
+// Multi-x fp8 GEMV (batched decode, B = 2..8 lockstep sequences):
+// ONE non-temporal weight stream feeds B accumulators — weight reuse
+// turns B-way decode into ~single-sequence step time (the 128-row MFMA
+// GEMM path wastes >90% of each tile at B<=8).  All staging modes of
+// the single-x GEMV, applied per row; XF32 stages fp32 pairs for
+// v_pk_fma_f32 (LDS B*K*4), falling back to bf16 staging (B*K*2) for
+// deep-K shapes.  BT = compile-time row capacity (2/4/8); the actual
+// B <= BT, with clamped source rows.
+// ====================================================================
+
+// per-row staging into xs + b*K (f32 or bf16 element type)
+template <bool XF32>
+DEVINL void gemv_stage_mx_row(char* smem, int b, const u16* xb,
+                              const u16* x2b, const float* g,
+                              const float* g2, u16* houtb, int K, int BT,
+                              int stage, int act, float eps, float escale,
+                              float* red) {
+  const int STRIDE = blockDim.x * 8;
+  float* xf = (float*)smem + (size_t)b * K;
+  u16* xh = (u16*)smem + (size_t)b * K;
+#define MXSTORE(i, v) do { if (XF32) xf[i] = (v); else xh[i] = f2b(v); } while (0)
+  if (stage == STAGE_RAW || stage == STAGE_NORM ||
+      stage == STAGE_NORM_EMBED) {
+    float ss = 0.f;
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
+      s8v v = *(const s8v*)(xb + i);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = b2f(((u16*)&v)[j]);
+        if (stage == STAGE_NORM_EMBED) {
+          f = b2f(f2b(f * escale));
+          if (blockIdx.x == 0) houtb[i + j] = f2b(f);
+        }
+        MXSTORE(i + j, f);
+        ss += f * f;
+      }
+    }
+    if (stage == STAGE_RAW) return;
+    ss = wave_reduce_sum(ss);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+    __syncthreads();
+    float rnorm = rsqrtf(stage_red_sum(red) / (float)K + eps);
+    __syncthreads();  // red reused by the next row's reduction
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = (XF32 ? xf[i + j] : b2f(xh[i + j])) * rnorm * g[i + j];
+        MXSTORE(i + j, f);
+      }
+    }
+  } else if (stage == STAGE_GLU) {
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
+      s8v gv = *(const s8v*)(xb + i);
+      s8v uv = *(const s8v*)(x2b + i);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float xx = b2f(((u16*)&gv)[j]);
+        float a;
+        if (act == 0) {
+          a = xx / (1.f + __expf(-xx));
+        } else {
+          float c = 0.7978845608028654f * (xx + 0.044715f * xx * xx * xx);
+          a = 0.5f * xx * (1.f + tanhf(c));
+        }
+        MXSTORE(i + j, a * b2f(((u16*)&uv)[j]));
+      }
+    }
+  } else if (stage == STAGE_NORM2) {
+    const u16* h = x2b;
+    float ss = 0.f;
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
+      s8v v = *(const s8v*)(xb + i);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = b2f(((u16*)&v)[j]);
+        ss += f * f;
+      }
+    }
+    ss = wave_reduce_sum(ss);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+    __syncthreads();
+    float rnorm_a = rsqrtf(stage_red_sum(red) / (float)K + eps);
+    __syncthreads();
+    float ss2 = 0.f;
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
+      s8v v = *(const s8v*)(xb + i);
+      s8v hv = *(const s8v*)(h + i);
+      f4v ga = *(const f4v*)(g + i);
+      f4v gb = *(const f4v*)(g + i + 4);
+      u16 o[8];
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = b2f(((u16*)&v)[j]) * rnorm_a * (j < 4 ? ga[j] : gb[j - 4])
+                  + b2f(((u16*)&hv)[j]);
+        o[j] = f2b(f);
+        float fr = b2f(o[j]);
+        MXSTORE(i + j, fr);
+        ss2 += fr * fr;
+      }
+      if (blockIdx.x == 0) *(s8v*)(houtb + i) = *(s8v*)o;
+    }
+    ss2 = wave_reduce_sum(ss2);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss2;
+    __syncthreads();
+    float rnorm_b = rsqrtf(stage_red_sum(red) / (float)K + eps);
+    __syncthreads();
+    for (int i = threadIdx.x * 8; i < K; i += STRIDE) {
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = (XF32 ? xf[i + j] : b2f(xh[i + j])) * rnorm_b * g2[i + j];
+        MXSTORE(i + j, f);
+      }
+    }
+  }
+#undef MXSTORE
+}
+
+template <int BT, bool XF32>
+__global__ void __launch_bounds__(256)
+k_gemv_fp8_mx(const uint8_t* __restrict__ W,
+              const float* __restrict__ scales, const u16* __restrict__ x,
+              long xstride, const u16* __restrict__ x2, long x2stride,
+              const float* __restrict__ g, const float* __restrict__ g2,
+              void* __restrict__ y, long ystride,
+              const u16* __restrict__ res, long rstride,
+              u16* __restrict__ hout, long hstride, int N, int K, int B,
+              int stage, int act, float eps, int out_f32, float softcap,
+              float escale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* red = (float*)(smem + (size_t)BT * K * (XF32 ? 4 : 2));
+  for (int b = 0; b < BT; b++) {
+    int bs = b < B ? b : B - 1;  // clamp padded rows to a valid source
+    const u16* xb = x + (size_t)bs * xstride;
+    if (stage == STAGE_NORM_EMBED)  // x = table, x2 = per-row token ids
+      xb = x + (size_t)((const int*)x2)[bs] * (size_t)K;
+    const u16* x2b = (stage == STAGE_NORM_EMBED || x2 == nullptr)
+                         ? nullptr : x2 + (size_t)bs * x2stride;
+    u16* houtb = hout ? hout + (size_t)bs * hstride : nullptr;
+    __syncthreads();
+    gemv_stage_mx_row<XF32>(smem, b, xb, x2b, g, g2, houtb, K, BT, stage,
+                            act, eps, escale, red);
+  }
+  __syncthreads();
+
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int wpb = blockDim.x >> 6;
+  const float* xsf = (const float*)smem;
+  const u16* xsh = (const u16*)smem;
+  for (int row = blockIdx.x * wpb + wave; row < N; row += gridDim.x * wpb) {
+    const uint8_t* Wr = W + (size_t)row * K;
+    f2v a0[BT], a1[BT];
+#pragma unroll
+    for (int b = 0; b < BT; b++) {
+      a0[b] = (f2v){0.f, 0.f};
+      a1[b] = (f2v){0.f, 0.f};
+    }
+    for (int k = lane * 16; k < K; k += 1024) {
+      u4v w0 = __builtin_nontemporal_load((const u4v*)(Wr + k));
+#pragma unroll
+      for (int q = 0; q < 4; q++) {
+        f2v ca = __builtin_amdgcn_cvt_pk_f32_fp8(w0[q], false);
+        f2v cb = __builtin_amdgcn_cvt_pk_f32_fp8(w0[q], true);
+#pragma unroll
+        for (int b = 0; b < BT; b++) {
+          f2v xa, xb2;
+          if (XF32) {
+            xa = *(const f2v*)(xsf + (size_t)b * K + k + q * 4);
+            xb2 = *(const f2v*)(xsf + (size_t)b * K + k + q * 4 + 2);
+          } else {
+            const u16* p = xsh + (size_t)b * K + k + q * 4;
+            xa = (f2v){b2f(p[0]), b2f(p[1])};
+            xb2 = (f2v){b2f(p[2]), b2f(p[3])};
+          }
+          a0[b] += ca * xa;
+          a1[b] += cb * xb2;
+        }
+      }
+    }
+    const float sc = scales[row];
+#pragma unroll
+    for (int b = 0; b < BT; b++) {
+      f2v s = a0[b] + a1[b];
+      float acc = wave_reduce_sum(s[0] + s[1]) * sc;
+      if (lane == 0 && b < B) {
+        if (softcap > 0.f) acc = softcap * tanhf(acc / softcap);
+        if (res && stage != STAGE_NORM2 && stage != STAGE_NORM_EMBED)
+          acc += b2f(res[(size_t)b * rstride + row]);
+        if (out_f32) ((float*)y)[(size_t)b * ystride + row] = acc;
+        else ((u16*)y)[(size_t)b * ystride + row] = f2b(acc);
+      }
+    }
+  }
+}
+
+extern "C" hipError_t launch_gemv_fp8_mx(
+    const void* W, const void* scales, const void* x, long xstride,
+    const void* x2, long x2stride, const void* g, const void* g2, void* y,
+    long ystride, const void* res, long rstride, void* hout, long hstride,
+    int N, int K, int B, int stage, int act, float eps, int out_f32,
+    float softcap, float escale, hipStream_t stream) {
+  if (B < 1 || B > 8 || K % 16 != 0) return hipErrorInvalidValue;
+  int BT = B <= 2 ? 2 : (B <= 4 ? 4 : 8);
+  bool xf32 = ((size_t)BT * K * 4 + 64) <= 128 * 1024;
+  size_t lds = (size_t)BT * K * (xf32 ? 4 : 2) + 64;
+  if (lds > 160 * 1024) return hipErrorInvalidValue;
+  static bool raised = false;
+  if (lds > 65536 && !raised) {
+#define MXRAISE(BTV, XFV)                                                   \
+    hipFuncSetAttribute((const void*)&k_gemv_fp8_mx<BTV, XFV>,              \
+                        hipFuncAttributeMaxDynamicSharedMemorySize,         \
+                        160 * 1024)
+    MXRAISE(2, true); MXRAISE(2, false);
+    MXRAISE(4, true); MXRAISE(4, false);
+    MXRAISE(8, true); MXRAISE(8, false);
+#undef MXRAISE
+    raised = true;
+  }
+  int threads = 256;
+  int wpb = threads / 64;
+  int blocks = (N + wpb - 1) / wpb;
+  if (blocks > 1024) blocks = 1024;
+#define MX_CASE(BTV, XFV)                                                   \
+  hipLaunchKernelGGL((k_gemv_fp8_mx<BTV, XFV>), dim3(blocks),               \
+                     dim3(threads), lds, stream, (const uint8_t*)W,         \
+                     (const float*)scales, (const u16*)x, xstride,          \
+                     (const u16*)x2, x2stride, (const float*)g,             \
+                     (const float*)g2, y, ystride, (const u16*)res,         \
+                     rstride, (u16*)hout, hstride, N, K, B, stage, act,     \
+                     eps, out_f32, softcap, escale)
+  if (BT == 2) { if (xf32) MX_CASE(2, true); else MX_CASE(2, false); }
+  else if (BT == 4) { if (xf32) MX_CASE(4, true); else MX_CASE(4, false); }
+  else { if (xf32) MX_CASE(8, true); else MX_CASE(8, false); }
+#undef MX_CASE
+  return hipGetLastError();
+}
+
+// ====================================================================
 // RMSNorm: mode 0: y = norm(x)*g ; mode 1: y = res + norm(x)*g
 // x bf16 [M,H]; g fp32[H] (Gemma's gamma+1 pre-folded on host); one block
 // per row; fused single pass (x kept in registers between reduce+scale).

@@ -343,6 +343,11 @@ class GPUModel:
                 B * self.nh_l * self.attn_split * (hd + 2),
                 dtype=torch.float32, device=dev)
             self.bt_attn_cnt = torch.zeros(B * self.nh_l, **i32)
+            # fused multi-x path scratch (fp8, B<=8)
+            self.bt_gu = torch.zeros(B, 2 * I, **bf)
+            self.bt_t1 = torch.zeros(B, H, **bf)
+            self.bt_t2 = torch.zeros(B, H, **bf)
+            self.bt_hb = torch.zeros(B, H, **bf)
 
         self.act = 0 if cfg.hidden_act == "silu" else 1
         self.gemma = cfg.model_type == "gemma2"
@@ -824,11 +829,14 @@ class GPUModel:
     # ------------------------------------------------------------------
     def _decode_batch_step(self, B: int, greedy: bool, min_p: float,
                            temperature: float = 1.0):
-        """One decode step for B lockstep sequences: the layer stack
-        runs with M=B rows (MFMA GEMMs amortize each weight read over
-        the batch), attention/sampling are per-row.  Graph-replayable
+        """One decode step for B lockstep sequences.  fp8 weights with
+        B<=8 take the fused MULTI-X GEMV path (one weight stream feeds
+        B accumulators — near single-sequence step time); otherwise the
+        layer stack runs as M=B-row MFMA GEMMs.  Graph-replayable
         (all state device-side)."""
         cfg = self.config
+        if self.fp8 and B <= 8 and not cfg.attention_bias:
+            return self._decode_batch_step_mx(B, greedy, min_p, temperature)
         ho.embed(self.embed, self.bt_next, self.b_h, B, cfg.embed_scale)
         self._layers_forward(B, batch_attn=True)
         ho.rmsnorm(self.b_h[:B], self.g_final, self.b_xn[:B],
@@ -843,6 +851,85 @@ class GPUModel:
                     accbuf=self.b_gemm_acc)
         if self.final_softcap:
             ho.softcap(self.bt_logits[:B], self.final_softcap)
+        ho.sample(self.bt_logits[:B], min_p, greedy, self.seed,
+                  self.rng_ctr, self.bt_gmax, self.bt_pick, self.bt_next,
+                  self.bt_ring, self.bt_nout, self.bt_lens, bump_len=True,
+                  temperature=temperature, cnt=self.bt_cnt, batch=B)
+
+    def _decode_batch_step_mx(self, B: int, greedy: bool, min_p: float,
+                              temperature: float = 1.0):
+        """Fused batched decode (fp8, B<=8): mirrors _decode_step's
+        4-5 kernels/layer with multi-x GEMVs over B rows."""
+        cfg = self.config
+        eps = cfg.rms_norm_eps
+        H, I = self.H, self.inter_l
+        h = self.b_h            # (PC, H): rows 0..B = per-seq h
+        hb = self.bt_hb         # gemma ping-pong
+        t1, t2, gu = self.bt_t1, self.bt_t2, self.bt_gu
+        qkvw = (self.nh_l + 2 * self.kvh_l) * self.hd
+        nhh = self.nh_l * self.hd
+        prev = None
+        for i, lw in enumerate(self.layers):
+            window = cfg.sliding_window if cfg.is_sliding(i) else 0
+            if self.gemma and prev is not None:
+                ho.gemv_fp8_mx(lw["wqkv_q"], lw["wqkv_s"], t2, self.bt_qkv,
+                               B, H, qkvw, stage=ho.STAGE_NORM2, x2=h,
+                               x2stride=H, g=prev["g_postffn"],
+                               g2=lw["g_in"], hout=hb, hstride=H, eps=eps)
+                h, hb = hb, h
+            elif i == 0:
+                ho.gemv_fp8_mx(lw["wqkv_q"], lw["wqkv_s"], self.embed,
+                               self.bt_qkv, B, 0, qkvw,
+                               stage=ho.STAGE_NORM_EMBED,
+                               x2=self.bt_next, g=lw["g_in"], hout=h,
+                               hstride=H, eps=eps,
+                               escale=cfg.embed_scale)
+            else:
+                ho.gemv_fp8_mx(lw["wqkv_q"], lw["wqkv_s"], h, self.bt_qkv,
+                               B, H, qkvw, stage=ho.STAGE_NORM,
+                               g=lw["g_in"], eps=eps)
+            prev = lw
+            ho.attn_dec(self.bt_qkv, self.k_cache[i], self.v_cache[i],
+                        self.b_att[0], self.bt_lens, self.cos_t, self.sin_t,
+                        self.bt_scratch, self.bt_attn_cnt,
+                        self.nh_l, self.kvh_l, self.hd, self.scale,
+                        softcap=self.attn_softcap, window=window or 0,
+                        split=self.attn_split, kS=self.k_scale[i],
+                        vS=self.v_scale[i], batch=B)
+            if self.gemma:
+                ho.gemv_fp8_mx(lw["wo_q"], lw["wo_s"], self.b_att, t1, B,
+                               nhh, H)
+                ho.gemv_fp8_mx(lw["wgu_q"], lw["wgu_s"], t1, gu, B, H,
+                               2 * I, stage=ho.STAGE_NORM2, x2=h,
+                               x2stride=H, g=lw["g_post"],
+                               g2=lw["g_preffn"], hout=hb, hstride=H,
+                               eps=eps)
+                h, hb = hb, h
+                ho.gemv_fp8_mx(lw["wdown_q"], lw["wdown_s"], gu, t2, B,
+                               2 * I, H, stage=ho.STAGE_GLU,
+                               x2=gu[:, I:], x2stride=2 * I, act=self.act)
+            else:
+                ho.gemv_fp8_mx(lw["wo_q"], lw["wo_s"], self.b_att, h, B,
+                               nhh, H, res=h, rstride=H)
+                ho.gemv_fp8_mx(lw["wgu_q"], lw["wgu_s"], h, gu, B, H,
+                               2 * I, stage=ho.STAGE_NORM, g=lw["g_post"],
+                               eps=eps)
+                ho.gemv_fp8_mx(lw["wdown_q"], lw["wdown_s"], gu, h, B,
+                               2 * I, H, stage=ho.STAGE_GLU, x2=gu[:, I:],
+                               x2stride=2 * I, act=self.act, res=h,
+                               rstride=H)
+        if self.gemma:
+            ho.gemv_fp8_mx(self.lm_head_q, self.lm_head_s, t2,
+                           self.bt_logits, B, H, self.vocab_l,
+                           stage=ho.STAGE_NORM2, x2=h, x2stride=H,
+                           g=self.layers[-1]["g_postffn"], g2=self.g_final,
+                           hout=hb, hstride=H, eps=eps,
+                           softcap=self.final_softcap)
+        else:
+            ho.gemv_fp8_mx(self.lm_head_q, self.lm_head_s, h,
+                           self.bt_logits, B, H, self.vocab_l,
+                           stage=ho.STAGE_NORM, g=self.g_final, eps=eps,
+                           softcap=self.final_softcap)
         ho.sample(self.bt_logits[:B], min_p, greedy, self.seed,
                   self.rng_ctr, self.bt_gmax, self.bt_pick, self.bt_next,
                   self.bt_ring, self.bt_nout, self.bt_lens, bump_len=True,

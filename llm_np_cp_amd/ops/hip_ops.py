@@ -70,6 +70,14 @@ _SIGS = {
     "launch_i32_add": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
     "launch_quant_fp8": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 +
                         [ctypes.c_void_p],
+    "launch_gemv_fp8_mx": [ctypes.c_void_p] * 3 + [ctypes.c_long] +
+                          [ctypes.c_void_p, ctypes.c_long] +
+                          [ctypes.c_void_p] * 3 + [ctypes.c_long] +
+                          [ctypes.c_void_p, ctypes.c_long] +
+                          [ctypes.c_void_p, ctypes.c_long] +
+                          [ctypes.c_int] * 5 +
+                          [ctypes.c_float, ctypes.c_int, ctypes.c_float,
+                           ctypes.c_float, ctypes.c_void_p],
     "launch_gemm_fp8": [ctypes.c_void_p] * 7 + [ctypes.c_int] * 3 +
                        [ctypes.c_void_p],
     # one-shot xGMI collectives (csrc/xgmi_comm.hip)
@@ -337,6 +345,29 @@ def gemm_fp8(xq: torch.Tensor, sx: torch.Tensor, Wq: torch.Tensor,
     _check(lib().launch_gemm_fp8(
         _ptr(xq), _ptr(sx), _ptr(Wq), _ptr(sw), _ptr(y), _ptr(res),
         _ptr(accbuf), M, N, K, _stream()), "gemm_fp8")
+
+
+def gemv_fp8_mx(Wq: torch.Tensor, scales: torch.Tensor, x: torch.Tensor,
+                y: torch.Tensor, B: int, xstride: int, ystride: int,
+                res: torch.Tensor | None = None, rstride: int = 0,
+                hout: torch.Tensor | None = None, hstride: int = 0,
+                softcap: float = 0.0, stage: int = 0,
+                x2: torch.Tensor | None = None, x2stride: int = 0,
+                g: torch.Tensor | None = None,
+                g2: torch.Tensor | None = None, act: int = 0,
+                eps: float = 1e-5, escale: float = 1.0):
+    """Multi-x fp8 GEMV: y[b, N] = scales * (Wq @ stage(x_b)) for B
+    lockstep rows — one weight stream, B accumulators.  Strides are in
+    ELEMENTS.  NORM_EMBED: x = embed table, x2 = int32 token ids."""
+    N, K = Wq.shape
+    out_f32 = 1 if y.dtype == torch.float32 else 0
+    _check(lib().launch_gemv_fp8_mx(
+        _ptr(Wq), _ptr(scales), _ptr(x), ctypes.c_long(xstride),
+        _ptr(x2), ctypes.c_long(x2stride), _ptr(g), _ptr(g2),
+        _ptr(y), ctypes.c_long(ystride), _ptr(res), ctypes.c_long(rstride),
+        _ptr(hout), ctypes.c_long(hstride), N, K, B, stage, act,
+        ctypes.c_float(eps), out_f32, ctypes.c_float(softcap),
+        ctypes.c_float(escale), _stream()), "gemv_fp8_mx")
 
 
 def prefetch(t: torch.Tensor, sink: torch.Tensor):

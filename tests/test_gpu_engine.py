@@ -511,3 +511,32 @@ def test_batch_decode_ragged_lengths():
     assert ids.shape == (3, N)
     for b in range(3):
         np.testing.assert_array_equal(ids[b], singles[b])
+
+
+@pytest.mark.parametrize("preset", ["tiny-llama", "tiny-gemma2"])
+def test_batch_decode_mx_path_matches_single_fp8(preset):
+    """The fused multi-x fp8 batch path (B<=8) must reproduce each
+    row's single-sequence fp8 rollout (ragged lengths)."""
+    import llm_np_cp_amd as L
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.engine import GPUModel
+
+    cfg = L.preset_config(preset)
+    w = random_weights(cfg, seed=90)
+    rng = np.random.default_rng(91)
+    lens = [6, 10, 7, 9]
+    prompts = [rng.integers(0, cfg.vocab_size, size=p) for p in lens]
+    N = 5
+
+    singles = []
+    single = GPUModel(cfg, dict(w), max_seq=64, dtype="fp8")
+    for p in prompts:
+        single.prefill(p)
+        singles.append(single.decode(N, greedy=True, use_graph=False))
+    del single
+    torch.cuda.empty_cache()
+
+    m = GPUModel(cfg, dict(w), max_seq=64, max_batch=4, dtype="fp8")
+    ids = m.generate_tokens_batch(prompts, N, greedy=True)
+    for b in range(4):
+        np.testing.assert_array_equal(ids[b], singles[b]), b
