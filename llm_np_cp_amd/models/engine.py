@@ -835,7 +835,12 @@ class GPUModel:
         layer stack runs as M=B-row MFMA GEMMs.  Graph-replayable
         (all state device-side)."""
         cfg = self.config
-        if self.fp8 and B <= 8 and not cfg.attention_bias:
+        # measured: the multi-x GEMV wins at B<=2 (1.44 ms vs ~2.4 GEMM
+        # at B=2) but is VALU-ISSUE-bound beyond that (64 irreducible
+        # v_pk_fma per 16 weight bytes at B=8 -> ~2 ms floor); B>=3
+        # stays on the MFMA GEMM path (matrix cores own the MACs).
+        # See profiles/decode_kernels_r02.md.
+        if self.fp8 and B <= 2 and not cfg.attention_bias:
             return self._decode_batch_step_mx(B, greedy, min_p, temperature)
         ho.embed(self.embed, self.bt_next, self.b_h, B, cfg.embed_scale)
         self._layers_forward(B, batch_attn=True)
