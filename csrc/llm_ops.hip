@@ -669,7 +669,7 @@ DEVINL void gemv_stage_mx_row(char* smem, int b, const u16* xb,
                               const u16* x2b, const float* g,
                               const float* g2, u16* houtb, int K, int BT,
                               int stage, int act, float eps, float escale,
-                              float* red) {
+                              float* red, bool persist) {
   const int STRIDE = blockDim.x * 8;
   float* xf = (float*)smem + (size_t)b * K;
   u16* xh = (u16*)smem + (size_t)b * K;
@@ -684,7 +684,7 @@ DEVINL void gemv_stage_mx_row(char* smem, int b, const u16* xb,
         float f = b2f(((u16*)&v)[j]);
         if (stage == STAGE_NORM_EMBED) {
           f = b2f(f2b(f * escale));
-          if (blockIdx.x == 0) houtb[i + j] = f2b(f);
+          if (persist) houtb[i + j] = f2b(f);
         }
         MXSTORE(i + j, f);
         ss += f * f;
@@ -752,7 +752,7 @@ DEVINL void gemv_stage_mx_row(char* smem, int b, const u16* xb,
         MXSTORE(i + j, fr);
         ss2 += fr * fr;
       }
-      if (blockIdx.x == 0) *(s8v*)(houtb + i) = *(s8v*)o;
+      if (persist) *(s8v*)(houtb + i) = *(s8v*)o;
     }
     ss2 = wave_reduce_sum(ss2);
     if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss2;
@@ -793,7 +793,7 @@ k_gemv_fp8_mx(const uint8_t* __restrict__ W,
     u16* houtb = hout ? hout + (size_t)bs * hstride : nullptr;
     __syncthreads();
     gemv_stage_mx_row<XF32>(smem, b, xb, x2b, g, g2, houtb, K, BT, stage,
-                            act, eps, escale, red);
+                            act, eps, escale, red, blockIdx.x == 0);
   }
   __syncthreads();
 
@@ -886,6 +886,155 @@ extern "C" hipError_t launch_gemv_fp8_mx(
   else if (BT == 4) { if (xf32) MX_CASE(4, true); else MX_CASE(4, false); }
   else { if (xf32) MX_CASE(8, true); else MX_CASE(8, false); }
 #undef MX_CASE
+  return hipGetLastError();
+}
+
+// ====================================================================
+// Skinny fp8 MFMA GEMM pair (batched decode, B = 3..16): matrix cores
+// own the MACs — the multi-x GEMV above is VALU-ISSUE-bound beyond B=2
+// (64 irreducible v_pk_fma per 16 weight bytes at B=8).
+//   1. k_stage_quant_mx: per sequence row, apply the staging op
+//      (RAW/NORM/GLU/NORM2/NORM_EMBED) and quantize to e4m3 + scale
+//      (one block per row; row stats need full-K visibility).
+//   2. k_gemm_fp8_skinny: Y[B,N] = (sx_b*sw_n) * Xq[B,K] @ Wq[N,K]^T
+//      with ONE 16-row mfma_f32_16x16x32_fp8_fp8 tile per wave — W is
+//      nt-streamed exactly once; Xq re-reads stay L2-hot (~N*K/4 B).
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_stage_quant_mx(const u16* __restrict__ x, long xstride,
+                 const u16* __restrict__ x2, long x2stride,
+                 const float* __restrict__ g, const float* __restrict__ g2,
+                 uint8_t* __restrict__ xq, float* __restrict__ sx,
+                 u16* __restrict__ hout, long hstride, int K, int B,
+                 int stage, int act, float eps, float escale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* red = (float*)(smem + (size_t)K * 4);
+  const int b = blockIdx.x;
+  const u16* xb = x + (size_t)b * xstride;
+  if (stage == STAGE_NORM_EMBED)
+    xb = x + (size_t)((const int*)x2)[b] * (size_t)K;
+  const u16* x2b = (stage == STAGE_NORM_EMBED || x2 == nullptr)
+                       ? nullptr : x2 + (size_t)b * x2stride;
+  u16* houtb = hout ? hout + (size_t)b * hstride : nullptr;
+  // stage into f32 LDS (row stats via the shared per-row helper)
+  gemv_stage_mx_row<true>(smem, 0, xb, x2b, g, g2, houtb, K, 1, stage,
+                          act, eps, escale, red, true);
+  __syncthreads();
+  // absmax -> scale -> quantize to global
+  const float* xf = (const float*)smem;
+  float am = 0.f;
+  for (int i = threadIdx.x; i < K; i += 256) am = fmaxf(am, fabsf(xf[i]));
+#pragma unroll
+  for (int w = 1; w < 64; w <<= 1) am = fmaxf(am, __shfl_xor(am, w));
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = am;
+  __syncthreads();
+  am = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+  float sc = fmaxf(am, 1e-8f) / 448.0f;
+  if (threadIdx.x == 0) sx[b] = sc;
+  const float rs = 1.0f / sc;
+  uint8_t* q = xq + (size_t)b * K;
+  for (int i = threadIdx.x * 8; i < K; i += 2048) {
+    uint32_t lo = 0, hi = 0;
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(xf[i] * rs, xf[i + 1] * rs, lo,
+                                         false);
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(xf[i + 2] * rs, xf[i + 3] * rs,
+                                         lo, true);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(xf[i + 4] * rs, xf[i + 5] * rs,
+                                         hi, false);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(xf[i + 6] * rs, xf[i + 7] * rs,
+                                         hi, true);
+    *(uint32_t*)(q + i) = lo;
+    *(uint32_t*)(q + i + 4) = hi;
+  }
+}
+
+extern "C" hipError_t launch_stage_quant_mx(
+    const void* x, long xstride, const void* x2, long x2stride,
+    const void* g, const void* g2, void* xq, void* sx, void* hout,
+    long hstride, int K, int B, int stage, int act, float eps,
+    float escale, hipStream_t stream) {
+  size_t lds = (size_t)K * 4 + 64;
+  if (lds > 160 * 1024 || K % 16 != 0) return hipErrorInvalidValue;
+  static bool raised = false;
+  if (lds > 65536 && !raised) {
+    hipFuncSetAttribute((const void*)&k_stage_quant_mx,
+                        hipFuncAttributeMaxDynamicSharedMemorySize,
+                        160 * 1024);
+    raised = true;
+  }
+  hipLaunchKernelGGL(k_stage_quant_mx, dim3(B), dim3(256), lds, stream,
+                     (const u16*)x, xstride, (const u16*)x2, x2stride,
+                     (const float*)g, (const float*)g2, (uint8_t*)xq,
+                     (float*)sx, (u16*)hout, hstride, K, B, stage, act,
+                     eps, escale);
+  return hipGetLastError();
+}
+
+// one 16x16 output tile per wave; acc row = X row (B dim), col = W row
+extern "C" __global__ void __launch_bounds__(256)
+k_gemm_fp8_skinny(const uint8_t* __restrict__ Xq,
+                  const float* __restrict__ sx,
+                  const uint8_t* __restrict__ Wq,
+                  const float* __restrict__ sw, void* __restrict__ y,
+                  long ystride, const u16* __restrict__ res, long rstride,
+                  const u16* __restrict__ bias, int out_f32, float softcap,
+                  int B, int N, int K) {
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int fr = lane & 15, fk8 = (lane >> 4) * 8;
+  const int asrc = fr < B ? fr : 0;  // padded A rows read row 0
+  const uint8_t* Xr = Xq + (size_t)asrc * K + fk8;
+  for (int n0 = (blockIdx.x * 4 + wave) * 16; n0 < N;
+       n0 += gridDim.x * 4 * 16) {
+    int wr = n0 + fr;
+    if (wr >= N) wr = N - 1;
+    const uint8_t* Wr = Wq + (size_t)wr * K + fk8;
+    f4v acc = {0.f, 0.f, 0.f, 0.f};
+    int k = 0;
+    for (; k + 64 <= K; k += 64) {  // 2-step unroll for load ILP
+      long a0 = *(const long*)(Xr + k);
+      long b0 = __builtin_nontemporal_load((const long*)(Wr + k));
+      long a1 = *(const long*)(Xr + k + 32);
+      long b1 = __builtin_nontemporal_load((const long*)(Wr + k + 32));
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a0, b0, acc, 0, 0,
+                                                       0);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a1, b1, acc, 0, 0,
+                                                       0);
+    }
+    for (; k < K; k += 32) {
+      long a0 = *(const long*)(Xr + k);
+      long b0 = __builtin_nontemporal_load((const long*)(Wr + k));
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a0, b0, acc, 0, 0,
+                                                       0);
+    }
+    const int col = n0 + fr;
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      const int b = (lane >> 4) * 4 + r;
+      if (b < B && col < N) {
+        float v = acc[r] * sx[b] * sw[col];
+        if (softcap > 0.f) v = softcap * tanhf(v / softcap);
+        if (bias) v += b2f(bias[col]);
+        if (res) v += b2f(res[(size_t)b * rstride + col]);
+        if (out_f32) ((float*)y)[(size_t)b * ystride + col] = v;
+        else ((u16*)y)[(size_t)b * ystride + col] = f2b(v);
+      }
+    }
+  }
+}
+
+extern "C" hipError_t launch_gemm_fp8_skinny(
+    const void* Xq, const void* sx, const void* Wq, const void* sw,
+    void* y, long ystride, const void* res, long rstride, const void* bias,
+    int out_f32, float softcap, int B, int N, int K, hipStream_t stream) {
+  if (B < 1 || B > 16 || K % 32 != 0) return hipErrorInvalidValue;
+  int blocks = (N + 63) / 64;
+  if (blocks > 1024) blocks = 1024;
+  hipLaunchKernelGGL(k_gemm_fp8_skinny, dim3(blocks), dim3(256), 0, stream,
+                     (const uint8_t*)Xq, (const float*)sx,
+                     (const uint8_t*)Wq, (const float*)sw, y, ystride,
+                     (const u16*)res, rstride, (const u16*)bias, out_f32,
+                     softcap, B, N, K);
   return hipGetLastError();
 }
 

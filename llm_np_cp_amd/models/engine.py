@@ -837,11 +837,14 @@ class GPUModel:
         cfg = self.config
         # measured: the multi-x GEMV wins at B<=2 (1.44 ms vs ~2.4 GEMM
         # at B=2) but is VALU-ISSUE-bound beyond that (64 irreducible
-        # v_pk_fma per 16 weight bytes at B=8 -> ~2 ms floor); B>=3
-        # stays on the MFMA GEMM path (matrix cores own the MACs).
-        # See profiles/decode_kernels_r02.md.
+        # v_pk_fma per 16 weight bytes at B=8 -> ~2 ms floor); B=3..16
+        # takes the SKINNY fp8 MFMA GEMM (matrix cores own the MACs,
+        # W nt-streamed once).  See profiles/decode_kernels_r02.md.
         if self.fp8 and B <= 2 and not cfg.attention_bias:
             return self._decode_batch_step_mx(B, greedy, min_p, temperature)
+        if self.fp8 and B <= 16:
+            return self._decode_batch_step_skinny(B, greedy, min_p,
+                                                  temperature)
         ho.embed(self.embed, self.bt_next, self.b_h, B, cfg.embed_scale)
         self._layers_forward(B, batch_attn=True)
         ho.rmsnorm(self.b_h[:B], self.g_final, self.b_xn[:B],
@@ -934,6 +937,90 @@ class GPUModel:
             ho.gemv_fp8_mx(self.lm_head_q, self.lm_head_s, h,
                            self.bt_logits, B, H, self.vocab_l,
                            stage=ho.STAGE_NORM, g=self.g_final, eps=eps,
+                           softcap=self.final_softcap)
+        ho.sample(self.bt_logits[:B], min_p, greedy, self.seed,
+                  self.rng_ctr, self.bt_gmax, self.bt_pick, self.bt_next,
+                  self.bt_ring, self.bt_nout, self.bt_lens, bump_len=True,
+                  temperature=temperature, cnt=self.bt_cnt, batch=B)
+
+    def _decode_batch_step_skinny(self, B: int, greedy: bool,
+                                  min_p: float, temperature: float = 1.0):
+        """Batched decode via (stage+quant, skinny fp8 MFMA GEMM) pairs:
+        B = 3..16 lockstep rows, W streamed once per projection."""
+        cfg = self.config
+        eps = cfg.rms_norm_eps
+        H, I = self.H, self.inter_l
+        h = self.b_h
+        hb = self.bt_hb
+        t1, t2, gu = self.bt_t1, self.bt_t2, self.bt_gu
+        qkvw = (self.nh_l + 2 * self.kvh_l) * self.hd
+        nhh = self.nh_l * self.hd
+        xq, sx = self.b_xq, self.b_sx
+        prev = None
+        for i, lw in enumerate(self.layers):
+            window = cfg.sliding_window if cfg.is_sliding(i) else 0
+            if self.gemma and prev is not None:
+                ho.stage_quant_mx(t2, H, xq, sx, B, H, stage=ho.STAGE_NORM2,
+                                  x2=h, x2stride=H, g=prev["g_postffn"],
+                                  g2=lw["g_in"], hout=hb, hstride=H,
+                                  eps=eps)
+                h, hb = hb, h
+            elif i == 0:
+                ho.stage_quant_mx(self.embed, 0, xq, sx, B, H,
+                                  stage=ho.STAGE_NORM_EMBED,
+                                  x2=self.bt_next, g=lw["g_in"], hout=h,
+                                  hstride=H, eps=eps,
+                                  escale=cfg.embed_scale)
+            else:
+                ho.stage_quant_mx(h, H, xq, sx, B, H, stage=ho.STAGE_NORM,
+                                  g=lw["g_in"], eps=eps)
+            ho.gemm_fp8_skinny(xq, sx, lw["wqkv_q"], lw["wqkv_s"],
+                               self.bt_qkv, B, qkvw, bias=lw.get("bqkv"))
+            prev = lw
+            ho.attn_dec(self.bt_qkv, self.k_cache[i], self.v_cache[i],
+                        self.b_att[0], self.bt_lens, self.cos_t, self.sin_t,
+                        self.bt_scratch, self.bt_attn_cnt,
+                        self.nh_l, self.kvh_l, self.hd, self.scale,
+                        softcap=self.attn_softcap, window=window or 0,
+                        split=self.attn_split, kS=self.k_scale[i],
+                        vS=self.v_scale[i], batch=B)
+            ho.stage_quant_mx(self.b_att, nhh, xq, sx, B, nhh)
+            if self.gemma:
+                ho.gemm_fp8_skinny(xq, sx, lw["wo_q"], lw["wo_s"], t1, B, H)
+                ho.stage_quant_mx(t1, H, xq, sx, B, H,
+                                  stage=ho.STAGE_NORM2, x2=h, x2stride=H,
+                                  g=lw["g_post"], g2=lw["g_preffn"],
+                                  hout=hb, hstride=H, eps=eps)
+                h, hb = hb, h
+                ho.gemm_fp8_skinny(xq, sx, lw["wgu_q"], lw["wgu_s"], gu, B,
+                                   2 * I)
+                ho.stage_quant_mx(gu, 2 * I, xq, sx, B, I,
+                                  stage=ho.STAGE_GLU, x2=gu[:, I:],
+                                  x2stride=2 * I, act=self.act)
+                ho.gemm_fp8_skinny(xq, sx, lw["wdown_q"], lw["wdown_s"],
+                                   t2, B, H)
+            else:
+                ho.gemm_fp8_skinny(xq, sx, lw["wo_q"], lw["wo_s"], h, B, H,
+                                   res=h, rstride=H)
+                ho.stage_quant_mx(h, H, xq, sx, B, H, stage=ho.STAGE_NORM,
+                                  g=lw["g_post"], eps=eps)
+                ho.gemm_fp8_skinny(xq, sx, lw["wgu_q"], lw["wgu_s"], gu, B,
+                                   2 * I)
+                ho.stage_quant_mx(gu, 2 * I, xq, sx, B, I,
+                                  stage=ho.STAGE_GLU, x2=gu[:, I:],
+                                  x2stride=2 * I, act=self.act)
+                ho.gemm_fp8_skinny(xq, sx, lw["wdown_q"], lw["wdown_s"], h,
+                                   B, H, res=h, rstride=H)
+        if self.gemma:
+            ho.stage_quant_mx(t2, H, xq, sx, B, H, stage=ho.STAGE_NORM2,
+                              x2=h, x2stride=H,
+                              g=self.layers[-1]["g_postffn"],
+                              g2=self.g_final, hout=hb, hstride=H, eps=eps)
+        else:
+            ho.stage_quant_mx(h, H, xq, sx, B, H, stage=ho.STAGE_NORM,
+                              g=self.g_final, eps=eps)
+        ho.gemm_fp8_skinny(xq, sx, self.lm_head_q, self.lm_head_s,
+                           self.bt_logits, B, self.vocab_l,
                            softcap=self.final_softcap)
         ho.sample(self.bt_logits[:B], min_p, greedy, self.seed,
                   self.rng_ctr, self.bt_gmax, self.bt_pick, self.bt_next,

@@ -70,6 +70,18 @@ _SIGS = {
     "launch_i32_add": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
     "launch_quant_fp8": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 +
                         [ctypes.c_void_p],
+    "launch_stage_quant_mx": [ctypes.c_void_p, ctypes.c_long,
+                              ctypes.c_void_p, ctypes.c_long] +
+                             [ctypes.c_void_p] * 4 +
+                             [ctypes.c_void_p, ctypes.c_long] +
+                             [ctypes.c_int] * 4 +
+                             [ctypes.c_float, ctypes.c_float,
+                              ctypes.c_void_p],
+    "launch_gemm_fp8_skinny": [ctypes.c_void_p] * 5 + [ctypes.c_long] +
+                              [ctypes.c_void_p, ctypes.c_long] +
+                              [ctypes.c_void_p, ctypes.c_int,
+                               ctypes.c_float] + [ctypes.c_int] * 3 +
+                              [ctypes.c_void_p],
     "launch_gemv_fp8_mx": [ctypes.c_void_p] * 3 + [ctypes.c_long] +
                           [ctypes.c_void_p, ctypes.c_long] +
                           [ctypes.c_void_p] * 3 + [ctypes.c_long] +
@@ -368,6 +380,38 @@ def gemv_fp8_mx(Wq: torch.Tensor, scales: torch.Tensor, x: torch.Tensor,
         _ptr(hout), ctypes.c_long(hstride), N, K, B, stage, act,
         ctypes.c_float(eps), out_f32, ctypes.c_float(softcap),
         ctypes.c_float(escale), _stream()), "gemv_fp8_mx")
+
+
+def stage_quant_mx(x: torch.Tensor, xstride: int, xq: torch.Tensor,
+                   sx: torch.Tensor, B: int, K: int, stage: int = 0,
+                   x2: torch.Tensor | None = None, x2stride: int = 0,
+                   g: torch.Tensor | None = None,
+                   g2: torch.Tensor | None = None,
+                   hout: torch.Tensor | None = None, hstride: int = 0,
+                   act: int = 0, eps: float = 1e-5, escale: float = 1.0):
+    """Per-sequence-row staging op + e4m3 quantization -> xq[B,K], sx[B]
+    (feeds the skinny fp8 MFMA GEMM).  Strides in elements."""
+    _check(lib().launch_stage_quant_mx(
+        _ptr(x), ctypes.c_long(xstride), _ptr(x2), ctypes.c_long(x2stride),
+        _ptr(g), _ptr(g2), _ptr(xq), _ptr(sx), _ptr(hout),
+        ctypes.c_long(hstride), K, B, stage, act, ctypes.c_float(eps),
+        ctypes.c_float(escale), _stream()), "stage_quant_mx")
+
+
+def gemm_fp8_skinny(xq: torch.Tensor, sx: torch.Tensor, Wq: torch.Tensor,
+                    sw: torch.Tensor, y: torch.Tensor, B: int,
+                    ystride: int, res: torch.Tensor | None = None,
+                    rstride: int = 0, bias: torch.Tensor | None = None,
+                    softcap: float = 0.0):
+    """Y[B<=16, N] = (sx_b*sw_n) * Xq @ Wq^T: one 16-row fp8 MFMA tile
+    per wave, W nt-streamed once (batched decode B=3..16)."""
+    N, K = Wq.shape
+    out_f32 = 1 if y.dtype == torch.float32 else 0
+    _check(lib().launch_gemm_fp8_skinny(
+        _ptr(xq), _ptr(sx), _ptr(Wq), _ptr(sw), _ptr(y),
+        ctypes.c_long(ystride), _ptr(res), ctypes.c_long(rstride),
+        _ptr(bias), out_f32, ctypes.c_float(softcap), B, N, K,
+        _stream()), "gemm_fp8_skinny")
 
 
 def prefetch(t: torch.Tensor, sink: torch.Tensor):

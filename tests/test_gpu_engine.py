@@ -513,10 +513,15 @@ def test_batch_decode_ragged_lengths():
         np.testing.assert_array_equal(ids[b], singles[b])
 
 
-@pytest.mark.parametrize("preset", ["tiny-llama", "tiny-gemma2"])
-def test_batch_decode_mx_path_matches_single_fp8(preset):
-    """The fused multi-x fp8 batch path (B<=8) must reproduce each
-    row's single-sequence fp8 rollout (ragged lengths)."""
+@pytest.mark.parametrize("preset,B", [("tiny-llama", 2),
+                                      ("tiny-gemma2", 2),
+                                      ("tiny-llama", 4),
+                                      ("tiny-gemma2", 4),
+                                      ("tiny-llama", 7)])
+def test_batch_decode_fused_paths_match_single_fp8(preset, B):
+    """The fused fp8 batch paths (multi-x GEMV at B<=2, skinny MFMA
+    GEMM at B=3..16) must reproduce each row's single-sequence fp8
+    rollout (ragged lengths)."""
     import llm_np_cp_amd as L
     from llm_np_cp_amd.io.loader import random_weights
     from llm_np_cp_amd.models.engine import GPUModel
@@ -524,7 +529,7 @@ def test_batch_decode_mx_path_matches_single_fp8(preset):
     cfg = L.preset_config(preset)
     w = random_weights(cfg, seed=90)
     rng = np.random.default_rng(91)
-    lens = [6, 10, 7, 9]
+    lens = ([6, 10, 7, 9, 5, 8, 11])[:B]
     prompts = [rng.integers(0, cfg.vocab_size, size=p) for p in lens]
     N = 5
 
@@ -536,7 +541,7 @@ def test_batch_decode_mx_path_matches_single_fp8(preset):
     del single
     torch.cuda.empty_cache()
 
-    m = GPUModel(cfg, dict(w), max_seq=64, max_batch=4, dtype="fp8")
+    m = GPUModel(cfg, dict(w), max_seq=64, max_batch=B, dtype="fp8")
     ids = m.generate_tokens_batch(prompts, N, greedy=True)
-    for b in range(4):
+    for b in range(B):
         np.testing.assert_array_equal(ids[b], singles[b]), b
