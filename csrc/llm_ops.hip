@@ -901,6 +901,8 @@ extern "C" hipError_t launch_gemv_fp8_mx(
 //      nt-streamed exactly once; Xq re-reads stay L2-hot (~N*K/4 B).
 // ====================================================================
 
+extern "C" __global__ void k_zero_f32(float* p, long total);
+
 extern "C" __global__ void __launch_bounds__(256)
 k_stage_quant_mx(const u16* __restrict__ x, long xstride,
                  const u16* __restrict__ x2, long x2stride,
@@ -971,70 +973,132 @@ extern "C" hipError_t launch_stage_quant_mx(
   return hipGetLastError();
 }
 
-// one 16x16 output tile per wave; acc row = X row (B dim), col = W row
+// one 16x16 output tile per wave; acc row = X row (B dim), col = W row.
+// TWO interleaved 64-byte K-chains per wave (independent mfma
+// accumulators — the dependent-chain latency was the v1 bound) and a
+// K-split grid.y for small-N occupancy (atomicAdd into an fp32 accbuf,
+// scales/epilogue applied by k_skinny_fin).
 extern "C" __global__ void __launch_bounds__(256)
 k_gemm_fp8_skinny(const uint8_t* __restrict__ Xq,
                   const float* __restrict__ sx,
                   const uint8_t* __restrict__ Wq,
                   const float* __restrict__ sw, void* __restrict__ y,
                   long ystride, const u16* __restrict__ res, long rstride,
-                  const u16* __restrict__ bias, int out_f32, float softcap,
-                  int B, int N, int K) {
+                  const u16* __restrict__ bias, float* __restrict__ accbuf,
+                  int out_f32, float softcap, int B, int N, int K) {
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int fr = lane & 15, fk8 = (lane >> 4) * 8;
   const int asrc = fr < B ? fr : 0;  // padded A rows read row 0
   const uint8_t* Xr = Xq + (size_t)asrc * K + fk8;
+  const int SK = gridDim.y;
+  const int kchunks = (K / 32 + SK - 1) / SK;
+  const int k_lo = blockIdx.y * kchunks * 32;
+  int k_hi = k_lo + kchunks * 32;
+  if (k_hi > K) k_hi = K;
+  if (k_hi <= k_lo) return;
   for (int n0 = (blockIdx.x * 4 + wave) * 16; n0 < N;
        n0 += gridDim.x * 4 * 16) {
     int wr = n0 + fr;
     if (wr >= N) wr = N - 1;
     const uint8_t* Wr = Wq + (size_t)wr * K + fk8;
-    f4v acc = {0.f, 0.f, 0.f, 0.f};
-    int k = 0;
-    for (; k + 64 <= K; k += 64) {  // 2-step unroll for load ILP
+    f4v acc_a = {0.f, 0.f, 0.f, 0.f};
+    f4v acc_b = {0.f, 0.f, 0.f, 0.f};
+    int k = k_lo;
+    for (; k + 128 <= k_hi; k += 128) {  // 2 independent 64-B chains
       long a0 = *(const long*)(Xr + k);
       long b0 = __builtin_nontemporal_load((const long*)(Wr + k));
       long a1 = *(const long*)(Xr + k + 32);
       long b1 = __builtin_nontemporal_load((const long*)(Wr + k + 32));
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a0, b0, acc, 0, 0,
-                                                       0);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a1, b1, acc, 0, 0,
-                                                       0);
+      long a2 = *(const long*)(Xr + k + 64);
+      long b2 = __builtin_nontemporal_load((const long*)(Wr + k + 64));
+      long a3 = *(const long*)(Xr + k + 96);
+      long b3 = __builtin_nontemporal_load((const long*)(Wr + k + 96));
+      acc_a = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a0, b0, acc_a,
+                                                         0, 0, 0);
+      acc_b = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a1, b1, acc_b,
+                                                         0, 0, 0);
+      acc_a = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a2, b2, acc_a,
+                                                         0, 0, 0);
+      acc_b = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a3, b3, acc_b,
+                                                         0, 0, 0);
     }
-    for (; k < K; k += 32) {
+    for (; k < k_hi; k += 32) {
       long a0 = *(const long*)(Xr + k);
       long b0 = __builtin_nontemporal_load((const long*)(Wr + k));
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a0, b0, acc, 0, 0,
-                                                       0);
+      acc_a = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a0, b0, acc_a,
+                                                         0, 0, 0);
     }
     const int col = n0 + fr;
 #pragma unroll
     for (int r = 0; r < 4; r++) {
       const int b = (lane >> 4) * 4 + r;
       if (b < B && col < N) {
-        float v = acc[r] * sx[b] * sw[col];
-        if (softcap > 0.f) v = softcap * tanhf(v / softcap);
-        if (bias) v += b2f(bias[col]);
-        if (res) v += b2f(res[(size_t)b * rstride + col]);
-        if (out_f32) ((float*)y)[(size_t)b * ystride + col] = v;
-        else ((u16*)y)[(size_t)b * ystride + col] = f2b(v);
+        float v = acc_a[r] + acc_b[r];
+        if (SK > 1) {
+          atomicAdd(accbuf + (size_t)b * N + col, v);
+        } else {
+          v *= sx[b] * sw[col];
+          if (softcap > 0.f) v = softcap * tanhf(v / softcap);
+          if (bias) v += b2f(bias[col]);
+          if (res) v += b2f(res[(size_t)b * rstride + col]);
+          if (out_f32) ((float*)y)[(size_t)b * ystride + col] = v;
+          else ((u16*)y)[(size_t)b * ystride + col] = f2b(v);
+        }
       }
     }
   }
 }
 
+// SK>1 epilogue: raw sums -> scales/softcap/bias/res -> y
+extern "C" __global__ void __launch_bounds__(256)
+k_skinny_fin(const float* __restrict__ accbuf, const float* __restrict__ sx,
+             const float* __restrict__ sw, void* __restrict__ y,
+             long ystride, const u16* __restrict__ res, long rstride,
+             const u16* __restrict__ bias, int out_f32, float softcap,
+             int B, int N) {
+  const long i = (long)blockIdx.x * 256 + threadIdx.x;
+  if (i >= (long)B * N) return;
+  const int b = (int)(i / N), col = (int)(i % N);
+  float v = accbuf[(size_t)b * N + col] * sx[b] * sw[col];
+  if (softcap > 0.f) v = softcap * tanhf(v / softcap);
+  if (bias) v += b2f(bias[col]);
+  if (res) v += b2f(res[(size_t)b * rstride + col]);
+  if (out_f32) ((float*)y)[(size_t)b * ystride + col] = v;
+  else ((u16*)y)[(size_t)b * ystride + col] = f2b(v);
+}
+
 extern "C" hipError_t launch_gemm_fp8_skinny(
     const void* Xq, const void* sx, const void* Wq, const void* sw,
     void* y, long ystride, const void* res, long rstride, const void* bias,
-    int out_f32, float softcap, int B, int N, int K, hipStream_t stream) {
+    void* accbuf, int out_f32, float softcap, int B, int N, int K,
+    hipStream_t stream) {
   if (B < 1 || B > 16 || K % 32 != 0) return hipErrorInvalidValue;
   int blocks = (N + 63) / 64;
   if (blocks > 1024) blocks = 1024;
-  hipLaunchKernelGGL(k_gemm_fp8_skinny, dim3(blocks), dim3(256), 0, stream,
-                     (const uint8_t*)Xq, (const float*)sx,
+  int sk = 1;
+  if (accbuf && blocks < 512) {
+    sk = 512 / blocks;
+    if (sk > 8) sk = 8;
+    while (sk > 1 && K / 32 < sk) sk--;
+  }
+  if (sk > 1) {
+    long total = (long)B * N;
+    hipLaunchKernelGGL(k_zero_f32, dim3((uint32_t)((total / 4 + 255) / 256)),
+                       dim3(256), 0, stream, (float*)accbuf, total);
+  }
+  hipLaunchKernelGGL(k_gemm_fp8_skinny, dim3(blocks, sk), dim3(256), 0,
+                     stream, (const uint8_t*)Xq, (const float*)sx,
                      (const uint8_t*)Wq, (const float*)sw, y, ystride,
-                     (const u16*)res, rstride, (const u16*)bias, out_f32,
-                     softcap, B, N, K);
+                     (const u16*)res, rstride, (const u16*)bias,
+                     (float*)accbuf, out_f32, softcap, B, N, K);
+  if (sk > 1) {
+    long total = (long)B * N;
+    hipLaunchKernelGGL(k_skinny_fin, dim3((uint32_t)((total + 255) / 256)),
+                       dim3(256), 0, stream, (const float*)accbuf,
+                       (const float*)sx, (const float*)sw, y, ystride,
+                       (const u16*)res, rstride, (const u16*)bias, out_f32,
+                       softcap, B, N);
+  }
   return hipGetLastError();
 }
 

@@ -975,7 +975,8 @@ class GPUModel:
                 ho.stage_quant_mx(h, H, xq, sx, B, H, stage=ho.STAGE_NORM,
                                   g=lw["g_in"], eps=eps)
             ho.gemm_fp8_skinny(xq, sx, lw["wqkv_q"], lw["wqkv_s"],
-                               self.bt_qkv, B, qkvw, bias=lw.get("bqkv"))
+                               self.bt_qkv, B, qkvw, bias=lw.get("bqkv"),
+                               accbuf=self.b_gemm_acc)
             prev = lw
             ho.attn_dec(self.bt_qkv, self.k_cache[i], self.v_cache[i],
                         self.b_att[0], self.bt_lens, self.cos_t, self.sin_t,
@@ -986,31 +987,34 @@ class GPUModel:
                         vS=self.v_scale[i], batch=B)
             ho.stage_quant_mx(self.b_att, nhh, xq, sx, B, nhh)
             if self.gemma:
-                ho.gemm_fp8_skinny(xq, sx, lw["wo_q"], lw["wo_s"], t1, B, H)
+                ho.gemm_fp8_skinny(xq, sx, lw["wo_q"], lw["wo_s"], t1, B, H,
+                                   accbuf=self.b_gemm_acc)
                 ho.stage_quant_mx(t1, H, xq, sx, B, H,
                                   stage=ho.STAGE_NORM2, x2=h, x2stride=H,
                                   g=lw["g_post"], g2=lw["g_preffn"],
                                   hout=hb, hstride=H, eps=eps)
                 h, hb = hb, h
                 ho.gemm_fp8_skinny(xq, sx, lw["wgu_q"], lw["wgu_s"], gu, B,
-                                   2 * I)
+                                   2 * I, accbuf=self.b_gemm_acc)
                 ho.stage_quant_mx(gu, 2 * I, xq, sx, B, I,
                                   stage=ho.STAGE_GLU, x2=gu[:, I:],
                                   x2stride=2 * I, act=self.act)
                 ho.gemm_fp8_skinny(xq, sx, lw["wdown_q"], lw["wdown_s"],
-                                   t2, B, H)
+                                   t2, B, H, accbuf=self.b_gemm_acc)
             else:
                 ho.gemm_fp8_skinny(xq, sx, lw["wo_q"], lw["wo_s"], h, B, H,
-                                   res=h, rstride=H)
+                                   res=h, rstride=H,
+                                   accbuf=self.b_gemm_acc)
                 ho.stage_quant_mx(h, H, xq, sx, B, H, stage=ho.STAGE_NORM,
                                   g=lw["g_post"], eps=eps)
                 ho.gemm_fp8_skinny(xq, sx, lw["wgu_q"], lw["wgu_s"], gu, B,
-                                   2 * I)
+                                   2 * I, accbuf=self.b_gemm_acc)
                 ho.stage_quant_mx(gu, 2 * I, xq, sx, B, I,
                                   stage=ho.STAGE_GLU, x2=gu[:, I:],
                                   x2stride=2 * I, act=self.act)
                 ho.gemm_fp8_skinny(xq, sx, lw["wdown_q"], lw["wdown_s"], h,
-                                   B, H, res=h, rstride=H)
+                                   B, H, res=h, rstride=H,
+                                   accbuf=self.b_gemm_acc)
         if self.gemma:
             ho.stage_quant_mx(t2, H, xq, sx, B, H, stage=ho.STAGE_NORM2,
                               x2=h, x2stride=H,
@@ -1021,7 +1025,8 @@ class GPUModel:
                               g=self.g_final, eps=eps)
         ho.gemm_fp8_skinny(xq, sx, self.lm_head_q, self.lm_head_s,
                            self.bt_logits, B, self.vocab_l,
-                           softcap=self.final_softcap)
+                           softcap=self.final_softcap,
+                           accbuf=self.b_gemm_acc)
         ho.sample(self.bt_logits[:B], min_p, greedy, self.seed,
                   self.rng_ctr, self.bt_gmax, self.bt_pick, self.bt_next,
                   self.bt_ring, self.bt_nout, self.bt_lens, bump_len=True,

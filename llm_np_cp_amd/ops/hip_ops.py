@@ -79,7 +79,8 @@ _SIGS = {
                               ctypes.c_void_p],
     "launch_gemm_fp8_skinny": [ctypes.c_void_p] * 5 + [ctypes.c_long] +
                               [ctypes.c_void_p, ctypes.c_long] +
-                              [ctypes.c_void_p, ctypes.c_int,
+                              [ctypes.c_void_p, ctypes.c_void_p,
+                               ctypes.c_int,
                                ctypes.c_float] + [ctypes.c_int] * 3 +
                               [ctypes.c_void_p],
     "launch_gemv_fp8_mx": [ctypes.c_void_p] * 3 + [ctypes.c_long] +
@@ -402,16 +403,20 @@ def gemm_fp8_skinny(xq: torch.Tensor, sx: torch.Tensor, Wq: torch.Tensor,
                     sw: torch.Tensor, y: torch.Tensor, B: int,
                     ystride: int, res: torch.Tensor | None = None,
                     rstride: int = 0, bias: torch.Tensor | None = None,
-                    softcap: float = 0.0):
+                    softcap: float = 0.0,
+                    accbuf: torch.Tensor | None = None):
     """Y[B<=16, N] = (sx_b*sw_n) * Xq @ Wq^T: one 16-row fp8 MFMA tile
-    per wave, W nt-streamed once (batched decode B=3..16)."""
+    per wave, W nt-streamed once (batched decode B=3..16).  accbuf
+    (fp32, >= B*N) enables the K-split path for small-N occupancy."""
     N, K = Wq.shape
     out_f32 = 1 if y.dtype == torch.float32 else 0
+    if accbuf is not None and accbuf.numel() < B * N:
+        accbuf = None
     _check(lib().launch_gemm_fp8_skinny(
         _ptr(xq), _ptr(sx), _ptr(Wq), _ptr(sw), _ptr(y),
         ctypes.c_long(ystride), _ptr(res), ctypes.c_long(rstride),
-        _ptr(bias), out_f32, ctypes.c_float(softcap), B, N, K,
-        _stream()), "gemm_fp8_skinny")
+        _ptr(bias), _ptr(accbuf), out_f32, ctypes.c_float(softcap),
+        B, N, K, _stream()), "gemm_fp8_skinny")
 
 
 def prefetch(t: torch.Tensor, sink: torch.Tensor):
