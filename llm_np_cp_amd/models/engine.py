@@ -840,7 +840,9 @@ class GPUModel:
         # v_pk_fma per 16 weight bytes at B=8 -> ~2 ms floor); B=3..16
         # takes the SKINNY fp8 MFMA GEMM (matrix cores own the MACs,
         # W nt-streamed once).  See profiles/decode_kernels_r02.md.
-        if self.fp8 and B <= 2 and not cfg.attention_bias:
+        import os as _os
+        mx_max = int(_os.environ.get("LLM_BATCH_MX_MAX", "2"))
+        if self.fp8 and B <= mx_max and not cfg.attention_bias:
             return self._decode_batch_step_mx(B, greedy, min_p, temperature)
         if self.fp8 and B <= 16:
             return self._decode_batch_step_skinny(B, greedy, min_p,
