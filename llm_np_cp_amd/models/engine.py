@@ -835,13 +835,13 @@ class GPUModel:
         layer stack runs as M=B-row MFMA GEMMs.  Graph-replayable
         (all state device-side)."""
         cfg = self.config
-        # measured: the multi-x GEMV wins at B<=2 (1.44 ms vs ~2.4 GEMM
-        # at B=2) but is VALU-ISSUE-bound beyond that (64 irreducible
-        # v_pk_fma per 16 weight bytes at B=8 -> ~2 ms floor); B=3..16
-        # takes the SKINNY fp8 MFMA GEMM (matrix cores own the MACs,
-        # W nt-streamed once).  See profiles/decode_kernels_r02.md.
+        # measured (profiles/decode_kernels_r02.md): the SKINNY fp8 MFMA
+        # GEMM wins at every B (even B=2: 1.34 vs the multi-x GEMV's
+        # 1.44 ms — the GEMV is VALU-ISSUE-bound, 64 irreducible
+        # v_pk_fma per 16 weight bytes at B=8).  The multi-x path stays
+        # available via LLM_BATCH_MX_MAX for comparison.
         import os as _os
-        mx_max = int(_os.environ.get("LLM_BATCH_MX_MAX", "2"))
+        mx_max = int(_os.environ.get("LLM_BATCH_MX_MAX", "0"))
         if self.fp8 and B <= mx_max and not cfg.attention_bias:
             return self._decode_batch_step_mx(B, greedy, min_p, temperature)
         if self.fp8 and B <= 16:
