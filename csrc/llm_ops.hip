@@ -2312,25 +2312,32 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 #define BN 128
 #define BK 64
 
-// BM=BN=128, BK=64, 4 waves (2x2), each wave a 64x64 sub-tile of
-// 4x4 16x16 fragments.  Double-buffered LDS filled by
-// global_load_lds (async global->LDS DMA, 16 B/lane), one barrier per
-// K-tile (guide §5 "Minimum 2-phase": STAGE next ahead of ds_read+MFMA).
-// LDS rows are XOR-swizzled st_8x16 (slot ^= row&7) to keep
-// ds_read_b128 fragment reads conflict-free; glds writes lane-linear,
-// so the swizzle is applied to the per-lane GLOBAL source address
-// (guide §5.4 rule 21) and to the read offsets — never the LDS dest.
-extern "C" __global__ void __launch_bounds__(256)
-k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
-            u16* __restrict__ Y, const u16* __restrict__ res,
-            float* __restrict__ accbuf, int M, int N, int K) {
-  __shared__ u16 As[2][128 * 64];
-  __shared__ u16 Bs[2][128 * 64];
+// TS=BM=BN tile template (128 or 64), BK=64, 4 waves (2x2), each wave
+// a (TS/2)^2 sub-tile of 16x16 fragments.  Double-buffered LDS filled
+// by global_load_lds (async global->LDS DMA, 16 B/lane), one barrier
+// per K-tile (guide §5 "Minimum 2-phase": STAGE next ahead of
+// ds_read+MFMA).  LDS rows are XOR-swizzled st_8x16 (slot ^= row&7) so
+// ds_read_b128 fragment reads stay conflict-free; glds writes
+// lane-linear, so the swizzle is applied to the per-lane GLOBAL source
+// address (guide §5.4 rule 21) and to the read offsets — never the LDS
+// dest.  The 64-tile variant exists because M=N=2048-class prefill
+// GEMMs give the 128-tile only ~256 blocks = 1 block/CU — no occupancy
+// to hide the DMA/barrier latency (measured 250 TF; see
+// profiles/prefill_gemm_r02.md).
+template <int TS>
+__global__ void __launch_bounds__(256)
+k_gemm_bf16_t(const u16* __restrict__ X, const u16* __restrict__ W,
+              u16* __restrict__ Y, const u16* __restrict__ res,
+              float* __restrict__ accbuf, int M, int N, int K) {
+  constexpr int F = TS / 32;        // 16x16 frags per wave dim
+  constexpr int SUB = TS / 2;       // wave sub-tile span
+  __shared__ u16 As[2][TS * 64];
+  __shared__ u16 Bs[2][TS * 64];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int wrow = wave >> 1, wcol = wave & 1;  // 2x2 waves
-  const int bm = blockIdx.x * BM, bn = blockIdx.y * BN;
+  const int bm = blockIdx.x * TS, bn = blockIdx.y * TS;
 
   const int SK = gridDim.z;
   const int kslices = (K / BK + SK - 1) / SK;
@@ -2340,11 +2347,11 @@ k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
   const int nt = (k_hi - k_lo) / BK;
   if (nt <= 0) return;
 
-  f4v acc[4][4];
+  f4v acc[F][F];
 #pragma unroll
-  for (int i = 0; i < 4; i++)
+  for (int i = 0; i < F; i++)
 #pragma unroll
-    for (int j = 0; j < 4; j++) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < F; j++) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   // glds staging: per call a wave fills 8 rows x 128 B (lane: row
   // lane/8, slot lane%8); source col-slot pre-swizzled by row&7.
@@ -2353,8 +2360,8 @@ k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
   auto stage = [&](int buf, int kt) {
     const int k0 = k_lo + kt * BK;
 #pragma unroll
-    for (int i = 0; i < 4; i++) {
-      int r = wave * 32 + i * 8 + g_r;   // tile row 0..127
+    for (int i = 0; i < TS / 32; i++) {
+      int r = wave * (TS / 4) + i * 8 + g_r;   // tile row 0..TS-1
       int cs = g_s ^ (r & 7);            // source col-slot (involution)
       int gr = bm + r;
       int grc = gr < M ? gr : (M > 0 ? M - 1 : 0);
@@ -2362,7 +2369,7 @@ k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
           (const __attribute__((address_space(1))) uint32_t*)(
               X + (size_t)grc * K + k0 + cs * 8),
           (__attribute__((address_space(3))) uint32_t*)(
-              &As[buf][(wave * 32 + i * 8) * 64]),
+              &As[buf][(wave * (TS / 4) + i * 8) * 64]),
           16, 0, 0);
       int gb = bn + r;
       int gbc = gb < N ? gb : N - 1;
@@ -2370,7 +2377,7 @@ k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
           (const __attribute__((address_space(1))) uint32_t*)(
               W + (size_t)gbc * K + k0 + cs * 8),
           (__attribute__((address_space(3))) uint32_t*)(
-              &Bs[buf][(wave * 32 + i * 8) * 64]),
+              &Bs[buf][(wave * (TS / 4) + i * 8) * 64]),
           16, 0, 0);
     }
   };
@@ -2384,20 +2391,20 @@ k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
     if (t + 1 < nt) stage(cur ^ 1, t + 1);   // issue next tile's DMA
 #pragma unroll
     for (int sl = 0; sl < 2; sl++) {         // two 32-deep k-slices
-      bf16x8 a[4], b[4];
+      bf16x8 a[F], b[F];
 #pragma unroll
-      for (int i = 0; i < 4; i++) {
-        int ra = wrow * 64 + i * 16 + fr;
+      for (int i = 0; i < F; i++) {
+        int ra = wrow * SUB + i * 16 + fr;
         int sa = (sl * 4 + fk) ^ (ra & 7);
         a[i] = *(bf16x8*)(&As[cur][ra * 64 + sa * 8]);
-        int rb = wcol * 64 + i * 16 + fr;
+        int rb = wcol * SUB + i * 16 + fr;
         int sb = (sl * 4 + fk) ^ (rb & 7);
         b[i] = *(bf16x8*)(&Bs[cur][rb * 64 + sb * 8]);
       }
 #pragma unroll
-      for (int i = 0; i < 4; i++)
+      for (int i = 0; i < F; i++)
 #pragma unroll
-        for (int j = 0; j < 4; j++)
+        for (int j = 0; j < F; j++)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a[i], b[j], acc[i][j], 0, 0, 0);
     }
@@ -2408,13 +2415,13 @@ k_gemm_bf16(const u16* __restrict__ X, const u16* __restrict__ W,
   // epilogue: C layout col = lane&15, row = (lane>>4)*4 + reg
   const int cc = lane & 15, cr = (lane >> 4) * 4;
 #pragma unroll
-  for (int i = 0; i < 4; i++) {
+  for (int i = 0; i < F; i++) {
 #pragma unroll
-    for (int j = 0; j < 4; j++) {
+    for (int j = 0; j < F; j++) {
 #pragma unroll
       for (int r = 0; r < 4; r++) {
-        int row = bm + wrow * 64 + i * 16 + cr + r;
-        int col = bn + wcol * 64 + j * 16 + cc;
+        int row = bm + wrow * SUB + i * 16 + cr + r;
+        int col = bn + wcol * SUB + j * 16 + cc;
         if (row < M && col < N) {
           float v = acc[i][j][r];
           if (SK > 1) {
@@ -2451,9 +2458,15 @@ extern "C" hipError_t launch_gemm_bf16(const void* X, const void* W, void* Y,
                                        const void* res, void* accbuf, int M,
                                        int N, int K, hipStream_t stream) {
   int gm = (M + BM - 1) / BM, gn = (N + BN - 1) / BN;
+  int ts = 128;
+  if (gm * gn < 400 && K % 64 == 0) {
+    // small grids underfill 256 CUs at 1 block/CU with the 128-tile
+    ts = 64;
+    gm = (M + 63) / 64;
+    gn = (N + 63) / 64;
+  }
   int sk = 1;
   if (accbuf && gm * gn < 160) {  // split K while the grid underfills
-    // target up to 2 blocks/CU (64 KB LDS each -> 2 resident max)
     while (sk < 8 && gm * gn * sk * 2 <= 512 && (K / BK) % (sk * 2) == 0)
       sk *= 2;
   }
@@ -2463,9 +2476,14 @@ extern "C" hipError_t launch_gemm_bf16(const void* X, const void* W, void* Y,
                        dim3(256), 0, stream, (float*)accbuf, total);
   }
   dim3 grid(gm, gn, sk);
-  hipLaunchKernelGGL(k_gemm_bf16, grid, dim3(256), 0, stream, (const u16*)X,
-                     (const u16*)W, (u16*)Y, (const u16*)res, (float*)accbuf,
-                     M, N, K);
+  if (ts == 64)
+    hipLaunchKernelGGL((k_gemm_bf16_t<64>), grid, dim3(256), 0, stream,
+                       (const u16*)X, (const u16*)W, (u16*)Y, (const u16*)res,
+                       (float*)accbuf, M, N, K);
+  else
+    hipLaunchKernelGGL((k_gemm_bf16_t<128>), grid, dim3(256), 0, stream,
+                       (const u16*)X, (const u16*)W, (u16*)Y, (const u16*)res,
+                       (float*)accbuf, M, N, K);
   if (sk > 1) {
     long total = (long)M * N;
     hipLaunchKernelGGL(k_gemm_fin, dim3((uint32_t)((total + 255) / 256)),
