@@ -1313,6 +1313,11 @@ extern "C" hipError_t launch_rope_cache(void* q, const void* kin,
 
 // ====================================================================
 // GQA attention with online softmax (decode + per-query prefill).
+// NOTE: with every shipped preset at head_dim in {64,128,256} the MFMA
+// prefill kernel below is taken instead; this per-query VALU kernel is
+// kept as the reachable fallback for odd head_dims and as the simple
+// DEBUGGING reference the fused kernels are compared against in tests
+// (SURVEY §2.2 K1).
 // grid (nh, M): block = head h, query row m (absolute pos = *len_ptr + m,
 // attends keys [start, pos+1)).  4 waves split the KV range; each lane
 // owns (sub-position, 8-dim chunk); merge within wave then across waves.
@@ -2285,6 +2290,12 @@ extern "C" hipError_t launch_sample(const void* logits, int V, int lbf16,
                                     hipStream_t stream) {
   int blocks = (V + 255) / 256;
   if (blocks > 512) blocks = 512;
+  // batched rows: keep TOTAL thread count constant (the B=8 sampler
+  // measured 102 us with a full grid per row)
+  if (batch > 1) {
+    blocks = blocks / batch + 1;
+    if (blocks < 16) blocks = 16;
+  }
   if (!greedy)
     hipLaunchKernelGGL(k_logit_max, dim3(blocks, batch), dim3(256), 0,
                        stream, logits, V, lbf16, (unsigned long long*)gmax);
