@@ -1118,6 +1118,114 @@ class GPUModel:
         out = self.bt_ring[:B, :n].cpu().numpy()
         return out[:, max(0, n - n_tokens):]
 
+    # ---- continuous batching primitives (rows join/leave the group) ----
+    def prefill_row(self, b: int, ids, greedy: bool = True,
+                    min_p: float = 0.1, temperature: float = 1.0):
+        """Prefill ONE sequence into slot b (other rows untouched) and
+        sample its first token — a request JOINING the lockstep group
+        between decode chunks (server continuous batching)."""
+        ids = np.asarray(ids, dtype=np.int32).ravel()
+        P = len(ids)
+        assert 0 <= b < self.max_batch and self.max_batch > 1
+        assert self.world == 1
+        if P + 1 >= self.max_seq:
+            raise ValueError(f"prompt {P} fills the {self.max_seq} pool")
+        self._pb = b
+        done = 0
+        while done < P:
+            M = min(P - done, self.PC)
+            self.ids_buf[:M].copy_(torch.from_numpy(ids[done:done + M]))
+            ho.i32_set(self.len_buf, done)
+            ho.embed(self.embed, self.ids_buf, self.b_h, M,
+                     self.config.embed_scale)
+            self._layers_forward(M)
+            done += M
+        self._lm_head_last(M)
+        self._pb = 0
+        self.bt_lens[b:b + 1].fill_(P)
+        self.bt_nout[b:b + 1].zero_()
+        # first token for this row only (row-sliced sampler state)
+        ho.sample(self.b_logits, min_p, greedy, self.seed, self.rng_ctr,
+                  self.bt_gmax[b:b + 1], self.bt_pick[b:b + 1],
+                  self.bt_next[b:b + 1], self.bt_ring[b:b + 1],
+                  self.bt_nout[b:b + 1], self.bt_lens[b:b + 1],
+                  bump_len=False, temperature=temperature,
+                  cnt=self.bt_cnt[b:b + 1], batch=1)
+        if not hasattr(self, "_host_lens") or self._host_lens is None:
+            self._host_lens = [0] * self.max_batch
+        while len(self._host_lens) < self.max_batch:
+            self._host_lens.append(0)
+        self._host_lens[b] = P
+        self._batch_n = max(getattr(self, "_batch_n", 0), b + 1)
+
+    def compact_row(self, dst: int, src: int):
+        """Move sequence slot src -> dst (KV pools, scales, device row
+        state) so retired rows leave the group dense."""
+        if dst == src:
+            return
+        for i in range(self.config.num_hidden_layers):
+            self.k_cache[i][dst].copy_(self.k_cache[i][src])
+            self.v_cache[i][dst].copy_(self.v_cache[i][src])
+            if self.kv8:
+                self.k_scale[i][dst].copy_(self.k_scale[i][src])
+                self.v_scale[i][dst].copy_(self.v_scale[i][src])
+        for t in (self.bt_lens, self.bt_next, self.bt_nout, self.bt_gmax,
+                  self.bt_pick, self.bt_cnt):
+            t[dst:dst + 1].copy_(t[src:src + 1])
+        self.bt_ring[dst].copy_(self.bt_ring[src])
+        self._host_lens[dst] = self._host_lens[src]
+
+    def decode_rows(self, B: int, n: int, greedy: bool = True,
+                    min_p: float = 0.1, temperature: float = 1.0,
+                    use_graph: bool = True):
+        """n lockstep steps for rows [0, B) (each already holding a
+        sampled next token from prefill_row / a previous chunk).
+        Returns a list of B arrays: each row's n new token ids."""
+        assert 1 <= B <= getattr(self, "_batch_n", 0) or B <= self.max_batch
+        if max(self._host_lens[:B]) + n > self.max_seq:
+            raise ValueError("decode_rows would overflow the KV pool")
+        for b in range(B):
+            self._host_lens[b] += n
+        key = ("batchc", B, greedy, min_p, temperature)
+        if not hasattr(self, "_bgraphs"):
+            self._bgraphs = {}
+        g = self._bgraphs.get(key)
+        executed = 0
+        if (use_graph and g is None and n > 1
+                and not getattr(self, "_graph_failed", False)):
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                self._decode_batch_step(B, greedy, min_p, temperature)
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            executed += 1  # the warm-up was a real step
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._decode_batch_step(B, greedy, min_p, temperature)
+                self._bgraphs[key] = g
+            except Exception as e:
+                self._graph_failed = True
+                self._graph_error = e
+                g = None
+        remaining = n - executed
+        if g is not None and use_graph:
+            for _ in range(remaining):
+                g.replay()
+        else:
+            for _ in range(remaining):
+                self._decode_batch_step(B, greedy, min_p, temperature)
+        torch.cuda.synchronize()
+        comm = tpu.xgmi_comm()
+        if comm is not None:
+            comm.check()
+        nouts = self.bt_nout[:B].cpu().numpy()
+        ring = self.bt_ring[:B].cpu().numpy()
+        # each step appends exactly one id per row
+        return [ring[b, max(0, int(nouts[b]) - n):int(nouts[b])]
+                for b in range(B)]
+
     def generate_tokens_batch(self, prompts, max_tokens: int,
                               greedy: bool = True, min_p: float = 0.1,
                               temperature: float = 1.0) -> np.ndarray:

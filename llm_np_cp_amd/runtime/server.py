@@ -1,16 +1,17 @@
-"""HTTP serving layer with dynamic request batching.
+"""HTTP serving layer with CONTINUOUS request batching.
 
 The reference has no server (no CLI, no server); this is the
 deployment-facing wrapper around the engine: an OpenAI-style
 ``/v1/completions`` endpoint plus ``/health`` and ``/stats``.
 
 Concurrent requests are NOT serialized: a scheduler thread owns the
-engine and groups compatible in-flight requests (same sampling mode)
-into one RAGGED lockstep batch (``GPUModel.prefill_batch`` /
-``decode_batch`` — per-sequence KV pools, per-row device positions),
-decoding in chunks and retiring rows on EOS / per-request max_tokens.
-Requests with CPU-only sampling strategies (top_k/top_p) or a non-batch
-engine fall back to single-sequence generate().
+engine and runs a ragged lockstep group (per-sequence KV pools, per-row
+device positions).  Between 16-token chunks, finished rows RETIRE (EOS
+or per-request max_tokens; the last row compacts into the freed slot)
+and queued compatible requests JOIN mid-flight (``prefill_row`` into a
+free slot) — continuous batching, not fixed groups.  Requests with
+CPU-only sampling strategies (top_k/top_p) or a non-batch engine fall
+back to single-sequence generate().
 
     python -m llm_np_cp_amd.runtime.server --model llama-3.2-1b \
         --port 8080 --max-batch 8
@@ -47,20 +48,22 @@ class _Pending:
     done: threading.Event = field(default_factory=threading.Event)
     result: Optional[dict] = None
     error: Optional[Exception] = None
+    t0: float = field(default_factory=time.time)
 
 
 class BatchScheduler:
-    """Groups compatible concurrent requests into ragged lockstep
-    batches on a single engine-owning thread."""
+    """Owns the engine thread; continuous ragged batching for
+    compatible concurrent requests."""
 
     def __init__(self, generate_one, run_group, max_batch: int,
                  window_s: float = 0.004):
         self.generate_one = generate_one
-        self.run_group = run_group
+        self.run_group = run_group  # (pendings, poll_fn) -> resolves them
         self.max_batch = max_batch
         self.window_s = window_s
         self.q: "queue.Queue[_Pending]" = queue.Queue()
-        self.stats = {"requests": 0, "batches": 0, "max_group": 0}
+        self.stats = {"requests": 0, "batches": 0, "max_group": 0,
+                      "joined_mid_flight": 0}
         t = threading.Thread(target=self._loop, daemon=True)
         t.start()
 
@@ -80,44 +83,58 @@ class BatchScheduler:
     def _batchable(self, req) -> bool:
         return self.max_batch > 1 and req.strategy in ("greedy", "min_p")
 
+    def _poll_compatible(self, key, deferred):
+        """Non-blocking: next queued pending with this sampling key;
+        others are deferred (re-queued after the group)."""
+        while True:
+            try:
+                p = self.q.get_nowait()
+            except queue.Empty:
+                return None
+            if self._batchable(p.req) and self._key(p.req) == key:
+                return p
+            deferred.append(p)
+
     def _loop(self):
         while True:
             first = self.q.get()
-            group = [first]
-            if self._batchable(first.req):
-                key = self._key(first.req)
-                deadline = time.monotonic() + self.window_s
-                while len(group) < self.max_batch:
-                    tmo = deadline - time.monotonic()
-                    if tmo <= 0:
-                        break
-                    try:
-                        nxt = self.q.get(timeout=tmo)
-                    except queue.Empty:
-                        break
-                    if self._batchable(nxt.req) and \
-                            self._key(nxt.req) == key:
-                        group.append(nxt)
-                    else:
-                        self.q.put(nxt)
-                        break
-            self.stats["requests"] += len(group)
             self.stats["batches"] += 1
-            self.stats["max_group"] = max(self.stats["max_group"],
-                                          len(group))
-            try:
-                if len(group) == 1 and not self._batchable(first.req):
-                    group[0].result = self.generate_one(group[0].req)
+            if not self._batchable(first.req):
+                self.stats["requests"] += 1
+                self.stats["max_group"] = max(self.stats["max_group"], 1)
+                try:
+                    first.result = self.generate_one(first.req)
+                except Exception as e:
+                    first.error = e
+                first.done.set()
+                continue
+            key = self._key(first.req)
+            group = [first]
+            deadline = time.monotonic() + self.window_s
+            deferred = []
+            while len(group) < self.max_batch:
+                tmo = deadline - time.monotonic()
+                if tmo <= 0:
+                    break
+                try:
+                    nxt = self.q.get(timeout=tmo)
+                except queue.Empty:
+                    break
+                if self._batchable(nxt.req) and self._key(nxt.req) == key:
+                    group.append(nxt)
                 else:
-                    results = self.run_group([p.req for p in group])
-                    for p, r in zip(group, results):
-                        p.result = r
-            except Exception as e:  # surface to every waiter
+                    deferred.append(nxt)
+            try:
+                self.run_group(
+                    group, lambda: self._poll_compatible(key, deferred),
+                    self.stats)
+            except Exception as e:
                 for p in group:
-                    p.error = e
-            finally:
-                for p in group:
-                    p.done.set()
+                    if not p.done.is_set():
+                        p.error = e
+                        p.done.set()
+            for p in deferred:  # re-enter scheduling
+                self.q.put(p)
 
 
 def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
@@ -133,11 +150,9 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
                                    kv_dtype=kv_dtype)
     can_batch = False
     if max_batch > 1 and type(model).__name__ == "GPUModel":
-        # rebuild with batch pools (load_model has no max_batch knob to
-        # keep its reference-parity signature lean)
-        from ..core.config import PRESETS
-        from ..io.loader import random_weights, load_weights_numpy, \
-            load_config
+        # rebuild with batch pools (load_model keeps its reference-parity
+        # signature lean)
+        from ..io.loader import random_weights, load_weights_numpy
         from ..models.engine import GPUModel
         import os
         if os.path.isdir(model_name):
@@ -151,60 +166,6 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
     eos = getattr(cfg, "eos_token_id", None)
     eos_set = (set() if eos is None else {int(eos)}
                if isinstance(eos, (int, float)) else {int(e) for e in eos})
-
-    def generate_one(req) -> dict:
-        params = L.SamplingParams(strategy=req.strategy, min_p=req.min_p,
-                                  temperature=req.temperature,
-                                  seed=req.seed)
-        t0 = time.time()
-        out = L.generate(req.prompt, tok, model,
-                         max_tokens=req.max_tokens, params=params,
-                         stream=False, stop_on_eos=req.stop_on_eos)
-        return _payload(req, out.token_ids, out.text, time.time() - t0,
-                        out.prefill_time_s, out.decode_tokens_per_s)
-
-    def run_group(reqs) -> list:
-        if not can_batch or len(reqs) == 1:
-            return [generate_one(r) for r in reqs]
-        t0 = time.time()
-        prompts = [tok.encode(r.prompt) for r in reqs]
-        B = len(reqs)
-        greedy = reqs[0].strategy == "greedy"
-        min_p = reqs[0].min_p
-        temp = reqs[0].temperature
-        stop_eos = reqs[0].stop_on_eos
-        maxn = max(r.max_tokens for r in reqs)
-        room = model.max_seq - max(len(p) for p in prompts) - 1
-        maxn = min(maxn, room)
-        model.prefill_batch(prompts)
-        tp = time.time() - t0
-        rows = [[] for _ in range(B)]
-        done = [False] * B
-        produced = 0
-        first = True
-        while produced < maxn and not all(done):
-            n = min(32, maxn - produced)
-            ids = model.decode_batch(n, greedy=greedy, min_p=min_p,
-                                     temperature=temp,
-                                     first_from_logits=first)
-            first = False
-            produced += n
-            for b in range(B):
-                if done[b]:
-                    continue
-                for t in ids[b].tolist():
-                    rows[b].append(int(t))
-                    if len(rows[b]) >= reqs[b].max_tokens:
-                        done[b] = True
-                        break
-                    if stop_eos and int(t) in eos_set:
-                        done[b] = True
-                        break
-        dt = time.time() - t0
-        n_out = sum(len(r) for r in rows)
-        return [_payload(reqs[b], rows[b], tok.decode(rows[b]), dt, tp,
-                         n_out / max(dt - tp, 1e-9))
-                for b in range(B)]
 
     def _payload(req, ids, text, total_s, prefill_s, tps) -> dict:
         return {
@@ -222,6 +183,118 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
                 "decode_tokens_per_s": tps,
             },
         }
+
+    def generate_one(req) -> dict:
+        params = L.SamplingParams(strategy=req.strategy, min_p=req.min_p,
+                                  temperature=req.temperature,
+                                  seed=req.seed)
+        t0 = time.time()
+        out = L.generate(req.prompt, tok, model,
+                         max_tokens=req.max_tokens, params=params,
+                         stream=False, stop_on_eos=req.stop_on_eos)
+        return _payload(req, out.token_ids, out.text, time.time() - t0,
+                        out.prefill_time_s, out.decode_tokens_per_s)
+
+    class _Row:
+        __slots__ = ("pending", "ids", "first_consumed")
+
+        def __init__(self, pending):
+            self.pending = pending
+            self.ids = []
+            self.first_consumed = False
+
+    def run_group(pendings, poll, stats) -> None:
+        """Continuous ragged batching: chunks of 16 steps; retired rows
+        compact; compatible queued requests join between chunks."""
+        if not can_batch:
+            for p in pendings:
+                stats["requests"] += 1
+                try:
+                    p.result = generate_one(p.req)
+                except Exception as e:
+                    p.error = e
+                p.done.set()
+            return
+        r0 = pendings[0].req
+        greedy = r0.strategy == "greedy"
+        min_p, temp, stop_eos = r0.min_p, r0.temperature, r0.stop_on_eos
+        rows: list = []  # slot -> _Row
+
+        def admit(p):
+            slot = len(rows)
+            t_p = time.time()
+            model.prefill_row(slot, tok.encode(p.req.prompt),
+                              greedy=greedy, min_p=min_p, temperature=temp)
+            p.prefill_s = time.time() - t_p
+            rows.append(_Row(p))
+            stats["requests"] += 1
+
+        def finish(row):
+            p = row.pending
+            dt = time.time() - p.t0
+            ids = row.ids[:p.req.max_tokens]
+            p.result = _payload(p.req, ids, tok.decode(ids), dt,
+                                getattr(p, "prefill_s", 0.0),
+                                len(ids) / max(dt, 1e-9))
+            p.done.set()
+
+        for p in pendings:
+            admit(p)
+        stats["max_group"] = max(stats["max_group"], len(rows))
+        room = model.max_seq - 2
+        while rows:
+            # mid-flight admission into free slots
+            while len(rows) < max_batch:
+                p2 = poll()
+                if p2 is None:
+                    break
+                admit(p2)
+                stats["joined_mid_flight"] += 1
+                stats["max_group"] = max(stats["max_group"], len(rows))
+            B = len(rows)
+            chunk = min(16, room - max(model._host_lens[:B]))
+            if chunk <= 0:  # a row ran out of KV room: retire longest
+                longest = max(range(B), key=lambda b: model._host_lens[b])
+                finish(rows[longest])
+                model.compact_row(longest, B - 1)
+                rows[longest] = rows[B - 1]
+                rows.pop()
+                continue
+            out = model.decode_rows(B, chunk, greedy=greedy, min_p=min_p,
+                                    temperature=temp)
+            # the first sampled token (from prefill_row) precedes chunk 1
+            for b in range(B):
+                row = rows[b]
+                if not row.first_consumed:
+                    # ring holds first+chunk ids; decode_rows returned
+                    # the last `chunk` — prepend the first token once
+                    n0 = int(model.bt_nout[b].item())
+                    first_id = int(model.bt_ring[b, 0].item()) \
+                        if n0 >= 1 else None
+                    if first_id is not None:
+                        row.ids.append(first_id)
+                    row.first_consumed = True
+                row.ids.extend(int(t) for t in out[b])
+            # retire rows (EOS / max_tokens), compacting from the end
+            b = 0
+            while b < len(rows):
+                row = rows[b]
+                req = row.pending.req
+                hit_eos = (stop_eos and
+                           any(int(t) in eos_set for t in row.ids))
+                if hit_eos:
+                    cut = next(i for i, t in enumerate(row.ids)
+                               if int(t) in eos_set)
+                    row.ids = row.ids[:cut + 1]
+                if hit_eos or len(row.ids) >= req.max_tokens:
+                    finish(row)
+                    last = len(rows) - 1
+                    model.compact_row(b, last)
+                    rows[b] = rows[last]
+                    rows.pop()
+                    # re-check slot b (now holds the moved row)
+                    continue
+                b += 1
 
     sched = BatchScheduler(generate_one, run_group,
                            max_batch if can_batch else 1,

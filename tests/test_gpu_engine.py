@@ -545,3 +545,54 @@ def test_batch_decode_fused_paths_match_single_fp8(preset, B):
     ids = m.generate_tokens_batch(prompts, N, greedy=True)
     for b in range(B):
         np.testing.assert_array_equal(ids[b], singles[b]), b
+
+
+def test_continuous_batching_primitives():
+    """prefill_row / decode_rows / compact_row: rows join a live group,
+    decode in lockstep from their own positions, retire and compact —
+    surviving rows must continue their exact single-sequence rollout."""
+    import llm_np_cp_amd as L
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.engine import GPUModel
+
+    cfg = L.preset_config("tiny-llama")
+    w = random_weights(cfg, seed=95)
+    rng = np.random.default_rng(96)
+    p0 = rng.integers(0, cfg.vocab_size, size=6)
+    p1 = rng.integers(0, cfg.vocab_size, size=9)
+    p2 = rng.integers(0, cfg.vocab_size, size=4)
+
+    # single-sequence references
+    single = GPUModel(cfg, dict(w), max_seq=64, dtype="fp8")
+    refs = {}
+    for name, p, n in (("p0", p0, 4), ("p1", p1, 9), ("p2", p2, 6)):
+        single.prefill(p)
+        refs[name] = list(single.decode(n, greedy=True, use_graph=False))
+    del single
+    torch.cuda.empty_cache()
+
+    m = GPUModel(cfg, dict(w), max_seq=64, max_batch=4, dtype="fp8")
+    m.reset()
+    m.bt_nout.zero_()
+    m._host_lens = [0] * 4
+    m._batch_n = 0
+    ids = {0: [], 1: []}
+    m.prefill_row(0, p0, greedy=True)
+    m.prefill_row(1, p1, greedy=True)
+    ids[0].append(int(m.bt_ring[0, 0].item()))
+    ids[1].append(int(m.bt_ring[1, 0].item()))
+    out = m.decode_rows(2, 3, greedy=True)
+    ids[0] += [int(t) for t in out[0]]
+    ids[1] += [int(t) for t in out[1]]
+    assert ids[0] == refs["p0"][:4]
+    assert ids[1] == refs["p1"][:4]
+    # retire row 0; row 1 compacts into slot 0; a NEW row joins slot 1
+    m.compact_row(0, 1)
+    ids = {0: ids[1], 1: []}
+    m.prefill_row(1, p2, greedy=True)
+    ids[1].append(int(m.bt_ring[1, 0].item()))
+    out = m.decode_rows(2, 5, greedy=True)
+    ids[0] += [int(t) for t in out[0]]
+    ids[1] += [int(t) for t in out[1]]
+    assert ids[0] == refs["p1"][:9], (ids[0], refs["p1"])
+    assert ids[1] == refs["p2"][:6], (ids[1], refs["p2"])
