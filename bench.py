@@ -32,7 +32,7 @@ def main():
     ap.add_argument("--prompt-len", type=int, default=64)
     ap.add_argument("--max-seq", type=int, default=None)
     ap.add_argument("--dtype", type=str, default="bf16",
-                    choices=["bf16", "fp8"],
+                    choices=["bf16", "fp8", "fp4"],
                     help="weight dtype (constant across N for honest "
                          "scaling; fp8 = BASELINE config 4)")
     ap.add_argument("--kv-dtype", type=str, default="bf16",

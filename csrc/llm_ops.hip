@@ -1103,6 +1103,206 @@ extern "C" hipError_t launch_gemm_fp8_skinny(
 }
 
 // ====================================================================
+// MXFP4 GEMV: y[N] = W4[N,K] @ stage(x)[K] (+res); 2 fp4/byte with one
+// e8m0 scale per 32 elements.  Same packed structure as the fp8 GEMV
+// (fp32 x staging or XDIR direct reads feeding v_pk_fma_f32) — the
+// block scale rides inside v_cvt_scalef32_pk_f32_fp4, so per-ELEMENT
+// VALU matches fp8 at HALF the weight bytes.
+// ====================================================================
+
+typedef uint32_t u2v_ __attribute__((ext_vector_type(2)));
+
+DEVINL void fp4x8_to_f32p(uint32_t dw, float sc, f2v* o) {
+  o[0] = __builtin_amdgcn_cvt_scalef32_pk_f32_fp4(dw, sc, 0);
+  o[1] = __builtin_amdgcn_cvt_scalef32_pk_f32_fp4(dw, sc, 1);
+  o[2] = __builtin_amdgcn_cvt_scalef32_pk_f32_fp4(dw, sc, 2);
+  o[3] = __builtin_amdgcn_cvt_scalef32_pk_f32_fp4(dw, sc, 3);
+}
+
+template <bool NT, int RPW, bool XDIR>
+__global__ void __launch_bounds__(512)
+k_gemv_fp4_t(const uint8_t* __restrict__ W, const uint8_t* __restrict__ E,
+             const u16* __restrict__ x, const u16* __restrict__ x2,
+             const float* __restrict__ g, const float* __restrict__ g2,
+             void* __restrict__ y, const u16* __restrict__ res, int N, int K,
+             int stage, int act, float eps, int out_f32, float softcap,
+             float escale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const float* xv = XDIR ? nullptr
+                         : gemv_stage_f32(smem, x, x2, g, g2, (u16*)res, K,
+                                          stage, act, eps, escale);
+  const u16* eres = (stage == STAGE_NORM2 || stage == STAGE_NORM_EMBED)
+                        ? nullptr : res;
+
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int wpb = blockDim.x >> 6;
+  const int rstride = gridDim.x * wpb * RPW;
+  const int K2 = K / 2, K32 = K / 32;
+  for (int row0 = (blockIdx.x * wpb + wave) * RPW; row0 < N;
+       row0 += rstride) {
+  const uint8_t* Wr[RPW];
+  const uint8_t* Er[RPW];
+  f2v a0[RPW], a1[RPW];
+#pragma unroll
+  for (int r = 0; r < RPW; r++) {
+    int rr = row0 + r < N ? row0 + r : N - 1;
+    Wr[r] = W + (size_t)rr * K2;
+    Er[r] = E + (size_t)rr * K32;
+    a0[r] = (f2v){0.f, 0.f};
+    a1[r] = (f2v){0.f, 0.f};
+  }
+  int k = lane * 16;
+  for (; k + 1024 + 16 <= K; k += 2048) {
+    s8v xda, xdb, xdc, xdd;
+    if (XDIR) {
+      xda = *(const s8v*)(x + k); xdb = *(const s8v*)(x + k + 8);
+      xdc = *(const s8v*)(x + k + 1024); xdd = *(const s8v*)(x + k + 1032);
+    }
+#pragma unroll
+    for (int r = 0; r < RPW; r++) {
+      u2v_ w0 = NT ? __builtin_nontemporal_load((const u2v_*)(Wr[r] + k / 2))
+                    : *(const u2v_*)(Wr[r] + k / 2);
+      u2v_ w1 = NT ? __builtin_nontemporal_load(
+                          (const u2v_*)(Wr[r] + (k + 1024) / 2))
+                    : *(const u2v_*)(Wr[r] + (k + 1024) / 2);
+      union { float f; uint32_t u; } s0, s1;
+      s0.u = (uint32_t)Er[r][k / 32] << 23;
+      s1.u = (uint32_t)Er[r][(k + 1024) / 32] << 23;
+      f2v c0[8], c1[8];
+      fp4x8_to_f32p(w0[0], s0.f, c0);
+      fp4x8_to_f32p(w0[1], s0.f, c0 + 4);
+      fp4x8_to_f32p(w1[0], s1.f, c1);
+      fp4x8_to_f32p(w1[1], s1.f, c1 + 4);
+#pragma unroll
+      for (int q = 0; q < 4; q++) {
+        f2v ca = c0[q * 2];
+        f2v cb = c0[q * 2 + 1];
+        f2v xa, xb, xc, xd;
+        if (XDIR) {
+          const u16* p0 = (q < 2) ? (const u16*)&xda : (const u16*)&xdb;
+          const int o0 = (q & 1) * 4;
+          xa = (f2v){b2f(p0[o0]), b2f(p0[o0 + 1])};
+          xb = (f2v){b2f(p0[o0 + 2]), b2f(p0[o0 + 3])};
+          const u16* p1 = (q < 2) ? (const u16*)&xdc : (const u16*)&xdd;
+          xc = (f2v){b2f(p1[o0]), b2f(p1[o0 + 1])};
+          xd = (f2v){b2f(p1[o0 + 2]), b2f(p1[o0 + 3])};
+        } else {
+          xa = *(const f2v*)(xv + k + q * 4);
+          xb = *(const f2v*)(xv + k + q * 4 + 2);
+          xc = *(const f2v*)(xv + k + 1024 + q * 4);
+          xd = *(const f2v*)(xv + k + 1024 + q * 4 + 2);
+        }
+        a0[r] += ca * xa;
+        a1[r] += cb * xb;
+        a0[r] += c1[q * 2] * xc;
+        a1[r] += c1[q * 2 + 1] * xd;
+      }
+    }
+  }
+  for (; k < K; k += 1024) {
+    s8v xda, xdb;
+    if (XDIR) {
+      xda = *(const s8v*)(x + k); xdb = *(const s8v*)(x + k + 8);
+    }
+#pragma unroll
+    for (int r = 0; r < RPW; r++) {
+      u2v_ w0 = NT ? __builtin_nontemporal_load((const u2v_*)(Wr[r] + k / 2))
+                    : *(const u2v_*)(Wr[r] + k / 2);
+      union { float f; uint32_t u; } s0;
+      s0.u = (uint32_t)Er[r][k / 32] << 23;
+      f2v c0[8];
+      fp4x8_to_f32p(w0[0], s0.f, c0);
+      fp4x8_to_f32p(w0[1], s0.f, c0 + 4);
+#pragma unroll
+      for (int q = 0; q < 4; q++) {
+        f2v ca = c0[q * 2];
+        f2v cb = c0[q * 2 + 1];
+        f2v xa, xb;
+        if (XDIR) {
+          const u16* p0 = (q < 2) ? (const u16*)&xda : (const u16*)&xdb;
+          const int o0 = (q & 1) * 4;
+          xa = (f2v){b2f(p0[o0]), b2f(p0[o0 + 1])};
+          xb = (f2v){b2f(p0[o0 + 2]), b2f(p0[o0 + 3])};
+        } else {
+          xa = *(const f2v*)(xv + k + q * 4);
+          xb = *(const f2v*)(xv + k + q * 4 + 2);
+        }
+        a0[r] += ca * xa;
+        a1[r] += cb * xb;
+      }
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < RPW; r++) {
+    f2v s = a0[r] + a1[r];
+    float acc = wave_reduce_sum(s[0] + s[1]);
+    if (lane == 0 && row0 + r < N)
+      gemv_epilogue(acc, row0 + r, y, eres, out_f32, softcap);
+  }
+  }  // row0 grid-stride loop
+}
+
+extern "C" hipError_t launch_gemv_fp4(const void* W, const void* E,
+                                      const void* x, const void* x2,
+                                      const void* g, const void* g2, void* y,
+                                      const void* res, int N, int K,
+                                      int stage, int act, float eps,
+                                      int out_f32, float softcap, int nt,
+                                      int rpw, int maxblocks, float escale,
+                                      hipStream_t stream) {
+  if (K % 32 != 0) return hipErrorInvalidValue;
+  static int xdir_raw4 = -1, rpw_env4 = -1;
+  if (xdir_raw4 < 0) {
+    const char* e = getenv("LLM_GEMV_XDIR");
+    xdir_raw4 = e ? atoi(e) : 1;
+    const char* p = getenv("LLM_GEMV_RPW");
+    rpw_env4 = p ? atoi(p) : 0;
+  }
+  if (rpw_env4 > 0) rpw = rpw_env4;
+  else if (rpw <= 1)
+    rpw = (K >= 3584 && N >= 3072) ? 2 : 1;  // same policy as fp8
+  const int xdir = (stage == STAGE_RAW) && xdir_raw4;
+  size_t lds = xdir ? 0 : ((size_t)K * 4 + 32);
+  if (lds > 65536) {
+    static bool raised4 = false;
+    if (!raised4) {
+#define GEMV4_RAISE(NTV, RPWV, XDV)                                         \
+      hipFuncSetAttribute((const void*)&k_gemv_fp4_t<NTV, RPWV, XDV>,       \
+                          hipFuncAttributeMaxDynamicSharedMemorySize,       \
+                          160 * 1024)
+      GEMV4_RAISE(true, 1, false); GEMV4_RAISE(true, 2, false);
+      GEMV4_RAISE(false, 1, false); GEMV4_RAISE(false, 2, false);
+#undef GEMV4_RAISE
+      raised4 = true;
+    }
+  }
+  int threads = 256;
+  int wpb = threads / 64;
+  int blocks = (N + wpb * rpw - 1) / (wpb * rpw);
+  int cap = maxblocks > 0 ? maxblocks : 1024;
+  if (blocks > cap) blocks = cap;
+#define GEMV4_CASE(NTV, RPWV, XD)                                           \
+  hipLaunchKernelGGL((k_gemv_fp4_t<NTV, RPWV, XD>), dim3(blocks),           \
+                     dim3(threads), lds, stream, (const uint8_t*)W,         \
+                     (const uint8_t*)E, (const u16*)x, (const u16*)x2,      \
+                     (const float*)g, (const float*)g2, y, (const u16*)res, \
+                     N, K, stage, act, eps, out_f32, softcap, escale)
+  if (xdir) {
+    if (nt && rpw == 2) GEMV4_CASE(true, 2, true);
+    else if (nt) GEMV4_CASE(true, 1, true);
+    else if (rpw == 2) GEMV4_CASE(false, 2, true);
+    else GEMV4_CASE(false, 1, true);
+  } else {
+    if (nt && rpw == 2) GEMV4_CASE(true, 2, false);
+    else if (nt) GEMV4_CASE(true, 1, false);
+    else if (rpw == 2) GEMV4_CASE(false, 2, false);
+    else GEMV4_CASE(false, 1, false);
+  }
+#undef GEMV4_CASE
+  return hipGetLastError();
+}
+
+// ====================================================================
 // RMSNorm: mode 0: y = norm(x)*g ; mode 1: y = res + norm(x)*g
 // x bf16 [M,H]; g fp32[H] (Gemma's gamma+1 pre-folded on host); one block
 // per row; fused single pass (x kept in registers between reduce+scale).
@@ -2504,6 +2704,70 @@ extern "C" hipError_t launch_gemm_bf16(const void* X, const void* W, void* Y,
                        dim3(256), 0, stream, (const float*)accbuf,
                        (const u16*)res, (u16*)Y, total);
   }
+  return hipGetLastError();
+}
+
+// ====================================================================
+// MXFP4 weight quantization: 4-bit e2m1 values packed 2/byte with one
+// e8m0 (power-of-two) scale per 32-element block — the OCP MX format.
+// Decode is weights-only quantization: the GEMV dequantizes via
+// v_cvt_scalef32_pk_f32_fp4, which applies the block scale INSIDE the
+// hardware convert (same VALU per element as the fp8 path, HALF the
+// weight bytes -> ~1.5x on stream-bound shapes).
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_quant_fp4_rows(const u16* __restrict__ X, uint8_t* __restrict__ Q,
+                 uint8_t* __restrict__ E, int K) {
+  const int row = blockIdx.x;
+  const u16* xr = X + (size_t)row * K;
+  uint8_t* qr = Q + (size_t)row * (K / 2);
+  uint8_t* er = E + (size_t)row * (K / 32);
+  // each thread owns whole 32-element blocks
+  for (int blk = threadIdx.x; blk < K / 32; blk += 256) {
+    float v[32];
+    float am = 0.f;
+#pragma unroll
+    for (int j = 0; j < 32; j++) {
+      v[j] = b2f(xr[blk * 32 + j]);
+      am = fmaxf(am, fabsf(v[j]));
+    }
+    // e2m1 max magnitude = 6.0: scale = 2^e with absmax/2^e <= 6
+    int e = 0;
+    if (am > 0.f) {
+      int ee;
+      frexpf(am / 6.0f, &ee);  // am/6 = m * 2^ee, m in [0.5, 1)
+      e = ee;                  // 2^ee >= am/6
+    }
+    if (e < -126) e = -126;
+    if (e > 127) e = 127;
+    union { float f; uint32_t u; } sc;
+    sc.u = (uint32_t)(e + 127) << 23;       // 2^e
+    er[blk] = (uint8_t)(e + 127);
+    const float rs = 1.0f / sc.f;
+#pragma unroll
+    for (int d = 0; d < 4; d++) {           // 4 dwords of 8 nibbles
+      // HW: fp4 = RNE(value * scale_operand) -> pass 1/2^e; the pair
+      // selector must be a literal
+      uint32_t w = 0;
+      w = __builtin_amdgcn_cvt_scalef32_pk_fp4_f32(
+          w, v[d * 8 + 0], v[d * 8 + 1], rs, 0);
+      w = __builtin_amdgcn_cvt_scalef32_pk_fp4_f32(
+          w, v[d * 8 + 2], v[d * 8 + 3], rs, 1);
+      w = __builtin_amdgcn_cvt_scalef32_pk_fp4_f32(
+          w, v[d * 8 + 4], v[d * 8 + 5], rs, 2);
+      w = __builtin_amdgcn_cvt_scalef32_pk_fp4_f32(
+          w, v[d * 8 + 6], v[d * 8 + 7], rs, 3);
+      *(uint32_t*)(qr + blk * 16 + d * 4) = w;
+    }
+  }
+}
+
+extern "C" hipError_t launch_quant_fp4(const void* X, void* Q, void* E,
+                                       int M, int K, hipStream_t stream) {
+  if (K % 32 != 0) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(k_quant_fp4_rows, dim3(M), dim3(256), 0, stream,
+                     (const u16*)X, (uint8_t*)Q, (uint8_t*)E, K);
   return hipGetLastError();
 }
 

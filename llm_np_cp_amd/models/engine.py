@@ -59,9 +59,13 @@ class GPUModel:
         self.max_batch = max_batch
         self._pb = 0  # current sequence row during per-sequence prefill
         self.dtype = dtype
-        if dtype not in ("bf16", "fp8"):
+        if dtype not in ("bf16", "fp8", "fp4"):
             raise ValueError(f"unsupported dtype {dtype!r}")
-        self.fp8 = dtype == "fp8"
+        # fp4 = MXFP4 weights on the single-sequence DECODE GEMVs (half
+        # the fp8 stream); prefill and batched decode run on the fp8
+        # copy (the fp8 machinery flag covers both quantized modes)
+        self.fp8 = dtype in ("fp8", "fp4")
+        self.wq4 = dtype == "fp4"
         if kv_dtype not in ("bf16", "fp8"):
             raise ValueError(f"unsupported kv_dtype {kv_dtype!r}")
         self.kv_dtype = kv_dtype
@@ -152,6 +156,16 @@ class GPUModel:
             g = a.astype(np.float32) + (1.0 if gemma else 0.0)
             return torch.from_numpy(g).to(dev)
 
+        def quant4(a: np.ndarray):
+            """MXFP4: e2m1 packed 2/byte + e8m0 scale per 32 elems."""
+            t = bf16(a)
+            N, K = t.shape
+            q = torch.empty(N, K // 2, dtype=torch.uint8, device=dev)
+            e = torch.empty(N, K // 32, dtype=torch.uint8, device=dev)
+            ho.quant_fp4(t, q, e)
+            torch.cuda.synchronize()
+            return q, e
+
         def quant8(a: np.ndarray):
             """Per-output-row e4m3fn quantization ON DEVICE (k_quant_
             fp8_rows): one bf16 upload, quantize, drop the bf16 copy —
@@ -171,6 +185,8 @@ class GPUModel:
             lmw = lm_w if tp == 1 else tpu.shard_rows(lm_w, r, tp)
             self.lm_head_q, self.lm_head_s = quant8(lmw)
             self.lm_head = None  # single-copy: no bf16 duplicate
+            if self.wq4:
+                self.lm_head_q4, self.lm_head_e4 = quant4(lmw)
         elif tp > 1:
             self.lm_head = bf16(tpu.shard_rows(lm_w, r, tp))
         elif lm_w is w.get("model.embed_tokens.weight", None) or \
@@ -210,6 +226,8 @@ class GPUModel:
                 for name, arr in [("wqkv", qkv_np), ("wgu", gu_np),
                                   ("wo", o_np), ("wdown", down_np)]:
                     lw[name + "_q"], lw[name + "_s"] = quant8(arr)
+                    if self.wq4:
+                        lw[name + "_q4"], lw[name + "_e4"] = quant4(arr)
                 lw.update({
                     "wq_q": lw["wqkv_q"][:nq], "wq_s": lw["wqkv_s"][:nq],
                     "wk_q": lw["wqkv_q"][nq:nq + nkv],
@@ -694,8 +712,10 @@ class GPUModel:
     # fast device-side decode loop (graph-replayable)
     # ------------------------------------------------------------------
     def _dgemv(self, lw, name, x, y, **kw):
-        """Decode GEMV: fp8 weights when dtype=fp8, else bf16."""
-        if self.fp8:
+        """Decode GEMV: MXFP4 / fp8 / bf16 weights by engine dtype."""
+        if self.wq4:
+            ho.gemv_fp4(lw[name + "_q4"], lw[name + "_e4"], x, y, **kw)
+        elif self.fp8:
             ho.gemv_fp8(lw[name + "_q"], lw[name + "_s"], x, y, **kw)
         else:
             ho.gemv(lw[name], x, y, **kw)
@@ -803,7 +823,10 @@ class GPUModel:
             kw = dict(stage=ho.STAGE_NORM2, x2=h,
                       g=self.layers[-1]["g_postffn"], g2=self.g_final,
                       res=hnext, eps=eps, softcap=self.final_softcap)
-            if self.fp8:
+            if self.wq4:
+                ho.gemv_fp4(self.lm_head_q4, self.lm_head_e4, t2,
+                            self.b_logits_l, **kw)
+            elif self.fp8:
                 ho.gemv_fp8(self.lm_head_q, self.lm_head_s, t2,
                             self.b_logits_l, **kw)
             else:
@@ -812,7 +835,10 @@ class GPUModel:
             # final norm fused into the lm_head staging pass
             kw = dict(stage=ho.STAGE_NORM, g=self.g_final, eps=eps,
                       softcap=self.final_softcap)
-            if self.fp8:
+            if self.wq4:
+                ho.gemv_fp4(self.lm_head_q4, self.lm_head_e4, h,
+                            self.b_logits_l, **kw)
+            elif self.fp8:
                 ho.gemv_fp8(self.lm_head_q, self.lm_head_s, h,
                             self.b_logits_l, **kw)
             else:

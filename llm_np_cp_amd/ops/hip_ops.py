@@ -70,6 +70,12 @@ _SIGS = {
     "launch_i32_add": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
     "launch_quant_fp8": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 +
                         [ctypes.c_void_p],
+    "launch_quant_fp4": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 +
+                        [ctypes.c_void_p],
+    "launch_gemv_fp4": [ctypes.c_void_p] * 8 + [ctypes.c_int] * 4 +
+                       [ctypes.c_float, ctypes.c_int, ctypes.c_float,
+                        ctypes.c_int, ctypes.c_int, ctypes.c_int,
+                        ctypes.c_float, ctypes.c_void_p],
     "launch_stage_quant_mx": [ctypes.c_void_p, ctypes.c_long,
                               ctypes.c_void_p, ctypes.c_long] +
                              [ctypes.c_void_p] * 4 +
@@ -343,6 +349,36 @@ def quant_fp8(x: torch.Tensor, q: torch.Tensor, s: torch.Tensor):
     assert q.numel() >= M * K and s.numel() >= M
     _check(lib().launch_quant_fp8(_ptr(x), _ptr(q), _ptr(s), M, K,
                                   _stream()), "quant_fp8")
+
+
+def quant_fp4(x: torch.Tensor, q: torch.Tensor, e: torch.Tensor):
+    """MXFP4 (OCP MX) weight quantization on device: e2m1 nibbles packed
+    2/byte + one e8m0 scale byte per 32 elements.  x: (M,K) bf16;
+    q: >= M*K/2 uint8; e: >= M*K/32 uint8."""
+    M = 1 if x.dim() == 1 else x.shape[0]
+    K = x.shape[-1]
+    assert q.numel() >= M * K // 2 and e.numel() >= M * K // 32
+    _check(lib().launch_quant_fp4(_ptr(x), _ptr(q), _ptr(e), M, K,
+                                  _stream()), "quant_fp4")
+
+
+def gemv_fp4(Wq: torch.Tensor, We: torch.Tensor, x: torch.Tensor,
+             y: torch.Tensor, res: torch.Tensor | None = None,
+             softcap: float = 0.0, stage: int = 0,
+             x2: torch.Tensor | None = None, g: torch.Tensor | None = None,
+             act: int = 0, eps: float = 1e-5, nt: int = 1, rpw: int = 1,
+             maxblocks: int = 0, g2: torch.Tensor | None = None,
+             escale: float = 1.0):
+    """y[N] = W4[N,K] @ stage(x): MXFP4 weights (Wq: N x K/2 packed
+    bytes, We: N x K/32 e8m0 scale bytes) — half the fp8 stream."""
+    N = Wq.shape[0]
+    K = Wq.shape[1] * 2
+    out_f32 = 1 if y.dtype == torch.float32 else 0
+    _check(lib().launch_gemv_fp4(
+        _ptr(Wq), _ptr(We), _ptr(x), _ptr(x2), _ptr(g), _ptr(g2),
+        _ptr(y), _ptr(res), N, K, stage, act, ctypes.c_float(eps), out_f32,
+        ctypes.c_float(softcap), nt, rpw, maxblocks,
+        ctypes.c_float(escale), _stream()), "gemv_fp4")
 
 
 def gemm_fp8(xq: torch.Tensor, sx: torch.Tensor, Wq: torch.Tensor,
