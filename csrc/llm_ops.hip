@@ -2738,26 +2738,27 @@ k_quant_fp4_rows(const u16* __restrict__ X, uint8_t* __restrict__ Q,
       int ee;
       frexpf(am / 6.0f, &ee);  // am/6 = m * 2^ee, m in [0.5, 1)
       e = ee;                  // 2^ee >= am/6
+      if (ldexpf(6.0f, e - 1) >= am) e -= 1;  // exact-power tightening
     }
     if (e < -126) e = -126;
     if (e > 127) e = 127;
     union { float f; uint32_t u; } sc;
     sc.u = (uint32_t)(e + 127) << 23;       // 2^e
     er[blk] = (uint8_t)(e + 127);
-    const float rs = 1.0f / sc.f;
 #pragma unroll
     for (int d = 0; d < 4; d++) {           // 4 dwords of 8 nibbles
-      // HW: fp4 = RNE(value * scale_operand) -> pass 1/2^e; the pair
+      // HW semantics (device-verified): encode fp4 = RNE(v / scale),
+      // decode = v * scale -> pass the SAME 2^e both ways; the pair
       // selector must be a literal
       uint32_t w = 0;
       w = __builtin_amdgcn_cvt_scalef32_pk_fp4_f32(
-          w, v[d * 8 + 0], v[d * 8 + 1], rs, 0);
+          w, v[d * 8 + 0], v[d * 8 + 1], sc.f, 0);
       w = __builtin_amdgcn_cvt_scalef32_pk_fp4_f32(
-          w, v[d * 8 + 2], v[d * 8 + 3], rs, 1);
+          w, v[d * 8 + 2], v[d * 8 + 3], sc.f, 1);
       w = __builtin_amdgcn_cvt_scalef32_pk_fp4_f32(
-          w, v[d * 8 + 4], v[d * 8 + 5], rs, 2);
+          w, v[d * 8 + 4], v[d * 8 + 5], sc.f, 2);
       w = __builtin_amdgcn_cvt_scalef32_pk_fp4_f32(
-          w, v[d * 8 + 6], v[d * 8 + 7], rs, 3);
+          w, v[d * 8 + 6], v[d * 8 + 7], sc.f, 3);
       *(uint32_t*)(qr + blk * 16 + d * 4) = w;
     }
   }
