@@ -1,5 +1,7 @@
 """Serving-layer test (CPU: NumPy backend, tiny preset)."""
 
+import time
+
 import pytest
 
 
@@ -84,3 +86,43 @@ def test_server_dynamic_batching_groups_requests():
     stats = client.get("/stats").json()
     assert stats["requests"] == 4
     assert stats["max_group"] >= 2, stats  # batching actually engaged
+
+
+@pytest.mark.gpu
+def test_server_mid_flight_join():
+    """Continuous batching: a request arriving while a long request is
+    mid-decode JOINS the running group (joined_mid_flight > 0) instead
+    of waiting for it to finish."""
+    fastapi = pytest.importorskip("fastapi")
+    import threading
+    from fastapi.testclient import TestClient
+
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="gpu", max_seq=512,
+                    max_batch=4, batch_window_ms=1.0)
+    client = TestClient(app)
+
+    results = {}
+
+    def long_req():
+        results["long"] = client.post("/v1/completions", json={
+            "prompt": "a very long story begins", "max_tokens": 320,
+            "strategy": "greedy", "stop_on_eos": False}).json()
+
+    def short_req():
+        results["short"] = client.post("/v1/completions", json={
+            "prompt": "hi", "max_tokens": 8,
+            "strategy": "greedy", "stop_on_eos": False}).json()
+
+    t1 = threading.Thread(target=long_req)
+    t1.start()
+    time.sleep(1.0)  # the long request is decoding by now
+    t2 = threading.Thread(target=short_req)
+    t2.start()
+    t1.join(timeout=120)
+    t2.join(timeout=120)
+    assert results["long"]["usage"]["completion_tokens"] == 320
+    assert results["short"]["usage"]["completion_tokens"] == 8
+    stats = client.get("/stats").json()
+    assert stats["joined_mid_flight"] >= 1, stats
