@@ -100,29 +100,36 @@ def test_server_mid_flight_join():
     from llm_np_cp_amd.runtime.server import build_app
 
     app = build_app("tiny-llama", backend="gpu", max_seq=512,
-                    max_batch=4, batch_window_ms=1.0)
+                    max_batch=4, batch_window_ms=0.5)
     client = TestClient(app)
 
     results = {}
 
     def long_req():
         results["long"] = client.post("/v1/completions", json={
-            "prompt": "a very long story begins", "max_tokens": 320,
+            "prompt": "a very long story begins", "max_tokens": 480,
             "strategy": "greedy", "stop_on_eos": False}).json()
 
-    def short_req():
-        results["short"] = client.post("/v1/completions", json={
+    def short_req(i):
+        results[i] = client.post("/v1/completions", json={
             "prompt": "hi", "max_tokens": 8,
             "strategy": "greedy", "stop_on_eos": False}).json()
 
     t1 = threading.Thread(target=long_req)
     t1.start()
-    time.sleep(1.0)  # the long request is decoding by now
-    t2 = threading.Thread(target=short_req)
-    t2.start()
+    # the 480-token decode spans dozens of 16-token chunks (>=10 ms);
+    # stagger several short requests into that window
+    joiners = []
+    for i in range(3):
+        time.sleep(0.01)
+        t = threading.Thread(target=short_req, args=(i,))
+        t.start()
+        joiners.append(t)
     t1.join(timeout=120)
-    t2.join(timeout=120)
-    assert results["long"]["usage"]["completion_tokens"] == 320
-    assert results["short"]["usage"]["completion_tokens"] == 8
+    for t in joiners:
+        t.join(timeout=120)
+    assert results["long"]["usage"]["completion_tokens"] == 480
+    for i in range(3):
+        assert results[i]["usage"]["completion_tokens"] == 8
     stats = client.get("/stats").json()
     assert stats["joined_mid_flight"] >= 1, stats
