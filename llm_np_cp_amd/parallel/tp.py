@@ -38,6 +38,12 @@ def init_distributed(backend: Optional[str] = None):
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29517")
+    # keep RCCL capture-friendly: the watchdog/monitoring threads make
+    # collectives refuse hipGraph capture on some stacks (the one-shot
+    # xGMI kernels are the decode-path primary; this is fallback
+    # insurance for RCCL-in-graph)
+    os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "0")
+    os.environ.setdefault("TORCH_NCCL_ENABLE_MONITORING", "0")
     dist.init_process_group(backend=backend)
     return dist.get_rank(), dist.get_world_size()
 

@@ -194,6 +194,10 @@ def load_model(model_dir_or_preset: str, backend: str = "auto",
         try:
             from transformers import AutoTokenizer
             tokenizer = AutoTokenizer.from_pretrained(tok_dir)
+            # a dir with config.json but no tokenizer files can yield a
+            # "working" object that encodes everything to [] — verify
+            if not tokenizer.encode("test"):
+                tokenizer = None
         except Exception:
             tokenizer = None
     if tokenizer is None:

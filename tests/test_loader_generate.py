@@ -165,3 +165,19 @@ def test_numpy_forward_hf_tuple_surface():
     from llm_np_cp_amd.models.numpy_ref import NumpyKVCache
     ref = model.forward(ids, NumpyKVCache(cfg, 16), 0)
     np.testing.assert_allclose(logits, ref, rtol=1e-5, atol=1e-6)
+
+
+def test_qwen2_synthetic_checkpoint_roundtrip(tmp_path):
+    """Qwen-2 bias tensors survive the checkpoint writer -> loader ->
+    oracle path (config round-trips attention_bias)."""
+    from llm_np_cp_amd.io.loader import write_synthetic_checkpoint
+
+    d = str(tmp_path / "qwen")
+    write_synthetic_checkpoint(d, "tiny-qwen2", seed=9)
+    tok, model, cfg = L.load_model(d, backend="numpy")
+    assert cfg.model_type == "qwen2" and cfg.attention_bias
+    assert "model.layers.0.self_attn.q_proj.bias" in model.w
+    r = L.generate("ab", tok, model, max_tokens=3, stream=False,
+                   params=SamplingParams(strategy="greedy"),
+                   stop_on_eos=False)
+    assert len(r.token_ids) == 3
