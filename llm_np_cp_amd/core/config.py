@@ -287,6 +287,13 @@ PRESETS = {
         rope_theta=10000.0, max_position_embeddings=512, hidden_act="silu",
         tie_word_embeddings=True,
     ),
+    "tiny-llama-tp4": dict(  # TP=4-friendly (kvh=4, sharded K %64)
+        model_type="llama", vocab_size=512, hidden_size=256,
+        intermediate_size=512, num_hidden_layers=2, num_attention_heads=8,
+        num_key_value_heads=4, head_dim=64, rms_norm_eps=1e-5,
+        rope_theta=10000.0, max_position_embeddings=512, hidden_act="silu",
+        tie_word_embeddings=True,
+    ),
     "tiny-llama-tp": dict(   # TP=2-friendly: sharded o/down K stays %64
         model_type="llama", vocab_size=512, hidden_size=256,
         intermediate_size=256, num_hidden_layers=2, num_attention_heads=4,

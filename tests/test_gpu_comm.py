@@ -241,7 +241,7 @@ def test_xgmi_comm_two_processes_one_gpu():
 # collectives inside the graph
 # ----------------------------------------------------------------------
 
-def _tp2_worker(rank, world, port, q):
+def _tp2_worker(rank, world, port, q):  # noqa: C901
     os.environ.update({
         "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
         "RANK": str(rank), "WORLD_SIZE": str(world),
@@ -258,7 +258,8 @@ def _tp2_worker(rank, world, port, q):
         from llm_np_cp_amd.models.engine import GPUModel
         from llm_np_cp_amd.parallel import tp
 
-        cfg = L.preset_config("tiny-llama-tp")
+        preset = "tiny-llama-tp" if world <= 2 else "tiny-llama-tp4"
+        cfg = L.preset_config(preset)
         w = random_weights(cfg, seed=0)
         model = GPUModel(cfg, w, max_seq=128, device="cuda:0")
         assert tp.xgmi_comm() is not None, \
@@ -285,7 +286,8 @@ def test_tp_engine_graph_decode_one_gpu(world):
     from llm_np_cp_amd.models.engine import GPUModel
 
     # TP=1 reference in the parent (same weights)
-    cfg = L.preset_config("tiny-llama-tp")
+    cfg = L.preset_config("tiny-llama-tp" if world <= 2
+                          else "tiny-llama-tp4")
     w = random_weights(cfg, seed=0)
     ref = GPUModel(cfg, w, max_seq=128, device="cuda:0")
     prompt = np.arange(1, 9, dtype=np.int32)
