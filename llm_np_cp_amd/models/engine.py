@@ -1152,6 +1152,8 @@ class GPUModel:
         between decode chunks (server continuous batching)."""
         ids = np.asarray(ids, dtype=np.int32).ravel()
         P = len(ids)
+        if P == 0:
+            raise ValueError("empty prompt")
         assert 0 <= b < self.max_batch and self.max_batch > 1
         assert self.world == 1
         if P + 1 >= self.max_seq:

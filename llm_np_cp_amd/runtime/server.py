@@ -85,7 +85,11 @@ class BatchScheduler:
 
     def _poll_compatible(self, key, deferred):
         """Non-blocking: next queued pending with this sampling key;
-        others are deferred (re-queued after the group)."""
+        others are deferred (re-queued after the group).  Once anything
+        is deferred, stop admitting joiners so the running group drains
+        and the deferred requests cannot starve behind it."""
+        if deferred:
+            return None
         while True:
             try:
                 p = self.q.get_nowait()
