@@ -1,5 +1,6 @@
 #!/usr/bin/env python3
-"""Quantify fp8-weight decode accuracy vs the bf16 engine (llama-1b)."""
+"""Quantify quantized-weight decode accuracy vs the bf16 engine
+(llama-1b).  Usage: python tools/fp8_accuracy.py [fp8|fp4]"""
 import os, sys
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import numpy as np
@@ -9,16 +10,17 @@ import llm_np_cp_amd as L
 from llm_np_cp_amd.io.loader import LazyRandomWeights
 from llm_np_cp_amd.models.engine import GPUModel
 
+qdtype = sys.argv[1] if len(sys.argv) > 1 else "fp8"
 cfg = L.preset_config("llama-3.2-1b")
 w = LazyRandomWeights(cfg, 0)
 bf = GPUModel(cfg, w, max_seq=512)
-f8 = GPUModel(cfg, w, max_seq=512, dtype="fp8")
+f8 = GPUModel(cfg, w, max_seq=512, dtype=qdtype)
 rng = np.random.default_rng(0)
 prompt = rng.integers(0, cfg.vocab_size, size=64)
 
 cb = bf.make_cache(512); cf = f8.make_cache(512)
 lb = bf.forward(prompt, cb, 0)[0]
-lf = f8.forward(prompt, cf, 0)[0]  # prefill uses bf16 weights in both
+lf = f8.forward(prompt, cf, 0)[0]
 agree = 0; kl_sum = 0.0; n_steps = 64
 ids_b, ids_f = [], []
 for i in range(n_steps):
@@ -32,4 +34,4 @@ for i in range(n_steps):
     lf = f8.forward(np.asarray([tf]), cf, cf.seq_len)[0]
 print(f"greedy token agreement over {n_steps} separate rollouts: "
       f"{agree}/{n_steps}")
-print(f"mean KL(bf16 || fp8) per step: {kl_sum/n_steps:.5f} nats")
+print(f"mean KL(bf16 || {qdtype}) per step: {kl_sum/n_steps:.5f} nats")
