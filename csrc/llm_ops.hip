@@ -2966,6 +2966,173 @@ extern "C" hipError_t launch_gemm_fp8(const void* X, const void* sx,
 }
 
 // ====================================================================
+// Prefill GEMM with MXFP4 weights: Y[M,N] = X[M,K] @ dequant(W4)[N,K]^T
+// (+res).  bf16 MFMA (activations stay bf16 — no activation quant, so
+// this path is MORE accurate than the fp8 prefill); the B operand is
+// dequantized fp4 -> bf16 in the staging pass (register loads + scaled
+// hardware converts), A keeps the async glds DMA.  Completes the
+// single-copy MXFP4 story for max_batch=1 engines.
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_gemm_fp4w(const u16* __restrict__ X, const uint8_t* __restrict__ W4,
+            const uint8_t* __restrict__ WE, u16* __restrict__ Y,
+            const u16* __restrict__ res, float* __restrict__ accbuf,
+            int M, int N, int K) {
+  __shared__ u16 As[2][128 * 64];
+  __shared__ u16 Bs[2][128 * 64];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wrow = wave >> 1, wcol = wave & 1;
+  const int bm = blockIdx.x * BM, bn = blockIdx.y * BN;
+
+  const int SK = gridDim.z;
+  const int kslices = (K / BK + SK - 1) / SK;
+  const int k_lo = blockIdx.z * kslices * BK;
+  int k_hi = k_lo + kslices * BK;
+  if (k_hi > K) k_hi = K;
+  const int nt = (k_hi - k_lo) / BK;
+  if (nt <= 0) return;
+
+  f4v acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; i++)
+#pragma unroll
+    for (int j = 0; j < 4; j++) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int g_r = lane >> 3;
+  const int g_s = lane & 7;
+  // A side: async glds DMA with pre-swizzled source slots (as bf16 GEMM)
+  auto stage_a = [&](int buf, int kt) {
+    const int k0 = k_lo + kt * BK;
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+      int r = wave * 32 + i * 8 + g_r;
+      int cs = g_s ^ (r & 7);
+      int gr = bm + r;
+      int grc = gr < M ? gr : (M > 0 ? M - 1 : 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)(
+              X + (size_t)grc * K + k0 + cs * 8),
+          (__attribute__((address_space(3))) uint32_t*)(
+              &As[buf][(wave * 32 + i * 8) * 64]),
+          16, 0, 0);
+    }
+  };
+  // B side: 2 threads per row (halves of BK=64), fp4 -> bf16 into the
+  // swizzled LDS slots
+  auto stage_b = [&](int buf, int kt) {
+    const int k0 = k_lo + kt * BK;
+    const int r = tid >> 1, h = tid & 1;
+    int gb = bn + r;
+    int gbc = gb < N ? gb : N - 1;
+    const uint8_t* wp = W4 + (size_t)gbc * (K / 2) + (k0 + h * 32) / 2;
+    union { float f; uint32_t u; } sc;
+    sc.u = (uint32_t)WE[(size_t)gbc * (K / 32) + (k0 + h * 32) / 32] << 23;
+    u4v_ pk = *(const u4v_*)wp;          // 16 bytes = 32 fp4 values
+#pragma unroll
+    for (int d = 0; d < 4; d++) {        // 8 values -> one 16-B slot
+      f2v p[4];
+      fp4x8_to_f32p(pk[d], sc.f, p);
+      u16 o[8];
+#pragma unroll
+      for (int j = 0; j < 4; j++) {
+        o[j * 2] = f2b(p[j][0]);
+        o[j * 2 + 1] = f2b(p[j][1]);
+      }
+      const int slot = (h * 4 + d) ^ (r & 7);
+      *(s8v*)(&Bs[buf][r * 64 + slot * 8]) = *(s8v*)o;
+    }
+  };
+
+  stage_a(0, 0);
+  stage_b(0, 0);
+  __syncthreads();
+
+  const int fr = lane & 15, fk = lane >> 4;
+  int cur = 0;
+  for (int t = 0; t < nt; t++) {
+    if (t + 1 < nt) {
+      stage_a(cur ^ 1, t + 1);
+      stage_b(cur ^ 1, t + 1);
+    }
+#pragma unroll
+    for (int sl = 0; sl < 2; sl++) {
+      bf16x8 a[4], b[4];
+#pragma unroll
+      for (int i = 0; i < 4; i++) {
+        int ra = wrow * 64 + i * 16 + fr;
+        int sa = (sl * 4 + fk) ^ (ra & 7);
+        a[i] = *(bf16x8*)(&As[cur][ra * 64 + sa * 8]);
+        int rb = wcol * 64 + i * 16 + fr;
+        int sb = (sl * 4 + fk) ^ (rb & 7);
+        b[i] = *(bf16x8*)(&Bs[cur][rb * 64 + sb * 8]);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; i++)
+#pragma unroll
+        for (int j = 0; j < 4; j++)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int cc = lane & 15, cr = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        int row = bm + wrow * 64 + i * 16 + cr + r;
+        int col = bn + wcol * 64 + j * 16 + cc;
+        if (row < M && col < N) {
+          float v = acc[i][j][r];
+          if (SK > 1) {
+            atomicAdd(accbuf + (size_t)row * N + col, v);
+          } else {
+            if (res) v += b2f(res[(size_t)row * N + col]);
+            Y[(size_t)row * N + col] = f2b(v);
+          }
+        }
+      }
+    }
+  }
+}
+
+extern "C" hipError_t launch_gemm_fp4w(const void* X, const void* W4,
+                                       const void* WE, void* Y,
+                                       const void* res, void* accbuf, int M,
+                                       int N, int K, hipStream_t stream) {
+  if (K % 64 != 0) return hipErrorInvalidValue;
+  int gm = (M + BM - 1) / BM, gn = (N + BN - 1) / BN;
+  int sk = 1;
+  if (accbuf && gm * gn < 160) {
+    while (sk < 8 && gm * gn * sk * 2 <= 512 && (K / BK) % (sk * 2) == 0)
+      sk *= 2;
+  }
+  if (sk > 1) {
+    long total = (long)M * N;
+    hipLaunchKernelGGL(k_zero_f32, dim3((uint32_t)((total / 4 + 255) / 256)),
+                       dim3(256), 0, stream, (float*)accbuf, total);
+  }
+  dim3 grid(gm, gn, sk);
+  hipLaunchKernelGGL(k_gemm_fp4w, grid, dim3(256), 0, stream, (const u16*)X,
+                     (const uint8_t*)W4, (const uint8_t*)WE, (u16*)Y,
+                     (const u16*)res, (float*)accbuf, M, N, K);
+  if (sk > 1) {
+    long total = (long)M * N;
+    hipLaunchKernelGGL(k_gemm_fin, dim3((uint32_t)((total + 255) / 256)),
+                       dim3(256), 0, stream, (const float*)accbuf,
+                       (const u16*)res, (u16*)Y, total);
+  }
+  return hipGetLastError();
+}
+
+// ====================================================================
 // Weight prefetcher (side-stream): stream a tensor through L2/MALL with
 // plain (retaining) loads so the 256 MB Infinity Cache holds the NEXT
 // layer's weights before its compute kernel issues.  The sink write

@@ -66,6 +66,9 @@ class GPUModel:
         # copy (the fp8 machinery flag covers both quantized modes)
         self.fp8 = dtype in ("fp8", "fp4")
         self.wq4 = dtype == "fp4"
+        # fp4 + batch pools: the skinny MFMA GEMM consumes fp8, so keep
+        # the fp8 copy only then; max_batch=1 fp4 engines are SINGLE-COPY
+        self.keep_fp8 = (dtype == "fp8") or (self.wq4 and max_batch > 1)
         if kv_dtype not in ("bf16", "fp8"):
             raise ValueError(f"unsupported kv_dtype {kv_dtype!r}")
         self.kv_dtype = kv_dtype
@@ -183,7 +186,8 @@ class GPUModel:
         lm_w = w.get("lm_head.weight", w["model.embed_tokens.weight"])
         if self.fp8:
             lmw = lm_w if tp == 1 else tpu.shard_rows(lm_w, r, tp)
-            self.lm_head_q, self.lm_head_s = quant8(lmw)
+            if self.keep_fp8:
+                self.lm_head_q, self.lm_head_s = quant8(lmw)
             self.lm_head = None  # single-copy: no bf16 duplicate
             if self.wq4:
                 self.lm_head_q4, self.lm_head_e4 = quant4(lmw)
@@ -225,7 +229,8 @@ class GPUModel:
                 # both read these (no bf16 duplicate in HBM)
                 for name, arr in [("wqkv", qkv_np), ("wgu", gu_np),
                                   ("wo", o_np), ("wdown", down_np)]:
-                    lw[name + "_q"], lw[name + "_s"] = quant8(arr)
+                    if self.keep_fp8:
+                        lw[name + "_q"], lw[name + "_s"] = quant8(arr)
                     if self.wq4:
                         lw[name + "_q4"], lw[name + "_e4"] = quant4(arr)
                 lw.update({
@@ -400,11 +405,18 @@ class GPUModel:
         place: bf16 GEMV/MFMA-GEMM, or (fp8) GEMV / fp8-MFMA GEMM with
         on-the-fly activation row quantization."""
         if M == 1:
-            if self.fp8:
+            if self.wq4:
+                ho.gemv_fp4(lw[name + "_q4"], lw[name + "_e4"], x, y,
+                            res=res, softcap=softcap)
+            elif self.fp8:
                 ho.gemv_fp8(lw[name + "_q"], lw[name + "_s"], x, y,
                             res=res, softcap=softcap)
             else:
                 ho.gemv(lw[name], x, y, res=res, softcap=softcap)
+        elif self.wq4:
+            ho.gemm_fp4w(x[:M], lw[name + "_q4"], lw[name + "_e4"], y[:M],
+                         res=res[:M] if res is not None else None,
+                         accbuf=self.b_gemm_acc)
         elif self.fp8:
             K = x.shape[-1]
             ho.quant_fp8(x[:M], self.b_xq, self.b_sx)
@@ -544,7 +556,10 @@ class GPUModel:
         hrow = self.b_h[M - 1] if M > 1 else self.b_h[0]
         kw = dict(stage=ho.STAGE_NORM, g=self.g_final,
                   eps=self.config.rms_norm_eps, softcap=self.final_softcap)
-        if self.fp8:
+        if self.wq4:
+            ho.gemv_fp4(self.lm_head_q4, self.lm_head_e4, hrow,
+                        self.b_logits_l, **kw)
+        elif self.fp8:
             ho.gemv_fp8(self.lm_head_q, self.lm_head_s, hrow,
                         self.b_logits_l, **kw)
         else:
@@ -589,7 +604,11 @@ class GPUModel:
                 self._layers_forward(M)
             ho.rmsnorm(self.b_h[:M], self.g_final, self.b_xn[:M],
                        eps=self.config.rms_norm_eps)
-            if self.fp8:
+            if self.wq4:
+                ho.gemm_fp4w(self.b_xn[:M], self.lm_head_q4,
+                             self.lm_head_e4, logits_buf[:M],
+                             accbuf=self.b_gemm_acc)
+            elif self.fp8:
                 ho.quant_fp8(self.b_xn[:M], self.b_xq, self.b_sx)
                 ho.gemm_fp8(self.b_xq, self.b_sx, self.lm_head_q,
                             self.lm_head_s, logits_buf[:M], M, self.H,

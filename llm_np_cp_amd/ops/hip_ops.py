@@ -99,6 +99,8 @@ _SIGS = {
                            ctypes.c_float, ctypes.c_void_p],
     "launch_gemm_fp8": [ctypes.c_void_p] * 7 + [ctypes.c_int] * 3 +
                        [ctypes.c_void_p],
+    "launch_gemm_fp4w": [ctypes.c_void_p] * 6 + [ctypes.c_int] * 3 +
+                        [ctypes.c_void_p],
     # one-shot xGMI collectives (csrc/xgmi_comm.hip)
     "launch_xgmi_coll": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                          ctypes.c_int, ctypes.c_int, ctypes.c_long,
@@ -453,6 +455,22 @@ def gemm_fp8_skinny(xq: torch.Tensor, sx: torch.Tensor, Wq: torch.Tensor,
         ctypes.c_long(ystride), _ptr(res), ctypes.c_long(rstride),
         _ptr(bias), _ptr(accbuf), out_f32, ctypes.c_float(softcap),
         B, N, K, _stream()), "gemm_fp8_skinny")
+
+
+def gemm_fp4w(X: torch.Tensor, Wq4: torch.Tensor, We4: torch.Tensor,
+              Y: torch.Tensor, res: torch.Tensor | None = None,
+              accbuf: torch.Tensor | None = None):
+    """Y[M,N] = X[M,K] @ dequant(W4)[N,K]^T (+res): bf16-MFMA prefill
+    over MXFP4 weights (single-copy fp4 engines; activations unquantized
+    -> more accurate than the fp8 prefill path)."""
+    M, K = X.shape
+    N = Wq4.shape[0]
+    assert Wq4.shape[1] * 2 == K and K % 64 == 0
+    if accbuf is not None and accbuf.numel() < M * N:
+        accbuf = None
+    _check(lib().launch_gemm_fp4w(
+        _ptr(X), _ptr(Wq4), _ptr(We4), _ptr(Y), _ptr(res), _ptr(accbuf),
+        M, N, K, _stream()), "gemm_fp4w")
 
 
 def prefetch(t: torch.Tensor, sink: torch.Tensor):

@@ -619,11 +619,31 @@ def test_fp4_engine_decode_on_grid_weights():
             w[k] = a.astype(np.float32)
     bf = GPUModel(cfg, dict(w), max_seq=128)
     f4 = GPUModel(cfg, dict(w), max_seq=128, dtype="fp4")
+    assert not hasattr(f4, "lm_head_q"), "max_batch=1 fp4 is single-copy"
     prompt = np.arange(1, 9)
-    bf.prefill(prompt)
+    _, la = bf.prefill(prompt)
     a = bf.decode(6, greedy=True, use_graph=False)
-    f4.prefill(prompt)
+    _, lb = f4.prefill(prompt)
     b = f4.decode(6, greedy=True, use_graph=True)
+    # grid weights quantize losslessly: the fp4w bf16-MFMA prefill must
+    # track the bf16 engine tightly
+    np.testing.assert_allclose(lb, la, rtol=5e-2, atol=5e-2)
     assert a[0] == b[0], (a, b)
     # at least the early steps agree (kernel rounding may diverge later)
     assert list(a[:3]) == list(b[:3]), (a, b)
+
+
+def test_fp4_batch_decode_keeps_fp8_copy():
+    """fp4 + max_batch>1: batch pools require the fp8 skinny path, so
+    the fp8 copy is retained and batched decode works."""
+    import llm_np_cp_amd as L
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.engine import GPUModel
+
+    cfg = L.preset_config("tiny-llama")
+    w = random_weights(cfg, seed=99)
+    m = GPUModel(cfg, w, max_seq=64, max_batch=4, dtype="fp4")
+    assert hasattr(m, "lm_head_q")
+    ids = m.generate_tokens_batch(
+        [np.arange(1, 7), np.arange(2, 9)], 4, greedy=True)
+    assert ids.shape == (2, 4)
