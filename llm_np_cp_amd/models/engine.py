@@ -233,15 +233,31 @@ class GPUModel:
                         lw[name + "_q"], lw[name + "_s"] = quant8(arr)
                     if self.wq4:
                         lw[name + "_q4"], lw[name + "_e4"] = quant4(arr)
-                lw.update({
-                    "wq_q": lw["wqkv_q"][:nq], "wq_s": lw["wqkv_s"][:nq],
-                    "wk_q": lw["wqkv_q"][nq:nq + nkv],
-                    "wk_s": lw["wqkv_s"][nq:nq + nkv],
-                    "wv_q": lw["wqkv_q"][nq + nkv:],
-                    "wv_s": lw["wqkv_s"][nq + nkv:],
-                    "wgate_q": lw["wgu_q"][:I], "wgate_s": lw["wgu_s"][:I],
-                    "wup_q": lw["wgu_q"][I:], "wup_s": lw["wgu_s"][I:],
-                })
+                if self.keep_fp8:
+                    lw.update({
+                        "wq_q": lw["wqkv_q"][:nq],
+                        "wq_s": lw["wqkv_s"][:nq],
+                        "wk_q": lw["wqkv_q"][nq:nq + nkv],
+                        "wk_s": lw["wqkv_s"][nq:nq + nkv],
+                        "wv_q": lw["wqkv_q"][nq + nkv:],
+                        "wv_s": lw["wqkv_s"][nq + nkv:],
+                        "wgate_q": lw["wgu_q"][:I],
+                        "wgate_s": lw["wgu_s"][:I],
+                        "wup_q": lw["wgu_q"][I:], "wup_s": lw["wgu_s"][I:],
+                    })
+                if self.wq4:
+                    lw.update({
+                        "wq_q4": lw["wqkv_q4"][:nq],
+                        "wq_e4": lw["wqkv_e4"][:nq],
+                        "wk_q4": lw["wqkv_q4"][nq:nq + nkv],
+                        "wk_e4": lw["wqkv_e4"][nq:nq + nkv],
+                        "wv_q4": lw["wqkv_q4"][nq + nkv:],
+                        "wv_e4": lw["wqkv_e4"][nq + nkv:],
+                        "wgate_q4": lw["wgu_q4"][:I],
+                        "wgate_e4": lw["wgu_e4"][:I],
+                        "wup_q4": lw["wgu_q4"][I:],
+                        "wup_e4": lw["wgu_e4"][I:],
+                    })
             else:
                 wqkv = bf16(qkv_np)
                 wgu = bf16(gu_np)
