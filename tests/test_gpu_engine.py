@@ -613,7 +613,10 @@ def test_fp4_engine_decode_on_grid_weights():
                     dtype=np.float32) * 0.02
     rng = np.random.default_rng(98)
     for k in w:
-        if "proj.weight" in k or k in ("lm_head.weight",):
+        # embed_tokens included: it is the TIED lm_head, which the fp4
+        # engine quantizes
+        if ("proj.weight" in k or "embed_tokens" in k
+                or k in ("lm_head.weight",)):
             a = grid[rng.integers(0, len(grid), size=w[k].shape)]
             a.reshape(-1, 32)[:, 0] = 6.0 * 0.02  # pin block scales
             w[k] = a.astype(np.float32)
