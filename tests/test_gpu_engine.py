@@ -608,9 +608,11 @@ def test_fp4_engine_decode_on_grid_weights():
 
     cfg = L.preset_config("tiny-llama")
     w = random_weights(cfg, seed=97)
+    # scale must be a POWER OF TWO (e8m0): grid/32 keeps every value
+    # exactly on the e2m1-times-2^-5 lattice -> lossless quantization
     grid = np.array([0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0,
                      -0.5, -1.0, -1.5, -2.0, -3.0, -4.0, -6.0],
-                    dtype=np.float32) * 0.02
+                    dtype=np.float32) / 32.0
     rng = np.random.default_rng(98)
     for k in w:
         # embed_tokens included: it is the TIED lm_head, which the fp4
@@ -618,7 +620,7 @@ def test_fp4_engine_decode_on_grid_weights():
         if ("proj.weight" in k or "embed_tokens" in k
                 or k in ("lm_head.weight",)):
             a = grid[rng.integers(0, len(grid), size=w[k].shape)]
-            a.reshape(-1, 32)[:, 0] = 6.0 * 0.02  # pin block scales
+            a.reshape(-1, 32)[:, 0] = 6.0 / 32.0  # pin block scales
             w[k] = a.astype(np.float32)
     bf = GPUModel(cfg, dict(w), max_seq=128)
     f4 = GPUModel(cfg, dict(w), max_seq=128, dtype="fp4")
