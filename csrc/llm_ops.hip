@@ -416,7 +416,9 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
   }
   if (rpw_env_b > 0) rpw = rpw_env_b;
   else if (rpw <= 1)
-    rpw = (K >= 3584 && N >= 3072) ? 2 : 1;  // same policy as fp8 GEMV
+    // deep-K only: the fp8 GEMV's large-N rpw rule measured NEGATIVE
+    // for bf16 (1b 1234->1218, gemma-2b 536->526)
+    rpw = (K >= 3584 && N >= 3072) ? 2 : 1;
   int threads = 256;  // 512 measured slower (fp8 1391->1328)
   int wpb = threads / 64;
   int blocks = (N + wpb * rpw - 1) / (wpb * rpw);
@@ -595,8 +597,10 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
   else if (rpw <= 1)
     // deep-K rows: 2x loads in flight (gemma-9b 240->300, llama-8b
     // 393->442, qwen-7b 352->442); N floor keeps small-N shapes (1B
-    // down-proj) at full block count
-    rpw = (K >= 3584 && N >= 3072) ? 2 : 1;
+    // down-proj) at full block count.  Very large N (gate+up, lm_head)
+    // also gains: waves already grid-stride many rows, RPW doubles the
+    // in-flight rows per iteration.
+    rpw = ((K >= 3584 && N >= 3072) || N >= 12288) ? 2 : 1;
   const int xdir = (stage == STAGE_RAW) && xdir_raw;
   // fp32 staging for packed math (all non-RAW stages; RAW reads direct)
   size_t lds = xdir ? 0 : ((size_t)K * 4 + 32);
