@@ -1838,39 +1838,39 @@ k_attn_dec_t(const u16* __restrict__ qkv, void* __restrict__ kc,
   {
     int t0 = c0 + wave * PP;
     if (t0 < c1) {
-      int t = t0 + p;
-      bool valid = t < c1;
-      int tl = valid ? t : c0;
-      s8v kv, vv;
-      unsigned long long kraw = 0, vraw = 0;
-      float ksc = 1.f, vsc = 1.f;
-      if (KV8) {
-        kraw = *(const unsigned long long*)(K08 + (size_t)tl * hd + d0);
-        vraw = *(const unsigned long long*)(V08 + (size_t)tl * hd + d0);
-        ksc = kS0[tl]; vsc = vS0[tl];
-      } else {
-        kv = *(const s8v*)(K0 + (size_t)tl * hd + d0);
-        vv = *(const s8v*)(V0 + (size_t)tl * hd + d0);
-      }
-      for (; t0 < c1; t0 += 4 * PP) {
-        int t0n = t0 + 4 * PP;
-        s8v kvn, vvn;
-        unsigned long long krawn = 0, vrawn = 0;
-        float kscn = 1.f, vscn = 1.f;
-        bool validn = false;
-        if (t0n < c1) {
-          int tn = t0n + p;
-          validn = tn < c1;
-          int tln = validn ? tn : c0;
+      // 2-AHEAD software pipeline: K/V for iterations i+1 and i+2 are
+      // in flight while iteration i computes (the serial
+      // K->dot->softmax->V chain is the latency bound at small T)
+      auto ldkv = [&](int tt, bool& vld, s8v& ko, s8v& vo,
+                      unsigned long long& kro, unsigned long long& vro,
+                      float& kso, float& vso) {
+        vld = false;
+        if (tt < c1) {
+          int t_ = tt + p;
+          vld = t_ < c1;
+          int tl_ = vld ? t_ : c0;
           if (KV8) {
-            krawn = *(const unsigned long long*)(K08 + (size_t)tln * hd + d0);
-            vrawn = *(const unsigned long long*)(V08 + (size_t)tln * hd + d0);
-            kscn = kS0[tln]; vscn = vS0[tln];
+            kro = *(const unsigned long long*)(K08 + (size_t)tl_ * hd + d0);
+            vro = *(const unsigned long long*)(V08 + (size_t)tl_ * hd + d0);
+            kso = kS0[tl_]; vso = vS0[tl_];
           } else {
-            kvn = *(const s8v*)(K0 + (size_t)tln * hd + d0);
-            vvn = *(const s8v*)(V0 + (size_t)tln * hd + d0);
+            ko = *(const s8v*)(K0 + (size_t)tl_ * hd + d0);
+            vo = *(const s8v*)(V0 + (size_t)tl_ * hd + d0);
           }
         }
+      };
+      s8v kv{}, vv{}, kvn{}, vvn{};
+      unsigned long long kraw = 0, vraw = 0, krawn = 0, vrawn = 0;
+      float ksc = 1.f, vsc = 1.f, kscn = 1.f, vscn = 1.f;
+      bool valid, validn;
+      ldkv(t0, valid, kv, vv, kraw, vraw, ksc, vsc);
+      ldkv(t0 + 4 * PP, validn, kvn, vvn, krawn, vrawn, kscn, vscn);
+      for (; t0 < c1; t0 += 4 * PP) {
+        s8v kv2{}, vv2{};
+        unsigned long long kraw2 = 0, vraw2 = 0;
+        float ksc2 = 1.f, vsc2 = 1.f;
+        bool valid2;
+        ldkv(t0 + 8 * PP, valid2, kv2, vv2, kraw2, vraw2, ksc2, vsc2);
         float partial = 0.f;
         if (KV8) {
           // dequant straight to f32 in the dot (4 cvt + 8 fma per 8
@@ -1917,6 +1917,8 @@ k_attn_dec_t(const u16* __restrict__ qkv, void* __restrict__ kc,
         if (mnew != -INFINITY) mrun = mnew;
         kv = kvn; vv = vvn; kraw = krawn; vraw = vrawn;
         valid = validn; ksc = kscn; vsc = vscn;
+        kvn = kv2; vvn = vv2; krawn = kraw2; vrawn = vraw2;
+        validn = valid2; kscn = ksc2; vscn = vsc2;
       }
     }
   }
