@@ -456,12 +456,15 @@ class GPUModel:
             return
         import torch.distributed as dist
         K = x.shape[-1]
-        if self.fp8:
+        if self.fp8 and not self.wq4:
             ho.quant_fp8(x[:M], self.b_xq, self.b_sx)
         works = []
         M2 = M // 2
         for a, b in ((0, M2), (M2, M)):
-            if self.fp8:
+            if self.wq4:
+                ho.gemm_fp4w(x[a:b], lw[name + "_q4"], lw[name + "_e4"],
+                             y[a:b], accbuf=self.b_gemm_acc)
+            elif self.fp8:
                 ho.gemm_fp8(self.b_xq[a * K:], self.b_sx[a:],
                             lw[name + "_q"], lw[name + "_s"], y[a:b],
                             b - a, K, accbuf=self.b_gemm_acc)
