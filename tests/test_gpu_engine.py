@@ -513,15 +513,18 @@ def test_batch_decode_ragged_lengths():
         np.testing.assert_array_equal(ids[b], singles[b])
 
 
-@pytest.mark.parametrize("preset,B", [("tiny-llama", 2),
-                                      ("tiny-gemma2", 2),
-                                      ("tiny-llama", 4),
-                                      ("tiny-gemma2", 4),
-                                      ("tiny-llama", 7)])
-def test_batch_decode_fused_paths_match_single_fp8(preset, B):
-    """The fused fp8 batch paths (multi-x GEMV at B<=2, skinny MFMA
-    GEMM at B=3..16) must reproduce each row's single-sequence fp8
-    rollout (ragged lengths)."""
+@pytest.mark.parametrize("preset,B,force_mx", [
+    ("tiny-llama", 2, False), ("tiny-gemma2", 2, False),
+    ("tiny-llama", 4, False), ("tiny-gemma2", 4, False),
+    ("tiny-llama", 7, False),
+    ("tiny-llama", 2, True), ("tiny-gemma2", 4, True)])
+def test_batch_decode_fused_paths_match_single_fp8(preset, B, force_mx,
+                                                   monkeypatch):
+    """The fused fp8 batch paths (skinny MFMA GEMM default; the multi-x
+    GEMV reference path via LLM_BATCH_MX_MAX) must reproduce each row's
+    single-sequence fp8 rollout (ragged lengths)."""
+    if force_mx:
+        monkeypatch.setenv("LLM_BATCH_MX_MAX", "16")
     import llm_np_cp_amd as L
     from llm_np_cp_amd.io.loader import random_weights
     from llm_np_cp_amd.models.engine import GPUModel
