@@ -448,16 +448,6 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
 
 typedef uint32_t u4v __attribute__((ext_vector_type(4)));  // = u4v_ alias
 
-DEVINL void fp8x16_to_f32(u4v w, float* o) {
-#pragma unroll
-  for (int q = 0; q < 4; q++) {
-    f2v a = __builtin_amdgcn_cvt_pk_f32_fp8(w[q], false);
-    f2v b = __builtin_amdgcn_cvt_pk_f32_fp8(w[q], true);
-    o[q * 4 + 0] = a[0]; o[q * 4 + 1] = a[1];
-    o[q * 4 + 2] = b[0]; o[q * 4 + 3] = b[1];
-  }
-}
-
 // Inner loop is PACKED: x is staged as fp32 in LDS (gemv_stage_f32),
 // each v_cvt_pk_f32_fp8 output pair feeds one v_pk_fma_f32 against an
 // LDS fp32 pair — ~16 VALU per 16 weight bytes vs ~40 for the round-1

@@ -759,6 +759,9 @@ class GPUModel:
             ho.gemv(lw[name], x, y, **kw)
 
     def _pf_tensors(self, lw):
+        if self.wq4:
+            return (lw["wqkv_q4"], lw["wo_q4"], lw["wgu_q4"],
+                    lw["wdown_q4"])
         if self.fp8:
             return (lw["wqkv_q"], lw["wo_q"], lw["wgu_q"], lw["wdown_q"])
         return (lw["wqkv"], lw["wo"], lw["wgu"], lw["wdown"])
