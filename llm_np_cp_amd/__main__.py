@@ -20,7 +20,9 @@ def main(argv=None):
     ap.add_argument("--max-tokens", type=int, default=200)
     ap.add_argument("--backend", default="auto",
                     choices=["auto", "gpu", "numpy"])
-    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp8"])
+    ap.add_argument("--dtype", default="bf16",
+                    choices=["bf16", "fp8", "fp4"])
+    ap.add_argument("--kv-dtype", default="bf16", choices=["bf16", "fp8"])
     ap.add_argument("--strategy", default="min_p",
                     choices=["min_p", "greedy", "top_k", "top_p",
                              "temperature"])
@@ -36,7 +38,8 @@ def main(argv=None):
     import llm_np_cp_amd as L
 
     tok, model, cfg = L.load_model(args.model, backend=args.backend,
-                                   dtype=args.dtype, max_seq=args.max_seq)
+                                   dtype=args.dtype, max_seq=args.max_seq,
+                                   kv_dtype=args.kv_dtype)
     params = L.SamplingParams(strategy=args.strategy, min_p=args.min_p,
                               temperature=args.temperature, seed=args.seed)
     t0 = time.time()

@@ -181,3 +181,13 @@ def test_qwen2_synthetic_checkpoint_roundtrip(tmp_path):
                    params=SamplingParams(strategy="greedy"),
                    stop_on_eos=False)
     assert len(r.token_ids) == 3
+
+
+def test_cli_main_numpy(capsys):
+    """`python -m llm_np_cp_amd` parity entry (reference __main__)."""
+    from llm_np_cp_amd.__main__ import main
+
+    main(["hello world", "--model", "tiny-llama", "--backend", "numpy",
+          "--max-tokens", "4", "--strategy", "greedy"])
+    out = capsys.readouterr()
+    assert len(out.out) > 0  # streamed something
