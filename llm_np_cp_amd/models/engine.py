@@ -346,8 +346,10 @@ class GPUModel:
         # split-T decode attention scratch (partials + per-head tickets)
         import os as _os
         _sp = _os.environ.get("LLM_ATTN_SPLIT")
+        # measured (fp8 llama-1b): T~100 split2 1493 > split4 1483;
+        # T~4k split8 1108 > split16 1050; T~8k split16 963 > split32
         self.attn_split = (int(_sp) if _sp else
-                           min(16, max(4, (self.max_seq + 255) // 256)))
+                           min(16, max(2, self.max_seq // 512)))
         self.attn_scratch = torch.zeros(
             self.nh_l * self.attn_split * (hd + 2), dtype=torch.float32,
             device=dev)
