@@ -909,7 +909,7 @@ class GPUModel:
         # 1.44 ms — the GEMV is VALU-ISSUE-bound, 64 irreducible
         # v_pk_fma per 16 weight bytes at B=8).  The multi-x path stays
         # available via LLM_BATCH_MX_MAX for comparison.
-        import os as _os
+        import os as _os  # noqa: delayed so env changes apply per call
         mx_max = int(_os.environ.get("LLM_BATCH_MX_MAX", "0"))
         if self.fp8 and B <= mx_max and not cfg.attention_bias:
             return self._decode_batch_step_mx(B, greedy, min_p, temperature)
@@ -1252,7 +1252,7 @@ class GPUModel:
         """n lockstep steps for rows [0, B) (each already holding a
         sampled next token from prefill_row / a previous chunk).
         Returns a list of B arrays: each row's n new token ids."""
-        assert 1 <= B <= getattr(self, "_batch_n", 0) or B <= self.max_batch
+        assert 1 <= B <= self.max_batch
         if max(self._host_lens[:B]) + n > self.max_seq:
             raise ValueError("decode_rows would overflow the KV pool")
         for b in range(B):
