@@ -997,8 +997,12 @@ k_gemm_fp8_skinny(const uint8_t* __restrict__ Xq,
     const uint8_t* Wr = Wq + (size_t)wr * K + fk8;
     f4v acc_a = {0.f, 0.f, 0.f, 0.f};
     f4v acc_b = {0.f, 0.f, 0.f, 0.f};
+    f4v acc_c = {0.f, 0.f, 0.f, 0.f};
+    f4v acc_d = {0.f, 0.f, 0.f, 0.f};
     int k = k_lo;
-    for (; k + 128 <= k_hi; k += 128) {  // 2 independent 64-B chains
+    // 4 independent chains x 256 B: PMC showed waves PARKED on memory
+    // 69% of cycles with only 2 chains in flight
+    for (; k + 256 <= k_hi; k += 256) {
       long a0 = *(const long*)(Xr + k);
       long b0 = __builtin_nontemporal_load((const long*)(Wr + k));
       long a1 = *(const long*)(Xr + k + 32);
@@ -1007,13 +1011,29 @@ k_gemm_fp8_skinny(const uint8_t* __restrict__ Xq,
       long b2 = __builtin_nontemporal_load((const long*)(Wr + k + 64));
       long a3 = *(const long*)(Xr + k + 96);
       long b3 = __builtin_nontemporal_load((const long*)(Wr + k + 96));
+      long a4 = *(const long*)(Xr + k + 128);
+      long b4 = __builtin_nontemporal_load((const long*)(Wr + k + 128));
+      long a5 = *(const long*)(Xr + k + 160);
+      long b5 = __builtin_nontemporal_load((const long*)(Wr + k + 160));
+      long a6 = *(const long*)(Xr + k + 192);
+      long b6 = __builtin_nontemporal_load((const long*)(Wr + k + 192));
+      long a7 = *(const long*)(Xr + k + 224);
+      long b7 = __builtin_nontemporal_load((const long*)(Wr + k + 224));
       acc_a = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a0, b0, acc_a,
                                                          0, 0, 0);
       acc_b = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a1, b1, acc_b,
                                                          0, 0, 0);
-      acc_a = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a2, b2, acc_a,
+      acc_c = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a2, b2, acc_c,
                                                          0, 0, 0);
-      acc_b = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a3, b3, acc_b,
+      acc_d = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a3, b3, acc_d,
+                                                         0, 0, 0);
+      acc_a = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a4, b4, acc_a,
+                                                         0, 0, 0);
+      acc_b = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a5, b5, acc_b,
+                                                         0, 0, 0);
+      acc_c = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a6, b6, acc_c,
+                                                         0, 0, 0);
+      acc_d = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a7, b7, acc_d,
                                                          0, 0, 0);
     }
     for (; k < k_hi; k += 32) {
@@ -1027,7 +1047,7 @@ k_gemm_fp8_skinny(const uint8_t* __restrict__ Xq,
     for (int r = 0; r < 4; r++) {
       const int b = (lane >> 4) * 4 + r;
       if (b < B && col < N) {
-        float v = acc_a[r] + acc_b[r];
+        float v = (acc_a[r] + acc_b[r]) + (acc_c[r] + acc_d[r]);
         if (SK > 1) {
           atomicAdd(accbuf + (size_t)b * N + col, v);
         } else {
