@@ -142,3 +142,47 @@ def test_hf_config_field_variants():
     assert cfg2.head_dim == 16  # derived hidden/heads
     plain = ModelConfig.from_hf_dict({**d2, "rope_scaling": None})
     assert np.any(cfg2.rope_inv_freq() < plain.rope_inv_freq())
+
+
+def test_all_presets_hf_dict_roundtrip():
+    """Every preset survives to_hf_dict -> from_hf_dict (config drift
+    guard across llama/gemma2/qwen2/mistral families)."""
+    import llm_np_cp_amd as L
+    from llm_np_cp_amd.core.config import PRESETS, ModelConfig
+
+    for name in PRESETS:
+        a = L.preset_config(name)
+        b = ModelConfig.from_hf_dict(a.to_hf_dict())
+        for f in ("model_type", "vocab_size", "hidden_size",
+                  "intermediate_size", "num_hidden_layers",
+                  "num_attention_heads", "num_key_value_heads", "head_dim",
+                  "rms_norm_eps", "rope_theta", "hidden_act",
+                  "tie_word_embeddings", "attention_bias",
+                  "query_pre_attn_scalar", "sliding_window",
+                  "attn_logit_softcapping", "final_logit_softcapping"):
+            assert getattr(a, f) == getattr(b, f), (name, f)
+        assert a.layer_types == b.layer_types, name
+
+
+def test_real_presets_tp_divisibility():
+    """The TP sharding preconditions (engine asserts) hold for the real
+    model presets at their natural TP degrees."""
+    import llm_np_cp_amd as L
+
+    expect = {
+        "llama-3.2-1b": [1, 2, 4, 8],
+        "llama-3.2-3b": [1, 2, 4, 8],
+        "llama-3.1-8b": [1, 2, 4, 8],
+        "gemma-2-2b": [1, 2, 4],
+        "gemma-2-9b": [1, 2, 4, 8],
+        "gemma-2-27b": [1, 2, 4, 8],
+        "qwen2.5-7b": [1, 2, 4],
+        "mistral-7b": [1, 2, 4, 8],
+    }
+    for name, tps in expect.items():
+        cfg = L.preset_config(name)
+        for tp in tps:
+            assert cfg.num_attention_heads % tp == 0, (name, tp)
+            assert cfg.num_key_value_heads % tp == 0, (name, tp)
+            assert cfg.intermediate_size % tp == 0, (name, tp)
+            assert cfg.vocab_size % tp == 0, (name, tp)
