@@ -1112,6 +1112,8 @@ class GPUModel:
         assert 1 <= B <= self.max_batch and self.max_batch > 1
         assert self.world == 1, "batched decode is single-GPU for now"
         lens = [len(sq) for sq in seqs]
+        if min(lens) == 0:
+            raise ValueError("empty prompt in batch")
         if max(lens) + 1 >= self.max_seq:
             raise ValueError(f"prompt {max(lens)} fills the "
                              f"{self.max_seq} pool")
