@@ -598,11 +598,13 @@ class GPUModel:
         llama3.2_model.py:682-716) as fp32 numpy arrays."""
         ids = np.asarray(ids, dtype=np.int32).ravel()
         self.reset()
+        if not return_hidden_states:
+            return self.forward_positions(ids, 0)
         n = len(ids)
         out = np.empty((n, self.config.vocab_size), dtype=np.float32)
         L = self.config.num_hidden_layers
-        hidden = ([np.empty((n, self.H), dtype=np.float32)
-                   for _ in range(L + 1)] if return_hidden_states else None)
+        hidden = [np.empty((n, self.H), dtype=np.float32)
+                  for _ in range(L + 1)]
         logits_buf = torch.empty(self.PC, self.vocab_l,
                                  dtype=torch.bfloat16, device=self.device)
         done = 0
@@ -613,16 +615,13 @@ class GPUModel:
             ho.i32_set(self.len_buf, done)
             ho.embed(self.embed, self.ids_buf, self.b_h, M,
                      self.config.embed_scale)
-            if return_hidden_states:
-                d0 = done
+            d0 = done
 
-                def hook(i, h):
-                    torch.cuda.synchronize()
-                    hidden[i + 1][d0:d0 + h.shape[0]] = \
-                        h.float().cpu().numpy()
-                self._layers_forward(M, layer_hook=hook)
-            else:
-                self._layers_forward(M)
+            def hook(i, h):
+                torch.cuda.synchronize()
+                hidden[i + 1][d0:d0 + h.shape[0]] = \
+                    h.float().cpu().numpy()
+            self._layers_forward(M, layer_hook=hook)
             ho.rmsnorm(self.b_h[:M], self.g_final, self.b_xn[:M],
                        eps=self.config.rms_norm_eps)
             if self.wq4:
@@ -659,6 +658,75 @@ class GPUModel:
         if return_hidden_states:
             return out, hidden
         return out
+
+    def forward_positions(self, ids: np.ndarray, pos0: int) -> np.ndarray:
+        """All-position logits (M, V) fp32, CONTINUING from pos0 (no
+        cache reset: KV for ids lands at [pos0, pos0+M) and the pool can
+        be rolled back afterwards with ``rewind``).  This is the
+        speculative-decoding VERIFY pass: the target model scores k
+        draft tokens in one prefill-shaped pass instead of k decode
+        steps (ROADMAP §5; the reference has nothing comparable)."""
+        ids = np.asarray(ids, dtype=np.int32).ravel()
+        n = len(ids)
+        if n == 0:
+            raise ValueError("empty ids")
+        if pos0 + n > self.max_seq:
+            raise ValueError(f"sequence {pos0}+{n} exceeds max_seq "
+                             f"{self.max_seq}")
+        out = np.empty((n, self.config.vocab_size), dtype=np.float32)
+        logits_buf = torch.empty(self.PC, self.vocab_l,
+                                 dtype=torch.bfloat16, device=self.device)
+        done = 0
+        while done < n:
+            M = min(n - done, self.PC)
+            self.ids_buf[:M].copy_(
+                torch.from_numpy(ids[done:done + M].astype(np.int32)))
+            ho.i32_set(self.len_buf, pos0 + done)
+            ho.embed(self.embed, self.ids_buf, self.b_h, M,
+                     self.config.embed_scale)
+            self._layers_forward(M)
+            ho.rmsnorm(self.b_h[:M], self.g_final, self.b_xn[:M],
+                       eps=self.config.rms_norm_eps)
+            if self.wq4:
+                ho.gemm_fp4w(self.b_xn[:M], self.lm_head_q4,
+                             self.lm_head_e4, logits_buf[:M],
+                             accbuf=self.b_gemm_acc)
+            elif self.fp8:
+                ho.quant_fp8(self.b_xn[:M], self.b_xq, self.b_sx)
+                ho.gemm_fp8(self.b_xq, self.b_sx, self.lm_head_q,
+                            self.lm_head_s, logits_buf[:M], M, self.H,
+                            accbuf=self.b_gemm_acc)
+            else:
+                ho.gemm(self.b_xn[:M], self.lm_head, logits_buf[:M],
+                        accbuf=self.b_gemm_acc)
+            if self.final_softcap:
+                ho.softcap(logits_buf[:M], self.final_softcap)
+            if self.world > 1:
+                # gather vocab shards on-device (each rank holds the
+                # columns [r*vocab_l, (r+1)*vocab_l) of rows [0, M))
+                import torch.distributed as dist
+                t = logits_buf[:M].float().contiguous()
+                chunks = [torch.empty_like(t) for _ in range(self.world)]
+                dist.all_gather(chunks, t)
+                loc = torch.cat(chunks, dim=1).cpu().numpy()
+            else:
+                torch.cuda.synchronize()
+                loc = logits_buf[:M].float().cpu().numpy()
+            out[done:done + M] = loc
+            done += M
+        ho.i32_set(self.len_buf, pos0 + n)
+        self._host_len = pos0 + n
+        return out
+
+    def rewind(self, n: int):
+        """Roll the sequence back to length n — O(1): the KV pool is
+        preallocated and the attention kernels read the device length
+        from len_buf, so positions >= n are dead until overwritten.
+        Speculative decoding uses this to discard rejected draft KV."""
+        if not 0 <= n <= self.max_seq:
+            raise ValueError(f"rewind {n} outside [0, {self.max_seq}]")
+        ho.i32_set(self.len_buf, n)
+        self._host_len = n
 
     def forward_hf(self, ids: np.ndarray):
         """Reference-parity output tuple ``(loss, logits, kv_cache,
