@@ -9,7 +9,8 @@ the ``lm_head.weight -> model.embed_tokens.weight`` tying alias.
 Differences by design (SURVEY §5 checkpoint/resume):
 - no bare ``except:`` fallback (the reference swallowed all errors,
   ``llama3.2_model.py:1063``) — missing files raise;
-- no network (`snapshot_download`): local directories only, plus a
+- this module reads LOCAL directories (hub repo ids are resolved one
+  level up: ``load_model`` -> ``_hub_download`` -> this reader), plus a
   synthetic random-init path for the no-network benchmark environment;
 - weights go straight to the target dtype (bf16/fp32), not through the
   reference's fp32-upcast-then-transfer detour (``llama3.2_model.py:1079``).
