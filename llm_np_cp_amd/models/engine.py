@@ -539,6 +539,9 @@ class GPUModel:
                         window=window or 0,
                         kS=self._ks(i), vS=self._vs(i))
                 else:
+                    # per-query VALU kernel: every production head_dim is
+                    # 64/128/256 (MFMA above); kept as the documented
+                    # debugging/odd-shape fallback
                     ho.attn(self.b_q, self._kc(i), self._vc(i),
                             self.b_att, self.len_buf, M, self.nh_l,
                             self.kvh_l, self.hd, self.scale,
