@@ -331,7 +331,8 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="llama-3.2-1b")
     ap.add_argument("--backend", default="auto")
-    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp8"])
+    ap.add_argument("--dtype", default="bf16",
+                    choices=["bf16", "fp8", "fp4"])
     ap.add_argument("--kv-dtype", default="bf16", choices=["bf16", "fp8"])
     ap.add_argument("--max-seq", type=int, default=4096)
     ap.add_argument("--max-batch", type=int, default=8)
