@@ -214,3 +214,32 @@ def test_mistral_all_layer_sliding_window():
     # early positions (inside the window) agree; late ones differ
     np.testing.assert_allclose(a[:8], b[:8], rtol=1e-5, atol=1e-6)
     assert not np.allclose(a[-1], b[-1])
+
+
+def test_sliding_window_geq_seq_equals_full_attention():
+    """window >= sequence length must degenerate to full attention
+    (guards the windowed-scan bounds in the oracle and, transitively,
+    the GPU kernels validated against it)."""
+    import numpy as np
+    from llm_np_cp_amd.core.config import preset_config
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.numpy_ref import NumpyModel, NumpyKVCache
+
+    base = preset_config("tiny-mistral")
+    w = random_weights(base, seed=5)
+    ids = np.arange(1, 13)
+
+    wide = preset_config("tiny-mistral")
+    wide.sliding_window = 64          # > seq: no position masked out
+    l_wide = NumpyModel(wide, w).forward(ids, NumpyKVCache(wide, 32), 0)
+
+    off = preset_config("tiny-mistral")
+    off.sliding_window = None         # layer_types already frozen in
+    off.layer_types = ["full_attention"] * off.num_hidden_layers
+    l_full = NumpyModel(off, w).forward(ids, NumpyKVCache(off, 32), 0)
+
+    assert np.allclose(l_wide, l_full, atol=1e-5)
+
+    tight = preset_config("tiny-mistral")  # window 8 < 12 really masks
+    l_tight = NumpyModel(tight, w).forward(ids, NumpyKVCache(tight, 32), 0)
+    assert not np.allclose(l_tight, l_full, atol=1e-3)
