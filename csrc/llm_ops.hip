@@ -23,7 +23,7 @@
 //                               k_rmsnorm, k_glu, k_softcap, k_addinto
 //  K4 KV concat              -> in-place pool writes
 //  K5 repeat_kv              -> attention indexes kv_head = q_head/groups
-//  K7 torch.multinomial      -> k_logit_max + k_sample_pick + k_sample_fin
+//  K7 torch.multinomial      -> k_logit_max + k_sample_pick (last-block commit)
 
 #include <hip/hip_runtime.h>
 #include <cstdint>
@@ -1097,7 +1097,7 @@ extern "C" hipError_t launch_gemm_fp8_skinny(
   }
   if (sk > 1) {
     long total = (long)B * N;
-    hipLaunchKernelGGL(k_zero_f32, dim3((uint32_t)((total / 4 + 255) / 256)),
+    hipLaunchKernelGGL(k_zero_f32, dim3((uint32_t)((total + 1023) / 1024)),
                        dim3(256), 0, stream, (float*)accbuf, total);
   }
   hipLaunchKernelGGL(k_gemm_fp8_skinny, dim3(blocks, sk), dim3(256), 0,
@@ -2390,9 +2390,14 @@ DEVINL int unpack_idx(uint64_t p) { return 0x7fffffff - (int)(uint32_t)p; }
 // batched lm_head GEMM path emits bf16 rows).
 extern "C" __global__ void __launch_bounds__(256)
 k_logit_max(const void* __restrict__ logits, int V, int lbf16,
-            unsigned long long* __restrict__ gmax) {
+            unsigned long long* __restrict__ gmax,
+            uint64_t* __restrict__ ctr) {
   const int b = blockIdx.y;
   gmax += b;
+  // advance the shared RNG counter HERE: stream order guarantees every
+  // k_sample_pick block then reads the same post-bump value (bumping
+  // from pick's commit raced with straggler blocks still reading it)
+  if (ctr && blockIdx.x == 0 && b == 0 && threadIdx.x == 0) *ctr += 1;
   const float* lf = (const float*)logits + (size_t)b * V;
   const u16* lh = (const u16*)logits + (size_t)b * V;
   float mv = -INFINITY;
@@ -2488,10 +2493,9 @@ k_sample_pick(const void* __restrict__ logits, int V, int lbf16,
   int n = *nout;
   out_ring[n] = winner;
   *nout = n + 1;
-  // per-row position advances (ragged batch); the RNG counter is
-  // shared and advanced once per step (row 0's committer)
+  // per-row position advances (ragged batch); the shared RNG counter
+  // was already advanced by k_logit_max ahead of this kernel
   if (bump_len) len_ptr[b] += 1;
-  if (b == 0 && !greedy) *ctr += 1;
   *pick = 0ull;
   *gmax = 0ull;
 }
@@ -2514,7 +2518,8 @@ extern "C" hipError_t launch_sample(const void* logits, int V, int lbf16,
   }
   if (!greedy)
     hipLaunchKernelGGL(k_logit_max, dim3(blocks, batch), dim3(256), 0,
-                       stream, logits, V, lbf16, (unsigned long long*)gmax);
+                       stream, logits, V, lbf16, (unsigned long long*)gmax,
+                       (uint64_t*)ctr);
   hipLaunchKernelGGL(k_sample_pick, dim3(blocks, batch), dim3(256), 0,
                      stream, logits, V, lbf16, ring_stride, min_p, greedy,
                      seed, inv_temp,
@@ -2677,8 +2682,11 @@ k_gemm_fin(const float* __restrict__ accbuf, const u16* __restrict__ res,
 extern "C" __global__ void __launch_bounds__(256)
 k_zero_f32(float* __restrict__ p, long total) {
   long i = ((long)blockIdx.x * 256 + threadIdx.x) * 4;
-  if (i >= total) return;
-  *(f4v*)(p + i) = {0.f, 0.f, 0.f, 0.f};
+  if (i + 4 <= total) {
+    *(f4v*)(p + i) = {0.f, 0.f, 0.f, 0.f};
+  } else {
+    for (; i < total; i++) p[i] = 0.f;  // tail: no over-write past total
+  }
 }
 
 extern "C" hipError_t launch_gemm_bf16(const void* X, const void* W, void* Y,
@@ -2702,7 +2710,7 @@ extern "C" hipError_t launch_gemm_bf16(const void* X, const void* W, void* Y,
   }
   if (sk > 1) {
     long total = (long)M * N;
-    hipLaunchKernelGGL(k_zero_f32, dim3((uint32_t)((total / 4 + 255) / 256)),
+    hipLaunchKernelGGL(k_zero_f32, dim3((uint32_t)((total + 1023) / 1024)),
                        dim3(256), 0, stream, (float*)accbuf, total);
   }
   dim3 grid(gm, gn, sk);
@@ -2964,7 +2972,7 @@ extern "C" hipError_t launch_gemm_fp8(const void* X, const void* sx,
   }
   if (sk > 1) {
     long total = (long)M * N;
-    hipLaunchKernelGGL(k_zero_f32, dim3((uint32_t)((total / 4 + 255) / 256)),
+    hipLaunchKernelGGL(k_zero_f32, dim3((uint32_t)((total + 1023) / 1024)),
                        dim3(256), 0, stream, (float*)accbuf, total);
   }
   dim3 grid(gm, gn, sk);
@@ -3132,7 +3140,7 @@ extern "C" hipError_t launch_gemm_fp4w(const void* X, const void* W4,
   }
   if (sk > 1) {
     long total = (long)M * N;
-    hipLaunchKernelGGL(k_zero_f32, dim3((uint32_t)((total / 4 + 255) / 256)),
+    hipLaunchKernelGGL(k_zero_f32, dim3((uint32_t)((total + 1023) / 1024)),
                        dim3(256), 0, stream, (float*)accbuf, total);
   }
   dim3 grid(gm, gn, sk);
