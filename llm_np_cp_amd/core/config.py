@@ -41,6 +41,10 @@ class ModelConfig:
     bos_token_id: int = 1
     eos_token_id: int = 2
     attention_bias: bool = False  # Qwen-2(.5): bias on q/k/v projections
+    # Mixtral sparse MoE: E experts per MLP, top-k routed per token
+    # (0 = dense MLP)
+    num_local_experts: int = 0
+    num_experts_per_tok: int = 2
     # Gemma-2 specifics
     query_pre_attn_scalar: Optional[float] = None
     sliding_window: Optional[int] = None
@@ -78,6 +82,10 @@ class ModelConfig:
         if self.model_type == "gemma2" and self.query_pre_attn_scalar is not None:
             return self.query_pre_attn_scalar ** -0.5
         return self.head_dim ** -0.5
+
+    @property
+    def is_moe(self) -> bool:
+        return self.num_local_experts > 0
 
     def is_sliding(self, layer_idx: int) -> bool:
         return (
@@ -117,6 +125,8 @@ class ModelConfig:
             # HF Qwen2Attention hardwires qkv bias=True (no config key)
             attention_bias=d.get("attention_bias",
                                  model_type == "qwen2"),
+            num_local_experts=d.get("num_local_experts", 0),
+            num_experts_per_tok=d.get("num_experts_per_tok", 2),
             query_pre_attn_scalar=d.get("query_pre_attn_scalar"),
             sliding_window=d.get("sliding_window"),
             attn_logit_softcapping=d.get("attn_logit_softcapping"),
@@ -150,6 +160,9 @@ class ModelConfig:
         }
         if self.attention_bias:
             d["attention_bias"] = True
+        if self.num_local_experts:
+            d["num_local_experts"] = self.num_local_experts
+            d["num_experts_per_tok"] = self.num_experts_per_tok
         if self.rope_scaling is not None:
             d["rope_scaling"] = self.rope_scaling
         for k in ("query_pre_attn_scalar", "sliding_window",
@@ -279,7 +292,26 @@ PRESETS = {
         query_pre_attn_scalar=144, sliding_window=4096,
         attn_logit_softcapping=50.0, final_logit_softcapping=30.0,
     ),
+    # Mixtral 8x7B sparse MoE: 8 experts, top-2 routing, GQA
+    # (shape table from the public mistralai/Mixtral-8x7B-v0.1 config.json)
+    "mixtral-8x7b": dict(
+        model_type="mixtral", vocab_size=32000, hidden_size=4096,
+        intermediate_size=14336, num_hidden_layers=32,
+        num_attention_heads=32, num_key_value_heads=8, head_dim=128,
+        rms_norm_eps=1e-5, rope_theta=1000000.0,
+        max_position_embeddings=32768, hidden_act="silu",
+        tie_word_embeddings=False, bos_token_id=1, eos_token_id=2,
+        num_local_experts=8, num_experts_per_tok=2,
+    ),
     # tiny configs for tests
+    "tiny-mixtral": dict(       # sparse MoE: 4 experts, top-2
+        model_type="mixtral", vocab_size=512, hidden_size=64,
+        intermediate_size=128, num_hidden_layers=2, num_attention_heads=4,
+        num_key_value_heads=2, head_dim=16, rms_norm_eps=1e-5,
+        rope_theta=10000.0, max_position_embeddings=512, hidden_act="silu",
+        tie_word_embeddings=True, num_local_experts=4,
+        num_experts_per_tok=2,
+    ),
     "tiny-llama-hd64": dict(    # exercises the hd=64 MFMA prefill path
         model_type="llama", vocab_size=512, hidden_size=128,
         intermediate_size=256, num_hidden_layers=2, num_attention_heads=2,

@@ -93,9 +93,20 @@ def hf_weight_shapes(cfg: ModelConfig) -> Dict[str, Tuple[int, ...]]:
             shapes[f"{p}.self_attn.k_proj.bias"] = (kvh * hd,)
             shapes[f"{p}.self_attn.v_proj.bias"] = (kvh * hd,)
         shapes[f"{p}.self_attn.o_proj.weight"] = (h, nh * hd)
-        shapes[f"{p}.mlp.gate_proj.weight"] = (im, h)
-        shapes[f"{p}.mlp.up_proj.weight"] = (im, h)
-        shapes[f"{p}.mlp.down_proj.weight"] = (h, im)
+        if cfg.is_moe:
+            # Mixtral hub naming: router + per-expert w1 (gate),
+            # w3 (up), w2 (down)
+            shapes[f"{p}.block_sparse_moe.gate.weight"] = \
+                (cfg.num_local_experts, h)
+            for e in range(cfg.num_local_experts):
+                q = f"{p}.block_sparse_moe.experts.{e}"
+                shapes[f"{q}.w1.weight"] = (im, h)
+                shapes[f"{q}.w3.weight"] = (im, h)
+                shapes[f"{q}.w2.weight"] = (h, im)
+        else:
+            shapes[f"{p}.mlp.gate_proj.weight"] = (im, h)
+            shapes[f"{p}.mlp.up_proj.weight"] = (im, h)
+            shapes[f"{p}.mlp.down_proj.weight"] = (h, im)
         shapes[f"{p}.input_layernorm.weight"] = (h,)
         shapes[f"{p}.post_attention_layernorm.weight"] = (h,)
         if cfg.model_type == "gemma2":

@@ -154,11 +154,37 @@ class NumpyModel:
         return out @ self.w[f"{p}.o_proj.weight"].T
 
     def _mlp(self, layer: int, h: np.ndarray) -> np.ndarray:
+        if self.config.is_moe:
+            return self._moe_mlp(layer, h)
         p = f"model.layers.{layer}.mlp"
         act = ACT2FN[self.config.hidden_act]
         gate = act(h @ self.w[f"{p}.gate_proj.weight"].T)
         up = h @ self.w[f"{p}.up_proj.weight"].T
         return (gate * up) @ self.w[f"{p}.down_proj.weight"].T
+
+    def _moe_mlp(self, layer: int, h: np.ndarray) -> np.ndarray:
+        """Mixtral sparse MoE (HF semantics, modeling_mixtral
+        MixtralTopKRouter/MixtralExperts): softmax over ALL experts ->
+        top-k probs renormalized by their own sum -> weighted sum of
+        the routed experts' SwiGLU outputs."""
+        cfg = self.config
+        p = f"model.layers.{layer}.block_sparse_moe"
+        act = ACT2FN[cfg.hidden_act]
+        E, K = cfg.num_local_experts, cfg.num_experts_per_tok
+        logits = h @ self.w[f"{p}.gate.weight"].T          # (M, E)
+        probs = softmax(logits.astype(np.float64), axis=-1)
+        top = np.argsort(-probs, axis=-1)[:, :K]           # (M, K)
+        out = np.zeros_like(h)
+        for m in range(h.shape[0]):
+            pe = probs[m, top[m]]
+            pe = pe / pe.sum()
+            for j, e in enumerate(top[m]):
+                q = f"{p}.experts.{e}"
+                gate = act(h[m] @ self.w[f"{q}.w1.weight"].T)
+                up = h[m] @ self.w[f"{q}.w3.weight"].T
+                out[m] += np.float32(pe[j]) * (
+                    (gate * up) @ self.w[f"{q}.w2.weight"].T)
+        return out
 
     # -- forward ---------------------------------------------------------
     def forward(self, input_ids: np.ndarray, cache: NumpyKVCache,

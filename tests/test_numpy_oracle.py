@@ -243,3 +243,101 @@ def test_sliding_window_geq_seq_equals_full_attention():
     tight = preset_config("tiny-mistral")  # window 8 < 12 really masks
     l_tight = NumpyModel(tight, w).forward(ids, NumpyKVCache(tight, 32), 0)
     assert not np.allclose(l_tight, l_full, atol=1e-3)
+
+
+def hf_mixtral(cfg):
+    from transformers import MixtralConfig, MixtralForCausalLM
+
+    hf_cfg = MixtralConfig(
+        vocab_size=cfg.vocab_size, hidden_size=cfg.hidden_size,
+        intermediate_size=cfg.intermediate_size,
+        num_hidden_layers=cfg.num_hidden_layers,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        head_dim=cfg.head_dim, rms_norm_eps=cfg.rms_norm_eps,
+        rope_theta=cfg.rope_theta,
+        max_position_embeddings=cfg.max_position_embeddings,
+        num_local_experts=cfg.num_local_experts,
+        num_experts_per_tok=cfg.num_experts_per_tok,
+        tie_word_embeddings=cfg.tie_word_embeddings,
+        attn_implementation="eager",
+    )
+    torch.manual_seed(11)
+    return MixtralForCausalLM(hf_cfg).eval()
+
+
+def np_weights_from_hf_mixtral(hf_model, cfg):
+    """transformers>=5 stores fused per-expert tensors
+    (mlp.experts.gate_up_proj (E,2I,H), .down_proj (E,H,I)); Mixtral hub
+    checkpoints (and this repo) use block_sparse_moe.experts.{e}.w1/w3/w2
+    — split them back out."""
+    I = cfg.intermediate_size
+    w = {}
+    for k, v in hf_model.state_dict().items():
+        a = v.detach().to(torch.float32).numpy()
+        if ".mlp.gate.weight" in k:
+            w[k.replace(".mlp.gate.", ".block_sparse_moe.gate.")] = a
+        elif ".mlp.experts.gate_up_proj" in k:
+            p = k.replace(".mlp.experts.gate_up_proj",
+                          ".block_sparse_moe.experts")
+            for e in range(a.shape[0]):
+                w[f"{p}.{e}.w1.weight"] = a[e, :I]
+                w[f"{p}.{e}.w3.weight"] = a[e, I:]
+        elif ".mlp.experts.down_proj" in k:
+            p = k.replace(".mlp.experts.down_proj",
+                          ".block_sparse_moe.experts")
+            for e in range(a.shape[0]):
+                w[f"{p}.{e}.w2.weight"] = a[e]
+        else:
+            w[k] = a
+    return w
+
+
+def test_mixtral_moe_forward_matches_transformers():
+    cfg = preset_config("tiny-mixtral")
+    hf = hf_mixtral(cfg)
+    model = NumpyModel(cfg, np_weights_from_hf_mixtral(hf, cfg))
+
+    rng = np.random.default_rng(3)
+    ids = rng.integers(0, cfg.vocab_size, size=12)
+    with torch.no_grad():
+        ref = hf(torch.tensor(ids[None])).logits[0].numpy()
+    got = model.forward(ids, NumpyKVCache(cfg, 64), 0)
+    np.testing.assert_allclose(got, ref, rtol=3e-4, atol=3e-4)
+
+
+def test_mixtral_incremental_decode_matches_prefill():
+    cfg = preset_config("tiny-mixtral")
+    hf = hf_mixtral(cfg)
+    model = NumpyModel(cfg, np_weights_from_hf_mixtral(hf, cfg))
+    rng = np.random.default_rng(4)
+    ids = rng.integers(0, cfg.vocab_size, size=10)
+    full = model.forward(ids, NumpyKVCache(cfg, 64), 0)
+    inc = NumpyKVCache(cfg, 64)
+    model.forward(ids[:4], inc, 0)
+    out = None
+    for t in range(4, 10):
+        out = model.forward(ids[t:t + 1], inc, t)
+    np.testing.assert_allclose(out[0], full[-1], rtol=1e-4, atol=1e-4)
+
+
+def test_mixtral_routing_is_sparse():
+    """Different tokens route to different experts; zeroing an UNROUTED
+    expert's weights must not change that token's output."""
+    from llm_np_cp_amd.io.loader import random_weights
+    cfg = preset_config("tiny-mixtral")
+    w = random_weights(cfg, seed=8)
+    model = NumpyModel(cfg, dict(w))
+    h = np.asarray(np.random.default_rng(0).standard_normal(
+        (1, cfg.hidden_size)), dtype=np.float32)
+    p = "model.layers.0.block_sparse_moe"
+    logits = h @ w[f"{p}.gate.weight"].T
+    order = np.argsort(-logits[0])
+    unused = order[-1]  # least-likely expert: not in top-2 (E=4)
+    out0 = model._moe_mlp(0, h)
+    w2 = dict(w)
+    for nm in ("w1", "w2", "w3"):
+        w2[f"{p}.experts.{unused}.{nm}.weight"] = \
+            np.zeros_like(w[f"{p}.experts.{unused}.{nm}.weight"])
+    out1 = NumpyModel(cfg, w2)._moe_mlp(0, h)
+    np.testing.assert_allclose(out0, out1)
