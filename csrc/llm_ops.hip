@@ -330,10 +330,16 @@ k_gemv_bf16_t(const u16* __restrict__ W, const u16* __restrict__ x,
               const u16* __restrict__ x2, const float* __restrict__ g,
               const float* __restrict__ g2, void* __restrict__ y,
               const u16* __restrict__ res, int N, int K, int stage, int act,
-              float eps, int out_f32, float softcap, float escale) {
+              float eps, int out_f32, float softcap, float escale,
+              const int* __restrict__ eidx, long wstride,
+              const float* __restrict__ oscale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const u16* xv = gemv_stage(smem, x, x2, g, g2, (u16*)res, K, stage, act,
                              eps, escale);
+  // MoE: expert-indexed weight base + device output scale (router
+  // prob); scalar loads, uniform branch — free on the dense path
+  if (eidx) W += (size_t)(*eidx) * wstride;
+  const float osc = oscale ? *oscale : 1.0f;
   // NORM2 / NORM_EMBED borrow `res` as the persisted-h output
   const u16* eres = (stage == STAGE_NORM2 || stage == STAGE_NORM_EMBED)
                         ? nullptr : res;
@@ -393,7 +399,7 @@ k_gemv_bf16_t(const u16* __restrict__ W, const u16* __restrict__ x,
   }
 #pragma unroll
   for (int r = 0; r < RPW; r++) {
-    float acc = wave_reduce_sum((a0[r] + a1[r]) + (a2[r] + a3[r]));
+    float acc = osc * wave_reduce_sum((a0[r] + a1[r]) + (a2[r] + a3[r]));
     if (lane == 0 && row0 + r < N)
       gemv_epilogue(acc, row0 + r, y, eres, out_f32, softcap);
   }
@@ -407,7 +413,9 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
                                        int stage, int act, float eps,
                                        int out_f32, float softcap,
                                        int nt, int rpw, int maxblocks,
-                                       float escale, hipStream_t stream) {
+                                       float escale, const void* eidx,
+                                       long wstride, const void* oscale,
+                                       hipStream_t stream) {
   size_t lds = (stage == STAGE_RAW) ? 0 : ((size_t)K * 2 + 32);
   static int rpw_env_b = -1;
   if (rpw_env_b < 0) {
@@ -429,7 +437,8 @@ extern "C" hipError_t launch_gemv_bf16(const void* W, const void* x,
                      lds, stream, (const u16*)W, (const u16*)x,              \
                      (const u16*)x2, (const float*)g, (const float*)g2, y,   \
                      (const u16*)res, N, K, stage, act, eps, out_f32,        \
-                     softcap, escale)
+                     softcap, escale, (const int*)eidx, wstride,             \
+                     (const float*)oscale)
   if (nt && rpw == 2) GEMV_CASE(true, 2);
   else if (nt) GEMV_CASE(true, 1);
   else if (rpw == 2) GEMV_CASE(false, 2);
@@ -460,7 +469,8 @@ k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
              const float* __restrict__ g, const float* __restrict__ g2,
              void* __restrict__ y, const u16* __restrict__ res, int N, int K,
              int stage, int act, float eps, int out_f32, float softcap,
-             float escale) {
+             float escale, const int* __restrict__ eidx, long wstride,
+             long sstride, const float* __restrict__ oscale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // XDIR (RAW stage only): x read from global per row iteration — the
   // f32 LDS staging pass costs ~2K B of LDS+global traffic per BLOCK,
@@ -471,6 +481,12 @@ k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
                                           stage, act, eps, escale);
   const u16* eres = (stage == STAGE_NORM2 || stage == STAGE_NORM_EMBED)
                         ? nullptr : res;
+  // MoE: expert-indexed weight/scale base + device output scale
+  if (eidx) {
+    W += (size_t)(*eidx) * wstride;
+    scales += (size_t)(*eidx) * sstride;
+  }
+  const float osc = oscale ? *oscale : 1.0f;
 
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int wpb = blockDim.x >> 6;
@@ -559,7 +575,7 @@ k_gemv_fp8_t(const uint8_t* __restrict__ W, const float* __restrict__ scales,
   for (int r = 0; r < RPW; r++) {
     int rr = row0 + r < N ? row0 + r : N - 1;
     f2v s = a0[r] + a1[r];
-    float acc = wave_reduce_sum(s[0] + s[1]) * scales[rr];
+    float acc = osc * wave_reduce_sum(s[0] + s[1]) * scales[rr];
     if (lane == 0 && row0 + r < N)
       gemv_epilogue(acc, row0 + r, y, eres, out_f32, softcap);
   }
@@ -573,6 +589,8 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
                                       int stage, int act, float eps,
                                       int out_f32, float softcap, int nt,
                                       int rpw, int maxblocks, float escale,
+                                      const void* eidx, long wstride,
+                                      long sstride, const void* oscale,
                                       hipStream_t stream) {
   static int xdir_raw = -1, rows_min = -1, rpw_env = -1;
   if (xdir_raw < 0) {
@@ -626,7 +644,8 @@ extern "C" hipError_t launch_gemv_fp8(const void* W, const void* scales,
                      lds, stream, (const uint8_t*)W, (const float*)scales,  \
                      (const u16*)x, (const u16*)x2, (const float*)g,        \
                      (const float*)g2, y, (const u16*)res, N, K, stage,     \
-                     act, eps, out_f32, softcap, escale)
+                     act, eps, out_f32, softcap, escale, (const int*)eidx,  \
+                     wstride, sstride, (const float*)oscale)
   if (xdir) {
     if (nt && rpw == 4) GEMV8_CASE(true, 4, true);
     else if (nt && rpw == 2) GEMV8_CASE(true, 2, true);
@@ -3349,5 +3368,123 @@ extern "C" hipError_t launch_i32_set(void* p, int v, hipStream_t stream) {
 }
 extern "C" hipError_t launch_i32_add(void* p, int v, hipStream_t stream) {
   hipLaunchKernelGGL(k_i32_add, dim3(1), dim3(1), 0, stream, (int*)p, v);
+  return hipGetLastError();
+}
+
+// ====================================================================
+// Mixtral sparse-MoE helpers (beyond-parity: the reference has no MoE;
+// ROADMAP §5).  k_moe_route fuses the per-row RMSNorm, the router dots
+// against Wg[E,H] (f32 accum — bf16 router logits would risk top-k
+// flips), the softmax over E, and the HF top-k renormalization
+// (modeling_mixtral MixtralTopKRouter), emitting compact (idx, w)
+// pairs for the expert-indexed decode GEMVs and optionally a dense
+// M x E weight grid for the prefill per-expert GEMM loop.
+// One block per row; E <= 64, topk <= 8.
+// ====================================================================
+
+extern "C" __global__ void __launch_bounds__(256)
+k_moe_route(const u16* __restrict__ h, const float* __restrict__ g,
+            const u16* __restrict__ wg, int M, int H, int E, int topk,
+            float eps, int* __restrict__ idx, float* __restrict__ wout,
+            float* __restrict__ dense) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* xf = (float*)smem;           // H floats: normed row
+  float* red = xf + H;                // 4 wave partials
+  float* el = red + 4;                // E router probs
+  const int m = blockIdx.x;
+  const u16* hr = h + (size_t)m * H;
+  float ss = 0.f;
+  for (int i = threadIdx.x; i < H; i += 256) {
+    float v = b2f(hr[i]);
+    xf[i] = v;
+    ss += v * v;
+  }
+#pragma unroll
+  for (int s = 1; s < 64; s <<= 1) ss += __shfl_xor(ss, s);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+  __syncthreads();
+  ss = (red[0] + red[1]) + (red[2] + red[3]);
+  const float rn = __frsqrt_rn(ss / H + eps);
+  for (int i = threadIdx.x; i < H; i += 256) xf[i] = xf[i] * rn * g[i];
+  __syncthreads();
+  for (int e = 0; e < E; e++) {
+    const u16* wr = wg + (size_t)e * H;
+    float d = 0.f;
+    for (int i = threadIdx.x; i < H; i += 256) d += xf[i] * b2f(wr[i]);
+#pragma unroll
+    for (int s = 1; s < 64; s <<= 1) d += __shfl_xor(d, s);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = d;
+    __syncthreads();
+    if (threadIdx.x == 0) el[e] = (red[0] + red[1]) + (red[2] + red[3]);
+    __syncthreads();
+  }
+  if (threadIdx.x != 0) return;
+  float mx = el[0];
+  for (int e = 1; e < E; e++) mx = fmaxf(mx, el[e]);
+  float sum = 0.f;
+  for (int e = 0; e < E; e++) { el[e] = __expf(el[e] - mx); sum += el[e]; }
+  if (dense)
+    for (int e = 0; e < E; e++) dense[(size_t)m * E + e] = 0.f;
+  // top-k by repeated argmax (ties -> smallest index, matching the
+  // oracle's stable argsort); renormalize the kept probs by their sum
+  int sel[8];
+  float sw[8];
+  float wsum = 0.f;
+  for (int j = 0; j < topk; j++) {
+    int be = -1;
+    float bv = -1.f;
+    for (int e = 0; e < E; e++) {
+      bool taken = false;
+      for (int q = 0; q < j; q++) taken |= (sel[q] == e);
+      if (!taken && el[e] > bv) { bv = el[e]; be = e; }
+    }
+    sel[j] = be;
+    sw[j] = bv;
+    wsum += bv;
+  }
+  for (int j = 0; j < topk; j++) {
+    idx[(size_t)m * topk + j] = sel[j];
+    const float wv = sw[j] / wsum;
+    wout[(size_t)m * topk + j] = wv;
+    if (dense) dense[(size_t)m * E + sel[j]] = wv;
+  }
+}
+
+extern "C" hipError_t launch_moe_route(const void* h, const void* g,
+                                       const void* wg, int M, int H, int E,
+                                       int topk, float eps, void* idx,
+                                       void* wout, void* dense,
+                                       hipStream_t stream) {
+  if (E < 1 || E > 64 || topk < 1 || topk > 8 || topk > E)
+    return hipErrorInvalidValue;
+  size_t lds = (size_t)H * 4 + 16 + (size_t)E * 4;
+  if (lds > 64 * 1024) return hipErrorInvalidValue;
+  hipLaunchKernelGGL(k_moe_route, dim3(M), dim3(256), lds, stream,
+                     (const u16*)h, (const float*)g, (const u16*)wg, M, H,
+                     E, topk, eps, (int*)idx, (float*)wout, (float*)dense);
+  return hipGetLastError();
+}
+
+// y[m, :H] += w[m * wstride] * x[m, :H]  (bf16 I/O, f32 math) — the
+// prefill MoE combine: each expert's GEMM output is folded into the
+// residual stream weighted by that row's router prob (0 if unrouted).
+extern "C" __global__ void __launch_bounds__(256)
+k_moe_scale_add(u16* __restrict__ y, const u16* __restrict__ xin,
+                const float* __restrict__ w, long wstride, int M, int H) {
+  const long i = (long)blockIdx.x * 256 + threadIdx.x;
+  if (i >= (long)M * H) return;
+  const float wv = w[(size_t)(i / H) * wstride];
+  if (wv != 0.f) y[i] = f2b(b2f(y[i]) + wv * b2f(xin[i]));
+}
+
+extern "C" hipError_t launch_moe_scale_add(void* y, const void* x,
+                                           const void* w, long wstride,
+                                           int M, int H,
+                                           hipStream_t stream) {
+  long total = (long)M * H;
+  hipLaunchKernelGGL(k_moe_scale_add,
+                     dim3((uint32_t)((total + 255) / 256)), dim3(256), 0,
+                     stream, (u16*)y, (const u16*)x, (const float*)w,
+                     wstride, M, H);
   return hipGetLastError();
 }

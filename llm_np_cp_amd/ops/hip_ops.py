@@ -27,11 +27,13 @@ _SIGS = {
     "launch_gemv_bf16": [ctypes.c_void_p] * 7 + [ctypes.c_int] * 4 +
                         [ctypes.c_float, ctypes.c_int, ctypes.c_float,
                          ctypes.c_int, ctypes.c_int, ctypes.c_int,
-                         ctypes.c_float, ctypes.c_void_p],
+                         ctypes.c_float, ctypes.c_void_p, ctypes.c_long,
+                         ctypes.c_void_p, ctypes.c_void_p],
     "launch_gemv_fp8": [ctypes.c_void_p] * 8 + [ctypes.c_int] * 4 +
                        [ctypes.c_float, ctypes.c_int, ctypes.c_float,
                         ctypes.c_int, ctypes.c_int, ctypes.c_int,
-                        ctypes.c_float, ctypes.c_void_p],
+                        ctypes.c_float, ctypes.c_void_p, ctypes.c_long,
+                        ctypes.c_long, ctypes.c_void_p, ctypes.c_void_p],
     "launch_rmsnorm": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 2 +
                       [ctypes.c_float, ctypes.c_int, ctypes.c_void_p],
     "launch_rope_cache": [ctypes.c_void_p] * 7 + [ctypes.c_int] +
@@ -101,6 +103,11 @@ _SIGS = {
                        [ctypes.c_void_p],
     "launch_gemm_fp4w": [ctypes.c_void_p] * 6 + [ctypes.c_int] * 3 +
                         [ctypes.c_void_p],
+    "launch_moe_route": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 4 +
+                        [ctypes.c_float] + [ctypes.c_void_p] * 4,
+    "launch_moe_scale_add": [ctypes.c_void_p] * 3 +
+                            [ctypes.c_long, ctypes.c_int, ctypes.c_int,
+                             ctypes.c_void_p],
     # one-shot xGMI collectives (csrc/xgmi_comm.hip)
     "launch_xgmi_coll": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                          ctypes.c_int, ctypes.c_int, ctypes.c_long,
@@ -162,7 +169,9 @@ def gemv(W: torch.Tensor, x: torch.Tensor, y: torch.Tensor,
          stage: int = 0, x2: torch.Tensor | None = None,
          g: torch.Tensor | None = None, act: int = 0, eps: float = 1e-5,
          nt: int = 1, rpw: int = 1, maxblocks: int = 0,
-         g2: torch.Tensor | None = None, escale: float = 1.0):
+         g2: torch.Tensor | None = None, escale: float = 1.0,
+         eidx: torch.Tensor | None = None, wstride: int = 0,
+         oscale: torch.Tensor | None = None):
     """y[N] = W[N,K] @ stage(x)[K] (+res); stage fuses RMSNorm / GLU /
     the Gemma sandwich (NORM2: x2 = h_in, g/g2 = post/pre gammas, res =
     h_out ping-pong) / the embed gather (NORM_EMBED: x = embed table,
@@ -174,7 +183,8 @@ def gemv(W: torch.Tensor, x: torch.Tensor, y: torch.Tensor,
         _ptr(W), _ptr(x), _ptr(x2), _ptr(g), _ptr(g2), _ptr(y), _ptr(res),
         N, K, stage, act, ctypes.c_float(eps), out_f32,
         ctypes.c_float(softcap), nt, rpw, maxblocks,
-        ctypes.c_float(escale), _stream()), "gemv")
+        ctypes.c_float(escale), _ptr(eidx), ctypes.c_long(wstride),
+        _ptr(oscale), _stream()), "gemv")
 
 
 def gemv_fp8(Wq: torch.Tensor, scales: torch.Tensor, x: torch.Tensor,
@@ -183,15 +193,20 @@ def gemv_fp8(Wq: torch.Tensor, scales: torch.Tensor, x: torch.Tensor,
              x2: torch.Tensor | None = None, g: torch.Tensor | None = None,
              act: int = 0, eps: float = 1e-5, nt: int = 1, rpw: int = 1,
              maxblocks: int = 0, g2: torch.Tensor | None = None,
-             escale: float = 1.0):
-    """y[N] = scales * (Wq[N,K] @ stage(x)); Wq = e4m3fn bytes."""
+             escale: float = 1.0, eidx: torch.Tensor | None = None,
+             wstride: int = 0, sstride: int = 0,
+             oscale: torch.Tensor | None = None):
+    """y[N] = scales * (Wq[N,K] @ stage(x)); Wq = e4m3fn bytes.
+    eidx/wstride/sstride/oscale: MoE expert indexing — W and scales
+    offset by the device int32 *eidx, output scaled by *oscale."""
     N, K = Wq.shape
     out_f32 = 1 if y.dtype == torch.float32 else 0
     _check(lib().launch_gemv_fp8(
         _ptr(Wq), _ptr(scales), _ptr(x), _ptr(x2), _ptr(g), _ptr(g2),
         _ptr(y), _ptr(res), N, K, stage, act, ctypes.c_float(eps), out_f32,
         ctypes.c_float(softcap), nt, rpw, maxblocks,
-        ctypes.c_float(escale), _stream()), "gemv_fp8")
+        ctypes.c_float(escale), _ptr(eidx), ctypes.c_long(wstride),
+        ctypes.c_long(sstride), _ptr(oscale), _stream()), "gemv_fp8")
 
 
 def rmsnorm(x: torch.Tensor, g: torch.Tensor, y: torch.Tensor,
@@ -577,3 +592,27 @@ def xgmi_coll(dst_ptr: int, src_ptr: int, mybase: int, rank: int,
         ctypes.c_void_p(mybase), rank, world, ctypes.c_long(nbytes),
         ctypes.c_long(slot_bytes), mode, nstripes,
         ctypes.c_long(spin_limit), _stream()), "xgmi_coll")
+
+
+def moe_route(h: torch.Tensor, g: torch.Tensor, wg: torch.Tensor, M: int,
+              topk: int, idx: torch.Tensor, w: torch.Tensor,
+              dense: torch.Tensor | None = None, eps: float = 1e-5):
+    """Fused MoE router (Mixtral semantics): per row RMSNorm(h)*g ->
+    E dots vs wg[E,H] (f32) -> softmax -> top-k renormalized.
+    idx: (M*topk,) int32; w: (M*topk,) f32; dense: (M*E,) f32 per-expert
+    weights (0 if unrouted) for the prefill expert-GEMM loop."""
+    E, H = wg.shape
+    assert idx.numel() >= M * topk and w.numel() >= M * topk
+    if dense is not None:
+        assert dense.numel() >= M * E
+    _check(lib().launch_moe_route(
+        _ptr(h), _ptr(g), _ptr(wg), M, H, E, topk, ctypes.c_float(eps),
+        _ptr(idx), _ptr(w), _ptr(dense), _stream()), "moe_route")
+
+
+def moe_scale_add(y: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
+                  wstride: int, M: int, H: int):
+    """y[m, :H] += w[m * wstride] * x[m, :H] (bf16, f32 math)."""
+    _check(lib().launch_moe_scale_add(
+        _ptr(y), _ptr(x), _ptr(w), ctypes.c_long(wstride), M, H,
+        _stream()), "moe_scale_add")
