@@ -73,6 +73,16 @@ class GPUModel:
             raise ValueError(f"unsupported kv_dtype {kv_dtype!r}")
         self.kv_dtype = kv_dtype
         self.kv8 = kv_dtype == "fp8"  # e4m3 KV pool, per-(head,pos) scales
+        # Mixtral sparse MoE: per-token top-k routed experts (decode:
+        # expert-indexed GEMVs off device idx; prefill: dense per-expert
+        # GEMM loop weighted by router probs)
+        self.moe = config.is_moe
+        self.topk = config.num_experts_per_tok
+        if self.moe:
+            if self.wq4:
+                raise ValueError("MoE + fp4 weights not supported yet")
+            if config.model_type == "gemma2":
+                raise ValueError("MoE is a llama-family (Mixtral) feature")
         self.rank, self.world = tpu.init_distributed()
         # exercise the TP code path on 1 GPU (collectives no-op at
         # world=1, partial sums are then exact) — used by tests
@@ -89,6 +99,9 @@ class GPUModel:
             f"kv heads {cfg.num_key_value_heads} not divisible by tp {tp}"
         assert cfg.intermediate_size % tp == 0
         assert cfg.vocab_size % tp == 0
+        if self.moe and tp > 1:
+            raise ValueError("MoE tensor parallelism (expert parallel) "
+                             "is not implemented yet — run TP=1")
         self.nh_l = cfg.num_attention_heads // tp
         self.kvh_l = cfg.num_key_value_heads // tp
         self.inter_l = cfg.intermediate_size // tp
@@ -212,11 +225,26 @@ class GPUModel:
                 hrows(w[f"{a}.q_proj.weight"]),
                 hrows(w[f"{a}.k_proj.weight"]),
                 hrows(w[f"{a}.v_proj.weight"])], axis=0)
-            gu_np = _np.concatenate([
-                hrows(w[f"{p}.mlp.gate_proj.weight"]),
-                hrows(w[f"{p}.mlp.up_proj.weight"])], axis=0)
+            if self.moe:
+                # stacked per-expert [gate; up] rows (E*2I, H) and down
+                # rows (E*H, I) — ONE device tensor per projection so the
+                # decode GEMV can index expert e as a row offset
+                # (eidx * 2I resp. eidx * H) off a device int32
+                m = f"{p}.block_sparse_moe"
+                E = cfg.num_local_experts
+                gu_np = _np.concatenate(
+                    [x for e in range(E)
+                     for x in (w[f"{m}.experts.{e}.w1.weight"],
+                               w[f"{m}.experts.{e}.w3.weight"])], axis=0)
+                down_np = _np.concatenate(
+                    [w[f"{m}.experts.{e}.w2.weight"] for e in range(E)],
+                    axis=0)
+            else:
+                gu_np = _np.concatenate([
+                    hrows(w[f"{p}.mlp.gate_proj.weight"]),
+                    hrows(w[f"{p}.mlp.up_proj.weight"])], axis=0)
+                down_np = hcols(w[f"{p}.mlp.down_proj.weight"])
             o_np = hcols(w[f"{a}.o_proj.weight"])
-            down_np = hcols(w[f"{p}.mlp.down_proj.weight"])
             nq = self.nh_l * hd
             nkv = self.kvh_l * hd
             I = self.inter_l
@@ -270,6 +298,26 @@ class GPUModel:
                     "wo": bf16(o_np),
                     "wdown": bf16(down_np),
                 })
+            if self.moe:
+                # router weights + decode-shaped views: the GEMV sees the
+                # FIRST expert's rows (N=2I / N=H) and offsets the base
+                # pointer by eidx * wstride device-side
+                lw["wg"] = bf16(w[f"{p}.block_sparse_moe.gate.weight"])
+                I2, H2 = 2 * self.inter_l, self.H
+                if self.fp8:
+                    lw["wgu_q_all"] = lw.pop("wgu_q")
+                    lw["wgu_s_all"] = lw.pop("wgu_s")
+                    lw["wdown_q_all"] = lw.pop("wdown_q")
+                    lw["wdown_s_all"] = lw.pop("wdown_s")
+                    lw["wgu_q"] = lw["wgu_q_all"][:I2]
+                    lw["wgu_s"] = lw["wgu_s_all"][:I2]
+                    lw["wdown_q"] = lw["wdown_q_all"][:H2]
+                    lw["wdown_s"] = lw["wdown_s_all"][:H2]
+                else:
+                    lw["wgu_all"] = lw.pop("wgu")
+                    lw["wdown_all"] = lw.pop("wdown")
+                    lw["wgu"] = lw["wgu_all"][:I2]
+                    lw["wdown"] = lw["wdown_all"][:H2]
             if gemma:
                 lw["g_preffn"] = gamma(w[f"{p}.pre_feedforward_layernorm.weight"])
                 lw["g_postffn"] = gamma(w[f"{p}.post_feedforward_layernorm.weight"])
@@ -340,6 +388,14 @@ class GPUModel:
                          torch.zeros(cfg.vocab_size, dtype=torch.float32,
                                      device=dev))
         self.b_qkv = torch.zeros((self.nh_l + 2 * self.kvh_l) * hd, **bf)
+        if self.moe:
+            E = cfg.num_local_experts
+            self.moe_idx = torch.zeros(PC * self.topk, dtype=torch.int32,
+                                       device=dev)
+            self.moe_w = torch.zeros(PC * self.topk, dtype=torch.float32,
+                                     device=dev)
+            self.moe_dense = torch.zeros(PC * E, dtype=torch.float32,
+                                         device=dev)
         self.b_hb = torch.zeros(H, **bf)   # gemma residual ping-pong
         self.b_t2 = torch.zeros(H, **bf)   # gemma ffn-out delta
         self.b_gu = torch.zeros(2 * I, **bf)
@@ -559,6 +615,11 @@ class GPUModel:
                 else:
                     self._linear(lw, "wo", self.b_att, h, res=h, M=M)
                 ho.rmsnorm(h[:M], lw["g_post"], xn[:M], eps=eps)
+            if self.moe:
+                self._moe_mlp_rows(lw, M)
+                if layer_hook is not None:
+                    layer_hook(i, h[:M])
+                continue
             self._linear(lw, "wgate", xn, self.b_gate, M=M)
             self._linear(lw, "wup", xn, self.b_up, M=M)
             ho.glu(self.b_gate[:M], self.b_up[:M], self.b_gate[:M], self.act)
@@ -573,6 +634,50 @@ class GPUModel:
                     self._linear(lw, "wdown", self.b_gate, h, res=h, M=M)
             if layer_hook is not None:
                 layer_hook(i, h[:M])
+
+    def _moe_mlp_rows(self, lw, M: int):
+        """Prefill/batch MoE MLP over rows [0, M): router weights per
+        row (dense M x E grid), then ONE GEMM pass per expert with the
+        output folded into h weighted by that row's prob (0 if
+        unrouted).  Work is E/topk x the routed minimum — prefill is
+        compute-rich and this keeps the MFMA GEMM path; decode uses the
+        expert-indexed GEMVs instead (`_decode_step`)."""
+        cfg = self.config
+        E, I, H = cfg.num_local_experts, self.inter_l, self.H
+        eps = cfg.rms_norm_eps
+        xn, gate, up, t1 = self.b_xn, self.b_gate, self.b_up, self.b_t1
+        ho.moe_route(self.b_h, lw["g_post"], lw["wg"], M, self.topk,
+                     self.moe_idx, self.moe_w, dense=self.moe_dense,
+                     eps=eps)
+        for e in range(E):
+            g_lo, g_hi = e * 2 * I, e * 2 * I + I
+            u_hi = (e + 1) * 2 * I
+            d_lo, d_hi = e * H, (e + 1) * H
+            if self.fp8:
+                ho.quant_fp8(xn[:M], self.b_xq, self.b_sx)
+                ho.gemm_fp8(self.b_xq, self.b_sx,
+                            lw["wgu_q_all"][g_lo:g_hi],
+                            lw["wgu_s_all"][g_lo:g_hi], gate[:M], M, H,
+                            accbuf=self.b_gemm_acc)
+                ho.gemm_fp8(self.b_xq, self.b_sx,
+                            lw["wgu_q_all"][g_hi:u_hi],
+                            lw["wgu_s_all"][g_hi:u_hi], up[:M], M, H,
+                            accbuf=self.b_gemm_acc)
+                ho.glu(gate[:M], up[:M], gate[:M], self.act)
+                ho.quant_fp8(gate[:M], self.b_xq, self.b_sx)
+                ho.gemm_fp8(self.b_xq, self.b_sx,
+                            lw["wdown_q_all"][d_lo:d_hi],
+                            lw["wdown_s_all"][d_lo:d_hi], t1[:M], M, I,
+                            accbuf=self.b_gemm_acc)
+            else:
+                ho.gemm(xn[:M], lw["wgu_all"][g_lo:g_hi], gate[:M],
+                        accbuf=self.b_gemm_acc)
+                ho.gemm(xn[:M], lw["wgu_all"][g_hi:u_hi], up[:M],
+                        accbuf=self.b_gemm_acc)
+                ho.glu(gate[:M], up[:M], gate[:M], self.act)
+                ho.gemm(gate[:M], lw["wdown_all"][d_lo:d_hi], t1[:M],
+                        accbuf=self.b_gemm_acc)
+            ho.moe_scale_add(self.b_h, t1, self.moe_dense[e:], E, M, H)
 
     def _lm_head_last(self, M: int):
         """Final norm + lm_head on the last row -> self.b_logits (f32, V);
@@ -913,6 +1018,38 @@ class GPUModel:
                     ho.addinto(h, t1)
                 else:
                     self._dgemv(lw, "wo", self.b_att[0], h, res=h)
+                if self.moe:
+                    # router -> topk expert-indexed GEMV pairs: W base
+                    # offset by the device idx, output scaled by the
+                    # device router prob, accumulated into h via res.
+                    # xn is staged ONCE: expert 0's down-proj already
+                    # updates h, so expert 1 must not re-norm it
+                    ho.moe_route(h, lw["g_post"], lw["wg"], 1, self.topk,
+                                 self.moe_idx, self.moe_w, eps=eps)
+                    xn0 = self.b_xn[0]
+                    ho.rmsnorm(h, lw["g_post"], xn0, eps=eps)
+                    I, H = self.inter_l, self.H
+                    for j in range(self.topk):
+                        ei = self.moe_idx[j:j + 1]
+                        osc = self.moe_w[j:j + 1]
+                        if self.fp8:
+                            ho.gemv_fp8(lw["wgu_q"], lw["wgu_s"], xn0,
+                                        self.b_gu, eidx=ei,
+                                        wstride=2 * I * H, sstride=2 * I)
+                            ho.gemv_fp8(lw["wdown_q"], lw["wdown_s"],
+                                        self.b_gu[:I], h, res=h,
+                                        stage=ho.STAGE_GLU,
+                                        x2=self.b_gu[I:], act=self.act,
+                                        eidx=ei, wstride=H * I,
+                                        sstride=H, oscale=osc)
+                        else:
+                            ho.gemv(lw["wgu"], xn0, self.b_gu, eidx=ei,
+                                    wstride=2 * I * H)
+                            ho.gemv(lw["wdown"], self.b_gu[:I], h, res=h,
+                                    stage=ho.STAGE_GLU, x2=self.b_gu[I:],
+                                    act=self.act, eidx=ei,
+                                    wstride=H * I, oscale=osc)
+                    continue
                 self._dgemv(lw, "wgu", h, self.b_gu, stage=ho.STAGE_NORM,
                             g=lw["g_post"], eps=eps)
                 if self.tp_branch:
@@ -982,9 +1119,10 @@ class GPUModel:
         # available via LLM_BATCH_MX_MAX for comparison.
         import os as _os  # noqa: delayed so env changes apply per call
         mx_max = int(_os.environ.get("LLM_BATCH_MX_MAX", "0"))
-        if self.fp8 and B <= mx_max and not cfg.attention_bias:
+        if (self.fp8 and B <= mx_max and not cfg.attention_bias
+                and not self.moe):
             return self._decode_batch_step_mx(B, greedy, min_p, temperature)
-        if self.fp8 and B <= 16:
+        if self.fp8 and B <= 16 and not self.moe:
             return self._decode_batch_step_skinny(B, greedy, min_p,
                                                   temperature)
         ho.embed(self.embed, self.bt_next, self.b_h, B, cfg.embed_scale)
