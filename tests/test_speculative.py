@@ -92,3 +92,16 @@ def test_spec_k_validation():
     _, draft, _ = _load(seed=0)
     with pytest.raises(ValueError):
         generate_speculative(PROMPT, tok, draft, target, k=0)
+
+
+def test_spec_with_moe_target():
+    """Speculative decoding composes with the Mixtral MoE path (the
+    verify pass runs the per-row routed forward)."""
+    tok, target, _ = L.load_model("tiny-mixtral", backend="numpy", seed=0)
+    _, draft, _ = L.load_model("tiny-mixtral", backend="numpy", seed=2)
+    ref = L.generate(PROMPT, tok, target, max_tokens=16, stream=False,
+                     params=L.SamplingParams(strategy="greedy"),
+                     stop_on_eos=False).token_ids
+    res = generate_speculative(PROMPT, tok, draft, target, max_tokens=16,
+                               k=4, stop_on_eos=False)
+    assert res.token_ids == ref
