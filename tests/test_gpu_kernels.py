@@ -656,3 +656,53 @@ def test_gemv_fp4_fused_norm_stage():
     xn = xf * torch.rsqrt(xf.pow(2).mean() + eps) * g
     ref = torch.from_numpy(Wn).to(dev()) @ xn
     assert_close(y, ref, rtol=3e-2, atol=3e-2)
+
+
+def test_moe_route_matches_numpy():
+    """k_moe_route (fused rmsnorm + router dots + softmax + top-k
+    renorm) vs a torch fp32 replica of the HF Mixtral router."""
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    M, H, E, topk = 5, 256, 8, 2
+    h = randn_bf16(M, H, seed=20)
+    g = (0.9 + 0.2 * torch.rand(H, generator=torch.Generator()
+                                .manual_seed(21))).to(dev())
+    wg = randn_bf16(E, H, seed=22, scale=0.5)
+    idx = torch.zeros(M * topk, dtype=torch.int32, device=dev())
+    w = torch.zeros(M * topk, dtype=torch.float32, device=dev())
+    dense = torch.zeros(M * E, dtype=torch.float32, device=dev())
+    ho.moe_route(h, g, wg, M, topk, idx, w, dense=dense, eps=1e-5)
+    torch.cuda.synchronize()
+
+    hf = h.float()
+    xn = hf * torch.rsqrt(hf.pow(2).mean(-1, keepdim=True) + 1e-5) * g
+    logits = xn @ wg.float().T
+    probs = torch.softmax(logits, dim=-1)
+    top_v, top_i = torch.topk(probs, topk, dim=-1)
+    top_v = top_v / top_v.sum(-1, keepdim=True)
+
+    assert torch.equal(idx.view(M, topk).cpu(), top_i.int().cpu())
+    assert_close(w.view(M, topk), top_v, rtol=2e-2, atol=2e-3)
+    d = dense.view(M, E).cpu()
+    for m in range(M):
+        for e in range(E):
+            want = 0.0
+            for j in range(topk):
+                if int(top_i[m, j]) == e:
+                    want = float(top_v[m, j])
+            assert abs(float(d[m, e]) - want) < 2e-2
+
+
+def test_moe_scale_add_matches_torch():
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    M, H, E = 4, 192, 6
+    y = randn_bf16(M, H, seed=30)
+    x = randn_bf16(M, H, seed=31)
+    wd = torch.rand(M * E, generator=torch.Generator().manual_seed(32)
+                    ).to(dev())
+    e = 3
+    ref = y.float() + wd.view(M, E)[:, e:e + 1] * x.float()
+    ho.moe_scale_add(y, x, wd[e:], E, M, H)
+    torch.cuda.synchronize()
+    assert_close(y, ref.to(torch.bfloat16))
