@@ -1393,6 +1393,8 @@ k_rmsnorm(const u16* __restrict__ x, const float* __restrict__ g,
 extern "C" hipError_t launch_rmsnorm(const void* x, const void* g,
                                      const void* res, void* y, int M, int H,
                                      float eps, int mode, hipStream_t stream) {
+  // the kernel keeps each row in registers: 2 chunks x 1024 threads x 8
+  if (H > 16384 || H % 8 != 0) return hipErrorInvalidValue;
   hipLaunchKernelGGL(k_rmsnorm, dim3(M), dim3(1024), 0, stream,
                      (const u16*)x, (const float*)g, (const u16*)res, (u16*)y,
                      H, eps, mode);
