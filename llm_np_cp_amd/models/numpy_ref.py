@@ -1,4 +1,5 @@
-"""Pure-NumPy reference implementation of the Llama-3.2 / Gemma-2 forward pass.
+"""Pure-NumPy reference implementation of the decoder-LM forward pass
+(Llama-3.2 / Gemma-2 / Qwen-2 / Mistral / Mixtral sparse-MoE).
 
 This is the in-repo oracle (the role HF ``transformers`` played for the
 reference) and BASELINE config 0 (the reference's NumPy path,
@@ -234,8 +235,9 @@ class NumpyModel:
     def forward_hf(self, input_ids: np.ndarray, cache: NumpyKVCache = None):
         """Reference-parity output tuple ``(loss, logits, kv_cache,
         hidden_states, attentions)`` (llama3.2_model.py:726-822).
-        ``attentions`` is None by design: attention here is computed with
-        online softmax and the probability matrix never materializes."""
+        ``attentions`` is None, mirroring the GPU engine's surface
+        (its fused online-softmax kernels never materialize the
+        probability matrix; the oracle keeps the same tuple shape)."""
         cache = cache or NumpyKVCache(self.config,
                                       len(np.ravel(input_ids)) + 1)
         hidden = []
