@@ -1,0 +1,85 @@
+"""Edge-case unit tests for config, sampling, and wrapper validation —
+CPU-only, no kernels launched (failing validation paths raise before
+any HIP call)."""
+
+import numpy as np
+import pytest
+
+from llm_np_cp_amd.core.config import ModelConfig, preset_config
+from llm_np_cp_amd.runtime.generate import GenerateResult
+from llm_np_cp_amd.runtime.sampling import SamplingParams, sample_token
+
+
+def test_is_moe_and_expert_fields():
+    mx = preset_config("tiny-mixtral")
+    assert mx.is_moe and mx.num_local_experts == 4
+    assert mx.num_experts_per_tok == 2
+    assert not preset_config("tiny-llama").is_moe
+    d = mx.to_hf_dict()
+    assert d["num_local_experts"] == 4
+    back = ModelConfig.from_hf_dict(d)
+    assert back.is_moe and back.num_experts_per_tok == 2
+    # dense configs do not leak MoE keys into their HF dict
+    assert "num_local_experts" not in preset_config("tiny-llama").to_hf_dict()
+
+
+def test_gemma_attn_scale_uses_query_pre_attn_scalar():
+    g = preset_config("gemma-2-27b")
+    assert g.attn_scale == pytest.approx(144 ** -0.5)
+    assert g.head_dim ** -0.5 != pytest.approx(g.attn_scale)
+    ll = preset_config("tiny-llama")
+    assert ll.attn_scale == pytest.approx(ll.head_dim ** -0.5)
+
+
+def test_min_p_one_keeps_only_max():
+    logits = np.array([0.0, 1.0, 5.0, 4.9999], dtype=np.float32)
+    p = SamplingParams(strategy="min_p", min_p=1.0, seed=0)
+    picks = {sample_token(logits, p, np.random.default_rng(i))
+             for i in range(20)}
+    # min_p=1.0 keeps p >= p_max: the max and anything tied with it
+    assert picks <= {2, 3}
+
+
+def test_temperature_strategy_sharpens():
+    logits = np.array([0.0, 1.0, 2.0], dtype=np.float32)
+    cold = SamplingParams(strategy="temperature", temperature=0.05)
+    rng = np.random.default_rng(0)
+    picks = [sample_token(logits, cold, rng) for _ in range(50)]
+    assert picks.count(2) >= 48  # near-greedy at T->0
+
+
+def test_top_p_keeps_cumulative_mass():
+    logits = np.log(np.array([0.5, 0.3, 0.15, 0.05], dtype=np.float32))
+    p = SamplingParams(strategy="top_p", top_p=0.6, seed=1)
+    picks = {sample_token(logits, p, np.random.default_rng(i))
+             for i in range(50)}
+    assert picks <= {0, 1}  # 0.5 + 0.3 crosses 0.6; 2/3 masked
+
+
+def test_unknown_strategy_raises():
+    with pytest.raises(ValueError):
+        sample_token(np.zeros(4, dtype=np.float32),
+                     SamplingParams(strategy="beam"))
+
+
+def test_generate_result_zero_decode_time():
+    r = GenerateResult(text="", token_ids=[1, 2], decode_time_s=0.0)
+    assert r.decode_tokens_per_s == 0.0
+
+
+def test_hip_wrapper_shape_validation_raises_before_launch():
+    """Size-mismatch asserts fire on CPU before any HIP call."""
+    import torch
+    from llm_np_cp_amd.ops import hip_ops as ho
+
+    x = torch.zeros(4, 32, dtype=torch.bfloat16)
+    q_small = torch.zeros(10, dtype=torch.uint8)
+    s = torch.zeros(4)
+    with pytest.raises(AssertionError):
+        ho.quant_fp8(x, q_small, s)
+    with pytest.raises(AssertionError):
+        ho.quant_fp4(x, q_small, torch.zeros(1, dtype=torch.uint8))
+    wg = torch.zeros(4, 32, dtype=torch.bfloat16)
+    with pytest.raises(AssertionError):
+        ho.moe_route(x, s, wg, 4, 2, torch.zeros(1, dtype=torch.int32),
+                     torch.zeros(8))
