@@ -33,13 +33,16 @@ def _softmax(x: np.ndarray) -> np.ndarray:
     return e / e.sum()
 
 
-def sample_token(logits: np.ndarray, params: SamplingParams,
-                 rng: Optional[np.random.Generator] = None) -> int:
-    """logits: (vocab,) fp32. Returns a token id (int)."""
+def filter_probs(logits: np.ndarray, params: SamplingParams) -> np.ndarray:
+    """The sampling DISTRIBUTION the strategy defines: temperature-scaled
+    softmax with the strategy's support filter applied, renormalized.
+    (fp64, sums to 1.)  ``sample_token`` draws from this; speculative
+    sampling (runtime/speculative.py) needs the vector itself to compute
+    accept ratios p(x)/q(x) and the residual distribution."""
     if params.strategy == "greedy":
-        return int(np.argmax(logits))
-    if rng is None:
-        rng = np.random.default_rng(params.seed)
+        p = np.zeros(len(logits), dtype=np.float64)
+        p[int(np.argmax(logits))] = 1.0
+        return p
     if params.temperature != 1.0:
         logits = logits / max(params.temperature, 1e-6)
     probs = _softmax(logits)
@@ -61,5 +64,15 @@ def sample_token(logits: np.ndarray, params: SamplingParams,
         probs = np.where(mask, probs, 0.0)
     elif params.strategy != "temperature":
         raise ValueError(f"unknown sampling strategy {params.strategy!r}")
-    probs = probs / probs.sum()
+    return probs / probs.sum()
+
+
+def sample_token(logits: np.ndarray, params: SamplingParams,
+                 rng: Optional[np.random.Generator] = None) -> int:
+    """logits: (vocab,) fp32. Returns a token id (int)."""
+    if params.strategy == "greedy":
+        return int(np.argmax(logits))
+    if rng is None:
+        rng = np.random.default_rng(params.seed)
+    probs = filter_probs(logits, params)
     return int(rng.choice(len(probs), p=probs))

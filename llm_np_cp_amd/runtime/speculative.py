@@ -1,11 +1,26 @@
-"""Greedy speculative decoding: a small DRAFT model proposes k tokens,
-the TARGET verifies them in ONE prefill-shaped pass and accepts the
-longest matching prefix, emitting m accepted drafts plus one target
-token per verify pass (1..k+1 tokens per target forward).
+"""Speculative decoding: a small DRAFT model proposes k tokens, the
+TARGET verifies them in ONE prefill-shaped pass, emitting the accepted
+drafts plus one target token per verify pass (1..k+1 tokens per target
+forward).
 
-The output is IDENTICAL to the target's own greedy decode under the
-verify pass's numerics: every emitted token is an argmax of target
-logits (accepted drafts matched it; the correction/bonus token IS it).
+Two modes through ONE accept/reject machine (``spec_accept``):
+
+- **Stochastic** (``params`` with a sampling strategy): the standard
+  speculative-sampling rule — draft token x ~ q is accepted with
+  probability min(1, p(x)/q(x)); on the first rejection the replacement
+  is drawn from the residual distribution norm(max(p - q, 0)); if all k
+  survive, a bonus token is drawn from the target's next-position p.
+  Every emitted token is distributed EXACTLY as the target's own
+  (filtered) sampling distribution p — the draft changes speed, never
+  statistics (chi-square-verified in tests/test_speculative.py).
+  p and q are both warped by the SAME SamplingParams
+  (temperature + min-p/top-k/top-p filter, ``sampling.filter_probs``).
+
+- **Greedy** (default / ``strategy="greedy"``): the same machine with
+  one-hot p and q degenerates to longest-matching-prefix acceptance,
+  so the output is IDENTICAL to the target's own greedy decode: every
+  emitted token is an argmax of target logits (accepted drafts matched
+  it; the correction/bonus token IS it).
 
 MI355X-native fit: both engines keep preallocated KV pools whose live
 length is a device scalar, so rejecting draft tokens is an O(1) length
@@ -29,6 +44,42 @@ from typing import List, Optional
 import numpy as np
 
 from .generate import GenerateResult
+from .sampling import SamplingParams, filter_probs
+
+
+def spec_accept(target_probs: np.ndarray, draft_probs: np.ndarray,
+                drafts: List[int], rng: np.random.Generator):
+    """One verify pass of the speculative accept/reject rule.
+
+    ``target_probs``: (k+1, V) filtered target distributions — row i is
+    p_i, the target's distribution for draft slot i; the last row is
+    the bonus-position distribution.  ``draft_probs``: (k, V) filtered
+    draft distributions q_i.  ``drafts``: the k proposed ids, with
+    drafts[i] drawn from q_i.
+
+    Returns ``(m, next_token)``: the first m drafts are accepted and
+    ``next_token`` is drawn from the residual norm(max(p_m - q_m, 0))
+    (m < k, rejection) or from the bonus row (m == k).  Marginally each
+    emitted token ~ p exactly (Leviathan/Chen speculative sampling);
+    with one-hot p/q (greedy) this reduces to longest-matching-prefix
+    + argmax correction, bit-equal to target-only greedy decode.
+    """
+    k = len(drafts)
+    for i in range(k):
+        x = int(drafts[i])
+        p, q = target_probs[i], draft_probs[i]
+        qx = float(q[x])
+        px = float(p[x])
+        # accept with prob min(1, p(x)/q(x)); qx > 0 because x ~ q
+        if qx > 0.0 and rng.random() * qx < px:
+            continue
+        resid = np.maximum(p - q, 0.0)
+        s = float(resid.sum())
+        if s <= 0.0:          # p <= q everywhere at fp precision ⇒ p == q
+            resid, s = p, float(p.sum())
+        return i, int(rng.choice(len(resid), p=resid / s))
+    p = target_probs[k]
+    return k, int(rng.choice(len(p), p=p / float(p.sum())))
 
 
 def _cap(model, cache) -> int:
@@ -56,8 +107,14 @@ def _rewind(model, cache, n: int) -> None:
 def generate_speculative(prompt: str, tokenizer, draft, target,
                          max_tokens: int = 200, k: int = 4,
                          stop_on_eos: bool = True,
+                         params: Optional[SamplingParams] = None,
                          on_token=None) -> GenerateResult:
-    """Greedy decode of ``target`` accelerated by ``draft`` proposals.
+    """Decode ``target`` accelerated by ``draft`` proposals.
+
+    ``params=None`` (or strategy "greedy") reproduces target-only
+    greedy decode token-identically; any other SamplingParams runs
+    stochastic speculative sampling whose per-token distribution equals
+    target-only sampling under the same params (see module docstring).
 
     ``draft`` and ``target`` must share the tokenizer/vocab (standard
     speculative-decoding requirement).  Returns a GenerateResult; the
@@ -66,6 +123,15 @@ def generate_speculative(prompt: str, tokenizer, draft, target,
     """
     if k < 1:
         raise ValueError("k must be >= 1")
+    if params is None:
+        params = SamplingParams(strategy="greedy")
+    rng = np.random.default_rng(params.seed)
+
+    def _pick(logits_row: np.ndarray) -> int:
+        if params.strategy == "greedy":
+            return int(np.argmax(logits_row))
+        p = filter_probs(logits_row, params)
+        return int(rng.choice(len(p), p=p))
     prompt_ids = [int(t) for t in tokenizer.encode(prompt)]
     P0 = len(prompt_ids)
     eos = getattr(target.config, "eos_token_id", None)
@@ -85,7 +151,7 @@ def generate_speculative(prompt: str, tokenizer, draft, target,
 
     t0 = time.perf_counter()
     vl = _all_logits(target, t_cache, prompt_ids, 0)
-    pending = int(np.argmax(vl[-1]))      # first output token
+    pending = _pick(vl[-1])               # first output token
     _all_logits(draft, d_cache, prompt_ids, 0)
     t_prefill = time.perf_counter() - t0
 
@@ -103,7 +169,7 @@ def generate_speculative(prompt: str, tokenizer, draft, target,
         if kk < 1:
             if n + 1 < t_max:             # no draft room: plain step
                 vl = _all_logits(target, t_cache, [pending], n)
-                pending = int(np.argmax(vl[-1]))
+                pending = _pick(vl[-1])
                 out.append(pending)
                 all_tokens.append(pending)
                 n += 1
@@ -115,17 +181,19 @@ def generate_speculative(prompt: str, tokenizer, draft, target,
         # cache up on tokens it skipped when a verify pass over-ran it)
         dl = _all_logits(draft, d_cache, all_tokens[d_len:], d_len)
         d_len = len(all_tokens)
-        drafts = [int(np.argmax(dl[-1]))]
+        q_rows = [filter_probs(dl[-1], params)]
+        drafts = [int(rng.choice(len(q_rows[0]), p=q_rows[0]))]
         for _ in range(kk - 1):
             dl = _all_logits(draft, d_cache, [drafts[-1]], d_len)
             d_len += 1
-            drafts.append(int(np.argmax(dl[-1])))
-        # target verifies pending + kk drafts in one pass
+            q_rows.append(filter_probs(dl[-1], params))
+            drafts.append(int(rng.choice(len(q_rows[-1]), p=q_rows[-1])))
+        # target verifies pending + kk drafts in one pass; rows 0..kk of
+        # vl are the target distributions at each draft slot + bonus
         vl = _all_logits(target, t_cache, [pending] + drafts, n)
-        m = 0
-        while m < kk and int(np.argmax(vl[m])) == drafts[m]:
-            m += 1
-        pending = int(np.argmax(vl[m]))   # correction (m<kk) or bonus
+        p_rows = np.stack([filter_probs(vl[i], params)
+                           for i in range(kk + 1)])
+        m, pending = spec_accept(p_rows, np.stack(q_rows), drafts, rng)
         emitted = drafts[:m] + [pending]
         out.extend(emitted)
         all_tokens.extend(emitted)

@@ -105,3 +105,116 @@ def test_spec_with_moe_target():
     res = generate_speculative(PROMPT, tok, draft, target, max_tokens=16,
                                k=4, stop_on_eos=False)
     assert res.token_ids == ref
+
+
+# ---------- stochastic speculative sampling (accept/reject machine) ----------
+
+def _rand_dist(rng, V, conc=0.5):
+    p = rng.dirichlet(np.full(V, conc))
+    return p / p.sum()
+
+
+def test_spec_accept_marginal_matches_target_chi_square():
+    """The theorem the whole mode rests on: with draft tokens ~ q pushed
+    through accept/reject + residual, the emitted slot-0 token is
+    distributed EXACTLY as p — whatever q is.  Chi-square over 20k
+    single-draft trials against an adversarially different q."""
+    from scipy.stats import chi2
+    from llm_np_cp_amd.runtime.speculative import spec_accept
+
+    V, n = 8, 20000
+    rng = np.random.default_rng(11)
+    p = _rand_dist(rng, V)
+    q = _rand_dist(rng, V)          # independent => very unlike p
+    bonus = _rand_dist(rng, V)
+    tp = np.stack([p, bonus])
+    counts = np.zeros(V)
+    accepted = 0
+    for _ in range(n):
+        d = [int(rng.choice(V, p=q))]
+        m, tok = spec_accept(tp, q[None, :], d, rng)
+        counts[d[0] if m == 1 else tok] += 1
+        accepted += m
+    exp = p * n
+    mask = exp > 5
+    stat = (((counts - exp) ** 2) / exp)[mask].sum()
+    thresh = chi2.ppf(0.999, int(mask.sum()) - 1)
+    assert stat < thresh, (stat, thresh, counts, exp)
+    # sanity: the acceptance rate ~ sum(min(p, q)) (Leviathan Thm. 3.5)
+    a_exp = np.minimum(p, q).sum()
+    assert abs(accepted / n - a_exp) < 0.02
+
+
+def test_spec_accept_identical_models_accept_everything():
+    """p == q: accept probability min(1, p/q) = 1 at every slot."""
+    from llm_np_cp_amd.runtime.speculative import spec_accept
+
+    V, k = 16, 6
+    rng = np.random.default_rng(5)
+    rows = np.stack([_rand_dist(rng, V) for _ in range(k + 1)])
+    for _ in range(50):
+        drafts = [int(rng.choice(V, p=rows[i])) for i in range(k)]
+        m, tok = spec_accept(rows, rows[:k], drafts, rng)
+        assert m == k
+        assert 0 <= tok < V
+
+
+def test_spec_accept_one_hot_degenerates_to_greedy_prefix():
+    """One-hot p/q (greedy mode) == longest-matching-prefix + argmax
+    correction, with no dependence on the rng draws."""
+    from llm_np_cp_amd.runtime.speculative import spec_accept
+
+    V = 8
+
+    def onehot(i):
+        v = np.zeros(V); v[i] = 1.0
+        return v
+
+    p_ids, q_ids = [3, 5, 2, 6], [3, 5, 4]   # mismatch at slot 2
+    tp = np.stack([onehot(i) for i in p_ids])
+    qp = np.stack([onehot(i) for i in q_ids])
+    for seed in range(5):
+        rng = np.random.default_rng(seed)
+        m, tok = spec_accept(tp, qp, q_ids, rng)
+        assert (m, tok) == (2, 2)            # reject at 2, emit argmax(p_2)
+    # full match -> bonus row argmax
+    qp2 = np.stack([onehot(i) for i in p_ids[:3]])
+    m, tok = spec_accept(tp, qp2, p_ids[:3], np.random.default_rng(0))
+    assert (m, tok) == (3, 6)
+
+
+def test_spec_stochastic_end_to_end_seeded():
+    """min-p speculative runs end to end, is seed-reproducible, and
+    matching seeds give matching outputs across draft choices of k."""
+    tok, target, _ = _load(seed=0)
+    _, draft, _ = _load(seed=7)
+    params = L.SamplingParams(strategy="min_p", min_p=0.05,
+                              temperature=0.9, seed=123)
+    a = generate_speculative(PROMPT, tok, draft, target, max_tokens=20,
+                             k=4, stop_on_eos=False, params=params)
+    b = generate_speculative(PROMPT, tok, draft, target, max_tokens=20,
+                             k=4, stop_on_eos=False, params=params)
+    assert a.token_ids == b.token_ids
+    assert len(a.token_ids) == 20
+    s = a.spec_stats
+    assert s["verify_passes"] >= 1 and 0 <= s["accepted"] <= s["proposed"]
+    c = generate_speculative(PROMPT, tok, draft, target, max_tokens=20,
+                             k=4, stop_on_eos=False,
+                             params=L.SamplingParams(strategy="min_p",
+                                                     min_p=0.05,
+                                                     temperature=0.9,
+                                                     seed=999))
+    assert isinstance(c.token_ids, list) and len(c.token_ids) == 20
+
+
+def test_spec_stochastic_same_models_high_acceptance():
+    """Draft == target under stochastic sampling: accept prob is
+    min(1, p/x) = 1 every slot (same weights => p == q)."""
+    tok, target, _ = _load(seed=0)
+    _, draft, _ = _load(seed=0)
+    params = L.SamplingParams(strategy="temperature", temperature=1.0,
+                              seed=42)
+    res = generate_speculative(PROMPT, tok, draft, target, max_tokens=16,
+                               k=4, stop_on_eos=False, params=params)
+    s = res.spec_stats
+    assert s["accepted"] == s["proposed"] > 0
