@@ -55,6 +55,44 @@ def hf_gemma2(cfg):
     return Gemma2ForCausalLM(hf_cfg).eval()
 
 
+def hf_qwen2(cfg):
+    from transformers import Qwen2Config, Qwen2ForCausalLM
+
+    hf_cfg = Qwen2Config(
+        vocab_size=cfg.vocab_size, hidden_size=cfg.hidden_size,
+        intermediate_size=cfg.intermediate_size,
+        num_hidden_layers=cfg.num_hidden_layers,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        rms_norm_eps=cfg.rms_norm_eps, rope_theta=cfg.rope_theta,
+        max_position_embeddings=cfg.max_position_embeddings,
+        tie_word_embeddings=cfg.tie_word_embeddings,
+        attn_implementation="eager",
+    )
+    torch.manual_seed(3)
+    return Qwen2ForCausalLM(hf_cfg).eval()   # qkv biases ARE random-init
+
+
+def hf_mistral(cfg):
+    from transformers import MistralConfig, MistralForCausalLM
+
+    hf_cfg = MistralConfig(
+        vocab_size=cfg.vocab_size, hidden_size=cfg.hidden_size,
+        intermediate_size=cfg.intermediate_size,
+        num_hidden_layers=cfg.num_hidden_layers,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        head_dim=cfg.head_dim, rms_norm_eps=cfg.rms_norm_eps,
+        rope_theta=cfg.rope_theta,
+        max_position_embeddings=cfg.max_position_embeddings,
+        tie_word_embeddings=cfg.tie_word_embeddings,
+        sliding_window=cfg.sliding_window,   # 8 < test seq: window active
+        attn_implementation="eager",
+    )
+    torch.manual_seed(5)
+    return MistralForCausalLM(hf_cfg).eval()
+
+
 def np_weights_from_hf(hf_model):
     return {k: v.detach().to(torch.float32).numpy()
             for k, v in hf_model.state_dict().items()}
@@ -63,6 +101,8 @@ def np_weights_from_hf(hf_model):
 @pytest.mark.parametrize("preset,builder", [
     ("tiny-llama", hf_llama),
     ("tiny-gemma2", hf_gemma2),
+    ("tiny-qwen2", hf_qwen2),
+    ("tiny-mistral", hf_mistral),
 ])
 def test_forward_matches_transformers(preset, builder):
     cfg = preset_config(preset)
@@ -83,6 +123,8 @@ def test_forward_matches_transformers(preset, builder):
 @pytest.mark.parametrize("preset,builder", [
     ("tiny-llama", hf_llama),
     ("tiny-gemma2", hf_gemma2),
+    ("tiny-qwen2", hf_qwen2),
+    ("tiny-mistral", hf_mistral),
 ])
 def test_incremental_decode_matches_prefill(preset, builder):
     """KV-cache decode path == full-prefill logits (cache correctness)."""
