@@ -1,8 +1,11 @@
 """HTTP serving layer with CONTINUOUS request batching.
 
 The reference has no server (no CLI, no server); this is the
-deployment-facing wrapper around the engine: an OpenAI-style
-``/v1/completions`` endpoint plus ``/health`` and ``/stats``.
+deployment-facing wrapper around the engine: OpenAI-style
+``/v1/completions`` and ``/v1/chat/completions`` endpoints (both with
+``"stream": true`` SSE token streaming) plus ``/health`` and ``/stats``.
+Chat messages go through the tokenizer's chat template when the
+checkpoint ships one, else a neutral ``<|role|>`` fallback template.
 
 Concurrent requests are NOT serialized: a scheduler thread owns the
 engine and runs a ragged lockstep group (per-sequence KV pools, per-row
@@ -21,6 +24,7 @@ back to single-sequence generate().
 
 from __future__ import annotations
 
+import json
 import queue
 import threading
 import time
@@ -38,8 +42,25 @@ try:  # pydantic model must live at module scope (ForwardRef resolution)
         strategy: str = "min_p"   # min_p | greedy | top_k | top_p
         seed: Optional[int] = None
         stop_on_eos: bool = True
+        stream: bool = False      # SSE token stream (OpenAI-style)
+
+    class ChatMessage(BaseModel):
+        role: str
+        content: str
+
+    class ChatCompletionRequest(BaseModel):
+        messages: list
+        max_tokens: int = 128
+        temperature: float = 1.0
+        min_p: float = 0.1
+        strategy: str = "min_p"
+        seed: Optional[int] = None
+        stop_on_eos: bool = True
+        stream: bool = False
 except ImportError:  # pragma: no cover - serving is optional
     CompletionRequest = None
+    ChatMessage = None
+    ChatCompletionRequest = None
 
 
 @dataclass
@@ -49,6 +70,7 @@ class _Pending:
     result: Optional[dict] = None
     error: Optional[Exception] = None
     t0: float = field(default_factory=time.time)
+    on_token: Optional[object] = None   # str -> None; SSE streaming hook
 
 
 class BatchScheduler:
@@ -75,13 +97,23 @@ class BatchScheduler:
             raise p.error
         return p.result
 
+    def submit_async(self, req, on_token=None) -> _Pending:
+        """Enqueue without blocking; the caller streams from on_token
+        and watches pending.done (SSE path)."""
+        p = _Pending(req, on_token=on_token)
+        self.q.put(p)
+        return p
+
     @staticmethod
     def _key(req):
         return (req.strategy, round(req.min_p, 6),
                 round(req.temperature, 6), req.stop_on_eos)
 
     def _batchable(self, req) -> bool:
-        return self.max_batch > 1 and req.strategy in ("greedy", "min_p")
+        # streaming requests take the generate_one path (per-token
+        # callback); the lockstep group only surfaces ids per chunk
+        return (self.max_batch > 1 and req.strategy in ("greedy", "min_p")
+                and not getattr(req, "stream", False))
 
     def _poll_compatible(self, key, deferred):
         """Non-blocking: next queued pending with this sampling key;
@@ -107,7 +139,11 @@ class BatchScheduler:
                 self.stats["requests"] += 1
                 self.stats["max_group"] = max(self.stats["max_group"], 1)
                 try:
-                    first.result = self.generate_one(first.req)
+                    if first.on_token is not None:
+                        first.result = self.generate_one(
+                            first.req, on_token=first.on_token)
+                    else:
+                        first.result = self.generate_one(first.req)
                 except Exception as e:
                     first.error = e
                 first.done.set()
@@ -172,11 +208,12 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
                if isinstance(eos, (int, float)) else {int(e) for e in eos})
 
     def _payload(req, ids, text, total_s, prefill_s, tps) -> dict:
+        hit_eos = bool(ids) and int(ids[-1]) in eos_set
         return {
             "object": "text_completion",
             "model": model_name,
             "choices": [{"text": text, "index": 0,
-                         "finish_reason": "stop"}],
+                         "finish_reason": "stop" if hit_eos else "length"}],
             "usage": {
                 "prompt_tokens": len(tok.encode(req.prompt)),
                 "completion_tokens": len(ids),
@@ -188,14 +225,15 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
             },
         }
 
-    def generate_one(req) -> dict:
+    def generate_one(req, on_token=None) -> dict:
         params = L.SamplingParams(strategy=req.strategy, min_p=req.min_p,
                                   temperature=req.temperature,
                                   seed=req.seed)
         t0 = time.time()
         out = L.generate(req.prompt, tok, model,
                          max_tokens=req.max_tokens, params=params,
-                         stream=False, stop_on_eos=req.stop_on_eos)
+                         stream=False, stop_on_eos=req.stop_on_eos,
+                         on_token=on_token)
         return _payload(req, out.token_ids, out.text, time.time() - t0,
                         out.prefill_time_s, out.decode_tokens_per_s)
 
@@ -316,9 +354,71 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
     def stats():
         return dict(sched.stats)
 
+    def _sse(req, chunk_of):
+        """Run req through the scheduler, yielding one SSE event per
+        decoded token piece, then a final [DONE].  Token callbacks fire
+        on the scheduler thread; all of them happen-before done.set(),
+        so `done and empty` is a safe termination check."""
+        from fastapi.responses import StreamingResponse
+
+        q: "queue.Queue[str]" = queue.Queue()
+        p = sched.submit_async(req, on_token=q.put)
+
+        def gen():
+            while not (p.done.is_set() and q.empty()):
+                try:
+                    piece = q.get(timeout=0.05)
+                except queue.Empty:
+                    continue
+                yield "data: " + json.dumps(chunk_of(piece)) + "\n\n"
+            if p.error is not None:
+                yield ("data: " + json.dumps({"error": str(p.error)})
+                       + "\n\n")
+            yield "data: [DONE]\n\n"
+
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
     @app.post("/v1/completions")
     def completions(req: CompletionRequest = Body(...)):
+        if req.stream:
+            return _sse(req, lambda piece: {
+                "object": "text_completion.chunk", "model": model_name,
+                "choices": [{"text": piece, "index": 0}]})
         return sched.submit(req)
+
+    def _chat_prompt(messages) -> str:
+        msgs = [{"role": m["role"], "content": m["content"]}
+                if isinstance(m, dict) else
+                {"role": m.role, "content": m.content} for m in messages]
+        template = getattr(tok, "chat_template", None)
+        if template and hasattr(tok, "apply_chat_template"):
+            return tok.apply_chat_template(msgs, tokenize=False,
+                                           add_generation_prompt=True)
+        # checkpoint-agnostic fallback (byte tokenizer / no template)
+        lines = [f"<|{m['role']}|>\n{m['content']}" for m in msgs]
+        return "\n".join(lines) + "\n<|assistant|>\n"
+
+    @app.post("/v1/chat/completions")
+    def chat_completions(req: ChatCompletionRequest = Body(...)):
+        creq = CompletionRequest(
+            prompt=_chat_prompt(req.messages), max_tokens=req.max_tokens,
+            temperature=req.temperature, min_p=req.min_p,
+            strategy=req.strategy, seed=req.seed,
+            stop_on_eos=req.stop_on_eos, stream=req.stream)
+        if req.stream:
+            return _sse(creq, lambda piece: {
+                "object": "chat.completion.chunk", "model": model_name,
+                "choices": [{"delta": {"content": piece}, "index": 0}]})
+        out = sched.submit(creq)
+        choice = out["choices"][0]
+        return {
+            "object": "chat.completion", "model": model_name,
+            "choices": [{"index": 0,
+                         "message": {"role": "assistant",
+                                     "content": choice["text"]},
+                         "finish_reason": choice["finish_reason"]}],
+            "usage": out["usage"], "timings": out["timings"],
+        }
 
     return app
 

@@ -184,3 +184,73 @@ def test_batch_scheduler_grouping_and_deferral():
         sched.submit(req("boom"))
     assert sched.stats["requests"] >= 4
     assert sched.stats["max_group"] == 0 or True  # stats sanity only
+
+
+def test_chat_completions_endpoint():
+    """OpenAI-shaped chat endpoint: messages assemble through the
+    (fallback) chat template, response carries an assistant message."""
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy")
+    client = TestClient(app)
+    r = client.post("/v1/chat/completions", json={
+        "messages": [{"role": "system", "content": "be brief"},
+                     {"role": "user", "content": "hello"}],
+        "max_tokens": 4, "strategy": "greedy", "stop_on_eos": False})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    msg = body["choices"][0]["message"]
+    assert msg["role"] == "assistant" and isinstance(msg["content"], str)
+    assert body["usage"]["completion_tokens"] == 4
+    assert body["choices"][0]["finish_reason"] in ("stop", "length")
+
+
+def test_completions_streaming_sse():
+    """"stream": true returns SSE chunks whose concatenated text equals
+    the non-streaming greedy completion, terminated by [DONE]."""
+    fastapi = pytest.importorskip("fastapi")
+    import json as _json
+    from fastapi.testclient import TestClient
+
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy")
+    client = TestClient(app)
+    base = {"prompt": "Once upon a time", "max_tokens": 6,
+            "strategy": "greedy", "stop_on_eos": False}
+
+    plain = client.post("/v1/completions", json=base).json()
+    with client.stream("POST", "/v1/completions",
+                       json={**base, "stream": True}) as r:
+        assert r.status_code == 200
+        assert "text/event-stream" in r.headers["content-type"]
+        events = [ln for ln in r.iter_lines() if ln.startswith("data: ")]
+    assert events[-1] == "data: [DONE]"
+    pieces = [_json.loads(e[len("data: "):]) for e in events[:-1]]
+    text = "".join(p["choices"][0]["text"] for p in pieces)
+    assert text == plain["choices"][0]["text"]
+    assert len(pieces) >= 2  # actually streamed per token
+
+
+def test_chat_streaming_sse_delta_format():
+    fastapi = pytest.importorskip("fastapi")
+    import json as _json
+    from fastapi.testclient import TestClient
+
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy")
+    client = TestClient(app)
+    with client.stream("POST", "/v1/chat/completions", json={
+            "messages": [{"role": "user", "content": "hi"}],
+            "max_tokens": 4, "strategy": "greedy",
+            "stop_on_eos": False, "stream": True}) as r:
+        events = [ln for ln in r.iter_lines() if ln.startswith("data: ")]
+    assert events[-1] == "data: [DONE]"
+    chunks = [_json.loads(e[len("data: "):]) for e in events[:-1]]
+    assert all(c["object"] == "chat.completion.chunk" for c in chunks)
+    assert all("content" in c["choices"][0]["delta"] for c in chunks)
