@@ -215,3 +215,73 @@ def test_mixtral_synthetic_checkpoint_roundtrip(tmp_path):
                      params=L.SamplingParams(strategy="greedy"),
                      stop_on_eos=False)
     assert len(out.token_ids) == 4
+
+
+def test_generate_greedy_matches_hf_generate():
+    """SURVEY §4 integration check: the full generate() loop (prefill +
+    cached feedback of sampled ids) reproduces transformers' own
+    greedy `generate` token-for-token on a tiny random-init Llama."""
+    import torch
+    from tests.test_numpy_oracle import hf_llama, np_weights_from_hf
+    from llm_np_cp_amd.core.config import preset_config
+    from llm_np_cp_amd.models.numpy_ref import NumpyKVCache, NumpyModel
+
+    cfg = preset_config("tiny-llama")
+    hf = hf_llama(cfg)
+    model = NumpyModel(cfg, np_weights_from_hf(hf))
+    model.make_cache = lambda n: NumpyKVCache(cfg, n)
+
+    prompt_ids = [5, 17, 99, 3, 250]
+
+    class IdTok:
+        def encode(self, s):
+            return list(prompt_ids)
+
+        def decode(self, ids):
+            return " ".join(str(int(i)) for i in ids)
+
+    n_new = 12
+    with torch.no_grad():
+        ref = hf.generate(torch.tensor([prompt_ids]),
+                          max_new_tokens=n_new, do_sample=False,
+                          use_cache=True)[0][len(prompt_ids):].tolist()
+    out = L.generate("x", IdTok(), model, max_tokens=n_new, stream=False,
+                     params=SamplingParams(strategy="greedy"),
+                     stop_on_eos=False)
+    assert out.token_ids == ref
+
+
+def test_filter_probs_properties():
+    """filter_probs (the distribution speculative sampling builds on):
+    normalization, strategy support rules, greedy one-hot."""
+    from llm_np_cp_amd.runtime.sampling import filter_probs
+
+    rng = np.random.default_rng(4)
+    logits = rng.standard_normal(64).astype(np.float32) * 3
+
+    for strat, kw in [("min_p", {}), ("top_k", {"top_k": 7}),
+                      ("top_p", {"top_p": 0.8}), ("temperature", {}),
+                      ("greedy", {})]:
+        p = filter_probs(logits, SamplingParams(strategy=strat, **kw))
+        assert abs(p.sum() - 1.0) < 1e-12
+        assert (p >= 0).all()
+
+    g = filter_probs(logits, SamplingParams(strategy="greedy"))
+    assert g[np.argmax(logits)] == 1.0 and (g > 0).sum() == 1
+
+    k7 = filter_probs(logits, SamplingParams(strategy="top_k", top_k=7))
+    assert (k7 > 0).sum() <= 7
+
+    mp = filter_probs(logits, SamplingParams(strategy="min_p", min_p=0.2))
+    nz = mp[mp > 0]
+    assert nz.min() >= 0.2 * nz.max() * (1 - 1e-9)
+
+    tp = filter_probs(logits, SamplingParams(strategy="top_p", top_p=0.8))
+    # kept mass covers >= 0.8 of the unfiltered temperature distribution
+    base = filter_probs(logits, SamplingParams(strategy="temperature"))
+    assert base[tp > 0].sum() >= 0.8 - 1e-9
+
+    # temperature scaling sharpens: lower T raises p_max
+    hot = filter_probs(logits, SamplingParams(strategy="temperature",
+                                              temperature=0.5))
+    assert hot.max() > base.max()
