@@ -851,12 +851,16 @@ class GPUModel:
     def generate_tokens(self, prompt_ids, max_tokens: int,
                         greedy: bool = True, min_p: float = 0.1,
                         eos_id=None, chunk: int = 16, on_ids=None,
-                        temperature: float = 1.0):
+                        temperature: float = 1.0, stop_fn=None):
         """Fast generate: device-side hipGraph decode in chunks, host
         sees ids every `chunk` tokens (streaming + EOS stop).  Used by
         runtime.generate() for greedy/min-p on GPU models.
         ``eos_id`` may be an int or a collection of ints (HF configs
-        often store a list, e.g. Llama-3.2-Instruct)."""
+        often store a list, e.g. Llama-3.2-Instruct).  ``stop_fn``
+        (optional) sees the accumulated ids after each chunk and returns
+        True to end generation early (stop-string support; the caller
+        truncates — chunking may have produced up to chunk-1 extra
+        ids)."""
         import time as _time
         prompt_ids = np.asarray(prompt_ids)
         room = self.max_seq - len(prompt_ids.ravel()) - 1
@@ -891,6 +895,8 @@ class GPUModel:
                 if on_ids:
                     on_ids(take)
                 if stop:
+                    break
+                if stop_fn is not None and stop_fn(out):
                     break
         return out
 

@@ -32,6 +32,7 @@ class GenerateResult:
     token_ids: List[int]
     prefill_time_s: float = 0.0
     decode_time_s: float = 0.0
+    finish_reason: str = "length"   # "stop" (EOS / stop string) | "length"
 
     @property
     def decode_tokens_per_s(self) -> float:
@@ -39,15 +40,78 @@ class GenerateResult:
         return n / self.decode_time_s if self.decode_time_s > 0 else 0.0
 
 
+class _StopScan:
+    """Incremental stop-sequence scanner for streaming: emits only text
+    that can no longer become part of a stop string (holds back
+    max(len(stop))-1 chars), cuts the stream at the earliest match."""
+
+    def __init__(self, stops: List[str], emit):
+        self.stops = stops
+        self.emit = emit
+        self.hold = max(len(s) for s in stops) - 1
+        self.buf = ""
+        self.hit = False
+
+    def feed(self, piece: str) -> bool:
+        """Returns True once a stop string appeared (emission stops
+        BEFORE it)."""
+        if self.hit:
+            return True
+        self.buf += piece
+        found = [i for i in (self.buf.find(s) for s in self.stops)
+                 if i >= 0]
+        if found:
+            cut = min(found)
+            if cut and self.emit:
+                self.emit(self.buf[:cut])
+            self.hit = True
+            return True
+        if self.hold == 0 or len(self.buf) > self.hold:
+            safe = self.buf if self.hold == 0 else self.buf[:-self.hold]
+            if safe and self.emit:
+                self.emit(safe)
+            self.buf = self.buf[len(safe):]
+        return False
+
+    def flush(self):
+        if not self.hit and self.buf and self.emit:
+            self.emit(self.buf)
+            self.buf = ""
+
+
+def _truncate_at_stop(ids: List[int], tokenizer, stops: List[str]):
+    """Earliest stop occurrence in decode(ids) -> (ids', text',
+    hit).  The stop string itself is excluded (OpenAI semantics); ids'
+    is the minimal token prefix covering text'."""
+    text = tokenizer.decode(ids)
+    found = [i for i in (text.find(s) for s in stops) if i >= 0]
+    if not found:
+        return ids, text, False
+    keep = text[:min(found)]
+    out: List[int] = []
+    for t in ids:
+        if len(tokenizer.decode(out)) >= len(keep):
+            break
+        out.append(int(t))
+    return out, keep, True
+
+
 def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
              kv_cache=None, params: Optional[SamplingParams] = None,
              use_cache: bool = True, stream: bool = True,
              stop_on_eos: bool = True,
+             stop: Optional[List[str]] = None,
              on_token: Optional[Callable[[str], None]] = None) -> GenerateResult:
-    """Generate up to max_tokens continuation tokens of prompt."""
+    """Generate up to max_tokens continuation tokens of prompt.
+
+    ``stop``: optional stop strings — generation ends at the earliest
+    occurrence in the decoded text, which is excluded from the result
+    (OpenAI semantics).  Streaming holds back max(len(stop))-1 chars so
+    a stop string spanning token boundaries is never emitted."""
     params = params or SamplingParams()
     rng = np.random.default_rng(params.seed)
     prompt_ids = list(tokenizer.encode(prompt))
+    stops = [s for s in (stop or []) if s]
     # HF configs store eos_token_id as an int OR a list (e.g.
     # Llama-3.2-Instruct: [128001, 128008, 128009]) — normalize to a set
     eos = getattr(model.config, "eos_token_id", None)
@@ -60,44 +124,70 @@ def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
     emit = on_token
     if emit is None and stream:
         emit = lambda s: (sys.stdout.write(s), sys.stdout.flush())
+    scan = _StopScan(stops, emit) if stops else None
+
+    def _finish(ids: List[int], hit_eos: bool):
+        """Apply stop-string truncation; build the result text."""
+        ids = [int(i) for i in ids]
+        if stops:
+            ids, text, hit = _truncate_at_stop(ids, tokenizer, stops)
+            if scan is not None:
+                scan.flush()
+            return ids, text, "stop" if (hit or hit_eos) else "length"
+        return ids, tokenizer.decode(ids), "stop" if hit_eos else "length"
 
     # fast path: device-side chunked decode loop (GPU engine, greedy or
     # min-p, cached mode) — same sampler semantics, device RNG
     if (use_cache and hasattr(model, "generate_tokens")
             and params.strategy in ("greedy", "min_p")):
         t0 = time.perf_counter()
-        pieces: List[int] = []
 
         def _emit(ids_chunk):
-            if emit:
-                emit(tokenizer.decode(list(ids_chunk)))
+            piece = tokenizer.decode(list(ids_chunk))
+            if scan is not None:
+                scan.feed(piece)
+            elif emit:
+                emit(piece)
+
+        def _stop_fn(all_ids):
+            return any(s in tokenizer.decode(list(all_ids)) for s in stops)
 
         ids = model.generate_tokens(
             prompt_ids, max_tokens,
             greedy=params.strategy == "greedy", min_p=params.min_p,
             eos_id=eos_set if stop_on_eos else None, on_ids=_emit,
-            temperature=params.temperature)
+            temperature=params.temperature,
+            stop_fn=_stop_fn if stops else None)
         dt = time.perf_counter() - t0
         tp = getattr(model, "last_prefill_time_s", 0.0)
-        res = GenerateResult(text=tokenizer.decode(list(ids)),
-                             token_ids=[int(i) for i in ids],
-                             prefill_time_s=tp,
-                             decode_time_s=max(dt - tp, 1e-9))
-        return res
+        hit_eos = bool(ids) and stop_on_eos and int(ids[-1]) in eos_set
+        ids, text, reason = _finish(ids, hit_eos)
+        if not stops and len(ids) >= max_tokens and not hit_eos:
+            reason = "length"
+        return GenerateResult(text=text, token_ids=ids,
+                              prefill_time_s=tp,
+                              decode_time_s=max(dt - tp, 1e-9),
+                              finish_reason=reason)
 
     out_ids: List[int] = []
     t0 = time.perf_counter()
     logits = model.forward(np.asarray(prompt_ids, dtype=np.int64), kv_cache, 0)
     t_prefill = time.perf_counter() - t0
 
+    hit_eos = False
     t1 = time.perf_counter()
     for _ in range(max_tokens):
         next_id = sample_token(np.asarray(logits[-1], dtype=np.float32),
                                params, rng)
         out_ids.append(next_id)
-        if emit:
-            emit(tokenizer.decode([next_id]))
+        piece = tokenizer.decode([next_id])
+        if scan is not None:
+            if scan.feed(piece):
+                break
+        elif emit:
+            emit(piece)
         if stop_on_eos and next_id in eos_set:
+            hit_eos = True
             break
         if use_cache:
             logits = model.forward(np.asarray([next_id], dtype=np.int64),
@@ -108,11 +198,13 @@ def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
             logits = model.forward(np.asarray(full, dtype=np.int64), fresh, 0)
     t_decode = time.perf_counter() - t1
 
+    out_ids, text, reason = _finish(out_ids, hit_eos)
     return GenerateResult(
-        text=tokenizer.decode(out_ids),
+        text=text,
         token_ids=out_ids,
         prefill_time_s=t_prefill,
         decode_time_s=t_decode,
+        finish_reason=reason,
     )
 
 

@@ -43,6 +43,7 @@ try:  # pydantic model must live at module scope (ForwardRef resolution)
         seed: Optional[int] = None
         stop_on_eos: bool = True
         stream: bool = False      # SSE token stream (OpenAI-style)
+        stop: Optional[list] = None   # stop strings (OpenAI semantics)
 
     class ChatMessage(BaseModel):
         role: str
@@ -57,6 +58,7 @@ try:  # pydantic model must live at module scope (ForwardRef resolution)
         seed: Optional[int] = None
         stop_on_eos: bool = True
         stream: bool = False
+        stop: Optional[list] = None
 except ImportError:  # pragma: no cover - serving is optional
     CompletionRequest = None
     ChatMessage = None
@@ -110,10 +112,12 @@ class BatchScheduler:
                 round(req.temperature, 6), req.stop_on_eos)
 
     def _batchable(self, req) -> bool:
-        # streaming requests take the generate_one path (per-token
-        # callback); the lockstep group only surfaces ids per chunk
+        # streaming and stop-string requests take the generate_one
+        # path (per-token callback / text-level truncation); the
+        # lockstep group only surfaces ids per chunk
         return (self.max_batch > 1 and req.strategy in ("greedy", "min_p")
-                and not getattr(req, "stream", False))
+                and not getattr(req, "stream", False)
+                and not getattr(req, "stop", None))
 
     def _poll_compatible(self, key, deferred):
         """Non-blocking: next queued pending with this sampling key;
@@ -207,13 +211,16 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
     eos_set = (set() if eos is None else {int(eos)}
                if isinstance(eos, (int, float)) else {int(e) for e in eos})
 
-    def _payload(req, ids, text, total_s, prefill_s, tps) -> dict:
-        hit_eos = bool(ids) and int(ids[-1]) in eos_set
+    def _payload(req, ids, text, total_s, prefill_s, tps,
+                 finish_reason=None) -> dict:
+        if finish_reason is None:
+            hit_eos = bool(ids) and int(ids[-1]) in eos_set
+            finish_reason = "stop" if hit_eos else "length"
         return {
             "object": "text_completion",
             "model": model_name,
             "choices": [{"text": text, "index": 0,
-                         "finish_reason": "stop" if hit_eos else "length"}],
+                         "finish_reason": finish_reason}],
             "usage": {
                 "prompt_tokens": len(tok.encode(req.prompt)),
                 "completion_tokens": len(ids),
@@ -233,9 +240,11 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
         out = L.generate(req.prompt, tok, model,
                          max_tokens=req.max_tokens, params=params,
                          stream=False, stop_on_eos=req.stop_on_eos,
+                         stop=getattr(req, "stop", None),
                          on_token=on_token)
         return _payload(req, out.token_ids, out.text, time.time() - t0,
-                        out.prefill_time_s, out.decode_tokens_per_s)
+                        out.prefill_time_s, out.decode_tokens_per_s,
+                        finish_reason=out.finish_reason)
 
     class _Row:
         __slots__ = ("pending", "ids", "first_consumed")
@@ -404,7 +413,8 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
             prompt=_chat_prompt(req.messages), max_tokens=req.max_tokens,
             temperature=req.temperature, min_p=req.min_p,
             strategy=req.strategy, seed=req.seed,
-            stop_on_eos=req.stop_on_eos, stream=req.stream)
+            stop_on_eos=req.stop_on_eos, stream=req.stream,
+            stop=req.stop)
         if req.stream:
             return _sse(creq, lambda piece: {
                 "object": "chat.completion.chunk", "model": model_name,

@@ -285,3 +285,70 @@ def test_filter_probs_properties():
     hot = filter_probs(logits, SamplingParams(strategy="temperature",
                                               temperature=0.5))
     assert hot.max() > base.max()
+
+
+def test_stop_sequences_truncate_and_exclude():
+    """OpenAI stop semantics: generation ends at the earliest stop
+    occurrence; the stop string is excluded from text and the returned
+    ids are the minimal prefix covering it."""
+    tok, model, cfg = L.load_model("tiny-llama", backend="numpy", seed=0)
+    p = SamplingParams(strategy="greedy")
+    base = L.generate("Once upon a time", tok, model, max_tokens=24,
+                      stream=False, params=p, stop_on_eos=False)
+    assert base.finish_reason == "length"
+    # pick a stop string from the middle of the greedy continuation
+    mid = base.text[8:11]
+    assert mid  # non-empty
+    res = L.generate("Once upon a time", tok, model, max_tokens=24,
+                     stream=False, params=p, stop_on_eos=False,
+                     stop=[mid])
+    assert res.finish_reason == "stop"
+    assert mid not in res.text
+    assert res.text == base.text[:base.text.find(mid)]
+    assert len(res.token_ids) < len(base.token_ids)
+    assert tok.decode(res.token_ids).startswith(res.text)
+
+
+def test_stop_sequence_streaming_never_emits_stop():
+    """Streaming with a stop string spanning token boundaries: emitted
+    chunks concatenate to exactly the truncated text (the holdback
+    buffer keeps partial stop matches back)."""
+    tok, model, cfg = L.load_model("tiny-llama", backend="numpy", seed=0)
+    p = SamplingParams(strategy="greedy")
+    base = L.generate("Once upon a time", tok, model, max_tokens=20,
+                      stream=False, params=p, stop_on_eos=False)
+    stop = base.text[6:10]  # spans >1 byte-token
+    chunks = []
+    res = L.generate("Once upon a time", tok, model, max_tokens=20,
+                     stream=False, params=p, stop_on_eos=False,
+                     stop=[stop], on_token=chunks.append)
+    assert "".join(chunks) == res.text
+    assert stop not in "".join(chunks)
+
+
+def test_stop_sequence_no_match_flushes_everything():
+    tok, model, cfg = L.load_model("tiny-llama", backend="numpy", seed=0)
+    p = SamplingParams(strategy="greedy")
+    chunks = []
+    res = L.generate("abc", tok, model, max_tokens=8, stream=False,
+                     params=p, stop_on_eos=False,
+                     stop=["ZXQW-NEVER"], on_token=chunks.append)
+    assert res.finish_reason == "length"
+    assert "".join(chunks) == res.text
+    assert len(res.token_ids) == 8
+
+
+def test_stop_earliest_of_multiple():
+    tok, model, cfg = L.load_model("tiny-llama", backend="numpy", seed=0)
+    p = SamplingParams(strategy="greedy")
+    base = L.generate("Once upon a time", tok, model, max_tokens=24,
+                      stream=False, params=p, stop_on_eos=False)
+    s_early, s_late = base.text[4:7], base.text[12:15]
+    res = L.generate("Once upon a time", tok, model, max_tokens=24,
+                     stream=False, params=p, stop_on_eos=False,
+                     stop=[s_late, s_early])
+    # earliest occurrence of EITHER stop wins (repetitive greedy text
+    # may contain the slice earlier than where it was taken from)
+    want_cut = min(c for c in (base.text.find(s_early),
+                               base.text.find(s_late)) if c >= 0)
+    assert res.text == base.text[:want_cut]

@@ -254,3 +254,26 @@ def test_chat_streaming_sse_delta_format():
     chunks = [_json.loads(e[len("data: "):]) for e in events[:-1]]
     assert all(c["object"] == "chat.completion.chunk" for c in chunks)
     assert all("content" in c["choices"][0]["delta"] for c in chunks)
+
+
+def test_completions_stop_strings():
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy")
+    client = TestClient(app)
+    base = client.post("/v1/completions", json={
+        "prompt": "Once upon a time", "max_tokens": 16,
+        "strategy": "greedy", "stop_on_eos": False}).json()
+    full = base["choices"][0]["text"]
+    stop = full[3:6]
+    r = client.post("/v1/completions", json={
+        "prompt": "Once upon a time", "max_tokens": 16,
+        "strategy": "greedy", "stop_on_eos": False,
+        "stop": [stop]}).json()
+    got = r["choices"][0]["text"]
+    assert got == full[:full.find(stop)]
+    assert r["choices"][0]["finish_reason"] == "stop"
+    assert base["choices"][0]["finish_reason"] == "length"
