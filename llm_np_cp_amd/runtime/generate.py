@@ -33,6 +33,10 @@ class GenerateResult:
     prefill_time_s: float = 0.0
     decode_time_s: float = 0.0
     finish_reason: str = "length"   # "stop" (EOS / stop string) | "length"
+    # per emitted token, when requested: {"id", "token", "logprob",
+    # "top": [{"id", "token", "logprob"}, ...]} — log-softmax of the RAW
+    # model logits (pre-temperature/filter, OpenAI semantics)
+    logprobs: Optional[List[dict]] = None
 
     @property
     def decode_tokens_per_s(self) -> float:
@@ -79,6 +83,24 @@ class _StopScan:
             self.buf = ""
 
 
+def _logprob_entry(logits: np.ndarray, chosen: int, top_n: int,
+                   tokenizer) -> dict:
+    """Stable log-softmax of the raw logits row -> chosen-token logprob
+    + the top_n most likely alternatives."""
+    z = logits.astype(np.float64)
+    z = z - z.max()
+    lse = np.log(np.exp(z).sum())
+    lp = z - lse
+    top_ids = np.argsort(-lp)[:max(top_n, 0)]
+    return {
+        "id": int(chosen),
+        "token": tokenizer.decode([int(chosen)]),
+        "logprob": float(lp[chosen]),
+        "top": [{"id": int(i), "token": tokenizer.decode([int(i)]),
+                 "logprob": float(lp[i])} for i in top_ids],
+    }
+
+
 def _truncate_at_stop(ids: List[int], tokenizer, stops: List[str]):
     """Earliest stop occurrence in decode(ids) -> (ids', text',
     hit).  The stop string itself is excluded (OpenAI semantics); ids'
@@ -101,13 +123,19 @@ def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
              use_cache: bool = True, stream: bool = True,
              stop_on_eos: bool = True,
              stop: Optional[List[str]] = None,
+             logprobs: Optional[int] = None,
              on_token: Optional[Callable[[str], None]] = None) -> GenerateResult:
     """Generate up to max_tokens continuation tokens of prompt.
 
     ``stop``: optional stop strings — generation ends at the earliest
     occurrence in the decoded text, which is excluded from the result
     (OpenAI semantics).  Streaming holds back max(len(stop))-1 chars so
-    a stop string spanning token boundaries is never emitted."""
+    a stop string spanning token boundaries is never emitted.
+
+    ``logprobs``: when an int N >= 0, GenerateResult.logprobs carries
+    per-token raw-model logprobs with the top-N alternatives (takes the
+    logits-visible step loop — the device-side sampler never surfaces
+    the full distribution)."""
     params = params or SamplingParams()
     rng = np.random.default_rng(params.seed)
     prompt_ids = list(tokenizer.encode(prompt))
@@ -138,7 +166,8 @@ def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
 
     # fast path: device-side chunked decode loop (GPU engine, greedy or
     # min-p, cached mode) — same sampler semantics, device RNG
-    if (use_cache and hasattr(model, "generate_tokens")
+    if (use_cache and logprobs is None
+            and hasattr(model, "generate_tokens")
             and params.strategy in ("greedy", "min_p")):
         t0 = time.perf_counter()
 
@@ -175,11 +204,15 @@ def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
     t_prefill = time.perf_counter() - t0
 
     hit_eos = False
+    lp_entries: List[dict] = [] if logprobs is not None else None
     t1 = time.perf_counter()
     for _ in range(max_tokens):
-        next_id = sample_token(np.asarray(logits[-1], dtype=np.float32),
-                               params, rng)
+        row = np.asarray(logits[-1], dtype=np.float32)
+        next_id = sample_token(row, params, rng)
         out_ids.append(next_id)
+        if lp_entries is not None:
+            lp_entries.append(_logprob_entry(row, next_id, logprobs,
+                                             tokenizer))
         piece = tokenizer.decode([next_id])
         if scan is not None:
             if scan.feed(piece):
@@ -199,12 +232,15 @@ def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
     t_decode = time.perf_counter() - t1
 
     out_ids, text, reason = _finish(out_ids, hit_eos)
+    if lp_entries is not None:
+        lp_entries = lp_entries[:len(out_ids)]  # stop truncation
     return GenerateResult(
         text=text,
         token_ids=out_ids,
         prefill_time_s=t_prefill,
         decode_time_s=t_decode,
         finish_reason=reason,
+        logprobs=lp_entries,
     )
 
 

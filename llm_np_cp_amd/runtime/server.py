@@ -44,6 +44,7 @@ try:  # pydantic model must live at module scope (ForwardRef resolution)
         stop_on_eos: bool = True
         stream: bool = False      # SSE token stream (OpenAI-style)
         stop: Optional[list] = None   # stop strings (OpenAI semantics)
+        logprobs: Optional[int] = None  # top-N per-token logprobs
 
     class ChatMessage(BaseModel):
         role: str
@@ -117,7 +118,8 @@ class BatchScheduler:
         # lockstep group only surfaces ids per chunk
         return (self.max_batch > 1 and req.strategy in ("greedy", "min_p")
                 and not getattr(req, "stream", False)
-                and not getattr(req, "stop", None))
+                and not getattr(req, "stop", None)
+                and getattr(req, "logprobs", None) is None)
 
     def _poll_compatible(self, key, deferred):
         """Non-blocking: next queued pending with this sampling key;
@@ -241,10 +243,20 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
                          max_tokens=req.max_tokens, params=params,
                          stream=False, stop_on_eos=req.stop_on_eos,
                          stop=getattr(req, "stop", None),
+                         logprobs=getattr(req, "logprobs", None),
                          on_token=on_token)
-        return _payload(req, out.token_ids, out.text, time.time() - t0,
-                        out.prefill_time_s, out.decode_tokens_per_s,
-                        finish_reason=out.finish_reason)
+        pay = _payload(req, out.token_ids, out.text, time.time() - t0,
+                       out.prefill_time_s, out.decode_tokens_per_s,
+                       finish_reason=out.finish_reason)
+        if out.logprobs is not None:
+            pay["choices"][0]["logprobs"] = {
+                "tokens": [e["token"] for e in out.logprobs],
+                "token_logprobs": [e["logprob"] for e in out.logprobs],
+                "top_logprobs": [{t["token"]: t["logprob"]
+                                  for t in e["top"]}
+                                 for e in out.logprobs],
+            }
+        return pay
 
     class _Row:
         __slots__ = ("pending", "ids", "first_consumed")

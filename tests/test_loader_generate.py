@@ -352,3 +352,43 @@ def test_stop_earliest_of_multiple():
     want_cut = min(c for c in (base.text.find(s_early),
                                base.text.find(s_late)) if c >= 0)
     assert res.text == base.text[:want_cut]
+
+
+def test_logprobs_surface():
+    """logprobs=N: per-token raw log-softmax of the model distribution,
+    chosen-token logprob consistent with greedy argmax, top list sorted
+    and containing the chosen token at rank 0 for greedy."""
+    tok, model, cfg = L.load_model("tiny-llama", backend="numpy", seed=0)
+    res = L.generate("Once upon a time", tok, model, max_tokens=5,
+                     stream=False, params=SamplingParams(strategy="greedy"),
+                     stop_on_eos=False, logprobs=3)
+    assert res.logprobs is not None and len(res.logprobs) == 5
+    for e, tid in zip(res.logprobs, res.token_ids):
+        assert e["id"] == tid
+        assert e["logprob"] <= 0.0
+        tops = [t["logprob"] for t in e["top"]]
+        assert len(tops) == 3 and tops == sorted(tops, reverse=True)
+        # greedy: chosen == most likely
+        assert e["top"][0]["id"] == tid
+        assert abs(e["top"][0]["logprob"] - e["logprob"]) < 1e-12
+    # probabilities of the full vocab sum to 1 at every step is implied
+    # by log-softmax; spot-check top-3 mass <= 1
+    import math
+    assert sum(math.exp(t["logprob"])
+               for t in res.logprobs[0]["top"]) <= 1.0 + 1e-9
+
+
+def test_logprobs_via_server():
+    import pytest as _pytest
+    fastapi = _pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy")
+    client = TestClient(app)
+    r = client.post("/v1/completions", json={
+        "prompt": "hi", "max_tokens": 3, "strategy": "greedy",
+        "stop_on_eos": False, "logprobs": 2}).json()
+    lp = r["choices"][0]["logprobs"]
+    assert len(lp["tokens"]) == 3 == len(lp["token_logprobs"])
+    assert all(len(d) <= 2 for d in lp["top_logprobs"])
