@@ -364,6 +364,17 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
                            window_s=batch_window_ms / 1e3)
     app = FastAPI(title="llm_np_cp_amd", version=L.__version__)
 
+    @app.get("/v1/models")
+    def models():
+        return {"object": "list",
+                "data": [{"id": model_name, "object": "model",
+                          "owned_by": "llm_np_cp_amd",
+                          "meta": {"model_type": cfg.model_type,
+                                   "dtype": dtype,
+                                   "max_seq": model.max_seq
+                                   if hasattr(model, "max_seq")
+                                   else max_seq}}]}
+
     @app.get("/health")
     def health():
         return {"status": "ok", "model": model_name,

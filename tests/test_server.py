@@ -277,3 +277,16 @@ def test_completions_stop_strings():
     assert got == full[:full.find(stop)]
     assert r["choices"][0]["finish_reason"] == "stop"
     assert base["choices"][0]["finish_reason"] == "length"
+
+
+def test_models_endpoint():
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy")
+    client = TestClient(app)
+    r = client.get("/v1/models").json()
+    assert r["object"] == "list"
+    assert r["data"][0]["id"] == "tiny-llama"
+    assert r["data"][0]["meta"]["model_type"] == "llama"
