@@ -32,14 +32,18 @@ from dataclasses import dataclass, field
 from typing import Optional
 
 try:  # pydantic model must live at module scope (ForwardRef resolution)
-    from pydantic import BaseModel
+    from typing import Literal
+
+    from pydantic import BaseModel, Field
+
+    _Strategy = Literal["min_p", "greedy", "top_k", "top_p", "temperature"]
 
     class CompletionRequest(BaseModel):
         prompt: str
-        max_tokens: int = 128
-        temperature: float = 1.0
-        min_p: float = 0.1
-        strategy: str = "min_p"   # min_p | greedy | top_k | top_p
+        max_tokens: int = Field(128, ge=1)
+        temperature: float = Field(1.0, gt=0)
+        min_p: float = Field(0.1, ge=0, le=1)
+        strategy: _Strategy = "min_p"
         seed: Optional[int] = None
         stop_on_eos: bool = True
         stream: bool = False      # SSE token stream (OpenAI-style)
@@ -51,11 +55,11 @@ try:  # pydantic model must live at module scope (ForwardRef resolution)
         content: str
 
     class ChatCompletionRequest(BaseModel):
-        messages: list
-        max_tokens: int = 128
-        temperature: float = 1.0
-        min_p: float = 0.1
-        strategy: str = "min_p"
+        messages: list = Field(..., min_length=1)
+        max_tokens: int = Field(128, ge=1)
+        temperature: float = Field(1.0, gt=0)
+        min_p: float = Field(0.1, ge=0, le=1)
+        strategy: _Strategy = "min_p"
         seed: Optional[int] = None
         stop_on_eos: bool = True
         stream: bool = False
@@ -416,7 +420,11 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
             return _sse(req, lambda piece: {
                 "object": "text_completion.chunk", "model": model_name,
                 "choices": [{"text": piece, "index": 0}]})
-        return sched.submit(req)
+        try:
+            return sched.submit(req)
+        except ValueError as e:  # engine-level rejection (e.g. KV room)
+            from fastapi import HTTPException
+            raise HTTPException(status_code=400, detail=str(e))
 
     def _chat_prompt(messages) -> str:
         msgs = [{"role": m["role"], "content": m["content"]}
@@ -442,7 +450,11 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
             return _sse(creq, lambda piece: {
                 "object": "chat.completion.chunk", "model": model_name,
                 "choices": [{"delta": {"content": piece}, "index": 0}]})
-        out = sched.submit(creq)
+        try:
+            out = sched.submit(creq)
+        except ValueError as e:
+            from fastapi import HTTPException
+            raise HTTPException(status_code=400, detail=str(e))
         choice = out["choices"][0]
         return {
             "object": "chat.completion", "model": model_name,

