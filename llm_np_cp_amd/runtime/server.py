@@ -49,6 +49,7 @@ try:  # pydantic model must live at module scope (ForwardRef resolution)
         stream: bool = False      # SSE token stream (OpenAI-style)
         stop: Optional[list] = None   # stop strings (OpenAI semantics)
         logprobs: Optional[int] = None  # top-N per-token logprobs
+        n: int = Field(1, ge=1, le=16)  # independent samples
 
     class ChatMessage(BaseModel):
         role: str
@@ -123,7 +124,8 @@ class BatchScheduler:
         return (self.max_batch > 1 and req.strategy in ("greedy", "min_p")
                 and not getattr(req, "stream", False)
                 and not getattr(req, "stop", None)
-                and getattr(req, "logprobs", None) is None)
+                and getattr(req, "logprobs", None) is None
+                and getattr(req, "n", 1) == 1)
 
     def _poll_compatible(self, key, deferred):
         """Non-blocking: next queued pending with this sampling key;
@@ -239,27 +241,41 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
         }
 
     def generate_one(req, on_token=None) -> dict:
-        params = L.SamplingParams(strategy=req.strategy, min_p=req.min_p,
-                                  temperature=req.temperature,
-                                  seed=req.seed)
+        n = getattr(req, "n", 1) or 1
         t0 = time.time()
-        out = L.generate(req.prompt, tok, model,
-                         max_tokens=req.max_tokens, params=params,
-                         stream=False, stop_on_eos=req.stop_on_eos,
-                         stop=getattr(req, "stop", None),
-                         logprobs=getattr(req, "logprobs", None),
-                         on_token=on_token)
-        pay = _payload(req, out.token_ids, out.text, time.time() - t0,
-                       out.prefill_time_s, out.decode_tokens_per_s,
-                       finish_reason=out.finish_reason)
-        if out.logprobs is not None:
-            pay["choices"][0]["logprobs"] = {
-                "tokens": [e["token"] for e in out.logprobs],
-                "token_logprobs": [e["logprob"] for e in out.logprobs],
-                "top_logprobs": [{t["token"]: t["logprob"]
-                                  for t in e["top"]}
-                                 for e in out.logprobs],
-            }
+        pay = None
+        for idx in range(n):
+            seed = (req.seed + idx) if (req.seed is not None and n > 1) \
+                else req.seed
+            params = L.SamplingParams(strategy=req.strategy,
+                                      min_p=req.min_p,
+                                      temperature=req.temperature,
+                                      seed=seed)
+            out = L.generate(req.prompt, tok, model,
+                             max_tokens=req.max_tokens, params=params,
+                             stream=False, stop_on_eos=req.stop_on_eos,
+                             stop=getattr(req, "stop", None),
+                             logprobs=getattr(req, "logprobs", None),
+                             on_token=on_token if idx == 0 else None)
+            if pay is None:
+                pay = _payload(req, out.token_ids, out.text,
+                               time.time() - t0, out.prefill_time_s,
+                               out.decode_tokens_per_s,
+                               finish_reason=out.finish_reason)
+            else:
+                pay["choices"].append({"text": out.text, "index": idx,
+                                       "finish_reason": out.finish_reason})
+                pay["usage"]["completion_tokens"] += len(out.token_ids)
+            if out.logprobs is not None:
+                pay["choices"][idx]["logprobs"] = {
+                    "tokens": [e["token"] for e in out.logprobs],
+                    "token_logprobs": [e["logprob"]
+                                       for e in out.logprobs],
+                    "top_logprobs": [{t["token"]: t["logprob"]
+                                      for t in e["top"]}
+                                     for e in out.logprobs],
+                }
+        pay["timings"]["total_s"] = time.time() - t0
         return pay
 
     class _Row:

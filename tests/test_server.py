@@ -317,3 +317,24 @@ def test_request_validation_and_engine_errors():
         "prompt": "x", "max_tokens": 2, "strategy": "greedy",
         "stop_on_eos": False})
     assert ok.status_code == 200
+
+
+def test_n_completions():
+    """n>1 returns n independent choices (distinct under min_p seeds)."""
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy")
+    client = TestClient(app)
+    r = client.post("/v1/completions", json={
+        "prompt": "Once upon a time", "max_tokens": 6, "n": 3,
+        "strategy": "min_p", "seed": 5, "stop_on_eos": False}).json()
+    ch = r["choices"]
+    assert len(ch) == 3
+    assert [c["index"] for c in ch] == [0, 1, 2]
+    assert r["usage"]["completion_tokens"] == 18
+    assert len({c["text"] for c in ch}) >= 2  # seeds 5,6,7 differ
+    r2 = client.post("/v1/completions", json={
+        "prompt": "x", "max_tokens": 2, "n": 99})
+    assert r2.status_code == 422
