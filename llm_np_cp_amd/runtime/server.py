@@ -192,7 +192,11 @@ class BatchScheduler:
 def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
               dtype: str = "bf16", max_seq: int = 4096,
               max_batch: int = 8, kv_dtype: str = "bf16",
-              batch_window_ms: float = 4.0):
+              batch_window_ms: float = 4.0, _engine=None):
+    """``_engine``: test hook — a pre-built batch-capable engine
+    (GPUModel interface: prefill_row/decode_rows/compact_row/bt_*)
+    injected in place of the loaded model, so the continuous-batching
+    scheduler logic is CPU-testable against a deterministic fake."""
     from fastapi import Body, FastAPI
 
     import llm_np_cp_amd as L
@@ -201,7 +205,10 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
                                    dtype=dtype, max_seq=max_seq,
                                    kv_dtype=kv_dtype)
     can_batch = False
-    if max_batch > 1 and type(model).__name__ == "GPUModel":
+    if _engine is not None:
+        model = _engine
+        can_batch = max_batch > 1
+    elif max_batch > 1 and type(model).__name__ == "GPUModel":
         # rebuild with batch pools (load_model keeps its reference-parity
         # signature lean)
         from ..io.loader import random_weights, load_weights_numpy

@@ -338,3 +338,141 @@ def test_n_completions():
     r2 = client.post("/v1/completions", json={
         "prompt": "x", "max_tokens": 2, "n": 99})
     assert r2.status_code == 422
+
+
+class FakeBatchEngine:
+    """Deterministic CPU stand-in for GPUModel's continuous-batching
+    surface (prefill_row/decode_rows/compact_row/bt_*): row b emits
+    stream(key_b, 0), stream(key_b, 1), ... where key_b is derived from
+    its prompt ids — so the scheduler's bookkeeping (first-token
+    consumption, chunk accounting, retirement, compaction, mid-flight
+    joins) is verifiable token-for-token without a GPU."""
+
+    def __init__(self, max_batch=4, max_seq=256, step_delay=0.0):
+        import torch
+        self.max_batch, self.max_seq = max_batch, max_seq
+        self.step_delay = step_delay
+        B, S = max_batch, max_seq
+        self.bt_ring = torch.zeros(B, S + 16, dtype=torch.int32)
+        self.bt_nout = torch.zeros(B, dtype=torch.int32)
+        self._host_lens = [0] * B
+        self._keys = [0] * B
+        self._step = [0] * B
+
+    @staticmethod
+    def tok_at(key, i):
+        return 1 + (key * 31 + i * 7) % 200
+
+    @staticmethod
+    def key_of(ids):
+        return sum(int(x) for x in ids) % 1009
+
+    def prefill_row(self, slot, ids, greedy=True, min_p=0.1,
+                    temperature=1.0):
+        self._keys[slot] = self.key_of(ids)
+        self._host_lens[slot] = len(ids)
+        self.bt_nout[slot] = 1
+        self.bt_ring[slot, 0] = self.tok_at(self._keys[slot], 0)
+        self._step[slot] = 1
+
+    def decode_rows(self, B, n, greedy=True, min_p=0.1, temperature=1.0):
+        import numpy as np
+        if self.step_delay:
+            time.sleep(self.step_delay * n)
+        outs = []
+        for b in range(B):
+            ids = [self.tok_at(self._keys[b], self._step[b] + j)
+                   for j in range(n)]
+            base = int(self.bt_nout[b])
+            for j, t in enumerate(ids):
+                self.bt_ring[b, base + j] = t
+            self.bt_nout[b] = base + n
+            self._step[b] += n
+            self._host_lens[b] += n
+            outs.append(np.array(ids, dtype=np.int32))
+        return outs
+
+    def compact_row(self, dst, src):
+        if dst == src:
+            return
+        self.bt_ring[dst] = self.bt_ring[src].clone()
+        self.bt_nout[dst] = self.bt_nout[src].clone()
+        self._host_lens[dst] = self._host_lens[src]
+        self._keys[dst] = self._keys[src]
+        self._step[dst] = self._step[src]
+
+
+def _expected_text(prompt: str, m: int) -> str:
+    from llm_np_cp_amd.runtime.generate import ByteTokenizer
+    key = FakeBatchEngine.key_of(prompt.encode("utf-8"))
+    ids = [FakeBatchEngine.tok_at(key, i) for i in range(m)]
+    return ByteTokenizer().decode(ids)
+
+
+def test_scheduler_continuous_batching_token_exact_cpu():
+    """Continuous batching against the deterministic fake engine: every
+    request (grouped, deferred, or joined mid-flight) receives EXACTLY
+    its own stream's first max_tokens tokens; joins happen while the
+    long request is still decoding."""
+    fastapi = pytest.importorskip("fastapi")
+    import threading
+    from fastapi.testclient import TestClient
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy", max_seq=512,
+                    max_batch=3, batch_window_ms=20.0,
+                    _engine=FakeBatchEngine(3, 512, step_delay=0.002))
+    client = TestClient(app)
+
+    results = {}
+
+    def req(name, prompt, m):
+        results[name] = client.post("/v1/completions", json={
+            "prompt": prompt, "max_tokens": m, "strategy": "greedy",
+            "stop_on_eos": False}).json()
+
+    t_long = threading.Thread(target=req, args=("long", "a long story", 200))
+    t_long.start()
+    joiners = []
+    for i in range(4):
+        time.sleep(0.02)
+        t = threading.Thread(target=req,
+                             args=(i, f"short prompt {i}", 9 + i))
+        t.start()
+        joiners.append(t)
+    t_long.join(timeout=60)
+    for t in joiners:
+        t.join(timeout=60)
+
+    assert results["long"]["usage"]["completion_tokens"] == 200
+    assert results["long"]["choices"][0]["text"] == \
+        _expected_text("a long story", 200)
+    for i in range(4):
+        assert results[i]["usage"]["completion_tokens"] == 9 + i
+        assert results[i]["choices"][0]["text"] == \
+            _expected_text(f"short prompt {i}", 9 + i), i
+    stats = client.get("/stats").json()
+    assert stats["requests"] == 5
+    assert stats["joined_mid_flight"] >= 1, stats
+
+
+def test_scheduler_kv_room_retirement_cpu():
+    """A request whose decode would overflow the KV pool is retired at
+    the room boundary (chunk <= 0 path) with the tokens it did get."""
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy", max_seq=64,
+                    max_batch=2, batch_window_ms=1.0,
+                    _engine=FakeBatchEngine(2, 64, step_delay=0.0))
+    client = TestClient(app)
+    prompt = "pppp"
+    r = client.post("/v1/completions", json={
+        "prompt": prompt, "max_tokens": 500, "strategy": "greedy",
+        "stop_on_eos": False}).json()
+    n = r["usage"]["completion_tokens"]
+    assert 0 < n < 500
+    # room = max_seq - 2; prefill used len(prompt) positions
+    assert n <= 64 - 2 - len(prompt) + 16  # chunk granularity slack
+    assert r["choices"][0]["text"] == _expected_text(prompt, n)
