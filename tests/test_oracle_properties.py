@@ -1,0 +1,99 @@
+"""Property-based oracle invariants (hypothesis): random tiny
+architectures across all families must satisfy cache/mask/rope
+invariants that the fixed-shape tests only pin at preset shapes."""
+
+import numpy as np
+import pytest
+
+hypothesis = pytest.importorskip("hypothesis")
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+from llm_np_cp_amd.core.config import ModelConfig  # noqa: E402
+from llm_np_cp_amd.io.loader import random_weights  # noqa: E402
+from llm_np_cp_amd.models.numpy_ref import NumpyKVCache, NumpyModel  # noqa: E402
+
+
+def _draw_config(data):
+    model_type = data.draw(st.sampled_from(
+        ["llama", "gemma2", "mistral", "qwen2"]), label="model_type")
+    nh = data.draw(st.sampled_from([1, 2, 4]), label="nh")
+    kvh = data.draw(st.sampled_from(
+        [d for d in (1, 2, 4) if nh % d == 0 and d <= nh]), label="kvh")
+    hd = data.draw(st.sampled_from([4, 8, 16]), label="hd")
+    kw = dict(
+        model_type=model_type, vocab_size=64,
+        hidden_size=data.draw(st.sampled_from([8, 24]), label="h"),
+        intermediate_size=data.draw(st.sampled_from([16, 40]), label="i"),
+        num_hidden_layers=data.draw(st.sampled_from([1, 2, 3]), label="L"),
+        num_attention_heads=nh, num_key_value_heads=kvh, head_dim=hd,
+        rms_norm_eps=1e-6, rope_theta=10000.0,
+        max_position_embeddings=128, tie_word_embeddings=True,
+    )
+    if model_type == "gemma2":
+        kw.update(hidden_act="gelu_pytorch_tanh",
+                  query_pre_attn_scalar=float(hd),
+                  sliding_window=data.draw(st.sampled_from([3, 5, 64]),
+                                           label="win"),
+                  attn_logit_softcapping=data.draw(
+                      st.sampled_from([None, 50.0]), label="asc"),
+                  final_logit_softcapping=data.draw(
+                      st.sampled_from([None, 30.0]), label="fsc"))
+    else:
+        kw["hidden_act"] = "silu"
+        if model_type == "mistral":
+            kw["sliding_window"] = data.draw(
+                st.sampled_from([3, 5, 64]), label="win")
+        if model_type == "qwen2":
+            kw["attention_bias"] = True
+    return ModelConfig(**kw)
+
+
+@settings(max_examples=15, deadline=None)
+@given(data=st.data())
+def test_incremental_decode_equals_prefill_any_shape(data):
+    """The defining KV-cache invariant at arbitrary (nh, kvh, hd, H, I,
+    L, window, softcap, bias) combinations: feeding a sequence in any
+    prefill/decode split produces the same last-position logits as one
+    full prefill."""
+    cfg = _draw_config(data)
+    w = random_weights(cfg, seed=3)
+    m = NumpyModel(cfg, dict(w))
+    rng = np.random.default_rng(7)
+    n = 10
+    ids = rng.integers(0, cfg.vocab_size, size=n)
+
+    full = m.forward(ids, NumpyKVCache(cfg, 32), 0)
+
+    split = data.draw(st.integers(min_value=1, max_value=n - 1),
+                      label="split")
+    c = NumpyKVCache(cfg, 32)
+    m.forward(ids[:split], c, 0)
+    out = None
+    for t in range(split, n):
+        out = m.forward(ids[t:t + 1], c, t)
+    np.testing.assert_allclose(out[0], full[-1], rtol=2e-4, atol=2e-5)
+
+
+@settings(max_examples=10, deadline=None)
+@given(data=st.data())
+def test_window_geq_seq_is_full_attention_any_shape(data):
+    """sliding_window >= sequence length must not change any logits,
+    for any architecture that windows (gemma2 alternating / mistral
+    all-layer)."""
+    cfg = _draw_config(data)
+    if cfg.sliding_window is None:
+        cfg = ModelConfig(**{**cfg.__dict__, "model_type": "mistral",
+                             "hidden_act": "silu", "sliding_window": 64,
+                             "layer_types": [],
+                             "attn_logit_softcapping": None,
+                             "final_logit_softcapping": None,
+                             "query_pre_attn_scalar": None,
+                             "attention_bias": False})
+    cfg.sliding_window = 64          # >= n below
+    w = random_weights(cfg, seed=5)
+    ids = np.random.default_rng(1).integers(0, cfg.vocab_size, size=9)
+    a = NumpyModel(cfg, dict(w)).forward(ids, NumpyKVCache(cfg, 32), 0)
+    cfg2 = ModelConfig(**{**cfg.__dict__, "sliding_window": None,
+                          "layer_types": []})
+    b = NumpyModel(cfg2, dict(w)).forward(ids, NumpyKVCache(cfg2, 32), 0)
+    np.testing.assert_allclose(a, b, rtol=1e-5, atol=1e-6)
