@@ -189,6 +189,21 @@ class BatchScheduler:
                 self.q.put(p)
 
 
+def chat_prompt(tok, messages) -> str:
+    """Assemble chat messages into a prompt: the tokenizer's own
+    chat_template when the checkpoint ships one (AutoTokenizer), else a
+    neutral <|role|> fallback usable with any tokenizer."""
+    msgs = [{"role": m["role"], "content": m["content"]}
+            if isinstance(m, dict) else
+            {"role": m.role, "content": m.content} for m in messages]
+    template = getattr(tok, "chat_template", None)
+    if template and hasattr(tok, "apply_chat_template"):
+        return tok.apply_chat_template(msgs, tokenize=False,
+                                       add_generation_prompt=True)
+    lines = [f"<|{m['role']}|>\n{m['content']}" for m in msgs]
+    return "\n".join(lines) + "\n<|assistant|>\n"
+
+
 def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
               dtype: str = "bf16", max_seq: int = 4096,
               max_batch: int = 8, kv_dtype: str = "bf16",
@@ -449,22 +464,11 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
             from fastapi import HTTPException
             raise HTTPException(status_code=400, detail=str(e))
 
-    def _chat_prompt(messages) -> str:
-        msgs = [{"role": m["role"], "content": m["content"]}
-                if isinstance(m, dict) else
-                {"role": m.role, "content": m.content} for m in messages]
-        template = getattr(tok, "chat_template", None)
-        if template and hasattr(tok, "apply_chat_template"):
-            return tok.apply_chat_template(msgs, tokenize=False,
-                                           add_generation_prompt=True)
-        # checkpoint-agnostic fallback (byte tokenizer / no template)
-        lines = [f"<|{m['role']}|>\n{m['content']}" for m in msgs]
-        return "\n".join(lines) + "\n<|assistant|>\n"
-
     @app.post("/v1/chat/completions")
     def chat_completions(req: ChatCompletionRequest = Body(...)):
         creq = CompletionRequest(
-            prompt=_chat_prompt(req.messages), max_tokens=req.max_tokens,
+            prompt=chat_prompt(tok, req.messages),
+            max_tokens=req.max_tokens,
             temperature=req.temperature, min_p=req.min_p,
             strategy=req.strategy, seed=req.seed,
             stop_on_eos=req.stop_on_eos, stream=req.stream,

@@ -476,3 +476,36 @@ def test_scheduler_kv_room_retirement_cpu():
     # room = max_seq - 2; prefill used len(prompt) positions
     assert n <= 64 - 2 - len(prompt) + 16  # chunk granularity slack
     assert r["choices"][0]["text"] == _expected_text(prompt, n)
+
+
+def test_chat_prompt_template_and_fallback():
+    """chat_prompt uses the tokenizer's chat template when present
+    (AutoTokenizer-style apply_chat_template), else the role-tag
+    fallback."""
+    from llm_np_cp_amd.runtime.server import chat_prompt
+
+    msgs = [{"role": "system", "content": "S"},
+            {"role": "user", "content": "U"}]
+
+    class TemplTok:
+        chat_template = "{{ bos }}..."
+
+        def apply_chat_template(self, m, tokenize, add_generation_prompt):
+            assert not tokenize and add_generation_prompt
+            return "".join(f"[{x['role']}]{x['content']}" for x in m) + "[assistant]"
+
+    out = chat_prompt(TemplTok(), msgs)
+    assert out == "[system]S[user]U[assistant]"
+
+    class PlainTok:
+        pass
+
+    out = chat_prompt(PlainTok(), msgs)
+    assert out == "<|system|>\nS\n<|user|>\nU\n<|assistant|>\n"
+
+    # pydantic-message objects work too
+    from llm_np_cp_amd.runtime.server import ChatMessage
+    if ChatMessage is not None:
+        out2 = chat_prompt(PlainTok(),
+                           [ChatMessage(role="user", content="hi")])
+        assert out2 == "<|user|>\nhi\n<|assistant|>\n"
