@@ -30,6 +30,16 @@ def main(argv=None):
     ap.add_argument("--temperature", type=float, default=1.0)
     ap.add_argument("--seed", type=int, default=None)
     ap.add_argument("--max-seq", type=int, default=4096)
+    ap.add_argument("--stop", action="append", default=None,
+                    help="stop string (repeatable); excluded from output")
+    ap.add_argument("--draft", default=None,
+                    help="draft model (dir/preset) — enables speculative "
+                         "decoding (greedy mode is token-identical to the "
+                         "target's own decode)")
+    ap.add_argument("--draft-dtype", default=None,
+                    help="draft weight dtype (default: same as --dtype)")
+    ap.add_argument("--spec-k", type=int, default=4,
+                    help="draft tokens per verify pass")
     ap.add_argument("--no-cache", action="store_true",
                     help="stateless re-prefill each step (debug mode, "
                          "reference use_cache=False path)")
@@ -43,8 +53,23 @@ def main(argv=None):
     params = L.SamplingParams(strategy=args.strategy, min_p=args.min_p,
                               temperature=args.temperature, seed=args.seed)
     t0 = time.time()
-    out = L.generate(args.prompt, tok, model, max_tokens=args.max_tokens,
-                     params=params, use_cache=not args.no_cache, stream=True)
+    if args.draft is not None:
+        _, draft, _ = L.load_model(args.draft, backend=args.backend,
+                                   dtype=args.draft_dtype or args.dtype,
+                                   max_seq=args.max_seq)
+        out = L.generate_speculative(
+            args.prompt, tok, draft, model, max_tokens=args.max_tokens,
+            k=args.spec_k, params=params,
+            on_token=lambda s: (sys.stdout.write(s), sys.stdout.flush()))
+        s = out.spec_stats
+        print(f"\n[speculative: {s['accepted']}/{s['proposed']} drafts "
+              f"accepted over {s['verify_passes']} verify passes]",
+              file=sys.stderr)
+    else:
+        out = L.generate(args.prompt, tok, model,
+                         max_tokens=args.max_tokens, params=params,
+                         use_cache=not args.no_cache, stream=True,
+                         stop=args.stop)
     dt = time.time() - t0
     print(f"\n[{len(out.token_ids)} tokens in {dt:.2f}s; "
           f"prefill {out.prefill_time_s * 1e3:.0f} ms, "

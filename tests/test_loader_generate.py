@@ -392,3 +392,18 @@ def test_logprobs_via_server():
     lp = r["choices"][0]["logprobs"]
     assert len(lp["tokens"]) == 3 == len(lp["token_logprobs"])
     assert all(len(d) <= 2 for d in lp["top_logprobs"])
+
+
+def test_cli_speculative_and_stop(capsys):
+    from llm_np_cp_amd.__main__ import main
+
+    main(["Once upon a time", "--model", "tiny-llama", "--backend",
+          "numpy", "--draft", "tiny-llama", "--strategy", "greedy",
+          "--max-tokens", "8"])
+    err = capsys.readouterr().err
+    assert "speculative:" in err and "verify passes" in err
+
+    main(["abc", "--model", "tiny-llama", "--backend", "numpy",
+          "--strategy", "greedy", "--max-tokens", "6",
+          "--stop", "ZXQNEVER"])
+    assert "6 tokens" in capsys.readouterr().err
