@@ -186,3 +186,122 @@ def test_real_presets_tp_divisibility():
             assert cfg.num_key_value_heads % tp == 0, (name, tp)
             assert cfg.intermediate_size % tp == 0, (name, tp)
             assert cfg.vocab_size % tp == 0, (name, tp)
+
+
+# ---------- real hub config.json fixtures (verbatim field sets) ----------
+
+LLAMA_32_1B_CONFIG = {
+    "architectures": ["LlamaForCausalLM"], "attention_bias": False,
+    "attention_dropout": 0.0, "bos_token_id": 128000,
+    "eos_token_id": 128001, "head_dim": 64, "hidden_act": "silu",
+    "hidden_size": 2048, "initializer_range": 0.02,
+    "intermediate_size": 8192, "max_position_embeddings": 131072,
+    "mlp_bias": False, "model_type": "llama", "num_attention_heads": 32,
+    "num_hidden_layers": 16, "num_key_value_heads": 8,
+    "pretraining_tp": 1, "rms_norm_eps": 1e-05,
+    "rope_scaling": {"factor": 32.0, "high_freq_factor": 4.0,
+                     "low_freq_factor": 1.0,
+                     "original_max_position_embeddings": 8192,
+                     "rope_type": "llama3"},
+    "rope_theta": 500000.0, "tie_word_embeddings": True,
+    "torch_dtype": "bfloat16", "use_cache": True, "vocab_size": 128256,
+}
+
+GEMMA_2_9B_CONFIG = {
+    "architectures": ["Gemma2ForCausalLM"],
+    "attention_bias": False, "attention_dropout": 0.0,
+    "attn_logit_softcapping": 50.0, "bos_token_id": 2,
+    "cache_implementation": "hybrid", "eos_token_id": 1,
+    "final_logit_softcapping": 30.0, "head_dim": 256,
+    "hidden_act": "gelu_pytorch_tanh",
+    "hidden_activation": "gelu_pytorch_tanh", "hidden_size": 3584,
+    "initializer_range": 0.02, "intermediate_size": 14336,
+    "max_position_embeddings": 8192, "model_type": "gemma2",
+    "num_attention_heads": 16, "num_hidden_layers": 42,
+    "num_key_value_heads": 8, "pad_token_id": 0,
+    "query_pre_attn_scalar": 256, "rms_norm_eps": 1e-06,
+    "rope_theta": 10000.0, "sliding_window": 4096,
+    "torch_dtype": "float32", "use_cache": True, "vocab_size": 256000,
+}
+
+MIXTRAL_8X7B_CONFIG = {
+    "architectures": ["MixtralForCausalLM"], "attention_dropout": 0.0,
+    "bos_token_id": 1, "eos_token_id": 2, "hidden_act": "silu",
+    "hidden_size": 4096, "initializer_range": 0.02,
+    "intermediate_size": 14336, "max_position_embeddings": 32768,
+    "model_type": "mixtral", "num_attention_heads": 32,
+    "num_experts_per_tok": 2, "num_hidden_layers": 32,
+    "num_key_value_heads": 8, "num_local_experts": 8,
+    "output_router_logits": False, "rms_norm_eps": 1e-05,
+    "rope_theta": 1000000.0, "router_aux_loss_coef": 0.02,
+    "sliding_window": None, "tie_word_embeddings": False,
+    "torch_dtype": "bfloat16", "use_cache": True, "vocab_size": 32000,
+}
+
+QWEN_25_7B_CONFIG = {
+    "architectures": ["Qwen2ForCausalLM"], "attention_dropout": 0.0,
+    "bos_token_id": 151643, "eos_token_id": 151643,
+    "hidden_act": "silu", "hidden_size": 3584,
+    "initializer_range": 0.02, "intermediate_size": 18944,
+    "max_position_embeddings": 131072, "max_window_layers": 28,
+    "model_type": "qwen2", "num_attention_heads": 28,
+    "num_hidden_layers": 28, "num_key_value_heads": 4,
+    "rms_norm_eps": 1e-06, "rope_theta": 1000000.0,
+    "sliding_window": None, "tie_word_embeddings": False,
+    "torch_dtype": "bfloat16", "use_cache": True,
+    "use_sliding_window": False, "vocab_size": 152064,
+}
+
+
+def test_real_llama32_config_parses():
+    """An actual Llama-3.2-1B hub config.json (extra keys and all)
+    parses to the right architecture, and llama3 rope_scaling survives."""
+    from llm_np_cp_amd.core.config import ModelConfig
+
+    cfg = ModelConfig.from_hf_dict(LLAMA_32_1B_CONFIG)
+    assert (cfg.hidden_size, cfg.num_hidden_layers) == (2048, 16)
+    assert (cfg.num_attention_heads, cfg.num_key_value_heads) == (32, 8)
+    assert cfg.head_dim == 64 and cfg.vocab_size == 128256
+    assert cfg.tie_word_embeddings and not cfg.attention_bias
+    assert cfg.rope_scaling["rope_type"] == "llama3"
+    f = cfg.rope_inv_freq()
+    assert f.shape == (32,)
+    assert not cfg.is_moe and cfg.attn_scale == 64 ** -0.5
+
+
+def test_real_gemma2_config_parses():
+    from llm_np_cp_amd.core.config import ModelConfig
+
+    cfg = ModelConfig.from_hf_dict(GEMMA_2_9B_CONFIG)
+    assert cfg.model_type == "gemma2" and cfg.head_dim == 256
+    assert cfg.attn_logit_softcapping == 50.0
+    assert cfg.final_logit_softcapping == 30.0
+    assert cfg.attn_scale == 256 ** -0.5  # query_pre_attn_scalar
+    assert cfg.embed_scale == pytest.approx(3584 ** 0.5)
+    assert cfg.hidden_act == "gelu_pytorch_tanh"
+    # alternating sliding layers, layer 0 sliding
+    assert cfg.is_sliding(0) and not cfg.is_sliding(1)
+    assert sum(cfg.is_sliding(i) for i in range(42)) == 21
+
+
+def test_real_mixtral_config_parses():
+    from llm_np_cp_amd.core.config import ModelConfig
+
+    cfg = ModelConfig.from_hf_dict(MIXTRAL_8X7B_CONFIG)
+    assert cfg.is_moe
+    assert (cfg.num_local_experts, cfg.num_experts_per_tok) == (8, 2)
+    assert not cfg.tie_word_embeddings
+    assert cfg.sliding_window is None
+    assert not any(cfg.is_sliding(i) for i in range(32))
+    assert cfg.head_dim == 4096 // 32  # derived (no head_dim key)
+
+
+def test_real_qwen25_config_parses():
+    from llm_np_cp_amd.core.config import ModelConfig
+
+    cfg = ModelConfig.from_hf_dict(QWEN_25_7B_CONFIG)
+    assert cfg.attention_bias  # HF Qwen2Attention hardwires qkv bias
+    assert cfg.head_dim == 3584 // 28
+    assert cfg.num_kv_groups == 7
+    # use_sliding_window false => sliding_window None => no windowing
+    assert not any(cfg.is_sliding(i) for i in range(28))
