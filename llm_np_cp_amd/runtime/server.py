@@ -71,6 +71,11 @@ except ImportError:  # pragma: no cover - serving is optional
     ChatCompletionRequest = None
 
 
+class ClientDisconnected(Exception):
+    """Raised inside the token callback to abort generation for a
+    streaming client that went away (SSE cancellation)."""
+
+
 @dataclass
 class _Pending:
     req: "CompletionRequest"
@@ -432,23 +437,39 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
         """Run req through the scheduler, yielding one SSE event per
         decoded token piece, then a final [DONE].  Token callbacks fire
         on the scheduler thread; all of them happen-before done.set(),
-        so `done and empty` is a safe termination check."""
+        so `done and empty` is a safe termination check.
+
+        CANCELLATION: when the client disconnects, Starlette closes the
+        generator; the finally block sets the cancel flag and the next
+        token callback raises inside generate() on the scheduler thread
+        — generation stops instead of decoding to max_tokens for a
+        reader that is gone."""
         from fastapi.responses import StreamingResponse
 
         q: "queue.Queue[str]" = queue.Queue()
-        p = sched.submit_async(req, on_token=q.put)
+        cancelled = threading.Event()
+
+        def push(piece: str):
+            if cancelled.is_set():
+                raise ClientDisconnected()
+            q.put(piece)
+
+        p = sched.submit_async(req, on_token=push)
 
         def gen():
-            while not (p.done.is_set() and q.empty()):
-                try:
-                    piece = q.get(timeout=0.05)
-                except queue.Empty:
-                    continue
-                yield "data: " + json.dumps(chunk_of(piece)) + "\n\n"
-            if p.error is not None:
-                yield ("data: " + json.dumps({"error": str(p.error)})
-                       + "\n\n")
-            yield "data: [DONE]\n\n"
+            try:
+                while not (p.done.is_set() and q.empty()):
+                    try:
+                        piece = q.get(timeout=0.05)
+                    except queue.Empty:
+                        continue
+                    yield "data: " + json.dumps(chunk_of(piece)) + "\n\n"
+                if p.error is not None:
+                    yield ("data: " + json.dumps({"error": str(p.error)})
+                           + "\n\n")
+                yield "data: [DONE]\n\n"
+            finally:
+                cancelled.set()
 
         return StreamingResponse(gen(), media_type="text/event-stream")
 

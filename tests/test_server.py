@@ -509,3 +509,68 @@ def test_chat_prompt_template_and_fallback():
         out2 = chat_prompt(PlainTok(),
                            [ChatMessage(role="user", content="hi")])
         assert out2 == "<|user|>\nhi\n<|assistant|>\n"
+
+
+def test_streaming_disconnect_cancels_generation():
+    """A client that drops its SSE connection mid-generation aborts the
+    scheduler-side generate (ClientDisconnected raised in the token
+    callback) instead of decoding to max_tokens.  Needs a REAL server:
+    the TestClient ASGI shim buffers the whole response, so this runs
+    uvicorn on a loopback port and hard-closes the socket."""
+    pytest.importorskip("fastapi")
+    uvicorn = pytest.importorskip("uvicorn")
+    import http.client
+    import json as _json
+    import socket
+    import threading
+
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy")
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    server = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1",
+                                           port=port, log_level="error"))
+    th = threading.Thread(target=server.run, daemon=True)
+    th.start()
+    for _ in range(200):
+        if server.started:
+            break
+        time.sleep(0.05)
+    assert server.started
+    try:
+        conn = http.client.HTTPConnection("127.0.0.1", port, timeout=60)
+        body = _json.dumps({"prompt": "Once upon a time",
+                            "max_tokens": 200000, "strategy": "greedy",
+                            "stop_on_eos": False, "stream": True})
+        conn.request("POST", "/v1/completions", body=body,
+                     headers={"Content-Type": "application/json"})
+        resp = conn.getresponse()
+        assert resp.status == 200
+        buf = b""
+        while buf.count(b"data: ") < 3:
+            chunk = resp.read(64)
+            assert chunk, "stream ended before 3 events"
+            buf += chunk
+        conn.close()  # hard disconnect mid-stream (~200k tokens left)
+
+        # the scheduler thread must become free promptly; a follow-up
+        # request would otherwise queue behind minutes of decode
+        t0 = time.time()
+        c2 = http.client.HTTPConnection("127.0.0.1", port, timeout=60)
+        c2.request("POST", "/v1/completions",
+                   body=_json.dumps({"prompt": "hi", "max_tokens": 3,
+                                     "strategy": "greedy",
+                                     "stop_on_eos": False}),
+                   headers={"Content-Type": "application/json"})
+        r2 = c2.getresponse()
+        assert r2.status == 200
+        out = _json.loads(r2.read())
+        assert out["usage"]["completion_tokens"] == 3
+        assert time.time() - t0 < 60
+        c2.close()
+    finally:
+        server.should_exit = True
+        th.join(timeout=15)
