@@ -246,8 +246,28 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
     eos_set = (set() if eos is None else {int(eos)}
                if isinstance(eos, (int, float)) else {int(e) for e in eos})
 
+    # Prometheus metrics (SURVEY §5 observability); per-app registry so
+    # repeated build_app calls (tests) never collide on timeseries names
+    try:
+        from prometheus_client import (CollectorRegistry, Counter,
+                                       Histogram)
+        metrics_reg = CollectorRegistry()
+        m_requests = Counter("llm_requests_total",
+                             "completed requests", ["endpoint"],
+                             registry=metrics_reg)
+        m_tokens = Counter("llm_completion_tokens_total",
+                           "tokens generated", registry=metrics_reg)
+        m_latency = Histogram("llm_request_seconds",
+                              "end-to-end request latency",
+                              registry=metrics_reg)
+    except ImportError:  # pragma: no cover - metrics are optional
+        metrics_reg = None
+
     def _payload(req, ids, text, total_s, prefill_s, tps,
                  finish_reason=None) -> dict:
+        if metrics_reg is not None:
+            m_tokens.inc(len(ids))
+            m_latency.observe(total_s)
         if finish_reason is None:
             hit_eos = bool(ids) and int(ids[-1]) in eos_set
             finish_reason = "stop" if hit_eos else "length"
@@ -433,6 +453,16 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
     def stats():
         return dict(sched.stats)
 
+    @app.get("/metrics")
+    def metrics():
+        from fastapi import HTTPException, Response
+        if metrics_reg is None:
+            raise HTTPException(status_code=404,
+                                detail="prometheus_client not installed")
+        from prometheus_client import CONTENT_TYPE_LATEST, generate_latest
+        return Response(generate_latest(metrics_reg),
+                        media_type=CONTENT_TYPE_LATEST)
+
     def _sse(req, chunk_of):
         """Run req through the scheduler, yielding one SSE event per
         decoded token piece, then a final [DONE].  Token callbacks fire
@@ -475,6 +505,8 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
 
     @app.post("/v1/completions")
     def completions(req: CompletionRequest = Body(...)):
+        if metrics_reg is not None:
+            m_requests.labels(endpoint="completions").inc()
         if req.stream:
             return _sse(req, lambda piece: {
                 "object": "text_completion.chunk", "model": model_name,
@@ -487,6 +519,8 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
 
     @app.post("/v1/chat/completions")
     def chat_completions(req: ChatCompletionRequest = Body(...)):
+        if metrics_reg is not None:
+            m_requests.labels(endpoint="chat").inc()
         creq = CompletionRequest(
             prompt=chat_prompt(tok, req.messages),
             max_tokens=req.max_tokens,

@@ -574,3 +574,27 @@ def test_streaming_disconnect_cancels_generation():
     finally:
         server.should_exit = True
         th.join(timeout=15)
+
+
+def test_metrics_endpoint_prometheus():
+    fastapi = pytest.importorskip("fastapi")
+    pytest.importorskip("prometheus_client")
+    from fastapi.testclient import TestClient
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy")
+    client = TestClient(app)
+    for _ in range(2):
+        client.post("/v1/completions", json={
+            "prompt": "x", "max_tokens": 3, "strategy": "greedy",
+            "stop_on_eos": False})
+    r = client.get("/metrics")
+    assert r.status_code == 200
+    body = r.text
+    assert 'llm_requests_total{endpoint="completions"} 2.0' in body
+    assert "llm_completion_tokens_total 6.0" in body
+    assert "llm_request_seconds_count 2.0" in body
+    # a second app instance must not collide on timeseries (per-app
+    # registry)
+    app2 = build_app("tiny-llama", backend="numpy")
+    assert TestClient(app2).get("/metrics").status_code == 200
