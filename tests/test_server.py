@@ -598,3 +598,48 @@ def test_metrics_endpoint_prometheus():
     # registry)
     app2 = build_app("tiny-llama", backend="numpy")
     assert TestClient(app2).get("/metrics").status_code == 200
+
+
+@pytest.mark.parametrize("seed", [0, 1, 2])
+def test_scheduler_fuzz_random_schedules(seed):
+    """Randomized request schedules (sizes, lengths, arrival times)
+    against the deterministic fake engine: whatever grouping, joining,
+    retirement and compaction the scheduler chooses, every request must
+    receive exactly its own stream's first max_tokens tokens."""
+    fastapi = pytest.importorskip("fastapi")
+    import random
+    import threading
+    from fastapi.testclient import TestClient
+    from llm_np_cp_amd.runtime.server import build_app
+
+    rng = random.Random(seed)
+    mb = rng.choice([2, 3, 4])
+    app = build_app("tiny-llama", backend="numpy", max_seq=512,
+                    max_batch=mb, batch_window_ms=rng.choice([1.0, 15.0]),
+                    _engine=FakeBatchEngine(mb, 512, step_delay=0.001))
+    client = TestClient(app)
+
+    n_req = rng.randint(4, 9)
+    plan = [(f"prompt {seed}-{i}-" + "x" * rng.randint(0, 20),
+             rng.randint(1, 60), rng.uniform(0, 0.03))
+            for i in range(n_req)]
+    results = {}
+
+    def fire(i, prompt, m, delay):
+        time.sleep(delay)
+        results[i] = client.post("/v1/completions", json={
+            "prompt": prompt, "max_tokens": m, "strategy": "greedy",
+            "stop_on_eos": False}).json()
+
+    threads = [threading.Thread(target=fire, args=(i, p, m, d))
+               for i, (p, m, d) in enumerate(plan)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=120)
+
+    for i, (prompt, m, _) in enumerate(plan):
+        body = results[i]
+        assert body["usage"]["completion_tokens"] == m, (i, body)
+        assert body["choices"][0]["text"] == _expected_text(prompt, m), i
+    assert client.get("/stats").json()["requests"] == n_req
