@@ -643,3 +643,53 @@ def test_scheduler_fuzz_random_schedules(seed):
         assert body["usage"]["completion_tokens"] == m, (i, body)
         assert body["choices"][0]["text"] == _expected_text(prompt, m), i
     assert client.get("/stats").json()["requests"] == n_req
+
+
+def test_scheduler_eos_retirement_cpu():
+    """EOS retirement in the lockstep group: rows whose stream emits the
+    config's eos token (tiny-llama default eos_token_id=2) retire at its
+    FIRST occurrence with the eos included, exactly as single-sequence
+    generate would truncate."""
+    fastapi = pytest.importorskip("fastapi")
+    import threading
+    from fastapi.testclient import TestClient
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy", max_seq=512,
+                    max_batch=3, batch_window_ms=25.0,
+                    _engine=FakeBatchEngine(3, 512, step_delay=0.001))
+    client = TestClient(app)
+
+    # find prompts whose stream does / does not hit eos=2 within 80
+    def stream(prompt, m):
+        key = FakeBatchEngine.key_of(prompt.encode("utf-8"))
+        return [FakeBatchEngine.tok_at(key, i) for i in range(m)]
+
+    with_eos = next(f"p{j}" for j in range(500)
+                    if 2 in stream(f"p{j}", 80))
+    no_eos = next(f"q{j}" for j in range(500)
+                  if 2 not in stream(f"q{j}", 80))
+    ids = stream(with_eos, 80)
+    cut = ids.index(2)
+
+    results = {}
+
+    def fire(name, prompt, eos_stop):
+        results[name] = client.post("/v1/completions", json={
+            "prompt": prompt, "max_tokens": 80, "strategy": "greedy",
+            "stop_on_eos": eos_stop}).json()
+
+    ts = [threading.Thread(target=fire, args=("a", with_eos, True)),
+          threading.Thread(target=fire, args=("b", no_eos, True))]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=60)
+
+    assert results["a"]["usage"]["completion_tokens"] == cut + 1
+    assert results["a"]["choices"][0]["finish_reason"] == "stop"
+    assert results["b"]["usage"]["completion_tokens"] == 80
+    assert results["b"]["choices"][0]["finish_reason"] == "length"
+    from llm_np_cp_amd.runtime.generate import ByteTokenizer
+    assert results["a"]["choices"][0]["text"] == \
+        ByteTokenizer().decode(ids[:cut + 1])
