@@ -83,3 +83,20 @@ def test_hip_wrapper_shape_validation_raises_before_launch():
     with pytest.raises(AssertionError):
         ho.moe_route(x, s, wg, 4, 2, torch.zeros(1, dtype=torch.int32),
                      torch.zeros(8))
+
+
+def test_examples_run_on_cpu(tmp_path):
+    """examples/*.py (the CPU-capable ones) run end to end via
+    subprocess against the tiny preset."""
+    import subprocess
+    import sys as _sys
+    import os as _os
+
+    root = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+    for script in ("basic_generate.py", "speculative_decode.py"):
+        r = subprocess.run(
+            [_sys.executable, _os.path.join(root, "examples", script),
+             "tiny-llama"],
+            capture_output=True, text=True, timeout=180)
+        assert r.returncode == 0, (script, r.stderr[-800:])
+        assert r.stdout.strip(), script
