@@ -117,6 +117,42 @@ def hf_weight_shapes(cfg: ModelConfig) -> Dict[str, Tuple[int, ...]]:
     return shapes
 
 
+def validate_weights(cfg: ModelConfig, w) -> None:
+    """Fail with an ACTIONABLE error when a checkpoint does not match
+    the architecture: lists every missing tensor name and every shape
+    mismatch instead of a bare KeyError deep inside the forward pass
+    (the reference swallowed loader errors entirely —
+    ``llama3.2_model.py:1063``).  Extra tensors are ignored (real
+    checkpoints ship rotary buffers etc.).  Lazy weight mappings that
+    expose their own ``shapes`` table (``LazyRandomWeights``) are
+    trusted — materializing every tensor just to check it would double
+    the load cost."""
+    expect = hf_weight_shapes(cfg)
+    own = getattr(w, "shapes", None)
+    if own is not None:
+        missing = [k for k in expect if k not in own]
+        bad = [(k, tuple(own[k]), expect[k]) for k in expect
+               if k in own and tuple(own[k]) != expect[k]]
+    else:
+        missing = [k for k in expect if k not in w]
+        bad = [(k, tuple(np.shape(w[k])), expect[k]) for k in expect
+               if k in w and tuple(np.shape(w[k])) != expect[k]]
+    msgs = []
+    if missing:
+        shown = ", ".join(missing[:6])
+        more = f" (+{len(missing) - 6} more)" if len(missing) > 6 else ""
+        msgs.append(f"{len(missing)} missing tensors: {shown}{more}")
+    if bad:
+        shown = "; ".join(f"{k}: got {g}, want {e}" for k, g, e in bad[:4])
+        more = f" (+{len(bad) - 4} more)" if len(bad) > 4 else ""
+        msgs.append(f"{len(bad)} shape mismatches: {shown}{more}")
+    if msgs:
+        raise ValueError(
+            f"checkpoint does not match {cfg.model_type} config "
+            f"(hidden={cfg.hidden_size}, layers={cfg.num_hidden_layers}): "
+            + "; ".join(msgs))
+
+
 def random_weights(cfg: ModelConfig, seed: int = 0,
                    scale: float = 0.02) -> Dict[str, np.ndarray]:
     """Random-init fp32 weights with HF names (norm gammas ~= identity)."""

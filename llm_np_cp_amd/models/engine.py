@@ -160,6 +160,9 @@ class GPUModel:
         return int(min(cfg.max_position_embeddings, n))
 
     def _upload_weights(self, w: Dict[str, np.ndarray]):
+        from ..io.loader import validate_weights
+
+        validate_weights(self.config, w)  # actionable error vs KeyError
         cfg, dev = self.config, self.device
         tp, r = self.world, self.rank
         gemma = cfg.model_type == "gemma2"

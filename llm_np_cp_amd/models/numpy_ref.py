@@ -80,6 +80,9 @@ class NumpyModel:
     """
 
     def __init__(self, config: ModelConfig, weights: Dict[str, np.ndarray]):
+        from ..io.loader import validate_weights
+
+        validate_weights(config, weights)  # actionable error vs KeyError
         self.config = config
         self.w = weights
         if "lm_head.weight" not in self.w and config.tie_word_embeddings:

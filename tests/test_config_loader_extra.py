@@ -305,3 +305,41 @@ def test_real_qwen25_config_parses():
     assert cfg.num_kv_groups == 7
     # use_sliding_window false => sliding_window None => no windowing
     assert not any(cfg.is_sliding(i) for i in range(28))
+
+
+def test_validate_weights_actionable_errors():
+    """Missing tensors and shape mismatches produce a clear ValueError
+    naming the offenders (vs a KeyError deep in the forward)."""
+    import numpy as np
+    from llm_np_cp_amd.core.config import preset_config
+    from llm_np_cp_amd.io.loader import (LazyRandomWeights,
+                                         random_weights, validate_weights)
+    from llm_np_cp_amd.models.numpy_ref import NumpyModel
+
+    cfg = preset_config("tiny-llama")
+    w = random_weights(cfg, seed=0)
+    validate_weights(cfg, w)  # complete: no raise
+    validate_weights(cfg, LazyRandomWeights(cfg))  # lazy table trusted
+
+    w2 = dict(w)
+    del w2["model.layers.1.mlp.down_proj.weight"]
+    with pytest.raises(ValueError, match="missing.*down_proj"):
+        NumpyModel(cfg, w2)
+
+    w3 = dict(w)
+    w3["model.norm.weight"] = np.zeros(7, dtype=np.float32)
+    with pytest.raises(ValueError, match="shape mismatch.*model.norm"):
+        validate_weights(cfg, w3)
+
+    # extra tensors are fine (real checkpoints ship rotary buffers)
+    w4 = dict(w)
+    w4["model.rotary_emb.inv_freq"] = np.zeros(8, dtype=np.float32)
+    validate_weights(cfg, w4)
+
+    # mixtral naming goes through the same table
+    mcfg = preset_config("tiny-mixtral")
+    mw = random_weights(mcfg, seed=1)
+    validate_weights(mcfg, mw)
+    del mw["model.layers.0.block_sparse_moe.experts.2.w3.weight"]
+    with pytest.raises(ValueError, match="experts.2.w3"):
+        validate_weights(mcfg, mw)
