@@ -260,6 +260,21 @@ PRESETS = {
         tie_word_embeddings=False, bos_token_id=128000,
         eos_token_id=128001,
     ),
+    # Llama-3.1-70B: the 288 GB HBM3E sizing case — fp8 weights are
+    # ~70 GB, so TP=1 fits with >200 GB left for KV; bf16 (141 GB)
+    # also fits on ONE MI355X (impossible on 80-192 GB parts)
+    "llama-3.1-70b": dict(
+        model_type="llama", vocab_size=128256, hidden_size=8192,
+        intermediate_size=28672, num_hidden_layers=80,
+        num_attention_heads=64, num_key_value_heads=8, head_dim=128,
+        rms_norm_eps=1e-5, rope_theta=500000.0,
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192},
+        max_position_embeddings=131072, hidden_act="silu",
+        tie_word_embeddings=False, bos_token_id=128000,
+        eos_token_id=128001,
+    ),
     # Qwen-2.5: Llama-family arch + qkv bias + untied lm_head
     # (shape table from the public Qwen/Qwen2.5-7B config.json)
     "qwen2.5-7b": dict(

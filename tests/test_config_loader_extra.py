@@ -343,3 +343,24 @@ def test_validate_weights_actionable_errors():
     del mw["model.layers.0.block_sparse_moe.experts.2.w3.weight"]
     with pytest.raises(ValueError, match="experts.2.w3"):
         validate_weights(mcfg, mw)
+
+
+def test_llama_70b_preset_shapes_and_memory_math():
+    """The 288 GB sizing preset: weight-byte math confirms fp8 70B fits
+    TP=1 with KV headroom (and bf16 fits at all — the MI355X story)."""
+    from llm_np_cp_amd.core.config import preset_config
+    from llm_np_cp_amd.io.loader import hf_weight_shapes
+
+    cfg = preset_config("llama-3.1-70b")
+    assert (cfg.hidden_size, cfg.num_hidden_layers) == (8192, 80)
+    assert cfg.num_kv_groups == 8
+    shapes = hf_weight_shapes(cfg)
+    n_params = sum(int(np.prod(s)) for s in shapes.values())
+    assert 69e9 < n_params < 72e9          # ~70.6B
+    assert n_params * 1 / 2**30 < 70       # fp8 bytes fit easily
+    assert n_params * 2 / 2**30 < 288      # bf16 fits in HBM too
+    # every TP degree the node offers divides the head/vocab dims
+    for tp in (1, 2, 4, 8):
+        assert cfg.num_key_value_heads % tp == 0
+        assert cfg.vocab_size % tp == 0
+        assert cfg.intermediate_size % tp == 0
