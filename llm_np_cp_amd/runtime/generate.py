@@ -282,7 +282,7 @@ class ByteTokenizer:
 def load_model(model_dir_or_preset: str, backend: str = "auto",
                device: str = "cuda", dtype: str = "bf16",
                max_seq: int = 4096, seed: int = 0,
-               kv_dtype: str = "bf16"):
+               kv_dtype: str = "bf16", max_batch: int = 1):
     """Reference-parity entry (``load_model`` -> (tokenizer, model, config),
     ``llama3.2_model.py:1082-1099``).
 
@@ -293,6 +293,9 @@ def load_model(model_dir_or_preset: str, backend: str = "auto",
     (``llama3.2_model.py:1090``) and fails with a clear error when the
     machine has no network access.
     ``backend``: "numpy" (CPU oracle), "gpu" (HIP engine), or "auto".
+    ``max_batch``: >1 allocates per-sequence KV pools on the GPU engine
+    (lockstep batched decode / continuous-batching server); ignored by
+    the NumPy oracle.
     """
     import os
 
@@ -347,7 +350,7 @@ def load_model(model_dir_or_preset: str, backend: str = "auto",
         from ..models.engine import GPUModel
 
         model = GPUModel(config, weights, dtype=dtype, max_seq=max_seq,
-                         kv_dtype=kv_dtype)
+                         kv_dtype=kv_dtype, max_batch=max_batch)
     else:
         raise ValueError(f"unknown backend {backend!r}")
     return tokenizer, model, config

@@ -223,24 +223,14 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
 
     tok, model, cfg = L.load_model(model_name, backend=backend,
                                    dtype=dtype, max_seq=max_seq,
-                                   kv_dtype=kv_dtype)
-    can_batch = False
+                                   kv_dtype=kv_dtype, max_batch=max_batch)
     if _engine is not None:
         model = _engine
         can_batch = max_batch > 1
-    elif max_batch > 1 and type(model).__name__ == "GPUModel":
-        # rebuild with batch pools (load_model keeps its reference-parity
-        # signature lean)
-        from ..io.loader import random_weights, load_weights_numpy
-        from ..models.engine import GPUModel
-        import os
-        if os.path.isdir(model_name):
-            weights = load_weights_numpy(model_name)
-        else:
-            weights = random_weights(cfg, seed=0)
-        model = GPUModel(cfg, weights, dtype=dtype, max_seq=max_seq,
-                         kv_dtype=kv_dtype, max_batch=max_batch)
-        can_batch = True
+    else:
+        # GPU engines built with batch pools serve the lockstep group
+        can_batch = (max_batch > 1
+                     and type(model).__name__ == "GPUModel")
 
     eos = getattr(cfg, "eos_token_id", None)
     eos_set = (set() if eos is None else {int(eos)}
