@@ -17,7 +17,7 @@ from .runtime.generate import generate, load_model, ByteTokenizer, GenerateResul
 from .runtime.sampling import SamplingParams, sample_token
 from .runtime.speculative import generate_speculative
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 __all__ = [
     "ModelConfig", "preset_config", "PRESETS",
     "generate", "load_model", "ByteTokenizer", "GenerateResult",
