@@ -30,6 +30,9 @@ def main(argv=None):
     ap.add_argument("--temperature", type=float, default=1.0)
     ap.add_argument("--seed", type=int, default=None)
     ap.add_argument("--max-seq", type=int, default=4096)
+    ap.add_argument("--lora", default=None,
+                    help="PEFT adapter dir merged into the weights at "
+                         "load (W' = W + (alpha/r) B@A)")
     ap.add_argument("--stop", action="append", default=None,
                     help="stop string (repeatable); excluded from output")
     ap.add_argument("--draft", default=None,
@@ -49,7 +52,7 @@ def main(argv=None):
 
     tok, model, cfg = L.load_model(args.model, backend=args.backend,
                                    dtype=args.dtype, max_seq=args.max_seq,
-                                   kv_dtype=args.kv_dtype)
+                                   kv_dtype=args.kv_dtype, lora=args.lora)
     params = L.SamplingParams(strategy=args.strategy, min_p=args.min_p,
                               temperature=args.temperature, seed=args.seed)
     t0 = time.time()

@@ -282,7 +282,8 @@ class ByteTokenizer:
 def load_model(model_dir_or_preset: str, backend: str = "auto",
                device: str = "cuda", dtype: str = "bf16",
                max_seq: int = 4096, seed: int = 0,
-               kv_dtype: str = "bf16", max_batch: int = 1):
+               kv_dtype: str = "bf16", max_batch: int = 1,
+               lora: Optional[str] = None):
     """Reference-parity entry (``load_model`` -> (tokenizer, model, config),
     ``llama3.2_model.py:1082-1099``).
 
@@ -296,6 +297,10 @@ def load_model(model_dir_or_preset: str, backend: str = "auto",
     ``max_batch``: >1 allocates per-sequence KV pools on the GPU engine
     (lockstep batched decode / continuous-batching server); ignored by
     the NumPy oracle.
+    ``lora``: optional PEFT adapter directory (adapter_config.json +
+    adapter_model.safetensors) merged into the base weights at load
+    (W' = W + (alpha/r)·B@A) — the adapted model then runs at full
+    native speed on either engine.
     """
     import os
 
@@ -319,6 +324,10 @@ def load_model(model_dir_or_preset: str, backend: str = "auto",
         raise FileNotFoundError(
             f"{model_dir_or_preset!r} is neither a directory, a preset "
             f"({sorted(PRESETS)}), nor an HF repo id")
+
+    if lora is not None:
+        from ..io.loader import apply_lora
+        apply_lora(weights, lora)
 
     tokenizer = None
     if tok_dir is not None:

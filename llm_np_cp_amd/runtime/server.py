@@ -212,7 +212,8 @@ def chat_prompt(tok, messages) -> str:
 def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
               dtype: str = "bf16", max_seq: int = 4096,
               max_batch: int = 8, kv_dtype: str = "bf16",
-              batch_window_ms: float = 4.0, _engine=None):
+              batch_window_ms: float = 4.0, lora: str = None,
+              _engine=None):
     """``_engine``: test hook — a pre-built batch-capable engine
     (GPUModel interface: prefill_row/decode_rows/compact_row/bt_*)
     injected in place of the loaded model, so the continuous-batching
@@ -223,7 +224,8 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
 
     tok, model, cfg = L.load_model(model_name, backend=backend,
                                    dtype=dtype, max_seq=max_seq,
-                                   kv_dtype=kv_dtype, max_batch=max_batch)
+                                   kv_dtype=kv_dtype, max_batch=max_batch,
+                                   lora=lora)
     if _engine is not None:
         model = _engine
         can_batch = max_batch > 1
@@ -554,12 +556,14 @@ def main():
     ap.add_argument("--max-seq", type=int, default=4096)
     ap.add_argument("--max-batch", type=int, default=8)
     ap.add_argument("--batch-window-ms", type=float, default=4.0)
+    ap.add_argument("--lora", default=None,
+                    help="PEFT adapter dir merged at load")
     ap.add_argument("--host", default="127.0.0.1")
     ap.add_argument("--port", type=int, default=8080)
     args = ap.parse_args()
     app = build_app(args.model, args.backend, args.dtype, args.max_seq,
                     max_batch=args.max_batch, kv_dtype=args.kv_dtype,
-                    batch_window_ms=args.batch_window_ms)
+                    batch_window_ms=args.batch_window_ms, lora=args.lora)
     uvicorn.run(app, host=args.host, port=args.port)
 
 
