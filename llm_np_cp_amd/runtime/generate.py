@@ -166,7 +166,7 @@ def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
 
     # fast path: device-side chunked decode loop (GPU engine, greedy or
     # min-p, cached mode) — same sampler semantics, device RNG
-    if (use_cache and logprobs is None
+    if (use_cache and logprobs is None and not params.logit_bias
             and hasattr(model, "generate_tokens")
             and params.strategy in ("greedy", "min_p")):
         t0 = time.perf_counter()

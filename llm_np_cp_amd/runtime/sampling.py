@@ -24,6 +24,20 @@ class SamplingParams:
     top_k: int = 50
     top_p: float = 0.9
     seed: Optional[int] = None
+    # OpenAI-style per-token logit offsets {token_id: bias}; applied to
+    # the raw logits before temperature/filtering (+-100 ~ ban/force)
+    logit_bias: Optional[dict] = None
+
+
+def _apply_bias(logits: np.ndarray, params: "SamplingParams") -> np.ndarray:
+    if not params.logit_bias:
+        return logits
+    logits = np.array(logits, dtype=np.float32, copy=True)
+    for tid, b in params.logit_bias.items():
+        t = int(tid)
+        if 0 <= t < len(logits):
+            logits[t] += float(b)
+    return logits
 
 
 def _softmax(x: np.ndarray) -> np.ndarray:
@@ -38,7 +52,10 @@ def filter_probs(logits: np.ndarray, params: SamplingParams) -> np.ndarray:
     softmax with the strategy's support filter applied, renormalized.
     (fp64, sums to 1.)  ``sample_token`` draws from this; speculative
     sampling (runtime/speculative.py) needs the vector itself to compute
-    accept ratios p(x)/q(x) and the residual distribution."""
+    accept ratios p(x)/q(x) and the residual distribution.  logit_bias
+    applies first, so every consumer (host sampler, speculative
+    accept/reject) sees the same biased distribution."""
+    logits = _apply_bias(logits, params)
     if params.strategy == "greedy":
         p = np.zeros(len(logits), dtype=np.float64)
         p[int(np.argmax(logits))] = 1.0
@@ -71,7 +88,7 @@ def sample_token(logits: np.ndarray, params: SamplingParams,
                  rng: Optional[np.random.Generator] = None) -> int:
     """logits: (vocab,) fp32. Returns a token id (int)."""
     if params.strategy == "greedy":
-        return int(np.argmax(logits))
+        return int(np.argmax(_apply_bias(logits, params)))
     if rng is None:
         rng = np.random.default_rng(params.seed)
     probs = filter_probs(logits, params)

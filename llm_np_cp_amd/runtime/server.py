@@ -50,6 +50,7 @@ try:  # pydantic model must live at module scope (ForwardRef resolution)
         stop: Optional[list] = None   # stop strings (OpenAI semantics)
         logprobs: Optional[int] = None  # top-N per-token logprobs
         n: int = Field(1, ge=1, le=16)  # independent samples
+        logit_bias: Optional[dict] = None  # {token_id: bias}
 
     class ChatMessage(BaseModel):
         role: str
@@ -130,6 +131,7 @@ class BatchScheduler:
                 and not getattr(req, "stream", False)
                 and not getattr(req, "stop", None)
                 and getattr(req, "logprobs", None) is None
+                and not getattr(req, "logit_bias", None)
                 and getattr(req, "n", 1) == 1)
 
     def _poll_compatible(self, key, deferred):
@@ -286,10 +288,12 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
         for idx in range(n):
             seed = (req.seed + idx) if (req.seed is not None and n > 1) \
                 else req.seed
-            params = L.SamplingParams(strategy=req.strategy,
-                                      min_p=req.min_p,
-                                      temperature=req.temperature,
-                                      seed=seed)
+            bias = getattr(req, "logit_bias", None)
+            params = L.SamplingParams(
+                strategy=req.strategy, min_p=req.min_p,
+                temperature=req.temperature, seed=seed,
+                logit_bias={int(k): float(vv) for k, vv in bias.items()}
+                if bias else None)
             out = L.generate(req.prompt, tok, model,
                              max_tokens=req.max_tokens, params=params,
                              stream=False, stop_on_eos=req.stop_on_eos,

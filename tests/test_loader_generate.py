@@ -407,3 +407,47 @@ def test_cli_speculative_and_stop(capsys):
           "--strategy", "greedy", "--max-tokens", "6",
           "--stop", "ZXQNEVER"])
     assert "6 tokens" in capsys.readouterr().err
+
+
+def test_logit_bias_bans_and_forces():
+    """OpenAI logit_bias: -100 bans a token from the whole decode; +100
+    forces it under every strategy, including greedy; speculative
+    sampling inherits the bias through filter_probs."""
+    tok, model, cfg = L.load_model("tiny-llama", backend="numpy", seed=0)
+    p0 = SamplingParams(strategy="greedy")
+    base = L.generate("Once upon a time", tok, model, max_tokens=8,
+                      stream=False, params=p0, stop_on_eos=False)
+    banned = int(base.token_ids[0])
+    res = L.generate("Once upon a time", tok, model, max_tokens=8,
+                     stream=False, stop_on_eos=False,
+                     params=SamplingParams(strategy="greedy",
+                                           logit_bias={banned: -1e4}))
+    assert banned not in res.token_ids
+    forced = 7
+    res2 = L.generate("x", tok, model, max_tokens=5, stream=False,
+                      stop_on_eos=False,
+                      params=SamplingParams(strategy="min_p", seed=3,
+                                            logit_bias={forced: 1e4}))
+    assert res2.token_ids == [forced] * 5
+
+    from llm_np_cp_amd.runtime.sampling import filter_probs
+    import numpy as np
+    logits = np.zeros(16, dtype=np.float32)
+    pr = filter_probs(logits, SamplingParams(strategy="temperature",
+                                             logit_bias={3: 100.0}))
+    assert pr[3] > 0.999
+
+
+def test_logit_bias_via_server():
+    import pytest as _pytest
+    fastapi = _pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy")
+    client = TestClient(app)
+    r = client.post("/v1/completions", json={
+        "prompt": "x", "max_tokens": 4, "strategy": "min_p", "seed": 1,
+        "stop_on_eos": False, "logit_bias": {"9": 10000.0}}).json()
+    ids = [ord(c) for c in r["choices"][0]["text"]]
+    assert ids == [9, 9, 9, 9]  # byte tokenizer: token 9 == "\t"
