@@ -15,11 +15,12 @@ Public API (reference parity):
 from .core.config import ModelConfig, preset_config, PRESETS
 from .runtime.generate import generate, load_model, ByteTokenizer, GenerateResult
 from .runtime.sampling import SamplingParams, sample_token
+from .runtime.session import ChatSession
 from .runtime.speculative import generate_speculative
 
 __version__ = "0.2.0"
 __all__ = [
     "ModelConfig", "preset_config", "PRESETS",
     "generate", "load_model", "ByteTokenizer", "GenerateResult",
-    "SamplingParams", "sample_token", "generate_speculative",
+    "SamplingParams", "sample_token", "generate_speculative", "ChatSession",
 ]
