@@ -66,6 +66,9 @@ try:  # pydantic model must live at module scope (ForwardRef resolution)
         stop_on_eos: bool = True
         stream: bool = False
         stop: Optional[list] = None
+        logprobs: Optional[int] = None
+        n: int = Field(1, ge=1, le=16)
+        logit_bias: Optional[dict] = None
 except ImportError:  # pragma: no cover - serving is optional
     CompletionRequest = None
     ChatMessage = None
@@ -523,7 +526,8 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
             temperature=req.temperature, min_p=req.min_p,
             strategy=req.strategy, seed=req.seed,
             stop_on_eos=req.stop_on_eos, stream=req.stream,
-            stop=req.stop)
+            stop=req.stop, logprobs=req.logprobs, n=req.n,
+            logit_bias=req.logit_bias)
         if req.stream:
             return _sse(creq, lambda piece: {
                 "object": "chat.completion.chunk", "model": model_name,
@@ -533,13 +537,18 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
         except ValueError as e:
             from fastapi import HTTPException
             raise HTTPException(status_code=400, detail=str(e))
-        choice = out["choices"][0]
+        choices = []
+        for i, ch in enumerate(out["choices"]):
+            c = {"index": i,
+                 "message": {"role": "assistant",
+                             "content": ch["text"]},
+                 "finish_reason": ch["finish_reason"]}
+            if "logprobs" in ch:
+                c["logprobs"] = ch["logprobs"]
+            choices.append(c)
         return {
             "object": "chat.completion", "model": model_name,
-            "choices": [{"index": 0,
-                         "message": {"role": "assistant",
-                                     "content": choice["text"]},
-                         "finish_reason": choice["finish_reason"]}],
+            "choices": choices,
             "usage": out["usage"], "timings": out["timings"],
         }
 

@@ -693,3 +693,22 @@ def test_scheduler_eos_retirement_cpu():
     from llm_np_cp_amd.runtime.generate import ByteTokenizer
     assert results["a"]["choices"][0]["text"] == \
         ByteTokenizer().decode(ids[:cut + 1])
+
+
+def test_chat_field_symmetry_n_and_bias():
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy")
+    client = TestClient(app)
+    r = client.post("/v1/chat/completions", json={
+        "messages": [{"role": "user", "content": "hi"}],
+        "max_tokens": 3, "n": 2, "strategy": "min_p", "seed": 4,
+        "stop_on_eos": False, "logit_bias": {"9": 10000.0},
+        "logprobs": 2}).json()
+    assert len(r["choices"]) == 2
+    for i, c in enumerate(r["choices"]):
+        assert c["index"] == i
+        assert c["message"]["content"] == "\t\t\t"  # forced token 9
+        assert len(c["logprobs"]["tokens"]) == 3
