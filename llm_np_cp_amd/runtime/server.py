@@ -312,6 +312,8 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
                 pay["choices"].append({"text": out.text, "index": idx,
                                        "finish_reason": out.finish_reason})
                 pay["usage"]["completion_tokens"] += len(out.token_ids)
+                if metrics_reg is not None:  # extra choices count too
+                    m_tokens.inc(len(out.token_ids))
             if out.logprobs is not None:
                 pay["choices"][idx]["logprobs"] = {
                     "tokens": [e["token"] for e in out.logprobs],
