@@ -97,3 +97,37 @@ def test_window_geq_seq_is_full_attention_any_shape(data):
                           "layer_types": []})
     b = NumpyModel(cfg2, dict(w)).forward(ids, NumpyKVCache(cfg2, 32), 0)
     np.testing.assert_allclose(a, b, rtol=1e-5, atol=1e-6)
+
+
+@settings(max_examples=8, deadline=None)
+@given(data=st.data())
+def test_moe_incremental_decode_any_shape(data):
+    """Mixtral-style sparse MoE at random (E, topk, shape) combos:
+    routing must be a pure function of position state — incremental
+    decode equals full prefill."""
+    E = data.draw(st.sampled_from([2, 4, 8]), label="E")
+    topk = data.draw(st.integers(min_value=1, max_value=min(E, 3)),
+                     label="topk")
+    nh = data.draw(st.sampled_from([2, 4]), label="nh")
+    kvh = data.draw(st.sampled_from([d for d in (1, 2) if nh % d == 0]),
+                    label="kvh")
+    cfg = ModelConfig(
+        model_type="mixtral", vocab_size=64, hidden_size=16,
+        intermediate_size=24, num_hidden_layers=2,
+        num_attention_heads=nh, num_key_value_heads=kvh, head_dim=8,
+        rms_norm_eps=1e-6, rope_theta=10000.0,
+        max_position_embeddings=64, hidden_act="silu",
+        tie_word_embeddings=False, num_local_experts=E,
+        num_experts_per_tok=topk)
+    w = random_weights(cfg, seed=11)
+    m = NumpyModel(cfg, dict(w))
+    ids = np.random.default_rng(2).integers(0, 64, size=9)
+    full = m.forward(ids, NumpyKVCache(cfg, 32), 0)
+    split = data.draw(st.integers(min_value=1, max_value=8),
+                      label="split")
+    c = NumpyKVCache(cfg, 32)
+    m.forward(ids[:split], c, 0)
+    out = full[split - 1:split]
+    for t in range(split, 9):
+        out = m.forward(ids[t:t + 1], c, t)
+    np.testing.assert_allclose(out[0], full[-1], rtol=2e-4, atol=2e-5)
