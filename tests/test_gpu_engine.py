@@ -850,3 +850,23 @@ def test_lora_merged_gpu_matches_oracle(tmp_path):
     c = L.generate("Once upon", tok, base, max_tokens=12, stream=False,
                    params=p, stop_on_eos=False)
     assert isinstance(c.token_ids, list)
+
+
+def test_stop_sequences_on_device_fast_path():
+    """Stop strings with the hipGraph chunked loop: stop_fn ends the
+    chunk loop early and the host truncation yields exactly the base
+    rollout's prefix before the stop (finish_reason 'stop')."""
+    import llm_np_cp_amd as L
+
+    cfg, gpu, _ref = make_pair("tiny-llama", seed=6)
+    tok = L.ByteTokenizer()
+    p = L.SamplingParams(strategy="greedy")
+    base = L.generate("Once upon", tok, gpu, max_tokens=24, stream=False,
+                      params=p, stop_on_eos=False)
+    stop = base.text[5:8]
+    assert stop
+    res = L.generate("Once upon", tok, gpu, max_tokens=24, stream=False,
+                     params=p, stop_on_eos=False, stop=[stop])
+    assert res.finish_reason == "stop"
+    assert res.text == base.text[:base.text.find(stop)]
+    assert stop not in res.text
