@@ -109,6 +109,16 @@ class ChatSession:
                     take = take[:hit[0] + 1]
                     hit_eos = True
                 out.extend(take)
+            if out:
+                # the device loop leaves the LAST sampled token pending
+                # (its KV is written by the step that would consume it)
+                # — settle it so the next send() attends over complete
+                # KV: O(1) rewind + one M=1 forward
+                self.model.rewind(pos0 + len(new_ids) + len(out) - 1)
+                self.cache.seq_len = pos0 + len(new_ids) + len(out) - 1
+                self.model.forward(
+                    np.asarray([out[-1]], dtype=np.int64), self.cache,
+                    self.cache.seq_len)
             self.cache.seq_len = len(self.token_ids) + len(out)
         else:
             for _ in range(max_tokens):

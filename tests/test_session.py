@@ -73,3 +73,75 @@ def test_session_eos_stops_turn():
         assert r2.finish_reason == "stop"
     finally:
         cfg.eos_token_id = None
+
+
+class DeviceLoopAdapter:
+    """Reproduces GPUModel's decode-loop CONTRACT over the NumPy oracle
+    so ChatSession's GPU branch is CPU-testable: forward() leaves
+    last-position logits; decode(n, first_from_logits) samples token 0
+    from them, then each further step PROCESSES the pending token
+    (writing its KV / advancing the cache) and samples the next — the
+    final sampled token is left pending with NO KV written, exactly
+    like the hipGraph loop.  rewind() is an O(1) length set."""
+
+    def __init__(self, cfg, ref, max_seq=256):
+        self.config = cfg
+        self.ref = ref
+        self.max_seq = max_seq
+        self._logits = None
+        self._pending = None
+
+    def make_cache(self, n):
+        self._cache = NumpyKVCache(self.config, n)
+        return self._cache
+
+    def forward(self, ids, cache, pos0):
+        self._logits = self.ref.forward(np.asarray(ids), cache, pos0)
+        return self._logits
+
+    def rewind(self, n):
+        self._cache.seq_len = n
+
+    def decode(self, n, greedy=True, min_p=0.1, use_graph=True,
+               first_from_logits=True, temperature=1.0):
+        assert greedy, "adapter models the greedy device loop"
+        out = []
+        steps = n
+        if first_from_logits:
+            self._pending = int(np.argmax(self._logits[-1]))
+            out.append(self._pending)
+            steps -= 1
+        for _ in range(steps):
+            self._logits = self.ref.forward(
+                np.asarray([self._pending]), self._cache,
+                self._cache.seq_len)
+            self._pending = int(np.argmax(self._logits[-1]))
+            out.append(self._pending)
+        return np.asarray(out, dtype=np.int32)
+
+
+def test_session_device_loop_branch_matches_plain_oracle():
+    """The GPU branch of ChatSession (device decode loop + pending-token
+    settle) must be turn-identical to the plain oracle branch — in
+    particular turn 2 attends over the KV of turn 1's LAST token, which
+    the device loop leaves unwritten until settled."""
+    tok, model, cfg = L.load_model("tiny-llama", backend="numpy", seed=0)
+    from llm_np_cp_amd.models.numpy_ref import NumpyModel
+    ref2 = NumpyModel(cfg, dict(model.w))
+    dev = DeviceLoopAdapter(cfg, ref2, max_seq=256)
+
+    p = L.SamplingParams(strategy="greedy")
+    s_plain = L.ChatSession(tok, model, params=p, max_seq=256)
+    s_dev = L.ChatSession(tok, dev, params=p, max_seq=256)
+    for text, k in (("Hello there.", 9), (" And then?", 9),
+                    (" The end?", 6)):
+        a = s_plain.send(text, max_tokens=k, stop_on_eos=False)
+        b = s_dev.send(text, max_tokens=k, stop_on_eos=False)
+        assert a.token_ids == b.token_ids, text
+        # the settle must leave KV WRITTEN for every resident position
+        # (without it the last token of each turn has a zeroed KV row —
+        # argmax can mask that on tiny models, the cache cannot)
+        kc = dev._cache.k[:, :, :s_dev.seq_len]
+        norms = np.abs(kc).sum(axis=(0, 1, 3))
+        assert (norms > 0).all(), np.where(norms == 0)
+    assert s_plain.token_ids == s_dev.token_ids
