@@ -126,12 +126,15 @@ class ChatSession:
                                               dtype=np.float32),
                                    self.params, self._rng)
                 out.append(nid)
-                if stop_on_eos and nid in self.eos_set:
-                    hit_eos = True
-                    break
+                hit_eos = stop_on_eos and nid in self.eos_set
+                # forward EVERY sampled token (the eos included) so the
+                # cache holds KV for the whole transcript — the next
+                # turn attends over all of it
                 logits = self.model.forward(
                     np.asarray([nid], dtype=np.int64), self.cache,
                     self.cache.seq_len)
+                if hit_eos:
+                    break
         t_decode = time.perf_counter() - t1
 
         self.token_ids.extend(out)

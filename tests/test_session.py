@@ -71,6 +71,13 @@ def test_session_eos_stops_turn():
         r2 = s.send("Once upon", max_tokens=8, stop_on_eos=True)
         assert r2.token_ids == r.token_ids[:first + 1]
         assert r2.finish_reason == "stop"
+        # the eos token's KV must be resident so the NEXT turn attends
+        # over the complete transcript
+        kc = s.cache.k[:, :, :s.seq_len]
+        norms = np.abs(kc).sum(axis=(0, 1, 3))
+        assert (norms > 0).all(), np.where(norms == 0)
+        r3 = s.send(" next turn", max_tokens=4, stop_on_eos=False)
+        assert len(r3.token_ids) == 4
     finally:
         cfg.eos_token_id = None
 
