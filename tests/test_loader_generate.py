@@ -451,3 +451,101 @@ def test_logit_bias_via_server():
         "stop_on_eos": False, "logit_bias": {"9": 10000.0}}).json()
     ids = [ord(c) for c in r["choices"][0]["text"]]
     assert ids == [9, 9, 9, 9]  # byte tokenizer: token 9 == "\t"
+
+
+class FastLoopAdapter:
+    """Emulates GPUModel's generate_tokens CONTRACT over the NumPy
+    oracle (chunked ids, EOS-set truncation inside a chunk, stop_fn
+    polled between chunks, on_ids per chunk) so generate()'s fast-path
+    glue — chunk-granular _StopScan feeding, hit_eos detection,
+    truncation — is CPU-testable."""
+
+    last_prefill_time_s = 0.0
+
+    def __init__(self, cfg, ref):
+        self.config = cfg
+        self.ref = ref
+
+    def make_cache(self, n):
+        from llm_np_cp_amd.models.numpy_ref import NumpyKVCache
+        self._cache = NumpyKVCache(self.config, max(n, 64))
+        return self._cache
+
+    def generate_tokens(self, prompt_ids, max_tokens, greedy=True,
+                        min_p=0.1, eos_id=None, chunk=4, on_ids=None,
+                        temperature=1.0, stop_fn=None):
+        assert greedy
+        eos = set() if eos_id is None else set(int(e) for e in eos_id)
+        cache = self.make_cache(len(prompt_ids) + max_tokens + 2)
+        logits = self.ref.forward(np.asarray(prompt_ids), cache, 0)
+        out, pending = [], int(np.argmax(logits[-1]))
+        while len(out) < max_tokens:
+            take = []
+            for _ in range(min(chunk, max_tokens - len(out))):
+                take.append(pending)
+                logits = self.ref.forward(np.asarray([pending]), cache,
+                                          cache.seq_len)
+                pending = int(np.argmax(logits[-1]))
+            hit = [j for j, t in enumerate(take) if t in eos]
+            stop = bool(hit)
+            if hit:
+                take = take[:hit[0] + 1]
+            out.extend(take)
+            if on_ids:
+                on_ids(take)
+            if stop:
+                break
+            if stop_fn is not None and stop_fn(out):
+                break
+        return out
+
+
+def _fast_adapter(seed=0):
+    from llm_np_cp_amd.core.config import preset_config
+    from llm_np_cp_amd.io.loader import random_weights
+    from llm_np_cp_amd.models.numpy_ref import NumpyModel
+
+    cfg = preset_config("tiny-llama")
+    ref = NumpyModel(cfg, random_weights(cfg, seed=seed))
+    return cfg, FastLoopAdapter(cfg, ref)
+
+
+def test_fast_path_stop_strings_chunk_granular():
+    """Fast path + stop strings: stop_fn ends between chunks, the
+    truncation is exact, and streamed chunks (fed to _StopScan in
+    MULTI-TOKEN pieces) concatenate to the truncated text."""
+    cfg, model = _fast_adapter(seed=0)
+    tok = L.ByteTokenizer()
+    p = SamplingParams(strategy="greedy")
+    base = L.generate("Once upon", tok, model, max_tokens=20,
+                      stream=False, params=p, stop_on_eos=False)
+    assert len(base.token_ids) == 20  # fast path engaged (adapter)
+    stop = base.text[6:9]
+    chunks = []
+    res = L.generate("Once upon", tok, model, max_tokens=20,
+                     stream=False, params=p, stop_on_eos=False,
+                     stop=[stop], on_token=chunks.append)
+    assert res.finish_reason == "stop"
+    assert res.text == base.text[:base.text.find(stop)]
+    assert "".join(chunks) == res.text
+    assert stop not in "".join(chunks)
+
+
+def test_fast_path_eos_list_mid_chunk():
+    """EOS inside a chunk: the fast path truncates at the FIRST eos hit
+    and reports finish_reason 'stop' (list-valued eos handled)."""
+    cfg, model = _fast_adapter(seed=0)
+    tok = L.ByteTokenizer()
+    p = SamplingParams(strategy="greedy")
+    base = L.generate("abc", tok, model, max_tokens=12, stream=False,
+                      params=p, stop_on_eos=False)
+    eos_tok = int(base.token_ids[2])
+    first = base.token_ids.index(eos_tok)
+    cfg.eos_token_id = [999999, eos_tok]
+    try:
+        r = L.generate("abc", tok, model, max_tokens=12, stream=False,
+                       params=p, stop_on_eos=True)
+        assert r.token_ids == base.token_ids[:first + 1]
+        assert r.finish_reason == "stop"
+    finally:
+        cfg.eos_token_id = None
