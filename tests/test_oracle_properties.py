@@ -131,3 +131,70 @@ def test_moe_incremental_decode_any_shape(data):
     for t in range(split, 9):
         out = m.forward(ids[t:t + 1], c, t)
     np.testing.assert_allclose(out[0], full[-1], rtol=2e-4, atol=2e-5)
+
+
+# ---------- stop-string machinery invariants (tricky string logic) ----------
+
+@settings(max_examples=120, deadline=None)
+@given(data=st.data())
+def test_stopscan_emits_exactly_truncated_text(data):
+    """_StopScan fed ANY chunking of ANY text must emit exactly
+    text[:first stop occurrence] (or all of it after flush when no stop
+    occurs), never any part of a stop string."""
+    from llm_np_cp_amd.runtime.generate import _StopScan
+
+    alphabet = "abcX"
+    text = data.draw(st.text(alphabet=alphabet, min_size=0, max_size=40),
+                     label="text")
+    stops = data.draw(st.lists(st.text(alphabet=alphabet, min_size=1,
+                                       max_size=4),
+                               min_size=1, max_size=3, unique=True),
+                      label="stops")
+    pieces = []
+    i = 0
+    while i < len(text):
+        n = data.draw(st.integers(min_value=1, max_value=6))
+        pieces.append(text[i:i + n])
+        i += n
+
+    emitted = []
+    scan = _StopScan(stops, emitted.append)
+    hit = False
+    for p in pieces:
+        if scan.feed(p):
+            hit = True
+            break
+    if not hit:
+        scan.flush()
+
+    cuts = [c for c in (text.find(s) for s in stops) if c >= 0]
+    expect = text[:min(cuts)] if cuts else text
+    assert "".join(emitted) == expect
+    assert hit == bool(cuts)
+
+
+@settings(max_examples=80, deadline=None)
+@given(data=st.data())
+def test_truncate_at_stop_invariants(data):
+    """_truncate_at_stop over the byte tokenizer: text' is the exact
+    prefix before the earliest stop, ids' decode covers it, and without
+    a match everything is returned unchanged."""
+    from llm_np_cp_amd.runtime.generate import (ByteTokenizer,
+                                                _truncate_at_stop)
+
+    tok = ByteTokenizer()
+    ids = data.draw(st.lists(st.integers(min_value=97, max_value=101),
+                             min_size=0, max_size=30), label="ids")
+    stops = data.draw(st.lists(st.text(alphabet="abcde", min_size=1,
+                                       max_size=3),
+                               min_size=1, max_size=2, unique=True),
+                      label="stops")
+    text = tok.decode(ids)
+    out, kept, hit = _truncate_at_stop(list(ids), tok, stops)
+    cuts = [c for c in (text.find(s) for s in stops) if c >= 0]
+    if not cuts:
+        assert (out, kept, hit) == (list(ids), text, False)
+    else:
+        assert hit
+        assert kept == text[:min(cuts)]
+        assert tok.decode(out) == kept  # byte tokenizer: 1 id = 1 char
