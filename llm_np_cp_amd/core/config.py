@@ -101,9 +101,25 @@ class ModelConfig:
         return 1.0
 
     # ------------------------------------------------------------------
+    SUPPORTED_FAMILIES = ("llama", "gemma2", "qwen2", "mistral",
+                          "mixtral")
+
     @classmethod
     def from_hf_dict(cls, d: dict) -> "ModelConfig":
         model_type = d.get("model_type", "llama")
+        if model_type not in cls.SUPPORTED_FAMILIES:
+            raise ValueError(
+                f"unsupported model_type {model_type!r}; this framework "
+                f"implements the {'/'.join(cls.SUPPORTED_FAMILIES)} "
+                f"decoder families (treating an unknown architecture as "
+                f"Llama would silently produce wrong results)")
+        required = ("vocab_size", "hidden_size", "intermediate_size",
+                    "num_hidden_layers", "num_attention_heads")
+        missing = [k for k in required if k not in d]
+        if missing:
+            raise ValueError(
+                f"config.json is missing required fields {missing} — "
+                f"not a decoder-LM config this framework can load")
         heads = d["num_attention_heads"]
         cfg = cls(
             model_type=model_type,

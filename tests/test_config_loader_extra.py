@@ -364,3 +364,19 @@ def test_llama_70b_preset_shapes_and_memory_math():
         assert cfg.num_key_value_heads % tp == 0
         assert cfg.vocab_size % tp == 0
         assert cfg.intermediate_size % tp == 0
+
+
+def test_from_hf_dict_rejects_foreign_configs():
+    """A non-decoder or unknown-architecture config.json fails with an
+    actionable error instead of silently parsing as Llama."""
+    from llm_np_cp_amd.core.config import ModelConfig
+
+    with pytest.raises(ValueError, match="unsupported model_type 'gpt2'"):
+        ModelConfig.from_hf_dict({"model_type": "gpt2",
+                                  "vocab_size": 50257, "hidden_size": 768,
+                                  "intermediate_size": 3072,
+                                  "num_hidden_layers": 12,
+                                  "num_attention_heads": 12})
+    with pytest.raises(ValueError, match="missing required fields"):
+        ModelConfig.from_hf_dict({"model_type": "llama",
+                                  "vocab_size": 128256})
