@@ -341,6 +341,9 @@ def load_model(model_dir_or_preset: str, backend: str = "auto",
         except Exception:
             tokenizer = None
     if tokenizer is None:
+        if tok_dir is not None:
+            print(f"# no usable tokenizer files in {tok_dir}; using the "
+                  f"byte-level fallback tokenizer", file=sys.stderr)
         tokenizer = ByteTokenizer()
 
     if backend == "auto":
