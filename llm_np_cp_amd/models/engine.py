@@ -913,6 +913,8 @@ class GPUModel:
         if pos0 != cache.seq_len:
             raise ValueError("GPU engine requires sequential positions")
         n = len(ids)
+        if n == 0:
+            raise ValueError("empty ids")
         if pos0 + n > self.max_seq:
             raise ValueError(f"sequence {pos0}+{n} exceeds max_seq "
                              f"{self.max_seq}")

@@ -40,7 +40,7 @@ class GenerateResult:
 
     @property
     def decode_tokens_per_s(self) -> float:
-        n = max(len(self.token_ids) - 0, 0)
+        n = len(self.token_ids)
         return n / self.decode_time_s if self.decode_time_s > 0 else 0.0
 
 
