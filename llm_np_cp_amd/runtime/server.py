@@ -199,13 +199,23 @@ class BatchScheduler:
                 self.q.put(p)
 
 
+def _msg_text(c) -> str:
+    """OpenAI allows content to be a string OR a list of typed parts
+    ({"type": "text", "text": ...}); flatten the text parts."""
+    if isinstance(c, list):
+        return "".join(p.get("text", "") if isinstance(p, dict)
+                       else str(p) for p in c)
+    return str(c)
+
+
 def chat_prompt(tok, messages) -> str:
     """Assemble chat messages into a prompt: the tokenizer's own
     chat_template when the checkpoint ships one (AutoTokenizer), else a
     neutral <|role|> fallback usable with any tokenizer."""
-    msgs = [{"role": m["role"], "content": m["content"]}
+    msgs = [{"role": m["role"], "content": _msg_text(m["content"])}
             if isinstance(m, dict) else
-            {"role": m.role, "content": m.content} for m in messages]
+            {"role": m.role, "content": _msg_text(m.content)}
+            for m in messages]
     template = getattr(tok, "chat_template", None)
     if template and hasattr(tok, "apply_chat_template"):
         return tok.apply_chat_template(msgs, tokenize=False,

@@ -712,3 +712,18 @@ def test_chat_field_symmetry_n_and_bias():
         assert c["index"] == i
         assert c["message"]["content"] == "\t\t\t"  # forced token 9
         assert len(c["logprobs"]["tokens"]) == 3
+
+
+def test_chat_prompt_content_part_arrays():
+    """OpenAI content-part arrays ({'type':'text','text':...}) flatten
+    to their text in the assembled prompt."""
+    from llm_np_cp_amd.runtime.server import chat_prompt
+
+    class PlainTok:
+        pass
+
+    out = chat_prompt(PlainTok(), [
+        {"role": "user",
+         "content": [{"type": "text", "text": "Hello "},
+                     {"type": "text", "text": "world"}]}])
+    assert out == "<|user|>\nHello world\n<|assistant|>\n"
