@@ -51,6 +51,7 @@ try:  # pydantic model must live at module scope (ForwardRef resolution)
         logprobs: Optional[int] = None  # top-N per-token logprobs
         n: int = Field(1, ge=1, le=16)  # independent samples
         logit_bias: Optional[dict] = None  # {token_id: bias}
+        echo: bool = False        # prepend the prompt to the text
 
     class ChatMessage(BaseModel):
         role: str
@@ -135,6 +136,7 @@ class BatchScheduler:
                 and not getattr(req, "stop", None)
                 and getattr(req, "logprobs", None) is None
                 and not getattr(req, "logit_bias", None)
+                and not getattr(req, "echo", False)
                 and getattr(req, "n", 1) == 1)
 
     def _poll_compatible(self, key, deferred):
@@ -313,13 +315,15 @@ def build_app(model_name: str = "llama-3.2-1b", backend: str = "auto",
                              stop=getattr(req, "stop", None),
                              logprobs=getattr(req, "logprobs", None),
                              on_token=on_token if idx == 0 else None)
+            text = (req.prompt + out.text
+                    if getattr(req, "echo", False) else out.text)
             if pay is None:
-                pay = _payload(req, out.token_ids, out.text,
+                pay = _payload(req, out.token_ids, text,
                                time.time() - t0, out.prefill_time_s,
                                out.decode_tokens_per_s,
                                finish_reason=out.finish_reason)
             else:
-                pay["choices"].append({"text": out.text, "index": idx,
+                pay["choices"].append({"text": text, "index": idx,
                                        "finish_reason": out.finish_reason})
                 pay["usage"]["completion_tokens"] += len(out.token_ids)
                 if metrics_reg is not None:  # extra choices count too

@@ -727,3 +727,20 @@ def test_chat_prompt_content_part_arrays():
          "content": [{"type": "text", "text": "Hello "},
                      {"type": "text", "text": "world"}]}])
     assert out == "<|user|>\nHello world\n<|assistant|>\n"
+
+
+def test_echo_prepends_prompt():
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+    from llm_np_cp_amd.runtime.server import build_app
+
+    app = build_app("tiny-llama", backend="numpy")
+    client = TestClient(app)
+    base = client.post("/v1/completions", json={
+        "prompt": "abc", "max_tokens": 3, "strategy": "greedy",
+        "stop_on_eos": False}).json()
+    r = client.post("/v1/completions", json={
+        "prompt": "abc", "max_tokens": 3, "strategy": "greedy",
+        "stop_on_eos": False, "echo": True}).json()
+    assert r["choices"][0]["text"] == "abc" + base["choices"][0]["text"]
+    assert r["usage"]["completion_tokens"] == 3
