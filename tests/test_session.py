@@ -152,3 +152,17 @@ def test_session_device_loop_branch_matches_plain_oracle():
         norms = np.abs(kc).sum(axis=(0, 1, 3))
         assert (norms > 0).all(), np.where(norms == 0)
     assert s_plain.token_ids == s_dev.token_ids
+
+
+def test_session_min_p_seeded_reset_determinism():
+    """Stochastic sessions: reset() restores the RNG stream, so the
+    same seed replays the same multi-turn sampling."""
+    tok, model, cfg = L.load_model("tiny-llama", backend="numpy", seed=2)
+    p = L.SamplingParams(strategy="min_p", min_p=0.05, seed=77)
+    s = L.ChatSession(tok, model, params=p, max_seq=128)
+    a1 = s.send("hello", max_tokens=6, stop_on_eos=False).token_ids
+    a2 = s.send(" more", max_tokens=6, stop_on_eos=False).token_ids
+    s.reset()
+    b1 = s.send("hello", max_tokens=6, stop_on_eos=False).token_ids
+    b2 = s.send(" more", max_tokens=6, stop_on_eos=False).token_ids
+    assert (a1, a2) == (b1, b2)
