@@ -3362,17 +3362,11 @@ extern "C" hipError_t launch_gbar_bench(void* cnt, void* seq, void* err,
 // ====================================================================
 
 extern "C" __global__ void k_i32_set(int* p, int v) { *p = v; }
-extern "C" __global__ void k_i32_add(int* p, int v) { *p += v; }
 
 extern "C" hipError_t launch_i32_set(void* p, int v, hipStream_t stream) {
   hipLaunchKernelGGL(k_i32_set, dim3(1), dim3(1), 0, stream, (int*)p, v);
   return hipGetLastError();
 }
-extern "C" hipError_t launch_i32_add(void* p, int v, hipStream_t stream) {
-  hipLaunchKernelGGL(k_i32_add, dim3(1), dim3(1), 0, stream, (int*)p, v);
-  return hipGetLastError();
-}
-
 // ====================================================================
 // Mixtral sparse-MoE helpers (beyond-parity: the reference has no MoE;
 // ROADMAP §5).  k_moe_route fuses the per-row RMSNorm, the router dots

@@ -69,7 +69,6 @@ _SIGS = {
     "launch_bias_add": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
                         ctypes.c_int, ctypes.c_void_p],
     "launch_i32_set": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
-    "launch_i32_add": [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
     "launch_quant_fp8": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 +
                         [ctypes.c_void_p],
     "launch_quant_fp4": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 +
@@ -519,10 +518,6 @@ def addinto(y: torch.Tensor, a: torch.Tensor):
 
 def i32_set(buf: torch.Tensor, v: int):
     _check(lib().launch_i32_set(_ptr(buf), v, _stream()), "i32_set")
-
-
-def i32_add(buf: torch.Tensor, v: int):
-    _check(lib().launch_i32_add(_ptr(buf), v, _stream()), "i32_add")
 
 
 # ----------------------------------------------------------------------
