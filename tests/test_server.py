@@ -744,3 +744,118 @@ def test_echo_prepends_prompt():
         "stop_on_eos": False, "echo": True}).json()
     assert r["choices"][0]["text"] == "abc" + base["choices"][0]["text"]
     assert r["usage"]["completion_tokens"] == 3
+
+
+def _extend_fake_single_sequence():
+    """Give FakeBatchEngine the single-sequence surface (forward /
+    make_cache / generate_tokens / config) over the SAME deterministic
+    per-prompt stream, with one-hot logits so every sampling strategy
+    picks the stream token — heterogeneous requests become
+    token-exactly checkable."""
+    from types import SimpleNamespace
+
+    def make_cache(self, n):
+        return SimpleNamespace(seq_len=0, max_seq=n, key=None, gen=-1)
+
+    def forward(self, ids, cache, pos0):
+        import numpy as _np
+        ids = [int(x) for x in _np.ravel(ids)]
+        if cache.key is None:
+            cache.key = self.key_of(ids)
+            cache.gen = 0
+        else:
+            cache.gen += 1
+        cache.seq_len = pos0 + len(ids)
+        row = _np.zeros((1, 256), dtype=_np.float32)
+        row[0, self.tok_at(cache.key, cache.gen)] = 100.0
+        return row
+
+    def generate_tokens(self, prompt_ids, max_tokens, greedy=True,
+                        min_p=0.1, eos_id=None, chunk=4, on_ids=None,
+                        temperature=1.0, stop_fn=None):
+        import numpy as _np
+        key = self.key_of(int(x) for x in _np.ravel(prompt_ids))
+        out = []
+        while len(out) < max_tokens:
+            take = [self.tok_at(key, len(out) + j)
+                    for j in range(min(chunk, max_tokens - len(out)))]
+            hit = [j for j, t in enumerate(take)
+                   if eos_id and t in eos_id]
+            if hit:
+                take = take[:hit[0] + 1]
+            out.extend(take)
+            if on_ids:
+                on_ids(take)
+            if hit:
+                break
+            if stop_fn is not None and stop_fn(out):
+                break
+        return out
+
+    FakeBatchEngine.make_cache = make_cache
+    FakeBatchEngine.forward = forward
+    FakeBatchEngine.generate_tokens = generate_tokens
+    FakeBatchEngine.last_prefill_time_s = 0.0
+    FakeBatchEngine.config = type("Cfg", (), {"eos_token_id": None})()
+
+
+def test_scheduler_heterogeneous_load_token_exact():
+    """Concurrent MIX of batchable (greedy), deferred (top_k), SSE
+    streaming, and stop-string requests against the fake engine: every
+    path routes correctly (group / generate_one) and every response is
+    token-exact for its own stream."""
+    fastapi = pytest.importorskip("fastapi")
+    import json as _json
+    import threading
+    from fastapi.testclient import TestClient
+    from llm_np_cp_amd.runtime.server import build_app
+
+    _extend_fake_single_sequence()
+    app = build_app("tiny-llama", backend="numpy", max_seq=256,
+                    max_batch=3, batch_window_ms=20.0,
+                    _engine=FakeBatchEngine(3, 256, step_delay=0.001))
+    client = TestClient(app)
+    results = {}
+
+    def plain(name, prompt, m, extra=None):
+        body = {"prompt": prompt, "max_tokens": m, "strategy": "greedy",
+                "stop_on_eos": False}
+        body.update(extra or {})
+        results[name] = client.post("/v1/completions", json=body).json()
+
+    def stream(name, prompt, m):
+        with client.stream("POST", "/v1/completions", json={
+                "prompt": prompt, "max_tokens": m, "strategy": "greedy",
+                "stop_on_eos": False, "stream": True}) as r:
+            ev = [ln for ln in r.iter_lines() if ln.startswith("data: ")]
+        txt = "".join(_json.loads(e[6:])["choices"][0]["text"]
+                      for e in ev[:-1])
+        results[name] = txt
+
+    full = _expected_text("stopper", 40)
+    stop_frag = full[20:23]
+    threads = [
+        threading.Thread(target=plain, args=("b1", "batch one", 30)),
+        threading.Thread(target=plain, args=("b2", "batch two", 25)),
+        threading.Thread(target=plain, args=("tk", "topk req", 10),
+                         kwargs={"extra": {"strategy": "top_k"}}),
+        threading.Thread(target=stream, args=("st", "streamer", 15)),
+        threading.Thread(target=plain, args=("sp", "stopper", 40),
+                         kwargs={"extra": {"stop": [stop_frag]}}),
+    ]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=120)
+
+    assert results["b1"]["choices"][0]["text"] == \
+        _expected_text("batch one", 30)
+    assert results["b2"]["choices"][0]["text"] == \
+        _expected_text("batch two", 25)
+    assert results["tk"]["choices"][0]["text"] == \
+        _expected_text("topk req", 10)
+    assert results["st"] == _expected_text("streamer", 15)
+    want_sp = full[:full.find(stop_frag)]
+    assert results["sp"]["choices"][0]["text"] == want_sp
+    assert results["sp"]["choices"][0]["finish_reason"] == "stop"
+    assert client.get("/stats").json()["requests"] == 5
