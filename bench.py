@@ -70,6 +70,13 @@ def main():
     model = GPUModel(cfg, w, max_seq=max_seq, seed=0, dtype=dtype,
                      kv_dtype=args.kv_dtype, max_batch=args.batch)
 
+    if rank == 0 and world > 1:
+        from llm_np_cp_amd.parallel import tp as _tp
+        mode = ("one-shot xGMI" if _tp.xgmi_comm() is not None
+                else "RCCL fallback")
+        print(f"# decode collectives: {mode} (world={world})",
+              file=sys.stderr)
+
     rng = np.random.default_rng(0)
     prompt = rng.integers(0, cfg.vocab_size, size=args.prompt_len)
 
