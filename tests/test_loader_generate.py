@@ -549,3 +549,22 @@ def test_fast_path_eos_list_mid_chunk():
         assert r.finish_reason == "stop"
     finally:
         cfg.eos_token_id = None
+
+
+def test_logit_bias_is_exact_odds_multiplier():
+    """Under softmax, adding bias b to token j multiplies its odds
+    against any other token by exactly e^b (the defining property of
+    logit_bias); other tokens' relative odds are untouched."""
+    from llm_np_cp_amd.runtime.sampling import filter_probs
+
+    rng = np.random.default_rng(8)
+    logits = rng.standard_normal(32).astype(np.float32)
+    base = filter_probs(logits, SamplingParams(strategy="temperature"))
+    b = 1.7
+    biased = filter_probs(logits, SamplingParams(
+        strategy="temperature", logit_bias={5: b}))
+    np.testing.assert_allclose(
+        (biased[5] / biased[11]) / (base[5] / base[11]), np.exp(b),
+        rtol=1e-5)
+    np.testing.assert_allclose(biased[7] / biased[11],
+                               base[7] / base[11], rtol=1e-6)
