@@ -91,3 +91,22 @@ def test_world1_collectives_are_noops():
     out = torch.zeros(4)
     tp.all_gather_into(out, torch.arange(4.0))
     np.testing.assert_array_equal(out.numpy(), np.arange(4.0))
+
+
+def test_xgmi_fits_predicate():
+    """Collective routing predicate (decode-sized -> one-shot, else
+    RCCL): contiguity, 16-byte granularity, slot capacity, dtype."""
+    from types import SimpleNamespace
+
+    from llm_np_cp_amd.parallel.xgmi import XgmiComm
+
+    d = SimpleNamespace(slot_bytes=1024)
+    fits = XgmiComm.fits
+    assert fits(d, torch.zeros(512, dtype=torch.bfloat16))   # == slot
+    assert fits(d, torch.zeros(256, dtype=torch.float32))
+    assert not fits(d, torch.zeros(600, dtype=torch.float32))   # > slot
+    assert not fits(d, torch.zeros(520, dtype=torch.bfloat16))  # > slot
+    assert not fits(d, torch.zeros(12, dtype=torch.bfloat16))   # % 16
+    assert not fits(d, torch.zeros(8, dtype=torch.int32))       # dtype
+    assert not fits(d, torch.zeros(8, 8,
+                                   dtype=torch.bfloat16)[:, :4])  # strided
