@@ -5,11 +5,18 @@ A from-scratch re-design of the capabilities of ``githubpradeep/llm_np_cp``
 ``generate()`` with per-token streaming, KV cache, min-p sampling) built
 MI355X-first: hand-written HIP/CDNA4 (gfx950) kernels for the whole
 forward pass, PyTorch-ROCm tensors as memory containers, hipGraph-captured
-decode, tensor parallelism over RCCL/xGMI, fp8 weight path.
+decode, tensor parallelism over RCCL/xGMI, fp8/MXFP4 weight paths.
 
-Public API (reference parity):
-    load_model(name_or_dir) -> (tokenizer, model, config)
-    generate(prompt, tokenizer, model, ...) -> GenerateResult (streams)
+Families: Llama-3.x, Gemma-2, Qwen-2.5, Mistral, Mixtral sparse MoE.
+
+Public API (reference parity + extensions; see docs/MIGRATING.md):
+    load_model(name_or_dir, ..., lora=) -> (tokenizer, model, config)
+    generate(prompt, tokenizer, model, ..., stop=, logprobs=)
+        -> GenerateResult (streams)
+    generate_speculative(prompt, tok, draft, target, ...) — greedy or
+        stochastic speculative sampling
+    ChatSession(tok, model) — multi-turn KV reuse
+    runtime.server — OpenAI-style continuous-batching HTTP server
 """
 
 from .core.config import ModelConfig, preset_config, PRESETS
