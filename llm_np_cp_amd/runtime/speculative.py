@@ -128,9 +128,11 @@ def generate_speculative(prompt: str, tokenizer, draft, target,
     rng = np.random.default_rng(params.seed)
 
     def _pick(logits_row: np.ndarray) -> int:
-        if params.strategy == "greedy":
-            return int(np.argmax(logits_row))
+        # route greedy through filter_probs too so logit_bias applies
+        # identically here and in the accept/reject distributions
         p = filter_probs(logits_row, params)
+        if params.strategy == "greedy":
+            return int(np.argmax(p))
         return int(rng.choice(len(p), p=p))
     prompt_ids = [int(t) for t in tokenizer.encode(prompt)]
     P0 = len(prompt_ids)

@@ -218,3 +218,26 @@ def test_spec_stochastic_same_models_high_acceptance():
                                k=4, stop_on_eos=False, params=params)
     s = res.spec_stats
     assert s["accepted"] == s["proposed"] > 0
+
+
+def test_spec_logit_bias_applies_in_both_modes():
+    """logit_bias flows through filter_probs into drafting, verify AND
+    the plain-step/first-token picks — banning the greedy chain's first
+    token changes the output; forcing one token pins the stream."""
+    tok, target, _ = _load(seed=0)
+    _, draft, _ = _load(seed=0)
+    base = generate_speculative(PROMPT, tok, draft, target,
+                                max_tokens=8, k=3, stop_on_eos=False)
+    banned = int(base.token_ids[0])
+    res = generate_speculative(
+        PROMPT, tok, draft, target, max_tokens=8, k=3, stop_on_eos=False,
+        params=L.SamplingParams(strategy="greedy",
+                                logit_bias={banned: -1e4}))
+    assert banned not in res.token_ids
+    forced = 13
+    res2 = generate_speculative(
+        PROMPT, tok, draft, target, max_tokens=6, k=3, stop_on_eos=False,
+        params=L.SamplingParams(strategy="min_p", seed=5,
+                                logit_bias={forced: 1e4}))
+    assert res2.token_ids == [forced] * 6
+    assert res2.spec_stats["accepted"] == res2.spec_stats["proposed"]
