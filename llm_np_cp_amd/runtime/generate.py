@@ -135,7 +135,13 @@ def generate(prompt: str, tokenizer, model, max_tokens: int = 200,
     ``logprobs``: when an int N >= 0, GenerateResult.logprobs carries
     per-token raw-model logprobs with the top-N alternatives (takes the
     logits-visible step loop — the device-side sampler never surfaces
-    the full distribution)."""
+    the full distribution).
+
+    Seeds: the GPU fast path's device sampler derives its stream from
+    the ENGINE seed + a device counter (baked into the captured graph),
+    so ``params.seed`` is honored on the logits-visible paths (CPU
+    oracle; logprobs / logit_bias / top-k / top-p requests) but not by
+    the in-graph device sampler."""
     params = params or SamplingParams()
     rng = np.random.default_rng(params.seed)
     prompt_ids = list(tokenizer.encode(prompt))
