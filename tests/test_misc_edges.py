@@ -117,3 +117,28 @@ def test_timing_and_tracing_utils():
     with trace_range("unit-test"):   # no roctx in CI -> must no-op
         x = 1
     assert x == 1
+
+
+def test_gpu_model_fails_loudly_without_gpu():
+    """The HIP engine must never silently fall back on CPU-only hosts
+    (driver requirement: no eager/CPU fallback masquerading as the
+    native path)."""
+    import torch
+    if torch.cuda.is_available():
+        pytest.skip("host has a GPU")
+    from llm_np_cp_amd.models.engine import GPUModel
+
+    cfg = preset_config("tiny-llama")
+    with pytest.raises(RuntimeError, match="requires a GPU"):
+        GPUModel(cfg, {})
+
+
+def test_hip_ops_fail_loudly_without_library(monkeypatch):
+    """ops.lib() refuses to run without the built extension (no silent
+    eager fallback)."""
+    from llm_np_cp_amd.ops import hip_ops
+
+    monkeypatch.setattr(hip_ops, "_LIB", None)
+    monkeypatch.setattr(hip_ops, "_LIB_PATH", "/nonexistent/_lib.so")
+    with pytest.raises(RuntimeError, match="not built"):
+        hip_ops.lib()
