@@ -3,7 +3,6 @@ stop strings on the device loop).  Kept in a file that sorts AFTER the
 core engine/kernel suites so an unexpected failure here cannot mask
 them under pytest -x."""
 
-import numpy as np
 import pytest
 
 from tests.test_gpu_engine import make_pair
