@@ -568,3 +568,28 @@ def test_logit_bias_is_exact_odds_multiplier():
         rtol=1e-5)
     np.testing.assert_allclose(biased[7] / biased[11],
                                base[7] / base[11], rtol=1e-6)
+
+
+@pytest.mark.parametrize("preset", ["tiny-gemma2", "tiny-mistral"])
+def test_synthetic_checkpoint_roundtrip_remaining_families(tmp_path,
+                                                          preset):
+    """Checkpoint-dir round trip for the families not covered by the
+    dedicated llama/qwen/mixtral tests: written safetensors + config
+    reload to BIT-identical oracle logits (incl. Gemma gamma-storage
+    and Mistral window fields surviving to_hf_dict/from_json)."""
+    from llm_np_cp_amd.core.config import preset_config
+    from llm_np_cp_amd.io.loader import (random_weights,
+                                         write_synthetic_checkpoint)
+    from llm_np_cp_amd.models.numpy_ref import NumpyKVCache, NumpyModel
+
+    d = str(tmp_path / preset)
+    write_synthetic_checkpoint(d, preset, seed=3)
+    tok, model, cfg = L.load_model(d, backend="numpy")
+    assert cfg.model_type == preset_config(preset).model_type
+
+    ref_cfg = preset_config(preset)
+    ref = NumpyModel(ref_cfg, random_weights(ref_cfg, seed=3))
+    ids = np.arange(1, 7)
+    a = model.forward(ids, NumpyKVCache(cfg, 16), 0)
+    b = ref.forward(ids, NumpyKVCache(ref_cfg, 16), 0)
+    np.testing.assert_array_equal(a, b)  # fp32 survives the round trip
