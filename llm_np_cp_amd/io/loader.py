@@ -70,6 +70,66 @@ def load_weights_numpy(model_dir: str) -> Dict[str, np.ndarray]:
     return w
 
 
+class LazyCheckpointWeights:
+    """Mapping that loads each HF-named tensor ON DEMAND from the
+    checkpoint's safetensors (fp32) and holds no reference afterward.
+
+    The GPU upload path converts each tensor to bf16/fp8 as it lands in
+    HBM, so an eager fp32 host dict is pure peak-memory waste — ~280 GB
+    of host RAM for the Llama-3.1-70B preset that the 288 GB MI355X
+    otherwise fits comfortably.  With this mapping host memory stays at
+    one-tensor peak.  ``shapes`` is read from the safetensors headers
+    (no tensor data), so ``validate_weights`` stays O(metadata).
+
+    Read-only: the NumPy oracle path (which mutates the dict for
+    lm_head tying and computes from host memory) keeps the eager
+    loader."""
+
+    def __init__(self, model_dir: str):
+        from safetensors import safe_open
+
+        self.model_dir = model_dir
+        index = os.path.join(model_dir, "model.safetensors.index.json")
+        self._shard_of: Dict[str, str] = {}
+        if os.path.exists(index):
+            with open(index) as f:
+                self._shard_of = dict(json.load(f)["weight_map"])
+        else:
+            single = os.path.join(model_dir, "model.safetensors")
+            if not os.path.exists(single):
+                raise FileNotFoundError(
+                    f"no model.safetensors[.index.json] in {model_dir}")
+            with safe_open(single, framework="pt") as f:
+                for name in f.keys():
+                    self._shard_of[name] = "model.safetensors"
+        self.shapes: Dict[str, tuple] = {}
+        for shard in sorted(set(self._shard_of.values())):
+            with safe_open(os.path.join(model_dir, shard),
+                           framework="pt") as f:
+                for name in f.keys():
+                    self.shapes[name] = tuple(
+                        f.get_slice(name).get_shape())
+
+    def keys(self):
+        return self._shard_of.keys()
+
+    def __contains__(self, name):
+        return name in self._shard_of
+
+    def __getitem__(self, name) -> np.ndarray:
+        import torch
+        from safetensors import safe_open
+
+        if name not in self._shard_of:
+            raise KeyError(name)
+        path = os.path.join(self.model_dir, self._shard_of[name])
+        with safe_open(path, framework="pt") as f:
+            return f.get_tensor(name).to(torch.float32).numpy()
+
+    def get(self, name, default=None):
+        return self[name] if name in self._shard_of else default
+
+
 # ----------------------------------------------------------------------
 # Synthetic checkpoints (random init — no network, BASELINE.json terms)
 # ----------------------------------------------------------------------

@@ -313,10 +313,10 @@ def load_model(model_dir_or_preset: str, backend: str = "auto",
     from ..core.config import preset_config, PRESETS
     from ..io.loader import load_config, load_weights_numpy, random_weights
 
+    weights_dir = None
     if os.path.isdir(model_dir_or_preset):
         config = load_config(model_dir_or_preset)
-        weights = load_weights_numpy(model_dir_or_preset)
-        tok_dir = model_dir_or_preset
+        weights_dir = tok_dir = model_dir_or_preset
     elif model_dir_or_preset.lower() in PRESETS:
         config = preset_config(model_dir_or_preset)
         weights = random_weights(config, seed=seed)
@@ -324,12 +324,30 @@ def load_model(model_dir_or_preset: str, backend: str = "auto",
     elif "/" in model_dir_or_preset:
         local = _hub_download(model_dir_or_preset)
         config = load_config(local)
-        weights = load_weights_numpy(local)
-        tok_dir = local
+        weights_dir = tok_dir = local
     else:
         raise FileNotFoundError(
             f"{model_dir_or_preset!r} is neither a directory, a preset "
             f"({sorted(PRESETS)}), nor an HF repo id")
+
+    if backend == "auto":
+        try:
+            import torch
+            backend = "gpu" if torch.cuda.is_available() else "numpy"
+        except ImportError:
+            backend = "numpy"
+
+    if weights_dir is not None:
+        if backend == "gpu" and lora is None:
+            # one-tensor host-memory peak: the upload converts each
+            # tensor to bf16/fp8 on arrival, so an eager fp32 dict
+            # (~280 GB host RAM for 70B) is pure waste
+            from ..io.loader import LazyCheckpointWeights
+            weights = LazyCheckpointWeights(weights_dir)
+        else:
+            # the oracle computes from (and mutates) the host dict;
+            # LoRA merge also needs a writable dict
+            weights = load_weights_numpy(weights_dir)
 
     if lora is not None:
         from ..io.loader import apply_lora
@@ -351,13 +369,6 @@ def load_model(model_dir_or_preset: str, backend: str = "auto",
             print(f"# no usable tokenizer files in {tok_dir}; using the "
                   f"byte-level fallback tokenizer", file=sys.stderr)
         tokenizer = ByteTokenizer()
-
-    if backend == "auto":
-        try:
-            import torch
-            backend = "gpu" if torch.cuda.is_available() else "numpy"
-        except ImportError:
-            backend = "numpy"
 
     if backend == "numpy":
         from ..models.numpy_ref import NumpyModel, NumpyKVCache
