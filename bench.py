@@ -2,9 +2,12 @@
 """Flagship decode benchmark (driver contract).
 
 Measures the BASELINE.json metric: decode tokens/sec, batch 1, synthetic
-prompt, random-init weights — Llama-3.2-1B TP=1 (N=1) or Gemma-2-9B TP=N
-(N>1), bf16.  A "step" is one decoded token; the timed region is K
-hipGraph-replayed decode steps (full forward incl. lm_head + sampling).
+prompt, random-init weights, bf16.  Default model is Llama-3.2-1B at
+EVERY N (TP=N) so the driver's 1/2/4/8-GPU curve compares like with
+like; the metric's second config (Gemma-2-9B TP=8, fp8) runs via
+``--model gemma-2-9b --dtype fp8``.  A "step" is one decoded token; the
+timed region is K hipGraph-replayed decode steps (full forward incl.
+lm_head + sampling).
 
 Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--model M]
 Launched by the driver for N>1 via torch.distributed.run (one rank/GPU).
