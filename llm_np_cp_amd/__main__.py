@@ -13,7 +13,11 @@ import time
 
 
 def main(argv=None):
+    import llm_np_cp_amd as _pkg
+
     ap = argparse.ArgumentParser(prog="llm_np_cp_amd")
+    ap.add_argument("--version", action="version",
+                    version=f"llm_np_cp_amd {_pkg.__version__}")
     ap.add_argument("prompt", nargs="?", default="Once upon a time")
     ap.add_argument("--model", default="llama-3.2-1b",
                     help="checkpoint directory or preset name")
